@@ -1,0 +1,41 @@
+"""Bias + dropout + residual-add (reference fused_bias_dropout.py:49-62).
+
+torch's dropout on ROCm is already a single HIP kernel; the fusion win is
+folding bias-add and the residual add around it, which torch fuses poorly.
+We keep the torch composition (3 HBM-bound kernels) on CPU and use it on
+GPU too until profiling shows it on the critical path — the MLP/attention
+epilogues already defer their bias here (skip_bias_add)."""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+
+def _bias_dropout_add_func(x_with_bias, residual, prob, training):
+    x, bias = x_with_bias
+    if bias is not None:
+        x = x + bias
+    out = torch.nn.functional.dropout(x, p=prob, training=training)
+    return residual + out
+
+
+def bias_dropout_add_unfused(training):
+    def _fn(x_with_bias, residual, prob):
+        return _bias_dropout_add_func(x_with_bias, residual, prob, training)
+    return _fn
+
+
+def bias_dropout_add_fused_train(x_with_bias, residual, prob):
+    return _bias_dropout_add_func(x_with_bias, residual, prob, True)
+
+
+def bias_dropout_add_fused_inference(x_with_bias, residual, prob):
+    return _bias_dropout_add_func(x_with_bias, residual, prob, False)
+
+
+def get_bias_dropout_add(training, fused):
+    if fused:
+        return bias_dropout_add_fused_train if training else bias_dropout_add_fused_inference
+    return bias_dropout_add_unfused(training)

@@ -1,0 +1,124 @@
+"""Fused LayerNorm / RMSNorm (HIP kernels, reference fused_layer_norm.py).
+
+GPU path: one-pass Welford/−sum-of-squares HIP kernels (ops/csrc/norms.hip)
+vectorised bf16x8, one workgroup per row group.  CPU path: fp32 torch
+reference (the numerics oracle the GPU tests compare against).
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+
+from ... import ops as _ops
+
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        if x.is_cuda:
+            y, invrms = _ops.get_ops().rmsnorm_fwd(x, weight, eps)
+        else:
+            xf = x.float()
+            invrms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+            y = (xf * invrms * weight.float()).to(x.dtype)
+            invrms = invrms.squeeze(-1)
+        ctx.save_for_backward(x, weight, invrms)
+        ctx.eps = eps
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, invrms = ctx.saved_tensors
+        if dy.is_cuda:
+            dx, dw = _ops.get_ops().rmsnorm_bwd(dy.contiguous(), x, weight, invrms)
+            return dx, dw.to(weight.dtype), None
+        xf = x.float()
+        dyf = dy.float()
+        wf = weight.float()
+        r = invrms.unsqueeze(-1)
+        xhat = xf * r
+        H = x.shape[-1]
+        dxhat = dyf * wf
+        dx = r * (dxhat - xhat * (dxhat * xhat).mean(-1, keepdim=True))
+        dw = (dyf * xhat).reshape(-1, H).sum(0)
+        return dx.to(x.dtype), dw.to(weight.dtype), None
+
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        if x.is_cuda:
+            y, mean, invstd = _ops.get_ops().layernorm_fwd(x, weight, bias, eps)
+        else:
+            xf = x.float()
+            mean = xf.mean(-1)
+            var = xf.var(-1, unbiased=False)
+            invstd = torch.rsqrt(var + eps)
+            xhat = (xf - mean.unsqueeze(-1)) * invstd.unsqueeze(-1)
+            y = (xhat * weight.float() + bias.float()).to(x.dtype)
+        ctx.save_for_backward(x, weight, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, invstd = ctx.saved_tensors
+        if dy.is_cuda:
+            dx, dw, db = _ops.get_ops().layernorm_bwd(
+                dy.contiguous(), x, weight, mean, invstd)
+            return dx, dw.to(weight.dtype), db.to(weight.dtype), None
+        xf = x.float()
+        dyf = dy.float()
+        wf = weight.float()
+        H = x.shape[-1]
+        xhat = (xf - mean.unsqueeze(-1)) * invstd.unsqueeze(-1)
+        dxhat = dyf * wf
+        dx = invstd.unsqueeze(-1) * (
+            dxhat - dxhat.mean(-1, keepdim=True)
+            - xhat * (dxhat * xhat).mean(-1, keepdim=True))
+        dw = (dyf * xhat).reshape(-1, H).sum(0)
+        db = dyf.reshape(-1, H).sum(0)
+        return dx.to(x.dtype), dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+class FusedLayerNorm(nn.Module):
+    """LayerNorm over the hidden dim; weights marked sequence_parallel so
+    finalize_model_grads all-reduces their grads across TP when SP is on."""
+
+    def __init__(self, config, hidden_size: int, eps: float = 1e-5, **kwargs):
+        super().__init__()
+        self.config = config
+        self.hidden_size = hidden_size
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(hidden_size, dtype=config.params_dtype))
+        self.bias = nn.Parameter(torch.zeros(hidden_size, dtype=config.params_dtype))
+        if config.sequence_parallel:
+            self.weight.sequence_parallel = True
+            self.bias.sequence_parallel = True
+
+    def forward(self, x):
+        return _LayerNormFn.apply(x, self.weight, self.bias, self.eps)
+
+
+class FusedRMSNorm(nn.Module):
+    def __init__(self, config, hidden_size: int, eps: float = 1e-5, **kwargs):
+        super().__init__()
+        self.config = config
+        self.hidden_size = hidden_size
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(hidden_size, dtype=config.params_dtype))
+        if config.sequence_parallel:
+            self.weight.sequence_parallel = True
+
+    def forward(self, x):
+        return _RMSNormFn.apply(x, self.weight, self.eps)
+
+
+def get_norm_cls(normalization: str):
+    if normalization == "RMSNorm":
+        return FusedRMSNorm
+    return FusedLayerNorm
+
+
+class WrappedTorchNorm(FusedLayerNorm):
+    """Alias kept for reference-API parity (core/transformer/torch_norm.py)."""

@@ -1,0 +1,115 @@
+"""Fused scale + mask + softmax (reference fused_softmax.py:12-96 +
+FusedScaleMaskSoftmax:97).
+
+GPU path: wave64 HIP kernels (ops/csrc/softmax.hip) — causal
+(upper-triangular) and generic-mask variants, fp32 accumulation in
+registers/LDS, bf16/fp16 I/O.  Used by the "fused" attention backend;
+the flash backend fuses softmax into the attention kernel itself.
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+
+from ...core.enums import AttnMaskType
+from ... import ops as _ops
+
+
+class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, inputs, scale):
+        if inputs.is_cuda:
+            probs = _ops.get_ops().scaled_upper_triang_masked_softmax_fwd(
+                inputs, scale)
+        else:
+            b, sq, sk = inputs.shape
+            x = inputs.float() * scale
+            mask = torch.triu(torch.ones(sq, sk, dtype=torch.bool,
+                                         device=inputs.device), diagonal=1)
+            x = x.masked_fill(mask, float("-inf"))
+            probs = torch.softmax(x, dim=-1).to(inputs.dtype)
+        ctx.save_for_backward(probs)
+        ctx.scale = scale
+        return probs
+
+    @staticmethod
+    def backward(ctx, dy):
+        (probs,) = ctx.saved_tensors
+        if dy.is_cuda:
+            dx = _ops.get_ops().scaled_softmax_bwd(dy.contiguous(), probs,
+                                                   ctx.scale)
+        else:
+            dyf = dy.float()
+            p = probs.float()
+            dx = (p * (dyf - (dyf * p).sum(-1, keepdim=True)) * ctx.scale).to(dy.dtype)
+        return dx, None
+
+
+class ScaledMaskedSoftmax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, inputs, mask, scale):
+        if inputs.is_cuda:
+            probs = _ops.get_ops().scaled_masked_softmax_fwd(inputs, mask, scale)
+        else:
+            x = inputs.float() * scale
+            if mask is not None:
+                x = x.masked_fill(mask, -10000.0)
+            probs = torch.softmax(x, dim=-1).to(inputs.dtype)
+        ctx.save_for_backward(probs)
+        ctx.scale = scale
+        return probs
+
+    @staticmethod
+    def backward(ctx, dy):
+        (probs,) = ctx.saved_tensors
+        if dy.is_cuda:
+            dx = _ops.get_ops().scaled_softmax_bwd(dy.contiguous(), probs, ctx.scale)
+        else:
+            dyf = dy.float()
+            p = probs.float()
+            dx = (p * (dyf - (dyf * p).sum(-1, keepdim=True)) * ctx.scale).to(dy.dtype)
+        return dx, None, None
+
+
+class FusedScaleMaskSoftmax(nn.Module):
+    """Dispatcher mirroring reference FusedScaleMaskSoftmax:97."""
+
+    def __init__(self, input_in_fp16, input_in_bf16, attn_mask_type,
+                 scaled_masked_softmax_fusion, mask_func, softmax_in_fp32,
+                 scale):
+        super().__init__()
+        self.input_in_float16 = input_in_fp16 or input_in_bf16
+        self.attn_mask_type = attn_mask_type
+        self.fusion = scaled_masked_softmax_fusion
+        self.mask_func = mask_func
+        self.softmax_in_fp32 = softmax_in_fp32
+        self.scale = scale
+
+    def forward(self, input, mask):
+        # input: [b, np, sq, sk]
+        scale = self.scale if self.scale is not None else 1.0
+        if self.fusion and self.input_in_float16:
+            b, np_, sq, sk = input.shape
+            if self.attn_mask_type == AttnMaskType.causal and sq == sk:
+                probs = ScaledUpperTriangMaskedSoftmax.apply(
+                    input.view(-1, sq, sk), scale)
+                return probs.view(b, np_, sq, sk)
+            return ScaledMaskedSoftmax.apply(input, mask, scale)
+        # unfused fallback
+        orig_dtype = input.dtype
+        if self.input_in_float16 and self.softmax_in_fp32:
+            input = input.float()
+        if self.scale is not None:
+            input = input * self.scale
+        if self.attn_mask_type == AttnMaskType.causal and mask is None:
+            sq, sk = input.shape[-2], input.shape[-1]
+            mask = torch.triu(torch.ones(sq, sk, dtype=torch.bool,
+                                         device=input.device), diagonal=1)
+            mask = mask.view(1, 1, sq, sk)
+        if mask is not None:
+            input = self.mask_func(input, mask)
+        probs = torch.softmax(input, dim=-1)
+        if probs.dtype != orig_dtype:
+            probs = probs.to(orig_dtype)
+        return probs

@@ -1,0 +1,261 @@
+"""Buffer-aligned mixed-precision Adam, optionally ZeRO-1 sharded.
+
+MI355X-first redesign of the reference's two optimizer classes
+(Float16OptimizerWithFloat16Params optimizer.py:504 and
+DistributedOptimizer distrib_optimizer.py:80): instead of per-param fp32
+master copies and apex multi_tensor_apply over thousands of tensors, the
+optimizer state lives in FLAT fp32 buffers exactly aligned with the DDP
+grad/param buffers, so one fused HIP Adam kernel call per buffer updates
+everything (ops/csrc/adam.hip), and the ZeRO-1 shard is just a slice of
+the same flat space:
+
+  grad buffer (fp32)  --reduce-scatter-->  local shard grads
+  master (fp32 flat)  --adam kernel   -->  master shard updated
+  param buffer (bf16) <--cast shard   ---  then all-gather params
+
+With dp==1 the "shard" is the whole buffer and the collectives are no-ops,
+giving one unified code path.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from .. import parallel_state
+from ..tensor_parallel.layers import param_is_not_tensor_parallel_duplicate
+from ..trace_hooks import trace_scope
+from ... import ops as _ops
+from .clip_grads import clip_grad_by_total_norm_fp32, count_zeros_fp32, get_grad_norm_fp32
+from .optimizer_config import OptimizerConfig
+
+
+def _adam_step_flat(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
+                    v: torch.Tensor, lr: float, beta1: float, beta2: float,
+                    eps: float, weight_decay: float, step: int):
+    """One Adam(W) step over flat fp32 tensors; HIP kernel on GPU."""
+    if master.is_cuda and _ops.have_ops() and hasattr(_ops.get_ops(), "adamw_flat"):
+        _ops.get_ops().adamw_flat(master, grad, m, v, lr, beta1, beta2, eps,
+                                  weight_decay, step)
+        return
+    bias_correction1 = 1 - beta1 ** step
+    bias_correction2 = 1 - beta2 ** step
+    if weight_decay != 0:
+        master.mul_(1 - lr * weight_decay)
+    m.mul_(beta1).add_(grad, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
+    denom = (v / bias_correction2).sqrt_().add_(eps)
+    master.addcdiv_(m, denom, value=-lr / bias_correction1)
+
+
+class ParamGroup:
+    """A (wd_mult, lr_mult) bucket of shard ranges within one buffer."""
+
+    def __init__(self, wd_mult: float, lr_mult: float):
+        self.wd_mult = wd_mult
+        self.lr_mult = lr_mult
+        self.ranges: List[tuple] = []  # (buffer_idx, start, end) in shard coords
+
+
+class DistributedOptimizer:
+    """Mixed-precision Adam over DDP buffers; ZeRO-1 when the DDP config
+    enabled use_distributed_optimizer, full-replica otherwise."""
+
+    def __init__(self, config: OptimizerConfig, model_chunks: List):
+        self.config = config
+        self.model_chunks = model_chunks
+        self.buffers = []
+        for chunk in model_chunks:
+            self.buffers.extend(chunk.buffers)
+        self.sharded = any(b.ddp_config.use_distributed_optimizer
+                           for b in self.buffers)
+        self.step_count = 0
+        self.lr = config.lr or 0.0
+        self.weight_decay = config.weight_decay
+
+        # per-buffer shard state
+        self.shard_master: List[torch.Tensor] = []
+        self.shard_m: List[torch.Tensor] = []
+        self.shard_v: List[torch.Tensor] = []
+        self.shard_bounds: List[tuple] = []
+        # per-(buffer,param) metadata for grad-norm / wd grouping
+        self.no_wd_ranges: List[List[tuple]] = []  # per buffer: shard-coord ranges w/o wd
+        self.norm_ranges: List[List[tuple]] = []   # per buffer: ranges counted in grad norm
+
+        for buf in self.buffers:
+            if buf.ddp_config.use_distributed_optimizer:
+                lo, hi = buf.local_shard_bounds()
+            else:
+                lo, hi = 0, buf.numel
+            self.shard_bounds.append((lo, hi))
+            if buf.param_data is not None:
+                master = buf.param_data[lo:hi].float().clone()
+            else:
+                master = torch.empty(hi - lo, dtype=torch.float32,
+                                     device=buf.grad_data.device)
+                # gather initial values from the individual param tensors
+                for p, (s, e) in buf.param_index_map.items():
+                    os_, oe = max(s, lo), min(e, hi)
+                    if os_ < oe:
+                        flat = p.data.reshape(-1)
+                        master[os_ - lo:oe - lo].copy_(
+                            flat[os_ - s:oe - s].float())
+            self.shard_master.append(master)
+            self.shard_m.append(torch.zeros_like(master))
+            self.shard_v.append(torch.zeros_like(master))
+
+            no_wd, norm_r = [], []
+            for p, (s, e) in buf.param_index_map.items():
+                os_, oe = max(s, lo), min(e, hi)
+                if os_ >= oe:
+                    continue
+                r = (os_ - lo, oe - lo)
+                if p.dim() == 1 or getattr(p, "_no_weight_decay", False):
+                    no_wd.append(r)
+                if param_is_not_tensor_parallel_duplicate(p):
+                    norm_r.append(r)
+            self.no_wd_ranges.append(no_wd)
+            self.norm_ranges.append(norm_r)
+
+        # interface compat: param_groups for LR schedulers
+        self.param_groups = [
+            {"lr": self.lr, "wd_mult": 1.0, "lr_mult": 1.0, "is_decoupled_lr": False,
+             "params": [], "weight_decay": config.weight_decay},
+        ]
+
+    # --------------------------------------------------------------
+    def zero_grad(self, set_to_none: bool = True):
+        for chunk in self.model_chunks:
+            chunk.zero_grad_buffer()
+
+    def get_loss_scale(self) -> torch.Tensor:
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+        return torch.ones(1, dtype=torch.float32, device=device)
+
+    def scale_loss(self, loss):
+        return loss
+
+    def _shard_grad(self, i):
+        lo, hi = self.shard_bounds[i]
+        return self.buffers[i].grad_data[lo:hi]
+
+    def get_grad_norm(self) -> float:
+        grads_for_norm = []
+        for i, buf in enumerate(self.buffers):
+            g = self._shard_grad(i)
+            for (s, e) in self.norm_ranges[i]:
+                grads_for_norm.append(g[s:e])
+        extra = []
+        if self.sharded:
+            extra.append(parallel_state.get_data_parallel_group(
+                with_context_parallel=True))
+        return get_grad_norm_fp32(
+            grads_for_norm,
+            model_parallel_group=parallel_state.get_model_parallel_group(),
+            extra_groups=extra)
+
+    def count_zeros(self) -> float:
+        grads = [self._shard_grad(i) for i in range(len(self.buffers))]
+        return count_zeros_fp32(grads, parallel_state.get_model_parallel_group())
+
+    @torch.no_grad()
+    def step(self):
+        self.step_count += 1
+        lr = self.param_groups[0]["lr"]
+        wd = self.param_groups[0].get("weight_decay", self.weight_decay)
+
+        grad_norm = None
+        if self.config.clip_grad > 0:
+            grad_norm = self.get_grad_norm()
+            if not math.isfinite(grad_norm):
+                # skip update on inf/nan grad norm
+                return False, grad_norm, None
+            all_grads = [self._shard_grad(i) for i in range(len(self.buffers))]
+            clip_grad_by_total_norm_fp32(all_grads, self.config.clip_grad,
+                                         grad_norm)
+        num_zeros = self.count_zeros() if self.config.log_num_zeros_in_grad else None
+
+        with trace_scope("optimizer"):
+            for i, buf in enumerate(self.buffers):
+                grad = self._shard_grad(i)
+                master, m, v = self.shard_master[i], self.shard_m[i], self.shard_v[i]
+                no_wd = self.no_wd_ranges[i]
+                if wd == 0 or not no_wd:
+                    _adam_step_flat(master, grad, m, v, lr,
+                                    self.config.adam_beta1, self.config.adam_beta2,
+                                    self.config.adam_eps, wd, self.step_count)
+                else:
+                    # two-pass: run with wd over the whole shard is wrong for
+                    # no-wd params, so stitch: wd pass on full shard minus
+                    # no-wd ranges is complex — instead run per-range.
+                    cursor = 0
+                    events = sorted(no_wd)
+                    for (s, e) in events + [(master.numel(), master.numel())]:
+                        if cursor < s:
+                            _adam_step_flat(master[cursor:s], grad[cursor:s],
+                                            m[cursor:s], v[cursor:s], lr,
+                                            self.config.adam_beta1,
+                                            self.config.adam_beta2,
+                                            self.config.adam_eps, wd,
+                                            self.step_count)
+                        if s < e:
+                            _adam_step_flat(master[s:e], grad[s:e], m[s:e],
+                                            v[s:e], lr, self.config.adam_beta1,
+                                            self.config.adam_beta2,
+                                            self.config.adam_eps, 0.0,
+                                            self.step_count)
+                        cursor = max(cursor, e)
+
+                # cast master back into model params
+                lo, hi = self.shard_bounds[i]
+                if buf.param_data is not None:
+                    buf.param_data[lo:hi].copy_(master)
+                else:
+                    for p, (s, e) in buf.param_index_map.items():
+                        os_, oe = max(s, lo), min(e, hi)
+                        if os_ < oe:
+                            p.data.reshape(-1)[os_ - s:oe - s].copy_(
+                                master[os_ - lo:oe - lo])
+
+            # ZeRO-1: all-gather updated params
+            if self.sharded:
+                for buf in self.buffers:
+                    if buf.ddp_config.use_distributed_optimizer:
+                        buf.start_param_sync(async_op=False)
+
+        return True, grad_norm, num_zeros
+
+    # --- checkpointing ---------------------------------------------
+    def state_dict(self):
+        return {
+            "step": self.step_count,
+            "lr": self.param_groups[0]["lr"],
+            "shard_master": [t.cpu() for t in self.shard_master],
+            "shard_m": [t.cpu() for t in self.shard_m],
+            "shard_v": [t.cpu() for t in self.shard_v],
+        }
+
+    def load_state_dict(self, sd):
+        self.step_count = sd["step"]
+        self.param_groups[0]["lr"] = sd["lr"]
+        for dst, src in zip(self.shard_master, sd["shard_master"]):
+            dst.copy_(src.to(dst.device))
+        for dst, src in zip(self.shard_m, sd["shard_m"]):
+            dst.copy_(src.to(dst.device))
+        for dst, src in zip(self.shard_v, sd["shard_v"]):
+            dst.copy_(src.to(dst.device))
+
+    def reload_model_params(self):
+        for i, buf in enumerate(self.buffers):
+            lo, hi = self.shard_bounds[i]
+            if buf.param_data is not None:
+                self.shard_master[i].copy_(buf.param_data[lo:hi].float())
+            else:
+                for p, (s, e) in buf.param_index_map.items():
+                    os_, oe = max(s, lo), min(e, hi)
+                    if os_ < oe:
+                        self.shard_master[i][os_ - lo:oe - lo].copy_(
+                            p.data.reshape(-1)[os_ - s:oe - s].float())

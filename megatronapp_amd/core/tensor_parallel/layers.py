@@ -1,0 +1,331 @@
+"""Tensor-parallel linear layers + vocab-parallel embedding.
+
+Reference: tensor_parallel/layers.py (ColumnParallelLinear:675,
+RowParallelLinear:1019, VocabParallelEmbedding:172,
+LinearWithGradAccumulationAndAsyncCommunication:404-560).
+
+MI355X design notes:
+* GEMMs go through torch.matmul -> hipBLASLt (bf16 MFMA).  The framework's
+  hand-written HIP kernels cover the *fused* ops (norms, activations,
+  attention) — plain projection GEMMs are hipBLASLt's job.
+* The backward overlaps the TP grad all-reduce (latency-bound over xGMI)
+  with the weight-gradient GEMM on the compute stream: the all-reduce is
+  issued first on the communication side via async_op=True, the wgrad GEMM
+  fills the gap (reference behaviour of :404).
+* Gradient accumulation fuses into the DDP-owned fp32 ``main_grad`` buffer.
+"""
+
+from __future__ import annotations
+
+import warnings
+from typing import Callable, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+from torch.nn.parameter import Parameter
+
+from .. import parallel_state
+from .mappings import (
+    copy_to_tensor_model_parallel_region,
+    gather_from_tensor_model_parallel_region,
+    reduce_from_tensor_model_parallel_region,
+    reduce_scatter_to_sequence_parallel_region,
+    scatter_to_tensor_model_parallel_region,
+    _gather_along_first_dim,
+    _reduce_scatter_along_first_dim,
+)
+from .random import get_cuda_rng_tracker
+from .utils import VocabUtility, divide
+
+_MODEL_PARALLEL_ATTRIBUTE_DEFAULTS = {
+    "tensor_model_parallel": False,
+    "partition_dim": -1,
+    "partition_stride": 1,
+}
+
+
+def param_is_not_tensor_parallel_duplicate(param):
+    return getattr(param, "tensor_model_parallel", False) or (
+        parallel_state.get_tensor_model_parallel_rank() == 0)
+
+
+def set_tensor_model_parallel_attributes(tensor, is_parallel, dim, stride=1):
+    tensor.tensor_model_parallel = is_parallel
+    tensor.partition_dim = dim
+    tensor.partition_stride = stride
+
+
+def set_defaults_if_not_set_tensor_model_parallel_attributes(tensor):
+    for attr, default in _MODEL_PARALLEL_ATTRIBUTE_DEFAULTS.items():
+        if not hasattr(tensor, attr):
+            setattr(tensor, attr, default)
+
+
+def copy_tensor_model_parallel_attributes(dst, src):
+    for attr in _MODEL_PARALLEL_ATTRIBUTE_DEFAULTS:
+        if hasattr(src, attr):
+            setattr(dst, attr, getattr(src, attr))
+
+
+def _initialize_affine_weight(weight, init_method, partition_dim, stride=1,
+                              expert_parallel=False):
+    set_tensor_model_parallel_attributes(weight, True, partition_dim, stride)
+    if torch.cuda.is_available() and weight.is_cuda:
+        with get_cuda_rng_tracker().fork():
+            init_method(weight)
+    else:
+        init_method(weight)
+
+
+class VocabParallelEmbedding(torch.nn.Module):
+    """Embedding sharded along the vocab dimension (reference layers.py:172).
+
+    Forward masks out-of-shard tokens, local lookup, then one TP
+    all-reduce (or reduce-scatter when sequence-parallel) merges shards.
+    """
+
+    def __init__(self, num_embeddings, embedding_dim, *, init_method,
+                 config, reduce_scatter_embeddings: bool = False):
+        super().__init__()
+        self.num_embeddings = num_embeddings
+        self.embedding_dim = embedding_dim
+        self.reduce_scatter_embeddings = reduce_scatter_embeddings
+        self.tensor_model_parallel_size = parallel_state.get_tensor_model_parallel_world_size()
+        (self.vocab_start_index, self.vocab_end_index) = (
+            VocabUtility.vocab_range_from_global_vocab_size(
+                num_embeddings, parallel_state.get_tensor_model_parallel_rank(),
+                self.tensor_model_parallel_size))
+        self.num_embeddings_per_partition = (
+            self.vocab_end_index - self.vocab_start_index)
+        self.weight = Parameter(torch.empty(
+            self.num_embeddings_per_partition, embedding_dim,
+            dtype=config.params_dtype))
+        _initialize_affine_weight(self.weight, init_method, partition_dim=0)
+
+    def forward(self, input_):
+        if self.tensor_model_parallel_size > 1:
+            input_mask = (input_ < self.vocab_start_index) | (
+                input_ >= self.vocab_end_index)
+            masked_input = input_.clone() - self.vocab_start_index
+            masked_input[input_mask] = 0
+        else:
+            masked_input = input_
+        output_parallel = F.embedding(masked_input, self.weight)
+        if self.tensor_model_parallel_size > 1:
+            output_parallel[input_mask, :] = 0.0
+        if self.reduce_scatter_embeddings:
+            # [b, s, h] -> [s, b, h] -> [s/tp, b, h]
+            output_parallel = output_parallel.transpose(0, 1).contiguous()
+            output = reduce_scatter_to_sequence_parallel_region(output_parallel)
+        else:
+            output = reduce_from_tensor_model_parallel_region(output_parallel)
+        return output
+
+
+class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
+    """Fused linear core (reference layers.py:404-560).
+
+    forward:  (SP? all-gather input) ; out = x @ W^T (+b)
+    backward: dgrad GEMM -> issue async TP all-reduce / SP reduce-scatter of
+              dgrad -> wgrad GEMM overlaps the collective -> wait.
+              wgrad accumulates straight into param.main_grad when the DDP
+              grad buffer owns one (gradient_accumulation_fusion).
+    """
+
+    @staticmethod
+    def forward(ctx, input, weight, bias, gradient_accumulation_fusion,
+                async_grad_allreduce, sequence_parallel):
+        ctx.use_bias = bias is not None
+        ctx.gradient_accumulation_fusion = gradient_accumulation_fusion
+        ctx.async_grad_allreduce = async_grad_allreduce
+        ctx.sequence_parallel = sequence_parallel
+
+        if sequence_parallel:
+            total_input = _gather_along_first_dim(input)
+        else:
+            total_input = input
+        ctx.save_for_backward(input, weight)
+        output = torch.matmul(total_input, weight.t())
+        if bias is not None:
+            output = output + bias
+        return output
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        input, weight = ctx.saved_tensors
+        use_bias = ctx.use_bias
+
+        if ctx.sequence_parallel:
+            total_input = _gather_along_first_dim(input)
+        else:
+            total_input = input
+
+        grad_input = grad_output.matmul(weight)
+
+        handle = None
+        if ctx.sequence_parallel:
+            assert not ctx.async_grad_allreduce
+            sub_grad_input = torch.empty(
+                input.shape, dtype=input.dtype, device=input.device)
+            handle = dist.reduce_scatter_tensor(
+                sub_grad_input, grad_input.contiguous(),
+                group=parallel_state.get_tensor_model_parallel_group(),
+                async_op=True)
+        elif ctx.async_grad_allreduce:
+            handle = dist.all_reduce(
+                grad_input, group=parallel_state.get_tensor_model_parallel_group(),
+                async_op=True)
+
+        # wgrad GEMM overlaps the collective above
+        grad_output_2d = grad_output.reshape(-1, grad_output.shape[-1])
+        total_input_2d = total_input.reshape(-1, total_input.shape[-1])
+        if ctx.gradient_accumulation_fusion and hasattr(weight, "main_grad"):
+            weight.main_grad.add_(
+                torch.matmul(grad_output_2d.t(), total_input_2d))
+            grad_weight = None
+            weight.grad_added_to_main_grad = True
+        else:
+            grad_weight = grad_output_2d.t().matmul(total_input_2d)
+        grad_bias = grad_output_2d.sum(dim=0) if use_bias else None
+
+        if handle is not None:
+            handle.wait()
+        if ctx.sequence_parallel:
+            grad_input = sub_grad_input
+
+        return grad_input, grad_weight, grad_bias, None, None, None
+
+
+def linear_with_grad_accumulation_and_async_allreduce(
+        input, weight, bias, gradient_accumulation_fusion,
+        async_grad_allreduce, sequence_parallel):
+    return LinearWithGradAccumulationAndAsyncCommunication.apply(
+        input, weight, bias, gradient_accumulation_fusion,
+        async_grad_allreduce, sequence_parallel)
+
+
+class ColumnParallelLinear(torch.nn.Module):
+    """Y = XA^T with A sharded along its output dim (reference :675)."""
+
+    def __init__(self, input_size, output_size, *, config, init_method,
+                 bias=True, gather_output=False, stride=1,
+                 keep_master_weight_for_test=False, skip_bias_add=False,
+                 skip_weight_param_allocation=False, is_expert=False,
+                 tp_comm_buffer_name=None):
+        super().__init__()
+        self.input_size = input_size
+        self.output_size = output_size
+        self.gather_output = gather_output
+        self.skip_bias_add = skip_bias_add
+        self.config = config
+        self.is_expert = is_expert
+        world_size = (1 if is_expert else
+                      parallel_state.get_tensor_model_parallel_world_size())
+        self.output_size_per_partition = divide(output_size, world_size)
+
+        if not skip_weight_param_allocation:
+            self.weight = Parameter(torch.empty(
+                self.output_size_per_partition, input_size,
+                dtype=config.params_dtype))
+            _initialize_affine_weight(self.weight, init_method, partition_dim=0,
+                                      stride=stride)
+            setattr(self.weight, "allreduce", not is_expert)
+        else:
+            self.weight = None
+
+        if bias:
+            self.bias = Parameter(torch.zeros(
+                self.output_size_per_partition, dtype=config.params_dtype))
+            set_tensor_model_parallel_attributes(self.bias, True, 0, stride)
+            setattr(self.bias, "allreduce", not is_expert)
+        else:
+            self.register_parameter("bias", None)
+
+        self.sequence_parallel = config.sequence_parallel and not is_expert
+        self.async_tensor_model_parallel_allreduce = (
+            config.async_tensor_model_parallel_allreduce
+            and world_size > 1 and not self.sequence_parallel)
+        self.gradient_accumulation_fusion = config.gradient_accumulation_fusion
+
+    def forward(self, input_, weight=None):
+        weight = weight if weight is not None else self.weight
+        bias = None if self.skip_bias_add else self.bias
+
+        if (self.async_tensor_model_parallel_allreduce or
+                self.sequence_parallel or self.is_expert):
+            input_parallel = input_
+        else:
+            input_parallel = copy_to_tensor_model_parallel_region(input_)
+
+        output_parallel = linear_with_grad_accumulation_and_async_allreduce(
+            input_parallel, weight, bias,
+            self.gradient_accumulation_fusion,
+            self.async_tensor_model_parallel_allreduce,
+            self.sequence_parallel)
+        if self.gather_output:
+            output = gather_from_tensor_model_parallel_region(output_parallel)
+        else:
+            output = output_parallel
+        output_bias = self.bias if self.skip_bias_add else None
+        return output, output_bias
+
+
+class RowParallelLinear(torch.nn.Module):
+    """Y = XA^T with A sharded along its input dim (reference :1019).
+
+    Forward ends with the TP all-reduce (or SP reduce-scatter); bias is
+    added after the reduction (skip_bias_add defers it to the caller for
+    fusion into the following kernel)."""
+
+    def __init__(self, input_size, output_size, *, config, init_method,
+                 bias=True, input_is_parallel=True, stride=1,
+                 keep_master_weight_for_test=False, skip_bias_add=False,
+                 is_expert=False, tp_comm_buffer_name=None):
+        super().__init__()
+        self.input_size = input_size
+        self.output_size = output_size
+        self.input_is_parallel = input_is_parallel
+        self.skip_bias_add = skip_bias_add
+        self.config = config
+        self.is_expert = is_expert
+        world_size = (1 if is_expert else
+                      parallel_state.get_tensor_model_parallel_world_size())
+        self.input_size_per_partition = divide(input_size, world_size)
+
+        self.weight = Parameter(torch.empty(
+            output_size, self.input_size_per_partition,
+            dtype=config.params_dtype))
+        _initialize_affine_weight(self.weight, init_method, partition_dim=1,
+                                  stride=stride)
+        setattr(self.weight, "allreduce", not is_expert)
+        if bias:
+            self.bias = Parameter(torch.zeros(output_size, dtype=config.params_dtype))
+            setattr(self.bias, "allreduce", not is_expert)
+            setattr(self.bias, "sequence_parallel", config.sequence_parallel)
+        else:
+            self.register_parameter("bias", None)
+        self.sequence_parallel = config.sequence_parallel and not is_expert
+        self.gradient_accumulation_fusion = config.gradient_accumulation_fusion
+
+    def forward(self, input_):
+        if self.input_is_parallel or self.is_expert:
+            input_parallel = input_
+        else:
+            input_parallel = scatter_to_tensor_model_parallel_region(input_)
+
+        output_parallel = linear_with_grad_accumulation_and_async_allreduce(
+            input_parallel, self.weight, None,
+            self.gradient_accumulation_fusion, False, False)
+        if self.is_expert:
+            output = output_parallel
+        elif self.sequence_parallel:
+            output = reduce_scatter_to_sequence_parallel_region(output_parallel)
+        else:
+            output = reduce_from_tensor_model_parallel_region(output_parallel)
+        if not self.skip_bias_add:
+            output = output + self.bias if self.bias is not None else output
+            output_bias = None
+        else:
+            output_bias = self.bias
+        return output, output_bias

@@ -1,0 +1,147 @@
+"""Self-attention block: fused QKV projection, GQA, RoPE, KV cache.
+
+Reference: transformer/attention.py (Attention:88, SelfAttention:845).
+MegaScope taps (QKV / raw scores / context) hook via core.tensor_tracer.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional, Union
+
+import torch
+
+from .. import parallel_state
+from ..enums import AttnMaskType
+from ..models.common.embeddings.rotary_pos_embedding import apply_rotary_pos_emb
+from ..tensor_parallel.layers import ColumnParallelLinear, RowParallelLinear
+from ..tensor_parallel.utils import divide
+from ..transformer_config import TransformerConfig
+from ..trace_hooks import trace_scope
+from ..tensor_tracer import get_tensor_tracers, FlagType
+from .module import MegatronModule
+from .spec_utils import ModuleSpec, build_module
+
+
+@dataclass
+class SelfAttentionSubmodules:
+    linear_qkv: Union[ModuleSpec, type] = None
+    core_attention: Union[ModuleSpec, type] = None
+    linear_proj: Union[ModuleSpec, type] = None
+    q_layernorm: Union[ModuleSpec, type] = None
+    k_layernorm: Union[ModuleSpec, type] = None
+
+
+class SelfAttention(MegatronModule):
+    def __init__(self, config: TransformerConfig,
+                 submodules: SelfAttentionSubmodules, layer_number: int,
+                 attn_mask_type=AttnMaskType.causal, cp_comm_type: str = None):
+        super().__init__(config)
+        self.layer_number = layer_number
+        self.attn_mask_type = attn_mask_type
+
+        world_size = parallel_state.get_tensor_model_parallel_world_size()
+        self.hidden_size_per_attention_head = config.kv_channels
+        self.num_attention_heads_per_partition = divide(
+            config.num_attention_heads, world_size)
+        self.num_query_groups_per_partition = divide(
+            config.num_query_groups, world_size)
+        self.query_projection_size = config.kv_channels * config.num_attention_heads
+        self.kv_projection_size = config.kv_channels * config.num_query_groups
+
+        self.linear_qkv = build_module(
+            submodules.linear_qkv, config.hidden_size,
+            self.query_projection_size + 2 * self.kv_projection_size,
+            config=config, init_method=config.init_method,
+            bias=config.add_bias_linear or config.add_qkv_bias,
+            skip_bias_add=False, gather_output=False)
+
+        self.core_attention = build_module(
+            submodules.core_attention, config=config,
+            layer_number=layer_number, attn_mask_type=attn_mask_type,
+            attention_type="self", cp_comm_type=cp_comm_type)
+
+        self.linear_proj = build_module(
+            submodules.linear_proj, self.query_projection_size,
+            config.hidden_size, config=config,
+            init_method=config.output_layer_init_method,
+            bias=config.add_bias_linear, input_is_parallel=True,
+            skip_bias_add=True)
+
+        if submodules.q_layernorm is not None and config.qk_layernorm:
+            self.q_layernorm = build_module(
+                submodules.q_layernorm, config=config,
+                hidden_size=self.hidden_size_per_attention_head,
+                eps=config.layernorm_epsilon)
+        else:
+            self.q_layernorm = None
+        if submodules.k_layernorm is not None and config.qk_layernorm:
+            self.k_layernorm = build_module(
+                submodules.k_layernorm, config=config,
+                hidden_size=self.hidden_size_per_attention_head,
+                eps=config.layernorm_epsilon)
+        else:
+            self.k_layernorm = None
+
+    def _split_qkv(self, mixed_qkv):
+        """[sq, b, (np/g + 2) * g * hn] -> q [sq,b,np,hn], k/v [sq,b,ng,hn]."""
+        sq, b, _ = mixed_qkv.shape
+        ng = self.num_query_groups_per_partition
+        np_ = self.num_attention_heads_per_partition
+        hn = self.hidden_size_per_attention_head
+        mixed_qkv = mixed_qkv.view(sq, b, ng, (np_ // ng + 2) * hn)
+        q, k, v = torch.split(
+            mixed_qkv, [(np_ // ng) * hn, hn, hn], dim=3)
+        q = q.reshape(sq, b, np_, hn)
+        return q.contiguous(), k.contiguous(), v.contiguous()
+
+    def forward(self, hidden_states, attention_mask=None, key_value_states=None,
+                inference_context=None, rotary_pos_emb=None, rotary_pos_cos=None,
+                rotary_pos_sin=None, attention_bias=None, packed_seq_params=None,
+                sequence_len_offset=None):
+        # hidden_states: [sq, b, h]
+        mixed_qkv, _ = self.linear_qkv(hidden_states)
+        query, key, value = self._split_qkv(mixed_qkv)
+
+        if self.q_layernorm is not None:
+            query = self.q_layernorm(query)
+        if self.k_layernorm is not None:
+            key = self.k_layernorm(key)
+
+        # MegaScope QKV tap (reference attention.py:979-981)
+        tt = get_tensor_tracers()
+        if tt is not None and tt.enabled(FlagType.QKV, self.layer_number):
+            tt.report(FlagType.QKV, self.layer_number, (query, key, value))
+
+        if rotary_pos_emb is not None:
+            if isinstance(rotary_pos_emb, tuple):
+                q_pos_emb, k_pos_emb = rotary_pos_emb
+            else:
+                q_pos_emb = k_pos_emb = rotary_pos_emb
+            if inference_context is not None:
+                offset = inference_context.sequence_len_offset
+                q_pos_emb = q_pos_emb[offset:offset + query.shape[0]]
+            query = apply_rotary_pos_emb(query, q_pos_emb, config=self.config)
+            key = apply_rotary_pos_emb(
+                key, k_pos_emb if inference_context is None else
+                k_pos_emb[:inference_context.sequence_len_offset + key.shape[0]][-key.shape[0]:],
+                config=self.config)
+
+        attn_mask_type = self.attn_mask_type
+        if inference_context is not None:
+            key, value = inference_context.update_kv_cache(
+                self.layer_number, key, value)
+            if query.shape[0] == 1:
+                attn_mask_type = AttnMaskType.no_mask
+
+        with trace_scope("attention"):
+            core_attn_out = self.core_attention(
+                query, key, value, attention_mask=attention_mask,
+                attn_mask_type=attn_mask_type, attention_bias=attention_bias,
+                packed_seq_params=packed_seq_params)
+
+        if tt is not None and tt.enabled(FlagType.ContextLayer, self.layer_number):
+            tt.report(FlagType.ContextLayer, self.layer_number, core_attn_out)
+
+        output, bias = self.linear_proj(core_attn_out)
+        return output, bias
