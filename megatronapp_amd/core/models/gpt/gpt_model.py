@@ -93,8 +93,11 @@ class GPTModel(LanguageModule):
 
         rotary_pos_emb = None
         if self.position_embedding_type == "rope":
+            rope_input = decoder_input
+            if rope_input is None:
+                rope_input = self.decoder.input_tensor  # mid-pipeline stage
             rotary_seq_len = self.rotary_pos_emb.get_rotary_seq_len(
-                inference_context, self.decoder, decoder_input, self.config,
+                inference_context, self.decoder, rope_input, self.config,
                 packed_seq_params)
             rotary_pos_emb = self.rotary_pos_emb(rotary_seq_len)
 

@@ -25,6 +25,14 @@ import torch.distributed as dist
 
 _TENSOR_MODEL_PARALLEL_GROUP = None
 _PIPELINE_MODEL_PARALLEL_GROUP = None
+# Direction-split duplicates of the PP group: activations (toward next)
+# ride the FWD communicator, gradients (toward prev) the BWD one.  At
+# PP=2 next==prev, so without the split forward and backward traffic
+# between one rank pair aliases onto one channel and isend/irecv order
+# matching can pair an activation send with a gradient recv (the
+# reference's dual-group trick, p2p_communication.py:202-216).
+_PIPELINE_FWD_GROUP = None
+_PIPELINE_BWD_GROUP = None
 _MODEL_PARALLEL_GROUP = None
 _DATA_PARALLEL_GROUP = None
 _DATA_PARALLEL_GROUP_GLOO = None
@@ -180,10 +188,15 @@ def initialize_model_parallel(
             _TENSOR_MODEL_PARALLEL_GROUP = group
             _TENSOR_MODEL_PARALLEL_GLOBAL_RANKS = ranks
 
+    global _PIPELINE_FWD_GROUP, _PIPELINE_BWD_GROUP
     for ranks in gen.get_ranks("pp"):
         group = _new_group(ranks)
+        fwd_group = _new_group(ranks)
+        bwd_group = _new_group(ranks)
         if rank in ranks:
             _PIPELINE_MODEL_PARALLEL_GROUP = group
+            _PIPELINE_FWD_GROUP = fwd_group
+            _PIPELINE_BWD_GROUP = bwd_group
             _PIPELINE_GLOBAL_RANKS = ranks
         # Embedding group: first and last stage of each pipeline (tied
         # word embeddings grad all-reduce, finalize_model_grads.py:120).
@@ -290,6 +303,14 @@ def get_tensor_model_parallel_group(check_initialized=True):
 def get_pipeline_model_parallel_group():
     assert _PIPELINE_MODEL_PARALLEL_GROUP is not None
     return _PIPELINE_MODEL_PARALLEL_GROUP
+
+
+def get_pipeline_forward_group():
+    return _PIPELINE_FWD_GROUP or _PIPELINE_MODEL_PARALLEL_GROUP
+
+
+def get_pipeline_backward_group():
+    return _PIPELINE_BWD_GROUP or _PIPELINE_MODEL_PARALLEL_GROUP
 
 
 def get_model_parallel_group():

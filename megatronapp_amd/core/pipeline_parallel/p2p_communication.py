@@ -63,7 +63,7 @@ def _communicate_shapes(tensor_send_next, tensor_send_prev, recv_prev, recv_next
 
 
 def _p2p_ops_even_odd(tensor_send_prev, tensor_recv_prev, tensor_send_next,
-                      tensor_recv_next, group):
+                      tensor_recv_next, fwd_group, bwd_group):
     """Per-op isend/irecv with even/odd rank ordering (deadlock-free without
     batching; reference _p2p_ops :191-216)."""
     reqs = []
@@ -74,19 +74,19 @@ def _p2p_ops_even_odd(tensor_send_prev, tensor_recv_prev, tensor_send_next,
 
     def send_next():
         if tensor_send_next is not None:
-            reqs.append(dist.isend(tensor_send_next, next_rank, group=group))
+            reqs.append(dist.isend(tensor_send_next, next_rank, group=fwd_group))
 
     def recv_prev():
         if tensor_recv_prev is not None:
-            reqs.append(dist.irecv(tensor_recv_prev, prev_rank, group=group))
+            reqs.append(dist.irecv(tensor_recv_prev, prev_rank, group=fwd_group))
 
     def send_prev():
         if tensor_send_prev is not None:
-            reqs.append(dist.isend(tensor_send_prev, prev_rank, group=group))
+            reqs.append(dist.isend(tensor_send_prev, prev_rank, group=bwd_group))
 
     def recv_next():
         if tensor_recv_next is not None:
-            reqs.append(dist.irecv(tensor_recv_next, next_rank, group=group))
+            reqs.append(dist.irecv(tensor_recv_next, next_rank, group=bwd_group))
 
     if even:
         send_next(); recv_prev(); send_prev(); recv_next()
@@ -118,32 +118,41 @@ def _communicate(*, tensor_send_next: Optional[torch.Tensor],
         tensor_recv_next = torch.empty(recv_next_shape, device=device,
                                        dtype=dtype, requires_grad=True)
 
-    group = parallel_state.get_pipeline_model_parallel_group()
+    # Direction-split channels (see parallel_state): activation traffic
+    # (send→next / recv←prev) on the FWD group, gradient traffic
+    # (send→prev / recv←next) on the BWD group — mandatory at PP=2 where
+    # next == prev would otherwise alias both flows onto one channel.
+    fwd_group = parallel_state.get_pipeline_forward_group()
+    bwd_group = parallel_state.get_pipeline_backward_group()
     if config.batch_p2p_comm:
-        ops = []
+        fwd_ops, bwd_ops = [], []
         if tensor_send_prev is not None:
-            ops.append(dist.P2POp(dist.isend, tensor_send_prev.contiguous(),
-                                  parallel_state.get_pipeline_model_parallel_prev_rank(),
-                                  group))
+            bwd_ops.append(dist.P2POp(dist.isend, tensor_send_prev.contiguous(),
+                                      parallel_state.get_pipeline_model_parallel_prev_rank(),
+                                      bwd_group))
         if tensor_recv_prev is not None:
-            ops.append(dist.P2POp(dist.irecv, tensor_recv_prev,
-                                  parallel_state.get_pipeline_model_parallel_prev_rank(),
-                                  group))
+            fwd_ops.append(dist.P2POp(dist.irecv, tensor_recv_prev,
+                                      parallel_state.get_pipeline_model_parallel_prev_rank(),
+                                      fwd_group))
         if tensor_send_next is not None:
-            ops.append(dist.P2POp(dist.isend, tensor_send_next.contiguous(),
-                                  parallel_state.get_pipeline_model_parallel_next_rank(),
-                                  group))
+            fwd_ops.append(dist.P2POp(dist.isend, tensor_send_next.contiguous(),
+                                      parallel_state.get_pipeline_model_parallel_next_rank(),
+                                      fwd_group))
         if tensor_recv_next is not None:
-            ops.append(dist.P2POp(dist.irecv, tensor_recv_next,
-                                  parallel_state.get_pipeline_model_parallel_next_rank(),
-                                  group))
-        reqs = dist.batch_isend_irecv(ops) if ops else []
+            bwd_ops.append(dist.P2POp(dist.irecv, tensor_recv_next,
+                                      parallel_state.get_pipeline_model_parallel_next_rank(),
+                                      bwd_group))
+        reqs = []
+        if fwd_ops:
+            reqs.extend(dist.batch_isend_irecv(fwd_ops))
+        if bwd_ops:
+            reqs.extend(dist.batch_isend_irecv(bwd_ops))
     else:
         reqs = _p2p_ops_even_odd(
             tensor_send_prev.contiguous() if tensor_send_prev is not None else None,
             tensor_recv_prev,
             tensor_send_next.contiguous() if tensor_send_next is not None else None,
-            tensor_recv_next, group)
+            tensor_recv_next, fwd_group, bwd_group)
 
     if wait_on_reqs and reqs:
         for req in reqs:
