@@ -1,0 +1,257 @@
+// PyTorch bindings for the megatronapp_amd CDNA4 kernels.
+// Built in-tree to megatronapp_amd/ops/_C.so (see ops/setup.py).
+
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+
+// launchers (norms.hip / elementwise.hip / rope.hip / softmax.hip / adam.hip /
+// attention.hip)
+void launch_rmsnorm_fwd(const void*, const void*, void*, float*, int, int,
+                        float, hipStream_t);
+void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
+                        void*, float*, int, int, hipStream_t);
+void launch_layernorm_fwd(const void*, const void*, const void*, void*, float*,
+                          float*, int, int, float, hipStream_t);
+void launch_layernorm_bwd(const void*, const void*, const void*, const float*,
+                          const float*, void*, float*, float*, int, int,
+                          hipStream_t);
+void launch_bias_gelu_fwd(const void*, const void*, void*, long, int,
+                          hipStream_t);
+void launch_bias_gelu_bwd(const void*, const void*, const void*, void*, long,
+                          int, hipStream_t);
+void launch_bias_swiglu_fwd(const void*, const void*, void*, long, int,
+                            hipStream_t);
+void launch_bias_swiglu_bwd(const void*, const void*, const void*, void*, long,
+                            int, hipStream_t);
+void launch_rope(const void*, const float*, const float*, void*, long, int,
+                 int, bool, hipStream_t);
+void launch_softmax_causal_fwd(const void*, void*, long, int, int, float,
+                               hipStream_t);
+void launch_softmax_masked_fwd(const void*, const void*, void*, long, int, int,
+                               int, float, hipStream_t);
+void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
+                        hipStream_t);
+void launch_adamw_flat(float*, const float*, float*, float*, long, float,
+                       float, float, float, float, int, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// ------------------------------------------------------------------- norms
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const int H = (int)x.size(-1);
+  const long N = x.numel() / H;
+  auto y = torch::empty_like(x);
+  auto invrms = torch::empty({N}, x.options().dtype(torch::kFloat32));
+  launch_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                     invrms.data_ptr<float>(), (int)N, H, (float)eps,
+                     cur_stream());
+  return {y, invrms};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor invrms) {
+  check_bf16(dy, "dy");
+  check_bf16(x, "x");
+  const int H = (int)x.size(-1);
+  const long N = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     invrms.data_ptr<float>(), dx.data_ptr(),
+                     dw.data_ptr<float>(), (int)N, H, cur_stream());
+  return {dx, dw};
+}
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps) {
+  check_bf16(x, "x");
+  const int H = (int)x.size(-1);
+  const long N = x.numel() / H;
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({N}, x.options().dtype(torch::kFloat32));
+  auto invstd = torch::empty({N}, x.options().dtype(torch::kFloat32));
+  launch_layernorm_fwd(x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       (int)N, H, (float)eps, cur_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor invstd) {
+  check_bf16(dy, "dy");
+  const int H = (int)x.size(-1);
+  const long N = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto db = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  launch_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       dx.data_ptr(), dw.data_ptr<float>(),
+                       db.data_ptr<float>(), (int)N, H, cur_stream());
+  return {dx, dw, db};
+}
+
+// -------------------------------------------------------------- elementwise
+torch::Tensor bias_gelu_fwd(torch::Tensor x, c10::optional<torch::Tensor> bias) {
+  check_bf16(x, "x");
+  auto y = torch::empty_like(x);
+  const int F = (int)x.size(-1);
+  launch_bias_gelu_fwd(x.data_ptr(),
+                       bias.has_value() ? bias->data_ptr() : nullptr,
+                       y.data_ptr(), x.numel(), F, cur_stream());
+  return y;
+}
+
+torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                            c10::optional<torch::Tensor> bias) {
+  check_bf16(dy, "dy");
+  auto dx = torch::empty_like(x);
+  const int F = (int)x.size(-1);
+  launch_bias_gelu_bwd(dy.data_ptr(), x.data_ptr(),
+                       bias.has_value() ? bias->data_ptr() : nullptr,
+                       dx.data_ptr(), x.numel(), F, cur_stream());
+  return dx;
+}
+
+torch::Tensor bias_swiglu_fwd(torch::Tensor x, c10::optional<torch::Tensor> bias) {
+  check_bf16(x, "x");
+  const int F2 = (int)x.size(-1);
+  TORCH_CHECK(F2 % 2 == 0, "last dim must be even (gated)");
+  const int F = F2 / 2;
+  const long N = x.numel() / F2;
+  auto sizes = x.sizes().vec();
+  sizes.back() = F;
+  auto y = torch::empty(sizes, x.options());
+  launch_bias_swiglu_fwd(x.data_ptr(),
+                         bias.has_value() ? bias->data_ptr() : nullptr,
+                         y.data_ptr(), N, F, cur_stream());
+  return y;
+}
+
+torch::Tensor bias_swiglu_bwd(torch::Tensor dy, torch::Tensor x,
+                              c10::optional<torch::Tensor> bias) {
+  check_bf16(dy, "dy");
+  const int F2 = (int)x.size(-1);
+  const int F = F2 / 2;
+  const long N = x.numel() / F2;
+  auto dx = torch::empty_like(x);
+  launch_bias_swiglu_bwd(dy.data_ptr(), x.data_ptr(),
+                         bias.has_value() ? bias->data_ptr() : nullptr,
+                         dx.data_ptr(), N, F, cur_stream());
+  return dx;
+}
+
+// --------------------------------------------------------------------- rope
+torch::Tensor rope_apply(torch::Tensor t, torch::Tensor cs, torch::Tensor sn,
+                         bool bwd) {
+  check_bf16(t, "t");
+  TORCH_CHECK(cs.scalar_type() == torch::kFloat32, "cos table must be fp32");
+  const int d = (int)t.size(-1);
+  const long rows = t.numel() / d;
+  // cs/sn come in as [s, 1, 1, d]; rows per seq position = b*nh
+  const long s = cs.numel() / d;
+  const int bnh = (int)(rows / s);
+  auto out = torch::empty_like(t);
+  launch_rope(t.data_ptr(), cs.data_ptr<float>(), sn.data_ptr<float>(),
+              out.data_ptr(), rows, bnh, d, bwd, cur_stream());
+  return out;
+}
+
+torch::Tensor rope_fwd(torch::Tensor t, torch::Tensor cs, torch::Tensor sn) {
+  return rope_apply(t, cs.contiguous(), sn.contiguous(), false);
+}
+
+torch::Tensor rope_bwd(torch::Tensor dy, torch::Tensor cs, torch::Tensor sn) {
+  return rope_apply(dy, cs.contiguous(), sn.contiguous(), true);
+}
+
+// ------------------------------------------------------------------ softmax
+torch::Tensor scaled_upper_triang_masked_softmax_fwd(torch::Tensor x,
+                                                     double scale) {
+  check_bf16(x, "x");
+  const int sk = (int)x.size(-1);
+  const int sq = (int)x.size(-2);
+  const long rows = x.numel() / sk;
+  auto y = torch::empty_like(x);
+  launch_softmax_causal_fwd(x.data_ptr(), y.data_ptr(), rows, sq, sk,
+                            (float)scale, cur_stream());
+  return y;
+}
+
+torch::Tensor scaled_masked_softmax_fwd(torch::Tensor x,
+                                        c10::optional<torch::Tensor> mask,
+                                        double scale) {
+  check_bf16(x, "x");
+  // x: [b, np, sq, sk]
+  const int sk = (int)x.size(-1);
+  const int sq = (int)x.size(-2);
+  const int np = (int)x.size(1);
+  const long rows = x.numel() / sk;
+  auto y = torch::empty_like(x);
+  const void* mptr = nullptr;
+  torch::Tensor m8;
+  if (mask.has_value()) {
+    m8 = mask->to(torch::kUInt8).contiguous();
+    mptr = m8.data_ptr();
+  }
+  launch_softmax_masked_fwd(x.data_ptr(), mptr, y.data_ptr(), rows, np, sq, sk,
+                            (float)scale, cur_stream());
+  return y;
+}
+
+torch::Tensor scaled_softmax_bwd(torch::Tensor dy, torch::Tensor p,
+                                 double scale) {
+  check_bf16(dy, "dy");
+  const int sk = (int)dy.size(-1);
+  const long rows = dy.numel() / sk;
+  auto dx = torch::empty_like(dy);
+  launch_softmax_bwd(dy.data_ptr(), p.data_ptr(), dx.data_ptr(), rows, sk,
+                     (float)scale, cur_stream());
+  return dx;
+}
+
+// --------------------------------------------------------------------- adam
+void adamw_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, double lr, double beta1, double beta2,
+                double eps, double wd, long step) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
+  launch_adamw_flat(p.data_ptr<float>(), g.data_ptr<float>(),
+                    m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
+                    (float)lr, (float)beta1, (float)beta2, (float)eps,
+                    (float)wd, (int)step, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("rmsnorm_fwd", &rmsnorm_fwd);
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd);
+  mod.def("layernorm_fwd", &layernorm_fwd);
+  mod.def("layernorm_bwd", &layernorm_bwd);
+  mod.def("bias_gelu_fwd", &bias_gelu_fwd);
+  mod.def("bias_gelu_bwd", &bias_gelu_bwd);
+  mod.def("bias_swiglu_fwd", &bias_swiglu_fwd);
+  mod.def("bias_swiglu_bwd", &bias_swiglu_bwd);
+  mod.def("rope_fwd", &rope_fwd);
+  mod.def("rope_bwd", &rope_bwd);
+  mod.def("scaled_upper_triang_masked_softmax_fwd",
+          &scaled_upper_triang_masked_softmax_fwd);
+  mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
+  mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
+  mod.def("adamw_flat", &adamw_flat);
+}

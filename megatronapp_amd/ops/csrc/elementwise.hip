@@ -1,0 +1,177 @@
+// Fused bias + activation elementwise kernels (bf16, HBM-bound).
+// Replaces reference torch-JIT fusions (fused_bias_gelu.py,
+// fused_bias_swiglu.py — SURVEY.md §2.5).
+
+#include "common.h"
+
+#include <stdexcept>
+
+#define BLOCK 256
+#define VEC 8
+
+__device__ __forceinline__ float gelu_tanh(float x) {
+  float x3 = x * x * x;
+  return 0.5f * x * (1.f + tanhf(0.7978845608028654f * (x + 0.044715f * x3)));
+}
+
+__device__ __forceinline__ float gelu_tanh_grad(float x) {
+  float x2 = x * x;
+  float t = tanhf(0.7978845608028654f * (x + 0.044715f * x2 * x));
+  return 0.5f * (1.f + t) +
+         0.5f * x * (1.f - t * t) * 0.7978845608028654f * (1.f + 3.f * 0.044715f * x2);
+}
+
+// --------------------------------------------------------------- bias + gelu
+__global__ void bias_gelu_fwd_kernel(const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ bias,
+                                     unsigned short* __restrict__ y, long n,
+                                     int F) {
+  long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC;
+  const long stride = (long)gridDim.x * BLOCK * VEC;
+  for (; i < n; i += stride) {
+    short8v v = *(const short8v*)(x + i);
+    short8v o;
+    if (bias != nullptr) {
+      short8v bv = *(const short8v*)(bias + (i % F));
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o[j] = (short)f2bf(gelu_tanh(bf2f((unsigned short)v[j]) +
+                                     bf2f((unsigned short)bv[j])));
+    } else {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o[j] = (short)f2bf(gelu_tanh(bf2f((unsigned short)v[j])));
+    }
+    *(short8v*)(y + i) = o;
+  }
+}
+
+__global__ void bias_gelu_bwd_kernel(const unsigned short* __restrict__ dy,
+                                     const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ bias,
+                                     unsigned short* __restrict__ dx, long n,
+                                     int F) {
+  long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC;
+  const long stride = (long)gridDim.x * BLOCK * VEC;
+  for (; i < n; i += stride) {
+    short8v v = *(const short8v*)(x + i);
+    short8v d = *(const short8v*)(dy + i);
+    short8v o;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float xf = bf2f((unsigned short)v[j]);
+      if (bias != nullptr) xf += bf2f((unsigned short)bias[(i + j) % F]);
+      o[j] = (short)f2bf(bf2f((unsigned short)d[j]) * gelu_tanh_grad(xf));
+    }
+    *(short8v*)(dx + i) = o;
+  }
+}
+
+// ------------------------------------------------------------- bias + swiglu
+// x: [N, 2F] (x1 | x2 halves), y: [N, F] = silu(x1+b1) * (x2+b2)
+__global__ void bias_swiglu_fwd_kernel(const unsigned short* __restrict__ x,
+                                       const unsigned short* __restrict__ bias,
+                                       unsigned short* __restrict__ y, long N,
+                                       int F) {
+  long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
+  const long total = N * (F / VEC);
+  const long stride = (long)gridDim.x * BLOCK;
+  for (; idx < total; idx += stride) {
+    const long row = idx / (F / VEC);
+    const int col = (int)(idx % (F / VEC)) * VEC;
+    const unsigned short* x1 = x + row * 2L * F + col;
+    const unsigned short* x2 = x1 + F;
+    short8v v1 = *(const short8v*)x1;
+    short8v v2 = *(const short8v*)x2;
+    short8v o;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float a = bf2f((unsigned short)v1[j]);
+      float b = bf2f((unsigned short)v2[j]);
+      if (bias != nullptr) {
+        a += bf2f((unsigned short)bias[col + j]);
+        b += bf2f((unsigned short)bias[F + col + j]);
+      }
+      float sig = 1.f / (1.f + __expf(-a));
+      o[j] = (short)f2bf(a * sig * b);
+    }
+    *(short8v*)(y + row * F + col) = o;
+  }
+}
+
+__global__ void bias_swiglu_bwd_kernel(const unsigned short* __restrict__ dy,
+                                       const unsigned short* __restrict__ x,
+                                       const unsigned short* __restrict__ bias,
+                                       unsigned short* __restrict__ dx, long N,
+                                       int F) {
+  long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
+  const long total = N * (F / VEC);
+  const long stride = (long)gridDim.x * BLOCK;
+  for (; idx < total; idx += stride) {
+    const long row = idx / (F / VEC);
+    const int col = (int)(idx % (F / VEC)) * VEC;
+    const unsigned short* x1 = x + row * 2L * F + col;
+    const unsigned short* x2 = x1 + F;
+    short8v v1 = *(const short8v*)x1;
+    short8v v2 = *(const short8v*)x2;
+    short8v d = *(const short8v*)(dy + row * F + col);
+    short8v o1, o2;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float a = bf2f((unsigned short)v1[j]);
+      float b = bf2f((unsigned short)v2[j]);
+      if (bias != nullptr) {
+        a += bf2f((unsigned short)bias[col + j]);
+        b += bf2f((unsigned short)bias[F + col + j]);
+      }
+      float dyf = bf2f((unsigned short)d[j]);
+      float sig = 1.f / (1.f + __expf(-a));
+      float silu = a * sig;
+      o1[j] = (short)f2bf(dyf * b * sig * (1.f + a * (1.f - sig)));
+      o2[j] = (short)f2bf(dyf * silu);
+    }
+    *(short8v*)(dx + row * 2L * F + col) = o1;
+    *(short8v*)(dx + row * 2L * F + F + col) = o2;
+  }
+}
+
+// ------------------------------------------------------------------ launchers
+static int ew_grid(long work_items) {
+  long blocks = (work_items + BLOCK - 1) / BLOCK;
+  return (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
+}
+
+void launch_bias_gelu_fwd(const void* x, const void* bias, void* y, long n,
+                          int F, hipStream_t s) {
+  hipLaunchKernelGGL(bias_gelu_fwd_kernel, dim3(ew_grid(n / VEC)), dim3(BLOCK),
+                     0, s, (const unsigned short*)x,
+                     (const unsigned short*)bias, (unsigned short*)y, n, F);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_bias_gelu_bwd(const void* dy, const void* x, const void* bias,
+                          void* dx, long n, int F, hipStream_t s) {
+  hipLaunchKernelGGL(bias_gelu_bwd_kernel, dim3(ew_grid(n / VEC)), dim3(BLOCK),
+                     0, s, (const unsigned short*)dy, (const unsigned short*)x,
+                     (const unsigned short*)bias, (unsigned short*)dx, n, F);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_bias_swiglu_fwd(const void* x, const void* bias, void* y, long N,
+                            int F, hipStream_t s) {
+  if (F % VEC != 0) throw std::runtime_error("F must be divisible by 8");
+  hipLaunchKernelGGL(bias_swiglu_fwd_kernel, dim3(ew_grid(N * (F / VEC))),
+                     dim3(BLOCK), 0, s, (const unsigned short*)x,
+                     (const unsigned short*)bias, (unsigned short*)y, N, F);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_bias_swiglu_bwd(const void* dy, const void* x, const void* bias,
+                            void* dx, long N, int F, hipStream_t s) {
+  if (F % VEC != 0) throw std::runtime_error("F must be divisible by 8");
+  hipLaunchKernelGGL(bias_swiglu_bwd_kernel, dim3(ew_grid(N * (F / VEC))),
+                     dim3(BLOCK), 0, s, (const unsigned short*)dy,
+                     (const unsigned short*)x, (const unsigned short*)bias,
+                     (unsigned short*)dx, N, F);
+  HIP_CHECK_LAUNCH();
+}

@@ -1,0 +1,279 @@
+// Fused RMSNorm / LayerNorm forward + backward for gfx950.
+//
+// Design (CDNA4 guide Appendix B, G13): memory-bound; bf16 loads as
+// short8 (16 B/lane); one 256-thread block per row-group; fp32
+// accumulation with wave shuffle + LDS block reduction; dw/db partials
+// accumulated per block in registers over a grid-stride row loop, then
+// one fp32 atomicAdd per column per block.
+//
+// Replaces: reference core/fusions/fused_layer_norm.py (apex kernels) and
+// RMSNorm torch path.
+
+#include "common.h"
+
+#include <stdexcept>
+#include <string>
+
+#define BLOCK 256
+// each thread handles VEC bf16 elements per row-chunk
+#define VEC 8
+
+// ---------------------------------------------------------------- RMSNorm fwd
+__global__ void rmsnorm_fwd_kernel(const unsigned short* __restrict__ x,
+                                   const unsigned short* __restrict__ w,
+                                   unsigned short* __restrict__ y,
+                                   float* __restrict__ invrms, int N, int H,
+                                   float eps) {
+  __shared__ float lds[BLOCK / WAVE];
+  for (int row = blockIdx.x; row < N; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * H;
+    unsigned short* yr = y + (long)row * H;
+    float ss = 0.f;
+    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+      short8v v = *(const short8v*)(xr + base);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float f = bf2f((unsigned short)v[j]);
+        ss += f * f;
+      }
+    }
+    ss = block_reduce_sum<BLOCK>(ss, lds);
+    float r = rsqrtf(ss / H + eps);
+    if (threadIdx.x == 0) invrms[row] = r;
+    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+      short8v v = *(const short8v*)(xr + base);
+      short8v wv = *(const short8v*)(w + base);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o[j] = (short)f2bf(bf2f((unsigned short)v[j]) * r *
+                           bf2f((unsigned short)wv[j]));
+      *(short8v*)(yr + base) = o;
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------- RMSNorm bwd
+__global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
+                                   const unsigned short* __restrict__ x,
+                                   const unsigned short* __restrict__ w,
+                                   const float* __restrict__ invrms,
+                                   unsigned short* __restrict__ dx,
+                                   float* __restrict__ dw, int N, int H) {
+  __shared__ float lds[BLOCK / WAVE];
+  // per-thread dw partial over its fixed columns
+  constexpr int MAX_COLS = 64;  // supports H <= BLOCK*VEC*MAX_COLS/8...
+  float dwacc[MAX_COLS];
+  const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
+#pragma unroll 4
+  for (int c = 0; c < chunks * VEC && c < MAX_COLS; ++c) dwacc[c] = 0.f;
+
+  for (int row = blockIdx.x; row < N; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * H;
+    const unsigned short* dyr = dy + (long)row * H;
+    unsigned short* dxr = dx + (long)row * H;
+    const float r = invrms[row];
+    // pass 1: dot = sum(dy*w*xhat)
+    float dot = 0.f;
+    for (int base = threadIdx.x * VEC, c = 0; base < H;
+         base += BLOCK * VEC, ++c) {
+      short8v xv = *(const short8v*)(xr + base);
+      short8v dv = *(const short8v*)(dyr + base);
+      short8v wv = *(const short8v*)(w + base);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float xh = bf2f((unsigned short)xv[j]) * r;
+        float dxh = bf2f((unsigned short)dv[j]) * bf2f((unsigned short)wv[j]);
+        dot += dxh * xh;
+        dwacc[c * VEC + j] += bf2f((unsigned short)dv[j]) * xh;
+      }
+    }
+    dot = block_reduce_sum<BLOCK>(dot, lds) / H;
+    // pass 2: dx = r * (dy*w - xhat * dot)
+    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+      short8v xv = *(const short8v*)(xr + base);
+      short8v dv = *(const short8v*)(dyr + base);
+      short8v wv = *(const short8v*)(w + base);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float xh = bf2f((unsigned short)xv[j]) * r;
+        float dxh = bf2f((unsigned short)dv[j]) * bf2f((unsigned short)wv[j]);
+        o[j] = (short)f2bf(r * (dxh - xh * dot));
+      }
+      *(short8v*)(dxr + base) = o;
+    }
+    __syncthreads();
+  }
+  // flush dw partials
+  for (int base = threadIdx.x * VEC, c = 0; base < H; base += BLOCK * VEC, ++c)
+#pragma unroll
+    for (int j = 0; j < VEC; ++j)
+      atomicAdd(dw + base + j, dwacc[c * VEC + j]);
+}
+
+// -------------------------------------------------------------- LayerNorm fwd
+__global__ void layernorm_fwd_kernel(const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ w,
+                                     const unsigned short* __restrict__ b,
+                                     unsigned short* __restrict__ y,
+                                     float* __restrict__ mean,
+                                     float* __restrict__ invstd, int N, int H,
+                                     float eps) {
+  __shared__ float lds[BLOCK / WAVE];
+  for (int row = blockIdx.x; row < N; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * H;
+    unsigned short* yr = y + (long)row * H;
+    float s = 0.f, ss = 0.f;
+    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+      short8v v = *(const short8v*)(xr + base);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float f = bf2f((unsigned short)v[j]);
+        s += f;
+        ss += f * f;
+      }
+    }
+    s = block_reduce_sum<BLOCK>(s, lds);
+    ss = block_reduce_sum<BLOCK>(ss, lds);
+    float mu = s / H;
+    float var = ss / H - mu * mu;
+    float r = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean[row] = mu;
+      invstd[row] = r;
+    }
+    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+      short8v v = *(const short8v*)(xr + base);
+      short8v wv = *(const short8v*)(w + base);
+      short8v bv = *(const short8v*)(b + base);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o[j] = (short)f2bf((bf2f((unsigned short)v[j]) - mu) * r *
+                               bf2f((unsigned short)wv[j]) +
+                           bf2f((unsigned short)bv[j]));
+      *(short8v*)(yr + base) = o;
+    }
+    __syncthreads();
+  }
+}
+
+// -------------------------------------------------------------- LayerNorm bwd
+__global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
+                                     const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ w,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     unsigned short* __restrict__ dx,
+                                     float* __restrict__ dw,
+                                     float* __restrict__ db, int N, int H) {
+  __shared__ float lds[BLOCK / WAVE];
+  constexpr int MAX_COLS = 64;
+  float dwacc[MAX_COLS];
+  float dbacc[MAX_COLS];
+  const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
+  for (int c = 0; c < chunks * VEC && c < MAX_COLS; ++c) {
+    dwacc[c] = 0.f;
+    dbacc[c] = 0.f;
+  }
+
+  for (int row = blockIdx.x; row < N; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * H;
+    const unsigned short* dyr = dy + (long)row * H;
+    unsigned short* dxr = dx + (long)row * H;
+    const float mu = mean[row];
+    const float r = invstd[row];
+    float sum1 = 0.f, sum2 = 0.f;
+    for (int base = threadIdx.x * VEC, c = 0; base < H;
+         base += BLOCK * VEC, ++c) {
+      short8v xv = *(const short8v*)(xr + base);
+      short8v dv = *(const short8v*)(dyr + base);
+      short8v wv = *(const short8v*)(w + base);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float xh = (bf2f((unsigned short)xv[j]) - mu) * r;
+        float dyf = bf2f((unsigned short)dv[j]);
+        float dxh = dyf * bf2f((unsigned short)wv[j]);
+        sum1 += dxh;
+        sum2 += dxh * xh;
+        dwacc[c * VEC + j] += dyf * xh;
+        dbacc[c * VEC + j] += dyf;
+      }
+    }
+    sum1 = block_reduce_sum<BLOCK>(sum1, lds) / H;
+    sum2 = block_reduce_sum<BLOCK>(sum2, lds) / H;
+    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+      short8v xv = *(const short8v*)(xr + base);
+      short8v dv = *(const short8v*)(dyr + base);
+      short8v wv = *(const short8v*)(w + base);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float xh = (bf2f((unsigned short)xv[j]) - mu) * r;
+        float dxh =
+            bf2f((unsigned short)dv[j]) * bf2f((unsigned short)wv[j]);
+        o[j] = (short)f2bf(r * (dxh - sum1 - xh * sum2));
+      }
+      *(short8v*)(dxr + base) = o;
+    }
+    __syncthreads();
+  }
+  for (int base = threadIdx.x * VEC, c = 0; base < H; base += BLOCK * VEC, ++c)
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      atomicAdd(dw + base + j, dwacc[c * VEC + j]);
+      atomicAdd(db + base + j, dbacc[c * VEC + j]);
+    }
+}
+
+// ------------------------------------------------------------------ launchers
+void launch_rmsnorm_fwd(const void* x, const void* w, void* y, float* invrms,
+                        int N, int H, float eps, hipStream_t stream) {
+  if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
+  int grid = N < 2048 ? N : 2048;
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(grid), dim3(BLOCK), 0, stream,
+                     (const unsigned short*)x, (const unsigned short*)w,
+                     (unsigned short*)y, invrms, N, H, eps);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_rmsnorm_bwd(const void* dy, const void* x, const void* w,
+                        const float* invrms, void* dx, float* dw, int N, int H,
+                        hipStream_t stream) {
+  if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
+  if (H > BLOCK * VEC * 8) throw std::runtime_error("H too large for rmsnorm bwd");
+  int grid = N < 1024 ? N : 1024;
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(BLOCK), 0, stream,
+                     (const unsigned short*)dy, (const unsigned short*)x,
+                     (const unsigned short*)w, invrms, (unsigned short*)dx, dw,
+                     N, H);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_layernorm_fwd(const void* x, const void* w, const void* b, void* y,
+                          float* mean, float* invstd, int N, int H, float eps,
+                          hipStream_t stream) {
+  if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
+  int grid = N < 2048 ? N : 2048;
+  hipLaunchKernelGGL(layernorm_fwd_kernel, dim3(grid), dim3(BLOCK), 0, stream,
+                     (const unsigned short*)x, (const unsigned short*)w,
+                     (const unsigned short*)b, (unsigned short*)y, mean, invstd,
+                     N, H, eps);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_layernorm_bwd(const void* dy, const void* x, const void* w,
+                          const float* mean, const float* invstd, void* dx,
+                          float* dw, float* db, int N, int H,
+                          hipStream_t stream) {
+  if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
+  if (H > BLOCK * VEC * 8) throw std::runtime_error("H too large for layernorm bwd");
+  int grid = N < 1024 ? N : 1024;
+  hipLaunchKernelGGL(layernorm_bwd_kernel, dim3(grid), dim3(BLOCK), 0, stream,
+                     (const unsigned short*)dy, (const unsigned short*)x,
+                     (const unsigned short*)w, mean, invstd,
+                     (unsigned short*)dx, dw, db, N, H);
+  HIP_CHECK_LAUNCH();
+}

@@ -1,0 +1,169 @@
+// Fused scale + mask + softmax kernels (bf16 I/O, fp32 math).
+// Replaces apex scaled_{upper_triang_,}masked_softmax.
+//
+// Layout: scores [R, sk] where R = b*np*sq rows; causal variant derives
+// the valid length from the row's q index.  One 256-thread block per
+// row; grid-strided; online (max, sum) in fp32.
+
+#include "common.h"
+
+#include <stdexcept>
+
+#define BLOCK 256
+#define VEC 8
+
+// ------------------------------------------------- causal (upper-triangular)
+__global__ void softmax_causal_fwd_kernel(const unsigned short* __restrict__ x,
+                                          unsigned short* __restrict__ y,
+                                          long rows, int sq, int sk,
+                                          float scale) {
+  __shared__ float lds[BLOCK / WAVE];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int q = (int)(row % sq);
+    const int valid = q + 1 + (sk - sq);  // causal with kv prefix
+    const unsigned short* xr = x + row * sk;
+    unsigned short* yr = y + row * sk;
+
+    float m = -INFINITY;
+    for (int base = threadIdx.x * VEC; base < sk; base += BLOCK * VEC) {
+      short8v v = *(const short8v*)(xr + base);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        if (base + j < valid) m = fmaxf(m, bf2f((unsigned short)v[j]) * scale);
+    }
+    m = block_reduce_max<BLOCK>(m, lds);
+
+    float sum = 0.f;
+    for (int base = threadIdx.x * VEC; base < sk; base += BLOCK * VEC) {
+      short8v v = *(const short8v*)(xr + base);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        if (base + j < valid)
+          sum += __expf(bf2f((unsigned short)v[j]) * scale - m);
+    }
+    sum = block_reduce_sum<BLOCK>(sum, lds);
+    const float inv = 1.f / sum;
+
+    for (int base = threadIdx.x * VEC; base < sk; base += BLOCK * VEC) {
+      short8v v = *(const short8v*)(xr + base);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        o[j] = (base + j < valid)
+                   ? (short)f2bf(__expf(bf2f((unsigned short)v[j]) * scale - m) * inv)
+                   : (short)0;
+      *(short8v*)(yr + base) = o;
+    }
+    __syncthreads();
+  }
+}
+
+// --------------------------------------------------------- generic bool mask
+// mask: [mrows, sq, sk] uint8 (1 = masked out); broadcast over heads when
+// mrows == b (mask row = row / (np*sq) * sq + q).
+__global__ void softmax_masked_fwd_kernel(const unsigned short* __restrict__ x,
+                                          const unsigned char* __restrict__ mask,
+                                          unsigned short* __restrict__ y,
+                                          long rows, int np, int sq, int sk,
+                                          float scale) {
+  __shared__ float lds[BLOCK / WAVE];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int q = (int)(row % sq);
+    const long bidx = row / ((long)np * sq);
+    const unsigned char* mr =
+        mask == nullptr ? nullptr : mask + (bidx * sq + q) * (long)sk;
+    const unsigned short* xr = x + row * sk;
+    unsigned short* yr = y + row * sk;
+
+    float m = -INFINITY;
+    for (int i = threadIdx.x; i < sk; i += BLOCK) {
+      if (mr == nullptr || !mr[i])
+        m = fmaxf(m, bf2f(xr[i]) * scale);
+    }
+    m = block_reduce_max<BLOCK>(m, lds);
+    if (m == -INFINITY) m = 0.f;  // fully masked row
+
+    float sum = 0.f;
+    for (int i = threadIdx.x; i < sk; i += BLOCK) {
+      if (mr == nullptr || !mr[i]) sum += __expf(bf2f(xr[i]) * scale - m);
+    }
+    sum = block_reduce_sum<BLOCK>(sum, lds);
+    const float inv = sum > 0.f ? 1.f / sum : 0.f;
+
+    for (int i = threadIdx.x; i < sk; i += BLOCK) {
+      float p = (mr == nullptr || !mr[i])
+                    ? __expf(bf2f(xr[i]) * scale - m) * inv
+                    : 0.f;
+      yr[i] = f2bf(p);
+    }
+    __syncthreads();
+  }
+}
+
+// -------------------------------------------------------------------- bwd
+// dx = p * (dy - sum(dy*p)) * scale     (rowwise)
+__global__ void softmax_bwd_kernel(const unsigned short* __restrict__ dy,
+                                   const unsigned short* __restrict__ p,
+                                   unsigned short* __restrict__ dx, long rows,
+                                   int sk, float scale) {
+  __shared__ float lds[BLOCK / WAVE];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* dyr = dy + row * sk;
+    const unsigned short* pr = p + row * sk;
+    unsigned short* dxr = dx + row * sk;
+    float dot = 0.f;
+    for (int base = threadIdx.x * VEC; base < sk; base += BLOCK * VEC) {
+      short8v d = *(const short8v*)(dyr + base);
+      short8v pv = *(const short8v*)(pr + base);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        dot += bf2f((unsigned short)d[j]) * bf2f((unsigned short)pv[j]);
+    }
+    dot = block_reduce_sum<BLOCK>(dot, lds);
+    for (int base = threadIdx.x * VEC; base < sk; base += BLOCK * VEC) {
+      short8v d = *(const short8v*)(dyr + base);
+      short8v pv = *(const short8v*)(pr + base);
+      short8v o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float pf = bf2f((unsigned short)pv[j]);
+        o[j] = (short)f2bf(pf * (bf2f((unsigned short)d[j]) - dot) * scale);
+      }
+      *(short8v*)(dxr + base) = o;
+    }
+    __syncthreads();
+  }
+}
+
+// ------------------------------------------------------------------ launchers
+static int sm_grid(long rows) {
+  return (int)(rows < 4096 ? (rows < 1 ? 1 : rows) : 4096);
+}
+
+void launch_softmax_causal_fwd(const void* x, void* y, long rows, int sq,
+                               int sk, float scale, hipStream_t s) {
+  if (sk % VEC != 0) throw std::runtime_error("sk must be divisible by 8");
+  hipLaunchKernelGGL(softmax_causal_fwd_kernel, dim3(sm_grid(rows)),
+                     dim3(BLOCK), 0, s, (const unsigned short*)x,
+                     (unsigned short*)y, rows, sq, sk, scale);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_softmax_masked_fwd(const void* x, const void* mask, void* y,
+                               long rows, int np, int sq, int sk, float scale,
+                               hipStream_t s) {
+  hipLaunchKernelGGL(softmax_masked_fwd_kernel, dim3(sm_grid(rows)),
+                     dim3(BLOCK), 0, s, (const unsigned short*)x,
+                     (const unsigned char*)mask, (unsigned short*)y, rows, np,
+                     sq, sk, scale);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_softmax_bwd(const void* dy, const void* p, void* dx, long rows,
+                        int sk, float scale, hipStream_t s) {
+  if (sk % VEC != 0) throw std::runtime_error("sk must be divisible by 8");
+  hipLaunchKernelGGL(softmax_bwd_kernel, dim3(sm_grid(rows)), dim3(BLOCK), 0,
+                     s, (const unsigned short*)dy, (const unsigned short*)p,
+                     (unsigned short*)dx, rows, sk, scale);
+  HIP_CHECK_LAUNCH();
+}
