@@ -26,6 +26,7 @@ import torch.nn.functional as F
 from torch.nn.parameter import Parameter
 
 from .. import parallel_state
+from ... import ops as _ops
 from .mappings import (
     copy_to_tensor_model_parallel_region,
     gather_from_tensor_model_parallel_region,
@@ -181,10 +182,23 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         grad_output_2d = grad_output.reshape(-1, grad_output.shape[-1])
         total_input_2d = total_input.reshape(-1, total_input.shape[-1])
         if ctx.gradient_accumulation_fusion and hasattr(weight, "main_grad"):
-            weight.main_grad.add_(
-                torch.matmul(grad_output_2d.t(), total_input_2d))
-            grad_weight = None
-            weight.grad_added_to_main_grad = True
+            if grad_output.is_cuda and _ops.have_ops():
+                # fp32-accumulating hipblasLt wgrad straight into main_grad
+                _ops.get_ops().wgrad_accum(
+                    grad_output_2d.contiguous(), total_input_2d.contiguous(),
+                    weight.main_grad)
+            else:
+                weight.main_grad.add_(
+                    torch.matmul(grad_output_2d.t(), total_input_2d))
+            if hasattr(weight, "grad_added_to_main_grad"):
+                # DDP-managed: return a dummy grad so the param's
+                # post-accumulate hook still fires for bucket bookkeeping
+                grad_weight = torch.empty(
+                    weight.shape, dtype=weight.dtype, device=weight.device,
+                    requires_grad=False)
+                weight.grad_added_to_main_grad = True
+            else:
+                grad_weight = None
         else:
             grad_weight = grad_output_2d.t().matmul(total_input_2d)
         grad_bias = grad_output_2d.sum(dim=0) if use_bias else None

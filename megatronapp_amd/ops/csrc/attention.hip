@@ -1,0 +1,813 @@
+// Flash attention (forward + backward) for gfx950 — MFMA 16x16x32 bf16
+// tiles, online softmax, O(s) memory.  Replaces the reference's TE/cuDNN
+// fused attention and the local baddbmm+softmax+bmm path.
+//
+// Structure (CDNA4 guide §5/§B):
+//  * workgroup = 4 waves (256 threads); each wave owns QW=16 query rows,
+//    so a workgroup covers a 64-row Q tile of one (batch, head).
+//  * K/V tiles (KVBLK=64 x D) are staged cooperatively in LDS: K in
+//    row-major [kv][D] with the ((row&7)<<4) XOR byte-swizzle so the
+//    ds_read_b128 B-fragments are conflict-free (guide G4/T2), V stored
+//    TRANSPOSED [D][kv] (so PV B-fragments are contiguous 16-byte reads).
+//  * S tiles accumulate in fp32 MFMA accumulators; softmax row stats via
+//    16-lane __shfl_xor reductions (C-layout rows live in 16-lane groups).
+//  * P goes through a small LDS buffer to re-shape from C-layout to the
+//    A-fragment layout of the PV MFMAs.
+//
+// fragment maps (v_mfma_f32_16x16x32_bf16):
+//   A[m][k]: lane l holds A[l&15][(l>>4)*8 + j], j=0..7
+//   B[k][n]: lane l holds B[(l>>4)*8 + j][l&15]
+//   C[m][n]: lane l holds C[(l>>4)*4 + r][l&15], r=0..3
+
+#include "common.h"
+
+#include <stdexcept>
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+
+#define ATT_BLOCK 256
+#define QW 16    // q rows per wave
+#define QBLK 64  // q rows per workgroup
+#define KVBLK 64
+
+__device__ __forceinline__ unsigned swz(int row, int col_bytes) {
+  // XOR swizzle within a row: spreads the 16B slots of a 256B bank row
+  return (unsigned)(col_bytes ^ ((row & 7) << 4));
+}
+
+// row-stat reduction across the 16-lane C-layout group
+__device__ __forceinline__ float group16_max(float v) {
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_xor(v, off, 16));
+  return v;
+}
+
+__device__ __forceinline__ float group16_sum(float v) {
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) v += __shfl_xor(v, off, 16);
+  return v;
+}
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, unsigned short* __restrict__ o,
+    float* __restrict__ lse, int sq, int sk, int b, int nh, int ng,
+    float scale) {
+  // grid: (sq/QBLK, b*nh)
+  const int qtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / nh;
+  const int h = bh % nh;
+  const int hkv = h / (nh / ng);
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lrow = lane >> 4;  // 0..3 (lane group)
+  const int lcol = lane & 15;
+
+  // element strides along the seq dim
+  const long q_ss = (long)b * nh * D;
+  const long k_ss = (long)b * ng * D;
+  const unsigned short* qp = q + ((long)bi * nh + h) * D;
+  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
+  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+
+  // LDS: K [KVBLK][D] swizzled (bf16), Vt [D][KVBLK] swizzled, P [QBLK][KVBLK]
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* k_lds = (unsigned short*)smem;               // KVBLK*D
+  unsigned short* vt_lds = k_lds + KVBLK * D;                  // D*KVBLK
+  unsigned short* p_lds = vt_lds + (long)D * KVBLK;            // QBLK*KVBLK
+
+  const int q0 = qtile * QBLK + wid * QW;  // this wave's first q row
+
+  // ---- load Q fragments (A-layout): frag f covers d = f*32 + lrow*8 + j
+  constexpr int DF = D / 32;  // MFMA k-steps over d
+  bf16x8 qfrag[DF];
+  {
+    const int qrow = q0 + lcol;
+    const unsigned short* src = qp + (long)qrow * q_ss;
+#pragma unroll
+    for (int f = 0; f < DF; ++f) {
+      if (qrow < sq)
+        qfrag[f] = *(const bf16x8*)(src + f * 32 + lrow * 8);
+      else
+        qfrag[f] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  // ---- softmax state (per lane: 4 rows r=0..3 -> q row lrow*4+r)
+  float m_run[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
+  // O accumulators: 8 d-subtiles x f32x4
+  constexpr int DS = D / 16;
+  f32x4 oacc[DS];
+#pragma unroll
+  for (int dsub = 0; dsub < DS; ++dsub) oacc[dsub] = f32x4{0, 0, 0, 0};
+
+  const int q_hi = qtile * QBLK + QBLK - 1;  // last q row in workgroup
+  int kv_end = sk;
+  if (CAUSAL) kv_end = min(sk, q_hi + 1 + (sk - sq));
+  const int n_kv_tiles = (kv_end + KVBLK - 1) / KVBLK;
+
+  for (int t = 0; t < n_kv_tiles; ++t) {
+    const int kv0 = t * KVBLK;
+    // ---- cooperative K/V load: 256 threads, K tile KVBLK x D
+    // each thread loads (KVBLK*D/8)/256 short8 pieces
+    {
+      constexpr int pieces = KVBLK * D / 8 / ATT_BLOCK;  // e.g. 4 at D=128
+#pragma unroll
+      for (int pc = 0; pc < pieces; ++pc) {
+        const int idx = (pc * ATT_BLOCK + tid) * 8;  // element index
+        const int row = idx / D;
+        const int col = idx % D;
+        const int grow = kv0 + row;
+        bf16x8 kv8 = (grow < sk)
+                         ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                         : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) = kv8;
+        bf16x8 vv8 = (grow < sk)
+                         ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                         : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        // transpose into vt_lds[d][kv] (scalar stores, swizzled rows)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = col + j;
+          *(unsigned short*)((char*)(vt_lds + (long)d * KVBLK) +
+                             swz(d, row * 2)) = (unsigned short)vv8[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S = scale * Q K^T for this wave's 16 q rows, 4 kv subtiles
+    f32x4 sacc[KVBLK / 16];
+#pragma unroll
+    for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+      sacc[ksub] = f32x4{0, 0, 0, 0};
+#pragma unroll
+      for (int f = 0; f < DF; ++f) {
+        // B fragment: B[d][kv] = K[kv0+ksub*16 + lcol][f*32 + lrow*8 + j]
+        const int krow = ksub * 16 + lcol;
+        bf16x8 bfrag = *(const bf16x8*)((char*)(k_lds + (long)krow * D) +
+                                        swz(krow, (f * 32 + lrow * 8) * 2));
+        sacc[ksub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qfrag[f], bfrag, sacc[ksub], 0, 0, 0);
+      }
+    }
+
+    // ---- masking + online softmax
+    float mtile[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+#pragma unroll
+    for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + lrow * 4 + r;
+        const int kvcol = kv0 + ksub * 16 + lcol;
+        float s = sacc[ksub][r] * scale;
+        bool valid = (kvcol < sk) && (qrow < sq);
+        if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
+        s = valid ? s : -INFINITY;
+        sacc[ksub][r] = s;
+        mtile[r] = fmaxf(mtile[r], s);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) mtile[r] = group16_max(mtile[r]);
+
+    float alpha[4];
+    float lt[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float mn = fmaxf(m_run[r], mtile[r]);
+      alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - mn);
+      m_run[r] = mn;
+    }
+    // P = exp(S - m); write to p_lds in A layout (row=q, col=kv), swizzled
+#pragma unroll
+    for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = sacc[ksub][r];
+        float p = (s == -INFINITY) ? 0.f : __expf(s - m_run[r]);
+        lt[r] += p;
+        const int prow = wid * QW + lrow * 4 + r;       // q within block
+        const int pcol = ksub * 16 + lcol;              // kv within tile
+        *(unsigned short*)((char*)(p_lds + (long)prow * KVBLK) +
+                           swz(prow, pcol * 2)) = f2bf(p);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      lt[r] = group16_sum(lt[r]);
+      l_run[r] = l_run[r] * alpha[r] + lt[r];
+    }
+    // rescale O
+#pragma unroll
+    for (int dsub = 0; dsub < DS; ++dsub)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) oacc[dsub][r] *= alpha[r];
+
+    __syncthreads();  // p_lds writes visible within the wave anyway; keep
+                      // the barrier so k_lds/vt_lds reuse next tile is safe
+
+    // ---- O += P V : contraction over kv (2 MFMA k-steps of 32)
+#pragma unroll
+    for (int dsub = 0; dsub < DS; ++dsub) {
+#pragma unroll
+      for (int ks = 0; ks < KVBLK / 32; ++ks) {
+        // A fragment: P[q0w + l&15][ks*32 + lrow*8 + j]
+        const int prow = wid * QW + lcol;
+        bf16x8 pa;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int pcol = ks * 32 + lrow * 8 + j;
+          pa[j] = (short)*(const unsigned short*)(
+              (char*)(p_lds + (long)prow * KVBLK) + swz(prow, pcol * 2));
+        }
+        // B fragment: Vt[dsub*16 + lcol][ks*32 + lrow*8 + j] -> V[kv][d]
+        const int vrow = dsub * 16 + lcol;
+        bf16x8 vb;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int vcol = ks * 32 + lrow * 8 + j;
+          vb[j] = (short)*(const unsigned short*)(
+              (char*)(vt_lds + (long)vrow * KVBLK) + swz(vrow, vcol * 2));
+        }
+        // note operand order: C[q][d] = A(P[q][kv]) x B(V[kv][d]); our B
+        // fragment is indexed [d][kv] -> use transposed roles:
+        // mfma(A=pa over kv, B=vb over kv) with B[k][n]: k=kv, n=d: need
+        // B[kv][d] = Vt[d][kv] read with n=lcol over d -> OK as built.
+        oacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pa, vb, oacc[dsub], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: normalize, write O and LSE
+  // O C-layout: lane holds O[q=lrow*4+r][d=dsub*16+lcol]
+  // stage through p_lds (reuse as [QW][D] per wave? QBLK*KVBLK >= QW*D OK
+  // for D<=256) then coalesced store
+  // per-wave staging slice of k_lds (KVBLK*D = 4 waves x QW*D shorts)
+  unsigned short* o_stage = k_lds + (long)wid * QW * D;
+
+#pragma unroll
+  for (int dsub = 0; dsub < DS; ++dsub) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float denom = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+      o_stage[(long)(lrow * 4 + r) * D + dsub * 16 + lcol] =
+          f2bf(oacc[dsub][r] * denom);
+    }
+  }
+  if (lcol == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + lrow * 4 + r;
+      if (qrow < sq)
+        lse[((long)bi * nh + h) * sq + qrow] =
+            m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+    }
+  }
+  __syncthreads();
+  // coalesced store: each lane writes short8 rows of its wave's tile
+  {
+    constexpr int pieces = QW * D / 8 / WAVE;
+#pragma unroll
+    for (int pc = 0; pc < pieces; ++pc) {
+      const int idx = (pc * WAVE + lane) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int qrow = q0 + row;
+      if (qrow < sq)
+        *(bf16x8*)(o + ((long)qrow * b * nh + (long)bi * nh + h) * D + col) =
+            *(const bf16x8*)(o_stage + (long)row * D + col);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
+                     float* lse, int sq, int sk, int b, int nh, int ng, int d,
+                     float scale, bool causal, hipStream_t stream) {
+  if (sq % QBLK != 0 || sk % KVBLK != 0)
+    throw std::runtime_error("attn_fwd: sq/sk must be multiples of 64");
+  dim3 grid(sq / QBLK, b * nh);
+  dim3 block(ATT_BLOCK);
+  const size_t lds = (size_t)(KVBLK * d + d * KVBLK + QBLK * KVBLK) *
+                     sizeof(unsigned short);
+  if (d == 128) {
+    if (causal)
+      hipLaunchKernelGGL((attn_fwd_kernel<128, true>), grid, block, lds,
+                         stream, (const unsigned short*)q,
+                         (const unsigned short*)k, (const unsigned short*)v,
+                         (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+    else
+      hipLaunchKernelGGL((attn_fwd_kernel<128, false>), grid, block, lds,
+                         stream, (const unsigned short*)q,
+                         (const unsigned short*)k, (const unsigned short*)v,
+                         (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+  } else if (d == 64) {
+    if (causal)
+      hipLaunchKernelGGL((attn_fwd_kernel<64, true>), grid, block, lds,
+                         stream, (const unsigned short*)q,
+                         (const unsigned short*)k, (const unsigned short*)v,
+                         (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+    else
+      hipLaunchKernelGGL((attn_fwd_kernel<64, false>), grid, block, lds,
+                         stream, (const unsigned short*)q,
+                         (const unsigned short*)k, (const unsigned short*)v,
+                         (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+  } else {
+    throw std::runtime_error("attn_fwd: head dim must be 64 or 128");
+  }
+  HIP_CHECK_LAUNCH();
+}
+
+// ===========================================================================
+// Backward.  FA2-style split: a dQ kernel gridded over q tiles and a dK/dV
+// kernel gridded over kv tiles (no cross-block atomics; S/P recomputed from
+// the stored LSE).  Drow = rowsum(dO * O) precomputed by attn_bwd_pre.
+// ===========================================================================
+
+__global__ void attn_bwd_pre_kernel(const unsigned short* __restrict__ do_,
+                                    const unsigned short* __restrict__ o,
+                                    float* __restrict__ drow, long rows,
+                                    int d) {
+  // one wave per row; rows = sq*b*nh, layout [sq][b][nh][d]
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+  if (row >= rows) return;
+  const int lane = threadIdx.x % WAVE;
+  const unsigned short* dor = do_ + row * d;
+  const unsigned short* orow = o + row * d;
+  float acc = 0.f;
+  for (int i = lane * 2; i < d; i += WAVE * 2) {
+    acc += bf2f(dor[i]) * bf2f(orow[i]);
+    acc += bf2f(dor[i + 1]) * bf2f(orow[i + 1]);
+  }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) drow[row] = acc;
+}
+
+// ---------------------------------------------------------------- dQ kernel
+// grid (sq/QBLK, b*nh); per wave 16 q rows; loops kv tiles.
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    unsigned short* __restrict__ dq, int sq, int sk, int b, int nh, int ng,
+    float scale) {
+  const int qtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / nh;
+  const int h = bh % nh;
+  const int hkv = h / (nh / ng);
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lrow = lane >> 4;
+  const int lcol = lane & 15;
+
+  const long q_ss = (long)b * nh * D;
+  const long k_ss = (long)b * ng * D;
+  const unsigned short* qp = q + ((long)bi * nh + h) * D;
+  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
+  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+  const unsigned short* dop = dout + ((long)bi * nh + h) * D;
+  const float* lse_row = lse + ((long)bi * nh + h) * sq;
+  // drow layout matches [sq][b][nh]
+  const long dr_ss = (long)b * nh;
+  const float* dr_base = drow + (long)bi * nh + h;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* k_lds = (unsigned short*)smem;              // KVBLK*D
+  unsigned short* kt_lds = k_lds + KVBLK * D;                 // D*KVBLK
+  unsigned short* v_lds = kt_lds + (long)D * KVBLK;           // KVBLK*D
+  unsigned short* ds_lds = v_lds + KVBLK * D;                 // QBLK*KVBLK
+
+  const int q0 = qtile * QBLK + wid * QW;
+  constexpr int DF = D / 32;
+  constexpr int DS_ = D / 16;
+
+  bf16x8 qfrag[DF], dofrag[DF];
+  float lse_r[4], dr_r[4];
+  {
+    const int qrow = q0 + lcol;
+    const unsigned short* src = qp + (long)qrow * q_ss;
+    const unsigned short* dsrc = dop + (long)qrow * q_ss;
+#pragma unroll
+    for (int f = 0; f < DF; ++f) {
+      qfrag[f] = *(const bf16x8*)(src + f * 32 + lrow * 8);
+      dofrag[f] = *(const bf16x8*)(dsrc + f * 32 + lrow * 8);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qr = q0 + lrow * 4 + r;
+      lse_r[r] = lse_row[qr];
+      dr_r[r] = dr_base[(long)qr * dr_ss];
+    }
+  }
+
+  f32x4 dqacc[DS_];
+#pragma unroll
+  for (int s = 0; s < DS_; ++s) dqacc[s] = f32x4{0, 0, 0, 0};
+
+  const int q_hi = qtile * QBLK + QBLK - 1;
+  int kv_end = sk;
+  if (CAUSAL) kv_end = min(sk, q_hi + 1 + (sk - sq));
+  const int n_kv_tiles = (kv_end + KVBLK - 1) / KVBLK;
+
+  for (int t = 0; t < n_kv_tiles; ++t) {
+    const int kv0 = t * KVBLK;
+    {
+      constexpr int pieces = KVBLK * D / 8 / ATT_BLOCK;
+#pragma unroll
+      for (int pc = 0; pc < pieces; ++pc) {
+        const int idx = (pc * ATT_BLOCK + tid) * 8;
+        const int row = idx / D;
+        const int col = idx % D;
+        const int grow = kv0 + row;
+        bf16x8 k8 = (grow < sk)
+                        ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                        : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) = k8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d_ = col + j;
+          *(unsigned short*)((char*)(kt_lds + (long)d_ * KVBLK) +
+                             swz(d_, row * 2)) = (unsigned short)k8[j];
+        }
+        bf16x8 v8 = (grow < sk)
+                        ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                        : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        *(bf16x8*)((char*)(v_lds + (long)row * D) + swz(row, col * 2)) = v8;
+      }
+    }
+    __syncthreads();
+
+    // S and dP, per kv subtile
+#pragma unroll
+    for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+      f32x4 sacc = f32x4{0, 0, 0, 0};
+      f32x4 dpacc = f32x4{0, 0, 0, 0};
+#pragma unroll
+      for (int f = 0; f < DF; ++f) {
+        const int krow = ksub * 16 + lcol;
+        bf16x8 kb = *(const bf16x8*)((char*)(k_lds + (long)krow * D) +
+                                     swz(krow, (f * 32 + lrow * 8) * 2));
+        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[f], kb, sacc,
+                                                       0, 0, 0);
+        bf16x8 vb = *(const bf16x8*)((char*)(v_lds + (long)krow * D) +
+                                     swz(krow, (f * 32 + lrow * 8) * 2));
+        dpacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dofrag[f], vb, dpacc,
+                                                        0, 0, 0);
+      }
+      // dS = P * (dP - Drow) * scale ; P = exp(S*scale - lse)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + lrow * 4 + r;
+        const int kvcol = kv0 + ksub * 16 + lcol;
+        bool valid = kvcol < sk;
+        if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
+        float p = valid ? __expf(sacc[r] * scale - lse_r[r]) : 0.f;
+        float ds_ = p * (dpacc[r] - dr_r[r]) * scale;
+        const int prow = wid * QW + lrow * 4 + r;
+        const int pcol = ksub * 16 + lcol;
+        *(unsigned short*)((char*)(ds_lds + (long)prow * KVBLK) +
+                           swz(prow, pcol * 2)) = f2bf(ds_);
+      }
+    }
+    __syncthreads();
+
+    // dQ += dS . K  (contraction over kv; B = K[kv][d] from kt_lds)
+#pragma unroll
+    for (int dsub = 0; dsub < DS_; ++dsub) {
+#pragma unroll
+      for (int ks = 0; ks < KVBLK / 32; ++ks) {
+        const int prow = wid * QW + lcol;
+        bf16x8 pa;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int pcol = ks * 32 + lrow * 8 + j;
+          pa[j] = (short)*(const unsigned short*)(
+              (char*)(ds_lds + (long)prow * KVBLK) + swz(prow, pcol * 2));
+        }
+        const int krow = dsub * 16 + lcol;  // d index
+        bf16x8 kb;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int kcol = ks * 32 + lrow * 8 + j;  // kv index
+          kb[j] = (short)*(const unsigned short*)(
+              (char*)(kt_lds + (long)krow * KVBLK) + swz(krow, kcol * 2));
+        }
+        dqacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, kb,
+                                                              dqacc[dsub],
+                                                              0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue via k_lds staging
+  unsigned short* stage = k_lds + (long)wid * QW * D;
+#pragma unroll
+  for (int dsub = 0; dsub < DS_; ++dsub)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      stage[(long)(lrow * 4 + r) * D + dsub * 16 + lcol] =
+          f2bf(dqacc[dsub][r]);
+  __syncthreads();
+  {
+    constexpr int pieces = QW * D / 8 / WAVE;
+#pragma unroll
+    for (int pc = 0; pc < pieces; ++pc) {
+      const int idx = (pc * WAVE + lane) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int qrow = q0 + row;
+      *(bf16x8*)(dq + ((long)qrow * b * nh + (long)bi * nh + h) * D + col) =
+          *(const bf16x8*)(stage + (long)row * D + col);
+    }
+  }
+}
+
+// -------------------------------------------------------------- dK/dV kernel
+// grid (sk/KVBLK, b*ng); per wave 16 kv rows; loops q tiles and the q-head
+// group (GQA: dK/dV sum over the nh/ng q heads sharing this kv head).
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    unsigned short* __restrict__ dk, unsigned short* __restrict__ dv, int sq,
+    int sk, int b, int nh, int ng, float scale) {
+  const int kvtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / ng;
+  const int hkv = bh % ng;
+  const int group = nh / ng;
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lrow = lane >> 4;
+  const int lcol = lane & 15;
+
+  const long q_ss = (long)b * nh * D;
+  const long k_ss = (long)b * ng * D;
+  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
+  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* qr_lds = (unsigned short*)smem;              // QBLK*D (row)
+  unsigned short* qt_lds = qr_lds + QBLK * D;                  // D*QBLK
+  unsigned short* dor_lds = qt_lds + (long)D * QBLK;           // QBLK*D
+  unsigned short* dot_lds = dor_lds + QBLK * D;                // D*QBLK
+  unsigned short* p_lds = dot_lds + (long)D * QBLK;            // KVBLK*QBLK
+
+  const int kv0 = kvtile * KVBLK + wid * QW;  // this wave's 16 kv rows
+  constexpr int DF = D / 32;
+  constexpr int DS_ = D / 16;
+
+  // K,V fragments for this wave's kv rows (A operands, reused all q tiles)
+  bf16x8 kfrag[DF], vfrag[DF];
+  {
+    const int kvrow = kv0 + lcol;
+    const unsigned short* ksrc = kp + (long)kvrow * k_ss;
+    const unsigned short* vsrc = vp + (long)kvrow * k_ss;
+#pragma unroll
+    for (int f = 0; f < DF; ++f) {
+      kfrag[f] = *(const bf16x8*)(ksrc + f * 32 + lrow * 8);
+      vfrag[f] = *(const bf16x8*)(vsrc + f * 32 + lrow * 8);
+    }
+  }
+
+  f32x4 dkacc[DS_], dvacc[DS_];
+#pragma unroll
+  for (int s = 0; s < DS_; ++s) {
+    dkacc[s] = f32x4{0, 0, 0, 0};
+    dvacc[s] = f32x4{0, 0, 0, 0};
+  }
+
+  const int kv_lo = kvtile * KVBLK;  // first kv of the block's tile
+  int q_start = 0;
+  if (CAUSAL) q_start = max(0, (kv_lo - (sk - sq)) / QBLK * QBLK);
+
+  for (int hq = hkv * group; hq < (hkv + 1) * group; ++hq) {
+    const unsigned short* qp = q + ((long)bi * nh + hq) * D;
+    const unsigned short* dop = dout + ((long)bi * nh + hq) * D;
+    const float* lse_row = lse + ((long)bi * nh + hq) * sq;
+    const long dr_ss = (long)b * nh;
+    const float* dr_base = drow + (long)bi * nh + hq;
+
+    for (int qt = q_start; qt < sq; qt += QBLK) {
+      // cooperative load Q/dO tiles (row-major + transposed)
+      {
+        constexpr int pieces = QBLK * D / 8 / ATT_BLOCK;
+#pragma unroll
+        for (int pc = 0; pc < pieces; ++pc) {
+          const int idx = (pc * ATT_BLOCK + tid) * 8;
+          const int row = idx / D;
+          const int col = idx % D;
+          const int grow = qt + row;
+          bf16x8 q8 = (grow < sq)
+                          ? *(const bf16x8*)(qp + (long)grow * q_ss + col)
+                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+          *(bf16x8*)((char*)(qr_lds + (long)row * D) + swz(row, col * 2)) = q8;
+          bf16x8 d8 = (grow < sq)
+                          ? *(const bf16x8*)(dop + (long)grow * q_ss + col)
+                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+          *(bf16x8*)((char*)(dor_lds + (long)row * D) + swz(row, col * 2)) = d8;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int d_ = col + j;
+            *(unsigned short*)((char*)(qt_lds + (long)d_ * QBLK) +
+                               swz(d_, row * 2)) = (unsigned short)q8[j];
+            *(unsigned short*)((char*)(dot_lds + (long)d_ * QBLK) +
+                               swz(d_, row * 2)) = (unsigned short)d8[j];
+          }
+        }
+      }
+      __syncthreads();
+
+      // per q subtile: S^T = K.Q^T and dP^T = V.dO^T  (C[m=kv][n=q])
+#pragma unroll
+      for (int qsub = 0; qsub < QBLK / 16; ++qsub) {
+        f32x4 st = f32x4{0, 0, 0, 0};
+        f32x4 dpt = f32x4{0, 0, 0, 0};
+#pragma unroll
+        for (int f = 0; f < DF; ++f) {
+          const int qrow = qsub * 16 + lcol;
+          bf16x8 qb = *(const bf16x8*)((char*)(qr_lds + (long)qrow * D) +
+                                       swz(qrow, (f * 32 + lrow * 8) * 2));
+          st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag[f], qb, st,
+                                                       0, 0, 0);
+          bf16x8 db = *(const bf16x8*)((char*)(dor_lds + (long)qrow * D) +
+                                       swz(qrow, (f * 32 + lrow * 8) * 2));
+          dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[f], db, dpt,
+                                                        0, 0, 0);
+        }
+        // P^T and dS^T in C layout: row=kv (lrow*4+r), col=q (lcol)
+        const int qcol = qt + qsub * 16 + lcol;
+        const float lse_c = (qcol < sq) ? lse_row[qcol] : 0.f;
+        const float dr_c = (qcol < sq) ? dr_base[(long)qcol * dr_ss] : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kvrow = kv0 + lrow * 4 + r;
+          bool valid = (qcol < sq) && (kvrow < sk);
+          if (CAUSAL) valid = valid && (kvrow <= qcol + (sk - sq));
+          float p = valid ? __expf(st[r] * scale - lse_c) : 0.f;
+          float ds_ = p * (dpt[r] - dr_c) * scale;
+          const int prow = wid * QW + lrow * 4 + r;  // kv within block tile
+          const int pcol = qsub * 16 + lcol;         // q within tile
+          // pack p (low) and ds (high) into one 32-bit LDS word; the two
+          // MFMA passes below unpack their half each.
+          *(unsigned*)((char*)(p_lds + 2 * ((long)prow * QBLK + 0)) +
+                       swz(prow, pcol * 4)) =
+              ((unsigned)f2bf(ds_) << 16) | (unsigned)f2bf(p);
+        }
+      }
+      // NOTE: p_lds here holds packed (p, ds) as 32-bit words laid out
+      // [kv][q] with a 4-byte column stride; reads below unpack.
+      __syncthreads();
+
+      // dV += P^T . dO   and   dK += dS^T . Q
+#pragma unroll
+      for (int dsub = 0; dsub < DS_; ++dsub) {
+#pragma unroll
+        for (int qs = 0; qs < QBLK / 32; ++qs) {
+          const int prow = wid * QW + lcol;  // kv index within tile
+          bf16x8 pa, da;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int pcol = qs * 32 + lrow * 8 + j;
+            unsigned packed = *(const unsigned*)(
+                (char*)(p_lds + 2 * ((long)prow * QBLK + 0)) +
+                swz(prow, pcol * 4));
+            pa[j] = (short)(packed & 0xffffu);
+            da[j] = (short)(packed >> 16);
+          }
+          const int drow_ = dsub * 16 + lcol;  // d index
+          bf16x8 dob, qb;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int qcol_ = qs * 32 + lrow * 8 + j;
+            dob[j] = (short)*(const unsigned short*)(
+                (char*)(dot_lds + (long)drow_ * QBLK) + swz(drow_, qcol_ * 2));
+            qb[j] = (short)*(const unsigned short*)(
+                (char*)(qt_lds + (long)drow_ * QBLK) + swz(drow_, qcol_ * 2));
+          }
+          dvacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pa, dob, dvacc[dsub], 0, 0, 0);
+          dkacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              da, qb, dkacc[dsub], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // epilogue: stage dK/dV through qr_lds and store
+  unsigned short* stage = qr_lds + (long)wid * QW * D;
+#pragma unroll
+  for (int dsub = 0; dsub < DS_; ++dsub)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      stage[(long)(lrow * 4 + r) * D + dsub * 16 + lcol] =
+          f2bf(dkacc[dsub][r]);
+  __syncthreads();
+  {
+    constexpr int pieces = QW * D / 8 / WAVE;
+#pragma unroll
+    for (int pc = 0; pc < pieces; ++pc) {
+      const int idx = (pc * WAVE + lane) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int kvrow = kv0 + row;
+      if (kvrow < sk)
+        *(bf16x8*)(dk + ((long)kvrow * b * ng + (long)bi * ng + hkv) * D +
+                   col) = *(const bf16x8*)(stage + (long)row * D + col);
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int dsub = 0; dsub < DS_; ++dsub)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      stage[(long)(lrow * 4 + r) * D + dsub * 16 + lcol] =
+          f2bf(dvacc[dsub][r]);
+  __syncthreads();
+  {
+    constexpr int pieces = QW * D / 8 / WAVE;
+#pragma unroll
+    for (int pc = 0; pc < pieces; ++pc) {
+      const int idx = (pc * WAVE + lane) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int kvrow = kv0 + row;
+      if (kvrow < sk)
+        *(bf16x8*)(dv + ((long)kvrow * b * ng + (long)bi * ng + hkv) * D +
+                   col) = *(const bf16x8*)(stage + (long)row * D + col);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+void launch_attn_bwd(const void* dout, const void* q, const void* k,
+                     const void* v, const void* o, const float* lse,
+                     float* drow, void* dq, void* dk, void* dv, int sq, int sk,
+                     int b, int nh, int ng, int d, float scale, bool causal,
+                     hipStream_t stream) {
+  if (sq % QBLK != 0 || sk % KVBLK != 0)
+    throw std::runtime_error("attn_bwd: sq/sk must be multiples of 64");
+  // Drow
+  {
+    const long rows = (long)sq * b * nh;
+    const int waves_per_block = ATT_BLOCK / WAVE;
+    const long blocks = (rows + waves_per_block - 1) / waves_per_block;
+    hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3((unsigned)blocks),
+                       dim3(ATT_BLOCK), 0, stream,
+                       (const unsigned short*)dout, (const unsigned short*)o,
+                       drow, rows, d);
+    HIP_CHECK_LAUNCH();
+  }
+  const size_t lds_dq = (size_t)(2 * KVBLK * d + d * KVBLK + QBLK * KVBLK) *
+                        sizeof(unsigned short);
+  const size_t lds_dkv = (size_t)(2 * QBLK * d + 2 * d * QBLK) *
+                             sizeof(unsigned short) +
+                         (size_t)KVBLK * QBLK * sizeof(unsigned);
+#define ATT_BWD_LAUNCH(DD, CC)                                                \
+  do {                                                                        \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), dim3(sq / QBLK, b * nh), \
+                       dim3(ATT_BLOCK), lds_dq, stream,                       \
+                       (const unsigned short*)q, (const unsigned short*)k,    \
+                       (const unsigned short*)v,                              \
+                       (const unsigned short*)dout, lse, drow,                \
+                       (unsigned short*)dq, sq, sk, b, nh, ng, scale);        \
+    HIP_CHECK_LAUNCH();                                                       \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>),                         \
+                       dim3(sk / KVBLK, b * ng), dim3(ATT_BLOCK), lds_dkv,    \
+                       stream, (const unsigned short*)q,                      \
+                       (const unsigned short*)k, (const unsigned short*)v,    \
+                       (const unsigned short*)dout, lse, drow,                \
+                       (unsigned short*)dk, (unsigned short*)dv, sq, sk, b,   \
+                       nh, ng, scale);                                        \
+    HIP_CHECK_LAUNCH();                                                       \
+  } while (0)
+  if (d == 128) {
+    if (causal) ATT_BWD_LAUNCH(128, true);
+    else ATT_BWD_LAUNCH(128, false);
+  } else if (d == 64) {
+    if (causal) ATT_BWD_LAUNCH(64, true);
+    else ATT_BWD_LAUNCH(64, false);
+  } else {
+    throw std::runtime_error("attn_bwd: head dim must be 64 or 128");
+  }
+#undef ATT_BWD_LAUNCH
+}

@@ -10,12 +10,12 @@
 void launch_rmsnorm_fwd(const void*, const void*, void*, float*, int, int,
                         float, hipStream_t);
 void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
-                        void*, float*, int, int, hipStream_t);
+                        void*, float*, float*, int, int, int, hipStream_t);
 void launch_layernorm_fwd(const void*, const void*, const void*, void*, float*,
                           float*, int, int, float, hipStream_t);
 void launch_layernorm_bwd(const void*, const void*, const void*, const float*,
-                          const float*, void*, float*, float*, int, int,
-                          hipStream_t);
+                          const float*, void*, float*, float*, float*, float*,
+                          int, int, int, hipStream_t);
 void launch_bias_gelu_fwd(const void*, const void*, void*, long, int,
                           hipStream_t);
 void launch_bias_gelu_bwd(const void*, const void*, const void*, void*, long,
@@ -34,6 +34,12 @@ void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
                         hipStream_t);
 void launch_adamw_flat(float*, const float*, float*, float*, long, float,
                        float, float, float, float, int, hipStream_t);
+void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
+void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
+                     int, int, int, int, int, int, float, bool, hipStream_t);
+void launch_attn_bwd(const void*, const void*, const void*, const void*,
+                     const void*, const float*, float*, void*, void*, void*,
+                     int, int, int, int, int, int, float, bool, hipStream_t);
 
 namespace {
 
@@ -69,10 +75,13 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int H = (int)x.size(-1);
   const long N = x.numel() / H;
   auto dx = torch::empty_like(x);
-  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  const int grid = (int)std::min<long>(N, 512);
+  auto dw_part = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
   launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                      invrms.data_ptr<float>(), dx.data_ptr(),
-                     dw.data_ptr<float>(), (int)N, H, cur_stream());
+                     dw.data_ptr<float>(), dw_part.data_ptr<float>(), grid,
+                     (int)N, H, cur_stream());
   return {dx, dw};
 }
 
@@ -97,12 +106,16 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int H = (int)x.size(-1);
   const long N = x.numel() / H;
   auto dx = torch::empty_like(x);
-  auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
-  auto db = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  auto db = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  const int grid = (int)std::min<long>(N, 512);
+  auto part = torch::empty({2, grid, H}, x.options().dtype(torch::kFloat32));
   launch_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        dx.data_ptr(), dw.data_ptr<float>(),
-                       db.data_ptr<float>(), (int)N, H, cur_stream());
+                       db.data_ptr<float>(), part[0].data_ptr<float>(),
+                       part[1].data_ptr<float>(), grid, (int)N, H,
+                       cur_stream());
   return {dx, dw, db};
 }
 
@@ -225,6 +238,44 @@ torch::Tensor scaled_softmax_bwd(torch::Tensor dy, torch::Tensor p,
   return dx;
 }
 
+// ---------------------------------------------------------------- attention
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, double scale,
+                                    bool causal) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(v, "v");
+  const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
+            d = (int)q.size(3);
+  const int sk = (int)k.size(0), ng = (int)k.size(2);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({b, nh, sq}, q.options().dtype(torch::kFloat32));
+  launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse.data_ptr<float>(), sq, sk, b, nh, ng, d, (float)scale,
+                  causal, cur_stream());
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    double scale, bool causal) {
+  check_bf16(dout, "dout");
+  const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
+            d = (int)q.size(3);
+  const int sk = (int)k.size(0), ng = (int)k.size(2);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto drow = torch::empty({(long)sq * b * nh},
+                           q.options().dtype(torch::kFloat32));
+  launch_attn_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                  o.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                  dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), sq, sk, b, nh,
+                  ng, d, (float)scale, causal, cur_stream());
+  return {dq, dk, dv};
+}
+
 // --------------------------------------------------------------------- adam
 void adamw_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double beta1, double beta2,
@@ -254,4 +305,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
   mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
   mod.def("adamw_flat", &adamw_flat);
+  mod.def("wgrad_accum", &wgrad_accum);
+  mod.def("attn_fwd", &attn_fwd);
+  mod.def("attn_bwd", &attn_bwd);
 }

@@ -55,29 +55,32 @@ __global__ void rmsnorm_fwd_kernel(const unsigned short* __restrict__ x,
 }
 
 // ---------------------------------------------------------------- RMSNorm bwd
+// dw is accumulated into per-block partial buffers (dw_part[grid][H]) and
+// reduced by norm_col_reduce_kernel — fp32 atomics across a 1024-deep grid
+// serialized ~1000x on the same address (13% of step time in the first
+// profile), partials + a tiny second kernel are contention-free.
+template <int CHUNKS>
 __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
                                    const unsigned short* __restrict__ x,
                                    const unsigned short* __restrict__ w,
                                    const float* __restrict__ invrms,
                                    unsigned short* __restrict__ dx,
-                                   float* __restrict__ dw, int N, int H) {
+                                   float* __restrict__ dw_part, int N, int H) {
   __shared__ float lds[BLOCK / WAVE];
-  // per-thread dw partial over its fixed columns
-  constexpr int MAX_COLS = 64;  // supports H <= BLOCK*VEC*MAX_COLS/8...
-  float dwacc[MAX_COLS];
-  const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
-#pragma unroll 4
-  for (int c = 0; c < chunks * VEC && c < MAX_COLS; ++c) dwacc[c] = 0.f;
+  float dwacc[CHUNKS * VEC];
+#pragma unroll
+  for (int c = 0; c < CHUNKS * VEC; ++c) dwacc[c] = 0.f;
 
   for (int row = blockIdx.x; row < N; row += gridDim.x) {
     const unsigned short* xr = x + (long)row * H;
     const unsigned short* dyr = dy + (long)row * H;
     unsigned short* dxr = dx + (long)row * H;
     const float r = invrms[row];
-    // pass 1: dot = sum(dy*w*xhat)
     float dot = 0.f;
-    for (int base = threadIdx.x * VEC, c = 0; base < H;
-         base += BLOCK * VEC, ++c) {
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int base = threadIdx.x * VEC + c * BLOCK * VEC;
+      if (base >= H) break;
       short8v xv = *(const short8v*)(xr + base);
       short8v dv = *(const short8v*)(dyr + base);
       short8v wv = *(const short8v*)(w + base);
@@ -90,8 +93,10 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
       }
     }
     dot = block_reduce_sum<BLOCK>(dot, lds) / H;
-    // pass 2: dx = r * (dy*w - xhat * dot)
-    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int base = threadIdx.x * VEC + c * BLOCK * VEC;
+      if (base >= H) break;
       short8v xv = *(const short8v*)(xr + base);
       short8v dv = *(const short8v*)(dyr + base);
       short8v wv = *(const short8v*)(w + base);
@@ -106,11 +111,24 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
     }
     __syncthreads();
   }
-  // flush dw partials
-  for (int base = threadIdx.x * VEC, c = 0; base < H; base += BLOCK * VEC, ++c)
+  float* dwp = dw_part + (long)blockIdx.x * H;
 #pragma unroll
-    for (int j = 0; j < VEC; ++j)
-      atomicAdd(dw + base + j, dwacc[c * VEC + j]);
+  for (int c = 0; c < CHUNKS; ++c) {
+    const int base = threadIdx.x * VEC + c * BLOCK * VEC;
+    if (base >= H) break;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) dwp[base + j] = dwacc[c * VEC + j];
+  }
+}
+
+// column-reduce [G][H] fp32 partials into [H] (optionally two outputs)
+__global__ void norm_col_reduce_kernel(const float* __restrict__ part,
+                                       float* __restrict__ out, int G, int H) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H) return;
+  float acc = 0.f;
+  for (int g = 0; g < G; ++g) acc += part[(long)g * H + col];
+  out[col] = acc;
 }
 
 // -------------------------------------------------------------- LayerNorm fwd
@@ -161,20 +179,21 @@ __global__ void layernorm_fwd_kernel(const unsigned short* __restrict__ x,
 }
 
 // -------------------------------------------------------------- LayerNorm bwd
+template <int CHUNKS>
 __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
                                      const unsigned short* __restrict__ x,
                                      const unsigned short* __restrict__ w,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      unsigned short* __restrict__ dx,
-                                     float* __restrict__ dw,
-                                     float* __restrict__ db, int N, int H) {
+                                     float* __restrict__ dw_part,
+                                     float* __restrict__ db_part, int N,
+                                     int H) {
   __shared__ float lds[BLOCK / WAVE];
-  constexpr int MAX_COLS = 64;
-  float dwacc[MAX_COLS];
-  float dbacc[MAX_COLS];
-  const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
-  for (int c = 0; c < chunks * VEC && c < MAX_COLS; ++c) {
+  float dwacc[CHUNKS * VEC];
+  float dbacc[CHUNKS * VEC];
+#pragma unroll
+  for (int c = 0; c < CHUNKS * VEC; ++c) {
     dwacc[c] = 0.f;
     dbacc[c] = 0.f;
   }
@@ -186,8 +205,10 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
     const float mu = mean[row];
     const float r = invstd[row];
     float sum1 = 0.f, sum2 = 0.f;
-    for (int base = threadIdx.x * VEC, c = 0; base < H;
-         base += BLOCK * VEC, ++c) {
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int base = threadIdx.x * VEC + c * BLOCK * VEC;
+      if (base >= H) break;
       short8v xv = *(const short8v*)(xr + base);
       short8v dv = *(const short8v*)(dyr + base);
       short8v wv = *(const short8v*)(w + base);
@@ -204,7 +225,10 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
     }
     sum1 = block_reduce_sum<BLOCK>(sum1, lds) / H;
     sum2 = block_reduce_sum<BLOCK>(sum2, lds) / H;
-    for (int base = threadIdx.x * VEC; base < H; base += BLOCK * VEC) {
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int base = threadIdx.x * VEC + c * BLOCK * VEC;
+      if (base >= H) break;
       short8v xv = *(const short8v*)(xr + base);
       short8v dv = *(const short8v*)(dyr + base);
       short8v wv = *(const short8v*)(w + base);
@@ -220,12 +244,18 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
     }
     __syncthreads();
   }
-  for (int base = threadIdx.x * VEC, c = 0; base < H; base += BLOCK * VEC, ++c)
+  float* dwp = dw_part + (long)blockIdx.x * H;
+  float* dbp = db_part + (long)blockIdx.x * H;
+#pragma unroll
+  for (int c = 0; c < CHUNKS; ++c) {
+    const int base = threadIdx.x * VEC + c * BLOCK * VEC;
+    if (base >= H) break;
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
-      atomicAdd(dw + base + j, dwacc[c * VEC + j]);
-      atomicAdd(db + base + j, dbacc[c * VEC + j]);
+      dwp[base + j] = dwacc[c * VEC + j];
+      dbp[base + j] = dbacc[c * VEC + j];
     }
+  }
 }
 
 // ------------------------------------------------------------------ launchers
@@ -240,15 +270,27 @@ void launch_rmsnorm_fwd(const void* x, const void* w, void* y, float* invrms,
 }
 
 void launch_rmsnorm_bwd(const void* dy, const void* x, const void* w,
-                        const float* invrms, void* dx, float* dw, int N, int H,
+                        const float* invrms, void* dx, float* dw,
+                        float* dw_part, int grid, int N, int H,
                         hipStream_t stream) {
   if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
-  if (H > BLOCK * VEC * 8) throw std::runtime_error("H too large for rmsnorm bwd");
-  int grid = N < 1024 ? N : 1024;
-  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(BLOCK), 0, stream,
-                     (const unsigned short*)dy, (const unsigned short*)x,
-                     (const unsigned short*)w, invrms, (unsigned short*)dx, dw,
-                     N, H);
+  const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
+#define RMS_CASE(C)                                                           \
+  case C:                                                                     \
+    hipLaunchKernelGGL(rmsnorm_bwd_kernel<C>, dim3(grid), dim3(BLOCK), 0,     \
+                       stream, (const unsigned short*)dy,                     \
+                       (const unsigned short*)x, (const unsigned short*)w,    \
+                       invrms, (unsigned short*)dx, dw_part, N, H);           \
+    break;
+  switch (chunks) {
+    RMS_CASE(1) RMS_CASE(2) RMS_CASE(3) RMS_CASE(4)
+    default:
+      throw std::runtime_error("H too large for rmsnorm bwd");
+  }
+#undef RMS_CASE
+  HIP_CHECK_LAUNCH();
+  hipLaunchKernelGGL(norm_col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
+                     0, stream, dw_part, dw, grid, H);
   HIP_CHECK_LAUNCH();
 }
 
@@ -266,14 +308,28 @@ void launch_layernorm_fwd(const void* x, const void* w, const void* b, void* y,
 
 void launch_layernorm_bwd(const void* dy, const void* x, const void* w,
                           const float* mean, const float* invstd, void* dx,
-                          float* dw, float* db, int N, int H,
-                          hipStream_t stream) {
+                          float* dw, float* db, float* dw_part, float* db_part,
+                          int grid, int N, int H, hipStream_t stream) {
   if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
-  if (H > BLOCK * VEC * 8) throw std::runtime_error("H too large for layernorm bwd");
-  int grid = N < 1024 ? N : 1024;
-  hipLaunchKernelGGL(layernorm_bwd_kernel, dim3(grid), dim3(BLOCK), 0, stream,
-                     (const unsigned short*)dy, (const unsigned short*)x,
-                     (const unsigned short*)w, mean, invstd,
-                     (unsigned short*)dx, dw, db, N, H);
+  const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
+#define LN_CASE(C)                                                            \
+  case C:                                                                     \
+    hipLaunchKernelGGL(layernorm_bwd_kernel<C>, dim3(grid), dim3(BLOCK), 0,   \
+                       stream, (const unsigned short*)dy,                     \
+                       (const unsigned short*)x, (const unsigned short*)w,    \
+                       mean, invstd, (unsigned short*)dx, dw_part, db_part,   \
+                       N, H);                                                 \
+    break;
+  switch (chunks) {
+    LN_CASE(1) LN_CASE(2) LN_CASE(3) LN_CASE(4)
+    default:
+      throw std::runtime_error("H too large for layernorm bwd");
+  }
+#undef LN_CASE
+  HIP_CHECK_LAUNCH();
+  hipLaunchKernelGGL(norm_col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
+                     0, stream, dw_part, dw, grid, H);
+  hipLaunchKernelGGL(norm_col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
+                     0, stream, db_part, db, grid, H);
   HIP_CHECK_LAUNCH();
 }

@@ -23,6 +23,7 @@ SOURCES = [
     "rope.hip",
     "softmax.hip",
     "adam.hip",
+    "wgrad.cpp",
     "attention.hip",
 ]
 
@@ -41,6 +42,7 @@ def build(verbose: bool = True) -> str:
         build_directory=build_dir,
         extra_cflags=["-O3"],
         extra_cuda_cflags=["-O3", "--offload-arch=gfx950", "-std=c++17"],
+        extra_ldflags=["-L/opt/rocm/lib", "-lhipblaslt"],
         verbose=verbose,
         is_python_module=True,
         keep_intermediates=True,

@@ -223,3 +223,81 @@ def test_model_layer_gpu_matches_cpu_fallback():
     assert logits.shape == (2, 64, 1024)
     assert torch.isfinite(logits.float()).all()
     destroy()
+
+
+def _sdpa_ref(q, k, v, scale, causal):
+    """fp32 full-tensor reference; q [sq,b,nh,d], k/v [sk,b,ng,d]."""
+    sq, b, nh, d = q.shape
+    sk, _, ng, _ = k.shape
+    qf = q.float().permute(1, 2, 0, 3)          # [b,nh,sq,d]
+    kf = k.float().permute(1, 2, 0, 3)
+    vf = v.float().permute(1, 2, 0, 3)
+    if ng != nh:
+        rep = nh // ng
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.triu(torch.ones(sq, sk, dtype=torch.bool,
+                                     device=q.device), 1 + sk - sq)
+        s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    o = torch.matmul(p, vf)                     # [b,nh,sq,d]
+    lse = torch.logsumexp(s, dim=-1)            # [b,nh,sq]
+    return o.permute(2, 0, 1, 3), lse
+
+
+@pytest.mark.parametrize("d,gqa,sq", [(128, 1, 256), (128, 4, 256),
+                                      (64, 1, 128), (128, 1, 2048)])
+def test_flash_attn_fwd(d, gqa, sq):
+    torch.manual_seed(10)
+    b, nh = 2, 8
+    ng = nh // gqa
+    scale = d ** -0.5
+    q = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(sq, b, ng, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(sq, b, ng, d, device="cuda", dtype=torch.bfloat16)
+    o, lse = _ops().attn_fwd(q, k, v, scale, True)
+    o_ref, lse_ref = _sdpa_ref(q, k, v, scale, True)
+    err = (o.float() - o_ref).abs().max().item()
+    assert err < 3e-2, f"fwd max err {err}"
+    lse_err = (lse - lse_ref).abs().max().item()
+    assert lse_err < 1e-2, f"lse max err {lse_err}"
+
+
+@pytest.mark.parametrize("d,gqa", [(128, 1), (128, 4), (64, 1)])
+def test_flash_attn_bwd(d, gqa):
+    torch.manual_seed(11)
+    sq, b, nh = 256, 2, 8
+    ng = nh // gqa
+    scale = d ** -0.5
+    q = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(sq, b, ng, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(sq, b, ng, d, device="cuda", dtype=torch.bfloat16)
+    do = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
+
+    o, lse = _ops().attn_fwd(q, k, v, scale, True)
+    dq, dk, dv = _ops().attn_bwd(do, q, k, v, o, lse, scale, True)
+
+    qr = q.float().clone().requires_grad_(True)
+    kr = k.float().clone().requires_grad_(True)
+    vr = v.float().clone().requires_grad_(True)
+    o_ref, _ = _sdpa_ref(qr, kr, vr, scale, True)
+    o_ref.backward(do.float())
+    for got, ref, name in ((dq, qr.grad, "dq"), (dk, kr.grad, "dk"),
+                           (dv, vr.grad, "dv")):
+        err = (got.float() - ref).abs().max().item()
+        rel = err / (ref.abs().max().item() + 1e-6)
+        assert rel < 5e-2, f"{name} max err {err} rel {rel}"
+
+
+def test_flash_attn_noncausal():
+    torch.manual_seed(12)
+    sq, b, nh, d = 128, 2, 4, 128
+    scale = d ** -0.5
+    q = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
+    o, lse = _ops().attn_fwd(q, k, v, scale, False)
+    o_ref, _ = _sdpa_ref(q, k, v, scale, False)
+    assert (o.float() - o_ref).abs().max().item() < 3e-2
