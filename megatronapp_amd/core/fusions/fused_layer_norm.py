@@ -31,6 +31,16 @@ class _RMSNormFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, invrms = ctx.saved_tensors
         if dy.is_cuda:
+            if hasattr(weight, "main_grad") and \
+                    hasattr(weight, "grad_added_to_main_grad"):
+                # dw accumulates straight into the DDP fp32 grad buffer;
+                # autograd gets a dummy so the bucket hook still fires
+                dx, _ = _ops.get_ops().rmsnorm_bwd(
+                    dy.contiguous(), x, weight, invrms, weight.main_grad)
+                weight.grad_added_to_main_grad = True
+                dw = torch.empty(weight.shape, dtype=weight.dtype,
+                                 device=weight.device)
+                return dx, dw, None
             dx, dw = _ops.get_ops().rmsnorm_bwd(dy.contiguous(), x, weight, invrms)
             return dx, dw.to(weight.dtype), None
         xf = x.float()
@@ -48,6 +58,7 @@ class _RMSNormFn(torch.autograd.Function):
 class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
+        ctx.bias_param = bias
         if x.is_cuda:
             y, mean, invstd = _ops.get_ops().layernorm_fwd(x, weight, bias, eps)
         else:
@@ -63,7 +74,21 @@ class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, mean, invstd = ctx.saved_tensors
+        bias = ctx.bias_param
         if dy.is_cuda:
+            if hasattr(weight, "main_grad") and bias is not None and \
+                    hasattr(bias, "main_grad") and \
+                    hasattr(weight, "grad_added_to_main_grad"):
+                dx, _, _ = _ops.get_ops().layernorm_bwd(
+                    dy.contiguous(), x, weight, mean, invstd,
+                    weight.main_grad, bias.main_grad)
+                weight.grad_added_to_main_grad = True
+                bias.grad_added_to_main_grad = True
+                dw = torch.empty(weight.shape, dtype=weight.dtype,
+                                 device=weight.device)
+                db = torch.empty(bias.shape, dtype=bias.dtype,
+                                 device=bias.device)
+                return dx, dw, db, None
             dx, dw, db = _ops.get_ops().layernorm_bwd(
                 dy.contiguous(), x, weight, mean, invstd)
             return dx, dw.to(weight.dtype), db.to(weight.dtype), None

@@ -79,6 +79,37 @@ def _initialize_affine_weight(weight, init_method, partition_dim, stride=1,
         init_method(weight)
 
 
+class _EmbeddingWithMainGradAccum(torch.autograd.Function):
+    """Embedding lookup whose backward scatter-adds dE straight into the
+    fp32 main_grad buffer (one fused kernel instead of the eager
+    embedding_dense_backward + cast-add pass)."""
+
+    @staticmethod
+    def forward(ctx, masked_input, weight, input_mask):
+        out = F.embedding(masked_input, weight)
+        if input_mask is not None:
+            out = out.clone()
+            out[input_mask, :] = 0.0
+        tokens = masked_input.to(torch.int32)
+        if input_mask is not None:
+            tokens = tokens.masked_fill(input_mask, -1)
+        ctx.save_for_backward(tokens)
+        ctx.weight_ref = weight
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (tokens,) = ctx.saved_tensors
+        weight = ctx.weight_ref
+        _ops.get_ops().embedding_bwd_accum(
+            dy.contiguous().reshape(-1, dy.shape[-1]),
+            tokens.reshape(-1).contiguous(), weight.main_grad)
+        weight.grad_added_to_main_grad = True
+        dummy = torch.empty(weight.shape, dtype=weight.dtype,
+                            device=weight.device)
+        return None, dummy, None
+
+
 class VocabParallelEmbedding(torch.nn.Module):
     """Embedding sharded along the vocab dimension (reference layers.py:172).
 
@@ -111,10 +142,18 @@ class VocabParallelEmbedding(torch.nn.Module):
             masked_input = input_.clone() - self.vocab_start_index
             masked_input[input_mask] = 0
         else:
+            input_mask = None
             masked_input = input_
-        output_parallel = F.embedding(masked_input, self.weight)
-        if self.tensor_model_parallel_size > 1:
-            output_parallel[input_mask, :] = 0.0
+        if (input_.is_cuda and _ops.have_ops() and
+                hasattr(self.weight, "main_grad") and
+                hasattr(self.weight, "grad_added_to_main_grad")):
+            output_parallel = _EmbeddingWithMainGradAccum.apply(
+                masked_input, self.weight, input_mask)
+        else:
+            output_parallel = F.embedding(masked_input, self.weight)
+            if input_mask is not None:
+                output_parallel = output_parallel.clone()
+                output_parallel[input_mask, :] = 0.0
         if self.reduce_scatter_embeddings:
             # [b, s, h] -> [s, b, h] -> [s/tp, b, h]
             output_parallel = output_parallel.transpose(0, 1).contiguous()
@@ -138,6 +177,7 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
     def forward(ctx, input, weight, bias, gradient_accumulation_fusion,
                 async_grad_allreduce, sequence_parallel):
         ctx.use_bias = bias is not None
+        ctx.bias_param = bias
         ctx.gradient_accumulation_fusion = gradient_accumulation_fusion
         ctx.async_grad_allreduce = async_grad_allreduce
         ctx.sequence_parallel = sequence_parallel
@@ -201,7 +241,21 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
                 grad_weight = None
         else:
             grad_weight = grad_output_2d.t().matmul(total_input_2d)
-        grad_bias = grad_output_2d.sum(dim=0) if use_bias else None
+        grad_bias = None
+        if use_bias:
+            bias_param = ctx.bias_param
+            if (grad_output.is_cuda and _ops.have_ops() and
+                    bias_param is not None and
+                    hasattr(bias_param, "main_grad") and
+                    hasattr(bias_param, "grad_added_to_main_grad")):
+                _ops.get_ops().colsum_accum(grad_output_2d.contiguous(),
+                                            bias_param.main_grad)
+                bias_param.grad_added_to_main_grad = True
+                grad_bias = torch.empty(bias_param.shape,
+                                        dtype=bias_param.dtype,
+                                        device=bias_param.device)
+            else:
+                grad_bias = grad_output_2d.sum(dim=0)
 
         if handle is not None:
             handle.wait()

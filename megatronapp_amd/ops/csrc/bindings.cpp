@@ -35,6 +35,9 @@ void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
 void launch_adamw_flat(float*, const float*, float*, float*, long, float,
                        float, float, float, float, int, hipStream_t);
 void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
+void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
+void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
+                                hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, int, float, bool, hipStream_t);
 void launch_attn_bwd(const void*, const void*, const void*, const void*,
@@ -69,13 +72,16 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
 }
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
-                                       torch::Tensor w, torch::Tensor invrms) {
+                                       torch::Tensor w, torch::Tensor invrms,
+                                       c10::optional<torch::Tensor> mg_w) {
   check_bf16(dy, "dy");
   check_bf16(x, "x");
   const int H = (int)x.size(-1);
   const long N = x.numel() / H;
   auto dx = torch::empty_like(x);
-  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  auto dw = mg_w.has_value()
+                ? *mg_w
+                : torch::zeros({H}, x.options().dtype(torch::kFloat32));
   const int grid = (int)std::min<long>(N, 512);
   auto dw_part = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
   launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
@@ -101,13 +107,19 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
 
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor w, torch::Tensor mean,
-                                         torch::Tensor invstd) {
+                                         torch::Tensor invstd,
+                                         c10::optional<torch::Tensor> mg_w,
+                                         c10::optional<torch::Tensor> mg_b) {
   check_bf16(dy, "dy");
   const int H = (int)x.size(-1);
   const long N = x.numel() / H;
   auto dx = torch::empty_like(x);
-  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
-  auto db = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  auto dw = mg_w.has_value()
+                ? *mg_w
+                : torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto db = mg_b.has_value()
+                ? *mg_b
+                : torch::zeros({H}, x.options().dtype(torch::kFloat32));
   const int grid = (int)std::min<long>(N, 512);
   auto part = torch::empty({2, grid, H}, x.options().dtype(torch::kFloat32));
   launch_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
@@ -238,6 +250,30 @@ torch::Tensor scaled_softmax_bwd(torch::Tensor dy, torch::Tensor p,
   return dx;
 }
 
+torch::Tensor colsum_accum(torch::Tensor dy, torch::Tensor out) {
+  check_bf16(dy, "dy");
+  TORCH_CHECK(out.scalar_type() == torch::kFloat32 && out.is_contiguous());
+  const int F = (int)dy.size(-1);
+  const long R = dy.numel() / F;
+  launch_colsum_accum(dy.data_ptr(), out.data_ptr<float>(), R, F,
+                      cur_stream());
+  return out;
+}
+
+void embedding_bwd_accum(torch::Tensor dy, torch::Tensor tokens,
+                         torch::Tensor main_grad) {
+  check_bf16(dy, "dy");
+  TORCH_CHECK(tokens.scalar_type() == torch::kInt32 && tokens.is_contiguous());
+  TORCH_CHECK(main_grad.scalar_type() == torch::kFloat32 &&
+              main_grad.is_contiguous());
+  const int H = (int)dy.size(-1);
+  const long ntok = dy.numel() / H;
+  TORCH_CHECK(tokens.numel() == ntok);
+  launch_embedding_bwd_accum(dy.data_ptr(), tokens.data_ptr<int>(),
+                             main_grad.data_ptr<float>(), ntok, H,
+                             cur_stream());
+}
+
 // ---------------------------------------------------------------- attention
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, double scale,
@@ -291,9 +327,15 @@ void adamw_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);
-  mod.def("rmsnorm_bwd", &rmsnorm_bwd);
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd, pybind11::arg("dy"), pybind11::arg("x"),
+          pybind11::arg("w"), pybind11::arg("invrms"),
+          pybind11::arg("main_grad_w") = pybind11::none());
   mod.def("layernorm_fwd", &layernorm_fwd);
-  mod.def("layernorm_bwd", &layernorm_bwd);
+  mod.def("layernorm_bwd", &layernorm_bwd, pybind11::arg("dy"),
+          pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("mean"),
+          pybind11::arg("invstd"),
+          pybind11::arg("main_grad_w") = pybind11::none(),
+          pybind11::arg("main_grad_b") = pybind11::none());
   mod.def("bias_gelu_fwd", &bias_gelu_fwd);
   mod.def("bias_gelu_bwd", &bias_gelu_bwd);
   mod.def("bias_swiglu_fwd", &bias_swiglu_fwd);
@@ -308,4 +350,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("wgrad_accum", &wgrad_accum);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_bwd", &attn_bwd);
+  mod.def("colsum_accum", &colsum_accum);
+  mod.def("embedding_bwd_accum", &embedding_bwd_accum);
 }

@@ -175,3 +175,59 @@ void launch_bias_swiglu_bwd(const void* dy, const void* x, const void* bias,
                      (unsigned short*)dx, N, F);
   HIP_CHECK_LAUNCH();
 }
+
+
+// ----------------------------------------------------- bias-grad column sum
+// dbias[F] (fp32, += accumulate) from dy [R, F] bf16.  2D grid: x over
+// column chunks, y over row chunks; one atomicAdd per (block, col).
+__global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
+                                    float* __restrict__ out, long R, int F) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= F) return;
+  const long r0 = (long)blockIdx.y * 128;
+  const long r1 = min(R, r0 + 128);
+  float acc = 0.f;
+  for (long r = r0; r < r1; ++r) acc += bf2f(dy[r * F + col]);
+  atomicAdd(out + col, acc);
+}
+
+void launch_colsum_accum(const void* dy, float* out, long R, int F,
+                         hipStream_t s) {
+  dim3 grid((F + 255) / 256, (unsigned)((R + 127) / 128));
+  hipLaunchKernelGGL(colsum_accum_kernel, grid, dim3(256), 0, s,
+                     (const unsigned short*)dy, out, R, F);
+  HIP_CHECK_LAUNCH();
+}
+
+// ------------------------------------------- embedding backward scatter-add
+// dy [Ntok, H] bf16, tokens int32 [Ntok] (already shard-local; -1 = skip),
+// main_grad [V, H] fp32 +=.
+__global__ void embedding_bwd_accum_kernel(const unsigned short* __restrict__ dy,
+                                           const int* __restrict__ tokens,
+                                           float* __restrict__ grad, long ntok,
+                                           int H) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int per_row = H / 4;
+  const long row = i / per_row;
+  if (row >= ntok) return;
+  const int col = (int)(i % per_row) * 4;
+  const int tok = tokens[row];
+  if (tok < 0) return;
+  const unsigned short* src = dy + row * H + col;
+  float* dst = grad + (long)tok * H + col;
+  atomicAdd(dst + 0, bf2f(src[0]));
+  atomicAdd(dst + 1, bf2f(src[1]));
+  atomicAdd(dst + 2, bf2f(src[2]));
+  atomicAdd(dst + 3, bf2f(src[3]));
+}
+
+void launch_embedding_bwd_accum(const void* dy, const int* tokens, float* grad,
+                                long ntok, int H, hipStream_t s) {
+  if (H % 4 != 0) throw std::runtime_error("H must be divisible by 4");
+  const long work = ntok * (H / 4);
+  const long blocks = (work + 255) / 256;
+  hipLaunchKernelGGL(embedding_bwd_accum_kernel, dim3((unsigned)blocks),
+                     dim3(256), 0, s, (const unsigned short*)dy, tokens, grad,
+                     ntok, H);
+  HIP_CHECK_LAUNCH();
+}

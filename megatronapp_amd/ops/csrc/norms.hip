@@ -121,14 +121,19 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
   }
 }
 
-// column-reduce [G][H] fp32 partials into [H] (optionally two outputs)
+// column-reduce [G][H] fp32 partials into [H]: 2D grid (col chunk, g chunk)
+// with one fp32 atomicAdd per (block, col).  out must be pre-zeroed, OR be
+// the param's fp32 main_grad (gradient accumulation lands here directly).
+#define GCHUNK 32
 __global__ void norm_col_reduce_kernel(const float* __restrict__ part,
                                        float* __restrict__ out, int G, int H) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= H) return;
+  const int g0 = blockIdx.y * GCHUNK;
+  const int g1 = min(G, g0 + GCHUNK);
   float acc = 0.f;
-  for (int g = 0; g < G; ++g) acc += part[(long)g * H + col];
-  out[col] = acc;
+  for (int g = g0; g < g1; ++g) acc += part[(long)g * H + col];
+  atomicAdd(out + col, acc);
 }
 
 // -------------------------------------------------------------- LayerNorm fwd
@@ -289,8 +294,9 @@ void launch_rmsnorm_bwd(const void* dy, const void* x, const void* w,
   }
 #undef RMS_CASE
   HIP_CHECK_LAUNCH();
-  hipLaunchKernelGGL(norm_col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
-                     0, stream, dw_part, dw, grid, H);
+  hipLaunchKernelGGL(norm_col_reduce_kernel,
+                     dim3((H + 255) / 256, (grid + GCHUNK - 1) / GCHUNK),
+                     dim3(256), 0, stream, dw_part, dw, grid, H);
   HIP_CHECK_LAUNCH();
 }
 
@@ -327,9 +333,10 @@ void launch_layernorm_bwd(const void* dy, const void* x, const void* w,
   }
 #undef LN_CASE
   HIP_CHECK_LAUNCH();
-  hipLaunchKernelGGL(norm_col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
-                     0, stream, dw_part, dw, grid, H);
-  hipLaunchKernelGGL(norm_col_reduce_kernel, dim3((H + 255) / 256), dim3(256),
-                     0, stream, db_part, db, grid, H);
+  dim3 rg((H + 255) / 256, (grid + GCHUNK - 1) / GCHUNK);
+  hipLaunchKernelGGL(norm_col_reduce_kernel, rg, dim3(256), 0, stream,
+                     dw_part, dw, grid, H);
+  hipLaunchKernelGGL(norm_col_reduce_kernel, rg, dim3(256), 0, stream,
+                     db_part, db, grid, H);
   HIP_CHECK_LAUNCH();
 }
