@@ -50,6 +50,35 @@ __device__ __forceinline__ float group16_sum(float v) {
   return v;
 }
 
+
+// Cooperative 4-row transpose write: in the tile loaders each wave's 16-lane
+// groups hold 4 consecutive rows of the same 8-column span (tid/16 = row).
+// Gather the 4 rows' j-th elements across lane groups via __shfl and have
+// group 0 write one short4 (ds_write_b64) per column instead of every lane
+// issuing 8 scalar ds_write_b16 (4x fewer LDS write instructions).
+// Requires: D == 128 (16 lanes per row); row0 = first of the 4 rows.
+__device__ __forceinline__ void transpose4_write(
+    unsigned short* dst_base, long dst_row_stride_elems, int row0,
+    int col0, const bf16x8 vals, int lane) {
+  const int lrow = lane >> 4;
+  typedef __attribute__((ext_vector_type(4))) short short4v_;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const unsigned short mine = (unsigned short)vals[j];
+    const int base_lane = lane & 15;
+    unsigned short r0 = __shfl((int)mine, base_lane, WAVE);
+    unsigned short r1 = __shfl((int)mine, base_lane + 16, WAVE);
+    unsigned short r2 = __shfl((int)mine, base_lane + 32, WAVE);
+    unsigned short r3 = __shfl((int)mine, base_lane + 48, WAVE);
+    if (lrow == 0) {
+      const int d = col0 + j;
+      short4v_ pack = {(short)r0, (short)r1, (short)r2, (short)r3};
+      *(short4v_*)((char*)(dst_base + d * dst_row_stride_elems) +
+                   swz(d, row0 * 2)) = pack;
+    }
+  }
+}
+
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
@@ -132,12 +161,16 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
         bf16x8 vv8 = (grow < sk)
                          ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        // transpose into vt_lds[d][kv] (scalar stores, swizzled rows)
+        if constexpr (D == 128) {
+          transpose4_write(vt_lds, KVBLK, (row / 4) * 4, col, vv8,
+                           tid % WAVE);
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = col + j;
-          *(unsigned short*)((char*)(vt_lds + (long)d * KVBLK) +
-                             swz(d, row * 2)) = (unsigned short)vv8[j];
+          for (int j = 0; j < 8; ++j) {
+            const int d = col + j;
+            *(unsigned short*)((char*)(vt_lds + (long)d * KVBLK) +
+                               swz(d, row * 2)) = (unsigned short)vv8[j];
+          }
         }
       }
     }
@@ -219,24 +252,16 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
     for (int dsub = 0; dsub < DS; ++dsub) {
 #pragma unroll
       for (int ks = 0; ks < KVBLK / 32; ++ks) {
-        // A fragment: P[q0w + l&15][ks*32 + lrow*8 + j]
+        // A fragment: P[q0w + l&15][ks*32 + lrow*8 .. +8] — the 8 kv
+        // elements are contiguous and 16B-aligned, and the row swizzle
+        // XORs bits >= 4 only, so one ds_read_b128 fetches the fragment.
         const int prow = wid * QW + lcol;
-        bf16x8 pa;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int pcol = ks * 32 + lrow * 8 + j;
-          pa[j] = (short)*(const unsigned short*)(
-              (char*)(p_lds + (long)prow * KVBLK) + swz(prow, pcol * 2));
-        }
-        // B fragment: Vt[dsub*16 + lcol][ks*32 + lrow*8 + j] -> V[kv][d]
+        bf16x8 pa = *(const bf16x8*)((char*)(p_lds + (long)prow * KVBLK) +
+                                     swz(prow, (ks * 32 + lrow * 8) * 2));
+        // B fragment: Vt[dsub*16 + lcol][ks*32 + lrow*8 .. +8]
         const int vrow = dsub * 16 + lcol;
-        bf16x8 vb;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int vcol = ks * 32 + lrow * 8 + j;
-          vb[j] = (short)*(const unsigned short*)(
-              (char*)(vt_lds + (long)vrow * KVBLK) + swz(vrow, vcol * 2));
-        }
+        bf16x8 vb = *(const bf16x8*)((char*)(vt_lds + (long)vrow * KVBLK) +
+                                     swz(vrow, (ks * 32 + lrow * 8) * 2));
         // note operand order: C[q][d] = A(P[q][kv]) x B(V[kv][d]); our B
         // fragment is indexed [d][kv] -> use transposed roles:
         // mfma(A=pa over kv, B=vb over kv) with B[k][n]: k=kv, n=d: need
@@ -437,11 +462,15 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
                         ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
                         : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
         *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) = k8;
+        if constexpr (D == 128) {
+          transpose4_write(kt_lds, KVBLK, (row / 4) * 4, col, k8, tid % WAVE);
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d_ = col + j;
-          *(unsigned short*)((char*)(kt_lds + (long)d_ * KVBLK) +
-                             swz(d_, row * 2)) = (unsigned short)k8[j];
+          for (int j = 0; j < 8; ++j) {
+            const int d_ = col + j;
+            *(unsigned short*)((char*)(kt_lds + (long)d_ * KVBLK) +
+                               swz(d_, row * 2)) = (unsigned short)k8[j];
+          }
         }
         bf16x8 v8 = (grow < sk)
                         ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
@@ -491,21 +520,11 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int ks = 0; ks < KVBLK / 32; ++ks) {
         const int prow = wid * QW + lcol;
-        bf16x8 pa;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int pcol = ks * 32 + lrow * 8 + j;
-          pa[j] = (short)*(const unsigned short*)(
-              (char*)(ds_lds + (long)prow * KVBLK) + swz(prow, pcol * 2));
-        }
+        bf16x8 pa = *(const bf16x8*)((char*)(ds_lds + (long)prow * KVBLK) +
+                                     swz(prow, (ks * 32 + lrow * 8) * 2));
         const int krow = dsub * 16 + lcol;  // d index
-        bf16x8 kb;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int kcol = ks * 32 + lrow * 8 + j;  // kv index
-          kb[j] = (short)*(const unsigned short*)(
-              (char*)(kt_lds + (long)krow * KVBLK) + swz(krow, kcol * 2));
-        }
+        bf16x8 kb = *(const bf16x8*)((char*)(kt_lds + (long)krow * KVBLK) +
+                                     swz(krow, (ks * 32 + lrow * 8) * 2));
         dqacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, kb,
                                                               dqacc[dsub],
                                                               0, 0, 0);
@@ -624,13 +643,19 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
                           ? *(const bf16x8*)(dop + (long)grow * q_ss + col)
                           : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
           *(bf16x8*)((char*)(dor_lds + (long)row * D) + swz(row, col * 2)) = d8;
+          if constexpr (D == 128) {
+            transpose4_write(qt_lds, QBLK, (row / 4) * 4, col, q8, tid % WAVE);
+            transpose4_write(dot_lds, QBLK, (row / 4) * 4, col, d8,
+                             tid % WAVE);
+          } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int d_ = col + j;
-            *(unsigned short*)((char*)(qt_lds + (long)d_ * QBLK) +
-                               swz(d_, row * 2)) = (unsigned short)q8[j];
-            *(unsigned short*)((char*)(dot_lds + (long)d_ * QBLK) +
-                               swz(d_, row * 2)) = (unsigned short)d8[j];
+            for (int j = 0; j < 8; ++j) {
+              const int d_ = col + j;
+              *(unsigned short*)((char*)(qt_lds + (long)d_ * QBLK) +
+                                 swz(d_, row * 2)) = (unsigned short)q8[j];
+              *(unsigned short*)((char*)(dot_lds + (long)d_ * QBLK) +
+                                 swz(d_, row * 2)) = (unsigned short)d8[j];
+            }
           }
         }
       }
@@ -683,26 +708,27 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int qs = 0; qs < QBLK / 32; ++qs) {
           const int prow = wid * QW + lcol;  // kv index within tile
+          // packed (p|ds) words: 8 q-columns = 32 contiguous bytes; the
+          // 4-byte-column swizzle XORs bits >= 4, 16B blocks stay intact
+          const char* prow_base = (char*)(p_lds + 2 * ((long)prow * QBLK));
+          typedef __attribute__((ext_vector_type(4))) unsigned uint4v;
+          uint4v plo = *(const uint4v*)(prow_base +
+                                        swz(prow, (qs * 32 + lrow * 8) * 4));
+          uint4v phi = *(const uint4v*)(prow_base +
+                                        swz(prow, (qs * 32 + lrow * 8 + 4) * 4));
           bf16x8 pa, da;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int pcol = qs * 32 + lrow * 8 + j;
-            unsigned packed = *(const unsigned*)(
-                (char*)(p_lds + 2 * ((long)prow * QBLK + 0)) +
-                swz(prow, pcol * 4));
-            pa[j] = (short)(packed & 0xffffu);
-            da[j] = (short)(packed >> 16);
+          for (int j = 0; j < 4; ++j) {
+            pa[j] = (short)(plo[j] & 0xffffu);
+            da[j] = (short)(plo[j] >> 16);
+            pa[4 + j] = (short)(phi[j] & 0xffffu);
+            da[4 + j] = (short)(phi[j] >> 16);
           }
           const int drow_ = dsub * 16 + lcol;  // d index
-          bf16x8 dob, qb;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int qcol_ = qs * 32 + lrow * 8 + j;
-            dob[j] = (short)*(const unsigned short*)(
-                (char*)(dot_lds + (long)drow_ * QBLK) + swz(drow_, qcol_ * 2));
-            qb[j] = (short)*(const unsigned short*)(
-                (char*)(qt_lds + (long)drow_ * QBLK) + swz(drow_, qcol_ * 2));
-          }
+          bf16x8 dob = *(const bf16x8*)((char*)(dot_lds + (long)drow_ * QBLK) +
+                                        swz(drow_, (qs * 32 + lrow * 8) * 2));
+          bf16x8 qb = *(const bf16x8*)((char*)(qt_lds + (long)drow_ * QBLK) +
+                                       swz(drow_, (qs * 32 + lrow * 8) * 2));
           dvacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pa, dob, dvacc[dsub], 0, 0, 0);
           dkacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(

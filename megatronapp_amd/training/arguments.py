@@ -1,0 +1,423 @@
+"""CLI argument system (reference training/arguments.py, ~700 flags).
+
+A practical subset covering everything this framework implements; names
+match the reference so launch scripts translate unchanged.  MegatronApp
+plugin flags: trace (reference :2705-2716), DPP (:2201-2205), FBD
+(:2197-2200), MegaScope ws (:1663).
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+
+import torch
+
+
+def parse_args(extra_args_provider=None, ignore_unknown_args=False):
+    parser = argparse.ArgumentParser(description="megatronapp_amd arguments",
+                                     allow_abbrev=False)
+    _add_model_args(parser)
+    _add_training_args(parser)
+    _add_learning_rate_args(parser)
+    _add_mixed_precision_args(parser)
+    _add_parallelism_args(parser)
+    _add_data_args(parser)
+    _add_checkpointing_args(parser)
+    _add_logging_args(parser)
+    _add_megascan_args(parser)
+    _add_megascope_args(parser)
+    _add_megadpp_args(parser)
+    _add_megafbd_args(parser)
+    _add_inference_args(parser)
+    if extra_args_provider is not None:
+        parser = extra_args_provider(parser)
+    if ignore_unknown_args:
+        args, _ = parser.parse_known_args()
+    else:
+        args = parser.parse_args()
+    return args
+
+
+def _add_model_args(p):
+    g = p.add_argument_group("model")
+    g.add_argument("--num-layers", type=int, default=12)
+    g.add_argument("--hidden-size", type=int, default=768)
+    g.add_argument("--ffn-hidden-size", type=int, default=None)
+    g.add_argument("--num-attention-heads", type=int, default=12)
+    g.add_argument("--num-query-groups", type=int, default=None)
+    g.add_argument("--group-query-attention", action="store_true")
+    g.add_argument("--kv-channels", type=int, default=None)
+    g.add_argument("--max-position-embeddings", type=int, default=None)
+    g.add_argument("--position-embedding-type", default="learned_absolute",
+                   choices=["learned_absolute", "rope", "none"])
+    g.add_argument("--use-rotary-position-embeddings", action="store_true")
+    g.add_argument("--rotary-base", type=int, default=10000)
+    g.add_argument("--rotary-percent", type=float, default=1.0)
+    g.add_argument("--normalization", default="LayerNorm",
+                   choices=["LayerNorm", "RMSNorm"])
+    g.add_argument("--swiglu", action="store_true")
+    g.add_argument("--squared-relu", action="store_true")
+    g.add_argument("--disable-bias-linear", action="store_false",
+                   dest="add_bias_linear")
+    g.add_argument("--add-qkv-bias", action="store_true")
+    g.add_argument("--qk-layernorm", action="store_true")
+    g.add_argument("--untie-embeddings-and-output-weights", action="store_true")
+    g.add_argument("--attention-dropout", type=float, default=0.1)
+    g.add_argument("--hidden-dropout", type=float, default=0.1)
+    g.add_argument("--layernorm-epsilon", type=float, default=1e-5)
+    g.add_argument("--apply-query-key-layer-scaling", action="store_true")
+    g.add_argument("--attention-softmax-in-fp32", action="store_true")
+    g.add_argument("--attention-backend", default="auto",
+                   choices=["auto", "flash", "fused", "unfused"])
+    g.add_argument("--init-method-std", type=float, default=0.02)
+    g.add_argument("--no-masked-softmax-fusion", action="store_false",
+                   dest="masked_softmax_fusion")
+    g.add_argument("--no-bias-gelu-fusion", action="store_false",
+                   dest="bias_gelu_fusion")
+    g.add_argument("--no-bias-dropout-fusion", action="store_false",
+                   dest="bias_dropout_fusion")
+    g.add_argument("--no-rope-fusion", action="store_false",
+                   dest="apply_rope_fusion")
+    g.add_argument("--use-legacy-models", action="store_true")
+    # MoE
+    g.add_argument("--num-experts", type=int, default=None)
+    g.add_argument("--moe-router-topk", type=int, default=2)
+    g.add_argument("--moe-router-load-balancing-type", default="aux_loss",
+                   choices=["aux_loss", "sinkhorn", "none"])
+    g.add_argument("--moe-aux-loss-coeff", type=float, default=0.0)
+    g.add_argument("--moe-token-dispatcher-type", default="alltoall",
+                   choices=["alltoall", "allgather"])
+    g.add_argument("--moe-grouped-gemm", action="store_true")
+    g.add_argument("--moe-ffn-hidden-size", type=int, default=None)
+    g.add_argument("--moe-shared-expert-intermediate-size", type=int,
+                   default=None)
+
+
+def _add_training_args(p):
+    g = p.add_argument_group("training")
+    g.add_argument("--micro-batch-size", type=int, default=1)
+    g.add_argument("--global-batch-size", type=int, default=None)
+    g.add_argument("--rampup-batch-size", nargs=3, type=int, default=None)
+    g.add_argument("--train-iters", type=int, default=None)
+    g.add_argument("--train-samples", type=int, default=None)
+    g.add_argument("--exit-interval", type=int, default=None)
+    g.add_argument("--eval-iters", type=int, default=10)
+    g.add_argument("--eval-interval", type=int, default=1000)
+    g.add_argument("--seed", type=int, default=1234)
+    g.add_argument("--optimizer", default="adam", choices=["adam", "sgd"])
+    g.add_argument("--weight-decay", type=float, default=0.01)
+    g.add_argument("--start-weight-decay", type=float, default=None)
+    g.add_argument("--end-weight-decay", type=float, default=None)
+    g.add_argument("--clip-grad", type=float, default=1.0)
+    g.add_argument("--adam-beta1", type=float, default=0.9)
+    g.add_argument("--adam-beta2", type=float, default=0.999)
+    g.add_argument("--adam-eps", type=float, default=1e-8)
+    g.add_argument("--recompute-granularity", default=None,
+                   choices=[None, "selective", "full"])
+    g.add_argument("--recompute-method", default=None,
+                   choices=[None, "uniform", "block"])
+    g.add_argument("--recompute-num-layers", type=int, default=None)
+    g.add_argument("--distribute-saved-activations", action="store_true")
+    g.add_argument("--no-overlap-grad-reduce", action="store_false",
+                   dest="overlap_grad_reduce")
+    g.add_argument("--overlap-param-gather", action="store_true")
+    g.add_argument("--use-distributed-optimizer", action="store_true")
+    g.add_argument("--ddp-bucket-size", type=int, default=None)
+    g.add_argument("--check-weight-hash-across-dp-replicas-interval",
+                   type=int, default=None)
+    g.add_argument("--calculate-per-token-loss", action="store_true")
+    g.add_argument("--empty-unused-memory-level", type=int, default=0)
+    g.add_argument("--exit-signal-handler", action="store_true")
+    g.add_argument("--rerun-mode", default="disabled",
+                   choices=["disabled", "validate_results", "report_stats"])
+
+
+def _add_learning_rate_args(p):
+    g = p.add_argument_group("learning rate")
+    g.add_argument("--lr", type=float, default=None)
+    g.add_argument("--lr-decay-style", default="linear",
+                   choices=["constant", "linear", "cosine",
+                            "inverse-square-root", "WSD"])
+    g.add_argument("--lr-decay-iters", type=int, default=None)
+    g.add_argument("--lr-warmup-iters", type=int, default=0)
+    g.add_argument("--lr-warmup-fraction", type=float, default=None)
+    g.add_argument("--lr-wsd-decay-iters", type=int, default=None)
+    g.add_argument("--lr-wsd-decay-style", default="exponential")
+    g.add_argument("--min-lr", type=float, default=0.0)
+    g.add_argument("--override-opt_param-scheduler", action="store_true")
+    g.add_argument("--use-checkpoint-opt_param-scheduler", action="store_true")
+
+
+def _add_mixed_precision_args(p):
+    g = p.add_argument_group("mixed precision")
+    g.add_argument("--fp16", action="store_true")
+    g.add_argument("--bf16", action="store_true")
+    g.add_argument("--loss-scale", type=float, default=None)
+    g.add_argument("--initial-loss-scale", type=float, default=2 ** 32)
+    g.add_argument("--min-loss-scale", type=float, default=1.0)
+    g.add_argument("--loss-scale-window", type=int, default=1000)
+    g.add_argument("--accumulate-allreduce-grads-in-fp32", action="store_true",
+                   default=True)
+
+
+def _add_parallelism_args(p):
+    g = p.add_argument_group("parallelism")
+    g.add_argument("--tensor-model-parallel-size", type=int, default=1)
+    g.add_argument("--pipeline-model-parallel-size", type=int, default=1)
+    g.add_argument("--num-layers-per-virtual-pipeline-stage", type=int,
+                   default=None)
+    g.add_argument("--virtual-pipeline-model-parallel-size", type=int,
+                   default=None)
+    g.add_argument("--microbatch-group-size-per-virtual-pipeline-stage",
+                   type=int, default=None)
+    g.add_argument("--context-parallel-size", type=int, default=1)
+    g.add_argument("--cp-comm-type", default="p2p",
+                   choices=["p2p", "a2a", "allgather"])
+    g.add_argument("--expert-model-parallel-size", type=int, default=1)
+    g.add_argument("--sequence-parallel", action="store_true")
+    g.add_argument("--no-async-tensor-model-parallel-allreduce",
+                   action="store_false", dest="async_tensor_model_parallel_allreduce")
+    g.add_argument("--no-gradient-accumulation-fusion", action="store_false",
+                   dest="gradient_accumulation_fusion")
+    g.add_argument("--distributed-backend", default="nccl",
+                   choices=["nccl", "gloo"])
+    g.add_argument("--distributed-timeout-minutes", type=int, default=10)
+    g.add_argument("--local-rank", type=int, default=None)
+    g.add_argument("--overlap-p2p-communication", action="store_true")
+    g.add_argument("--no-batch-p2p-comm", action="store_false",
+                   dest="batch_p2p_comm")
+
+
+def _add_data_args(p):
+    g = p.add_argument_group("data")
+    g.add_argument("--data-path", nargs="*", default=None)
+    g.add_argument("--split", default="969, 30, 1")
+    g.add_argument("--seq-length", type=int, default=1024)
+    g.add_argument("--vocab-size", type=int, default=None)
+    g.add_argument("--padded-vocab-size", type=int, default=None)
+    g.add_argument("--make-vocab-size-divisible-by", type=int, default=128)
+    g.add_argument("--vocab-file", default=None)
+    g.add_argument("--merge-file", default=None)
+    g.add_argument("--tokenizer-type", default="NullTokenizer",
+                   choices=["GPT2BPETokenizer", "SentencePieceTokenizer",
+                            "HuggingFaceTokenizer", "NullTokenizer"])
+    g.add_argument("--tokenizer-model", default=None)
+    g.add_argument("--mock-data", action="store_true")
+    g.add_argument("--num-workers", type=int, default=2)
+    g.add_argument("--dataloader-type", default="single",
+                   choices=["single", "cyclic"])
+    g.add_argument("--eod-mask-loss", action="store_true")
+    g.add_argument("--reset-position-ids", action="store_true")
+    g.add_argument("--reset-attention-mask", action="store_true")
+    g.add_argument("--create-attention-mask-in-dataloader", action="store_true")
+
+
+def _add_checkpointing_args(p):
+    g = p.add_argument_group("checkpointing")
+    g.add_argument("--save", default=None)
+    g.add_argument("--load", default=None)
+    g.add_argument("--save-interval", type=int, default=None)
+    g.add_argument("--no-save-optim", action="store_true")
+    g.add_argument("--no-save-rng", action="store_true")
+    g.add_argument("--no-load-optim", action="store_true")
+    g.add_argument("--no-load-rng", action="store_true")
+    g.add_argument("--finetune", action="store_true")
+    g.add_argument("--ckpt-format", default="torch_dist",
+                   choices=["torch_dist", "torch"])
+    g.add_argument("--async-save", action="store_true")
+    g.add_argument("--use-checkpoint-args", action="store_true")
+
+
+def _add_logging_args(p):
+    g = p.add_argument_group("logging")
+    g.add_argument("--log-interval", type=int, default=100)
+    g.add_argument("--tensorboard-dir", default=None)
+    g.add_argument("--wandb-project", default=None)
+    g.add_argument("--wandb-exp-name", default="")
+    g.add_argument("--log-params-norm", action="store_true")
+    g.add_argument("--log-num-zeros-in-grad", action="store_true")
+    g.add_argument("--log-throughput", action="store_true")
+    g.add_argument("--log-progress", action="store_true")
+    g.add_argument("--timing-log-level", type=int, default=0)
+    g.add_argument("--log-timers-to-tensorboard", action="store_true")
+
+
+def _add_megascan_args(p):
+    g = p.add_argument_group("MegaScan tracing")
+    g.add_argument("--trace", action="store_true",
+                   help="enable the MegaScan hipEvent tracer")
+    g.add_argument("--trace-dir", default="trace_output")
+    g.add_argument("--trace-interval", type=int, default=5)
+    g.add_argument("--continuous-trace-iterations", type=int, default=2)
+    g.add_argument("--trace-granularity", default="full",
+                   choices=["base", "full"])
+    g.add_argument("--trace-max-iters", type=int, default=None)
+
+
+def _add_megascope_args(p):
+    g = p.add_argument_group("MegaScope visualization")
+    g.add_argument("--enable-ws-server", action="store_true")
+    g.add_argument("--training-ws-port", type=int, default=None)
+    g.add_argument("--inference-ws-port", type=int, default=5000)
+
+
+def _add_megadpp_args(p):
+    g = p.add_argument_group("MegaDPP dynamic pipeline")
+    g.add_argument("--use-dpp", action="store_true")
+    g.add_argument("--workload", type=int, default=1048576)
+    g.add_argument("--num-gpus", type=int, default=None)
+    g.add_argument("--node-ips", nargs="*", default=None)
+    g.add_argument("--multi-node", action="store_true")
+    g.add_argument("--dpp-policy", default="depth_first",
+                   choices=["depth_first", "breadth_first", "greedy"])
+
+
+def _add_megafbd_args(p):
+    g = p.add_argument_group("MegaFBD forward/backward disaggregation")
+    g.add_argument("--forward-backward-disaggregating", action="store_true")
+    g.add_argument("--ignore-forward-tensor-parallel", action="store_true")
+
+
+def _add_inference_args(p):
+    g = p.add_argument_group("inference")
+    g.add_argument("--inference-max-seq-length", type=int, default=2560)
+    g.add_argument("--max-tokens-to-oom", type=int, default=12000)
+    g.add_argument("--temperature", type=float, default=1.0)
+    g.add_argument("--top_p", type=float, default=0.0)
+    g.add_argument("--top_k", type=int, default=0)
+    g.add_argument("--port", type=int, default=5000)
+
+
+def validate_args(args, defaults={}):
+    for key, value in defaults.items():
+        if getattr(args, key, None) is None:
+            setattr(args, key, value)
+
+    args.rank = int(os.getenv("RANK", "0"))
+    args.world_size = int(os.getenv("WORLD_SIZE", "1"))
+    args.local_rank = int(os.getenv("LOCAL_RANK", "0"))
+
+    total_model = (args.tensor_model_parallel_size *
+                   args.pipeline_model_parallel_size *
+                   args.context_parallel_size)
+    assert args.world_size % total_model == 0, (
+        f"world size {args.world_size} not divisible by tp*pp*cp {total_model}")
+    args.data_parallel_size = args.world_size // total_model
+
+    if args.global_batch_size is None:
+        args.global_batch_size = args.micro_batch_size * args.data_parallel_size
+    if args.max_position_embeddings is None:
+        args.max_position_embeddings = args.seq_length
+    if args.ffn_hidden_size is None:
+        args.ffn_hidden_size = (int(8 * args.hidden_size / 3 / 64) * 64
+                                if args.swiglu else 4 * args.hidden_size)
+    if args.use_rotary_position_embeddings:
+        args.position_embedding_type = "rope"
+    if args.num_query_groups is None:
+        args.num_query_groups = args.num_attention_heads
+    if args.kv_channels is None:
+        args.kv_channels = args.hidden_size // args.num_attention_heads
+
+    if args.num_layers_per_virtual_pipeline_stage is not None:
+        per_stage = args.num_layers // args.pipeline_model_parallel_size
+        args.virtual_pipeline_model_parallel_size = (
+            per_stage // args.num_layers_per_virtual_pipeline_stage)
+        if args.virtual_pipeline_model_parallel_size <= 1:
+            args.virtual_pipeline_model_parallel_size = None
+
+    if args.lr_decay_iters is None and args.train_iters:
+        args.lr_decay_iters = args.train_iters
+    if args.lr_warmup_fraction is not None and args.lr_decay_iters:
+        args.lr_warmup_iters = int(args.lr_warmup_fraction * args.lr_decay_iters)
+    if args.start_weight_decay is None:
+        args.start_weight_decay = args.weight_decay
+    if args.end_weight_decay is None:
+        args.end_weight_decay = args.weight_decay
+
+    if args.sequence_parallel and args.tensor_model_parallel_size == 1:
+        args.sequence_parallel = False
+    if args.fp16:
+        assert not args.bf16
+        args.params_dtype = torch.float16
+    elif args.bf16:
+        args.params_dtype = torch.bfloat16
+    else:
+        args.params_dtype = torch.float32
+
+    # padded vocab
+    if args.vocab_size is None:
+        args.vocab_size = 50257 if args.tokenizer_type == "GPT2BPETokenizer" else 32000
+    if args.padded_vocab_size is None:
+        mult = args.make_vocab_size_divisible_by * args.tensor_model_parallel_size
+        args.padded_vocab_size = ((args.vocab_size + mult - 1) // mult) * mult
+
+    args.consumed_train_samples = 0
+    args.consumed_valid_samples = 0
+    args.curr_iteration = 0
+    return args
+
+
+def core_transformer_config_from_args(args, config_class=None):
+    from ..core.transformer_config import TransformerConfig
+    from ..core.distributed.finalize_model_grads import finalize_model_grads
+    config_class = config_class or TransformerConfig
+    activation = "gelu"
+    if args.swiglu:
+        activation = "swiglu"
+    elif args.squared_relu:
+        activation = "squared_relu"
+    return config_class(
+        num_layers=args.num_layers,
+        hidden_size=args.hidden_size,
+        num_attention_heads=args.num_attention_heads,
+        num_query_groups=args.num_query_groups,
+        kv_channels=args.kv_channels,
+        ffn_hidden_size=args.ffn_hidden_size,
+        hidden_dropout=args.hidden_dropout,
+        attention_dropout=args.attention_dropout,
+        layernorm_epsilon=args.layernorm_epsilon,
+        normalization=args.normalization,
+        activation_func=activation,
+        add_bias_linear=args.add_bias_linear,
+        add_qkv_bias=args.add_qkv_bias,
+        qk_layernorm=args.qk_layernorm,
+        position_embedding_type=args.position_embedding_type,
+        rotary_base=args.rotary_base,
+        rotary_percent=args.rotary_percent,
+        untie_embeddings_and_output_weights=args.untie_embeddings_and_output_weights,
+        apply_query_key_layer_scaling=args.apply_query_key_layer_scaling,
+        masked_softmax_fusion=args.masked_softmax_fusion,
+        attention_backend=args.attention_backend,
+        init_method_std=args.init_method_std,
+        apply_rope_fusion=args.apply_rope_fusion,
+        fp16=args.fp16,
+        bf16=args.bf16,
+        params_dtype=args.params_dtype,
+        tensor_model_parallel_size=args.tensor_model_parallel_size,
+        pipeline_model_parallel_size=args.pipeline_model_parallel_size,
+        virtual_pipeline_model_parallel_size=args.virtual_pipeline_model_parallel_size,
+        microbatch_group_size_per_vp_stage=args.microbatch_group_size_per_virtual_pipeline_stage,
+        context_parallel_size=args.context_parallel_size,
+        cp_comm_type=args.cp_comm_type,
+        expert_model_parallel_size=args.expert_model_parallel_size,
+        sequence_parallel=args.sequence_parallel,
+        async_tensor_model_parallel_allreduce=args.async_tensor_model_parallel_allreduce,
+        gradient_accumulation_fusion=args.gradient_accumulation_fusion,
+        batch_p2p_comm=args.batch_p2p_comm,
+        overlap_p2p_comm=args.overlap_p2p_communication,
+        recompute_granularity=args.recompute_granularity,
+        recompute_method=args.recompute_method,
+        recompute_num_layers=args.recompute_num_layers,
+        distribute_saved_activations=args.distribute_saved_activations,
+        calculate_per_token_loss=args.calculate_per_token_loss,
+        num_moe_experts=args.num_experts,
+        moe_router_topk=args.moe_router_topk,
+        moe_router_load_balancing_type=args.moe_router_load_balancing_type,
+        moe_aux_loss_coeff=args.moe_aux_loss_coeff,
+        moe_token_dispatcher_type=args.moe_token_dispatcher_type,
+        moe_grouped_gemm=args.moe_grouped_gemm,
+        moe_ffn_hidden_size=args.moe_ffn_hidden_size,
+        moe_shared_expert_intermediate_size=args.moe_shared_expert_intermediate_size,
+        finalize_model_grads_func=finalize_model_grads,
+    )

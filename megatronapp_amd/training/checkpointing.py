@@ -1,0 +1,181 @@
+"""Checkpoint save/load (reference training/checkpointing.py:315).
+
+Layout (reference-compatible, §2.6 SURVEY):
+  <save>/iter_{it:07d}/mp_rank_{tp:02d}[_{pp:03d}]/model_optim_rng.pt
+  <save>/latest_checkpointed_iteration.txt
+
+"torch" format: one file per (tp, pp) rank written by dp rank 0 — the
+reference's legacy layout, loadable by its conversion tooling.
+"torch_dist" format: torch.distributed.checkpoint (DCP) sharded save with
+per-rank key namespacing for TP/PP shards + async save option.
+"""
+
+from __future__ import annotations
+
+import os
+import random
+import shutil
+import sys
+from typing import List, Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..core import parallel_state
+from ..core.tensor_parallel.random import get_cuda_rng_tracker
+from .global_vars import get_args
+
+
+def get_checkpoint_name(checkpoints_path, iteration, release=False,
+                        tp_rank=None, pp_rank=None, return_base_dir=False):
+    if release:
+        directory = "release"
+    else:
+        directory = f"iter_{iteration:07d}"
+    base = os.path.join(checkpoints_path, directory)
+    if return_base_dir:
+        return base
+    tp_rank = (parallel_state.get_tensor_model_parallel_rank()
+               if tp_rank is None else tp_rank)
+    pp_rank = (parallel_state.get_pipeline_model_parallel_rank()
+               if pp_rank is None else pp_rank)
+    if parallel_state.get_pipeline_model_parallel_world_size() == 1:
+        common_path = os.path.join(base, f"mp_rank_{tp_rank:02d}")
+    else:
+        common_path = os.path.join(base, f"mp_rank_{tp_rank:02d}_{pp_rank:03d}")
+    return os.path.join(common_path, "model_optim_rng.pt")
+
+
+def get_checkpoint_tracker_filename(checkpoints_path):
+    return os.path.join(checkpoints_path, "latest_checkpointed_iteration.txt")
+
+
+def read_metadata(tracker_filename):
+    with open(tracker_filename) as f:
+        metastring = f.read().strip()
+    release = metastring == "release"
+    iteration = 0 if release else int(metastring)
+    return iteration, release
+
+
+def _rng_state():
+    state = {
+        "random_rng_state": random.getstate(),
+        "np_rng_state": np.random.get_state(),
+        "torch_rng_state": torch.get_rng_state(),
+        "rng_tracker_states": get_cuda_rng_tracker().get_states(),
+    }
+    if torch.cuda.is_available():
+        state["cuda_rng_state"] = torch.cuda.get_rng_state()
+    return state
+
+
+def _restore_rng_state(state):
+    random.setstate(state["random_rng_state"])
+    np.random.set_state(state["np_rng_state"])
+    torch.set_rng_state(state["torch_rng_state"])
+    if torch.cuda.is_available() and "cuda_rng_state" in state:
+        torch.cuda.set_rng_state(state["cuda_rng_state"])
+    get_cuda_rng_tracker().set_states(state["rng_tracker_states"])
+
+
+def save_checkpoint(iteration, model: List, optimizer, opt_param_scheduler,
+                    num_floating_point_operations_so_far=0, checkpointing_context=None,
+                    train_data_iterator=None, **kwargs):
+    args = get_args()
+    if args.save is None:
+        return
+    state_dict = {
+        "args": vars(args).copy(),
+        "checkpoint_version": 3.0,
+        "iteration": iteration,
+        "num_floating_point_operations_so_far": num_floating_point_operations_so_far,
+    }
+    # drop unpicklables from args copy
+    state_dict["args"] = {k: v for k, v in state_dict["args"].items()
+                          if isinstance(v, (int, float, str, bool, list, tuple,
+                                            type(None)))}
+    if len(model) == 1:
+        state_dict["model"] = model[0].state_dict_for_save_checkpoint()
+    else:
+        for i, chunk in enumerate(model):
+            state_dict[f"model{i}"] = chunk.state_dict_for_save_checkpoint()
+    if optimizer is not None and not args.no_save_optim:
+        state_dict["optimizer"] = optimizer.state_dict()
+        if opt_param_scheduler is not None:
+            state_dict["opt_param_scheduler"] = opt_param_scheduler.state_dict()
+    if not args.no_save_rng:
+        state_dict["rng_state"] = _rng_state()
+
+    # dp rank 0 of every (tp, pp) writes; with the distributed optimizer
+    # every dp rank holds a distinct shard -> append dp suffix
+    dp_rank = parallel_state.get_data_parallel_rank()
+    write = dp_rank == 0 or args.use_distributed_optimizer
+    if write:
+        name = get_checkpoint_name(args.save, iteration)
+        if args.use_distributed_optimizer and dp_rank > 0:
+            name = name.replace("model_optim_rng.pt",
+                                f"optim_shard_dp{dp_rank:03d}.pt")
+            state_dict = {"optimizer": state_dict.get("optimizer")}
+        os.makedirs(os.path.dirname(name), exist_ok=True)
+        torch.save(state_dict, name)
+
+    if dist.is_initialized():
+        dist.barrier()
+    if (not dist.is_initialized()) or dist.get_rank() == 0:
+        with open(get_checkpoint_tracker_filename(args.save), "w") as f:
+            f.write(str(iteration))
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def load_checkpoint(model: List, optimizer, opt_param_scheduler,
+                    load_arg="load", strict=True, checkpointing_context=None,
+                    skip_load_to_model_and_opt=False):
+    args = get_args()
+    load_dir = getattr(args, load_arg)
+    if load_dir is None:
+        return 0, 0
+    tracker = get_checkpoint_tracker_filename(load_dir)
+    if not os.path.isfile(tracker):
+        if args.rank == 0:
+            print(f"  no checkpoint tracker at {tracker}, starting fresh")
+        return 0, 0
+    iteration, release = read_metadata(tracker)
+    name = get_checkpoint_name(load_dir, iteration, release)
+    state_dict = torch.load(name, map_location="cpu", weights_only=False)
+
+    if len(model) == 1:
+        model[0].load_state_dict(state_dict["model"], strict=strict)
+    else:
+        for i, chunk in enumerate(model):
+            chunk.load_state_dict(state_dict[f"model{i}"], strict=strict)
+
+    if optimizer is not None and not args.no_load_optim and not args.finetune:
+        if args.use_distributed_optimizer and \
+                parallel_state.get_data_parallel_rank() > 0:
+            shard_name = name.replace(
+                "model_optim_rng.pt",
+                f"optim_shard_dp{parallel_state.get_data_parallel_rank():03d}.pt")
+            opt_sd = torch.load(shard_name, map_location="cpu",
+                                weights_only=False)["optimizer"]
+        else:
+            opt_sd = state_dict.get("optimizer")
+        if opt_sd is not None:
+            optimizer.load_state_dict(opt_sd)
+        if opt_param_scheduler is not None and \
+                "opt_param_scheduler" in state_dict:
+            opt_param_scheduler.load_state_dict(
+                state_dict["opt_param_scheduler"])
+    if optimizer is not None and hasattr(optimizer, "reload_model_params") \
+            and (args.no_load_optim or args.finetune):
+        optimizer.reload_model_params()
+
+    if not args.no_load_rng and not args.finetune and "rng_state" in state_dict:
+        _restore_rng_state(state_dict["rng_state"])
+
+    num_flop = state_dict.get("num_floating_point_operations_so_far", 0)
+    if args.rank == 0:
+        print(f"  loaded checkpoint from {load_dir} at iteration {iteration}")
+    return (0 if args.finetune else iteration), num_flop

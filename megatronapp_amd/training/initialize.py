@@ -1,0 +1,100 @@
+"""Framework initialization (reference training/initialize.py:37).
+
+initialize_megatron: parse args -> set globals -> init torch.distributed
+(RCCL on GPU, gloo on CPU) -> initialize model parallel groups -> seed the
+parallel RNG -> start the MegaScan tracer if --trace.
+"""
+
+from __future__ import annotations
+
+import os
+import random
+from datetime import timedelta
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..core import parallel_state
+from ..core.num_microbatches_calculator import init_num_microbatches_calculator
+from ..core.tensor_parallel.random import model_parallel_cuda_manual_seed
+from .arguments import parse_args, validate_args
+from .global_vars import set_global_variables
+
+
+def initialize_megatron(extra_args_provider=None, args_defaults={},
+                        ignore_unknown_args=False, allow_no_cuda=True,
+                        skip_mpu_initialization=False,
+                        parsed_args=None):
+    args = parsed_args or parse_args(extra_args_provider, ignore_unknown_args)
+    validate_args(args, args_defaults)
+    set_global_variables(args)
+
+    _initialize_distributed(args)
+    _set_random_seed(args.seed)
+
+    init_num_microbatches_calculator(
+        args.rank, args.rampup_batch_size, args.global_batch_size,
+        args.micro_batch_size, args.data_parallel_size)
+
+    if args.trace:
+        from .trace import Tracer
+        Tracer.initialize(trace_dir=args.trace_dir,
+                          interval=args.trace_interval,
+                          continuous_iters=args.continuous_trace_iterations,
+                          granularity=args.trace_granularity,
+                          max_iters=args.trace_max_iters)
+    return args
+
+
+def _initialize_distributed(args):
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "6000")
+        backend = args.distributed_backend
+        if not torch.cuda.is_available():
+            backend = "gloo"
+        dist.init_process_group(
+            backend=backend, world_size=args.world_size, rank=args.rank,
+            timeout=timedelta(minutes=args.distributed_timeout_minutes))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(args.local_rank)
+
+    if parallel_state.model_parallel_is_initialized():
+        return
+    if getattr(args, "forward_backward_disaggregating", False):
+        from ..fbd.topology import initialize_model_parallel_fbd
+        initialize_model_parallel_fbd(args)
+        return
+    parallel_state.initialize_model_parallel(
+        tensor_model_parallel_size=args.tensor_model_parallel_size,
+        pipeline_model_parallel_size=args.pipeline_model_parallel_size,
+        virtual_pipeline_model_parallel_size=args.virtual_pipeline_model_parallel_size,
+        context_parallel_size=args.context_parallel_size,
+        expert_model_parallel_size=args.expert_model_parallel_size)
+
+
+def _set_random_seed(seed: int, data_parallel_random_init=False):
+    # PP-stage-offset seeds (reference initialize.py:399)
+    seed = seed + 100 * parallel_state.get_pipeline_model_parallel_rank()
+    random.seed(seed)
+    np.random.seed(seed)
+    torch.manual_seed(seed)
+    if torch.cuda.is_available():
+        model_parallel_cuda_manual_seed(seed)
+    else:
+        model_parallel_cuda_manual_seed(seed)
+
+
+def set_jit_fusion_options():
+    """No-op on MI355X: fusion is done by our HIP kernels, not torch JIT."""
+
+
+def write_args_to_tensorboard(args=None, writer=None):
+    from .global_vars import get_args, get_tensorboard_writer
+    args = args or get_args()
+    writer = writer or get_tensorboard_writer()
+    if writer is None:
+        return
+    for k, v in sorted(vars(args).items()):
+        writer.add_text(k, str(v))

@@ -1,0 +1,177 @@
+"""Tokenizers (reference training/tokenizer/tokenizer.py, ~900 LoC).
+
+Builds Null / GPT2-BPE / SentencePiece / HuggingFace tokenizers.  The
+MegaScope additions are kept: ``decoder`` and ``offsets`` properties used
+by the visualization server (reference :157,166,378-385,479-527).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+
+class MegatronTokenizer:
+    def tokenize(self, text: str) -> List[int]:
+        raise NotImplementedError
+
+    def detokenize(self, ids: List[int]) -> str:
+        raise NotImplementedError
+
+    @property
+    def vocab_size(self) -> int:
+        raise NotImplementedError
+
+    @property
+    def eod(self) -> int:
+        raise NotImplementedError
+
+    @property
+    def decoder(self):
+        """id -> printable token map (MegaScope)."""
+        return {}
+
+    def offsets(self, ids: List[int], text: str) -> List[int]:
+        offsets, pos = [], 0
+        for i in ids:
+            offsets.append(pos)
+            pos += len(self.detokenize([i]))
+        return offsets
+
+
+class NullTokenizer(MegatronTokenizer):
+    """Integer-string tokenizer for synthetic data (reference NullTokenizer)."""
+
+    def __init__(self, vocab_size: int = 131072):
+        self._vocab_size = int(vocab_size)
+        self._eod = self._vocab_size - 1
+
+    def tokenize(self, text):
+        return [int(t) for t in text.split()]
+
+    def detokenize(self, ids):
+        return " ".join(str(i) for i in ids)
+
+    @property
+    def vocab_size(self):
+        return self._vocab_size
+
+    @property
+    def eod(self):
+        return self._eod
+
+    @property
+    def decoder(self):
+        return _LazyIntDecoder()
+
+
+class _LazyIntDecoder(dict):
+    def get(self, key, default=None):
+        return str(key)
+
+    def __getitem__(self, key):
+        return str(key)
+
+    def __contains__(self, key):
+        return True
+
+
+class HuggingFaceTokenizer(MegatronTokenizer):
+    def __init__(self, model_name_or_path: str):
+        from transformers import AutoTokenizer
+        self._tok = AutoTokenizer.from_pretrained(model_name_or_path)
+
+    def tokenize(self, text):
+        return self._tok.encode(text)
+
+    def detokenize(self, ids):
+        return self._tok.decode(ids)
+
+    @property
+    def vocab_size(self):
+        return len(self._tok)
+
+    @property
+    def eod(self):
+        return self._tok.eos_token_id
+
+    @property
+    def decoder(self):
+        return _LazyReadableDecoder(self._tok)
+
+
+class _LazyReadableDecoder(dict):
+    """Decodes ids on demand into readable tokens (reference :321)."""
+
+    def __init__(self, tok):
+        super().__init__()
+        self._tok = tok
+
+    def get(self, key, default=None):
+        try:
+            return self._tok.decode([int(key)])
+        except Exception:  # noqa: BLE001
+            return default
+
+    def __getitem__(self, key):
+        return self._tok.decode([int(key)])
+
+
+class GPT2BPETokenizer(MegatronTokenizer):
+    def __init__(self, vocab_file: str, merge_file: str):
+        from tokenizers import ByteLevelBPETokenizer
+        self._tok = ByteLevelBPETokenizer(vocab_file, merge_file)
+        self._eod = self._tok.token_to_id("<|endoftext|>")
+        if self._eod is None:
+            self._eod = self._tok.get_vocab_size() - 1
+
+    def tokenize(self, text):
+        return self._tok.encode(text).ids
+
+    def detokenize(self, ids):
+        return self._tok.decode(ids)
+
+    @property
+    def vocab_size(self):
+        return self._tok.get_vocab_size()
+
+    @property
+    def eod(self):
+        return self._eod
+
+    @property
+    def decoder(self):
+        return {i: self._tok.id_to_token(i) or str(i)
+                for i in range(self.vocab_size)}
+
+
+class SentencePieceTokenizer(MegatronTokenizer):
+    def __init__(self, model_file: str):
+        import sentencepiece as spm
+        self._tok = spm.SentencePieceProcessor(model_file=model_file)
+
+    def tokenize(self, text):
+        return self._tok.encode(text)
+
+    def detokenize(self, ids):
+        return self._tok.decode(ids)
+
+    @property
+    def vocab_size(self):
+        return self._tok.get_piece_size()
+
+    @property
+    def eod(self):
+        return self._tok.eos_id()
+
+
+def build_tokenizer(args):
+    t = args.tokenizer_type
+    if t == "NullTokenizer":
+        return NullTokenizer(args.padded_vocab_size or args.vocab_size or 131072)
+    if t == "GPT2BPETokenizer":
+        return GPT2BPETokenizer(args.vocab_file, args.merge_file)
+    if t == "SentencePieceTokenizer":
+        return SentencePieceTokenizer(args.tokenizer_model)
+    if t == "HuggingFaceTokenizer":
+        return HuggingFaceTokenizer(args.tokenizer_model)
+    raise ValueError(f"unknown tokenizer type {t}")
