@@ -86,12 +86,16 @@ class FusedScaleMaskSoftmax(nn.Module):
         self.softmax_in_fp32 = softmax_in_fp32
         self.scale = scale
 
-    def forward(self, input, mask):
-        # input: [b, np, sq, sk]
+    def forward(self, input, mask, attn_mask_type=None):
+        # input: [b, np, sq, sk]; attn_mask_type overrides the constructed
+        # default (decode steps pass no_mask: the single query row attends
+        # to the whole KV prefix)
+        mask_type = attn_mask_type if attn_mask_type is not None \
+            else self.attn_mask_type
         scale = self.scale if self.scale is not None else 1.0
         if self.fusion and self.input_in_float16:
             b, np_, sq, sk = input.shape
-            if self.attn_mask_type == AttnMaskType.causal and sq == sk:
+            if mask_type == AttnMaskType.causal and sq == sk:
                 probs = ScaledUpperTriangMaskedSoftmax.apply(
                     input.view(-1, sq, sk), scale)
                 return probs.view(b, np_, sq, sk)
@@ -102,10 +106,13 @@ class FusedScaleMaskSoftmax(nn.Module):
             input = input.float()
         if self.scale is not None:
             input = input * self.scale
-        if self.attn_mask_type == AttnMaskType.causal and mask is None:
+        if mask_type == AttnMaskType.causal and mask is None:
             sq, sk = input.shape[-2], input.shape[-1]
+            # causal with a KV prefix: query row i is global position
+            # i + (sk - sq)
             mask = torch.triu(torch.ones(sq, sk, dtype=torch.bool,
-                                         device=input.device), diagonal=1)
+                                         device=input.device),
+                              diagonal=1 + sk - sq)
             mask = mask.view(1, 1, sq, sk)
         if mask is not None:
             input = self.mask_func(input, mask)

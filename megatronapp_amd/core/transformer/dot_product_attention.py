@@ -82,7 +82,15 @@ class DotProductAttention(nn.Module):
         scores = scores.view(b, np_, sq, sk)
 
         mask_type = attn_mask_type if attn_mask_type is not None else self.attn_mask_type
-        probs = self.scale_mask_softmax(scores, attention_mask)
+        probs = self.scale_mask_softmax(scores, attention_mask, mask_type)
+
+        # MegaScope raw-attention-score tap (reference
+        # dot_product_attention.py:168-170)
+        from ..tensor_tracer import FlagType, get_tensor_tracers
+        tt = get_tensor_tracers()
+        if tt is not None and tt.enabled(FlagType.RawAttentionScore,
+                                         self.layer_number):
+            tt.report(FlagType.RawAttentionScore, self.layer_number, probs)
         if self.config.attention_dropout > 0 or self.training:
             from ..tensor_parallel.random import get_cuda_rng_tracker
             if torch.cuda.is_available():

@@ -161,7 +161,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
         bf16x8 vv8 = (grow < sk)
                          ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        if constexpr (D == 128) {
+        if constexpr (false) {  // shuffle-transpose measured SLOWER than scalar
           transpose4_write(vt_lds, KVBLK, (row / 4) * 4, col, vv8,
                            tid % WAVE);
         } else {
@@ -462,7 +462,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
                         ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
                         : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
         *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) = k8;
-        if constexpr (D == 128) {
+        if constexpr (false) {
           transpose4_write(kt_lds, KVBLK, (row / 4) * 4, col, k8, tid % WAVE);
         } else {
 #pragma unroll
@@ -643,7 +643,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
                           ? *(const bf16x8*)(dop + (long)grow * q_ss + col)
                           : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
           *(bf16x8*)((char*)(dor_lds + (long)row * D) + swz(row, col * 2)) = d8;
-          if constexpr (D == 128) {
+          if constexpr (false) {
             transpose4_write(qt_lds, QBLK, (row / 4) * 4, col, q8, tid % WAVE);
             transpose4_write(dot_lds, QBLK, (row / 4) * 4, col, d8,
                              tid % WAVE);
