@@ -76,6 +76,31 @@ def forward_backward_pipelining_with_interleaving(
     def is_first_microbatch_for_model_chunk(microbatch_id: int) -> bool:
         return microbatch_id < group_size
 
+    # ---- MegaDPP tags: (chunk, within-chunk microbatch), with the
+    # producer's chunk id on wrap-around hops (first rank's chunk c input
+    # is the LAST rank's chunk c-1 output, and symmetrically for grads).
+    def _wc(k: int) -> int:
+        return (k // (group_size * num_model_chunks)) * group_size + \
+            k % group_size
+
+    def fwd_send_tag(k):
+        return (get_model_chunk_id(k, True), _wc(k))
+
+    def fwd_recv_tag(k):
+        c = get_model_chunk_id(k, True)
+        if pp_rank == 0:
+            c -= 1  # wrap: produced by the last rank's previous chunk
+        return (c, _wc(k))
+
+    def bwd_send_tag(k):
+        return (get_model_chunk_id(k, False), _wc(k))
+
+    def bwd_recv_tag(k):
+        c = get_model_chunk_id(k, False)
+        if pp_rank == pp_size - 1:
+            c += 1  # wrap: produced by rank 0's next chunk
+        return (c, _wc(k))
+
     def forward_step_helper(microbatch_id: int):
         chunk = get_model_chunk_id(microbatch_id, forward=True)
         parallel_state.set_virtual_pipeline_model_parallel_rank(chunk)
@@ -121,6 +146,7 @@ def forward_backward_pipelining_with_interleaving(
     num_remaining = total_num_microbatches - num_warmup
 
     parallel_state.set_virtual_pipeline_model_parallel_rank(0)
+    p2p_communication.set_dpp_tags(fwd_recv=fwd_recv_tag(0))
     input_tensors[0].append(p2p_communication.recv_forward(
         tensor_shape, config, parallel_state.is_pipeline_first_stage()))
 
@@ -139,11 +165,16 @@ def forward_backward_pipelining_with_interleaving(
         if parallel_state.is_pipeline_last_stage():
             output_tensor = None
 
+        cur_chunk = get_model_chunk_id(k, forward=True)
         if (k == num_warmup - 1 and not forward_only and not all_warmup):
             input_tensor_grad = None
             recv_next = True
             if parallel_state.is_pipeline_last_stage(ignore_virtual=True):
                 recv_next = False
+            p2p_communication.set_dpp_tags(
+                fwd_send=fwd_send_tag(k),
+                fwd_recv=fwd_recv_tag(k + 1),
+                bwd_recv=bwd_recv_tag(0))
             (input_tensor, output_tensor_grad) = (
                 p2p_communication.send_forward_backward_recv_forward_backward(
                     output_tensor, input_tensor_grad, recv_prev=recv_prev,
@@ -152,6 +183,9 @@ def forward_backward_pipelining_with_interleaving(
             if recv_next:
                 output_tensor_grads[num_model_chunks - 1].append(output_tensor_grad)
         else:
+            p2p_communication.set_dpp_tags(
+                fwd_send=fwd_send_tag(k),
+                fwd_recv=fwd_recv_tag(k + 1))
             input_tensor = p2p_communication.send_forward_recv_forward(
                 output_tensor, recv_prev, tensor_shape, config)
         if recv_prev:
@@ -187,6 +221,11 @@ def forward_backward_pipelining_with_interleaving(
         if forward_k == total_num_microbatches - 1:
             recv_prev = False
 
+        p2p_communication.set_dpp_tags(
+            fwd_send=fwd_send_tag(forward_k),
+            bwd_send=bwd_send_tag(backward_k),
+            fwd_recv=fwd_recv_tag(forward_k + 1),
+            bwd_recv=bwd_recv_tag(backward_k + 1))
         (input_tensor, output_tensor_grad) = (
             p2p_communication.send_forward_backward_recv_forward_backward(
                 output_tensor, input_tensor_grad, recv_prev=recv_prev,
@@ -200,6 +239,7 @@ def forward_backward_pipelining_with_interleaving(
     # ---- cooldown backwards ----
     if not forward_only:
         if all_warmup:
+            p2p_communication.set_dpp_tags(bwd_recv=bwd_recv_tag(0))
             output_tensor_grads[num_model_chunks - 1].append(
                 p2p_communication.recv_backward(
                     tensor_shape, config,
@@ -217,6 +257,9 @@ def forward_backward_pipelining_with_interleaving(
             parallel_state.set_virtual_pipeline_model_parallel_rank(backward_chunk)
             if parallel_state.is_pipeline_first_stage():
                 input_tensor_grad = None
+            p2p_communication.set_dpp_tags(
+                bwd_send=bwd_send_tag(k),
+                bwd_recv=bwd_recv_tag(k + 1))
             output_tensor_grad = p2p_communication.send_backward_recv_backward(
                 input_tensor_grad, recv_next, tensor_shape, config)
             if recv_next:

@@ -23,6 +23,35 @@ import torch.distributed as dist
 from .. import parallel_state
 from ..trace_hooks import trace_collective, trace_scope
 
+# ---------------------------------------------------------------------------
+# MegaDPP integration: when a transport is active, pipeline sends go through
+# the tagged shm mailbox (sender-side policy ordering) instead of RCCL p2p.
+# The schedules set the (chunk, microbatch) tags before each call.
+# ---------------------------------------------------------------------------
+import threading as _threading
+
+_DPP_TAGS = _threading.local()
+
+
+def set_dpp_tags(fwd_send=None, fwd_recv=None, bwd_send=None, bwd_recv=None):
+    _DPP_TAGS.fwd_send = fwd_send
+    _DPP_TAGS.fwd_recv = fwd_recv
+    _DPP_TAGS.bwd_send = bwd_send
+    _DPP_TAGS.bwd_recv = bwd_recv
+
+
+def _tag(name):
+    return getattr(_DPP_TAGS, name, None)
+
+
+def _dpp():
+    from ...dpp.transport import get_transport
+    return get_transport()
+
+
+def _dpp_dtype(config):
+    return config.pipeline_dtype or torch.float32
+
 
 def _shape_numel(shape):
     n = 1
@@ -175,6 +204,11 @@ def _p2p_peers():
 def recv_forward(tensor_shape, config, is_first_stage: bool):
     if is_first_stage:
         return None
+    dpp = _dpp()
+    if dpp is not None and _tag("fwd_recv") is not None:
+        with trace_scope("recv-forward"):
+            return dpp.recv_forward(tensor_shape, _dpp_dtype(config),
+                                    *_tag("fwd_recv"))
     with trace_scope("recv-forward"):
         input_tensor, _, _ = _communicate(
             tensor_send_next=None, tensor_send_prev=None, recv_prev=True,
@@ -185,6 +219,11 @@ def recv_forward(tensor_shape, config, is_first_stage: bool):
 def recv_backward(tensor_shape, config, is_last_stage: bool):
     if is_last_stage:
         return None
+    dpp = _dpp()
+    if dpp is not None and _tag("bwd_recv") is not None:
+        with trace_scope("recv-backward"):
+            return dpp.recv_backward(tensor_shape, _dpp_dtype(config),
+                                     *_tag("bwd_recv"))
     with trace_scope("recv-backward"):
         _, output_tensor_grad, _ = _communicate(
             tensor_send_next=None, tensor_send_prev=None, recv_prev=False,
@@ -195,6 +234,11 @@ def recv_backward(tensor_shape, config, is_last_stage: bool):
 def send_forward(output_tensor, config, is_last_stage: bool):
     if is_last_stage:
         return
+    dpp = _dpp()
+    if dpp is not None and _tag("fwd_send") is not None:
+        with trace_scope("send-forward"):
+            dpp.send_forward(output_tensor, *_tag("fwd_send"))
+        return
     with trace_scope("send-forward", data=output_tensor.numel() * output_tensor.element_size(),
                      group=_p2p_peers()):
         _communicate(tensor_send_next=output_tensor, tensor_send_prev=None,
@@ -204,6 +248,11 @@ def send_forward(output_tensor, config, is_last_stage: bool):
 
 def send_backward(input_tensor_grad, config, is_first_stage: bool):
     if is_first_stage:
+        return
+    dpp = _dpp()
+    if dpp is not None and _tag("bwd_send") is not None:
+        with trace_scope("send-backward"):
+            dpp.send_backward(input_tensor_grad, *_tag("bwd_send"))
         return
     with trace_scope("send-backward", data=input_tensor_grad.numel() * input_tensor_grad.element_size(),
                      group=_p2p_peers()):
@@ -216,6 +265,11 @@ def send_forward_recv_backward(output_tensor, tensor_shape, config,
                                is_last_stage: bool):
     if is_last_stage:
         return None
+    dpp = _dpp()
+    if dpp is not None and _tag("fwd_send") is not None:
+        dpp.send_forward(output_tensor, *_tag("fwd_send"))
+        return dpp.recv_backward(tensor_shape, _dpp_dtype(config),
+                                 *_tag("bwd_recv"))
     with trace_scope("send-forward-recv-backward",
                      data=output_tensor.numel() * output_tensor.element_size(),
                      group=_p2p_peers()):
@@ -230,6 +284,13 @@ def send_backward_recv_forward(input_tensor_grad, tensor_shape, config,
                                is_first_stage: bool):
     if is_first_stage:
         return None
+    dpp = _dpp()
+    if dpp is not None and _tag("bwd_send") is not None:
+        dpp.send_backward(input_tensor_grad, *_tag("bwd_send"))
+        if _tag("fwd_recv") is None:
+            return None
+        return dpp.recv_forward(tensor_shape, _dpp_dtype(config),
+                                *_tag("fwd_recv"))
     with trace_scope("send-backward-recv-forward",
                      data=input_tensor_grad.numel() * input_tensor_grad.element_size(),
                      group=_p2p_peers()):
@@ -242,6 +303,18 @@ def send_backward_recv_forward(input_tensor_grad, tensor_shape, config,
 
 def send_forward_recv_forward(output_tensor, recv_prev, tensor_shape, config,
                               overlap_p2p_comm=False):
+    dpp = _dpp()
+    if dpp is not None and (_tag("fwd_send") is not None or
+                            (recv_prev and _tag("fwd_recv") is not None)):
+        if output_tensor is not None and _tag("fwd_send") is not None:
+            dpp.send_forward(output_tensor, *_tag("fwd_send"))
+        result = None
+        if recv_prev and _tag("fwd_recv") is not None:
+            result = dpp.recv_forward(tensor_shape, _dpp_dtype(config),
+                                      *_tag("fwd_recv"))
+        if overlap_p2p_comm:
+            return result, None
+        return result
     with trace_scope("exchange-next"):
         input_tensor, _, wait_handles = _communicate(
             tensor_send_next=output_tensor, tensor_send_prev=None,
@@ -254,6 +327,18 @@ def send_forward_recv_forward(output_tensor, recv_prev, tensor_shape, config,
 
 def send_backward_recv_backward(input_tensor_grad, recv_next, tensor_shape,
                                 config, overlap_p2p_comm=False):
+    dpp = _dpp()
+    if dpp is not None and (_tag("bwd_send") is not None or
+                            (recv_next and _tag("bwd_recv") is not None)):
+        if input_tensor_grad is not None and _tag("bwd_send") is not None:
+            dpp.send_backward(input_tensor_grad, *_tag("bwd_send"))
+        result = None
+        if recv_next and _tag("bwd_recv") is not None:
+            result = dpp.recv_backward(tensor_shape, _dpp_dtype(config),
+                                       *_tag("bwd_recv"))
+        if overlap_p2p_comm:
+            return result, None
+        return result
     with trace_scope("exchange-prev"):
         _, output_tensor_grad, wait_handles = _communicate(
             tensor_send_next=None, tensor_send_prev=input_tensor_grad,
@@ -267,6 +352,21 @@ def send_backward_recv_backward(input_tensor_grad, recv_next, tensor_shape,
 def send_forward_backward_recv_forward_backward(
         output_tensor, input_tensor_grad, recv_prev, recv_next, tensor_shape,
         config):
+    dpp = _dpp()
+    if dpp is not None and _tag("fwd_send") is not None or \
+            dpp is not None and _tag("bwd_send") is not None:
+        if output_tensor is not None and _tag("fwd_send") is not None:
+            dpp.send_forward(output_tensor, *_tag("fwd_send"))
+        if input_tensor_grad is not None and _tag("bwd_send") is not None:
+            dpp.send_backward(input_tensor_grad, *_tag("bwd_send"))
+        it = ot = None
+        if recv_prev and _tag("fwd_recv") is not None:
+            it = dpp.recv_forward(tensor_shape, _dpp_dtype(config),
+                                  *_tag("fwd_recv"))
+        if recv_next and _tag("bwd_recv") is not None:
+            ot = dpp.recv_backward(tensor_shape, _dpp_dtype(config),
+                                   *_tag("bwd_recv"))
+        return it, ot
     input_tensor, output_tensor_grad, _ = _communicate(
         tensor_send_next=output_tensor, tensor_send_prev=input_tensor_grad,
         recv_prev=recv_prev, recv_next=recv_next, tensor_shape=tensor_shape,

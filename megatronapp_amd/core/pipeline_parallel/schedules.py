@@ -256,6 +256,7 @@ def forward_backward_pipelining_without_interleaving(
 
     # --- warmup forwards ---
     for i in range(num_warmup):
+        p2p_communication.set_dpp_tags(fwd_recv=(0, i), fwd_send=(0, i))
         input_tensor = p2p_communication.recv_forward(tensor_shape, config, is_first)
         output_tensor = _fwd(input_tensor, i)
         p2p_communication.send_forward(output_tensor, config, is_last)
@@ -266,17 +267,22 @@ def forward_backward_pipelining_without_interleaving(
 
     # --- steady 1F1B ---
     if num_steady > 0:
+        p2p_communication.set_dpp_tags(fwd_recv=(0, num_warmup))
         input_tensor = p2p_communication.recv_forward(tensor_shape, config, is_first)
     for i in range(num_steady):
         last_iteration = (i == num_steady - 1)
-        output_tensor = _fwd(input_tensor, num_warmup + i)
+        f_mb = num_warmup + i
+        output_tensor = _fwd(input_tensor, f_mb)
         if forward_only:
+            p2p_communication.set_dpp_tags(fwd_send=(0, f_mb),
+                                           fwd_recv=(0, f_mb + 1))
             p2p_communication.send_forward(output_tensor, config, is_last)
             if not last_iteration:
                 input_tensor = p2p_communication.recv_forward(
                     tensor_shape, config, is_first)
             continue
 
+        p2p_communication.set_dpp_tags(fwd_send=(0, f_mb), bwd_recv=(0, i))
         output_tensor_grad = p2p_communication.send_forward_recv_backward(
             output_tensor, tensor_shape, config, is_last)
         input_tensors.append(input_tensor)
@@ -294,8 +300,11 @@ def forward_backward_pipelining_without_interleaving(
             config)
         if last_iteration:
             input_tensor = None
+            p2p_communication.set_dpp_tags(bwd_send=(0, i))
             p2p_communication.send_backward(input_tensor_grad, config, is_first)
         else:
+            p2p_communication.set_dpp_tags(bwd_send=(0, i),
+                                           fwd_recv=(0, f_mb + 1))
             input_tensor = p2p_communication.send_backward_recv_forward(
                 input_tensor_grad, tensor_shape, config, is_first)
 
@@ -306,6 +315,9 @@ def forward_backward_pipelining_without_interleaving(
                 sync_gate.release()
             input_tensor_b = input_tensors.pop(0)
             output_tensor_b = output_tensors.pop(0)
+            b_mb = num_steady + i
+            p2p_communication.set_dpp_tags(bwd_recv=(0, b_mb),
+                                           bwd_send=(0, b_mb))
             output_tensor_grad = p2p_communication.recv_backward(
                 tensor_shape, config, is_last)
             input_tensor_grad = backward_step(

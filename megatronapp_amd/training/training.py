@@ -367,6 +367,9 @@ def pretrain(train_valid_test_dataset_provider, model_provider,
                             model_provider, forward_step_func)
 
     config = core_transformer_config_from_args(args)
+    if getattr(args, "use_dpp", False):
+        from ..dpp.transport import initialize_dpp
+        initialize_dpp(args, config)
     model, optimizer, opt_param_scheduler = setup_model_and_optimizer(
         model_provider, model_type, args=args)
 
@@ -394,6 +397,9 @@ def pretrain(train_valid_test_dataset_provider, model_provider,
     tracer = get_tracer()
     if tracer is not None:
         tracer.shutdown()
+    if getattr(args, "use_dpp", False):
+        from ..dpp.transport import shutdown_dpp
+        shutdown_dpp()
     return model
 
 
