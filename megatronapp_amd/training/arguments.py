@@ -301,9 +301,15 @@ def validate_args(args, defaults={}):
     total_model = (args.tensor_model_parallel_size *
                    args.pipeline_model_parallel_size *
                    args.context_parallel_size)
-    assert args.world_size % total_model == 0, (
-        f"world size {args.world_size} not divisible by tp*pp*cp {total_model}")
-    args.data_parallel_size = args.world_size // total_model
+    logical_world = args.world_size
+    if getattr(args, "forward_backward_disaggregating", False):
+        # MegaFBD: every logical rank is a (forward, backward) pair
+        assert args.world_size % 2 == 0, "FBD needs an even world size"
+        logical_world = args.world_size // 2
+    assert logical_world % total_model == 0, (
+        f"logical world size {logical_world} not divisible by tp*pp*cp "
+        f"{total_model}")
+    args.data_parallel_size = logical_world // total_model
 
     if args.global_batch_size is None:
         args.global_batch_size = args.micro_batch_size * args.data_parallel_size
