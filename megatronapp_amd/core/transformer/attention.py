@@ -60,6 +60,10 @@ class SelfAttention(MegatronModule):
             submodules.core_attention, config=config,
             layer_number=layer_number, attn_mask_type=attn_mask_type,
             attention_type="self", cp_comm_type=cp_comm_type)
+        if config.context_parallel_size > 1:
+            from .cp_attention import ContextParallelAttention
+            self.core_attention = ContextParallelAttention(
+                self.core_attention, config)
 
         self.linear_proj = build_module(
             submodules.linear_proj, self.query_projection_size,
@@ -118,6 +122,16 @@ class SelfAttention(MegatronModule):
                 q_pos_emb, k_pos_emb = rotary_pos_emb
             else:
                 q_pos_emb = k_pos_emb = rotary_pos_emb
+            if parallel_state.get_context_parallel_world_size() > 1:
+                # CP: local rows are global chunks (r, 2cp-1-r); take their
+                # rows of the full-frequency table
+                from .cp_attention import _local_global_positions
+                cp = parallel_state.get_context_parallel_world_size()
+                cp_rank = parallel_state.get_context_parallel_rank()
+                pos = _local_global_positions(cp, cp_rank, query.shape[0],
+                                              q_pos_emb.device)
+                q_pos_emb = q_pos_emb.index_select(0, pos)
+                k_pos_emb = k_pos_emb.index_select(0, pos)
             if inference_context is not None:
                 offset = inference_context.sequence_len_offset
                 q_pos_emb = q_pos_emb[offset:offset + query.shape[0]]

@@ -1,0 +1,90 @@
+"""Context parallelism: CP=2 must reproduce the single-rank forward/loss."""
+
+import hashlib
+
+import pytest
+import torch
+
+from .utils import spawn_ranks
+
+SEQ = 32
+VOCAB = 64
+
+
+def _fill(model):
+    for name, p in model.named_parameters():
+        g = torch.Generator()
+        g.manual_seed(int(hashlib.md5(name.encode()).hexdigest()[:8], 16))
+        with torch.no_grad():
+            p.copy_(torch.randn(p.shape, generator=g) * 0.05)
+
+
+def _build(cp_comm_type):
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    from megatronapp_amd.core import parallel_state
+    cp = parallel_state.get_context_parallel_world_size()
+    config = TransformerConfig(
+        num_layers=2, hidden_size=32, num_attention_heads=4,
+        hidden_dropout=0.0, attention_dropout=0.0,
+        position_embedding_type="rope", normalization="RMSNorm",
+        activation_func="swiglu", add_bias_linear=False,
+        context_parallel_size=cp, cp_comm_type=cp_comm_type,
+        masked_softmax_fusion=True)
+    m = GPTModel(config=config,
+                 transformer_layer_spec=get_gpt_layer_local_spec(
+                     normalization="RMSNorm", use_flash=False),
+                 vocab_size=VOCAB, max_sequence_length=SEQ,
+                 position_embedding_type="rope")
+    _fill(m)
+    return m
+
+
+def _batch():
+    g = torch.Generator().manual_seed(5)
+    tokens = torch.randint(0, VOCAB, (2, SEQ), generator=g)
+    pos = torch.arange(SEQ).unsqueeze(0).expand(2, -1)
+    return tokens, pos
+
+
+def _reference_logits(cp_comm_type="p2p"):
+    from megatronapp_amd.core import parallel_state
+    from .utils import init_distributed
+    init_distributed()
+    if parallel_state.model_parallel_is_initialized():
+        parallel_state.destroy_model_parallel()
+    parallel_state.initialize_model_parallel()
+    m = _build("p2p")
+    tokens, pos = _batch()
+    with torch.no_grad():
+        logits = m(tokens, pos)
+    parallel_state.destroy_model_parallel()
+    return logits
+
+
+def _cp_run(rank, world, cp_comm_type, ref):
+    from megatronapp_amd.core import parallel_state
+    from megatronapp_amd.core.utils import get_batch_on_this_cp_rank
+    parallel_state.initialize_model_parallel(context_parallel_size=world)
+    m = _build(cp_comm_type)
+    tokens, pos = _batch()
+    batch = get_batch_on_this_cp_rank(
+        {"tokens": tokens, "position_ids": pos})
+    with torch.no_grad():
+        logits_local = m(batch["tokens"], batch["position_ids"])
+    # local rows are global chunks (rank, 2cp-1-rank)
+    half = SEQ // (2 * world)
+    c0, c1 = rank, 2 * world - 1 - rank
+    expected = torch.cat([ref[:, c0 * half:(c0 + 1) * half],
+                          ref[:, c1 * half:(c1 + 1) * half]], dim=1)
+    err = (logits_local - expected).abs().max().item()
+    assert err < 1e-4, f"cp rank {rank} {cp_comm_type} err {err}"
+    parallel_state.destroy_model_parallel()
+
+
+@pytest.mark.parametrize("mode", ["allgather", "a2a"])
+def test_cp2_matches_single_rank(mode):
+    ref = _reference_logits()
+    spawn_ranks(_cp_run, world_size=2, args=(mode, ref))
