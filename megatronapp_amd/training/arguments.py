@@ -131,6 +131,9 @@ def _add_training_args(p):
     g.add_argument("--exit-signal-handler", action="store_true")
     g.add_argument("--rerun-mode", default="disabled",
                    choices=["disabled", "validate_results", "report_stats"])
+    g.add_argument("--rerun-validate-interval", type=int, default=10)
+    g.add_argument("--log-straggler", action="store_true")
+    g.add_argument("--straggler-report-interval", type=int, default=10)
 
 
 def _add_learning_rate_args(p):

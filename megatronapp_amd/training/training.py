@@ -134,19 +134,30 @@ def setup_model_and_optimizer(model_provider_func, model_type,
 
 def train_step(forward_step_func, data_iterator, model, optimizer,
                opt_param_scheduler, config, args=None):
-    """One optimizer step (reference training.py:1367)."""
+    """One optimizer step (reference training.py:1367), wrapped by the
+    rerun state machine (reference training.py:1387-1410)."""
     args = args or get_args()
-    for chunk in model:
-        chunk.zero_grad_buffer()
-    optimizer.zero_grad()
+    from ..core.rerun_state_machine import get_rerun_state_machine
+    rerun = get_rerun_state_machine()
 
     fb_func = get_forward_backward_func()
-    losses_reduced = fb_func(
-        forward_step_func=forward_step_func, data_iterator=data_iterator,
-        model=model if len(model) > 1 else model[0],
-        num_microbatches=get_num_microbatches(),
-        seq_length=args.seq_length, micro_batch_size=args.micro_batch_size,
-        forward_only=False)
+    losses_reduced = []
+    while rerun.should_run_forward_backward(data_iterator):
+        for chunk in model:
+            chunk.zero_grad_buffer()
+        optimizer.zero_grad()
+        losses_reduced = fb_func(
+            forward_step_func=forward_step_func, data_iterator=data_iterator,
+            model=model if len(model) > 1 else model[0],
+            num_microbatches=get_num_microbatches(),
+            seq_length=args.seq_length,
+            micro_batch_size=args.micro_batch_size,
+            forward_only=False)
+        if losses_reduced and "lm loss" in losses_reduced[0]:
+            rerun.record_result(sum(
+                float(d["lm loss"]) for d in losses_reduced))
+        else:
+            rerun.record_result(0.0)
 
     tracer = get_tracer()
     if tracer is not None and tracer.is_tracing_active():
@@ -280,6 +291,24 @@ def train(forward_step_func, model, optimizer, opt_param_scheduler,
     report_memory_flag = True
     tracer = get_tracer()
 
+    from ..core.rerun_state_machine import (
+        RerunDataIterator, initialize_rerun_state_machine)
+    initialize_rerun_state_machine(
+        args.rerun_mode,
+        args.rerun_validate_interval if args.rerun_mode != "disabled" else 0)
+    if args.rerun_mode == "validate_results" and train_data_iterator is not None:
+        if isinstance(train_data_iterator, list):
+            train_data_iterator = [RerunDataIterator(it)
+                                   for it in train_data_iterator]
+        else:
+            train_data_iterator = RerunDataIterator(train_data_iterator)
+
+    from ..core.straggler_detector import StragglerDetector
+    from ..core.utils import num_floating_point_operations as _nfpo
+    straggler = StragglerDetector(
+        args.straggler_report_interval if args.log_straggler else 0,
+        _nfpo(args, args.global_batch_size) / args.world_size)
+
     ws_server = None
     if args.enable_ws_server and args.training_ws_port and \
             parallel_state.get_tensor_model_parallel_rank() == 0 and \
@@ -303,9 +332,18 @@ def train(forward_step_func, model, optimizer, opt_param_scheduler,
                 dist.barrier()
             tracer.iteration_begin(iteration)
 
-        loss_dict, skipped_iter, grad_norm, num_zeros = train_step(
-            forward_step_func, train_data_iterator, model, optimizer,
-            opt_param_scheduler, config, args)
+        with straggler:
+            loss_dict, skipped_iter, grad_norm, num_zeros = train_step(
+                forward_step_func, train_data_iterator, model, optimizer,
+                opt_param_scheduler, config, args)
+        from ..core.rerun_state_machine import get_rerun_state_machine
+        if get_rerun_state_machine().should_checkpoint_and_exit():
+            print_rank_0("rerun state machine requested checkpoint + exit "
+                         "(irreproducible result detected)")
+            if args.save:
+                save_checkpoint(iteration + 1, model, optimizer,
+                                opt_param_scheduler)
+            break
         iteration += 1
         args.curr_iteration = iteration
         args.consumed_train_samples += get_current_global_batch_size()
