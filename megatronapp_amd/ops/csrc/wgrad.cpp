@@ -99,6 +99,9 @@ void wgrad_accum(torch::Tensor grad2d, torch::Tensor input2d,
   HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, N, K, N));
   HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lc, HIP_R_32F, M, N, M));
 
+  float alpha = 1.f, beta = 1.f;
+  hipStream_t stream = at::hip::getCurrentHIPStream().stream();
+
   AlgoKey key{M, N, K};
   auto it = algo_cache().find(key);
   if (it == algo_cache().end()) {
@@ -107,18 +110,54 @@ void wgrad_accum(torch::Tensor grad2d, torch::Tensor input2d,
     size_t ws = kWorkspaceBytes;
     HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
         pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
-    hipblasLtMatmulHeuristicResult_t results[4];
+    hipblasLtMatmulHeuristicResult_t results[16];
     int found = 0;
     HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
-        c.handle, op, la, lb, lc, lc, pref, 4, results, &found));
+        c.handle, op, la, lb, lc, lc, pref, 16, results, &found));
     hipblasLtMatmulPreferenceDestroy(pref);
     TORCH_CHECK(found > 0, "no hipblaslt algo for wgrad shape ", M, "x", N,
                 "x", K);
-    it = algo_cache().emplace(key, results[0].algo).first;
+    // one-shot autotune: time each candidate on a scratch C so the
+    // beta=1 accumulation into live gradients is not corrupted
+    int best = 0;
+    if (found > 1) {
+      void* scratch = nullptr;
+      const size_t cbytes = (size_t)M * N * sizeof(float);
+      if (hipMalloc(&scratch, cbytes) == hipSuccess) {
+        float best_ms = 1e30f;
+        hipEvent_t t0, t1;
+        hipEventCreate(&t0);
+        hipEventCreate(&t1);
+        for (int i = 0; i < found; ++i) {
+          // warm
+          if (hipblasLtMatmul(c.handle, op, &alpha, input2d.data_ptr(), la,
+                              grad2d.data_ptr(), lb, &beta, scratch, lc,
+                              scratch, lc, &results[i].algo, c.workspace,
+                              kWorkspaceBytes,
+                              stream) != HIPBLAS_STATUS_SUCCESS)
+            continue;
+          hipEventRecord(t0, stream);
+          for (int r = 0; r < 3; ++r)
+            hipblasLtMatmul(c.handle, op, &alpha, input2d.data_ptr(), la,
+                            grad2d.data_ptr(), lb, &beta, scratch, lc,
+                            scratch, lc, &results[i].algo, c.workspace,
+                            kWorkspaceBytes, stream);
+          hipEventRecord(t1, stream);
+          hipEventSynchronize(t1);
+          float ms = 1e30f;
+          hipEventElapsedTime(&ms, t0, t1);
+          if (ms < best_ms) {
+            best_ms = ms;
+            best = i;
+          }
+        }
+        hipEventDestroy(t0);
+        hipEventDestroy(t1);
+        hipFree(scratch);
+      }
+    }
+    it = algo_cache().emplace(key, results[best].algo).first;
   }
-
-  float alpha = 1.f, beta = 1.f;
-  hipStream_t stream = at::hip::getCurrentHIPStream().stream();
   HIPBLASLT_CHECK(hipblasLtMatmul(
       c.handle, op, &alpha, input2d.data_ptr(), la, grad2d.data_ptr(), lb,
       &beta, main_grad.data_ptr(), lc, main_grad.data_ptr(), lc, &it->second,
