@@ -154,6 +154,8 @@ class FlashAttention(nn.Module):
         if self.dropout_p > 0 and self.training:
             return False
         hn = query.shape[-1]
+        if query.shape[0] % 128 != 0 or key.shape[0] % 64 != 0:
+            return False
         return hn in (64, 128) and query.dtype == torch.bfloat16
 
     def forward(self, query, key, value, attention_mask=None,

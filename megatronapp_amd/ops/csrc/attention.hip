@@ -30,6 +30,13 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8;
 #define QW 16    // q rows per wave
 #define QBLK 64  // q rows per workgroup
 #define KVBLK 64
+// padded row stride (elements) for TRANSPOSED tiles ([d][kv] / [d][q]):
+// 72*2B = 144B rows make the 16-lane ds_read_b128 groups (16 distinct d
+// rows, same kv offset) land on 16 distinct banks (36*d mod 64 is a
+// bijection on d=0..15) and spread the scalar transpose writes — the
+// 256B-stride layout was an 8-way conflict on every write (PMC:
+// SQ_LDS_BANK_CONFLICT 1.5e9 per 20 fwd calls).
+#define TSTRIDE 72
 
 __device__ __forceinline__ unsigned swz(int row, int col_bytes) {
   // XOR swizzle within a row: spreads the 16B slots of a 256B bank row
@@ -79,8 +86,12 @@ __device__ __forceinline__ void transpose4_write(
   }
 }
 
-template <int D, bool CAUSAL>
-__global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
+#define FWD_BLOCK 512   // 8 waves
+#define FQBLK 128       // q rows per workgroup (halves K/V tile reloads
+                        // vs 64: the loader was 63% of kernel time)
+
+template <int D, bool CAUSAL, int ABLATE = 0>
+__global__ __launch_bounds__(FWD_BLOCK) void attn_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, unsigned short* __restrict__ o,
     float* __restrict__ lse, int sq, int sk, int b, int nh, int ng,
@@ -105,13 +116,15 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
   const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
   const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
 
-  // LDS: K [KVBLK][D] swizzled (bf16), Vt [D][KVBLK] swizzled, P [QBLK][KVBLK]
+  // LDS: K [KVBLK][D] swizzled (bf16), Vt [D][TSTRIDE], P/O staging.
+  // (measured: direct-from-L2 K fragments were ~2x SLOWER — 16 lanes
+  // gathering 16 different rows defeats coalescing — so K stays staged)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* k_lds = (unsigned short*)smem;               // KVBLK*D
-  unsigned short* vt_lds = k_lds + KVBLK * D;                  // D*KVBLK
-  unsigned short* p_lds = vt_lds + (long)D * KVBLK;            // QBLK*KVBLK
+  unsigned short* vt_lds = k_lds + KVBLK * D;                  // D*TSTRIDE
+  unsigned short* p_lds = vt_lds + (long)D * TSTRIDE;          // FQBLK*max(KVBLK,D)
 
-  const int q0 = qtile * QBLK + wid * QW;  // this wave's first q row
+  const int q0 = qtile * FQBLK + wid * QW;  // this wave's first q row
 
   // ---- load Q fragments (A-layout): frag f covers d = f*32 + lrow*8 + j
   constexpr int DF = D / 32;  // MFMA k-steps over d
@@ -137,50 +150,63 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
 #pragma unroll
   for (int dsub = 0; dsub < DS; ++dsub) oacc[dsub] = f32x4{0, 0, 0, 0};
 
-  const int q_hi = qtile * QBLK + QBLK - 1;  // last q row in workgroup
+  const int q_hi = qtile * FQBLK + FQBLK - 1;  // last q row in workgroup
   int kv_end = sk;
   if (CAUSAL) kv_end = min(sk, q_hi + 1 + (sk - sq));
   const int n_kv_tiles = (kv_end + KVBLK - 1) / KVBLK;
 
+  // T14 async-stage split: issue tile t+1's global loads into registers
+  // right after tile t's LDS image is consumed-safe, write them to LDS
+  // after the end-of-tile barrier — HBM latency and the transpose write
+  // pass hide under the tile-t MFMAs.
+  constexpr int PIECES = KVBLK * D / 8 / FWD_BLOCK;  // 2 at D=128
+  bf16x8 kreg[PIECES], vreg[PIECES];
+
+  auto stage_load = [&](int t) {
+    const int kv0 = t * KVBLK;
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int grow = kv0 + row;
+      kreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  };
+
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) =
+          kreg[pc];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[(long)(col + j) * TSTRIDE + row] = (unsigned short)vreg[pc][j];
+    }
+  };
+
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+
   for (int t = 0; t < n_kv_tiles; ++t) {
     const int kv0 = t * KVBLK;
-    // ---- cooperative K/V load: 256 threads, K tile KVBLK x D
-    // each thread loads (KVBLK*D/8)/256 short8 pieces
-    {
-      constexpr int pieces = KVBLK * D / 8 / ATT_BLOCK;  // e.g. 4 at D=128
-#pragma unroll
-      for (int pc = 0; pc < pieces; ++pc) {
-        const int idx = (pc * ATT_BLOCK + tid) * 8;  // element index
-        const int row = idx / D;
-        const int col = idx % D;
-        const int grow = kv0 + row;
-        bf16x8 kv8 = (grow < sk)
-                         ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
-                         : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) = kv8;
-        bf16x8 vv8 = (grow < sk)
-                         ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
-                         : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        if constexpr (false) {  // shuffle-transpose measured SLOWER than scalar
-          transpose4_write(vt_lds, KVBLK, (row / 4) * 4, col, vv8,
-                           tid % WAVE);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int d = col + j;
-            *(unsigned short*)((char*)(vt_lds + (long)d * KVBLK) +
-                               swz(d, row * 2)) = (unsigned short)vv8[j];
-          }
-        }
-      }
-    }
-    __syncthreads();
+    if (t + 1 < n_kv_tiles) stage_load(t + 1);
 
     // ---- S = scale * Q K^T for this wave's 16 q rows, 4 kv subtiles
     f32x4 sacc[KVBLK / 16];
 #pragma unroll
     for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
       sacc[ksub] = f32x4{0, 0, 0, 0};
+      if constexpr (ABLATE >= 3) continue;
 #pragma unroll
       for (int f = 0; f < DF; ++f) {
         // B fragment: B[d][kv] = K[kv0+ksub*16 + lcol][f*32 + lrow*8 + j]
@@ -192,20 +218,49 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
       }
     }
 
-    // ---- masking + online softmax
+    if constexpr (ABLATE >= 2) {
+      // keep sacc live so the S MFMAs aren't dead-code-eliminated
+#pragma unroll
+      for (int ksub = 0; ksub < KVBLK / 16; ++ksub)
+        asm volatile("" ::"v"(sacc[ksub]));
+      __syncthreads();
+      __syncthreads();
+      if (t + 1 < n_kv_tiles) {
+        stage_write();
+        __syncthreads();
+      }
+      continue;
+    }
+    // ---- masking + online softmax.  Fast path: a tile entirely below
+    // the causal diagonal for this wave's 16 q rows needs no per-element
+    // masking (only ~1 of the kv tiles straddles the diagonal).
     float mtile[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+    const int q_lo_wave = q0 + (sk - sq);     // smallest causal bound
+    const bool full_tile = !CAUSAL ? (kv0 + KVBLK <= sk)
+                                   : (kv0 + KVBLK - 1 <= q_lo_wave);
+    if (full_tile) {
 #pragma unroll
-    for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+      for (int ksub = 0; ksub < KVBLK / 16; ++ksub)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + lrow * 4 + r;
-        const int kvcol = kv0 + ksub * 16 + lcol;
-        float s = sacc[ksub][r] * scale;
-        bool valid = (kvcol < sk) && (qrow < sq);
-        if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
-        s = valid ? s : -INFINITY;
-        sacc[ksub][r] = s;
-        mtile[r] = fmaxf(mtile[r], s);
+        for (int r = 0; r < 4; ++r) {
+          float s = sacc[ksub][r] * scale;
+          sacc[ksub][r] = s;
+          mtile[r] = fmaxf(mtile[r], s);
+        }
+    } else {
+#pragma unroll
+      for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qrow = q0 + lrow * 4 + r;
+          const int kvcol = kv0 + ksub * 16 + lcol;
+          float s = sacc[ksub][r] * scale;
+          bool valid = (kvcol < sk) && (qrow < sq);
+          if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
+          s = valid ? s : -INFINITY;
+          sacc[ksub][r] = s;
+          mtile[r] = fmaxf(mtile[r], s);
+        }
       }
     }
 #pragma unroll
@@ -249,7 +304,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
 
     // ---- O += P V : contraction over kv (2 MFMA k-steps of 32)
 #pragma unroll
-    for (int dsub = 0; dsub < DS; ++dsub) {
+    for (int dsub = 0; dsub < DS && ABLATE < 1; ++dsub) {
 #pragma unroll
       for (int ks = 0; ks < KVBLK / 32; ++ks) {
         // A fragment: P[q0w + l&15][ks*32 + lrow*8 .. +8] — the 8 kv
@@ -260,8 +315,8 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
                                      swz(prow, (ks * 32 + lrow * 8) * 2));
         // B fragment: Vt[dsub*16 + lcol][ks*32 + lrow*8 .. +8]
         const int vrow = dsub * 16 + lcol;
-        bf16x8 vb = *(const bf16x8*)((char*)(vt_lds + (long)vrow * KVBLK) +
-                                     swz(vrow, (ks * 32 + lrow * 8) * 2));
+        bf16x8 vb = *(const bf16x8*)(vt_lds + (long)vrow * TSTRIDE +
+                                     ks * 32 + lrow * 8);
         // note operand order: C[q][d] = A(P[q][kv]) x B(V[kv][d]); our B
         // fragment is indexed [d][kv] -> use transposed roles:
         // mfma(A=pa over kv, B=vb over kv) with B[k][n]: k=kv, n=d: need
@@ -271,14 +326,18 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
       }
     }
     __syncthreads();
+    if (t + 1 < n_kv_tiles) {
+      stage_write();
+      __syncthreads();
+    }
   }
 
   // ---- epilogue: normalize, write O and LSE
   // O C-layout: lane holds O[q=lrow*4+r][d=dsub*16+lcol]
   // stage through p_lds (reuse as [QW][D] per wave? QBLK*KVBLK >= QW*D OK
   // for D<=256) then coalesced store
-  // per-wave staging slice of k_lds (KVBLK*D = 4 waves x QW*D shorts)
-  unsigned short* o_stage = k_lds + (long)wid * QW * D;
+  // per-wave O staging slice of p_lds (sized FQBLK*max(KVBLK, D))
+  unsigned short* o_stage = p_lds + (long)wid * QW * D;
 
 #pragma unroll
   for (int dsub = 0; dsub < DS; ++dsub) {
@@ -315,15 +374,39 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
   }
 }
 
+// ablation entry (perf diagnosis only; outputs wrong for level>0)
+void launch_attn_fwd_ablate(const void* q, const void* k, const void* v,
+                            void* o, float* lse, int sq, int sk, int b,
+                            int nh, int ng, int d, float scale, int level,
+                            hipStream_t stream) {
+  dim3 grid(sq / FQBLK, b * nh);
+  dim3 block(FWD_BLOCK);
+  const int p_elems = FQBLK * (d > KVBLK ? d : KVBLK);  // P tile / O staging
+  const size_t lds = (size_t)(KVBLK * d + d * TSTRIDE + p_elems) *
+                     sizeof(unsigned short);
+#define ABL_CASE(L)                                                         \
+  case L:                                                                   \
+    hipLaunchKernelGGL((attn_fwd_kernel<128, true, L>), grid, block, lds,   \
+                       stream, (const unsigned short*)q,                    \
+                       (const unsigned short*)k, (const unsigned short*)v,  \
+                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);  \
+    break;
+  switch (level) { ABL_CASE(0) ABL_CASE(1) ABL_CASE(2) ABL_CASE(3) }
+#undef ABL_CASE
+  HIP_CHECK_LAUNCH();
+}
+
 // ---------------------------------------------------------------------------
 void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
                      float* lse, int sq, int sk, int b, int nh, int ng, int d,
                      float scale, bool causal, hipStream_t stream) {
-  if (sq % QBLK != 0 || sk % KVBLK != 0)
-    throw std::runtime_error("attn_fwd: sq/sk must be multiples of 64");
-  dim3 grid(sq / QBLK, b * nh);
-  dim3 block(ATT_BLOCK);
-  const size_t lds = (size_t)(KVBLK * d + d * KVBLK + QBLK * KVBLK) *
+  if (sq % FQBLK != 0 || sk % KVBLK != 0)
+    throw std::runtime_error(
+        "attn_fwd: sq must be a multiple of 128, sk of 64");
+  dim3 grid(sq / FQBLK, b * nh);
+  dim3 block(FWD_BLOCK);
+  const int p_elems = FQBLK * (d > KVBLK ? d : KVBLK);  // P tile / O staging
+  const size_t lds = (size_t)(KVBLK * d + d * TSTRIDE + p_elems) *
                      sizeof(unsigned short);
   if (d == 128) {
     if (causal)
@@ -381,7 +464,7 @@ __global__ void attn_bwd_pre_kernel(const unsigned short* __restrict__ do_,
 // ---------------------------------------------------------------- dQ kernel
 // grid (sq/QBLK, b*nh); per wave 16 q rows; loops kv tiles.
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v,
     const unsigned short* __restrict__ dout,
@@ -412,11 +495,11 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* k_lds = (unsigned short*)smem;              // KVBLK*D
-  unsigned short* kt_lds = k_lds + KVBLK * D;                 // D*KVBLK
-  unsigned short* v_lds = kt_lds + (long)D * KVBLK;           // KVBLK*D
-  unsigned short* ds_lds = v_lds + KVBLK * D;                 // QBLK*KVBLK
+  unsigned short* kt_lds = k_lds + KVBLK * D;                 // D*TSTRIDE
+  unsigned short* v_lds = kt_lds + (long)D * TSTRIDE;         // KVBLK*D
+  unsigned short* ds_lds = v_lds + KVBLK * D;                 // FQBLK*KVBLK
 
-  const int q0 = qtile * QBLK + wid * QW;
+  const int q0 = qtile * FQBLK + wid * QW;
   constexpr int DF = D / 32;
   constexpr int DS_ = D / 16;
 
@@ -443,7 +526,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
 #pragma unroll
   for (int s = 0; s < DS_; ++s) dqacc[s] = f32x4{0, 0, 0, 0};
 
-  const int q_hi = qtile * QBLK + QBLK - 1;
+  const int q_hi = qtile * FQBLK + FQBLK - 1;
   int kv_end = sk;
   if (CAUSAL) kv_end = min(sk, q_hi + 1 + (sk - sq));
   const int n_kv_tiles = (kv_end + KVBLK - 1) / KVBLK;
@@ -451,10 +534,10 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
   for (int t = 0; t < n_kv_tiles; ++t) {
     const int kv0 = t * KVBLK;
     {
-      constexpr int pieces = KVBLK * D / 8 / ATT_BLOCK;
+      constexpr int pieces = KVBLK * D / 8 / FWD_BLOCK;
 #pragma unroll
       for (int pc = 0; pc < pieces; ++pc) {
-        const int idx = (pc * ATT_BLOCK + tid) * 8;
+        const int idx = (pc * FWD_BLOCK + tid) * 8;
         const int row = idx / D;
         const int col = idx % D;
         const int grow = kv0 + row;
@@ -462,15 +545,10 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
                         ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
                         : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
         *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) = k8;
-        if constexpr (false) {
-          transpose4_write(kt_lds, KVBLK, (row / 4) * 4, col, k8, tid % WAVE);
-        } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int d_ = col + j;
-            *(unsigned short*)((char*)(kt_lds + (long)d_ * KVBLK) +
-                               swz(d_, row * 2)) = (unsigned short)k8[j];
-          }
+        for (int j = 0; j < 8; ++j) {
+          const int d_ = col + j;
+          kt_lds[(long)d_ * TSTRIDE + row] = (unsigned short)k8[j];
         }
         bf16x8 v8 = (grow < sk)
                         ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
@@ -523,8 +601,8 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
         bf16x8 pa = *(const bf16x8*)((char*)(ds_lds + (long)prow * KVBLK) +
                                      swz(prow, (ks * 32 + lrow * 8) * 2));
         const int krow = dsub * 16 + lcol;  // d index
-        bf16x8 kb = *(const bf16x8*)((char*)(kt_lds + (long)krow * KVBLK) +
-                                     swz(krow, (ks * 32 + lrow * 8) * 2));
+        bf16x8 kb = *(const bf16x8*)(kt_lds + (long)krow * TSTRIDE +
+                                     ks * 32 + lrow * 8);
         dqacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, kb,
                                                               dqacc[dsub],
                                                               0, 0, 0);
@@ -533,7 +611,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
     __syncthreads();
   }
 
-  // epilogue via k_lds staging
+  // epilogue via LDS staging (reuse the k + kt span)
   unsigned short* stage = k_lds + (long)wid * QW * D;
 #pragma unroll
   for (int dsub = 0; dsub < DS_; ++dsub)
@@ -559,8 +637,10 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
 // -------------------------------------------------------------- dK/dV kernel
 // grid (sk/KVBLK, b*ng); per wave 16 kv rows; loops q tiles and the q-head
 // group (GQA: dK/dV sum over the nh/ng q heads sharing this kv head).
+#define DKVBLK 128  // kv rows per dkv workgroup (8 waves)
+
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v,
     const unsigned short* __restrict__ dout,
@@ -585,12 +665,12 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* qr_lds = (unsigned short*)smem;              // QBLK*D (row)
-  unsigned short* qt_lds = qr_lds + QBLK * D;                  // D*QBLK
-  unsigned short* dor_lds = qt_lds + (long)D * QBLK;           // QBLK*D
-  unsigned short* dot_lds = dor_lds + QBLK * D;                // D*QBLK
-  unsigned short* p_lds = dot_lds + (long)D * QBLK;            // KVBLK*QBLK
+  unsigned short* qt_lds = qr_lds + QBLK * D;                  // D*TSTRIDE
+  unsigned short* dor_lds = qt_lds + (long)D * TSTRIDE;        // QBLK*D
+  unsigned short* dot_lds = dor_lds + QBLK * D;                // D*TSTRIDE
+  unsigned short* p_lds = dot_lds + (long)D * TSTRIDE;         // DKVBLK*QBLK u32
 
-  const int kv0 = kvtile * KVBLK + wid * QW;  // this wave's 16 kv rows
+  const int kv0 = kvtile * DKVBLK + wid * QW;  // this wave's 16 kv rows
   constexpr int DF = D / 32;
   constexpr int DS_ = D / 16;
 
@@ -614,7 +694,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
     dvacc[s] = f32x4{0, 0, 0, 0};
   }
 
-  const int kv_lo = kvtile * KVBLK;  // first kv of the block's tile
+  const int kv_lo = kvtile * DKVBLK;  // first kv of the block's tile
   int q_start = 0;
   if (CAUSAL) q_start = max(0, (kv_lo - (sk - sq)) / QBLK * QBLK);
 
@@ -628,10 +708,10 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
     for (int qt = q_start; qt < sq; qt += QBLK) {
       // cooperative load Q/dO tiles (row-major + transposed)
       {
-        constexpr int pieces = QBLK * D / 8 / ATT_BLOCK;
+        constexpr int pieces = QBLK * D / 8 / FWD_BLOCK;
 #pragma unroll
         for (int pc = 0; pc < pieces; ++pc) {
-          const int idx = (pc * ATT_BLOCK + tid) * 8;
+          const int idx = (pc * FWD_BLOCK + tid) * 8;
           const int row = idx / D;
           const int col = idx % D;
           const int grow = qt + row;
@@ -643,19 +723,11 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
                           ? *(const bf16x8*)(dop + (long)grow * q_ss + col)
                           : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
           *(bf16x8*)((char*)(dor_lds + (long)row * D) + swz(row, col * 2)) = d8;
-          if constexpr (false) {
-            transpose4_write(qt_lds, QBLK, (row / 4) * 4, col, q8, tid % WAVE);
-            transpose4_write(dot_lds, QBLK, (row / 4) * 4, col, d8,
-                             tid % WAVE);
-          } else {
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int d_ = col + j;
-              *(unsigned short*)((char*)(qt_lds + (long)d_ * QBLK) +
-                                 swz(d_, row * 2)) = (unsigned short)q8[j];
-              *(unsigned short*)((char*)(dot_lds + (long)d_ * QBLK) +
-                                 swz(d_, row * 2)) = (unsigned short)d8[j];
-            }
+          for (int j = 0; j < 8; ++j) {
+            const int d_ = col + j;
+            qt_lds[(long)d_ * TSTRIDE + row] = (unsigned short)q8[j];
+            dot_lds[(long)d_ * TSTRIDE + row] = (unsigned short)d8[j];
           }
         }
       }
@@ -725,10 +797,10 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
             da[4 + j] = (short)(phi[j] >> 16);
           }
           const int drow_ = dsub * 16 + lcol;  // d index
-          bf16x8 dob = *(const bf16x8*)((char*)(dot_lds + (long)drow_ * QBLK) +
-                                        swz(drow_, (qs * 32 + lrow * 8) * 2));
-          bf16x8 qb = *(const bf16x8*)((char*)(qt_lds + (long)drow_ * QBLK) +
-                                       swz(drow_, (qs * 32 + lrow * 8) * 2));
+          bf16x8 dob = *(const bf16x8*)(dot_lds + (long)drow_ * TSTRIDE +
+                                        qs * 32 + lrow * 8);
+          bf16x8 qb = *(const bf16x8*)(qt_lds + (long)drow_ * TSTRIDE +
+                                       qs * 32 + lrow * 8);
           dvacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pa, dob, dvacc[dsub], 0, 0, 0);
           dkacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -739,7 +811,7 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkv_kernel(
     }
   }
 
-  // epilogue: stage dK/dV through qr_lds and store
+  // epilogue: stage dK/dV through the LDS span and store
   unsigned short* stage = qr_lds + (long)wid * QW * D;
 #pragma unroll
   for (int dsub = 0; dsub < DS_; ++dsub)
@@ -790,8 +862,8 @@ void launch_attn_bwd(const void* dout, const void* q, const void* k,
                      float* drow, void* dq, void* dk, void* dv, int sq, int sk,
                      int b, int nh, int ng, int d, float scale, bool causal,
                      hipStream_t stream) {
-  if (sq % QBLK != 0 || sk % KVBLK != 0)
-    throw std::runtime_error("attn_bwd: sq/sk must be multiples of 64");
+  if (sq % FQBLK != 0 || sk % DKVBLK != 0)
+    throw std::runtime_error("attn_bwd: sq/sk must be multiples of 128");
   // Drow
   {
     const long rows = (long)sq * b * nh;
@@ -803,22 +875,25 @@ void launch_attn_bwd(const void* dout, const void* q, const void* k,
                        drow, rows, d);
     HIP_CHECK_LAUNCH();
   }
-  const size_t lds_dq = (size_t)(2 * KVBLK * d + d * KVBLK + QBLK * KVBLK) *
-                        sizeof(unsigned short);
-  const size_t lds_dkv = (size_t)(2 * QBLK * d + 2 * d * QBLK) *
+  // dq: k + kt + v + ds(FQBLK x KVBLK); epilogue staging (FQBLK*d shorts)
+  // reuses the k+kt span
+  const size_t lds_dq = (size_t)(2 * KVBLK * d + d * TSTRIDE +
+                                 FQBLK * KVBLK) * sizeof(unsigned short);
+  const size_t lds_dkv = (size_t)(2 * QBLK * d + 2 * d * TSTRIDE) *
                              sizeof(unsigned short) +
-                         (size_t)KVBLK * QBLK * sizeof(unsigned);
+                         (size_t)DKVBLK * QBLK * sizeof(unsigned);
 #define ATT_BWD_LAUNCH(DD, CC)                                                \
   do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), dim3(sq / QBLK, b * nh), \
-                       dim3(ATT_BLOCK), lds_dq, stream,                       \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>),                          \
+                       dim3(sq / FQBLK, b * nh),                              \
+                       dim3(FWD_BLOCK), lds_dq, stream,                       \
                        (const unsigned short*)q, (const unsigned short*)k,    \
                        (const unsigned short*)v,                              \
                        (const unsigned short*)dout, lse, drow,                \
                        (unsigned short*)dq, sq, sk, b, nh, ng, scale);        \
     HIP_CHECK_LAUNCH();                                                       \
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, CC>),                         \
-                       dim3(sk / KVBLK, b * ng), dim3(ATT_BLOCK), lds_dkv,    \
+                       dim3(sk / DKVBLK, b * ng), dim3(FWD_BLOCK), lds_dkv,   \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k, (const unsigned short*)v,    \
                        (const unsigned short*)dout, lse, drow,                \

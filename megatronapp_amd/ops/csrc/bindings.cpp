@@ -40,6 +40,9 @@ void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, int, float, bool, hipStream_t);
+void launch_attn_fwd_ablate(const void*, const void*, const void*, void*,
+                            float*, int, int, int, int, int, int, float, int,
+                            hipStream_t);
 void launch_attn_bwd(const void*, const void*, const void*, const void*,
                      const void*, const float*, float*, void*, void*, void*,
                      int, int, int, int, int, int, float, bool, hipStream_t);
@@ -312,6 +315,20 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   return {dq, dk, dv};
 }
 
+std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
+                                           torch::Tensor v, double scale,
+                                           long level) {
+  const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
+            d = (int)q.size(3);
+  const int sk = (int)k.size(0), ng = (int)k.size(2);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({b, nh, sq}, q.options().dtype(torch::kFloat32));
+  launch_attn_fwd_ablate(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                         o.data_ptr(), lse.data_ptr<float>(), sq, sk, b, nh,
+                         ng, d, (float)scale, (int)level, cur_stream());
+  return {o, lse};
+}
+
 // --------------------------------------------------------------------- adam
 void adamw_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double beta1, double beta2,
@@ -349,6 +366,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_flat", &adamw_flat);
   mod.def("wgrad_accum", &wgrad_accum);
   mod.def("attn_fwd", &attn_fwd);
+  mod.def("attn_fwd_ablate", &attn_fwd_ablate);
   mod.def("attn_bwd", &attn_bwd);
   mod.def("colsum_accum", &colsum_accum);
   mod.def("embedding_bwd_accum", &embedding_bwd_accum);

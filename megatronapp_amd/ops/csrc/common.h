@@ -19,11 +19,10 @@ __device__ __forceinline__ float bf2f(unsigned short u) {
 }
 
 __device__ __forceinline__ unsigned short f2bf(float f) {
-  // round-to-nearest-even bf16 conversion
-  unsigned int x = __builtin_bit_cast(unsigned int, f);
-  unsigned int lsb = (x >> 16) & 1u;
-  x += 0x7fffu + lsb;
-  return (unsigned short)(x >> 16);
+  // hardware RNE conversion: lowers to ONE v_cvt_pk_bf16_f32 (the manual
+  // bit-math version cost 3 VALU ops and made attention VALU-bound)
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return __builtin_bit_cast(unsigned short, h);
 }
 
 // wave-level sum over all 64 lanes

@@ -25,6 +25,12 @@ def main():
         dq, dk, dv = ops.get_ops().attn_bwd(do, q, k, v, o, lse, scale, True)
     torch.cuda.synchronize(); dt = (time.perf_counter() - t0) / reps
     print(f"attn_bwd: {dt*1e6:.0f} us  {2.5*flops_fwd/dt/1e12:.0f} TF-equiv")
+    for lvl, name in ((1, "noPV"), (2, "noSM"), (3, "loader-only")):
+        torch.cuda.synchronize(); t0 = time.perf_counter()
+        for _ in range(reps):
+            ops.get_ops().attn_fwd_ablate(q, k, v, scale, lvl)
+        torch.cuda.synchronize(); dt = (time.perf_counter() - t0) / reps
+        print(f"attn_fwd ablate {name}: {dt*1e6:.0f} us")
 
 if __name__ == "__main__":
     main()
