@@ -10,6 +10,7 @@ causal labels, loss mask, position ids, attention mask.
 
 from __future__ import annotations
 
+import os
 from dataclasses import dataclass
 from typing import Optional
 
@@ -80,6 +81,11 @@ class GPTDataset(torch.utils.data.Dataset):
         self.name = name
         if token_file.endswith(".npy"):
             self.tokens = np.load(token_file, mmap_mode="r")
+        elif os.path.exists(token_file + ".idx"):
+            # reference-format indexed dataset: the .bin is a flat token
+            # stream; document boundaries live in the .idx
+            from .indexed_dataset import IndexedDataset
+            self.tokens = IndexedDataset(token_file).bin
         else:
             self.tokens = np.memmap(token_file, dtype=np.int32, mode="r")
         max_samples = (len(self.tokens) - 1) // config.sequence_length

@@ -42,3 +42,23 @@ def test_save_and_resume(tmp_path):
     assert "iteration        5/6" in out2 or "iteration        6/6" in out2
     # iterations 1-4 must NOT rerun
     assert "iteration        1/6" not in out2
+
+
+def test_indexed_dataset_roundtrip(tmp_path):
+    import numpy as np
+    from megatronapp_amd.core.datasets.indexed_dataset import (
+        IndexedDataset, IndexedDatasetBuilder)
+    prefix = str(tmp_path / "corpus")
+    b = IndexedDatasetBuilder(prefix, dtype=np.int32)
+    docs = [[1, 2, 3], [7, 8, 9, 10, 11], [42]]
+    for d in docs:
+        b.add_item(d)
+        b.end_document()
+    b.finalize()
+    ds = IndexedDataset(prefix)
+    assert len(ds) == 3
+    for i, d in enumerate(docs):
+        assert list(ds.get(i)) == d
+    assert list(ds.get(1, offset=1, length=2)) == [8, 9]
+    # flat .bin stream view used by GPTDataset
+    assert list(ds.bin[:3]) == [1, 2, 3]
