@@ -28,7 +28,9 @@ def _norm_spec(normalization: str):
 
 def get_gpt_layer_local_spec(num_experts: int = None, moe_grouped_gemm: bool = False,
                              qk_layernorm: bool = False, *, normalization: str = "LayerNorm",
-                             use_flash: bool = True) -> ModuleSpec:
+                             use_flash: bool = True,
+                             attn_mask_type: AttnMaskType = AttnMaskType.causal
+                             ) -> ModuleSpec:
     norm = _norm_spec(normalization)
     core_attn = FlashAttention if use_flash else DotProductAttention
     if num_experts is None:
@@ -51,7 +53,7 @@ def get_gpt_layer_local_spec(num_experts: int = None, moe_grouped_gemm: bool = F
             input_layernorm=norm,
             self_attention=ModuleSpec(
                 module=SelfAttention,
-                params={"attn_mask_type": AttnMaskType.causal},
+                params={"attn_mask_type": attn_mask_type},
                 submodules=SelfAttentionSubmodules(
                     linear_qkv=ColumnParallelLinear,
                     core_attention=core_attn,

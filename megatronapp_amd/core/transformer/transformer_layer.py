@@ -51,6 +51,15 @@ class TransformerLayer(MegatronModule):
             hidden_size=config.hidden_size, eps=config.layernorm_epsilon)
         self.self_attention = build_module(
             submodules.self_attention, config=config, layer_number=layer_number)
+        if submodules.cross_attention is not None:
+            self.pre_cross_attn_layernorm = build_module(
+                submodules.pre_cross_attn_layernorm, config=config,
+                hidden_size=config.hidden_size, eps=config.layernorm_epsilon)
+            self.cross_attention = build_module(
+                submodules.cross_attention, config=config,
+                layer_number=layer_number)
+        else:
+            self.cross_attention = None
         self.pre_mlp_layernorm = build_module(
             submodules.pre_mlp_layernorm, config=config,
             hidden_size=config.hidden_size, eps=config.layernorm_epsilon)
@@ -83,6 +92,17 @@ class TransformerLayer(MegatronModule):
                 with self.bias_dropout_add_exec_handler():
                     hidden_states = self._bda()(
                         attention_output_with_bias, residual, self.hidden_dropout)
+
+            # --- cross-attention block (encoder-decoder models) ---
+            if self.cross_attention is not None and context is not None:
+                residual = hidden_states
+                normed = self.pre_cross_attn_layernorm(hidden_states)
+                cross_out_with_bias = self.cross_attention(
+                    normed, attention_mask=context_mask,
+                    key_value_states=context)
+                with self.bias_dropout_add_exec_handler():
+                    hidden_states = self._bda()(
+                        cross_out_with_bias, residual, self.hidden_dropout)
 
             # --- MLP block ---
             with trace_scope("_forward_mlp"):
