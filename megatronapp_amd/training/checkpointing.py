@@ -86,6 +86,10 @@ def save_checkpoint(iteration, model: List, optimizer, opt_param_scheduler,
     args = get_args()
     if args.save is None:
         return
+    if getattr(args, "ckpt_format", "torch") == "torch_dist":
+        return _save_checkpoint_torch_dist(
+            iteration, model, optimizer, opt_param_scheduler,
+            num_floating_point_operations_so_far)
     state_dict = {
         "args": vars(args).copy(),
         "checkpoint_version": 3.0,
@@ -143,6 +147,12 @@ def load_checkpoint(model: List, optimizer, opt_param_scheduler,
             print(f"  no checkpoint tracker at {tracker}, starting fresh")
         return 0, 0
     iteration, release = read_metadata(tracker)
+    # sharded (torch_dist) checkpoints are recognised by their index file
+    base = get_checkpoint_name(load_dir, iteration, release,
+                               return_base_dir=True)
+    if os.path.exists(os.path.join(base, "index.json")):
+        return _load_checkpoint_torch_dist(
+            model, optimizer, opt_param_scheduler, base, iteration, strict)
     name = get_checkpoint_name(load_dir, iteration, release)
     state_dict = torch.load(name, map_location="cpu", weights_only=False)
 
@@ -178,4 +188,78 @@ def load_checkpoint(model: List, optimizer, opt_param_scheduler,
     num_flop = state_dict.get("num_floating_point_operations_so_far", 0)
     if args.rank == 0:
         print(f"  loaded checkpoint from {load_dir} at iteration {iteration}")
+    return (0 if args.finetune else iteration), num_flop
+
+
+# ---------------------------------------------------------------------------
+# torch_dist (sharded) format: model weights via dist_checkpointing with
+# cross-topology resharding; optimizer shards + rng per-rank alongside.
+# ---------------------------------------------------------------------------
+
+def _model_sharded_sd(model: List):
+    from ..core.dist_checkpointing import module_sharded_state_dict
+    sd = {}
+    for i, chunk in enumerate(model):
+        mod = chunk
+        while hasattr(mod, "module"):
+            mod = mod.module
+        prefix = "model." if len(model) == 1 else f"model{i}."
+        sd.update(module_sharded_state_dict(mod, prefix))
+    return sd
+
+
+def _save_checkpoint_torch_dist(iteration, model, optimizer,
+                                opt_param_scheduler, num_flop):
+    from ..core.dist_checkpointing import save as dist_save
+    args = get_args()
+    base = get_checkpoint_name(args.save, iteration, return_base_dir=True)
+    os.makedirs(base, exist_ok=True)
+    common = {
+        "args": {k: v for k, v in vars(args).items()
+                 if isinstance(v, (int, float, str, bool, list, tuple,
+                                   type(None)))},
+        "iteration": iteration,
+        "checkpoint_version": 3.0,
+        "num_floating_point_operations_so_far": num_flop,
+    }
+    if opt_param_scheduler is not None:
+        common["opt_param_scheduler"] = opt_param_scheduler.state_dict()
+    if not args.no_save_rng:
+        common["rng_state"] = _rng_state()
+    dist_save(_model_sharded_sd(model), base, common_state=common)
+    if optimizer is not None and not args.no_save_optim:
+        rank = dist.get_rank() if dist.is_initialized() else 0
+        torch.save({"optimizer": optimizer.state_dict()},
+                   os.path.join(base, f"optim_rank{rank:05d}.pt"))
+    if dist.is_initialized():
+        dist.barrier()
+    if (not dist.is_initialized()) or dist.get_rank() == 0:
+        with open(get_checkpoint_tracker_filename(args.save), "w") as f:
+            f.write(str(iteration))
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def _load_checkpoint_torch_dist(model, optimizer, opt_param_scheduler, base,
+                                iteration, strict):
+    from ..core.dist_checkpointing import load as dist_load, load_common
+    args = get_args()
+    dist_load(_model_sharded_sd(model), base, strict=strict)
+    common = load_common(base)
+    if optimizer is not None and not args.no_load_optim and not args.finetune:
+        rank = dist.get_rank() if dist.is_initialized() else 0
+        opt_path = os.path.join(base, f"optim_rank{rank:05d}.pt")
+        if os.path.exists(opt_path):
+            optimizer.load_state_dict(
+                torch.load(opt_path, map_location="cpu",
+                           weights_only=False)["optimizer"])
+        if opt_param_scheduler is not None and                 "opt_param_scheduler" in common:
+            opt_param_scheduler.load_state_dict(common["opt_param_scheduler"])
+    if optimizer is not None and hasattr(optimizer, "reload_model_params")             and (args.no_load_optim or args.finetune):
+        optimizer.reload_model_params()
+    if not args.no_load_rng and not args.finetune and "rng_state" in common:
+        _restore_rng_state(common["rng_state"])
+    if args.rank == 0:
+        print(f"  loaded checkpoint (torch_dist) at iteration {iteration}")
+    num_flop = common.get("num_floating_point_operations_so_far", 0)
     return (0 if args.finetune else iteration), num_flop

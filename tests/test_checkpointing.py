@@ -28,8 +28,8 @@ def _run(extra, port):
 
 def test_save_and_resume(tmp_path):
     save = str(tmp_path / "ckpt")
-    out1 = _run(["--train-iters", "4", "--save", save, "--save-interval", "2"],
-                29631)
+    out1 = _run(["--train-iters", "4", "--save", save, "--save-interval", "2",
+                 "--ckpt-format", "torch"], 29631)
     assert os.path.exists(os.path.join(save, "latest_checkpointed_iteration.txt"))
     with open(os.path.join(save, "latest_checkpointed_iteration.txt")) as f:
         assert f.read().strip() == "4"
@@ -37,7 +37,7 @@ def test_save_and_resume(tmp_path):
                                        "model_optim_rng.pt"))
     # resume: continues from iteration 4 to 6
     out2 = _run(["--train-iters", "6", "--save", save, "--load", save,
-                 "--save-interval", "100"], 29632)
+                 "--ckpt-format", "torch", "--save-interval", "100"], 29632)
     assert "loaded checkpoint" in out2
     assert "iteration        5/6" in out2 or "iteration        6/6" in out2
     # iterations 1-4 must NOT rerun
@@ -62,3 +62,15 @@ def test_indexed_dataset_roundtrip(tmp_path):
     assert list(ds.get(1, offset=1, length=2)) == [8, 9]
     # flat .bin stream view used by GPTDataset
     assert list(ds.bin[:3]) == [1, 2, 3]
+
+
+def test_save_and_resume_torch_dist(tmp_path):
+    save = str(tmp_path / "ckpt_dist")
+    out1 = _run(["--train-iters", "3", "--save", save, "--save-interval", "3",
+                 "--ckpt-format", "torch_dist"], 29633)
+    assert os.path.exists(os.path.join(save, "iter_0000003", "index.json"))
+    out2 = _run(["--train-iters", "5", "--save", save, "--load", save,
+                 "--ckpt-format", "torch_dist", "--save-interval", "100"],
+                29634)
+    assert "loaded checkpoint (torch_dist)" in out2
+    assert "iteration        1/5" not in out2
