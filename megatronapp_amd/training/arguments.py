@@ -36,6 +36,13 @@ def parse_args(extra_args_provider=None, ignore_unknown_args=False):
         args, _ = parser.parse_known_args()
     else:
         args = parser.parse_args()
+    if getattr(args, "yaml_cfg", None):
+        from .yaml_arguments import apply_yaml_config
+        import sys
+        explicit = {parser._option_string_actions[tok.split("=", 1)[0]].dest
+                    for tok in sys.argv[1:]
+                    if tok.split("=", 1)[0] in parser._option_string_actions}
+        apply_yaml_config(args, args.yaml_cfg, explicit)
     return args
 
 
@@ -96,6 +103,8 @@ def _add_model_args(p):
 
 def _add_training_args(p):
     g = p.add_argument_group("training")
+    g.add_argument("--yaml-cfg", default=None,
+                   help="YAML config file; CLI flags override its values")
     g.add_argument("--micro-batch-size", type=int, default=1)
     g.add_argument("--global-batch-size", type=int, default=None)
     g.add_argument("--rampup-batch-size", nargs=3, type=int, default=None)

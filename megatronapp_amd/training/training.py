@@ -207,6 +207,25 @@ def training_log(loss_dict, total_loss_dict, learning_rate, iteration,
             writer.add_scalar(k, v, iteration)
         writer.add_scalar("learning-rate", learning_rate, iteration)
         writer.add_scalar("throughput", throughput_tps, iteration)
+    if report_memory_flag:
+        # once, after optimizer state exists (reference training.py:1745)
+        if (not torch.distributed.is_initialized()
+                or torch.distributed.get_rank() == 0):
+            from .theoretical_memory_usage import report_theoretical_memory
+            try:
+                report_theoretical_memory(
+                    args, num_microbatches=get_num_microbatches())
+            except Exception as e:  # never let a report kill training
+                print(f"  (theoretical memory report failed: {e})")
+            if torch.cuda.is_available():
+                mem = torch.cuda.memory_stats()
+                print(f"Measured memory (rank 0): allocated "
+                      f"{mem.get('allocated_bytes.all.current', 0) / 2**20:.0f}"
+                      f" MiB | reserved "
+                      f"{mem.get('reserved_bytes.all.current', 0) / 2**20:.0f}"
+                      f" MiB | max allocated "
+                      f"{mem.get('allocated_bytes.all.peak', 0) / 2**20:.0f} MiB")
+        report_memory_flag = False
     return report_memory_flag
 
 

@@ -1,0 +1,69 @@
+"""yaml_arguments + theoretical_memory_usage unit tests."""
+import os
+import sys
+
+import pytest
+
+
+def test_yaml_config_overlay(tmp_path, monkeypatch):
+    cfg = tmp_path / "cfg.yml"
+    cfg.write_text(
+        "model:\n  num-layers: 6\n  hidden_size: 320\n"
+        "training:\n  micro_batch_size: 4\n  save: ${AUX_SAVE}\n")
+    monkeypatch.setenv("AUX_SAVE", "/tmp/aux_ck")
+    monkeypatch.setattr(sys, "argv",
+                        ["x", "--yaml-cfg", str(cfg), "--hidden-size", "640"])
+    from megatronapp_amd.training.arguments import parse_args
+    a = parse_args(ignore_unknown_args=True)
+    assert a.num_layers == 6
+    assert a.hidden_size == 640          # CLI overrides YAML
+    assert a.micro_batch_size == 4
+    assert a.save == "/tmp/aux_ck"       # env interpolation
+
+
+def test_yaml_config_unknown_key(tmp_path):
+    cfg = tmp_path / "bad.yml"
+    cfg.write_text("model:\n  not_a_real_flag: 1\n")
+    from megatronapp_amd.training.arguments import parse_args
+    from megatronapp_amd.training.yaml_arguments import apply_yaml_config
+    import argparse
+    ns = argparse.Namespace(num_layers=2)
+    with pytest.raises(ValueError):
+        apply_yaml_config(ns, str(cfg))
+
+
+def _gpt3_1p3b_args(**over):
+    from types import SimpleNamespace
+    d = dict(hidden_size=2048, num_attention_heads=16, kv_channels=128,
+             num_layers=24, ffn_hidden_size=8192, swiglu=False,
+             group_query_attention=False, num_query_groups=None,
+             num_experts=None, padded_vocab_size=51200, vocab_size=50257,
+             untie_embeddings_and_output_weights=False,
+             pipeline_model_parallel_size=1, tensor_model_parallel_size=1,
+             data_parallel_size=1, use_distributed_optimizer=False,
+             seq_length=2048, micro_batch_size=2,
+             virtual_pipeline_model_parallel_size=None)
+    d.update(over)
+    return SimpleNamespace(**d)
+
+
+def test_theoretical_memory_gpt3_1p3b():
+    from megatronapp_amd.training.theoretical_memory_usage import (
+        compute_weight_and_optimizer_memory, compute_activation_memory,
+        report_theoretical_memory)
+    args = _gpt3_1p3b_args()
+    w = compute_weight_and_optimizer_memory(args)
+    # ~1.3B params to within 15%, 18 B/param
+    assert 1.1e9 * 18 < w < 1.6e9 * 18, w
+    # distributed optimizer across dp=8 shrinks state
+    w8 = compute_weight_and_optimizer_memory(
+        _gpt3_1p3b_args(use_distributed_optimizer=True, data_parallel_size=8))
+    assert w8 < w * 0.55
+    a = compute_activation_memory(args, num_microbatches=8)
+    assert a > 0
+    # TP halves activations
+    a2 = compute_activation_memory(
+        _gpt3_1p3b_args(tensor_model_parallel_size=2), num_microbatches=8)
+    assert abs(a2 - a / 2) / a < 0.01
+    total = report_theoretical_memory(args, num_microbatches=8)
+    assert total == pytest.approx(w + a)
