@@ -1,0 +1,161 @@
+#!/usr/bin/env python3
+"""Checkpoint format / topology converter.
+
+Reference: tools/checkpoint/convert.py (loader/saver plugin protocol that
+streams FULL, unsplit weights between a loader and a saver).  Same design
+here, collapsed to the formats this framework writes:
+
+  loaders:  torch_dist   sharded dir (index.json) at any TP/PP -> full tensors
+            legacy       mp_rank_* layout (TP=PP=1 only; for sharded legacy
+                         checkpoints re-save with --ckpt-format torch_dist)
+            consolidated a single .pt of {name: tensor}
+  savers:   torch_dist   single-shard-per-key sharded dir; loadable at ANY
+                         TP/PP by the overlap-window loader
+            consolidated single .pt of {name: tensor}
+
+Typical uses:
+  # make a topology-free checkpoint from a TP=2,PP=2 run:
+  python tools/checkpoint/convert.py \
+      --load ckpt/iter_0001000 --loader torch_dist \
+      --save ckpt_univ/iter_0001000 --saver torch_dist
+  # export full weights for inspection / external tooling:
+  python tools/checkpoint/convert.py \
+      --load ckpt/iter_0001000 --loader torch_dist \
+      --save model_full.pt --saver consolidated
+"""
+
+import argparse
+import json
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
+
+
+def load_torch_dist(path):
+    """Merge a sharded checkpoint's pieces into full tensors."""
+    with open(os.path.join(path, "index.json")) as f:
+        index = json.load(f)
+    cache = {}
+
+    def piece(fname):
+        if fname not in cache:
+            cache[fname] = torch.load(os.path.join(path, fname),
+                                      map_location="cpu", weights_only=False)
+        return cache[fname]
+
+    full = {}
+    for key, metas in index.items():
+        gshape = metas[0]["global_shape"]
+        out = torch.empty(gshape, dtype=piece(metas[0]["file"])[key]["tensor"].dtype)
+        covered = 0
+        for meta in metas:
+            t = piece(meta["file"])[key]["tensor"]
+            slices = tuple(slice(o, o + s)
+                           for o, s in zip(meta["offset"], meta["shape"]))
+            out[slices] = t
+            covered += t.numel()
+        if covered < out.numel():
+            raise RuntimeError(f"{key}: shards cover {covered}/{out.numel()} "
+                               f"elements — checkpoint incomplete")
+        full[key] = out
+    common = {}
+    cpath = os.path.join(path, "common.pt")
+    if os.path.exists(cpath):
+        common = torch.load(cpath, map_location="cpu", weights_only=False)
+    return full, common
+
+
+def load_legacy(path):
+    """mp_rank_00/model_optim_rng.pt (TP=PP=1 checkpoints only)."""
+    ranks = sorted(d for d in os.listdir(path) if d.startswith("mp_rank_"))
+    if ranks != ["mp_rank_00"]:
+        raise SystemExit(
+            f"legacy loader handles TP=PP=1 checkpoints; found {ranks}. "
+            f"Re-save the run with --ckpt-format torch_dist and use "
+            f"--loader torch_dist instead.")
+    sd = torch.load(os.path.join(path, "mp_rank_00", "model_optim_rng.pt"),
+                    map_location="cpu", weights_only=False)
+    model_keys = [k for k in sd if k == "model" or k.startswith("model")]
+    full = {}
+    for mk in model_keys:
+        prefix = "model." if mk == "model" else f"{mk}."
+        for name, t in sd[mk].items():
+            if torch.is_tensor(t):
+                full[prefix + name] = t
+    common = {k: v for k, v in sd.items() if k not in model_keys
+              and k not in ("optimizer", "rng_state")}
+    return full, common
+
+
+def load_consolidated(path):
+    blob = torch.load(path, map_location="cpu", weights_only=False)
+    if "weights" in blob:
+        return blob["weights"], blob.get("common", {})
+    return {k: v for k, v in blob.items() if torch.is_tensor(v)}, {}
+
+
+def save_torch_dist(full, common, path):
+    os.makedirs(path, exist_ok=True)
+    payload = {k: {"offset": [0] * t.dim(), "global_shape": list(t.shape),
+                   "tensor": t} for k, t in full.items()}
+    fname = "shards_rank00000.pt"
+    torch.save(payload, os.path.join(path, fname))
+    index = {k: [{"file": fname, "offset": [0] * t.dim(),
+                  "shape": list(t.shape), "global_shape": list(t.shape)}]
+             for k, t in full.items()}
+    with open(os.path.join(path, "index.json"), "w") as f:
+        json.dump(index, f)
+    torch.save(common, os.path.join(path, "common.pt"))
+    # tracker so load_checkpoint finds it when pointed at the parent dir
+    parent, leaf = os.path.split(os.path.normpath(path))
+    if leaf.startswith("iter_"):
+        with open(os.path.join(parent,
+                               "latest_checkpointed_iteration.txt"), "w") as f:
+            f.write(str(int(leaf[len("iter_"):])))
+
+
+def save_consolidated(full, common, path):
+    os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+    torch.save({"weights": full, "common": common}, path)
+
+
+LOADERS = {"torch_dist": load_torch_dist, "legacy": load_legacy,
+           "consolidated": load_consolidated}
+SAVERS = {"torch_dist": save_torch_dist, "consolidated": save_consolidated}
+
+
+def main():
+    p = argparse.ArgumentParser(description=__doc__,
+                                formatter_class=argparse.RawDescriptionHelpFormatter)
+    p.add_argument("--load", required=True,
+                   help="checkpoint iter dir (or .pt for consolidated)")
+    p.add_argument("--loader", choices=sorted(LOADERS), default="torch_dist")
+    p.add_argument("--save", required=True)
+    p.add_argument("--saver", choices=sorted(SAVERS), default="torch_dist")
+    p.add_argument("--dtype", default=None,
+                   choices=[None, "float32", "bfloat16", "float16"],
+                   help="optionally cast all weights")
+    p.add_argument("--inspect", action="store_true",
+                   help="print every key/shape/dtype after loading")
+    args = p.parse_args()
+
+    full, common = LOADERS[args.loader](args.load)
+    n_params = sum(t.numel() for t in full.values())
+    print(f"loaded {len(full)} tensors, {n_params / 1e6:.1f} M elements")
+    if args.inspect:
+        for k in sorted(full):
+            t = full[k]
+            print(f"  {k:70s} {tuple(t.shape)!s:24s} {t.dtype}")
+    if args.dtype:
+        dt = getattr(torch, args.dtype)
+        full = {k: (t.to(dt) if t.is_floating_point() else t)
+                for k, t in full.items()}
+    SAVERS[args.saver](full, common, args.save)
+    print(f"wrote {args.saver} checkpoint to {args.save}")
+
+
+if __name__ == "__main__":
+    main()
