@@ -22,14 +22,31 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import torch
 import torch.distributed as dist
 
+# arch="gpt" -> learned positions, GeLU, LayerNorm, biases, tied embeddings
+# arch="llama" -> rope, SwiGLU, RMSNorm, no biases, untied embeddings (+GQA)
 MODELS = {
-    # name: (layers, hidden, heads, ffn, seq, vocab)
     "gpt3-1.3b": dict(num_layers=24, hidden_size=2048, num_attention_heads=16,
-                      ffn_hidden_size=8192, seq_length=2048, vocab_size=51200),
+                      ffn_hidden_size=8192, seq_length=2048, vocab_size=51200,
+                      arch="gpt"),
     "gpt3-345m": dict(num_layers=24, hidden_size=1024, num_attention_heads=16,
-                      ffn_hidden_size=4096, seq_length=2048, vocab_size=51200),
+                      ffn_hidden_size=4096, seq_length=2048, vocab_size=51200,
+                      arch="gpt"),
     "gpt-tiny": dict(num_layers=4, hidden_size=256, num_attention_heads=4,
-                     ffn_hidden_size=1024, seq_length=512, vocab_size=8192),
+                     ffn_hidden_size=1024, seq_length=512, vocab_size=8192,
+                     arch="gpt"),
+    "llama3-8b": dict(num_layers=32, hidden_size=4096,
+                      num_attention_heads=32, num_query_groups=8,
+                      ffn_hidden_size=14336, seq_length=2048,
+                      vocab_size=128256, arch="llama"),
+    "llama-1b": dict(num_layers=16, hidden_size=2048,
+                     num_attention_heads=32, num_query_groups=8,
+                     ffn_hidden_size=8192, seq_length=2048,
+                     vocab_size=32000, arch="llama"),
+    "mixtral-8x1b": dict(num_layers=16, hidden_size=2048,
+                         num_attention_heads=32, num_query_groups=8,
+                         ffn_hidden_size=8192, seq_length=2048,
+                         vocab_size=32000, arch="llama",
+                         num_moe_experts=8, moe_router_topk=2),
 }
 
 BASELINE_TOKENS_PER_S = 16 * 2048 / 0.722  # BASELINE.md row 2 (4x RTX4090)
@@ -109,6 +126,7 @@ def main():
     torch.manual_seed(1234)
 
     bf16 = on_gpu
+    llama_style = spec.get("arch") == "llama"
     config = TransformerConfig(
         num_layers=spec["num_layers"], hidden_size=spec["hidden_size"],
         num_attention_heads=spec["num_attention_heads"],
@@ -117,10 +135,17 @@ def main():
         bf16=bf16, params_dtype=torch.bfloat16 if bf16 else torch.float32,
         pipeline_dtype=torch.bfloat16 if bf16 else torch.float32,
         hidden_dropout=0.0, attention_dropout=0.0,
-        # classic GPT-3 architecture: learned positions, GeLU MLP, LayerNorm
-        position_embedding_type="learned_absolute",
-        normalization="LayerNorm", activation_func="gelu",
-        add_bias_linear=True, masked_softmax_fusion=True,
+        position_embedding_type=("learned_absolute" if llama_style is False
+                                 else "rope"),
+        normalization="LayerNorm" if not llama_style else "RMSNorm",
+        activation_func="gelu" if not llama_style else "silu",
+        gated_linear_unit=llama_style,
+        add_bias_linear=not llama_style, masked_softmax_fusion=True,
+        num_query_groups=spec.get("num_query_groups"),
+        num_moe_experts=spec.get("num_moe_experts"),
+        moe_router_topk=spec.get("moe_router_topk", 2),
+        moe_router_load_balancing_type="aux_loss",
+        moe_aux_loss_coeff=0.01 if spec.get("num_moe_experts") else 0.0,
         finalize_model_grads_func=finalize_model_grads,
         sequence_parallel=(tp > 1))
 
@@ -131,12 +156,13 @@ def main():
         model = GPTModel(
             config=config,
             transformer_layer_spec=get_gpt_layer_local_spec(
-                normalization="LayerNorm",
+                normalization=config.normalization,
+                num_experts=spec.get("num_moe_experts"),
                 use_flash=(args.attention == "flash")),
             vocab_size=vocab, max_sequence_length=seq,
-            position_embedding_type="learned_absolute",
+            position_embedding_type=config.position_embedding_type,
             pre_process=pre, post_process=post,
-            share_embeddings_and_output_weights=True)
+            share_embeddings_and_output_weights=not llama_style)
     ddp_config = DistributedDataParallelConfig(
         overlap_grad_reduce=not args.no_overlap_grad_reduce,
         use_distributed_optimizer=(dp > 1),
