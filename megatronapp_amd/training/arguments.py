@@ -206,6 +206,8 @@ def _add_data_args(p):
     g.add_argument("--data-path", nargs="*", default=None)
     g.add_argument("--split", default="969, 30, 1")
     g.add_argument("--seq-length", type=int, default=1024)
+    g.add_argument("--decoder-seq-length", type=int, default=None,
+                   help="decoder sequence length (encoder-decoder models)")
     g.add_argument("--vocab-size", type=int, default=None)
     g.add_argument("--padded-vocab-size", type=int, default=None)
     g.add_argument("--make-vocab-size-divisible-by", type=int, default=128)
