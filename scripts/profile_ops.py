@@ -1,0 +1,52 @@
+"""Attribute GPU kernels to Python ops with torch.profiler.
+
+Runs a few GPT-3 1.3B training steps single-GPU and prints the op->kernel
+table sorted by CUDA time, so eager PyTorch kernels showing up in
+rocprofv3 stats (cast-adds, cats, reduces) can be traced to their source.
+
+Usage (on a GPU box):  python scripts/profile_ops.py [--steps 3]
+"""
+
+import argparse
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--model", default="gpt3-1.3b")
+    p.add_argument("--out", default="gpurun_out/op_profile.txt")
+    args = p.parse_args()
+
+    import bench
+    bench_args = ["bench.py", "--model", args.model, "--steps", "2",
+                  "--warmup", "1", "--global-batch-size", "4",
+                  "--micro-batch-size", "2"]
+    sys.argv = bench_args
+
+    # run bench once to build everything, capturing its step fn via profiler
+    # simpler: replicate the bench setup by calling main() under profiler
+    # for the timed region only is invasive; instead profile the whole
+    # short run — warmup pollution is acceptable for attribution.
+    from torch.profiler import profile, ProfilerActivity
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                 with_stack=False) as prof:
+        try:
+            bench.main()
+        except SystemExit:
+            pass
+    table = prof.key_averages().table(
+        sort_by="self_cuda_time_total", row_limit=60, max_src_column_width=120)
+    os.makedirs(os.path.dirname(args.out), exist_ok=True)
+    with open(args.out, "w") as f:
+        f.write(table)
+    print(table[:8000])
+
+
+if __name__ == "__main__":
+    main()
