@@ -24,6 +24,21 @@ def _gelu_tanh_grad(x):
         1.0 + 3 * 0.044715 * x * x)
 
 
+def _bias_grad(bias, dx):
+    """Bias grad for a fused activation: colsum into main_grad on GPU
+    (with a dummy grad so the DDP post-accumulate hook still fires),
+    eager sum otherwise."""
+    if bias is None or not bias.requires_grad:
+        return None
+    if (dx.is_cuda and _ops.have_ops() and hasattr(bias, "main_grad")
+            and hasattr(bias, "grad_added_to_main_grad")):
+        _ops.get_ops().colsum_accum(
+            dx.reshape(-1, dx.shape[-1]).contiguous(), bias.main_grad)
+        bias.grad_added_to_main_grad = True
+        return torch.empty(bias.shape, dtype=bias.dtype, device=bias.device)
+    return dx.reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype)
+
+
 class _BiasGeluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, bias):
@@ -41,10 +56,7 @@ class _BiasGeluFn(torch.autograd.Function):
         else:
             xf = (x.float() + bias.float()) if bias is not None else x.float()
             dx = (dy.float() * _gelu_tanh_grad(xf)).to(x.dtype)
-        dbias = None
-        if bias is not None and bias.requires_grad:
-            dbias = dx.reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype) \
-                if not dy.is_cuda else dx.reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype)
+        dbias = _bias_grad(bias, dx)
         return dx, dbias
 
 
@@ -74,9 +86,7 @@ class _BiasSwigluFn(torch.autograd.Function):
             d1 = dyf * x2 * (sig * (1 + x1 * (1 - sig)))
             d2 = dyf * silu
             dx = torch.cat([d1, d2], dim=-1).to(x.dtype)
-        dbias = None
-        if bias is not None and bias.requires_grad:
-            dbias = dx.reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype)
+        dbias = _bias_grad(bias, dx)
         return dx, dbias
 
 

@@ -12,9 +12,36 @@ from typing import Optional, Tuple
 
 import torch
 
+from ... import ops as _ops
+
+
+class _BiasAddResidualFn(torch.autograd.Function):
+    """x + bias + residual with the bias grad accumulated by colsum_accum
+    straight into the param's fp32 main_grad (no eager [s*b,h]->[h] sum
+    kernel, no separate cast-add in the DDP hook)."""
+
+    @staticmethod
+    def forward(ctx, x, bias, residual):
+        ctx.bias_param = bias
+        return x + bias + residual
+
+    @staticmethod
+    def backward(ctx, grad):
+        bias = ctx.bias_param
+        _ops.get_ops().colsum_accum(
+            grad.reshape(-1, grad.shape[-1]).contiguous(), bias.main_grad)
+        bias.grad_added_to_main_grad = True
+        dummy = torch.empty(bias.shape, dtype=bias.dtype, device=bias.device)
+        return grad, dummy, grad
+
 
 def _bias_dropout_add_func(x_with_bias, residual, prob, training):
     x, bias = x_with_bias
+    if (prob == 0.0 and bias is not None and x.is_cuda
+            and torch.is_grad_enabled() and _ops.have_ops()
+            and hasattr(bias, "main_grad")
+            and hasattr(bias, "grad_added_to_main_grad")):
+        return _BiasAddResidualFn.apply(x, bias, residual)
     if bias is not None:
         x = x + bias
     out = torch.nn.functional.dropout(x, p=prob, training=training)

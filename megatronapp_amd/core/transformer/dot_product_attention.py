@@ -73,8 +73,11 @@ class DotProductAttention(nn.Module):
             key = key.repeat_interleave(rep, dim=2)
             value = value.repeat_interleave(rep, dim=2)
 
-        q = query.reshape(sq, b * np_, hn).transpose(0, 1)          # [b*np, sq, hn]
-        k = key.reshape(sk, b * np_, hn).transpose(0, 1)            # [b*np, sk, hn]
+        # permute+reshape stays a VIEW even for the strided _SplitQKV
+        # outputs (b,np adjacent with nested strides); hipBLASLt handles
+        # the resulting lda/batch strides without materializing copies
+        q = query.permute(1, 2, 0, 3).reshape(b * np_, sq, hn)      # [b*np, sq, hn]
+        k = key.permute(1, 2, 0, 3).reshape(b * np_, sk, hn)        # [b*np, sk, hn]
         scores = torch.empty(b * np_, sq, sk, dtype=query.dtype,
                              device=query.device)
         scores = torch.baddbmm(scores, q, k.transpose(1, 2), beta=0.0,
@@ -99,7 +102,7 @@ class DotProductAttention(nn.Module):
             else:
                 probs = self.attention_dropout(probs)
 
-        v = value.reshape(sk, b * np_, hn).transpose(0, 1)          # [b*np, sk, hn]
+        v = value.permute(1, 2, 0, 3).reshape(b * np_, sk, hn)      # [b*np, sk, hn]
         context = torch.bmm(probs.view(b * np_, sq, sk), v)         # [b*np, sq, hn]
         context = context.transpose(0, 1).reshape(sq, b, np_ * hn)
         return context
@@ -164,6 +167,7 @@ class FlashAttention(nn.Module):
             return self._fallback(query, key, value, attention_mask,
                                   attn_mask_type, attention_bias, packed_seq_params)
         sq, b, np_, hn = query.shape
-        o = _FlashAttnFn.apply(query, key, value, self.softmax_scale, True,
+        o = _FlashAttnFn.apply(query.contiguous(), key.contiguous(),
+                               value.contiguous(), self.softmax_scale, True,
                                self.dropout_p)
         return o.reshape(sq, b, np_ * hn)

@@ -35,7 +35,7 @@ def main():
     # short run — warmup pollution is acceptable for attribution.
     from torch.profiler import profile, ProfilerActivity
     with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
-                 with_stack=False) as prof:
+                 with_stack=True) as prof:
         try:
             bench.main()
         except SystemExit:
@@ -46,6 +46,20 @@ def main():
     with open(args.out, "w") as f:
         f.write(table)
     print(table[:8000])
+    # stack attribution for the eager suspects
+    want = ("aten::add", "aten::add_", "aten::cat", "aten::sum",
+            "aten::copy_", "aten::mul", "aten::fill_")
+    lines = []
+    for ev in prof.key_averages(group_by_stack_n=12):
+        if ev.key in want and ev.device_time_total > 2000:
+            lines.append(f"== {ev.key}  cuda_total={ev.device_time_total/1e3:.1f}ms  "
+                         f"calls={ev.count}")
+            for fr in (ev.stack or [])[:12]:
+                lines.append(f"    {fr}")
+    stacks = "\n".join(lines)
+    with open(args.out.replace(".txt", "_stacks.txt"), "w") as f:
+        f.write(stacks)
+    print(stacks[:12000])
 
 
 if __name__ == "__main__":
