@@ -161,6 +161,30 @@ class DistributedOptimizer:
         grads = [self._shard_grad(i) for i in range(len(self.buffers))]
         return count_zeros_fp32(grads, parallel_state.get_model_parallel_group())
 
+    def _ranges_aligned(self, i) -> bool:
+        """The single-launch ranged kernel decides wd per float4; it needs
+        every no-wd boundary 4-aligned (true for transformer param sizes)."""
+        return all(s % 4 == 0 and e % 4 == 0 for s, e in self.no_wd_ranges[i])
+
+    def _nowd_device_ranges(self, i, device):
+        """Merged, sorted no-wd ranges as cached device int64 tensors."""
+        cache = getattr(self, "_nowd_cache", None)
+        if cache is None:
+            cache = self._nowd_cache = {}
+        if i not in cache:
+            merged = []
+            for s, e in sorted(self.no_wd_ranges[i]):
+                if merged and s <= merged[-1][1]:
+                    merged[-1][1] = max(merged[-1][1], e)
+                else:
+                    merged.append([s, e])
+            cache[i] = (
+                torch.tensor([r[0] for r in merged], dtype=torch.int64,
+                             device=device),
+                torch.tensor([r[1] for r in merged], dtype=torch.int64,
+                             device=device))
+        return cache[i]
+
     @torch.no_grad()
     def step(self):
         self.step_count += 1
@@ -187,6 +211,14 @@ class DistributedOptimizer:
                     _adam_step_flat(master, grad, m, v, lr,
                                     self.config.adam_beta1, self.config.adam_beta2,
                                     self.config.adam_eps, wd, self.step_count)
+                elif (master.is_cuda and _ops.have_ops()
+                      and hasattr(_ops.get_ops(), "adamw_flat_ranged")
+                      and self._ranges_aligned(i)):
+                    nw_s, nw_e = self._nowd_device_ranges(i, master.device)
+                    _ops.get_ops().adamw_flat_ranged(
+                        master, grad, m, v, nw_s, nw_e, lr,
+                        self.config.adam_beta1, self.config.adam_beta2,
+                        self.config.adam_eps, wd, self.step_count)
                 else:
                     # two-pass: run with wd over the whole shard is wrong for
                     # no-wd params, so stitch: wd pass on full shard minus
