@@ -365,3 +365,46 @@ def test_train_step_grads_fused_vs_eager():
         rel = (f - e).abs().max() / denom
         assert rel < 0.06, f"{name}: rel diff {rel:.4f}"
     destroy()
+
+
+@pytest.mark.gpu
+def test_adamw_flat_ranged_matches_per_range():
+    """Single-launch ranged AdamW == stitched per-range adamw_flat."""
+    import megatronapp_amd.ops as _ops_mod
+    ops = _ops_mod.get_ops()
+    torch.manual_seed(5)
+    n = 40960
+    ranges = [(4096, 8192), (12288, 12288 + 2048), (20480, 24576)]
+    lr, b1, b2, eps, wd, step = 1e-3, 0.9, 0.95, 1e-8, 0.1, 7
+
+    def init():
+        p = torch.randn(n, device="cuda", dtype=torch.float32)
+        g = torch.randn(n, device="cuda", dtype=torch.float32)
+        m = torch.randn(n, device="cuda", dtype=torch.float32).abs()
+        v = torch.randn(n, device="cuda", dtype=torch.float32).abs()
+        return p, g, m, v
+
+    torch.manual_seed(5)
+    p1, g1, m1, v1 = init()
+    torch.manual_seed(5)
+    p2, g2, m2, v2 = init()
+
+    nw_s = torch.tensor([r[0] for r in ranges], dtype=torch.int64,
+                        device="cuda")
+    nw_e = torch.tensor([r[1] for r in ranges], dtype=torch.int64,
+                        device="cuda")
+    ops.adamw_flat_ranged(p1, g1, m1, v1, nw_s, nw_e, lr, b1, b2, eps, wd,
+                          step)
+
+    cursor = 0
+    for (s, e) in ranges + [(n, n)]:
+        if cursor < s:
+            ops.adamw_flat(p2[cursor:s], g2[cursor:s], m2[cursor:s],
+                           v2[cursor:s], lr, b1, b2, eps, wd, step)
+        if s < e:
+            ops.adamw_flat(p2[s:e], g2[s:e], m2[s:e], v2[s:e], lr, b1, b2,
+                           eps, 0.0, step)
+        cursor = max(cursor, e)
+
+    for a, b, name in ((p1, p2, "p"), (m1, m2, "m"), (v1, v2, "v")):
+        assert torch.equal(a, b), f"{name} differs"
