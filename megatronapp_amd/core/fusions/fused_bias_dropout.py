@@ -23,6 +23,10 @@ class _BiasAddResidualFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, bias, residual):
         ctx.bias_param = bias
+        if (x.is_contiguous() and residual.is_contiguous()
+                and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0
+                and hasattr(_ops.get_ops(), "bias_add_residual")):
+            return _ops.get_ops().bias_add_residual(x, bias, residual)
         return x + bias + residual
 
     @staticmethod

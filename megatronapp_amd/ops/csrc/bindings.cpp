@@ -16,6 +16,8 @@ void launch_layernorm_fwd(const void*, const void*, const void*, void*, float*,
 void launch_layernorm_bwd(const void*, const void*, const void*, const float*,
                           const float*, void*, float*, float*, float*, float*,
                           int, int, int, hipStream_t);
+void launch_bias_add_residual(const void*, const void*, const void*, void*,
+                              long, int, hipStream_t);
 void launch_bias_gelu_fwd(const void*, const void*, void*, long, int,
                           hipStream_t);
 void launch_bias_gelu_bwd(const void*, const void*, const void*, void*, long,
@@ -145,6 +147,21 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, c10::optional<torch::Tensor> bias) 
   launch_bias_gelu_fwd(x.data_ptr(),
                        bias.has_value() ? bias->data_ptr() : nullptr,
                        y.data_ptr(), x.numel(), F, cur_stream());
+  return y;
+}
+
+torch::Tensor bias_add_residual(torch::Tensor x, torch::Tensor bias,
+                                torch::Tensor res) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16);
+  TORCH_CHECK(x.is_contiguous() && res.is_contiguous() &&
+              bias.is_contiguous());
+  TORCH_CHECK(x.sizes() == res.sizes());
+  int F = bias.numel();
+  TORCH_CHECK(x.size(-1) == F && F % 8 == 0);
+  auto y = torch::empty_like(x);
+  launch_bias_add_residual(x.data_ptr(), bias.data_ptr(), res.data_ptr(),
+                           y.data_ptr(), x.numel(), F,
+                           at::cuda::getCurrentCUDAStream());
   return y;
 }
 
@@ -383,6 +400,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
   mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
   mod.def("adamw_flat", &adamw_flat);
+  mod.def("bias_add_residual", &bias_add_residual);
   mod.def("adamw_flat_ranged", &adamw_flat_ranged);
   mod.def("wgrad_accum", &wgrad_accum);
   mod.def("attn_fwd", &attn_fwd);

@@ -408,3 +408,16 @@ def test_adamw_flat_ranged_matches_per_range():
 
     for a, b, name in ((p1, p2, "p"), (m1, m2, "m"), (v1, v2, "v")):
         assert torch.equal(a, b), f"{name} differs"
+
+
+@pytest.mark.gpu
+def test_bias_add_residual():
+    import megatronapp_amd.ops as _ops_mod
+    ops = _ops_mod.get_ops()
+    torch.manual_seed(9)
+    x = torch.randn(128, 4, 256, device="cuda", dtype=torch.bfloat16)
+    r = torch.randn_like(x)
+    b = torch.randn(256, device="cuda", dtype=torch.bfloat16)
+    out = ops.bias_add_residual(x, b, r)
+    ref = (x.float() + b.float() + r.float()).to(torch.bfloat16)
+    assert (out.float() - ref.float()).abs().max() < 2e-2
