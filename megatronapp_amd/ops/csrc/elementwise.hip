@@ -214,8 +214,30 @@ void launch_bias_swiglu_bwd(const void* dy, const void* x, const void* bias,
 // ----------------------------------------------------- bias-grad column sum
 // dbias[F] (fp32, += accumulate) from dy [R, F] bf16.  2D grid: x over
 // column chunks, y over row chunks; one atomicAdd per (block, col).
+// each lane owns 8 adjacent columns (one bf16x8 = 16 B load, fully
+// coalesced across the 256-thread block = 4 KB per row) and a block strip
+// of rows; per-column fp32 partials land with one atomicAdd each.
 __global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
-                                    float* __restrict__ out, long R, int F) {
+                                    float* __restrict__ out, long R, int F,
+                                    int rows_per_block) {
+  const int col = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (col >= F) return;
+  const long r0 = (long)blockIdx.y * rows_per_block;
+  const long r1 = min(R, r0 + rows_per_block);
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (long r = r0; r < r1; ++r) {
+    short8v v = *(const short8v*)(dy + r * F + col);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf2f((unsigned short)v[j]);
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(out + col + j, acc[j]);
+}
+
+// scalar fallback for F % 8 != 0
+__global__ void colsum_accum_scalar_kernel(const unsigned short* __restrict__ dy,
+                                           float* __restrict__ out, long R,
+                                           int F) {
   const int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= F) return;
   const long r0 = (long)blockIdx.y * 128;
@@ -227,9 +249,21 @@ __global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
 
 void launch_colsum_accum(const void* dy, float* out, long R, int F,
                          hipStream_t s) {
-  dim3 grid((F + 255) / 256, (unsigned)((R + 127) / 128));
+  if (F % 8 != 0) {
+    dim3 grid((F + 255) / 256, (unsigned)((R + 127) / 128));
+    hipLaunchKernelGGL(colsum_accum_scalar_kernel, grid, dim3(256), 0, s,
+                       (const unsigned short*)dy, out, R, F);
+    HIP_CHECK_LAUNCH();
+    return;
+  }
+  const int gx = (F / 8 + 255) / 256;
+  int target_y = 768 / gx;
+  if (target_y < 1) target_y = 1;
+  int rpb = (int)((R + target_y - 1) / target_y);
+  if (rpb < 1) rpb = 1;
+  dim3 grid(gx, (unsigned)((R + rpb - 1) / rpb));
   hipLaunchKernelGGL(colsum_accum_kernel, grid, dim3(256), 0, s,
-                     (const unsigned short*)dy, out, R, F);
+                     (const unsigned short*)dy, out, R, F, rpb);
   HIP_CHECK_LAUNCH();
 }
 
