@@ -37,8 +37,9 @@ class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
     def backward(ctx, dy):
         (probs,) = ctx.saved_tensors
         if dy.is_cuda:
-            dx = _ops.get_ops().scaled_softmax_bwd(dy.contiguous(), probs,
-                                                   ctx.scale)
+            # causal-aware bwd: only the valid row prefix is read
+            dx = _ops.get_ops().scaled_upper_triang_masked_softmax_bwd(
+                dy.contiguous(), probs, ctx.scale)
         else:
             dyf = dy.float()
             p = probs.float()

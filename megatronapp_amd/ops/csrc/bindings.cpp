@@ -32,6 +32,8 @@ void launch_softmax_causal_fwd(const void*, void*, long, int, int, float,
                                hipStream_t);
 void launch_softmax_masked_fwd(const void*, const void*, void*, long, int, int,
                                int, float, hipStream_t);
+void launch_softmax_causal_bwd(const void*, const void*, void*, long, int,
+                               int, float, hipStream_t);
 void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
                         hipStream_t);
 void launch_adamw_flat_ranged(float*, const float*, float*, float*,
@@ -273,6 +275,19 @@ torch::Tensor scaled_softmax_bwd(torch::Tensor dy, torch::Tensor p,
   return dx;
 }
 
+torch::Tensor scaled_upper_triang_masked_softmax_bwd(torch::Tensor dy,
+                                                     torch::Tensor p,
+                                                     double scale) {
+  check_bf16(dy, "dy");
+  const int sk = (int)dy.size(-1);
+  const int sq = (int)dy.size(-2);
+  const long rows = dy.numel() / sk;
+  auto dx = torch::empty_like(dy);
+  launch_softmax_causal_bwd(dy.data_ptr(), p.data_ptr(), dx.data_ptr(), rows,
+                            sq, sk, (float)scale, cur_stream());
+  return dx;
+}
+
 torch::Tensor colsum_accum(torch::Tensor dy, torch::Tensor out) {
   check_bf16(dy, "dy");
   TORCH_CHECK(out.scalar_type() == torch::kFloat32 && out.is_contiguous());
@@ -400,6 +415,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
   mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
   mod.def("adamw_flat", &adamw_flat);
+  mod.def("scaled_upper_triang_masked_softmax_bwd",
+          &scaled_upper_triang_masked_softmax_bwd);
   mod.def("bias_add_residual", &bias_add_residual);
   mod.def("adamw_flat_ranged", &adamw_flat_ranged);
   mod.def("wgrad_accum", &wgrad_accum);
