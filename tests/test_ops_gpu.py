@@ -138,7 +138,7 @@ def test_rope_fwd_bwd():
     assert torch.allclose(dx.float(), tr.grad, atol=ATOL, rtol=RTOL)
 
 
-@pytest.mark.parametrize("sq", [128, 512])
+@pytest.mark.parametrize("sq", [128, 512, 2048, 4096])
 def test_softmax_causal(sq):
     torch.manual_seed(5)
     b_np = 8
@@ -150,10 +150,13 @@ def test_softmax_causal(sq):
     assert torch.allclose(y.float(), ref, atol=ATOL, rtol=RTOL)
 
     dy = torch.randn_like(x)
-    dx = _ops().scaled_softmax_bwd(dy, y, scale)
     xr = x.float().clone().requires_grad_(True)
     torch.softmax((xr * scale).masked_fill(mask, float("-inf")), -1).backward(dy.float())
+    dx = _ops().scaled_softmax_bwd(dy, y, scale)
     assert torch.allclose(dx.float(), xr.grad, atol=ATOL, rtol=RTOL)
+    # causal-aware bwd (register-cached prefix reads)
+    dx_c = _ops().scaled_upper_triang_masked_softmax_bwd(dy, y, scale)
+    assert torch.allclose(dx_c.float(), xr.grad, atol=ATOL, rtol=RTOL)
 
 
 def test_softmax_masked():
