@@ -1,0 +1,36 @@
+import torch, time
+import megatronapp_amd.ops as O
+ops = O.get_ops()
+def t(fn, iters=30):
+    for _ in range(5): fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters): fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6
+# softmax shapes: [b*np, sq, sk] = [32, 2048, 2048] bf16
+x = torch.randn(32, 2048, 2048, device="cuda", dtype=torch.bfloat16)
+y = ops.scaled_upper_triang_masked_softmax_fwd(x, 0.08)
+dy = torch.randn_like(x)
+us = t(lambda: ops.scaled_upper_triang_masked_softmax_fwd(x, 0.08))
+print(f"softmax causal fwd: {us:.1f} us  ({(x.numel()*2*1.5)/us/1e3:.0f} GB/s eff)")
+us = t(lambda: ops.scaled_upper_triang_masked_softmax_bwd(dy, y, 0.08))
+print(f"softmax causal bwd: {us:.1f} us")
+us = t(lambda: ops.scaled_softmax_bwd(dy, y, 0.08))
+print(f"softmax generic bwd: {us:.1f} us")
+# colsum: [4096, 2048] bf16 -> [2048] f32
+g = torch.randn(4096, 2048, device="cuda", dtype=torch.bfloat16)
+acc = torch.zeros(2048, device="cuda", dtype=torch.float32)
+us = t(lambda: ops.colsum_accum(g, acc))
+print(f"colsum 4096x2048: {us:.1f} us ({g.numel()*2/us/1e3:.0f} GB/s)")
+g2 = torch.randn(4096, 8192, device="cuda", dtype=torch.bfloat16)
+acc2 = torch.zeros(8192, device="cuda", dtype=torch.float32)
+us = t(lambda: ops.colsum_accum(g2, acc2))
+print(f"colsum 4096x8192: {us:.1f} us ({g2.numel()*2/us/1e3:.0f} GB/s)")
+# bias_add_residual
+xa = torch.randn(2048, 2, 2048, device="cuda", dtype=torch.bfloat16)
+ra = torch.randn_like(xa); ba = torch.randn(2048, device="cuda", dtype=torch.bfloat16)
+us = t(lambda: ops.bias_add_residual(xa, ba, ra))
+print(f"bias_add_residual: {us:.1f} us ({xa.numel()*2*3/us/1e3:.0f} GB/s)")
+us = t(lambda: (xa + ba + ra))
+print(f"eager x+b+r:       {us:.1f} us")
