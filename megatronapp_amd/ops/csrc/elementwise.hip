@@ -257,10 +257,11 @@ void launch_colsum_accum(const void* dy, float* out, long R, int F,
     return;
   }
   const int gx = (F / 8 + 255) / 256;
-  int target_y = 768 / gx;
-  if (target_y < 1) target_y = 1;
+  // atomic traffic per column == grid.y: keep it small (each block then
+  // streams a long sequential row strip, which is what saturates HBM)
+  int target_y = 64;
   int rpb = (int)((R + target_y - 1) / target_y);
-  if (rpb < 1) rpb = 1;
+  if (rpb < 8) rpb = 8;
   dim3 grid(gx, (unsigned)((R + rpb - 1) / rpb));
   hipLaunchKernelGGL(colsum_accum_kernel, grid, dim3(256), 0, s,
                      (const unsigned short*)dy, out, R, F, rpb);
