@@ -43,8 +43,6 @@ void launch_adamw_flat(float*, const float*, float*, float*, long, float,
                        float, float, float, float, int, hipStream_t);
 void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
 void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
-void launch_colsum_accum_staged(const void*, float*, float*, int, long, int,
-                                hipStream_t);
 void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
@@ -295,18 +293,8 @@ torch::Tensor colsum_accum(torch::Tensor dy, torch::Tensor out) {
   TORCH_CHECK(out.scalar_type() == torch::kFloat32 && out.is_contiguous());
   const int F = (int)dy.size(-1);
   const long R = dy.numel() / F;
-  if (F % 8 == 0 && R >= 256) {
-    // two-stage (atomic-free): partials scratch from the caching
-    // allocator, folded by a second tiny kernel
-    const int ny = 256;
-    auto part = torch::empty({ny, F}, out.options());
-    launch_colsum_accum_staged(dy.data_ptr(), out.data_ptr<float>(),
-                               part.data_ptr<float>(), ny, R, F,
-                               cur_stream());
-  } else {
-    launch_colsum_accum(dy.data_ptr(), out.data_ptr<float>(), R, F,
-                        cur_stream());
-  }
+  launch_colsum_accum(dy.data_ptr(), out.data_ptr<float>(), R, F,
+                      cur_stream());
   return out;
 }
 

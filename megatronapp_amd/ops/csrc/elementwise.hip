@@ -234,34 +234,6 @@ __global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
   for (int j = 0; j < 8; ++j) atomicAdd(out + col + j, acc[j]);
 }
 
-// two-stage variant: blocks write private [grid.y, F] fp32 partials (no
-// atomics), then a second kernel folds the partials into main_grad.
-__global__ void colsum_partials_kernel(const unsigned short* __restrict__ dy,
-                                       float* __restrict__ part, long R,
-                                       int F, int rows_per_block) {
-  const int col = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  if (col >= F) return;
-  const long r0 = (long)blockIdx.y * rows_per_block;
-  const long r1 = min(R, r0 + rows_per_block);
-  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  for (long r = r0; r < r1; ++r) {
-    short8v v = *(const short8v*)(dy + r * F + col);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) acc[j] += bf2f((unsigned short)v[j]);
-  }
-  float* dst = part + (long)blockIdx.y * F + col;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) dst[j] = acc[j];
-}
-
-__global__ void colsum_fold_kernel(const float* __restrict__ part,
-                                   float* __restrict__ out, int F, int ny) {
-  const int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= F) return;
-  float acc = 0.f;
-  for (int y = 0; y < ny; ++y) acc += part[(long)y * F + col];
-  out[col] += acc;
-}
 
 // scalar fallback for F % 8 != 0
 __global__ void colsum_accum_scalar_kernel(const unsigned short* __restrict__ dy,
@@ -274,19 +246,6 @@ __global__ void colsum_accum_scalar_kernel(const unsigned short* __restrict__ dy
   float acc = 0.f;
   for (long r = r0; r < r1; ++r) acc += bf2f(dy[r * F + col]);
   atomicAdd(out + col, acc);
-}
-
-void launch_colsum_accum_staged(const void* dy, float* out, float* part,
-                                int ny, long R, int F, hipStream_t s) {
-  const int gx = (F / 8 + 255) / 256;
-  int rpb = (int)((R + ny - 1) / ny);
-  if (rpb < 1) rpb = 1;
-  dim3 grid(gx, (unsigned)((R + rpb - 1) / rpb));
-  hipLaunchKernelGGL(colsum_partials_kernel, grid, dim3(256), 0, s,
-                     (const unsigned short*)dy, part, R, F, rpb);
-  hipLaunchKernelGGL(colsum_fold_kernel, dim3((F + 255) / 256), dim3(256), 0,
-                     s, part, out, F, (int)grid.y);
-  HIP_CHECK_LAUNCH();
 }
 
 void launch_colsum_accum(const void* dy, float* out, long R, int F,
