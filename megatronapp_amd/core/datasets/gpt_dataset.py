@@ -72,30 +72,107 @@ class MockGPTDataset(torch.utils.data.Dataset):
         return _build_sample(tokens, self.config)
 
 
+def _build_sample_idx_py(sizes, doc_idx, seq_length, num_epochs,
+                         tokens_per_epoch):
+    """Pure-python reference for the native build_sample_idx (tests)."""
+    num_samples = (num_epochs * tokens_per_epoch - 1) // seq_length
+    out = np.zeros((num_samples + 1, 2), dtype=np.int64)
+    doc_pos, doc_off = 0, 0
+    for s in range(1, num_samples + 1):
+        remaining = seq_length
+        while remaining > 0:
+            doc_len = sizes[doc_idx[doc_pos]] - doc_off
+            if doc_len > remaining:
+                doc_off += remaining
+                remaining = 0
+            else:
+                remaining -= doc_len
+                doc_pos += 1
+                doc_off = 0
+        out[s] = (doc_pos, doc_off)
+    return out
+
+
 class GPTDataset(torch.utils.data.Dataset):
-    """Causal-LM windows over a flat token stream (np.memmap .npy/.bin)."""
+    """Causal-LM samples (reference gpt_dataset.py three-index design).
+
+    With a reference-format .idx/.bin pair: documents are shuffled per
+    epoch (doc_idx), samples map to (document position, offset) honoring
+    document boundaries via the native ``build_sample_idx`` helper, and a
+    shuffle_idx permutes sample order — so a sample can span documents and
+    epochs differ in order, like the reference.  Flat .npy/.bin streams
+    fall back to fixed windows."""
 
     def __init__(self, config: GPTDatasetConfig, token_file: str,
                  num_samples: Optional[int] = None, name: str = "gpt"):
         self.config = config
         self.name = name
+        self.indexed = None
         if token_file.endswith(".npy"):
             self.tokens = np.load(token_file, mmap_mode="r")
         elif os.path.exists(token_file + ".idx"):
-            # reference-format indexed dataset: the .bin is a flat token
-            # stream; document boundaries live in the .idx
             from .indexed_dataset import IndexedDataset
-            self.tokens = IndexedDataset(token_file).bin
+            self.indexed = IndexedDataset(token_file)
+            self.tokens = self.indexed.bin
         else:
             self.tokens = np.memmap(token_file, dtype=np.int32, mode="r")
-        max_samples = (len(self.tokens) - 1) // config.sequence_length
-        self.num_samples = min(num_samples or max_samples, max_samples)
+
+        seq = config.sequence_length
+        if self.indexed is not None and self.indexed.document_count > 1:
+            self._build_doc_aware_indices(num_samples)
+        else:
+            max_samples = (len(self.tokens) - 1) // seq
+            self.num_samples = min(num_samples or max_samples, max_samples)
+            self.sample_idx = None
+
+    def _build_doc_aware_indices(self, num_samples):
+        cfg = self.config
+        seq = cfg.sequence_length
+        sizes = self.indexed.sequence_lengths
+        docs = np.arange(len(sizes), dtype=np.int32)
+        tokens_per_epoch = int(sizes.sum())
+        want = num_samples or max(1, (tokens_per_epoch - 1) // seq)
+        num_epochs = max(1, -(-(want * seq + 1) // tokens_per_epoch))
+        rng = np.random.RandomState(cfg.random_seed)
+        doc_idx = np.concatenate(
+            [rng.permutation(docs) for _ in range(num_epochs)])
+        try:
+            from .build_helpers import load_helpers
+            self.sample_idx = load_helpers().build_sample_idx(
+                sizes.astype(np.int32), doc_idx, seq, num_epochs,
+                tokens_per_epoch)
+        except Exception:
+            self.sample_idx = _build_sample_idx_py(
+                sizes, doc_idx, seq, num_epochs, tokens_per_epoch)
+        self.doc_idx = doc_idx
+        avail = self.sample_idx.shape[0] - 1
+        self.num_samples = min(want, avail)
+        self.shuffle_idx = rng.permutation(avail)[:self.num_samples]
 
     def __len__(self):
         return self.num_samples
 
+    def _doc_aware_window(self, idx):
+        seq = self.config.sequence_length
+        i = int(self.shuffle_idx[idx])
+        pos0, off0 = self.sample_idx[i]
+        pos1, off1 = self.sample_idx[i + 1]
+        parts = []
+        need = seq + 1
+        pos, off = int(pos0), int(off0)
+        while need > 0:
+            doc = int(self.doc_idx[pos])
+            chunk = self.indexed.get(doc, offset=off, length=need)
+            parts.append(chunk)
+            need -= len(chunk)
+            pos += 1
+            off = 0
+        return np.concatenate(parts)
+
     def __getitem__(self, idx):
         s = self.config.sequence_length
+        if self.sample_idx is not None:
+            return _build_sample(self._doc_aware_window(idx), self.config)
         start = idx * s
         window = np.asarray(self.tokens[start:start + s + 1])
         return _build_sample(window, self.config)

@@ -74,3 +74,45 @@ def test_save_and_resume_torch_dist(tmp_path):
                 29634)
     assert "loaded checkpoint (torch_dist)" in out2
     assert "iteration        1/5" not in out2
+
+
+def test_doc_aware_sample_idx(tmp_path):
+    """Native build_sample_idx == python reference; doc-aware GPTDataset
+    yields seq+1 windows that respect the shuffled document stream."""
+    import numpy as np
+    from megatronapp_amd.core.datasets.build_helpers import load_helpers
+    from megatronapp_amd.core.datasets.gpt_dataset import (
+        GPTDataset, GPTDatasetConfig, _build_sample_idx_py)
+    from megatronapp_amd.core.datasets.indexed_dataset import (
+        IndexedDatasetBuilder)
+
+    rng = np.random.RandomState(0)
+    sizes = rng.randint(3, 40, size=57).astype(np.int32)
+    doc_idx = np.concatenate([rng.permutation(57).astype(np.int32)
+                              for _ in range(3)])
+    tpe = int(sizes.sum())
+    h = load_helpers()
+    a = h.build_sample_idx(sizes, doc_idx, 16, 3, tpe)
+    b = _build_sample_idx_py(sizes, doc_idx, 16, 3, tpe)
+    assert np.array_equal(a, b)
+
+    # real dataset round trip
+    prefix = str(tmp_path / "docs")
+    builder = IndexedDatasetBuilder(prefix)
+    stream = {}
+    for d in range(10):
+        toks = rng.randint(0, 1000, size=rng.randint(20, 60)).astype(np.int32)
+        builder.add_item(toks)
+        builder.end_document()
+        stream[d] = toks
+    builder.finalize()
+    cfg = GPTDatasetConfig(sequence_length=16, random_seed=7, mock=False)
+    ds = GPTDataset(cfg, prefix, num_samples=12)
+    assert len(ds) == 12
+    for i in range(12):
+        s = ds[i]
+        assert s["tokens"].shape == (16,)
+        # labels are the stream shifted by one
+        full = ds._doc_aware_window(i)
+        assert np.array_equal(s["tokens"].numpy(), full[:-1])
+        assert np.array_equal(s["labels"].numpy(), full[1:])
