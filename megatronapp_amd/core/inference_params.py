@@ -30,7 +30,10 @@ class InferenceParams:
         if layer_number not in self.key_value_memory_dict:
             k_cache = torch.empty(self.max_sequence_length, b, ng, hd,
                                   dtype=key.dtype, device=key.device)
-            v_cache = torch.empty_like(k_cache)
+            # v head dim may differ from k (MLA: v_head_dim != qk dims)
+            v_cache = torch.empty(self.max_sequence_length, b, ng,
+                                  value.shape[-1], dtype=value.dtype,
+                                  device=value.device)
             self.key_value_memory_dict[layer_number] = (k_cache, v_cache)
         k_cache, v_cache = self.key_value_memory_dict[layer_number]
         start = self.sequence_len_offset

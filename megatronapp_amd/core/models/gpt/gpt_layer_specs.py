@@ -29,10 +29,16 @@ def _norm_spec(normalization: str):
 def get_gpt_layer_local_spec(num_experts: int = None, moe_grouped_gemm: bool = False,
                              qk_layernorm: bool = False, *, normalization: str = "LayerNorm",
                              use_flash: bool = True,
+                             multi_latent_attention: bool = False,
                              attn_mask_type: AttnMaskType = AttnMaskType.causal
                              ) -> ModuleSpec:
     norm = _norm_spec(normalization)
     core_attn = FlashAttention if use_flash else DotProductAttention
+    if multi_latent_attention:
+        # MLA's concatenated qk dim (192) never matches the flash kernel's
+        # supported head sizes; the fused baddbmm+softmax path handles the
+        # asymmetric v_head_dim
+        core_attn = DotProductAttention
     if num_experts is None:
         mlp = ModuleSpec(module=MLP, submodules=MLPSubmodules(
             linear_fc1=ColumnParallelLinear, linear_fc2=RowParallelLinear))
@@ -47,21 +53,39 @@ def get_gpt_layer_local_spec(num_experts: int = None, moe_grouped_gemm: bool = F
                                       submodules=MLPSubmodules(
                                           linear_fc1=ColumnParallelLinear,
                                           linear_fc2=RowParallelLinear))))
+    if multi_latent_attention:
+        from ...transformer.multi_latent_attention import (
+            MLASelfAttention, MLASelfAttentionSubmodules)
+        attn_spec = ModuleSpec(
+            module=MLASelfAttention,
+            params={"attn_mask_type": attn_mask_type},
+            submodules=MLASelfAttentionSubmodules(
+                linear_q_proj=ColumnParallelLinear,
+                linear_q_up_proj=ColumnParallelLinear,
+                linear_kv_up_proj=ColumnParallelLinear,
+                core_attention=core_attn,
+                linear_proj=RowParallelLinear,
+                q_layernorm=norm,
+                kv_layernorm=norm,
+            ),
+        )
+    else:
+        attn_spec = ModuleSpec(
+            module=SelfAttention,
+            params={"attn_mask_type": attn_mask_type},
+            submodules=SelfAttentionSubmodules(
+                linear_qkv=ColumnParallelLinear,
+                core_attention=core_attn,
+                linear_proj=RowParallelLinear,
+                q_layernorm=norm if qk_layernorm else None,
+                k_layernorm=norm if qk_layernorm else None,
+            ),
+        )
     return ModuleSpec(
         module=TransformerLayer,
         submodules=TransformerLayerSubmodules(
             input_layernorm=norm,
-            self_attention=ModuleSpec(
-                module=SelfAttention,
-                params={"attn_mask_type": attn_mask_type},
-                submodules=SelfAttentionSubmodules(
-                    linear_qkv=ColumnParallelLinear,
-                    core_attention=core_attn,
-                    linear_proj=RowParallelLinear,
-                    q_layernorm=norm if qk_layernorm else None,
-                    k_layernorm=norm if qk_layernorm else None,
-                ),
-            ),
+            self_attention=attn_spec,
             self_attn_bda=None,
             pre_mlp_layernorm=norm,
             mlp=mlp,

@@ -102,9 +102,10 @@ class DotProductAttention(nn.Module):
             else:
                 probs = self.attention_dropout(probs)
 
-        v = value.permute(1, 2, 0, 3).reshape(b * np_, sk, hn)      # [b*np, sk, hn]
-        context = torch.bmm(probs.view(b * np_, sq, sk), v)         # [b*np, sq, hn]
-        context = context.transpose(0, 1).reshape(sq, b, np_ * hn)
+        hn_v = value.shape[-1]  # MLA: v_head_dim may differ from qk dim
+        v = value.permute(1, 2, 0, 3).reshape(b * np_, sk, hn_v)    # [b*np, sk, hn_v]
+        context = torch.bmm(probs.view(b * np_, sq, sk), v)         # [b*np, sq, hn_v]
+        context = context.transpose(0, 1).reshape(sq, b, np_ * hn_v)
         return context
 
 
