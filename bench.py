@@ -85,7 +85,7 @@ def main():
         dist.init_process_group(backend=backend, rank=rank,
                                 world_size=world_size)
     if on_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
     device = torch.device("cuda", local_rank) if on_gpu else torch.device("cpu")
 
     n = world_size

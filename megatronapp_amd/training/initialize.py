@@ -58,7 +58,8 @@ def _initialize_distributed(args):
             backend=backend, world_size=args.world_size, rank=args.rank,
             timeout=timedelta(minutes=args.distributed_timeout_minutes))
     if torch.cuda.is_available():
-        torch.cuda.set_device(args.local_rank)
+        # modulo so oversubscribed test topologies (2 ranks, 1 GPU) run
+        torch.cuda.set_device(args.local_rank % torch.cuda.device_count())
 
     if parallel_state.model_parallel_is_initialized():
         return
