@@ -102,3 +102,42 @@ def test_pretrain_with_dpp(extra, tmp_path):
         capture_output=True, text=True, cwd=REPO, timeout=420)
     assert out.returncode == 0, out.stderr[-4000:]
     assert "lm loss" in out.stdout
+
+
+def _channel_proc_cuda(rank):
+    """Same exchange with CUDA tensors: exercises the hipMemcpy staging
+    in/out of the shm slot on both sides."""
+    _ensure_built()
+    sys.path.insert(0, REPO)
+    import torch
+    from megatronapp_amd.dpp.transport import _load
+    c = _load()
+    torch.cuda.set_device(0)
+    slot = 1024 * 4
+    if rank == 0:
+        c.init_channel("fwd", 10, 11, slot, 4, False)
+        for mb in (2, 0, 1):
+            t = torch.full((1024,), float(mb) + 0.5, dtype=torch.float32,
+                           device="cuda")
+            c.put_tensor("fwd", 10, 11, 0, mb, t)
+        torch.cuda.synchronize()
+    else:
+        c.init_channel("fwd", 10, 11, slot, 4, True)
+        for mb in (0, 1, 2):
+            out = torch.empty(1024, dtype=torch.float32, device="cuda")
+            c.get_tensor("fwd", 10, 11, 0, mb, out)
+            torch.cuda.synchronize()
+            assert torch.all(out == float(mb) + 0.5).item(), (mb, out[:4])
+        c.clean_channels()
+
+
+@pytest.mark.gpu
+def test_shm_channel_cuda_tensors():
+    _ensure_built()
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    recv = ctx.Process(target=_channel_proc_cuda, args=(1,))
+    send = ctx.Process(target=_channel_proc_cuda, args=(0,))
+    recv.start(); send.start()
+    recv.join(timeout=180); send.join(timeout=180)
+    assert recv.exitcode == 0 and send.exitcode == 0
