@@ -155,3 +155,25 @@ def test_kv_cache_matches_full_forward():
     cached = torch.cat(outs, dim=1)
     assert torch.allclose(full, cached, atol=1e-4), (full - cached).abs().max()
     destroy()
+
+
+@pytest.mark.gpu
+def test_inference_engine_generates_gpu():
+    """Serving path on MI355X: greedy decode through the KV cache on
+    cuda:0, deterministic across runs."""
+    initialize_model_parallel()
+    from megatronapp_amd.core.inference.static_engine import (
+        get_inference_engine, run_mcore_engine)
+    from megatronapp_amd.training.tokenizer import NullTokenizer
+
+    torch.manual_seed(0)
+    model = _tiny_model().cuda().eval()
+    tok = NullTokenizer(64)
+    engine = get_inference_engine(model, tok, max_batch_size=4)
+    out = run_mcore_engine(engine, ["1 2 3", "4 5"], tokens_to_generate=16,
+                           top_k=1, logprobs=True)
+    out2 = run_mcore_engine(engine, ["1 2 3", "4 5"], tokens_to_generate=16,
+                            top_k=1)
+    assert out["text"] == out2["text"]
+    assert len(out["logprobs"][0]) > 0
+    destroy()
