@@ -42,6 +42,7 @@ class _BiasAddResidualFn(torch.autograd.Function):
 def _bias_dropout_add_func(x_with_bias, residual, prob, training):
     x, bias = x_with_bias
     if (prob == 0.0 and bias is not None and x.is_cuda
+            and x.dtype == torch.bfloat16
             and torch.is_grad_enabled() and _ops.have_ops()
             and hasattr(bias, "main_grad")
             and hasattr(bias, "grad_added_to_main_grad")):

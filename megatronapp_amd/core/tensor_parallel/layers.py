@@ -145,6 +145,7 @@ class VocabParallelEmbedding(torch.nn.Module):
             input_mask = None
             masked_input = input_
         if (input_.is_cuda and _ops.have_ops() and
+                self.weight.dtype == torch.bfloat16 and
                 hasattr(self.weight, "main_grad") and
                 hasattr(self.weight, "grad_added_to_main_grad")):
             output_parallel = _EmbeddingWithMainGradAccum.apply(
@@ -222,7 +223,8 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         grad_output_2d = grad_output.reshape(-1, grad_output.shape[-1])
         total_input_2d = total_input.reshape(-1, total_input.shape[-1])
         if ctx.gradient_accumulation_fusion and hasattr(weight, "main_grad"):
-            if grad_output.is_cuda and _ops.have_ops():
+            if (grad_output.is_cuda and _ops.have_ops()
+                    and grad_output.dtype == torch.bfloat16):
                 # fp32-accumulating hipblasLt wgrad straight into main_grad
                 _ops.get_ops().wgrad_accum(
                     grad_output_2d.contiguous(), total_input_2d.contiguous(),
@@ -245,6 +247,7 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         if use_bias:
             bias_param = ctx.bias_param
             if (grad_output.is_cuda and _ops.have_ops() and
+                    grad_output.dtype == torch.bfloat16 and
                     bias_param is not None and
                     hasattr(bias_param, "main_grad") and
                     hasattr(bias_param, "grad_added_to_main_grad")):
