@@ -30,7 +30,8 @@ def _bias_grad(bias, dx):
     eager sum otherwise."""
     if bias is None or not bias.requires_grad:
         return None
-    if (dx.is_cuda and _ops.have_ops() and hasattr(bias, "main_grad")
+    if (dx.is_cuda and dx.dtype == torch.bfloat16
+            and _ops.have_ops() and hasattr(bias, "main_grad")
             and hasattr(bias, "grad_added_to_main_grad")):
         _ops.get_ops().colsum_accum(
             dx.reshape(-1, dx.shape[-1]).contiguous(), bias.main_grad)
