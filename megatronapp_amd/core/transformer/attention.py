@@ -165,19 +165,30 @@ class SelfAttention(MegatronModule):
                 q_pos_emb = q_pos_emb.index_select(0, pos)
                 k_pos_emb = k_pos_emb.index_select(0, pos)
             if inference_context is not None:
-                offset = inference_context.sequence_len_offset
-                q_pos_emb = q_pos_emb[offset:offset + query.shape[0]]
+                if getattr(inference_context, "is_graph_context", False):
+                    # hipGraph capture: position comes from a device tensor
+                    idx = (inference_context.cur_len +
+                           inference_context._arange[:query.shape[0]])
+                    q_pos_emb = q_pos_emb.index_select(0, idx)
+                    k_pos_emb = k_pos_emb.index_select(0, idx)
+                else:
+                    offset = inference_context.sequence_len_offset
+                    q_pos_emb = q_pos_emb[offset:offset + query.shape[0]]
+                    k_pos_emb = k_pos_emb[:inference_context.sequence_len_offset
+                                          + key.shape[0]][-key.shape[0]:]
             query = apply_rotary_pos_emb(query, q_pos_emb, config=self.config)
-            key = apply_rotary_pos_emb(
-                key, k_pos_emb if inference_context is None else
-                k_pos_emb[:inference_context.sequence_len_offset + key.shape[0]][-key.shape[0]:],
-                config=self.config)
+            key = apply_rotary_pos_emb(key, k_pos_emb, config=self.config)
 
         attn_mask_type = self.attn_mask_type
         if inference_context is not None:
             key, value = inference_context.update_kv_cache(
                 self.layer_number, key, value)
-            if query.shape[0] == 1:
+            if getattr(inference_context, "is_graph_context", False):
+                # fixed-shape window + padding mask (hipGraph decode)
+                attention_mask = inference_context.decode_padding_mask(
+                    query.shape[0], query.shape[1])
+                attn_mask_type = AttnMaskType.padding
+            elif query.shape[0] == 1:
                 attn_mask_type = AttnMaskType.no_mask
 
         with trace_scope("attention"):
