@@ -24,9 +24,11 @@ from .sampling_params import SamplingParams
 
 
 class TextGenerationController:
-    def __init__(self, inference_wrapped_model, tokenizer):
+    def __init__(self, inference_wrapped_model, tokenizer,
+                 use_hip_graphs: bool = True):
         self.model = inference_wrapped_model
         self.tokenizer = tokenizer
+        self.use_hip_graphs = use_hip_graphs
 
     # -------------------------------------------------------------- sampling
     @staticmethod
@@ -73,11 +75,20 @@ class TextGenerationController:
         for i, toks in enumerate(prompts_tokens):
             batch[i, :len(toks)] = torch.tensor(toks, device=device)
 
-        inference_params = InferenceParams(b, total)
-        logprobs = [] if sampling.return_log_probs else None
         pp_group = parallel_state.get_pipeline_model_parallel_group()
         pp_world = parallel_state.get_pipeline_model_parallel_world_size()
         is_last = parallel_state.is_pipeline_last_stage()
+        # hipGraph the single-token decode step (launch-bound) when shapes
+        # allow: one process, CUDA, replayable cache appends
+        use_graphs = (self.use_hip_graphs and device == "cuda"
+                      and pp_world == 1)
+        if use_graphs:
+            from ..hip_graphs import GraphDecodeContext
+            inference_params = GraphDecodeContext(b, total)
+        else:
+            inference_params = InferenceParams(b, total)
+        graph_step = None
+        logprobs = [] if sampling.return_log_probs else None
 
         pos = 0
         step_tokens = batch[:, :max_prompt]
@@ -85,9 +96,16 @@ class TextGenerationController:
             cur_len = step_tokens.shape[1]
             position_ids = torch.arange(
                 pos, pos + cur_len, device=device).unsqueeze(0).expand(b, -1)
-            logits = self.model(step_tokens, position_ids,
-                                inference_context=inference_params)
-            inference_params.increment_sequence_len_offset(cur_len)
+            if use_graphs and cur_len == 1:
+                if graph_step is None:
+                    from ..hip_graphs import GraphedDecodeStep
+                    graph_step = GraphedDecodeStep(self.model,
+                                                   inference_params, b)
+                logits = graph_step(step_tokens, position_ids)
+            else:
+                logits = self.model(step_tokens, position_ids,
+                                    inference_context=inference_params)
+                inference_params.increment_sequence_len_offset(cur_len)
             pos += cur_len
             if pos >= total:
                 break
