@@ -239,7 +239,16 @@ def _add_checkpointing_args(p):
     g.add_argument("--finetune", action="store_true")
     g.add_argument("--ckpt-format", default="torch_dist",
                    choices=["torch_dist", "torch"])
-    g.add_argument("--async-save", action="store_true")
+    g.add_argument("--async-save", action="store_true",
+                   help="write the checkpoint from a background thread")
+    g.add_argument("--config-logger-dir", default=None,
+                   help="dump resolved configs as JSON into this directory")
+    g.add_argument("--ft-heartbeat-dir", default=None,
+                   help="write per-rank heartbeat files for external watchdogs")
+    g.add_argument("--simulate-fault", default=None, choices=[None, "hang", "exit"],
+                   help="arm a simulated fault (failure-detection demos)")
+    g.add_argument("--simulate-fault-rank", type=int, default=0)
+    g.add_argument("--simulate-fault-delay", type=float, default=30.0)
     g.add_argument("--use-checkpoint-args", action="store_true")
 
 

@@ -80,12 +80,36 @@ def _restore_rng_state(state):
     get_cuda_rng_tracker().set_states(state["rng_tracker_states"])
 
 
+_async_save_thread = None
+
+
+def finalize_async_save(blocking: bool = True):
+    """Join any in-flight background checkpoint write."""
+    global _async_save_thread
+    if _async_save_thread is not None and blocking:
+        _async_save_thread.join()
+        _async_save_thread = None
+
+
 def save_checkpoint(iteration, model: List, optimizer, opt_param_scheduler,
                     num_floating_point_operations_so_far=0, checkpointing_context=None,
                     train_data_iterator=None, **kwargs):
     args = get_args()
     if args.save is None:
         return
+    from . import ft_integration
+    ft_integration.on_checkpointing_start()
+    try:
+        _save_checkpoint_impl(iteration, model, optimizer,
+                              opt_param_scheduler,
+                              num_floating_point_operations_so_far)
+    finally:
+        ft_integration.on_checkpointing_end()
+
+
+def _save_checkpoint_impl(iteration, model, optimizer, opt_param_scheduler,
+                          num_floating_point_operations_so_far):
+    args = get_args()
     if getattr(args, "ckpt_format", "torch") == "torch_dist":
         return _save_checkpoint_torch_dist(
             iteration, model, optimizer, opt_param_scheduler,
@@ -123,6 +147,34 @@ def save_checkpoint(iteration, model: List, optimizer, opt_param_scheduler,
                                 f"optim_shard_dp{dp_rank:03d}.pt")
             state_dict = {"optimizer": state_dict.get("optimizer")}
         os.makedirs(os.path.dirname(name), exist_ok=True)
+        if getattr(args, "async_save", False):
+            # deep-copy tensors to CPU inline (cheap vs HBM), write in a
+            # background thread so training resumes immediately
+            import threading
+
+            def _to_cpu(obj):
+                if torch.is_tensor(obj):
+                    return obj.detach().cpu().clone()
+                if isinstance(obj, dict):
+                    return {k: _to_cpu(v) for k, v in obj.items()}
+                if isinstance(obj, (list, tuple)):
+                    return type(obj)(_to_cpu(v) for v in obj)
+                return obj
+
+            snapshot = _to_cpu(state_dict)
+            tracker = get_checkpoint_tracker_filename(args.save)
+
+            def _write():
+                torch.save(snapshot, name)
+                with open(tracker, "w") as f:
+                    f.write(str(iteration))
+
+            global _async_save_thread
+            finalize_async_save(blocking=True)
+            _async_save_thread = threading.Thread(target=_write, daemon=False)
+            _async_save_thread.start()
+            return
+
         torch.save(state_dict, name)
 
     if dist.is_initialized():

@@ -44,6 +44,12 @@ def initialize_megatron(extra_args_provider=None, args_defaults={},
                           continuous_iters=args.continuous_trace_iterations,
                           granularity=args.trace_granularity,
                           max_iters=args.trace_max_iters)
+    if getattr(args, "config_logger_dir", None):
+        from ..core.config_logger import log_config_to_dir
+        log_config_to_dir(args.config_logger_dir, rank=args.rank, args=args)
+    from . import ft_integration
+    ft_integration.setup(args)
+    ft_integration.maybe_setup_simulated_fault(args)
     return args
 
 

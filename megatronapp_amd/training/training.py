@@ -351,10 +351,13 @@ def train(forward_step_func, model, optimizer, opt_param_scheduler,
                 dist.barrier()
             tracer.iteration_begin(iteration)
 
+        from . import ft_integration
+        ft_integration.on_training_step_start()
         with straggler:
             loss_dict, skipped_iter, grad_norm, num_zeros = train_step(
                 forward_step_func, train_data_iterator, model, optimizer,
                 opt_param_scheduler, config, args)
+        ft_integration.on_training_step_end()
         from ..core.rerun_state_machine import get_rerun_state_machine
         if get_rerun_state_machine().should_checkpoint_and_exit():
             print_rank_0("rerun state machine requested checkpoint + exit "
@@ -447,6 +450,8 @@ def pretrain(train_valid_test_dataset_provider, model_provider,
     if args.save:
         save_checkpoint(iteration, model, optimizer, opt_param_scheduler,
                         args.num_floating_point_operations_so_far)
+    from .checkpointing import finalize_async_save
+    finalize_async_save(blocking=True)
 
     if test_it is not None and args.eval_iters > 0:
         evaluate_and_print_results("the end of training", forward_step_func,

@@ -119,3 +119,26 @@ def test_checkpoint_converter(tmp_path):
     assert r.returncode == 0, r.stderr
     blob = torch.load(flat, weights_only=False)
     assert torch.equal(blob["weights"]["model.w"], w)
+
+
+def test_async_save_and_config_logger(tmp_path):
+    """--async-save writes a loadable checkpoint from a background thread;
+    --config-logger-dir dumps resolved args; heartbeat files appear."""
+    import subprocess
+    save = str(tmp_path / "ck")
+    cfgdir = str(tmp_path / "cfg")
+    hbdir = str(tmp_path / "hb")
+    from tests.test_checkpointing import _run
+    out = _run(["--train-iters", "3", "--save", save, "--save-interval", "3",
+                "--ckpt-format", "torch", "--async-save",
+                "--config-logger-dir", cfgdir,
+                "--ft-heartbeat-dir", hbdir], 29641)
+    assert os.path.exists(os.path.join(
+        save, "iter_0000003", "mp_rank_00", "model_optim_rng.pt"))
+    assert os.path.exists(os.path.join(cfgdir, "args.json"))
+    import json
+    hb = json.load(open(os.path.join(hbdir, "heartbeat_rank0.json")))
+    assert hb["section"] in ("idle", "checkpoint")
+    out2 = _run(["--train-iters", "5", "--save", save, "--load", save,
+                 "--ckpt-format", "torch", "--save-interval", "100"], 29642)
+    assert "loaded checkpoint" in out2
