@@ -75,8 +75,10 @@ def load_common(ckpt_dir: str) -> dict:
 
 
 def load(sharded_state_dict: Dict[str, ShardedTensor], ckpt_dir: str,
-         strict: bool = True) -> None:
-    """Fill each local shard in-place from overlapping saved pieces."""
+         strict: bool = True) -> Dict[str, "torch.Tensor"]:
+    """Fill each local shard in-place from overlapping saved pieces and
+    return {key: filled tensor} (reference API: feedable to
+    ``module.load_state_dict``)."""
     with open(os.path.join(ckpt_dir, "index.json")) as f:
         index = json.load(f)
     cache: Dict[str, dict] = {}
@@ -119,3 +121,4 @@ def load(sharded_state_dict: Dict[str, ShardedTensor], ckpt_dir: str,
             raise RuntimeError(
                 f"{key}: only {filled}/{st.data.numel()} elements found "
                 f"in checkpoint (topology mismatch?)")
+    return {k: st.data for k, st in sharded_state_dict.items()}
