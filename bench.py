@@ -67,6 +67,8 @@ def parse_args():
     p.add_argument("--trace", action="store_true", help="enable MegaScan tracing")
     p.add_argument("--trace-dir", default="trace_out")
     p.add_argument("--attention", default="fused", choices=["flash", "fused"])
+    p.add_argument("--seq-length", type=int, default=None,
+                   help="override the model's sequence length")
     p.add_argument("--no-overlap-grad-reduce", action="store_true")
     return p.parse_args()
 
@@ -98,7 +100,7 @@ def main():
     assert dp * pp * tp == n, f"world {n} != dp{dp}*pp{pp}*tp{tp}"
 
     spec = MODELS[args.model]
-    seq = spec["seq_length"]
+    seq = args.seq_length or spec["seq_length"]
     vocab = spec["vocab_size"]
     mbs = args.micro_batch_size
     gbs = args.global_batch_size or 16 * n
