@@ -58,7 +58,9 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="gpt3-1.3b", choices=list(MODELS))
-    p.add_argument("--micro-batch-size", type=int, default=2)
+    p.add_argument("--micro-batch-size", type=int, default=None,
+                   help="default: fattest microbatch the topology allows "
+                        "(16 at PP=1; 4 with a deep-enough pipeline)")
     p.add_argument("--global-batch-size", type=int, default=None,
                    help="default: 16 * n_gpus (weak scaling)")
     p.add_argument("--pp", type=int, default=None,
@@ -102,8 +104,18 @@ def main():
     spec = MODELS[args.model]
     seq = args.seq_length or spec["seq_length"]
     vocab = spec["vocab_size"]
-    mbs = args.micro_batch_size
     gbs = args.global_batch_size or 16 * n
+    mbs = args.micro_batch_size
+    if mbs is None:
+        # fatter microbatches feed fatter GEMMs (+15% measured at PP=1);
+        # with a pipeline keep enough microbatches to bound the bubble
+        spd = gbs // dp  # samples per dp rank
+        if pp == 1:
+            mbs = min(16, spd)
+        elif spd % 4 == 0 and spd // 4 >= pp:
+            mbs = 4
+        else:
+            mbs = 2
     num_microbatches = gbs // (mbs * dp)
     assert num_microbatches * mbs * dp == gbs
 
