@@ -258,9 +258,12 @@ void launch_colsum_accum(const void* dy, float* out, long R, int F,
     return;
   }
   const int gx = (F / 8 + 255) / 256;
-  // atomic traffic per column == grid.y: keep it small (each block then
-  // streams a long sequential row strip, which is what saturates HBM)
-  int target_y = 64;
+  // atomic traffic per column == grid.y.  Keep each block's row strip
+  // >=128 rows (low atomic rate) but scale grid.y with R so large
+  // activations (mbs16: R=32k) still fill the 256 CUs.
+  long target_y = R / 128;
+  if (target_y < 64) target_y = 64;
+  if (target_y > 256) target_y = 256;
   int rpb = (int)((R + target_y - 1) / target_y);
   if (rpb < 8) rpb = 8;
   dim3 grid(gx, (unsigned)((R + rpb - 1) / rpb));

@@ -25,6 +25,10 @@ g = torch.randn(4096, 2048, device="cuda", dtype=torch.bfloat16)
 acc = torch.zeros(2048, device="cuda", dtype=torch.float32)
 us = t(lambda: ops.colsum_accum(g, acc))
 print(f"colsum 4096x2048: {us:.1f} us ({g.numel()*2/us/1e3:.0f} GB/s)")
+g3 = torch.randn(32768, 2048, device="cuda", dtype=torch.bfloat16)
+acc3 = torch.zeros(2048, device="cuda", dtype=torch.float32)
+us = t(lambda: ops.colsum_accum(g3, acc3))
+print(f"colsum 32768x2048: {us:.1f} us ({g3.numel()*2/us/1e3:.0f} GB/s)")
 g2 = torch.randn(4096, 8192, device="cuda", dtype=torch.bfloat16)
 acc2 = torch.zeros(8192, device="cuda", dtype=torch.float32)
 us = t(lambda: ops.colsum_accum(g2, acc2))
