@@ -42,15 +42,20 @@ __global__ void softmax_causal_fwd_reg_kernel(
     }
     m = block_reduce_max<BLOCK>(m, lds);
 
+    // p kept in fp32 regs between the sum and write passes: exp is the
+    // dominant VALU cost and would otherwise run twice per element
+    float pf[NIT][VEC];
     float sum = 0.f;
 #pragma unroll
     for (int it = 0; it < NIT; ++it) {
       const int base = it * BLOCK * VEC + threadIdx.x * VEC;
-      if (base < valid) {
 #pragma unroll
-        for (int j = 0; j < VEC; ++j)
-          if (base + j < valid)
-            sum += __expf(bf2f((unsigned short)c[it][j]) * scale - m);
+      for (int j = 0; j < VEC; ++j) {
+        float p = (base + j < valid)
+                      ? __expf(bf2f((unsigned short)c[it][j]) * scale - m)
+                      : 0.f;
+        pf[it][j] = p;
+        sum += p;
       }
     }
     sum = block_reduce_sum<BLOCK>(sum, lds);
@@ -62,11 +67,7 @@ __global__ void softmax_causal_fwd_reg_kernel(
       short8v o;
 #pragma unroll
       for (int j = 0; j < VEC; ++j)
-        o[j] = (base + j < valid)
-                   ? (short)f2bf(
-                         __expf(bf2f((unsigned short)c[it][j]) * scale - m) *
-                         inv)
-                   : (short)0;
+        o[j] = (short)f2bf(pf[it][j] * inv);
       *(short8v*)(yr + base) = o;
     }
     __syncthreads();

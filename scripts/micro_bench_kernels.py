@@ -11,7 +11,7 @@ def t(fn, iters=30):
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / iters * 1e6
 # softmax shapes: [b*np, sq, sk] = [32, 2048, 2048] bf16
-x = torch.randn(32, 2048, 2048, device="cuda", dtype=torch.bfloat16)
+x = torch.randn(256, 2048, 2048, device="cuda", dtype=torch.bfloat16)
 y = ops.scaled_upper_triang_masked_softmax_fwd(x, 0.08)
 dy = torch.randn_like(x)
 us = t(lambda: ops.scaled_upper_triang_masked_softmax_fwd(x, 0.08))
