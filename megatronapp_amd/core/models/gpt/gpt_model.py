@@ -61,6 +61,14 @@ class GPTModel(LanguageModule):
             pre_process=self.pre_process, post_process=self.post_process,
             vp_stage=vp_stage)
 
+        self.mtp = None
+        if (self.post_process and self.pre_process
+                and getattr(config, "mtp_num_layers", None)):
+            from ...transformer.multi_token_prediction import (
+                MultiTokenPredictionBlock, get_gpt_mtp_block_spec)
+            self.mtp = MultiTokenPredictionBlock(
+                config, get_gpt_mtp_block_spec(config, transformer_layer_spec))
+
         if self.post_process:
             self.output_layer = ColumnParallelLinear(
                 config.hidden_size, vocab_size, config=config,
@@ -115,6 +123,17 @@ class GPTModel(LanguageModule):
         output_weight = None
         if self.share_embeddings_and_output_weights:
             output_weight = self.shared_embedding_or_output_weight()
+
+        if (self.mtp is not None and labels is not None
+                and inference_context is None):
+            hidden_states = self.mtp(
+                input_ids, position_ids, hidden_states,
+                attention_mask=attention_mask, rotary_pos_emb=rotary_pos_emb,
+                embedding=self.embedding, output_layer=self.output_layer,
+                output_weight=output_weight,
+                compute_loss=self.compute_language_model_loss, labels=labels,
+                loss_mask=loss_mask)
+
         logits, _ = self.output_layer(hidden_states, weight=output_weight)
 
         if labels is None:

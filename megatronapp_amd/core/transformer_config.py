@@ -104,6 +104,11 @@ class TransformerConfig(ModelParallelConfig):
     window_size: Optional[tuple] = None
 
     # ---- MoE ----
+    # multi-token prediction (DeepSeek-style; reference
+    # transformer_config.py mtp section)
+    mtp_num_layers: Optional[int] = None
+    mtp_loss_scaling_factor: float = 0.1
+
     num_moe_experts: Optional[int] = None
     moe_router_topk: int = 2
     moe_router_load_balancing_type: str = "aux_loss"  # | "sinkhorn" | "none"

@@ -99,6 +99,9 @@ def _add_model_args(p):
     g.add_argument("--moe-ffn-hidden-size", type=int, default=None)
     g.add_argument("--moe-shared-expert-intermediate-size", type=int,
                    default=None)
+    g.add_argument("--mtp-num-layers", type=int, default=None,
+                   help="multi-token-prediction depths (DeepSeek-style)")
+    g.add_argument("--mtp-loss-scaling-factor", type=float, default=0.1)
 
 
 def _add_training_args(p):
