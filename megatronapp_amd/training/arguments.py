@@ -444,6 +444,8 @@ def core_transformer_config_from_args(args, config_class=None):
         distribute_saved_activations=args.distribute_saved_activations,
         calculate_per_token_loss=args.calculate_per_token_loss,
         num_moe_experts=args.num_experts,
+        mtp_num_layers=args.mtp_num_layers,
+        mtp_loss_scaling_factor=args.mtp_loss_scaling_factor,
         moe_router_topk=args.moe_router_topk,
         moe_router_load_balancing_type=args.moe_router_load_balancing_type,
         moe_aux_loss_coeff=args.moe_aux_loss_coeff,
