@@ -1,0 +1,20 @@
+"""Activation functions selectable via config (reference
+training/activations.py).  No JIT fuser on this stack — the hot-path
+bias+activation fusions are HIP kernels (ops/csrc/elementwise.hip); these
+plain-torch forms serve CPU fallbacks and exotic configs."""
+
+import torch
+import torch.nn.functional as F
+
+
+def squared_relu(x: torch.Tensor) -> torch.Tensor:
+    return torch.pow(F.relu(x), 2)
+
+
+def quick_gelu(x: torch.Tensor) -> torch.Tensor:
+    return x * torch.sigmoid(1.702 * x)
+
+
+def fast_gelu(x: torch.Tensor) -> torch.Tensor:
+    return 0.5 * x * (1.0 + torch.tanh(
+        x * 0.7978845608 * (1.0 + 0.044715 * x * x)))
