@@ -131,7 +131,7 @@ class MLASelfAttention(MegatronModule):
         kv_combined = self.linear_kv_down_proj(hidden_states)
         kv_compressed, k_pos = torch.split(
             kv_combined, [cfg.kv_lora_rank, cfg.qk_pos_emb_head_dim], dim=-1)
-        kv_compressed = self.kv_layernorm(kv_compressed)
+        kv_compressed = self.kv_layernorm(kv_compressed.contiguous())
         kv, _ = self.linear_kv_up_proj(kv_compressed)
         kv = kv.view(sq, b, self.np_, cfg.qk_head_dim + cfg.v_head_dim)
         k_no_pe, value = torch.split(
