@@ -16,7 +16,8 @@ from ... import ops as _ops
 class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
-        if x.is_cuda and x.dtype == torch.bfloat16 and x.is_contiguous():
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            x = x.contiguous()
             y, invrms = _ops.get_ops().rmsnorm_fwd(x, weight, eps)
         else:
             xf = x.float()
@@ -30,7 +31,7 @@ class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, invrms = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16 and dy.is_contiguous():
+        if dy.is_cuda and dy.dtype == torch.bfloat16:
             if hasattr(weight, "main_grad") and \
                     hasattr(weight, "grad_added_to_main_grad"):
                 # dw accumulates straight into the DDP fp32 grad buffer;
@@ -59,7 +60,8 @@ class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
         ctx.bias_param = bias
-        if x.is_cuda and x.dtype == torch.bfloat16 and x.is_contiguous():
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            x = x.contiguous()
             y, mean, invstd = _ops.get_ops().layernorm_fwd(x, weight, bias, eps)
         else:
             xf = x.float()
@@ -75,7 +77,7 @@ class _LayerNormFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, mean, invstd = ctx.saved_tensors
         bias = ctx.bias_param
-        if dy.is_cuda and dy.dtype == torch.bfloat16 and dy.is_contiguous():
+        if dy.is_cuda and dy.dtype == torch.bfloat16:
             if hasattr(weight, "main_grad") and bias is not None and \
                     hasattr(bias, "main_grad") and \
                     hasattr(weight, "grad_added_to_main_grad"):
