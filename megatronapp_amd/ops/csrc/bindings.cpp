@@ -39,6 +39,9 @@ void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
 void launch_adamw_flat_ranged(float*, const float*, float*, float*,
                               const long*, const long*, int, long, float,
                               float, float, float, float, int, hipStream_t);
+void launch_selective_scan_fwd(const void*, const void*, const float*,
+                               const void*, const void*, const float*, float*,
+                               void*, int, int, int, int, hipStream_t);
 void launch_adamw_flat(float*, const float*, float*, float*, long, float,
                        float, float, float, float, int, hipStream_t);
 void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
@@ -375,6 +378,29 @@ void adamw_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                     (float)wd, (int)step, cur_stream());
 }
 
+torch::Tensor selective_scan_fwd(torch::Tensor x, torch::Tensor dt,
+                                 torch::Tensor A, torch::Tensor B,
+                                 torch::Tensor C, torch::Tensor D,
+                                 torch::Tensor h) {
+  check_bf16(x, "x");
+  check_bf16(dt, "dt");
+  check_bf16(B, "B");
+  check_bf16(C, "C");
+  TORCH_CHECK(A.dtype() == torch::kFloat32 && D.dtype() == torch::kFloat32 &&
+              h.dtype() == torch::kFloat32);
+  TORCH_CHECK(x.is_contiguous() && dt.is_contiguous() && B.is_contiguous() &&
+              C.is_contiguous() && h.is_contiguous());
+  const int b = (int)x.size(0), l = (int)x.size(1), d = (int)x.size(2);
+  const int n = (int)A.size(1);
+  auto y = torch::empty_like(x);
+  launch_selective_scan_fwd(x.data_ptr(), dt.data_ptr(),
+                            A.data_ptr<float>(), B.data_ptr(), C.data_ptr(),
+                            D.data_ptr<float>(), h.data_ptr<float>(),
+                            y.data_ptr(), b, l, d, n,
+                            at::cuda::getCurrentCUDAStream());
+  return y;
+}
+
 void adamw_flat_ranged(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        torch::Tensor v, torch::Tensor nw_s, torch::Tensor nw_e,
                        double lr, double beta1, double beta2, double eps,
@@ -415,6 +441,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
   mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
   mod.def("adamw_flat", &adamw_flat);
+  mod.def("selective_scan_fwd", &selective_scan_fwd);
   mod.def("scaled_upper_triang_masked_softmax_bwd",
           &scaled_upper_triang_masked_softmax_bwd);
   mod.def("bias_add_residual", &bias_add_residual);

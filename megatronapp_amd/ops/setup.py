@@ -22,7 +22,7 @@ SOURCES = [
     "elementwise.hip",
     "rope.hip",
     "softmax.hip",
-    "adam.hip",
+    "adam.hip", "scan.hip",
     "wgrad.cpp",
     "attention.hip",
 ]
