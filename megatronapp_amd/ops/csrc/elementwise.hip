@@ -9,14 +9,21 @@
 #define BLOCK 256
 #define VEC 8
 
+// tanh via one fast exp (tanhf is a multi-branch libcall on amdclang):
+// tanh(u) = 1 - 2 / (e^{2u} + 1); |u| clamped so e^{2u} cannot overflow
+__device__ __forceinline__ float fast_tanh(float u) {
+  u = fminf(fmaxf(u, -15.f), 15.f);
+  return 1.f - 2.f / (__expf(2.f * u) + 1.f);
+}
+
 __device__ __forceinline__ float gelu_tanh(float x) {
   float x3 = x * x * x;
-  return 0.5f * x * (1.f + tanhf(0.7978845608028654f * (x + 0.044715f * x3)));
+  return 0.5f * x * (1.f + fast_tanh(0.7978845608028654f * (x + 0.044715f * x3)));
 }
 
 __device__ __forceinline__ float gelu_tanh_grad(float x) {
   float x2 = x * x;
-  float t = tanhf(0.7978845608028654f * (x + 0.044715f * x2 * x));
+  float t = fast_tanh(0.7978845608028654f * (x + 0.044715f * x2 * x));
   return 0.5f * (1.f + t) +
          0.5f * x * (1.f - t * t) * 0.7978845608028654f * (1.f + 3.f * 0.044715f * x2);
 }
