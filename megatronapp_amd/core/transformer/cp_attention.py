@@ -13,9 +13,12 @@ load-balancing split.  Two comm modes, selected by config.cp_comm_type:
   each rank holds the FULL sequence for nh/cp heads; normal causal
   attention after chunk reorder; inverse a2a on the context output.
 
-The ring (p2p KV rotation) mode is the planned third backend; allgather
-is bandwidth-equivalent for cp<=8 on one xGMI-connected node since the
-all-gather saturates the same links the ring would.
+* "ring": flash-style ring attention — K/V rotate around the CP ring
+  (p2p over xGMI) while each rank folds the visiting block into its
+  running (O, logsumexp) online; backward re-rotates K/V, recomputes the
+  block softmax from the saved lse and circulates dK/dV accumulators one
+  full lap so every contribution lands back on its owner.  O(s_local)
+  activation memory — the long-context mode.
 """
 
 from __future__ import annotations
@@ -108,6 +111,8 @@ class ContextParallelAttention(torch.nn.Module):
 
         if mode == "a2a":
             return self._ulysses(query, key, value, group, cp, cp_rank)
+        if mode == "ring":
+            return self._ring(query, key, value, group, cp, cp_rank)
         return self._allgather(query, key, value, group, cp, cp_rank)
 
     # ----------------------------------------------------------- allgather
@@ -125,6 +130,16 @@ class ContextParallelAttention(torch.nn.Module):
         mask = mask.view(1, 1, sq_local, sk)
         return self.core(q, k_full, v_full, attention_mask=mask,
                          attn_mask_type=AttnMaskType.padding)
+
+    # ------------------------------------------------------------------ ring
+    def _ring(self, q, k, v, group, cp, cp_rank):
+        import math
+        sl, b, nh, hd = q.shape
+        assert k.shape[2] == nh, "ring CP requires num_query_groups == heads"
+        scale = 1.0 / math.sqrt(hd)
+        out = _RingAttention.apply(q.contiguous(), k.contiguous(),
+                                   v.contiguous(), group, cp, cp_rank, scale)
+        return out.reshape(sl, b, nh * hd)
 
     # -------------------------------------------------------------- ulysses
     def _ulysses(self, q, k, v, group, cp, cp_rank):
@@ -161,3 +176,93 @@ class ContextParallelAttention(torch.nn.Module):
         ctx = ctx.reshape(s_full, b, nh // cp, hd)
         out = gather_heads(ctx)
         return out.reshape(sq_local, b, nh * hd)
+
+
+# ------------------------------------------------------------------- ring
+def _ring_exchange(t: torch.Tensor, group, cp: int, rank: int):
+    """One hop: send t to (rank+1) % cp, return the tensor received from
+    (rank-1) % cp.  isend/irecv so odd ring sizes cannot deadlock."""
+    ranks = torch.distributed.get_process_group_ranks(group)
+    nxt = ranks[(rank + 1) % cp]
+    prv = ranks[(rank - 1) % cp]
+    recv = torch.empty_like(t)
+    reqs = [torch.distributed.P2POp(torch.distributed.isend, t.contiguous(),
+                                    nxt, group=group),
+            torch.distributed.P2POp(torch.distributed.irecv, recv, prv,
+                                    group=group)]
+    for r in torch.distributed.batch_isend_irecv(reqs):
+        r.wait()
+    return recv
+
+
+def _block_mask(cp, q_rank, kv_rank, sl, device):
+    """True = masked; causal between this rank's global q positions and
+    the visiting rank's global kv positions."""
+    q_pos = _local_global_positions(cp, q_rank, sl, device)
+    kv_pos = _local_global_positions(cp, kv_rank, sl, device)
+    return kv_pos.unsqueeze(0) > q_pos.unsqueeze(1)      # [sl, sl]
+
+
+class _RingAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, group, cp, rank, scale):
+        sl, b, nh, hd = q.shape
+        qf = q.float().permute(1, 2, 0, 3)                # [b, nh, sl, hd]
+        o = torch.zeros_like(qf)
+        lse = torch.full((b, nh, sl), float("-inf"), device=q.device)
+        kv = torch.stack([k.float(), v.float()])          # travels the ring
+        for step in range(cp):
+            src = (rank - step) % cp
+            kf = kv[0].permute(1, 2, 0, 3)                # [b, nh, sl, hd]
+            vf = kv[1].permute(1, 2, 0, 3)
+            scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+            scores = scores.masked_fill(
+                _block_mask(cp, rank, src, sl, q.device), float("-inf"))
+            lse_p = torch.logsumexp(scores, dim=-1)       # [b, nh, sl]
+            p = torch.exp(scores - lse_p.unsqueeze(-1)).nan_to_num(0.0)
+            o_p = torch.matmul(p, vf)
+            new_lse = torch.logaddexp(lse, lse_p)
+            a = torch.exp(lse - new_lse).nan_to_num(0.0).unsqueeze(-1)
+            c = torch.exp(lse_p - new_lse).nan_to_num(0.0).unsqueeze(-1)
+            o = o * a + o_p * c
+            lse = new_lse
+            if step + 1 < cp:
+                kv = _ring_exchange(kv, group, cp, rank)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.meta = (group, cp, rank, scale)
+        return o.permute(2, 0, 1, 3).to(q.dtype)          # [sl, b, nh, hd]
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, o, lse = ctx.saved_tensors
+        group, cp, rank, scale = ctx.meta
+        sl, b, nh, hd = q.shape
+        qf = q.float().permute(1, 2, 0, 3)
+        dof = dout.float().permute(1, 2, 0, 3)            # [b, nh, sl, hd]
+        drow = (dof * o).sum(-1, keepdim=True)            # [b, nh, sl, 1]
+        dq = torch.zeros_like(qf)
+        # kv and its grad accumulators travel one full lap together, so
+        # each rank's contribution rides home to the owner
+        kvd = torch.stack([k.float(), v.float(),
+                           torch.zeros(sl, b, nh, hd, device=q.device),
+                           torch.zeros(sl, b, nh, hd, device=q.device)])
+        for step in range(cp):
+            src = (rank - step) % cp
+            kf = kvd[0].permute(1, 2, 0, 3)
+            vf = kvd[1].permute(1, 2, 0, 3)
+            scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+            scores = scores.masked_fill(
+                _block_mask(cp, rank, src, sl, q.device), float("-inf"))
+            p = torch.exp(scores - lse.unsqueeze(-1)).nan_to_num(0.0)
+            dp = torch.matmul(dof, vf.transpose(-1, -2))
+            ds = p * (dp - drow) * scale                  # [b, nh, sl, sl]
+            dq += torch.matmul(ds, kf)
+            dk_p = torch.matmul(ds.transpose(-1, -2), qf) # [b, nh, sl_k, hd]
+            dv_p = torch.matmul(p.transpose(-1, -2), dof)
+            kvd[2] += dk_p.permute(2, 0, 1, 3)
+            kvd[3] += dv_p.permute(2, 0, 1, 3)
+            kvd = _ring_exchange(kvd, group, cp, rank)
+        # after cp hops the stack is home: kvd[2]/kvd[3] hold the full sums
+        dq = dq.permute(2, 0, 1, 3).to(q.dtype)
+        return (dq, kvd[2].to(k.dtype), kvd[3].to(v.dtype),
+                None, None, None, None)

@@ -84,7 +84,56 @@ def _cp_run(rank, world, cp_comm_type, ref):
     parallel_state.destroy_model_parallel()
 
 
-@pytest.mark.parametrize("mode", ["allgather", "a2a"])
+@pytest.mark.parametrize("mode", ["allgather", "a2a", "ring"])
 def test_cp2_matches_single_rank(mode):
     ref = _reference_logits()
     spawn_ranks(_cp_run, world_size=2, args=(mode, ref))
+
+
+def _ring_grad_run(rank, world, tmpdir):
+    """Ring CP backward: dQ/dK/dV equal the full-attention grads."""
+    import os
+    import math
+    from megatronapp_amd.core import parallel_state
+    from megatronapp_amd.core.transformer.cp_attention import (
+        _RingAttention, _local_global_positions)
+    parallel_state.initialize_model_parallel(context_parallel_size=world)
+    group = parallel_state.get_context_parallel_group()
+    torch.manual_seed(11)
+    sl_full, b, nh, hd = 32, 2, 2, 16
+    qg = torch.randn(sl_full, b, nh, hd)
+    kg = torch.randn(sl_full, b, nh, hd)
+    vg = torch.randn(sl_full, b, nh, hd)
+    dog = torch.randn(sl_full, b, nh, hd)
+
+    # full reference
+    qr = qg.clone().requires_grad_(True)
+    kr = kg.clone().requires_grad_(True)
+    vr = vg.clone().requires_grad_(True)
+    scale = 1.0 / math.sqrt(hd)
+    s = torch.einsum("qbnd,kbnd->bnqk", qr, kr) * scale
+    mask = torch.triu(torch.ones(sl_full, sl_full, dtype=torch.bool), 1)
+    s = s.masked_fill(mask, float("-inf"))
+    o = torch.einsum("bnqk,kbnd->qbnd", torch.softmax(s, -1), vr)
+    o.backward(dog)
+
+    # local chunks for this rank
+    half = sl_full // (2 * world)
+    c0, c1 = rank, 2 * world - 1 - rank
+    def take(t):
+        return torch.cat([t[c0 * half:(c0 + 1) * half],
+                          t[c1 * half:(c1 + 1) * half]], dim=0)
+    ql = take(qg).clone().requires_grad_(True)
+    kl = take(kg).clone().requires_grad_(True)
+    vl = take(vg).clone().requires_grad_(True)
+    out = _RingAttention.apply(ql, kl, vl, group, world, rank, scale)
+    out.backward(take(dog))
+    assert (out - take(o.detach())).abs().max() < 1e-4
+    assert (ql.grad - take(qr.grad)).abs().max() < 1e-4, "dq mismatch"
+    assert (kl.grad - take(kr.grad)).abs().max() < 1e-4, "dk mismatch"
+    assert (vl.grad - take(vr.grad)).abs().max() < 1e-4, "dv mismatch"
+    parallel_state.destroy_model_parallel()
+
+
+def test_ring_attention_grads_match_full(tmp_path):
+    spawn_ranks(_ring_grad_run, world_size=2, args=(str(tmp_path),))
