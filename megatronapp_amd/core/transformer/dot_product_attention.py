@@ -85,6 +85,15 @@ class DotProductAttention(nn.Module):
         scores = scores.view(b, np_, sq, sk)
 
         mask_type = attn_mask_type if attn_mask_type is not None else self.attn_mask_type
+        if packed_seq_params is not None and \
+                packed_seq_params.cu_seqlens_q is not None:
+            # packed (THD) stream: block-diagonal causal mask
+            from ..packed_seq_params import packed_attention_mask
+            from ..enums import AttnMaskType as _AMT
+            attention_mask = packed_attention_mask(
+                packed_seq_params.cu_seqlens_q, sk,
+                causal=(mask_type == _AMT.causal))
+            mask_type = _AMT.padding
         probs = self.scale_mask_softmax(scores, attention_mask, mask_type)
 
         # MegaScope raw-attention-score tap (reference

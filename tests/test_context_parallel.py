@@ -137,3 +137,36 @@ def _ring_grad_run(rank, world, tmpdir):
 
 def test_ring_attention_grads_match_full(tmp_path):
     spawn_ranks(_ring_grad_run, world_size=2, args=(str(tmp_path),))
+
+
+def test_packed_sequences_block_diagonal():
+    """Packed THD attention == running the segments separately."""
+    from tests.utils import initialize_model_parallel, destroy
+    from megatronapp_amd.core.packed_seq_params import PackedSeqParams
+    from megatronapp_amd.core.transformer.dot_product_attention import (
+        DotProductAttention)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    from megatronapp_amd.core.enums import AttnMaskType
+    initialize_model_parallel()
+    torch.manual_seed(2)
+    cfg = TransformerConfig(num_layers=1, hidden_size=32,
+                            num_attention_heads=2, hidden_dropout=0.0,
+                            attention_dropout=0.0,
+                            masked_softmax_fusion=False)
+    attn = DotProductAttention(cfg, 1, AttnMaskType.causal)
+    lens = [12, 8, 4]
+    total = sum(lens)
+    q = torch.randn(total, 1, 2, 16)
+    k = torch.randn(total, 1, 2, 16)
+    v = torch.randn(total, 1, 2, 16)
+    cu = torch.tensor([0, 12, 20, 24], dtype=torch.int32)
+    packed = attn(q, k, v,
+                  packed_seq_params=PackedSeqParams(cu_seqlens_q=cu,
+                                                    cu_seqlens_kv=cu))
+    off = 0
+    for ln in lens:
+        sep = attn(q[off:off + ln], k[off:off + ln], v[off:off + ln],
+                   attn_mask_type=AttnMaskType.causal)
+        assert torch.allclose(packed[off:off + ln], sep, atol=1e-5), off
+        off += ln
+    destroy()
