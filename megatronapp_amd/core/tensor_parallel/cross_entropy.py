@@ -12,6 +12,7 @@ import torch.distributed as dist
 
 from .. import parallel_state
 from .utils import VocabUtility
+from ... import ops as _ops
 
 
 class _VocabParallelCrossEntropy(torch.autograd.Function):
@@ -20,6 +21,40 @@ class _VocabParallelCrossEntropy(torch.autograd.Function):
         tp_group = parallel_state.get_tensor_model_parallel_group()
         tp_world = parallel_state.get_tensor_model_parallel_world_size()
         tp_rank = parallel_state.get_tensor_model_parallel_rank()
+
+        if (vocab_parallel_logits.is_cuda
+                and vocab_parallel_logits.dtype == torch.bfloat16
+                and label_smoothing == 0.0 and _ops.have_ops()
+                and hasattr(_ops.get_ops(), "ce_fwd")
+                and vocab_parallel_logits.size(-1) % 8 == 0):
+            # fused path: the bf16 logits are the only [N, V] tensor ever
+            # read or written — no fp32 softmax materialization
+            ops = _ops.get_ops()
+            Vp = vocab_parallel_logits.size(-1)
+            logits_2d = vocab_parallel_logits.reshape(-1, Vp).contiguous()
+            vocab_start, vocab_end = \
+                VocabUtility.vocab_range_from_per_partition_vocab_size(
+                    Vp, tp_rank, tp_world)
+            tgt = target.reshape(-1).to(torch.int32) - vocab_start
+            tgt = torch.where((target.reshape(-1) >= vocab_start)
+                              & (target.reshape(-1) < vocab_end), tgt,
+                              torch.full_like(tgt, -1))
+            tgt = tgt.contiguous()
+            rowmax = ops.ce_rowmax(logits_2d)
+            if tp_world > 1:
+                dist.all_reduce(rowmax, op=dist.ReduceOp.MAX, group=tp_group)
+            sumexp, predicted = ops.ce_fwd(logits_2d, rowmax, tgt)
+            if tp_world > 1:
+                dist.all_reduce(sumexp, group=tp_group)
+                dist.all_reduce(predicted, group=tp_group)
+            loss = (torch.log(sumexp) - predicted).view_as(
+                target).to(torch.float32)
+            ctx.fused = True
+            ctx.label_smoothing = 0.0
+            ctx.logits_shape = vocab_parallel_logits.shape
+            ctx.save_for_backward(logits_2d, rowmax, sumexp, tgt)
+            return loss
+        ctx.fused = False
 
         logits_max = torch.max(vocab_parallel_logits, dim=-1)[0]
         if tp_world > 1:
@@ -69,6 +104,12 @@ class _VocabParallelCrossEntropy(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_output):
+        if ctx.fused:
+            logits_2d, rowmax, sumexp, tgt = ctx.saved_tensors
+            dlogits = _ops.get_ops().ce_bwd(
+                logits_2d, rowmax, sumexp, tgt,
+                grad_output.reshape(-1).float())
+            return dlogits.view(ctx.logits_shape), None, None
         softmax, target_mask, masked_target_1d = ctx.saved_tensors
         grad_input = softmax
         partition_vocab_size = softmax.size(-1)

@@ -39,6 +39,11 @@ void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
 void launch_adamw_flat_ranged(float*, const float*, float*, float*,
                               const long*, const long*, int, long, float,
                               float, float, float, float, int, hipStream_t);
+void launch_ce_rowmax(const void*, float*, long, int, hipStream_t);
+void launch_ce_fwd(const void*, const float*, const int*, float*, float*,
+                   long, int, hipStream_t);
+void launch_ce_bwd(const void*, const float*, const float*, const int*,
+                   const float*, void*, long, int, hipStream_t);
 void launch_selective_scan_fwd(const void*, const void*, const float*,
                                const void*, const void*, const float*, float*,
                                void*, int, int, int, int, hipStream_t);
@@ -401,6 +406,48 @@ torch::Tensor selective_scan_fwd(torch::Tensor x, torch::Tensor dt,
   return y;
 }
 
+torch::Tensor ce_rowmax(torch::Tensor logits) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(logits.is_contiguous() && logits.dim() == 2);
+  const long rows = logits.size(0);
+  const int v = (int)logits.size(1);
+  TORCH_CHECK(v % 8 == 0);
+  auto out = torch::empty({rows}, logits.options().dtype(torch::kFloat32));
+  launch_ce_rowmax(logits.data_ptr(), out.data_ptr<float>(), rows, v,
+                   cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor rowmax,
+                                  torch::Tensor target) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(target.dtype() == torch::kInt32 && target.is_contiguous());
+  const long rows = logits.size(0);
+  const int v = (int)logits.size(1);
+  auto sumexp = torch::empty({rows},
+                             logits.options().dtype(torch::kFloat32));
+  auto predicted = torch::zeros({rows},
+                                logits.options().dtype(torch::kFloat32));
+  launch_ce_fwd(logits.data_ptr(), rowmax.data_ptr<float>(),
+                target.data_ptr<int>(), sumexp.data_ptr<float>(),
+                predicted.data_ptr<float>(), rows, v, cur_stream());
+  return {sumexp, predicted};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor rowmax,
+                     torch::Tensor sumexp, torch::Tensor target,
+                     torch::Tensor grad_row) {
+  check_bf16(logits, "logits");
+  const long rows = logits.size(0);
+  const int v = (int)logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  launch_ce_bwd(logits.data_ptr(), rowmax.data_ptr<float>(),
+                sumexp.data_ptr<float>(), target.data_ptr<int>(),
+                grad_row.contiguous().data_ptr<float>(), dlogits.data_ptr(),
+                rows, v, cur_stream());
+  return dlogits;
+}
+
 void adamw_flat_ranged(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        torch::Tensor v, torch::Tensor nw_s, torch::Tensor nw_e,
                        double lr, double beta1, double beta2, double eps,
@@ -442,6 +489,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
   mod.def("adamw_flat", &adamw_flat);
   mod.def("selective_scan_fwd", &selective_scan_fwd);
+  mod.def("ce_rowmax", &ce_rowmax);
+  mod.def("ce_fwd", &ce_fwd);
+  mod.def("ce_bwd", &ce_bwd);
   mod.def("scaled_upper_triang_masked_softmax_bwd",
           &scaled_upper_triang_masked_softmax_bwd);
   mod.def("bias_add_residual", &bias_add_residual);

@@ -22,7 +22,7 @@ SOURCES = [
     "elementwise.hip",
     "rope.hip",
     "softmax.hip",
-    "adam.hip", "scan.hip",
+    "adam.hip", "scan.hip", "ce.hip",
     "wgrad.cpp",
     "attention.hip",
 ]
