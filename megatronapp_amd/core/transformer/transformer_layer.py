@@ -78,8 +78,14 @@ class TransformerLayer(MegatronModule):
         with trace_scope("transformer_layer"):
             # --- attention block ---
             with trace_scope("_forward_attention"):
-                residual = hidden_states
-                input_layernorm_output = self.input_layernorm(hidden_states)
+                if hasattr(self.input_layernorm, "forward_with_residual"):
+                    input_layernorm_output, residual = \
+                        self.input_layernorm.forward_with_residual(
+                            hidden_states)
+                else:
+                    residual = hidden_states
+                    input_layernorm_output = self.input_layernorm(
+                        hidden_states)
                 attention_output_with_bias = self.self_attention(
                     input_layernorm_output, attention_mask=attention_mask,
                     inference_context=inference_context,
@@ -106,8 +112,14 @@ class TransformerLayer(MegatronModule):
 
             # --- MLP block ---
             with trace_scope("_forward_mlp"):
-                residual = hidden_states
-                pre_mlp_layernorm_output = self.pre_mlp_layernorm(hidden_states)
+                if hasattr(self.pre_mlp_layernorm, "forward_with_residual"):
+                    pre_mlp_layernorm_output, residual = \
+                        self.pre_mlp_layernorm.forward_with_residual(
+                            hidden_states)
+                else:
+                    residual = hidden_states
+                    pre_mlp_layernorm_output = self.pre_mlp_layernorm(
+                        hidden_states)
                 mlp_output_with_bias = self.mlp(pre_mlp_layernorm_output)
 
                 disturbance = get_disturbance()

@@ -10,12 +10,13 @@
 void launch_rmsnorm_fwd(const void*, const void*, void*, float*, int, int,
                         float, hipStream_t);
 void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
-                        void*, float*, float*, int, int, int, hipStream_t);
+                        void*, float*, float*, const void*, int, int, int,
+                        hipStream_t);
 void launch_layernorm_fwd(const void*, const void*, const void*, void*, float*,
                           float*, int, int, float, hipStream_t);
 void launch_layernorm_bwd(const void*, const void*, const void*, const float*,
                           const float*, void*, float*, float*, float*, float*,
-                          int, int, int, hipStream_t);
+                          const void*, int, int, int, hipStream_t);
 void launch_bias_add_residual(const void*, const void*, const void*, void*,
                               long, int, hipStream_t);
 void launch_bias_gelu_fwd(const void*, const void*, void*, long, int,
@@ -91,7 +92,8 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor invrms,
-                                       c10::optional<torch::Tensor> mg_w) {
+                                       c10::optional<torch::Tensor> mg_w,
+                                       c10::optional<torch::Tensor> dres) {
   check_bf16(dy, "dy");
   check_bf16(x, "x");
   const int H = (int)x.size(-1);
@@ -102,10 +104,17 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                 : torch::zeros({H}, x.options().dtype(torch::kFloat32));
   const int grid = (int)std::min<long>(N, 512);
   auto dw_part = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
+  const void* dres_p = nullptr;
+  torch::Tensor dres_c;
+  if (dres.has_value()) {
+    check_bf16(*dres, "dres");
+    dres_c = dres->contiguous();
+    dres_p = dres_c.data_ptr();
+  }
   launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                      invrms.data_ptr<float>(), dx.data_ptr(),
-                     dw.data_ptr<float>(), dw_part.data_ptr<float>(), grid,
-                     (int)N, H, cur_stream());
+                     dw.data_ptr<float>(), dw_part.data_ptr<float>(), dres_p,
+                     grid, (int)N, H, cur_stream());
   return {dx, dw};
 }
 
@@ -127,7 +136,8 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor w, torch::Tensor mean,
                                          torch::Tensor invstd,
                                          c10::optional<torch::Tensor> mg_w,
-                                         c10::optional<torch::Tensor> mg_b) {
+                                         c10::optional<torch::Tensor> mg_b,
+                                         c10::optional<torch::Tensor> dres) {
   check_bf16(dy, "dy");
   const int H = (int)x.size(-1);
   const long N = x.numel() / H;
@@ -140,11 +150,18 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                 : torch::zeros({H}, x.options().dtype(torch::kFloat32));
   const int grid = (int)std::min<long>(N, 512);
   auto part = torch::empty({2, grid, H}, x.options().dtype(torch::kFloat32));
+  const void* dres_p = nullptr;
+  torch::Tensor dres_c;
+  if (dres.has_value()) {
+    check_bf16(*dres, "dres");
+    dres_c = dres->contiguous();
+    dres_p = dres_c.data_ptr();
+  }
   launch_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        dx.data_ptr(), dw.data_ptr<float>(),
                        db.data_ptr<float>(), part[0].data_ptr<float>(),
-                       part[1].data_ptr<float>(), grid, (int)N, H,
+                       part[1].data_ptr<float>(), dres_p, grid, (int)N, H,
                        cur_stream());
   return {dx, dw, db};
 }
@@ -470,13 +487,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("rmsnorm_fwd", &rmsnorm_fwd);
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, pybind11::arg("dy"), pybind11::arg("x"),
           pybind11::arg("w"), pybind11::arg("invrms"),
-          pybind11::arg("main_grad_w") = pybind11::none());
+          pybind11::arg("main_grad_w") = pybind11::none(),
+          pybind11::arg("dres") = pybind11::none());
   mod.def("layernorm_fwd", &layernorm_fwd);
   mod.def("layernorm_bwd", &layernorm_bwd, pybind11::arg("dy"),
           pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("mean"),
           pybind11::arg("invstd"),
           pybind11::arg("main_grad_w") = pybind11::none(),
-          pybind11::arg("main_grad_b") = pybind11::none());
+          pybind11::arg("main_grad_b") = pybind11::none(),
+          pybind11::arg("dres") = pybind11::none());
   mod.def("bias_gelu_fwd", &bias_gelu_fwd);
   mod.def("bias_gelu_bwd", &bias_gelu_bwd);
   mod.def("bias_swiglu_fwd", &bias_swiglu_fwd);

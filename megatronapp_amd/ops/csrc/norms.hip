@@ -65,7 +65,9 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
                                    const unsigned short* __restrict__ w,
                                    const float* __restrict__ invrms,
                                    unsigned short* __restrict__ dx,
-                                   float* __restrict__ dw_part, int N, int H) {
+                                   float* __restrict__ dw_part,
+                                   const unsigned short* __restrict__ dres,
+                                   int N, int H) {
   __shared__ float lds[BLOCK / WAVE];
   float dwacc[CHUNKS * VEC];
 #pragma unroll
@@ -101,11 +103,17 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
       short8v dv = *(const short8v*)(dyr + base);
       short8v wv = *(const short8v*)(w + base);
       short8v o;
+      // residual-join fusion: dx += the BDA residual's grad in the same
+      // pass (saves a standalone [rows, H] add kernel per layer)
+      short8v rv;
+      if (dres != nullptr)
+        rv = *(const short8v*)(dres + (long)row * H + base);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         float xh = bf2f((unsigned short)xv[j]) * r;
         float dxh = bf2f((unsigned short)dv[j]) * bf2f((unsigned short)wv[j]);
-        o[j] = (short)f2bf(r * (dxh - xh * dot));
+        float add = (dres != nullptr) ? bf2f((unsigned short)rv[j]) : 0.f;
+        o[j] = (short)f2bf(r * (dxh - xh * dot) + add);
       }
       *(short8v*)(dxr + base) = o;
     }
@@ -192,8 +200,9 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
                                      const float* __restrict__ invstd,
                                      unsigned short* __restrict__ dx,
                                      float* __restrict__ dw_part,
-                                     float* __restrict__ db_part, int N,
-                                     int H) {
+                                     float* __restrict__ db_part,
+                                     const unsigned short* __restrict__ dres,
+                                     int N, int H) {
   __shared__ float lds[BLOCK / WAVE];
   float dwacc[CHUNKS * VEC];
   float dbacc[CHUNKS * VEC];
@@ -276,8 +285,8 @@ void launch_rmsnorm_fwd(const void* x, const void* w, void* y, float* invrms,
 
 void launch_rmsnorm_bwd(const void* dy, const void* x, const void* w,
                         const float* invrms, void* dx, float* dw,
-                        float* dw_part, int grid, int N, int H,
-                        hipStream_t stream) {
+                        float* dw_part, const void* dres, int grid, int N,
+                        int H, hipStream_t stream) {
   if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
   const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
 #define RMS_CASE(C)                                                           \
@@ -285,7 +294,8 @@ void launch_rmsnorm_bwd(const void* dy, const void* x, const void* w,
     hipLaunchKernelGGL(rmsnorm_bwd_kernel<C>, dim3(grid), dim3(BLOCK), 0,     \
                        stream, (const unsigned short*)dy,                     \
                        (const unsigned short*)x, (const unsigned short*)w,    \
-                       invrms, (unsigned short*)dx, dw_part, N, H);           \
+                       invrms, (unsigned short*)dx, dw_part,                  \
+                       (const unsigned short*)dres, N, H);                    \
     break;
   switch (chunks) {
     RMS_CASE(1) RMS_CASE(2) RMS_CASE(3) RMS_CASE(4)
@@ -315,7 +325,8 @@ void launch_layernorm_fwd(const void* x, const void* w, const void* b, void* y,
 void launch_layernorm_bwd(const void* dy, const void* x, const void* w,
                           const float* mean, const float* invstd, void* dx,
                           float* dw, float* db, float* dw_part, float* db_part,
-                          int grid, int N, int H, hipStream_t stream) {
+                          const void* dres, int grid, int N, int H,
+                          hipStream_t stream) {
   if (H % VEC != 0) throw std::runtime_error("H must be divisible by 8");
   const int chunks = (H + BLOCK * VEC - 1) / (BLOCK * VEC);
 #define LN_CASE(C)                                                            \
@@ -324,7 +335,7 @@ void launch_layernorm_bwd(const void* dy, const void* x, const void* w,
                        stream, (const unsigned short*)dy,                     \
                        (const unsigned short*)x, (const unsigned short*)w,    \
                        mean, invstd, (unsigned short*)dx, dw_part, db_part,   \
-                       N, H);                                                 \
+                       (const unsigned short*)dres, N, H);                    \
     break;
   switch (chunks) {
     LN_CASE(1) LN_CASE(2) LN_CASE(3) LN_CASE(4)
