@@ -13,6 +13,120 @@
 #define VEC 8
 
 // ------------------------------------------------- causal (upper-triangular)
+// Wave-per-row variants: 4 rows in flight per block, reductions are pure
+// wave shuffles — no LDS, no __syncthreads.  Used when the row fits
+// WAVE*VEC*NIT (sk <= 2048 at NIT 4).
+template <int NIT>
+__global__ void softmax_causal_fwd_wave_kernel(
+    const unsigned short* __restrict__ x, unsigned short* __restrict__ y,
+    long rows, int sq, int sk, float scale) {
+  const int nwaves = BLOCK / WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  for (long row = (long)blockIdx.x * nwaves + wid; row < rows;
+       row += (long)gridDim.x * nwaves) {
+    const int q = (int)(row % sq);
+    const int valid = q + 1 + (sk - sq);
+    const unsigned short* xr = x + row * sk;
+    unsigned short* yr = y + row * sk;
+
+    short8v c[NIT];
+    float m = -INFINITY;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int base = it * WAVE * VEC + lane * VEC;
+      if (base < valid) {
+        c[it] = *(const short8v*)(xr + base);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j)
+          if (base + j < valid)
+            m = fmaxf(m, bf2f((unsigned short)c[it][j]) * scale);
+      }
+    }
+    m = wave_reduce_max(m);
+
+    float pf[NIT][VEC];
+    float sum = 0.f;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int base = it * WAVE * VEC + lane * VEC;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float p = (base + j < valid)
+                      ? __expf(bf2f((unsigned short)c[it][j]) * scale - m)
+                      : 0.f;
+        pf[it][j] = p;
+        sum += p;
+      }
+    }
+    sum = wave_reduce_sum(sum);
+    const float inv = 1.f / sum;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int base = it * WAVE * VEC + lane * VEC;
+      if (base < sk) {
+        short8v o;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j)
+          o[j] = (short)f2bf(pf[it][j] * inv);
+        *(short8v*)(yr + base) = o;
+      }
+    }
+  }
+}
+
+template <int NIT, bool CAUSAL>
+__global__ void softmax_bwd_wave_kernel(const unsigned short* __restrict__ dy,
+                                        const unsigned short* __restrict__ p,
+                                        unsigned short* __restrict__ dx,
+                                        long rows, int sq, int sk,
+                                        float scale) {
+  const int nwaves = BLOCK / WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  for (long row = (long)blockIdx.x * nwaves + wid; row < rows;
+       row += (long)gridDim.x * nwaves) {
+    const int valid = CAUSAL ? (int)(row % sq) + 1 + (sk - sq) : sk;
+    const unsigned short* dyr = dy + row * sk;
+    const unsigned short* pr = p + row * sk;
+    unsigned short* dxr = dx + row * sk;
+
+    short8v cd[NIT], cp[NIT];
+    float dot = 0.f;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int base = it * WAVE * VEC + lane * VEC;
+      if (base < valid) {
+        cd[it] = *(const short8v*)(dyr + base);
+        cp[it] = *(const short8v*)(pr + base);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j)
+          dot += bf2f((unsigned short)cd[it][j]) *
+                 bf2f((unsigned short)cp[it][j]);
+      }
+    }
+    dot = wave_reduce_sum(dot);
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int base = it * WAVE * VEC + lane * VEC;
+      if (base >= sk) continue;
+      short8v o;
+      if (base < valid) {
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float pv = bf2f((unsigned short)cp[it][j]);
+          o[j] = (short)f2bf(
+              pv * (bf2f((unsigned short)cd[it][j]) - dot) * scale);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) o[j] = (short)0;
+      }
+      *(short8v*)(dxr + base) = o;
+    }
+  }
+}
+
 // Register-cached variants: one read of the row instead of three (fwd)
 // or two (bwd), and causal rows only read their valid prefix.  Used when
 // sk is a multiple of BLOCK*VEC and fits NIT<=4 iterations (sk <= 8192).
@@ -254,6 +368,23 @@ static int sm_grid(long rows) {
 void launch_softmax_causal_fwd(const void* x, void* y, long rows, int sq,
                                int sk, float scale, hipStream_t s) {
   if (sk % VEC != 0) throw std::runtime_error("sk must be divisible by 8");
+  const int wspan = WAVE * VEC;
+  if (sk % wspan == 0 && sk / wspan <= 4) {
+#define FWD_WAVE(NIT)                                                         \
+    hipLaunchKernelGGL((softmax_causal_fwd_wave_kernel<NIT>),                 \
+                       dim3(sm_grid(rows / (BLOCK / WAVE) + 1)),              \
+                       dim3(BLOCK), 0, s, (const unsigned short*)x,           \
+                       (unsigned short*)y, rows, sq, sk, scale)
+    switch (sk / wspan) {
+      case 1: FWD_WAVE(1); break;
+      case 2: FWD_WAVE(2); break;
+      case 3: FWD_WAVE(3); break;
+      default: FWD_WAVE(4); break;
+    }
+#undef FWD_WAVE
+    HIP_CHECK_LAUNCH();
+    return;
+  }
   const int span = BLOCK * VEC;
   if (sk % span == 0 && sk / span <= 4) {
 #define FWD_REG(NIT)                                                          \
@@ -291,6 +422,28 @@ void launch_softmax_bwd_impl(const void* dy, const void* p, void* dx,
                              long rows, int sq, int sk, bool causal,
                              float scale, hipStream_t s) {
   if (sk % VEC != 0) throw std::runtime_error("sk must be divisible by 8");
+  const int wspan = WAVE * VEC;
+  if (sk % wspan == 0 && sk / wspan <= 4) {
+#define BWD_WAVE(NIT, C)                                                      \
+    hipLaunchKernelGGL((softmax_bwd_wave_kernel<NIT, C>),                     \
+                       dim3(sm_grid(rows / (BLOCK / WAVE) + 1)),              \
+                       dim3(BLOCK), 0, s, (const unsigned short*)dy,          \
+                       (const unsigned short*)p, (unsigned short*)dx, rows,   \
+                       sq, sk, scale)
+    switch ((sk / wspan) * 2 + (causal ? 1 : 0)) {
+      case 2 * 1 + 1: BWD_WAVE(1, true); break;
+      case 2 * 1 + 0: BWD_WAVE(1, false); break;
+      case 2 * 2 + 1: BWD_WAVE(2, true); break;
+      case 2 * 2 + 0: BWD_WAVE(2, false); break;
+      case 2 * 3 + 1: BWD_WAVE(3, true); break;
+      case 2 * 3 + 0: BWD_WAVE(3, false); break;
+      case 2 * 4 + 1: BWD_WAVE(4, true); break;
+      default: BWD_WAVE(4, false); break;
+    }
+#undef BWD_WAVE
+    HIP_CHECK_LAUNCH();
+    return;
+  }
   const int span = BLOCK * VEC;
   if (sk % span == 0 && sk / span <= 4) {
 #define BWD_REG(NIT, C)                                                       \
