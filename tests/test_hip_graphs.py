@@ -106,3 +106,63 @@ def test_graphed_decode_matches_eager_gpu():
     for a, b in zip(got, ref):
         assert torch.equal(a, b), (a, b)
     destroy()
+
+
+@pytest.mark.gpu
+def test_graphed_decode_rope_model_gpu():
+    """hipGraph decode with rotary embeddings: the rope slice is fetched
+    by device-tensor index_select, so replays see the right positions."""
+    from megatronapp_amd.core.hip_graphs import (GraphDecodeContext,
+                                                 GraphedDecodeStep)
+    from megatronapp_amd.core.inference_params import InferenceParams
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    initialize_model_parallel()
+    model_parallel_cuda_manual_seed(4)
+    torch.manual_seed(4)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4,
+        ffn_hidden_size=128, hidden_dropout=0.0, attention_dropout=0.0,
+        params_dtype=torch.bfloat16, bf16=True, add_bias_linear=False,
+        position_embedding_type="rope", normalization="RMSNorm",
+        gated_linear_unit=True, activation_func="silu")
+    with torch.device("cuda"):
+        m = GPTModel(config=cfg,
+                     transformer_layer_spec=get_gpt_layer_local_spec(
+                         normalization="RMSNorm", use_flash=False),
+                     vocab_size=256, max_sequence_length=64,
+                     position_embedding_type="rope",
+                     pre_process=True, post_process=True).eval()
+    tok = torch.randint(0, 256, (2, 8), device="cuda")
+    pos = torch.arange(8, device="cuda").unsqueeze(0).expand(2, -1)
+    with torch.no_grad():
+        plain = InferenceParams(2, 32)
+        logits = m(tok, pos, None, inference_context=plain)
+        plain.sequence_len_offset = 8
+        cur = logits[:, -1].argmax(-1, keepdim=True)
+        ref = [cur.clone()]
+        for i in range(5):
+            p = torch.full((2, 1), 8 + i, device="cuda", dtype=torch.long)
+            logits = m(cur, p, None, inference_context=plain)
+            plain.sequence_len_offset += 1
+            cur = logits[:, -1].argmax(-1, keepdim=True)
+            ref.append(cur.clone())
+
+        g = GraphDecodeContext(2, 32)
+        logits = m(tok, pos, None, inference_context=g)
+        g.sequence_len_offset = 8
+        cur = logits[:, -1].argmax(-1, keepdim=True)
+        got = [cur.clone()]
+        step = GraphedDecodeStep(m, g, batch_size=2)
+        for i in range(5):
+            p = torch.full((2, 1), 8 + i, device="cuda", dtype=torch.long)
+            logits = step(cur, p)
+            cur = logits[:, -1].argmax(-1, keepdim=True)
+            got.append(cur.clone())
+    for a, b in zip(got, ref):
+        assert torch.equal(a, b), (got, ref)
+    destroy()
