@@ -103,7 +103,9 @@ class DotProductAttention(nn.Module):
         if tt is not None and tt.enabled(FlagType.RawAttentionScore,
                                          self.layer_number):
             tt.report(FlagType.RawAttentionScore, self.layer_number, probs)
-        if self.config.attention_dropout > 0 or self.training:
+        if self.training and self.attention_dropout.p > 0:
+            # fork only when dropout actually randomizes: the RNG state
+            # swap is illegal inside hipGraph capture and pointless in eval
             from ..tensor_parallel.random import get_cuda_rng_tracker
             if torch.cuda.is_available():
                 with get_cuda_rng_tracker().fork():
