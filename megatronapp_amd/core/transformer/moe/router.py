@@ -62,13 +62,16 @@ class TopKRouter(torch.nn.Module):
         self.config = config
         self.num_experts = config.num_moe_experts
         self.topk = config.moe_router_topk
+        # stored in params_dtype so it lives in the DDP flat param buffer
+        # like every other weight; gating MATH stays fp32 below
         self.weight = torch.nn.Parameter(torch.empty(
-            self.num_experts, config.hidden_size, dtype=torch.float32))
+            self.num_experts, config.hidden_size,
+            dtype=config.params_dtype))
         config.init_method(self.weight)
         setattr(self.weight, "sequence_parallel", config.sequence_parallel)
 
     def gating(self, x: torch.Tensor) -> torch.Tensor:
-        return torch.nn.functional.linear(x.float(), self.weight)
+        return torch.nn.functional.linear(x.float(), self.weight.float())
 
     def forward(self, hidden: torch.Tensor):
         """hidden [n_tokens, h] -> (probs [n, topk], indices [n, topk],
