@@ -48,3 +48,12 @@ us = t(lambda: ops.bias_add_residual(xa, ba, ra))
 print(f"bias_add_residual: {us:.1f} us ({xa.numel()*2*3/us/1e3:.0f} GB/s)")
 us = t(lambda: (xa + ba + ra))
 print(f"eager x+b+r:       {us:.1f} us")
+
+# hot fc GEMM shapes: torch.matmul baseline (its own hipBLASLt heuristics)
+for (M, K, N) in ((32768, 2048, 8192), (32768, 8192, 2048),
+                  (32768, 2048, 6144)):
+    a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    us = t(lambda: torch.matmul(a, w.t()), iters=20)
+    fl = 2 * M * K * N / us / 1e6
+    print(f"torch.mm {M}x{K}x{N}: {us:.1f} us ({fl:.0f} GFLOP/s)")
