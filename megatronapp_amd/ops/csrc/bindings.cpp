@@ -51,6 +51,8 @@ void launch_selective_scan_fwd(const void*, const void*, const float*,
 void launch_adamw_flat(float*, const float*, float*, float*, long, float,
                        float, float, float, float, int, hipStream_t);
 void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
+torch::Tensor gemm_nt(torch::Tensor, torch::Tensor);
+torch::Tensor gemm_nn(torch::Tensor, torch::Tensor);
 void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
 void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
@@ -509,6 +511,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_flat", &adamw_flat);
   mod.def("selective_scan_fwd", &selective_scan_fwd);
   mod.def("ce_rowmax", &ce_rowmax);
+  mod.def("gemm_nt", &gemm_nt);
+  mod.def("gemm_nn", &gemm_nn);
   mod.def("ce_fwd", &ce_fwd);
   mod.def("ce_bwd", &ce_bwd);
   mod.def("scaled_upper_triang_masked_softmax_bwd",

@@ -56,4 +56,18 @@ for (M, K, N) in ((32768, 2048, 8192), (32768, 8192, 2048),
     w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
     us = t(lambda: torch.matmul(a, w.t()), iters=20)
     fl = 2 * M * K * N / us / 1e6
-    print(f"torch.mm {M}x{K}x{N}: {us:.1f} us ({fl:.0f} GFLOP/s)")
+    print(f"torch.mm {M}x{K}x{N}: {us:.1f} us ({fl:.0f} TFLOP/s... unit: 1e12/s)")
+    ref = torch.matmul(a.float(), w.t().float())
+    o2 = ops.gemm_nt(a, w)
+    err = (o2.float() - ref).abs().max() / ref.abs().max()
+    us = t(lambda: ops.gemm_nt(a, w), iters=20)
+    fl = 2 * M * K * N / us / 1e6
+    print(f"gemm_nt  {M}x{K}x{N}: {us:.1f} us ({fl:.0f}) relerr={float(err):.4f}")
+    g = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
+    us = t(lambda: torch.matmul(g, w), iters=20)
+    print(f"torch.nn {M}x{N}x{K}: {us:.1f} us")
+    refnn = torch.matmul(g.float(), w.float())
+    o3 = ops.gemm_nn(g, w)
+    errnn = (o3.float() - refnn).abs().max() / refnn.abs().max()
+    us = t(lambda: ops.gemm_nn(g, w), iters=20)
+    print(f"gemm_nn  {M}x{N}x{K}: {us:.1f} us relerr={float(errnn):.4f}")
