@@ -459,3 +459,18 @@ def test_fused_cross_entropy_matches_eager():
     assert rel < 3e-3, rel
     assert (l1.grad.float() - l2.grad.float()).abs().max() < 2e-2
     destroy()
+
+
+@pytest.mark.gpu
+def test_swept_gemms_match_torch():
+    ops = _ops()
+    torch.manual_seed(17)
+    a = torch.randn(256, 512, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(384, 512, device="cuda", dtype=torch.bfloat16)
+    ref = torch.matmul(a.float(), w.t().float())
+    out = ops.gemm_nt(a, w)
+    assert (out.float() - ref).abs().max() / ref.abs().max() < 2e-2
+    g = torch.randn(256, 384, device="cuda", dtype=torch.bfloat16)
+    refnn = torch.matmul(g.float(), w.float())
+    outnn = ops.gemm_nn(g, w)
+    assert (outnn.float() - refnn).abs().max() / refnn.abs().max() < 2e-2
