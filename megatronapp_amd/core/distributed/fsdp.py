@@ -32,7 +32,6 @@ import torch
 import torch.distributed as dist
 
 from .. import parallel_state
-from ..optimizer.clip_grads import clip_grad_by_total_norm_fp32, get_grad_norm_fp32
 from ..optimizer.distrib_optimizer import _adam_step_flat
 
 

@@ -29,8 +29,6 @@ Usage::
 
 from __future__ import annotations
 
-from typing import Dict, Tuple
-
 import torch
 
 from .inference_params import InferenceParams

@@ -26,9 +26,6 @@ from .. import parallel_state
 from ..enums import AttnMaskType
 from ..models.common.embeddings.rotary_pos_embedding import (
     RotaryEmbedding, apply_rotary_pos_emb)
-from ..tensor_parallel.layers import ColumnParallelLinear, RowParallelLinear
-from ..tensor_parallel.mappings import (
-    gather_from_tensor_model_parallel_region)
 from ..tensor_parallel.utils import divide
 from ..transformer_config import MLATransformerConfig
 from .module import MegatronModule
