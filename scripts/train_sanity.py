@@ -50,7 +50,7 @@ def main():
            "--eval-iters", "0", "--hidden-dropout", "0",
            "--attention-dropout", "0"]
     import torch
-    if torch.cuda.is_available():
+    if torch.cuda.is_available() and os.environ.get("SANITY_FP32") != "1":
         cmd.append("--bf16")
     env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29673",
                RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
