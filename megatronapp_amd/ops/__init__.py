@@ -42,6 +42,8 @@ def _try_load():
 
 
 def have_ops() -> bool:
+    if os.environ.get("MEGATRONAPP_FORCE_EAGER") == "1":
+        return False
     return _try_load() is not None
 
 
