@@ -44,7 +44,7 @@ class _BiasGeluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, bias):
         ctx.save_for_backward(x, bias)
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        if _ops.fused_enabled(x, "bias_act"):
             return _ops.get_ops().bias_gelu_fwd(x, bias)
         xf = (x.float() + bias.float()) if bias is not None else x.float()
         return _gelu_tanh(xf).to(x.dtype)
@@ -52,7 +52,7 @@ class _BiasGeluFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, bias = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "bias_act"):
             dx = _ops.get_ops().bias_gelu_bwd(dy.contiguous(), x, bias)
         else:
             xf = (x.float() + bias.float()) if bias is not None else x.float()
@@ -67,7 +67,7 @@ class _BiasSwigluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, bias):
         ctx.save_for_backward(x, bias)
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        if _ops.fused_enabled(x, "bias_act"):
             return _ops.get_ops().bias_swiglu_fwd(x, bias)
         xf = (x.float() + bias.float()) if bias is not None else x.float()
         x1, x2 = xf.chunk(2, dim=-1)
@@ -76,7 +76,7 @@ class _BiasSwigluFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, bias = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "bias_act"):
             dx = _ops.get_ops().bias_swiglu_bwd(dy.contiguous(), x, bias)
         else:
             xf = (x.float() + bias.float()) if bias is not None else x.float()

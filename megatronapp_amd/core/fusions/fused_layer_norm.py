@@ -16,7 +16,7 @@ from ... import ops as _ops
 class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        if _ops.fused_enabled(x, "norms"):
             x = x.contiguous()
             y, invrms = _ops.get_ops().rmsnorm_fwd(x, weight, eps)
         else:
@@ -31,7 +31,7 @@ class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, invrms = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "norms"):
             if hasattr(weight, "main_grad") and \
                     hasattr(weight, "grad_added_to_main_grad"):
                 # dw accumulates straight into the DDP fp32 grad buffer;
@@ -60,7 +60,7 @@ class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
         ctx.bias_param = bias
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        if _ops.fused_enabled(x, "norms"):
             x = x.contiguous()
             y, mean, invstd = _ops.get_ops().layernorm_fwd(x, weight, bias, eps)
         else:
@@ -77,7 +77,7 @@ class _LayerNormFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, mean, invstd = ctx.saved_tensors
         bias = ctx.bias_param
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "norms"):
             if hasattr(weight, "main_grad") and bias is not None and \
                     hasattr(bias, "main_grad") and \
                     hasattr(weight, "grad_added_to_main_grad"):
@@ -116,7 +116,7 @@ class _RMSNormResidualFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, eps):
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        if _ops.fused_enabled(x, "norms"):
             x = x.contiguous()
             y, invrms = _ops.get_ops().rmsnorm_fwd(x, weight, eps)
         else:
@@ -130,7 +130,7 @@ class _RMSNormResidualFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy, dres):
         x, weight, invrms = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "norms"):
             mg = (weight.main_grad
                   if hasattr(weight, "main_grad")
                   and hasattr(weight, "grad_added_to_main_grad") else None)
@@ -154,7 +154,7 @@ class _LayerNormResidualFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
         ctx.bias_param = bias
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        if _ops.fused_enabled(x, "norms"):
             x = x.contiguous()
             y, mean, invstd = _ops.get_ops().layernorm_fwd(x, weight, bias,
                                                            eps)
@@ -171,7 +171,7 @@ class _LayerNormResidualFn(torch.autograd.Function):
     def backward(ctx, dy, dres):
         x, weight, mean, invstd = ctx.saved_tensors
         bias = ctx.bias_param
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "norms"):
             fuse_wgrad = (hasattr(weight, "main_grad") and bias is not None
                           and hasattr(bias, "main_grad")
                           and hasattr(weight, "grad_added_to_main_grad"))

@@ -19,7 +19,7 @@ from ... import ops as _ops
 class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, inputs, scale):
-        if inputs.is_cuda and inputs.dtype == torch.bfloat16:
+        if _ops.fused_enabled(inputs, "softmax"):
             probs = _ops.get_ops().scaled_upper_triang_masked_softmax_fwd(
                 inputs, scale)
         else:
@@ -36,7 +36,7 @@ class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (probs,) = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "softmax"):
             # causal-aware bwd: only the valid row prefix is read
             dx = _ops.get_ops().scaled_upper_triang_masked_softmax_bwd(
                 dy.contiguous(), probs, ctx.scale)
@@ -50,7 +50,7 @@ class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
 class ScaledMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, inputs, mask, scale):
-        if inputs.is_cuda and inputs.dtype == torch.bfloat16:
+        if _ops.fused_enabled(inputs, "softmax"):
             probs = _ops.get_ops().scaled_masked_softmax_fwd(inputs, mask, scale)
         else:
             x = inputs.float() * scale
@@ -64,7 +64,7 @@ class ScaledMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (probs,) = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if _ops.fused_enabled(dy, "softmax"):
             dx = _ops.get_ops().scaled_softmax_bwd(dy.contiguous(), probs, ctx.scale)
         else:
             dyf = dy.float()

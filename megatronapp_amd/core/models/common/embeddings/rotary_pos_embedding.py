@@ -90,7 +90,7 @@ def apply_rotary_pos_emb(t: torch.Tensor, freqs: torch.Tensor,
     cos = torch.cos(freqs)
     sin = torch.sin(freqs)
     if fused is None:
-        fused = (t.is_cuda and t.dtype == torch.bfloat16
+        fused = (_ops.fused_enabled(t, "rope")
                  and (config is None or config.apply_rope_fusion))
     if fused and t.is_cuda:
         out = _FusedRoPEFn.apply(t_rot.contiguous(), cos, sin)

@@ -57,3 +57,16 @@ def get_ops():
             "error: build it with `python -m megatronapp_amd.ops.setup` "
             "or __graft_entry__.build().")
     return mod
+
+
+def fused_enabled(t, group: str) -> bool:
+    """Gate for bf16 fused-kernel autograd paths.  MEGATRONAPP_DISABLE_FUSED
+    can name comma-separated groups (softmax, norms, bias_act, rope, all)
+    to force the torch fallback — used for GPU numerics bisection."""
+    import torch
+    if not (t.is_cuda and t.dtype == torch.bfloat16):
+        return False
+    dis = os.environ.get("MEGATRONAPP_DISABLE_FUSED", "")
+    if dis and ("all" in dis or group in dis):
+        return False
+    return have_ops()

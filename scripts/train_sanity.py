@@ -45,12 +45,14 @@ def main():
            "--max-position-embeddings", "256", "--micro-batch-size", "8",
            "--global-batch-size", "32", "--vocab-size", "512",
            "--data-path", prefix, "--train-iters", str(args.iters),
-           "--lr", "1e-3", "--lr-decay-style", "cosine",
+           "--lr", os.environ.get("SANITY_LR", "1e-3"), "--lr-decay-style", "cosine",
            "--lr-warmup-iters", "20", "--log-interval", "20",
            "--eval-iters", "0", "--hidden-dropout", "0",
            "--attention-dropout", "0"]
     import torch
-    if torch.cuda.is_available() and os.environ.get("SANITY_FP32") != "1":
+    if os.environ.get("SANITY_FORCE_BF16") == "1" or (
+            torch.cuda.is_available()
+            and os.environ.get("SANITY_FP32") != "1"):
         cmd.append("--bf16")
     env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29673",
                RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
