@@ -474,3 +474,59 @@ def test_swept_gemms_match_torch():
     refnn = torch.matmul(g.float(), w.float())
     outnn = ops.gemm_nn(g, w)
     assert (outnn.float() - refnn).abs().max() / refnn.abs().max() < 2e-2
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("rows,H", [(2048, 256), (512, 2048), (333, 512)])
+def test_norm_residual_backward_matches_eager(rows, H):
+    """The two-output norm (residual-grad fused into dx) vs fp32 autograd,
+    including the main_grad-accumulate branch."""
+    from megatronapp_amd.core.fusions.fused_layer_norm import (
+        _LayerNormResidualFn, _RMSNormResidualFn)
+    torch.manual_seed(3)
+    for fn, has_bias in ((_LayerNormResidualFn, True),
+                        (_RMSNormResidualFn, False)):
+        x = torch.randn(rows, H, device="cuda",
+                        dtype=torch.bfloat16).requires_grad_(True)
+        w = torch.nn.Parameter(torch.rand(H, device="cuda",
+                                          dtype=torch.bfloat16) + 0.5)
+        b = torch.nn.Parameter(torch.randn(H, device="cuda",
+                                           dtype=torch.bfloat16))
+        dy = torch.randn(rows, H, device="cuda", dtype=torch.bfloat16)
+        dres = torch.randn(rows, H, device="cuda", dtype=torch.bfloat16)
+
+        # fp32 reference
+        xf = x.detach().float().requires_grad_(True)
+        wf = w.detach().float().requires_grad_(True)
+        bf = b.detach().float().requires_grad_(True)
+        if has_bias:
+            yr = torch.nn.functional.layer_norm(xf, (H,), wf, bf, 1e-5)
+        else:
+            r = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5)
+            yr = xf * r * wf
+        (yr * dy.float()).sum().backward()
+        ref_dx = xf.grad + dres.float()
+
+        # fused two-output path, main_grad branch
+        w.main_grad = torch.zeros(H, device="cuda", dtype=torch.float32)
+        w.grad_added_to_main_grad = False
+        if has_bias:
+            b.main_grad = torch.zeros(H, device="cuda", dtype=torch.float32)
+            b.grad_added_to_main_grad = False
+            y, passthrough = fn.apply(x, w, b, 1e-5)
+        else:
+            y, passthrough = fn.apply(x, w, 1e-5)
+        (y.float() * dy.float()).sum().backward(retain_graph=True)
+        # feed the residual grad through the passthrough output
+        x.grad = None
+        if has_bias:
+            y2, p2 = fn.apply(x, w, b, 1e-5)
+        else:
+            y2, p2 = fn.apply(x, w, 1e-5)
+        ((y2.float() * dy.float()).sum()
+         + (p2.float() * dres.float()).sum()).backward()
+        got_dx = x.grad.float()
+        err = (got_dx - ref_dx).abs().max()
+        assert err < 0.1, (fn.__name__, rows, H, float(err))
+        werr = (w.main_grad - wf.grad).abs().max() / wf.grad.abs().max()
+        assert werr < 0.05, (fn.__name__, float(werr))
