@@ -247,12 +247,16 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
       short8v dv = *(const short8v*)(dyr + base);
       short8v wv = *(const short8v*)(w + base);
       short8v o;
+      short8v rv;
+      if (dres != nullptr)
+        rv = *(const short8v*)(dres + (long)row * H + base);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         float xh = (bf2f((unsigned short)xv[j]) - mu) * r;
         float dxh =
             bf2f((unsigned short)dv[j]) * bf2f((unsigned short)wv[j]);
-        o[j] = (short)f2bf(r * (dxh - sum1 - xh * sum2));
+        float add = (dres != nullptr) ? bf2f((unsigned short)rv[j]) : 0.f;
+        o[j] = (short)f2bf(r * (dxh - sum1 - xh * sum2) + add);
       }
       *(short8v*)(dxr + base) = o;
     }
