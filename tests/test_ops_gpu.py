@@ -519,6 +519,11 @@ def test_norm_residual_backward_matches_eager(rows, H):
         (y.float() * dy.float()).sum().backward(retain_graph=True)
         # feed the residual grad through the passthrough output
         x.grad = None
+        w.main_grad.zero_()              # first backward accumulated once
+        w.grad_added_to_main_grad = False
+        if has_bias:
+            b.main_grad.zero_()
+            b.grad_added_to_main_grad = False
         if has_bias:
             y2, p2 = fn.apply(x, w, b, 1e-5)
         else:
