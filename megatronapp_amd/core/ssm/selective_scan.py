@@ -85,6 +85,17 @@ def selective_scan(x, dt, A, B, C, D, h0=None, return_state=False):
     use_kernel = (x.is_cuda and not torch.is_grad_enabled()
                   and _ops.have_ops()
                   and hasattr(_ops.get_ops(), "selective_scan_fwd"))
+    if torch.is_grad_enabled() and x.requires_grad and h0 is None \
+            and not return_state:
+        # the chunk-parallel scan materializes ~6 [b, l/c, c, d, n] fp32
+        # intermediates; checkpointing recomputes them in backward so the
+        # layer only holds its inputs (without this a 24L/1024h Mamba
+        # OOMs 288 GB at mbs 8)
+        from torch.utils.checkpoint import checkpoint
+        return checkpoint(
+            lambda x_, dt_, B_, C_: selective_scan_chunked(
+                x_, dt_, A, B_, C_, D),
+            x, dt, B, C, use_reentrant=False)
     if use_kernel:
         state = (h0.float().contiguous() if h0 is not None else
                  torch.zeros(x.shape[0], x.shape[2], A.shape[1],
