@@ -8,3 +8,6 @@ from . import tensor_parallel
 from .inference_params import InferenceParams
 from .num_microbatches_calculator import get_num_microbatches
 from .packed_seq_params import PackedSeqParams
+
+# reference-API alias: `from megatron.core import mpu` is parallel_state
+from . import parallel_state as mpu
