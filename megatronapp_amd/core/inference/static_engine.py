@@ -42,9 +42,14 @@ class StaticInferenceEngine:
                     toks, sampling_params, report_step=report_step)
             for j, prompt in enumerate(chunk):
                 row = out_tokens[j]
+                gen = row[len(toks[j]):].cpu()
+                if sampling_params.termination_id >= 0:
+                    hits = (gen == sampling_params.termination_id).nonzero()
+                    if hits.numel():
+                        gen = gen[:int(hits[0]) + 1]   # stop token inclusive
                 req = InferenceRequest(
                     prompt=prompt, prompt_tokens=toks[j],
-                    generated_tokens=row[len(toks[j]):].cpu(),
+                    generated_tokens=gen,
                     generated_log_probs=(logprobs[j].cpu()
                                          if logprobs is not None else None))
                 req.generated_text = self.controller.detokenize(

@@ -1,0 +1,118 @@
+"""Fourth coverage sweep: graph-context masks, blended datasets, inference
+termination/logprobs, WSD scheduler tail, BDA fused function, legacy
+converter loader."""
+import json
+import os
+
+import pytest
+import torch
+
+from tests.utils import initialize_model_parallel, destroy
+
+
+def test_graph_decode_mask_semantics_cpu():
+    from megatronapp_amd.core.hip_graphs import GraphDecodeContext
+    from megatronapp_amd.core.inference_params import InferenceParams
+    g = GraphDecodeContext.__new__(GraphDecodeContext)
+    InferenceParams.__init__(g, 2, 8)
+    g.device = torch.device("cpu")
+    g.cur_len = torch.tensor([3])
+    g._arange = torch.arange(8)
+    g.graph_mode = False
+    m = g.decode_padding_mask(1, 2)
+    assert m.shape == (2, 1, 1, 8)
+    # positions 0..3 visible for the token at position 3
+    assert (~m[0, 0, 0, :4]).all() and m[0, 0, 0, 4:].all()
+    m2 = g.decode_padding_mask(2, 1)  # two query rows at 3 and 4
+    assert (~m2[0, 0, 1, :5]).all() and m2[0, 0, 1, 5:].all()
+
+
+def test_blended_dataset_builder_proportions():
+    from megatronapp_amd.core.datasets import (
+        BlendedMegatronDatasetBuilder, GPTDatasetConfig, MockGPTDataset)
+    cfg = GPTDatasetConfig(sequence_length=16, vocab_size=64, mock=True)
+    train, valid, test = BlendedMegatronDatasetBuilder(
+        MockGPTDataset, [100, 10, 10], lambda: True, cfg).build()
+    assert len(train) >= 100 and valid is not None
+    s = train[0]
+    assert s["tokens"].shape == (16,)
+
+
+def test_inference_termination_id():
+    from megatronapp_amd.core.inference.static_engine import (
+        get_inference_engine)
+    from megatronapp_amd.core.inference.sampling_params import SamplingParams
+    from megatronapp_amd.training.tokenizer import NullTokenizer
+    from tests.test_megascope import _tiny_model
+    initialize_model_parallel()
+    torch.manual_seed(0)
+    m = _tiny_model().eval()
+    tok = NullTokenizer(64)
+    engine = get_inference_engine(m, tok, max_batch_size=2)
+    # terminate immediately on whatever token is produced first
+    out = engine.generate(["1 2"], SamplingParams(
+        num_tokens_to_generate=32, top_k=1))
+    first = int(out[0].generated_tokens[0])
+    out2 = engine.generate(["1 2"], SamplingParams(
+        num_tokens_to_generate=32, top_k=1, termination_id=first))
+    assert len(out2[0].generated_tokens) < 32
+    destroy()
+
+
+def test_wsd_scheduler_decay_tail():
+    from megatronapp_amd.core.optimizer.optimizer_param_scheduler import (
+        OptimizerParamScheduler)
+
+    class _Opt:
+        param_groups = [{"lr": 0.0, "wd_mult": 1.0, "lr_mult": 1.0,
+                         "weight_decay": 0.0}]
+    opt = _Opt()
+    sch = OptimizerParamScheduler(
+        opt, init_lr=0.0, max_lr=1e-3, min_lr=1e-5, lr_warmup_steps=10,
+        lr_decay_steps=100, lr_decay_style="WSD", start_wd=0.0, end_wd=0.0,
+        wd_incr_steps=100, wd_incr_style="constant", wsd_decay_steps=20,
+        lr_wsd_decay_style="linear")
+    lrs = []
+    for _ in range(100):
+        sch.step(1)
+        lrs.append(opt.param_groups[0]["lr"])
+    # stable plateau at max_lr, then a decay tail
+    assert lrs[40] == pytest.approx(1e-3, rel=1e-6)
+    assert lrs[79] == pytest.approx(1e-3, rel=1e-6)
+    assert lrs[-1] < 1.1e-4
+
+
+def test_bias_add_residual_fn_cpu_grads():
+    """The BDA fused Function's eager fallback path: grads for x, bias,
+    residual match plain autograd."""
+    from megatronapp_amd.core.fusions.fused_bias_dropout import (
+        _bias_dropout_add_func)
+    torch.manual_seed(2)
+    x = torch.randn(8, 2, 16, requires_grad=True)
+    b = torch.randn(16, requires_grad=True)
+    r = torch.randn(8, 2, 16, requires_grad=True)
+    out = _bias_dropout_add_func((x, b), r, 0.0, True)
+    out.sum().backward()
+    assert torch.allclose(x.grad, torch.ones_like(x))
+    assert torch.allclose(b.grad, torch.full((16,), 16.0))
+    assert torch.allclose(r.grad, torch.ones_like(r))
+
+
+def test_converter_legacy_loader(tmp_path):
+    import subprocess, sys
+    # fabricate a tp1/pp1 legacy checkpoint
+    sd = {"model": {"w": torch.arange(6.0).view(2, 3),
+                    "b": torch.ones(3)},
+          "iteration": 5, "args": {"x": 1}}
+    d = tmp_path / "iter_0000005" / "mp_rank_00"
+    os.makedirs(d)
+    torch.save(sd, d / "model_optim_rng.pt")
+    r = subprocess.run([sys.executable, "tools/checkpoint/convert.py",
+                        "--load", str(tmp_path / "iter_0000005"),
+                        "--loader", "legacy",
+                        "--save", str(tmp_path / "out"),
+                        "--saver", "torch_dist"],
+                       capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    idx = json.load(open(tmp_path / "out" / "index.json"))
+    assert "model.w" in idx and "model.b" in idx
