@@ -188,6 +188,10 @@ class FullyShardedDataParallel(torch.nn.Module):
         def pre_bwd(module, grad_output):
             unit.begin_backward()
 
+        def pre_fwd_eval_guard(module, args, kwargs=None):
+            # in eval nothing re-gathers after the post-forward release;
+            # handled by pre_fwd on the next call
+
         sub.register_forward_pre_hook(pre_fwd)
         sub.register_forward_hook(post_fwd)
         sub.register_full_backward_pre_hook(pre_bwd)
@@ -202,7 +206,8 @@ class FullyShardedDataParallel(torch.nn.Module):
         # root unit (embeddings, final norm, head) stays live all step
         if self.root_unit is not None:
             self.root_unit.gather()
-            self.root_unit.begin_backward()
+            if self.training and torch.is_grad_enabled():
+                self.root_unit.begin_backward()
         return self.module(*args, **kwargs)
 
     def zero_grad_shards(self):

@@ -42,11 +42,15 @@ class _BiasAddResidualFn(torch.autograd.Function):
 def _bias_dropout_add_func(x_with_bias, residual, prob, training):
     x, bias = x_with_bias
     if (prob == 0.0 and bias is not None and x.is_cuda
-            and x.dtype == torch.bfloat16
-            and torch.is_grad_enabled() and _ops.have_ops()
-            and hasattr(bias, "main_grad")
-            and hasattr(bias, "grad_added_to_main_grad")):
-        return _BiasAddResidualFn.apply(x, bias, residual)
+            and x.dtype == torch.bfloat16 and _ops.have_ops()):
+        if (torch.is_grad_enabled() and hasattr(bias, "main_grad")
+                and hasattr(bias, "grad_added_to_main_grad")):
+            return _BiasAddResidualFn.apply(x, bias, residual)
+        if (not torch.is_grad_enabled() and x.is_contiguous()
+                and residual.is_contiguous() and x.shape[-1] % 8 == 0
+                and hasattr(_ops.get_ops(), "bias_add_residual")):
+            # decode path: one fused kernel instead of two adds
+            return _ops.get_ops().bias_add_residual(x, bias, residual)
     if bias is not None:
         x = x + bias
     out = torch.nn.functional.dropout(x, p=prob, training=training)
