@@ -188,10 +188,6 @@ class FullyShardedDataParallel(torch.nn.Module):
         def pre_bwd(module, grad_output):
             unit.begin_backward()
 
-        def pre_fwd_eval_guard(module, args, kwargs=None):
-            # in eval nothing re-gathers after the post-forward release;
-            # handled by pre_fwd on the next call
-
         sub.register_forward_pre_hook(pre_fwd)
         sub.register_forward_hook(post_fwd)
         sub.register_full_backward_pre_hook(pre_bwd)
