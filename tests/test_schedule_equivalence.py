@@ -41,7 +41,7 @@ def _make_config(pp, vpp, **kw):
     from megatronapp_amd.core.distributed.finalize_model_grads import (
         finalize_model_grads)
     from megatronapp_amd.core.transformer_config import TransformerConfig
-    return TransformerConfig(
+    defaults = dict(
         num_layers=LAYERS, hidden_size=32, num_attention_heads=4,
         ffn_hidden_size=64, pipeline_dtype=torch.float32,
         pipeline_model_parallel_size=pp,
@@ -49,7 +49,9 @@ def _make_config(pp, vpp, **kw):
         hidden_dropout=0.0, attention_dropout=0.0,
         position_embedding_type="rope", normalization="RMSNorm",
         activation_func="swiglu", add_bias_linear=False,
-        finalize_model_grads_func=finalize_model_grads, **kw)
+        finalize_model_grads_func=finalize_model_grads)
+    defaults.update(kw)
+    return TransformerConfig(**defaults)
 
 
 def _fixed_batches():
@@ -245,5 +247,151 @@ def _spawn_pp(rank, world_size, vpp, q, port):
     dist.init_process_group(backend="gloo", rank=rank, world_size=world_size)
     try:
         _pp_run(rank, world_size, vpp, q)
+    finally:
+        dist.destroy_process_group()
+
+
+def _tp_pp_run(rank, world_size, result_q):
+    """TP=2 x PP=2 grid: every shard grad must equal the corresponding
+    slice of the single-process reference grads."""
+    from megatronapp_amd.core import parallel_state
+    from megatronapp_amd.core.distributed import (
+        DistributedDataParallel, DistributedDataParallelConfig)
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.pipeline_parallel import get_forward_backward_func
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    import hashlib as _h
+
+    parallel_state.initialize_model_parallel(
+        tensor_model_parallel_size=2, pipeline_model_parallel_size=2)
+    tp_rank = parallel_state.get_tensor_model_parallel_rank()
+    pp_rank = parallel_state.get_pipeline_model_parallel_rank()
+    model_parallel_cuda_manual_seed(1)
+    # non-gated activation: TP sharding is then a contiguous row/col slice
+    config = _make_config(2, None, tensor_model_parallel_size=2,
+                          activation_func="gelu")
+
+    pre = parallel_state.is_pipeline_first_stage()
+    post = parallel_state.is_pipeline_last_stage()
+    m = GPTModel(config=config,
+                 transformer_layer_spec=get_gpt_layer_local_spec(
+                     normalization="RMSNorm", use_flash=False),
+                 vocab_size=VOCAB, max_sequence_length=SEQ,
+                 position_embedding_type="rope", pre_process=pre,
+                 post_process=post,
+                 share_embeddings_and_output_weights=False)
+    offset = pp_rank * (LAYERS // 2)
+
+    def full_shape_and_slice(p):
+        if getattr(p, "tensor_model_parallel", False):
+            dim = getattr(p, "partition_dim", 0)
+            shape = list(p.shape)
+            shape[dim] *= 2
+            sl = [slice(None)] * len(shape)
+            sl[dim] = slice(tp_rank * p.shape[dim],
+                            (tp_rank + 1) * p.shape[dim])
+            return tuple(shape), tuple(sl)
+        return tuple(p.shape), tuple([slice(None)] * max(p.dim(), 1))
+
+    for name, p in m.named_parameters():
+        canon = _canon_layer_name(name, offset)
+        shape, sl = full_shape_and_slice(p)
+        g = torch.Generator()
+        g.manual_seed(int(_h.md5(canon.encode()).hexdigest()[:8], 16))
+        full = torch.randn(shape, generator=g) * 0.02
+        with torch.no_grad():
+            p.copy_(full[sl])
+
+    ddp = DistributedDataParallel(config, DistributedDataParallelConfig(), m)
+    fb = get_forward_backward_func()
+    fb(forward_step_func=_forward_step_maker(), data_iterator=None,
+       model=ddp, num_microbatches=NUM_MICRO, seq_length=SEQ,
+       micro_batch_size=MBS, forward_only=False)
+
+    out = {}
+    for name, p in ddp.named_parameters():
+        canon = _canon_layer_name(name.replace("module.", ""), offset)
+        _, sl = full_shape_and_slice(p)
+        out[canon] = (p.main_grad.clone(), sl)
+    result_q.put((rank, out))
+    parallel_state.destroy_model_parallel()
+
+
+def test_tp2_pp2_grads_match_single():
+    import torch.multiprocessing as mp
+    from .utils import _free_port
+    import os
+
+    ref = None
+    # reference with the same (gelu) activation
+    global _make_config_orig
+    import functools
+    ref_cfg_kw = dict(activation_func="gelu")
+    from megatronapp_amd.core import parallel_state
+    from megatronapp_amd.core.distributed import (
+        DistributedDataParallel, DistributedDataParallelConfig)
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.pipeline_parallel import get_forward_backward_func
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    from .utils import init_distributed
+
+    init_distributed()
+    parallel_state.initialize_model_parallel()
+    model_parallel_cuda_manual_seed(1)
+    config = _make_config(1, None, **ref_cfg_kw)
+    mref = GPTModel(config=config,
+                    transformer_layer_spec=get_gpt_layer_local_spec(
+                        normalization="RMSNorm", use_flash=False),
+                    vocab_size=VOCAB, max_sequence_length=SEQ,
+                    position_embedding_type="rope",
+                    share_embeddings_and_output_weights=False)
+    _fill_params_deterministic(mref)
+    ddp = DistributedDataParallel(config, DistributedDataParallelConfig(),
+                                  mref)
+    fb = get_forward_backward_func()
+    fb(forward_step_func=_forward_step_maker(), data_iterator=None,
+       model=ddp, num_microbatches=NUM_MICRO, seq_length=SEQ,
+       micro_batch_size=MBS, forward_only=False)
+    ref = {n.replace("module.", ""): p.main_grad.clone()
+           for n, p in ddp.named_parameters()}
+    parallel_state.destroy_model_parallel()
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = []
+    for r in range(4):
+        env = dict(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                   RANK=str(r), WORLD_SIZE="4")
+        p = ctx.Process(target=_grid_entry, args=(r, 4, port, q))
+        p.start()
+        procs.append(p)
+    results = [q.get(timeout=300) for _ in range(4)]
+    for p in procs:
+        p.join(timeout=120)
+    for rank, grads in results:
+        for canon, (g, sl) in grads.items():
+            assert canon in ref, canon
+            expected = ref[canon][sl]
+            err = (g - expected).abs().max()
+            assert err < 2e-4, (rank, canon, float(err))
+
+
+def _grid_entry(rank, world, port, q):
+    import os
+    import torch.distributed as dist
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        _tp_pp_run(rank, world, q)
     finally:
         dist.destroy_process_group()
