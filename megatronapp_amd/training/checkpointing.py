@@ -278,6 +278,34 @@ def _save_checkpoint_torch_dist(iteration, model, optimizer,
         common["opt_param_scheduler"] = opt_param_scheduler.state_dict()
     if not args.no_save_rng:
         common["rng_state"] = _rng_state()
+    if getattr(args, "async_save", False) and not dist.is_initialized():
+        # single-rank async: snapshot shards to CPU inline, write in a
+        # background thread (multi-rank needs the barrier, stays sync)
+        import threading
+        sd = _model_sharded_sd(model)
+        from ..core.dist_checkpointing.mapping import ShardedTensor
+        cpu_sd = {k: ShardedTensor(st.key, st.data.detach().cpu().clone(),
+                                   st.global_shape, st.global_offset,
+                                   st.replica_id)
+                  for k, st in sd.items()}
+        opt_sd = (optimizer.state_dict()
+                  if optimizer is not None and not args.no_save_optim
+                  else None)
+        tracker = get_checkpoint_tracker_filename(args.save)
+
+        def _write():
+            dist_save(cpu_sd, base, common_state=common)
+            if opt_sd is not None:
+                torch.save({"optimizer": opt_sd},
+                           os.path.join(base, "optim_rank00000.pt"))
+            with open(tracker, "w") as f:
+                f.write(str(iteration))
+
+        global _async_save_thread
+        finalize_async_save(blocking=True)
+        _async_save_thread = threading.Thread(target=_write, daemon=False)
+        _async_save_thread.start()
+        return
     dist_save(_model_sharded_sd(model), base, common_state=common)
     if optimizer is not None and not args.no_save_optim:
         rank = dist.get_rank() if dist.is_initialized() else 0

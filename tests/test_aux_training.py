@@ -142,3 +142,15 @@ def test_async_save_and_config_logger(tmp_path):
     out2 = _run(["--train-iters", "5", "--save", save, "--load", save,
                  "--ckpt-format", "torch", "--save-interval", "100"], 29642)
     assert "loaded checkpoint" in out2
+
+
+def test_async_save_torch_dist(tmp_path):
+    from tests.test_checkpointing import _run
+    save = str(tmp_path / "ckd")
+    _run(["--train-iters", "3", "--save", save, "--save-interval", "3",
+          "--ckpt-format", "torch_dist", "--async-save"], 29643)
+    assert os.path.exists(os.path.join(save, "iter_0000003", "index.json"))
+    out = _run(["--train-iters", "5", "--save", save, "--load", save,
+                "--ckpt-format", "torch_dist", "--save-interval", "100"],
+               29644)
+    assert "loaded checkpoint (torch_dist)" in out
