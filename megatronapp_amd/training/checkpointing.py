@@ -278,7 +278,8 @@ def _save_checkpoint_torch_dist(iteration, model, optimizer,
         common["opt_param_scheduler"] = opt_param_scheduler.state_dict()
     if not args.no_save_rng:
         common["rng_state"] = _rng_state()
-    if getattr(args, "async_save", False) and not dist.is_initialized():
+    single = (not dist.is_initialized()) or dist.get_world_size() == 1
+    if getattr(args, "async_save", False) and single:
         # single-rank async: snapshot shards to CPU inline, write in a
         # background thread (multi-rank needs the barrier, stays sync)
         import threading
