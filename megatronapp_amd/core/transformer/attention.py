@@ -192,10 +192,26 @@ class SelfAttention(MegatronModule):
                 attn_mask_type = AttnMaskType.no_mask
 
         with trace_scope("attention"):
-            core_attn_out = self.core_attention(
-                query, key, value, attention_mask=attention_mask,
-                attn_mask_type=attn_mask_type, attention_bias=attention_bias,
-                packed_seq_params=packed_seq_params)
+            if (self.config.recompute_granularity == "selective"
+                    and self.training and torch.is_grad_enabled()
+                    and inference_context is None):
+                # selective recompute (reference attention.py checkpointed
+                # core attention): the [b,np,sq,sk] probs are recomputed in
+                # backward instead of stored
+                from ..tensor_parallel.random import checkpoint as tp_ckpt
+                core_attn_out = tp_ckpt(
+                    lambda q, k, v: self.core_attention(
+                        q, k, v, attention_mask=attention_mask,
+                        attn_mask_type=attn_mask_type,
+                        attention_bias=attention_bias,
+                        packed_seq_params=packed_seq_params),
+                    False, query, key, value)
+            else:
+                core_attn_out = self.core_attention(
+                    query, key, value, attention_mask=attention_mask,
+                    attn_mask_type=attn_mask_type,
+                    attention_bias=attention_bias,
+                    packed_seq_params=packed_seq_params)
 
         if tt is not None and tt.enabled(FlagType.ContextLayer, self.layer_number):
             tt.report(FlagType.ContextLayer, self.layer_number, core_attn_out)

@@ -116,3 +116,38 @@ def test_converter_legacy_loader(tmp_path):
     assert r.returncode == 0, r.stderr
     idx = json.load(open(tmp_path / "out" / "index.json"))
     assert "model.w" in idx and "model.b" in idx
+
+
+def test_selective_recompute_grads_exact():
+    """--recompute-granularity selective: same grads, probs not stored."""
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    initialize_model_parallel()
+
+    def run(selective):
+        model_parallel_cuda_manual_seed(9)
+        torch.manual_seed(9)
+        cfg = TransformerConfig(
+            num_layers=2, hidden_size=64, num_attention_heads=4,
+            ffn_hidden_size=128, hidden_dropout=0.0, attention_dropout=0.0,
+            recompute_granularity="selective" if selective else None)
+        m = GPTModel(config=cfg,
+                     transformer_layer_spec=get_gpt_layer_local_spec(
+                         use_flash=False),
+                     vocab_size=128, max_sequence_length=32,
+                     pre_process=True, post_process=True)
+        tok = torch.randint(0, 128, (2, 32),
+                            generator=torch.Generator().manual_seed(4))
+        pos = torch.arange(32).unsqueeze(0).expand(2, -1)
+        m(tok, pos, None, labels=tok).mean().backward()
+        return {n: p.grad.clone() for n, p in m.named_parameters()}
+
+    base = run(False)
+    sel = run(True)
+    for n in base:
+        assert torch.allclose(base[n], sel[n], atol=1e-6), n
+    destroy()
