@@ -395,3 +395,124 @@ def _grid_entry(rank, world, port, q):
         _tp_pp_run(rank, world, q)
     finally:
         dist.destroy_process_group()
+
+
+def _sp_run(rank, world_size, result_q):
+    """TP=2 with sequence parallelism: shard grads == reference slices."""
+    from megatronapp_amd.core import parallel_state
+    from megatronapp_amd.core.distributed import (
+        DistributedDataParallel, DistributedDataParallelConfig)
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.pipeline_parallel import get_forward_backward_func
+    import hashlib as _h
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+
+    parallel_state.initialize_model_parallel(tensor_model_parallel_size=2)
+    tp_rank = parallel_state.get_tensor_model_parallel_rank()
+    model_parallel_cuda_manual_seed(1)
+    config = _make_config(1, None, tensor_model_parallel_size=2,
+                          activation_func="gelu", sequence_parallel=True)
+    m = GPTModel(config=config,
+                 transformer_layer_spec=get_gpt_layer_local_spec(
+                     normalization="RMSNorm", use_flash=False),
+                 vocab_size=VOCAB, max_sequence_length=SEQ,
+                 position_embedding_type="rope",
+                 share_embeddings_and_output_weights=False)
+
+    def full_shape_and_slice(p):
+        if getattr(p, "tensor_model_parallel", False):
+            dim = getattr(p, "partition_dim", 0)
+            shape = list(p.shape)
+            shape[dim] *= 2
+            sl = [slice(None)] * len(shape)
+            sl[dim] = slice(tp_rank * p.shape[dim],
+                            (tp_rank + 1) * p.shape[dim])
+            return tuple(shape), tuple(sl)
+        return tuple(p.shape), tuple([slice(None)] * max(p.dim(), 1))
+
+    for name, p in m.named_parameters():
+        shape, sl = full_shape_and_slice(p)
+        g = torch.Generator()
+        g.manual_seed(int(_h.md5(name.encode()).hexdigest()[:8], 16))
+        full = torch.randn(shape, generator=g) * 0.02
+        with torch.no_grad():
+            p.copy_(full[sl])
+
+    ddp = DistributedDataParallel(config, DistributedDataParallelConfig(), m)
+    fb = get_forward_backward_func()
+    fb(forward_step_func=_forward_step_maker(), data_iterator=None,
+       model=ddp, num_microbatches=NUM_MICRO, seq_length=SEQ,
+       micro_batch_size=MBS, forward_only=False)
+    out = {}
+    for name, p in ddp.named_parameters():
+        _, sl = full_shape_and_slice(p)
+        out[name.replace("module.", "")] = (p.main_grad.clone(), sl)
+    result_q.put((rank, out))
+    parallel_state.destroy_model_parallel()
+
+
+def test_tp2_sequence_parallel_grads_match_single():
+    import torch.multiprocessing as mp
+    from .utils import _free_port, init_distributed
+    from megatronapp_amd.core import parallel_state
+    from megatronapp_amd.core.distributed import (
+        DistributedDataParallel, DistributedDataParallelConfig)
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.pipeline_parallel import get_forward_backward_func
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    import torch.distributed as dist
+
+    init_distributed()
+    parallel_state.initialize_model_parallel()
+    model_parallel_cuda_manual_seed(1)
+    config = _make_config(1, None, activation_func="gelu")
+    mref = GPTModel(config=config,
+                    transformer_layer_spec=get_gpt_layer_local_spec(
+                        normalization="RMSNorm", use_flash=False),
+                    vocab_size=VOCAB, max_sequence_length=SEQ,
+                    position_embedding_type="rope",
+                    share_embeddings_and_output_weights=False)
+    _fill_params_deterministic(mref)
+    ddp = DistributedDataParallel(config, DistributedDataParallelConfig(),
+                                  mref)
+    fb = get_forward_backward_func()
+    fb(forward_step_func=_forward_step_maker(), data_iterator=None,
+       model=ddp, num_microbatches=NUM_MICRO, seq_length=SEQ,
+       micro_batch_size=MBS, forward_only=False)
+    ref = {n.replace("module.", ""): p.main_grad.clone()
+           for n, p in ddp.named_parameters()}
+    parallel_state.destroy_model_parallel()
+    dist.destroy_process_group()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_sp_entry, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=120)
+    for rank, grads in results:
+        for canon, (g, sl) in grads.items():
+            err = (g - ref[canon][sl]).abs().max()
+            assert err < 2e-4, (rank, canon, float(err))
+
+
+def _sp_entry(rank, world, port, q):
+    import os
+    import torch.distributed as dist
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        _sp_run(rank, world, q)
+    finally:
+        dist.destroy_process_group()
