@@ -151,3 +151,57 @@ def test_selective_recompute_grads_exact():
     for n in base:
         assert torch.allclose(base[n], sel[n], atol=1e-6), n
     destroy()
+
+
+def test_hf_gpt2_export_logit_parity(tmp_path):
+    """Export a random-init GPT through the converter's hf_gpt2 saver and
+    verify transformers.GPT2LMHeadModel produces the SAME logits."""
+    transformers = pytest.importorskip("transformers")
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    initialize_model_parallel()
+    model_parallel_cuda_manual_seed(6)
+    torch.manual_seed(6)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4,
+        ffn_hidden_size=256, hidden_dropout=0.0, attention_dropout=0.0,
+        masked_softmax_fusion=False)
+    m = GPTModel(config=cfg,
+                 transformer_layer_spec=get_gpt_layer_local_spec(
+                     use_flash=False),
+                 vocab_size=96, max_sequence_length=32,
+                 pre_process=True, post_process=True,
+                 share_embeddings_and_output_weights=True).eval()
+
+    full = {"model." + k: v for k, v in m.state_dict().items()
+            if torch.is_tensor(v)}
+    import sys as _s
+    _s.path.insert(0, "tools/checkpoint")
+    from saver_hf_gpt2 import save_hf_gpt2
+    save_hf_gpt2(full, {"args": {"num_attention_heads": 4}},
+                 str(tmp_path / "hf"))
+
+    hf_cfg = transformers.GPT2Config(
+        vocab_size=96, n_positions=32, n_embd=64, n_layer=2, n_head=4,
+        n_inner=256, activation_function="gelu_new", resid_pdrop=0.0,
+        embd_pdrop=0.0, attn_pdrop=0.0)
+    hf = transformers.GPT2LMHeadModel(hf_cfg).eval()
+    sd = torch.load(tmp_path / "hf" / "pytorch_model.bin",
+                    weights_only=False)
+    missing, unexpected = hf.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("attn.bias" in k or "masked_bias" in k for k in missing), \
+        missing
+
+    tok = torch.randint(0, 96, (2, 24))
+    pos = torch.arange(24).unsqueeze(0).expand(2, -1)
+    with torch.no_grad():
+        ours = m(tok, pos, None)
+        theirs = hf(tok).logits
+    err = (ours - theirs).abs().max()
+    assert err < 2e-4, float(err)
+    destroy()

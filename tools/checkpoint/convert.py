@@ -32,6 +32,7 @@ import sys
 import torch
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 
 def load_torch_dist(path):
@@ -122,9 +123,15 @@ def save_consolidated(full, common, path):
     torch.save({"weights": full, "common": common}, path)
 
 
+def save_hf_gpt2(full, common, path):
+    from saver_hf_gpt2 import save_hf_gpt2 as impl
+    impl(full, common, path)
+
+
 LOADERS = {"torch_dist": load_torch_dist, "legacy": load_legacy,
            "consolidated": load_consolidated}
-SAVERS = {"torch_dist": save_torch_dist, "consolidated": save_consolidated}
+SAVERS = {"torch_dist": save_torch_dist, "consolidated": save_consolidated,
+          "hf_gpt2": save_hf_gpt2}
 
 
 def main():
