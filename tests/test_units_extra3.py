@@ -204,4 +204,10 @@ def test_hf_gpt2_export_logit_parity(tmp_path):
         theirs = hf(tok).logits
     err = (ours - theirs).abs().max()
     assert err < 2e-4, float(err)
+
+    # round trip: import the HF export back into OUR naming and compare
+    from saver_hf_gpt2 import load_hf_gpt2
+    full2, common2 = load_hf_gpt2(str(tmp_path / "hf"))
+    for k, v in full.items():
+        assert torch.allclose(full2[k].float(), v.float(), atol=1e-6), k
     destroy()

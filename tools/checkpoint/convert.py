@@ -128,8 +128,13 @@ def save_hf_gpt2(full, common, path):
     impl(full, common, path)
 
 
+def load_hf_gpt2(path):
+    from saver_hf_gpt2 import load_hf_gpt2 as impl
+    return impl(path)
+
+
 LOADERS = {"torch_dist": load_torch_dist, "legacy": load_legacy,
-           "consolidated": load_consolidated}
+           "consolidated": load_consolidated, "hf_gpt2": load_hf_gpt2}
 SAVERS = {"torch_dist": save_torch_dist, "consolidated": save_consolidated,
           "hf_gpt2": save_hf_gpt2}
 

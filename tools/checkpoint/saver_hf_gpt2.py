@@ -95,3 +95,61 @@ def save_hf_gpt2(full: dict, common: dict, path: str) -> None:
         json.dump(cfg, f, indent=2)
     print(f"wrote HF GPT-2 export: {n_layer} layers, {nh} heads, "
           f"vocab {vocab} -> {path}")
+
+
+def _qkv_from_hf(w, nh, hn):
+    """HF c_attn [in, 3h] -> our fused rows [(3hn)*nh, in] grouped/head."""
+    w = w.t().contiguous()                         # [3h, in]
+    h3 = w.shape[0]
+    q, k, v = w.split(h3 // 3, dim=0)
+    q = q.view(nh, hn, -1)
+    k = k.view(nh, hn, -1)
+    v = v.view(nh, hn, -1)
+    return torch.stack([q, k, v], dim=1).reshape(3 * nh * hn, -1).contiguous()
+
+
+def _qkv_bias_from_hf(b, nh, hn):
+    q, k, v = b.split(b.shape[0] // 3, dim=0)
+    q = q.view(nh, hn)
+    k = k.view(nh, hn)
+    v = v.view(nh, hn)
+    return torch.stack([q, k, v], dim=1).reshape(-1).contiguous()
+
+
+def load_hf_gpt2(path):
+    """transformers GPT-2 directory -> full tensors in this framework's
+    naming (inverse of save_hf_gpt2)."""
+    cfg = json.load(open(os.path.join(path, "config.json")))
+    nh = cfg["n_head"]
+    hn = cfg["n_embd"] // nh
+    bin_path = os.path.join(path, "pytorch_model.bin")
+    if os.path.exists(bin_path):
+        sd = torch.load(bin_path, map_location="cpu", weights_only=False)
+    else:
+        from safetensors.torch import load_file
+        sd = load_file(os.path.join(path, "model.safetensors"))
+    sd = {k.replace("transformer.", ""): v for k, v in sd.items()}
+
+    full = {"model.embedding.word_embeddings.weight": sd["wte.weight"],
+            "model.embedding.position_embeddings.weight": sd["wpe.weight"],
+            "model.decoder.final_layernorm.weight": sd["ln_f.weight"],
+            "model.decoder.final_layernorm.bias": sd["ln_f.bias"]}
+    for i in range(cfg["n_layer"]):
+        s_ = f"h.{i}."
+        d = f"model.decoder.layers.{i}."
+        full[d + "input_layernorm.weight"] = sd[s_ + "ln_1.weight"]
+        full[d + "input_layernorm.bias"] = sd[s_ + "ln_1.bias"]
+        full[d + "self_attention.linear_qkv.weight"] = _qkv_from_hf(
+            sd[s_ + "attn.c_attn.weight"], nh, hn)
+        full[d + "self_attention.linear_qkv.bias"] = _qkv_bias_from_hf(
+            sd[s_ + "attn.c_attn.bias"], nh, hn)
+        full[d + "self_attention.linear_proj.weight"] =             sd[s_ + "attn.c_proj.weight"].t().contiguous()
+        full[d + "self_attention.linear_proj.bias"] =             sd[s_ + "attn.c_proj.bias"]
+        full[d + "pre_mlp_layernorm.weight"] = sd[s_ + "ln_2.weight"]
+        full[d + "pre_mlp_layernorm.bias"] = sd[s_ + "ln_2.bias"]
+        full[d + "mlp.linear_fc1.weight"] =             sd[s_ + "mlp.c_fc.weight"].t().contiguous()
+        full[d + "mlp.linear_fc1.bias"] = sd[s_ + "mlp.c_fc.bias"]
+        full[d + "mlp.linear_fc2.weight"] =             sd[s_ + "mlp.c_proj.weight"].t().contiguous()
+        full[d + "mlp.linear_fc2.bias"] = sd[s_ + "mlp.c_proj.bias"]
+    common = {"hf_config": cfg}
+    return full, common
