@@ -9,9 +9,13 @@ here, collapsed to the formats this framework writes:
             legacy       mp_rank_* layout (TP=PP=1 only; for sharded legacy
                          checkpoints re-save with --ckpt-format torch_dist)
             consolidated a single .pt of {name: tensor}
+            hf_gpt2      a transformers GPT-2 directory (weights remapped
+                         into this framework's naming)
   savers:   torch_dist   single-shard-per-key sharded dir; loadable at ANY
                          TP/PP by the overlap-window loader
             consolidated single .pt of {name: tensor}
+            hf_gpt2      transformers-loadable GPT2LMHeadModel directory
+                         (verified logit-exact against transformers)
 
 Typical uses:
   # make a topology-free checkpoint from a TP=2,PP=2 run:
