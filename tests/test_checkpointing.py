@@ -116,3 +116,41 @@ def test_doc_aware_sample_idx(tmp_path):
         full = ds._doc_aware_window(i)
         assert np.array_equal(s["tokens"].numpy(), full[:-1])
         assert np.array_equal(s["labels"].numpy(), full[1:])
+
+
+def test_torch_dist_save_resume_pp2(tmp_path):
+    """Sharded save + resume across a 2-rank pipeline: each stage writes
+    its shards, the gathered index covers both, and resume continues."""
+    import subprocess
+    import sys
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    save = str(tmp_path / "ck_pp2")
+    base_args = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+        "--master-port", "29667", os.path.join(REPO, "pretrain_gpt.py"),
+        "--num-layers", "4", "--hidden-size", "64",
+        "--num-attention-heads", "4", "--seq-length", "32",
+        "--max-position-embeddings", "32", "--micro-batch-size", "2",
+        "--global-batch-size", "8", "--pipeline-model-parallel-size", "2",
+        "--mock-data", "--lr", "1e-3", "--log-interval", "1",
+        "--vocab-size", "128", "--eval-iters", "0", "--hidden-dropout", "0",
+        "--attention-dropout", "0", "--ckpt-format", "torch_dist",
+        "--save", save]
+    out = subprocess.run(base_args + ["--train-iters", "3",
+                                      "--save-interval", "3"],
+                         capture_output=True, text=True, cwd=REPO,
+                         timeout=420)
+    assert out.returncode == 0, out.stderr[-3000:]
+    idx = os.path.join(save, "iter_0000003", "index.json")
+    assert os.path.exists(idx)
+    import json
+    index = json.load(open(idx))
+    files = {m["file"] for metas in index.values() for m in metas}
+    assert len(files) == 2, files            # both stages wrote shards
+    out2 = subprocess.run(base_args + ["--train-iters", "5", "--load", save,
+                                       "--save-interval", "100"],
+                          capture_output=True, text=True, cwd=REPO,
+                          timeout=420)
+    assert out2.returncode == 0, out2.stderr[-3000:]
+    assert "loaded checkpoint (torch_dist)" in out2.stdout
