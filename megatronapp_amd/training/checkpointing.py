@@ -248,15 +248,43 @@ def load_checkpoint(model: List, optimizer, opt_param_scheduler,
 # cross-topology resharding; optimizer shards + rng per-rank alongside.
 # ---------------------------------------------------------------------------
 
+def _globalize_layer_keys(sd, mod):
+    """Rewrite local decoder layer indices to global layer numbers so PP
+    stages contribute DISJOINT keys to the sharded index (stage 1's local
+    layers.0 must not collide with stage 0's)."""
+    import re
+    block = getattr(mod, "decoder", None) or getattr(mod, "encoder", None)
+    layers = getattr(block, "layers", None) if block is not None else None
+    if layers is None:
+        return sd
+
+    def glob(m):
+        local = int(m.group(1))
+        layer = layers[local]
+        g = getattr(layer, "layer_number", local + 1) - 1
+        return f"layers.{g}."
+
+    out = {}
+    for key, st in sd.items():
+        new_key = re.sub(r"layers\.(\d+)\.", glob, key)
+        st.key = new_key
+        out[new_key] = st
+    return out
+
+
 def _model_sharded_sd(model: List):
     from ..core.dist_checkpointing import module_sharded_state_dict
     sd = {}
-    for i, chunk in enumerate(model):
+    for chunk in model:
         mod = chunk
         while hasattr(mod, "module"):
             mod = mod.module
-        prefix = "model." if len(model) == 1 else f"model{i}."
-        sd.update(module_sharded_state_dict(mod, prefix))
+        # one namespace for every chunk: layer keys are globally unique
+        # after _globalize_layer_keys, and embeddings / final norm / head
+        # exist in exactly one chunk — so VPP checkpoints reshard to any
+        # topology like plain PP ones
+        chunk_sd = module_sharded_state_dict(mod, "model.")
+        sd.update(_globalize_layer_keys(chunk_sd, mod))
     return sd
 
 
@@ -331,9 +359,19 @@ def _load_checkpoint_torch_dist(model, optimizer, opt_param_scheduler, base,
         rank = dist.get_rank() if dist.is_initialized() else 0
         opt_path = os.path.join(base, f"optim_rank{rank:05d}.pt")
         if os.path.exists(opt_path):
-            optimizer.load_state_dict(
-                torch.load(opt_path, map_location="cpu",
-                           weights_only=False)["optimizer"])
+            try:
+                optimizer.load_state_dict(
+                    torch.load(opt_path, map_location="cpu",
+                               weights_only=False)["optimizer"])
+            except (RuntimeError, ValueError, KeyError) as e:
+                # cross-topology resume: model weights reshard through the
+                # overlap-window loader, but optimizer shards are
+                # per-topology — start the optimizer fresh
+                if args.rank == 0:
+                    print(f"  optimizer state does not match this topology "
+                          f"({e}); reinitializing optimizer")
+                if hasattr(optimizer, "reload_model_params"):
+                    optimizer.reload_model_params()
         if opt_param_scheduler is not None and                 "opt_param_scheduler" in common:
             opt_param_scheduler.load_state_dict(common["opt_param_scheduler"])
     if optimizer is not None and hasattr(optimizer, "reload_model_params")             and (args.no_load_optim or args.finetune):

@@ -154,3 +154,14 @@ def test_torch_dist_save_resume_pp2(tmp_path):
                           timeout=420)
     assert out2.returncode == 0, out2.stderr[-3000:]
     assert "loaded checkpoint (torch_dist)" in out2.stdout
+
+    # cross-topology resume: the PP=2 sharded checkpoint loads at PP=1
+    out3 = _run(["--num-layers", "4", "--hidden-size", "64",
+                 "--num-attention-heads", "4", "--seq-length", "32",
+                 "--max-position-embeddings", "32", "--micro-batch-size",
+                 "2", "--global-batch-size", "8", "--vocab-size", "128",
+                 "--hidden-dropout", "0", "--attention-dropout", "0",
+                 "--ckpt-format", "torch_dist", "--train-iters", "5",
+                 "--load", save, "--save-interval", "100",
+                 "--eval-iters", "0"], 29668)
+    assert "loaded checkpoint (torch_dist)" in out3
