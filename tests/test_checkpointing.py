@@ -155,6 +155,24 @@ def test_torch_dist_save_resume_pp2(tmp_path):
     assert out2.returncode == 0, out2.stderr[-3000:]
     assert "loaded checkpoint (torch_dist)" in out2.stdout
 
+    # VPP save -> PP=1 resume (unified namespace + globalized layer keys)
+    save_v = str(tmp_path / "ck_vpp")
+    outv = subprocess.run(
+        [a if a != save else save_v for a in base_args]
+        + ["--train-iters", "2", "--save-interval", "2",
+           "--num-layers-per-virtual-pipeline-stage", "1"],
+        capture_output=True, text=True, cwd=REPO, timeout=420)
+    assert outv.returncode == 0, outv.stderr[-3000:]
+    outv2 = _run(["--num-layers", "4", "--hidden-size", "64",
+                  "--num-attention-heads", "4", "--seq-length", "32",
+                  "--max-position-embeddings", "32", "--micro-batch-size",
+                  "2", "--global-batch-size", "8", "--vocab-size", "128",
+                  "--hidden-dropout", "0", "--attention-dropout", "0",
+                  "--ckpt-format", "torch_dist", "--train-iters", "4",
+                  "--load", save_v, "--save-interval", "100",
+                  "--eval-iters", "0"], 29670)
+    assert "loaded checkpoint (torch_dist)" in outv2
+
     # cross-topology resume: the PP=2 sharded checkpoint loads at PP=1
     out3 = _run(["--num-layers", "4", "--hidden-size", "64",
                  "--num-attention-heads", "4", "--seq-length", "32",
