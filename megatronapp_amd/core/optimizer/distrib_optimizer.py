@@ -66,6 +66,19 @@ class DistributedOptimizer:
 
     def __init__(self, config: OptimizerConfig, model_chunks: List):
         self.config = config
+        self.grad_scaler = None
+        if getattr(config, "fp16", False):
+            from .grad_scaler import ConstantGradScaler, DynamicGradScaler
+            if config.loss_scale is not None:
+                self.grad_scaler = ConstantGradScaler(config.loss_scale)
+            else:
+                self.grad_scaler = DynamicGradScaler(
+                    initial_scale=config.initial_loss_scale,
+                    min_scale=config.min_loss_scale,
+                    growth_interval=config.loss_scale_window
+                    if hasattr(config, "loss_scale_window") else 1000,
+                    hysteresis=config.hysteresis
+                    if hasattr(config, "hysteresis") else 2)
         self.model_chunks = model_chunks
         self.buffers = []
         for chunk in model_chunks:
@@ -133,10 +146,17 @@ class DistributedOptimizer:
 
     def get_loss_scale(self) -> torch.Tensor:
         device = "cuda" if torch.cuda.is_available() else "cpu"
+        if self.grad_scaler is not None:
+            return self.grad_scaler.scale.to(device)
         return torch.ones(1, dtype=torch.float32, device=device)
 
-    def scale_loss(self, loss):
-        return loss
+    def scale_loss(self, loss: torch.Tensor) -> torch.Tensor:
+        """fp16: multiply the loss so small grads survive the fp16 range;
+        wired into config.grad_scale_func by the training setup."""
+        if self.grad_scaler is None:
+            return loss
+        return loss * self.grad_scaler.scale.to(loss.device)
+
 
     def _shard_grad(self, i):
         lo, hi = self.shard_bounds[i]
@@ -190,6 +210,20 @@ class DistributedOptimizer:
         self.step_count += 1
         lr = self.param_groups[0]["lr"]
         wd = self.param_groups[0].get("weight_decay", self.weight_decay)
+
+        if self.grad_scaler is not None:
+            inv = self.grad_scaler.inv_scale
+            found_inf = False
+            for i in range(len(self.buffers)):
+                g = self._shard_grad(i)
+                if not torch.isfinite(g).all():
+                    found_inf = True
+                g.mul_(inv)
+            self.grad_scaler.update(found_inf)
+            if found_inf:
+                for i in range(len(self.buffers)):
+                    self._shard_grad(i).zero_()
+                return False, None, None
 
         grad_norm = None
         if self.config.clip_grad > 0:

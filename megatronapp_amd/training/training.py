@@ -121,6 +121,13 @@ def setup_model_and_optimizer(model_provider_func, model_type,
         use_distributed_optimizer=args.use_distributed_optimizer,
         log_num_zeros_in_grad=args.log_num_zeros_in_grad)
     optimizer = get_megatron_optimizer(opt_config, model)
+    if args.fp16 and hasattr(optimizer, "scale_loss"):
+        # fp16: schedules scale the loss before backward; the optimizer
+        # unscales gradients and skips overflowed steps
+        config = core_transformer_config_from_args(args)
+        for chunk in model:
+            mod = chunk.module if hasattr(chunk, "module") else chunk
+            mod.config.grad_scale_func = optimizer.scale_loss
     opt_param_scheduler = get_optimizer_param_scheduler(optimizer, args)
 
     if args.load is not None:

@@ -102,3 +102,46 @@ def test_bf16_training_learns_on_gpu():
                         "--iters", "120"], capture_output=True, text=True,
                        timeout=400)
     assert "LEARNING SANITY OK" in r.stdout, r.stdout[-1500:] + r.stderr[-800:]
+
+
+@pytest.mark.gpu
+def test_fp16_training_step_gpu():
+    """fp16 with dynamic loss scaling: one step on MI355X (fp16 params,
+    scaled loss, unscaled master update)."""
+    from megatronapp_amd.core.distributed import (
+        DistributedDataParallel, DistributedDataParallelConfig)
+    from megatronapp_amd.core.optimizer import (
+        OptimizerConfig, get_megatron_optimizer)
+    initialize_model_parallel()
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    model_parallel_cuda_manual_seed(8)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=128, num_attention_heads=4,
+        ffn_hidden_size=256, hidden_dropout=0.0, attention_dropout=0.0,
+        fp16=True, params_dtype=torch.float16)
+    with torch.device("cuda"):
+        m = GPTModel(config=cfg,
+                     transformer_layer_spec=get_gpt_layer_local_spec(
+                         use_flash=False),
+                     vocab_size=256, max_sequence_length=64,
+                     pre_process=True, post_process=True)
+    ddp = DistributedDataParallel(
+        cfg, DistributedDataParallelConfig(overlap_grad_reduce=False), m)
+    opt = get_megatron_optimizer(
+        OptimizerConfig(optimizer="adam", lr=1e-4, min_lr=0.0, fp16=True,
+                        weight_decay=0.0, clip_grad=1.0,
+                        initial_loss_scale=2 ** 16), [ddp])
+    tok = torch.randint(0, 256, (2, 64), device="cuda")
+    pos = torch.arange(64, device="cuda").unsqueeze(0).expand(2, -1)
+    ddp.zero_grad_buffer()
+    loss = ddp(input_ids=tok, position_ids=pos, attention_mask=None,
+               labels=tok).float().mean()
+    opt.scale_loss(loss).backward()
+    ok, norm, _ = opt.step()
+    assert ok and norm is not None and torch.isfinite(loss)
+    destroy()

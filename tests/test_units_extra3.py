@@ -211,3 +211,61 @@ def test_hf_gpt2_export_logit_parity(tmp_path):
     for k, v in full.items():
         assert torch.allclose(full2[k].float(), v.float(), atol=1e-6), k
     destroy()
+
+
+def test_dynamic_grad_scaler_dynamics():
+    from megatronapp_amd.core.optimizer.grad_scaler import DynamicGradScaler
+    s = DynamicGradScaler(initial_scale=1024, growth_interval=3,
+                          hysteresis=2, min_scale=1.0)
+    assert float(s.scale) == 1024
+    s.update(True)                      # hysteresis absorbs the first inf
+    assert float(s.scale) == 1024
+    s.update(True)
+    assert float(s.scale) == 512       # backoff after hysteresis spent
+    for _ in range(3):
+        s.update(False)
+    assert float(s.scale) == 1024      # growth after interval clean steps
+
+
+def test_fp16_optimizer_skips_overflowed_step():
+    from megatronapp_amd.core.distributed import (
+        DistributedDataParallel, DistributedDataParallelConfig)
+    from megatronapp_amd.core.optimizer import (
+        OptimizerConfig, get_megatron_optimizer)
+    from tests.test_fsdp import _build
+    initialize_model_parallel()
+    model = _build(3)
+    ddp = DistributedDataParallel(
+        model.config, DistributedDataParallelConfig(
+            overlap_grad_reduce=False), model)
+    opt = get_megatron_optimizer(
+        OptimizerConfig(optimizer="adam", lr=1e-3, min_lr=0.0, fp16=True,
+                        weight_decay=0.0, clip_grad=0.0,
+                        initial_loss_scale=2 ** 8), [ddp])
+    assert opt.grad_scaler is not None
+    assert float(opt.get_loss_scale()) == 2 ** 8
+    loss = torch.tensor(2.0, requires_grad=True)
+    assert float(opt.scale_loss(loss)) == 2.0 * 2 ** 8
+
+    tok = torch.randint(0, 128, (2, 32))
+    pos = torch.arange(32).unsqueeze(0).expand(2, -1)
+    ddp.zero_grad_buffer()
+    ddp(input_ids=tok, position_ids=pos, attention_mask=None,
+        labels=tok).float().mean().backward()
+    before = {n: p.detach().clone() for n, p in ddp.named_parameters()}
+    # poison one grad buffer with inf -> step must skip and params hold
+    opt._shard_grad(0)[0] = float("inf")
+    ok, norm, _ = opt.step()
+    assert not ok
+    for n, p in ddp.named_parameters():
+        assert torch.equal(p.detach(), before[n]), n
+    # clean step succeeds
+    ddp.zero_grad_buffer()
+    ddp(input_ids=tok, position_ids=pos, attention_mask=None,
+        labels=tok).float().mean().backward()
+    ok, norm, _ = opt.step()
+    assert ok
+    changed = any(not torch.equal(p.detach(), before[n])
+                  for n, p in ddp.named_parameters())
+    assert changed
+    destroy()
