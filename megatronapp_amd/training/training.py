@@ -118,6 +118,11 @@ def setup_model_and_optimizer(model_provider_func, model_type,
         weight_decay=args.weight_decay, fp16=args.fp16, bf16=args.bf16,
         adam_beta1=args.adam_beta1, adam_beta2=args.adam_beta2,
         adam_eps=args.adam_eps, clip_grad=args.clip_grad,
+        loss_scale=args.loss_scale,
+        initial_loss_scale=args.initial_loss_scale,
+        min_loss_scale=args.min_loss_scale,
+        loss_scale_window=args.loss_scale_window,
+        hysteresis=args.hysteresis,
         use_distributed_optimizer=args.use_distributed_optimizer,
         log_num_zeros_in_grad=args.log_num_zeros_in_grad)
     optimizer = get_megatron_optimizer(opt_config, model)
