@@ -167,17 +167,13 @@ def _add_learning_rate_args(p):
 def _add_mixed_precision_args(p):
     g = p.add_argument_group("mixed precision")
     g.add_argument("--fp16", action="store_true")
+    g.add_argument("--bf16", action="store_true")
     g.add_argument("--loss-scale", type=float, default=None,
                    help="static fp16 loss scale (default: dynamic)")
     g.add_argument("--initial-loss-scale", type=float, default=2 ** 32)
     g.add_argument("--min-loss-scale", type=float, default=1.0)
     g.add_argument("--loss-scale-window", type=int, default=1000)
     g.add_argument("--hysteresis", type=int, default=2)
-    g.add_argument("--bf16", action="store_true")
-    g.add_argument("--loss-scale", type=float, default=None)
-    g.add_argument("--initial-loss-scale", type=float, default=2 ** 32)
-    g.add_argument("--min-loss-scale", type=float, default=1.0)
-    g.add_argument("--loss-scale-window", type=int, default=1000)
     g.add_argument("--accumulate-allreduce-grads-in-fp32", action="store_true",
                    default=True)
 
