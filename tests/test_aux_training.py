@@ -154,3 +154,12 @@ def test_async_save_torch_dist(tmp_path):
                 "--ckpt-format", "torch_dist", "--save-interval", "100"],
                29644)
     assert "loaded checkpoint (torch_dist)" in out
+
+
+def test_eval_loop_and_exit_interval(tmp_path):
+    from tests.test_checkpointing import _run
+    out = _run(["--train-iters", "6", "--eval-iters", "2",
+                "--eval-interval", "3", "--exit-interval", "4"], 29646)
+    # eval ran at iter 3 and training stopped at the exit interval
+    assert "validation loss" in out or "val loss" in out or "eval" in out.lower()
+    assert "iteration        5" not in out and "iteration        4" in out
