@@ -163,3 +163,26 @@ def test_eval_loop_and_exit_interval(tmp_path):
     # eval ran at iter 3 and training stopped at the exit interval
     assert "validation loss" in out or "val loss" in out or "eval" in out.lower()
     assert "iteration        5" not in out and "iteration        4" in out
+
+
+def test_train_samples_mode(tmp_path):
+    """--train-samples derives train_iters from the global batch."""
+    from tests.test_checkpointing import _run
+    out = _run(["--train-samples", "32", "--train-iters-none"], 29647) \
+        if False else None
+    import subprocess, sys, os
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29648",
+               RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "pretrain_gpt.py"),
+         "--num-layers", "2", "--hidden-size", "64",
+         "--num-attention-heads", "4", "--seq-length", "32",
+         "--max-position-embeddings", "32", "--micro-batch-size", "2",
+         "--global-batch-size", "8", "--vocab-size", "128", "--mock-data",
+         "--train-samples", "32", "--lr", "1e-3", "--log-interval", "1",
+         "--eval-iters", "0", "--hidden-dropout", "0",
+         "--attention-dropout", "0"],
+        capture_output=True, text=True, env=env, cwd=REPO, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "4/4" in r.stdout        # 32 samples / gbs 8 = 4 iters
