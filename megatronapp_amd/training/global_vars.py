@@ -67,10 +67,35 @@ def set_global_variables(args, build_tokenizer=True):
             _GLOBAL_TENSORBOARD_WRITER = SummaryWriter(
                 log_dir=args.tensorboard_dir)
         except ImportError:
-            pass
+            # tensorboard is not installed in this image: fall back to a
+            # JSONL scalar log with the same add_scalar surface so the
+            # metrics are still captured under --tensorboard-dir
+            _GLOBAL_TENSORBOARD_WRITER = _JsonlScalarWriter(
+                args.tensorboard_dir)
     if getattr(args, "exit_signal_handler", False):
         from .dist_signal_handler import DistributedSignalHandler
         _GLOBAL_SIGNAL_HANDLER = DistributedSignalHandler().__enter__()
+
+
+class _JsonlScalarWriter:
+    """SummaryWriter-shaped JSONL fallback (scalars.jsonl per run)."""
+
+    def __init__(self, log_dir):
+        import os
+        os.makedirs(log_dir, exist_ok=True)
+        self._f = open(os.path.join(log_dir, "scalars.jsonl"), "a")
+
+    def add_scalar(self, tag, value, step=None):
+        import json
+        self._f.write(json.dumps(
+            {"tag": tag, "value": float(value), "step": step}) + "\n")
+        self._f.flush()
+
+    def flush(self):
+        self._f.flush()
+
+    def close(self):
+        self._f.close()
 
 
 def unset_global_variables():

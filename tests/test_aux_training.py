@@ -186,3 +186,14 @@ def test_train_samples_mode(tmp_path):
         capture_output=True, text=True, env=env, cwd=REPO, timeout=300)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "4/4" in r.stdout        # 32 samples / gbs 8 = 4 iters
+
+
+def test_tensorboard_jsonl_fallback(tmp_path):
+    from tests.test_checkpointing import _run
+    tb = str(tmp_path / "tb")
+    _run(["--train-iters", "2", "--tensorboard-dir", tb], 29649)
+    import json
+    lines = [json.loads(l) for l in
+             open(os.path.join(tb, "scalars.jsonl"))]
+    tags = {l["tag"] for l in lines}
+    assert "lm loss" in tags and "learning-rate" in tags
