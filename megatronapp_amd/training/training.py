@@ -381,6 +381,11 @@ def train(forward_step_func, model, optimizer, opt_param_scheduler,
         iteration += 1
         args.curr_iteration = iteration
         args.consumed_train_samples += get_current_global_batch_size()
+        if args.empty_unused_memory_level >= 1 and torch.cuda.is_available():
+            # reference training.py empty_unused_memory: release cached
+            # blocks between iterations (level 2: every iteration)
+            if args.empty_unused_memory_level >= 2 or iteration % 50 == 0:
+                torch.cuda.empty_cache()
 
         if tracer is not None:
             tracer.iteration_end()
