@@ -422,6 +422,9 @@ def core_transformer_config_from_args(args, config_class=None):
         untie_embeddings_and_output_weights=args.untie_embeddings_and_output_weights,
         apply_query_key_layer_scaling=args.apply_query_key_layer_scaling,
         masked_softmax_fusion=args.masked_softmax_fusion,
+        bias_activation_fusion=args.bias_gelu_fusion,
+        bias_dropout_fusion=args.bias_dropout_fusion,
+        attention_softmax_in_fp32=args.attention_softmax_in_fp32,
         attention_backend=args.attention_backend,
         init_method_std=args.init_method_std,
         apply_rope_fusion=args.apply_rope_fusion,
