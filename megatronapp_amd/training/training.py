@@ -211,7 +211,9 @@ def training_log(loss_dict, total_loss_dict, learning_rate, iteration,
         f"elapsed (ms): {elapsed_per_iter*1000:.1f} | tokens/s: "
         f"{throughput_tps:.0f} | TFLOP/s/GPU: {tflops:.1f} | lr: "
         f"{learning_rate:.3E} | {loss_str} | grad norm: "
-        f"{grad_norm if grad_norm is not None else 0:.3f}")
+        f"{grad_norm if grad_norm is not None else 0:.3f}"
+        + (f" | params norm: {params_norm:.3f}"
+           if params_norm is not None else ""))
     from .global_vars import get_tensorboard_writer
     writer = get_tensorboard_writer()
     if writer is not None:
@@ -370,6 +372,15 @@ def train(forward_step_func, model, optimizer, opt_param_scheduler,
                 forward_step_func, train_data_iterator, model, optimizer,
                 opt_param_scheduler, config, args)
         ft_integration.on_training_step_end()
+        params_norm = None
+        if args.log_params_norm:
+            sq = sum(float(p.data.float().norm() ** 2)
+                     for chunk in model for p in chunk.parameters())
+            t = torch.tensor([sq])
+            if dist.is_initialized() and \
+                    parallel_state.get_model_parallel_group() is not None:
+                dist.all_reduce(t, group=parallel_state.get_model_parallel_group())
+            params_norm = float(t.sqrt())
         from ..core.rerun_state_machine import get_rerun_state_machine
         if get_rerun_state_machine().should_checkpoint_and_exit():
             print_rank_0("rerun state machine requested checkpoint + exit "
@@ -395,7 +406,8 @@ def train(forward_step_func, model, optimizer, opt_param_scheduler,
         lr = optimizer.param_groups[0]["lr"]
         report_memory_flag = training_log(
             loss_dict, total_loss_dict, lr, iteration, 1.0,
-            report_memory_flag, skipped_iter, grad_norm, None, num_zeros,
+            report_memory_flag, skipped_iter, grad_norm, params_norm,
+            num_zeros,
             elapsed, args)
         if ws_server is not None:
             ws_server.step_finished(iteration, loss_dict)

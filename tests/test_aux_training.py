@@ -197,3 +197,9 @@ def test_tensorboard_jsonl_fallback(tmp_path):
              open(os.path.join(tb, "scalars.jsonl"))]
     tags = {l["tag"] for l in lines}
     assert "lm loss" in tags and "learning-rate" in tags
+
+
+def test_log_params_norm(tmp_path):
+    from tests.test_checkpointing import _run
+    out = _run(["--train-iters", "2", "--log-params-norm"], 29650)
+    assert "params norm:" in out
