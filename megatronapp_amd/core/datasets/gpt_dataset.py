@@ -104,9 +104,11 @@ class GPTDataset(torch.utils.data.Dataset):
     fall back to fixed windows."""
 
     def __init__(self, config: GPTDatasetConfig, token_file: str,
-                 num_samples: Optional[int] = None, name: str = "gpt"):
+                 num_samples: Optional[int] = None, name: str = "gpt",
+                 sample_offset: int = 0):
         self.config = config
         self.name = name
+        self.sample_offset = sample_offset
         self.indexed = None
         if token_file.endswith(".npy"):
             self.tokens = np.load(token_file, mmap_mode="r")
@@ -122,7 +124,8 @@ class GPTDataset(torch.utils.data.Dataset):
             self._build_doc_aware_indices(num_samples)
         else:
             max_samples = (len(self.tokens) - 1) // seq
-            self.num_samples = min(num_samples or max_samples, max_samples)
+            avail = max(0, max_samples - sample_offset)
+            self.num_samples = min(num_samples or avail, avail)
             self.sample_idx = None
 
     def _build_doc_aware_indices(self, num_samples):
@@ -146,8 +149,10 @@ class GPTDataset(torch.utils.data.Dataset):
                 sizes, doc_idx, seq, num_epochs, tokens_per_epoch)
         self.doc_idx = doc_idx
         avail = self.sample_idx.shape[0] - 1
-        self.num_samples = min(want, avail)
-        self.shuffle_idx = rng.permutation(avail)[:self.num_samples]
+        perm = rng.permutation(avail)
+        off = min(self.sample_offset, avail)
+        self.num_samples = min(want, avail - off)
+        self.shuffle_idx = perm[off:off + self.num_samples]
 
     def __len__(self):
         return self.num_samples
@@ -173,7 +178,7 @@ class GPTDataset(torch.utils.data.Dataset):
         s = self.config.sequence_length
         if self.sample_idx is not None:
             return _build_sample(self._doc_aware_window(idx), self.config)
-        start = idx * s
+        start = (idx + self.sample_offset) * s
         window = np.asarray(self.tokens[start:start + s + 1])
         return _build_sample(window, self.config)
 
@@ -201,6 +206,79 @@ class BlendedMegatronDatasetBuilder:
                 out.append(MockGPTDataset(cfg, num_samples=size,
                                           name=f"split{split_idx}"))
             else:
-                paths = self.config.blend
-                out.append(self.cls(self.config, paths[0], num_samples=size))
+                out.append(self._build_real(split_idx, size))
         return out
+
+    def _split_offsets(self, path):
+        """Partition a source's sample space by config.split ratios."""
+        probe = self.cls(self.config, path)
+        total = len(probe)
+        if not self.config.split:
+            return {0: (0, total), 1: (0, 0), 2: (0, 0)}
+        parts = [float(x) for x in str(self.config.split).split(",")]
+        parts += [0.0] * (3 - len(parts))
+        norm = sum(parts) or 1.0
+        counts = [int(total * p / norm) for p in parts]
+        offs, off = {}, 0
+        for i, c in enumerate(counts):
+            offs[i] = (off, c)
+            off += c
+        return offs
+
+    def _build_real(self, split_idx, size):
+        """Real token files: split-ratio windows per source; multiple
+        sources interleave by the native blending indices."""
+        blend = self.config.blend
+        # reference blend format may interleave weights and paths
+        if all(isinstance(b, str) for b in blend):
+            paths = list(blend)
+            weights = [1.0] * len(paths)
+        else:
+            weights = [float(b) for b in blend[0::2]]
+            paths = list(blend[1::2])
+        sources = []
+        for path in paths:
+            off, avail = self._split_offsets(path)[split_idx]
+            n = min(size, avail) if avail else 0
+            sources.append(self.cls(self.config, path, num_samples=n,
+                                    sample_offset=off,
+                                    name=f"split{split_idx}"))
+        if len(sources) == 1:
+            return sources[0]
+        return BlendedDataset(sources, weights, size)
+
+
+class BlendedDataset(torch.utils.data.Dataset):
+    """Proportional interleave of sources via the native
+    build_blending_indices helper (reference blended_dataset.py)."""
+
+    def __init__(self, datasets, weights, size):
+        import numpy as _np
+        self.datasets = datasets
+        cap = sum(len(d) for d in datasets)
+        self.size = min(size, cap) if cap else 0
+        w = _np.array(weights, dtype=_np.float64)
+        w = w / w.sum()
+        try:
+            from .build_helpers import load_helpers
+            di, dsi = load_helpers().build_blending_indices(w, self.size)
+            self.dataset_index = _np.asarray(di)
+            self.dataset_sample_index = _np.asarray(dsi)
+        except Exception:
+            counts = _np.zeros(len(datasets), dtype=_np.int64)
+            self.dataset_index = _np.zeros(self.size, dtype=_np.int16)
+            self.dataset_sample_index = _np.zeros(self.size, dtype=_np.int64)
+            for i in range(self.size):
+                err = w * (i + 1) - counts
+                d = int(err.argmax())
+                self.dataset_index[i] = d
+                self.dataset_sample_index[i] = counts[d]
+                counts[d] += 1
+
+    def __len__(self):
+        return self.size
+
+    def __getitem__(self, idx):
+        d = int(self.dataset_index[idx])
+        s = int(self.dataset_sample_index[idx]) % max(len(self.datasets[d]), 1)
+        return self.datasets[d][s]

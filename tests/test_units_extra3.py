@@ -269,3 +269,36 @@ def test_fp16_optimizer_skips_overflowed_step():
                   for n, p in ddp.named_parameters())
     assert changed
     destroy()
+
+
+def test_blend_and_split_real_sources(tmp_path):
+    """Two token files blended 3:1 with 80/10/10 split ratios: proportions
+    hold and train/valid windows don't overlap."""
+    import numpy as np
+    from megatronapp_amd.core.datasets.gpt_dataset import (
+        BlendedMegatronDatasetBuilder, GPTDataset, GPTDatasetConfig)
+    rng = np.random.RandomState(0)
+    paths = []
+    for i in range(2):
+        p = str(tmp_path / f"src{i}.npy")
+        np.save(p, rng.randint(0, 100, size=8 * 401 + 1).astype(np.int32))
+        paths.append(p)
+    cfg = GPTDatasetConfig(sequence_length=8, vocab_size=100, mock=False,
+                           blend=[3.0, paths[0], 1.0, paths[1]],
+                           split="80,10,10")
+    train, valid, test = BlendedMegatronDatasetBuilder(
+        GPTDataset, [200, 20, 20], lambda: True, cfg).build()
+    assert len(train) == 200 and len(valid) == 20
+    from collections import Counter
+    c = Counter(int(train.dataset_index[i]) for i in range(len(train)))
+    assert abs(c[0] - 150) <= 2 and abs(c[1] - 50) <= 2   # 3:1 blend
+    # split windows disjoint within a source: train uses offsets < valid's
+    s0_train = train.datasets[0]
+    s0_valid = valid.datasets[0]
+    assert s0_train.sample_offset == 0
+    assert s0_valid.sample_offset >= len(s0_train)
+    # samples decode correctly from the right window
+    t = s0_valid[0]["tokens"]
+    raw = np.load(paths[0])
+    start = s0_valid.sample_offset * 8
+    assert np.array_equal(t.numpy(), raw[start:start + 8])
