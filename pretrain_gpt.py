@@ -72,7 +72,8 @@ def train_valid_test_datasets_provider(train_val_test_num_samples):
         reset_attention_mask=args.reset_attention_mask,
         eod_mask_loss=args.eod_mask_loss,
         create_attention_mask=args.create_attention_mask_in_dataloader,
-        blend=args.data_path, mock=args.mock_data or args.data_path is None)
+        blend=args.data_path, split=args.split,
+        mock=args.mock_data or args.data_path is None)
     builder = BlendedMegatronDatasetBuilder(
         GPTDataset, train_val_test_num_samples, lambda: True, config)
     return builder.build()
