@@ -133,7 +133,8 @@ def _add_training_args(p):
     g.add_argument("--distribute-saved-activations", action="store_true")
     g.add_argument("--no-overlap-grad-reduce", action="store_false",
                    dest="overlap_grad_reduce")
-    g.add_argument("--overlap-param-gather", action="store_true")
+    g.add_argument("--overlap-param-gather", action="store_true",
+                   help="accepted for parity; ZeRO-1 param all-gather is synchronous (roadmap)")
     g.add_argument("--use-distributed-optimizer", action="store_true")
     g.add_argument("--ddp-bucket-size", type=int, default=None)
     g.add_argument("--check-weight-hash-across-dp-replicas-interval",
@@ -225,6 +226,7 @@ def _add_data_args(p):
     g.add_argument("--mock-data", action="store_true")
     g.add_argument("--num-workers", type=int, default=2)
     g.add_argument("--dataloader-type", default="single",
+                   help="accepted for parity; the loader is cyclic by construction",
                    choices=["single", "cyclic"])
     g.add_argument("--eod-mask-loss", action="store_true")
     g.add_argument("--reset-position-ids", action="store_true")
@@ -254,7 +256,8 @@ def _add_checkpointing_args(p):
                    help="arm a simulated fault (failure-detection demos)")
     g.add_argument("--simulate-fault-rank", type=int, default=0)
     g.add_argument("--simulate-fault-delay", type=float, default=30.0)
-    g.add_argument("--use-checkpoint-args", action="store_true")
+    g.add_argument("--use-checkpoint-args", action="store_true",
+                   help="accepted for reference parity; args come from the CLI/YAML here")
 
 
 def _add_logging_args(p):
