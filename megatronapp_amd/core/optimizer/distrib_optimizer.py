@@ -181,6 +181,12 @@ class DistributedOptimizer:
         grads = [self._shard_grad(i) for i in range(len(self.buffers))]
         return count_zeros_fp32(grads, parallel_state.get_model_parallel_group())
 
+    def finish_param_sync(self):
+        """Wait for an in-flight async param all-gather (overlap mode)."""
+        for h in getattr(self, "_param_sync_handles", []):
+            h.wait()
+        self._param_sync_handles = []
+
     def _ranges_aligned(self, i) -> bool:
         """The single-launch ranged kernel decides wd per float4; it needs
         every no-wd boundary 4-aligned (true for transformer param sizes)."""
@@ -286,11 +292,16 @@ class DistributedOptimizer:
                             p.data.reshape(-1)[os_ - s:oe - s].copy_(
                                 master[os_ - lo:oe - lo])
 
-            # ZeRO-1: all-gather updated params
+            # ZeRO-1: all-gather updated params; with overlap_param_gather
+            # the gather runs async and the DDP forward pre-hook waits
             if self.sharded:
+                overlap = getattr(self.config, "overlap_param_gather", False)
+                self._param_sync_handles = []
                 for buf in self.buffers:
                     if buf.ddp_config.use_distributed_optimizer:
-                        buf.start_param_sync(async_op=False)
+                        h = buf.start_param_sync(async_op=overlap)
+                        if h is not None:
+                            self._param_sync_handles.append(h)
 
         return True, grad_norm, num_zeros
 

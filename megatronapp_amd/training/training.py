@@ -124,6 +124,7 @@ def setup_model_and_optimizer(model_provider_func, model_type,
         loss_scale_window=args.loss_scale_window,
         hysteresis=args.hysteresis,
         use_distributed_optimizer=args.use_distributed_optimizer,
+        overlap_param_gather=args.overlap_param_gather,
         log_num_zeros_in_grad=args.log_num_zeros_in_grad)
     optimizer = get_megatron_optimizer(opt_config, model)
     if args.fp16 and hasattr(optimizer, "scale_loss"):
@@ -152,6 +153,8 @@ def train_step(forward_step_func, data_iterator, model, optimizer,
     from ..core.rerun_state_machine import get_rerun_state_machine
     rerun = get_rerun_state_machine()
 
+    if hasattr(optimizer, "finish_param_sync"):
+        optimizer.finish_param_sync()   # overlap-param-gather from last step
     fb_func = get_forward_backward_func()
     losses_reduced = []
     while rerun.should_run_forward_backward(data_iterator):

@@ -101,9 +101,12 @@ def _zero1_worker(rank, world, tmpdir, use_dist_opt):
     opt = get_megatron_optimizer(
         OptimizerConfig(optimizer="adam", lr=1e-3, min_lr=0.0,
                         weight_decay=0.01, clip_grad=1.0,
+                        overlap_param_gather=use_dist_opt,
                         use_distributed_optimizer=use_dist_opt), [ddp])
     losses = []
     for step in range(4):
+        if hasattr(opt, "finish_param_sync"):
+            opt.finish_param_sync()
         tok, pos = _data(step, 0)
         ddp.zero_grad_buffer()
         loss = ddp(tok, pos, None, labels=tok).float().mean()
