@@ -134,7 +134,7 @@ def _add_training_args(p):
     g.add_argument("--no-overlap-grad-reduce", action="store_false",
                    dest="overlap_grad_reduce")
     g.add_argument("--overlap-param-gather", action="store_true",
-                   help="accepted for parity; ZeRO-1 param all-gather is synchronous (roadmap)")
+                   help="overlap the ZeRO-1 param all-gather into the next step")
     g.add_argument("--use-distributed-optimizer", action="store_true")
     g.add_argument("--ddp-bucket-size", type=int, default=None)
     g.add_argument("--check-weight-hash-across-dp-replicas-interval",
