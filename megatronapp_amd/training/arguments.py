@@ -74,7 +74,10 @@ def _add_model_args(p):
     g.add_argument("--hidden-dropout", type=float, default=0.1)
     g.add_argument("--layernorm-epsilon", type=float, default=1e-5)
     g.add_argument("--apply-query-key-layer-scaling", action="store_true")
-    g.add_argument("--attention-softmax-in-fp32", action="store_true")
+    g.add_argument("--attention-softmax-in-fp32", action="store_true",
+                   default=True)
+    g.add_argument("--no-attention-softmax-in-fp32", action="store_false",
+                   dest="attention_softmax_in_fp32")
     g.add_argument("--attention-backend", default="auto",
                    choices=["auto", "flash", "fused", "unfused"])
     g.add_argument("--init-method-std", type=float, default=0.02)
