@@ -59,6 +59,27 @@ class InferenceGenerate:
     def run(self, request: dict, websocket: Optional[WebSocketConnection]):
         tokenizer = self.engine.controller.tokenizer
         prompts = request.get("prompts") or []
+        # request-size guard (reference --max-tokens-to-oom): refuse
+        # generations that would exceed the configured token budget
+        try:
+            from ..training.global_vars import get_args
+            args = get_args()
+            budget = getattr(args, "max_tokens_to_oom", None)
+            max_seq = getattr(args, "inference_max_seq_length", None)
+        except Exception:
+            budget = max_seq = None
+        n_new = int(request.get("tokens_to_generate", 64))
+        total = sum(len(tokenizer.tokenize(p)) + n_new for p in prompts)
+        if budget and total > budget:
+            err = {"type": "error",
+                   "message": f"request of {total} tokens exceeds "
+                              f"--max-tokens-to-oom {budget}"}
+            if websocket is not None:
+                websocket.send(err)
+            return err
+        if max_seq:
+            request = dict(request)
+            request["tokens_to_generate"] = min(n_new, max_seq)
         report = (lambda msg: websocket.send(msg)) if websocket else (lambda msg: None)
         _apply_configs(request, self.num_layers, tokenizer, report)
 

@@ -177,3 +177,25 @@ def test_inference_engine_generates_gpu():
     assert out["text"] == out2["text"]
     assert len(out["logprobs"][0]) > 0
     destroy()
+
+
+def test_server_token_budget_guard():
+    """--max-tokens-to-oom rejects oversized generation requests."""
+    initialize_model_parallel()
+    from megatronapp_amd.core.inference.static_engine import (
+        get_inference_engine)
+    from megatronapp_amd.inference.text_generation_server import (
+        InferenceGenerate)
+    from megatronapp_amd.training.tokenizer import NullTokenizer
+    from megatronapp_amd.training import global_vars
+    import argparse
+    args = argparse.Namespace(max_tokens_to_oom=32,
+                              inference_max_seq_length=None, rank=0)
+    global_vars._GLOBAL_ARGS = args
+    torch.manual_seed(0)
+    engine = get_inference_engine(_tiny_model().eval(), NullTokenizer(64), 4)
+    srv = InferenceGenerate(engine, num_layers=2)
+    out = srv.run({"prompts": ["1 2 3"], "tokens_to_generate": 64}, None)
+    assert out and out.get("type") == "error"
+    global_vars._GLOBAL_ARGS = None
+    destroy()
