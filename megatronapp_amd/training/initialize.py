@@ -27,6 +27,8 @@ def initialize_megatron(extra_args_provider=None, args_defaults={},
                         skip_mpu_initialization=False,
                         parsed_args=None):
     args = parsed_args or parse_args(extra_args_provider, ignore_unknown_args)
+    if getattr(args, "use_checkpoint_args", False) and args.load:
+        _apply_checkpoint_args(args)
     validate_args(args, args_defaults)
     set_global_variables(args)
 
@@ -105,3 +107,45 @@ def write_args_to_tensorboard(args=None, writer=None):
         return
     for k, v in sorted(vars(args).items()):
         writer.add_text(k, str(v))
+
+
+_CKPT_ARCH_ARGS = [
+    "num_layers", "hidden_size", "ffn_hidden_size", "num_attention_heads",
+    "num_query_groups", "group_query_attention", "kv_channels",
+    "max_position_embeddings", "position_embedding_type", "normalization",
+    "swiglu", "add_bias_linear", "untie_embeddings_and_output_weights",
+    "vocab_size", "padded_vocab_size", "seq_length", "mtp_num_layers",
+    "num_experts", "moe_router_topk"]
+
+
+def _apply_checkpoint_args(args):
+    """--use-checkpoint-args: architecture flags come from the checkpoint
+    (reference checkpointing.load_args_from_checkpoint)."""
+    import torch as _t
+    from .checkpointing import (get_checkpoint_name,
+                                get_checkpoint_tracker_filename,
+                                read_metadata)
+    tracker = get_checkpoint_tracker_filename(args.load)
+    if not os.path.isfile(tracker):
+        return
+    iteration, release = read_metadata(tracker)
+    base = get_checkpoint_name(args.load, iteration, release,
+                               return_base_dir=True)
+    saved = None
+    if os.path.exists(os.path.join(base, "common.pt")):
+        saved = _t.load(os.path.join(base, "common.pt"), map_location="cpu",
+                        weights_only=False).get("args")
+    else:
+        legacy = os.path.join(base, "mp_rank_00", "model_optim_rng.pt")
+        if os.path.exists(legacy):
+            saved = _t.load(legacy, map_location="cpu",
+                            weights_only=False).get("args")
+    if not saved:
+        return
+    for key in _CKPT_ARCH_ARGS:
+        if key in saved and saved[key] is not None:
+            setattr(args, key, saved[key])
+    if int(os.environ.get("RANK", "0")) == 0:
+        # rank is set by validate_args later; use the env here
+        print(f"  architecture args restored from checkpoint at "
+              f"iteration {iteration}")

@@ -203,3 +203,18 @@ def test_log_params_norm(tmp_path):
     from tests.test_checkpointing import _run
     out = _run(["--train-iters", "2", "--log-params-norm"], 29650)
     assert "params norm:" in out
+
+
+def test_use_checkpoint_args(tmp_path):
+    """Resume with --use-checkpoint-args restores the architecture even if
+    the CLI disagrees."""
+    from tests.test_checkpointing import _run
+    save = str(tmp_path / "ca")
+    _run(["--train-iters", "2", "--save", save, "--save-interval", "2",
+          "--ckpt-format", "torch_dist"], 29652)
+    # resume with WRONG --hidden-size on the CLI; checkpoint args win
+    out = _run(["--train-iters", "4", "--load", save, "--hidden-size",
+                "128", "--use-checkpoint-args", "--ckpt-format",
+                "torch_dist", "--save-interval", "100"], 29653)
+    assert "architecture args restored" in out
+    assert "loaded checkpoint (torch_dist)" in out
