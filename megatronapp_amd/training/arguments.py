@@ -260,7 +260,7 @@ def _add_checkpointing_args(p):
     g.add_argument("--simulate-fault-rank", type=int, default=0)
     g.add_argument("--simulate-fault-delay", type=float, default=30.0)
     g.add_argument("--use-checkpoint-args", action="store_true",
-                   help="accepted for reference parity; args come from the CLI/YAML here")
+                   help="restore architecture args from the checkpoint")
 
 
 def _add_logging_args(p):
