@@ -224,6 +224,11 @@ def training_log(loss_dict, total_loss_dict, learning_rate, iteration,
             writer.add_scalar(k, v, iteration)
         writer.add_scalar("learning-rate", learning_rate, iteration)
         writer.add_scalar("throughput", throughput_tps, iteration)
+        if getattr(args, "log_timers_to_tensorboard", False):
+            from .global_vars import get_timers
+            timers = get_timers()
+            if timers is not None:
+                timers.write(list(timers._timers), writer, iteration)
     if report_memory_flag:
         # once, after optimizer state exists (reference training.py:1745)
         if (not torch.distributed.is_initialized()
