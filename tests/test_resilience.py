@@ -64,3 +64,42 @@ def test_straggler_detector_reports():
     out = det2.report()
     assert out is not None and out["min_rank"] == 0
     destroy()
+
+
+def test_sigterm_graceful_exit_with_checkpoint(tmp_path):
+    """--exit-signal-handler: SIGTERM mid-training saves a checkpoint and
+    exits cleanly (reference dist_signal_handler semantics)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+    import time
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    save = str(tmp_path / "sig_ck")
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29672",
+               RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
+    proc = subprocess.Popen(
+        [sys.executable, os.path.join(REPO, "pretrain_gpt.py"),
+         "--num-layers", "2", "--hidden-size", "64",
+         "--num-attention-heads", "4", "--seq-length", "32",
+         "--max-position-embeddings", "32", "--micro-batch-size", "2",
+         "--global-batch-size", "4", "--vocab-size", "128", "--mock-data",
+         "--train-iters", "100000", "--lr", "1e-3", "--log-interval", "5",
+         "--eval-iters", "0", "--hidden-dropout", "0",
+         "--attention-dropout", "0", "--exit-signal-handler",
+         "--save", save, "--save-interval", "100000"],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True, cwd=REPO,
+        env=env)
+    # wait until training is running, then SIGTERM
+    time.sleep(20)
+    proc.send_signal(signal.SIGTERM)
+    try:
+        out, err = proc.communicate(timeout=120)
+    except subprocess.TimeoutExpired:
+        proc.kill()
+        raise
+    assert proc.returncode == 0, err[-2000:]
+    assert "exit" in out.lower()
+    assert os.path.exists(
+        os.path.join(save, "latest_checkpointed_iteration.txt"))
