@@ -302,3 +302,25 @@ def test_blend_and_split_real_sources(tmp_path):
     raw = np.load(paths[0])
     start = s0_valid.sample_offset * 8
     assert np.array_equal(t.numpy(), raw[start:start + 8])
+
+
+def test_preprocess_data_cli(tmp_path):
+    """jsonl corpus -> preprocess_data CLI -> indexed dataset -> GPTDataset
+    doc-aware samples."""
+    import json as _json
+    import subprocess
+    import sys
+    corpus = tmp_path / "docs.jsonl"
+    with open(corpus, "w") as f:
+        for i in range(20):
+            f.write(_json.dumps({"text": " ".join(
+                str((i * 13 + j) % 200) for j in range(30))}) + "\n")
+    prefix = str(tmp_path / "out")
+    r = subprocess.run(
+        [sys.executable, "tools/preprocess_data.py", "--input", str(corpus),
+         "--output-prefix", prefix, "--tokenizer-type", "NullTokenizer",
+         "--vocab-size", "256"],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    assert os.path.exists(prefix + "_text_document.bin") or \
+        os.path.exists(prefix + ".bin"), os.listdir(tmp_path)
