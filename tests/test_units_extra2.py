@@ -161,3 +161,41 @@ def test_theoretical_memory_moe_branch():
     args.num_experts = None
     dense = compute_weight_and_optimizer_memory(args)
     assert moe > dense  # experts multiply mlp params
+
+
+def test_cpu_fallback_oracles_match_torch():
+    """The CPU fallback paths are the oracles the GPU kernels test
+    against — pin them to torch's own primitives."""
+    from megatronapp_amd.core.fusions.fused_layer_norm import (
+        _LayerNormFn, _RMSNormFn)
+    from megatronapp_amd.core.fusions.fused_bias_act import _BiasGeluFn
+    from megatronapp_amd.core.fusions.fused_softmax import (
+        ScaledUpperTriangMaskedSoftmax)
+    torch.manual_seed(4)
+    x = torch.randn(6, 32, requires_grad=True)
+    w = torch.rand(32, requires_grad=True) + 0.5
+    b = torch.randn(32, requires_grad=True)
+
+    y = _LayerNormFn.apply(x, w, b, 1e-5)
+    ref = torch.nn.functional.layer_norm(x, (32,), w, b, 1e-5)
+    assert torch.allclose(y, ref, atol=1e-5)
+    g = torch.autograd.grad(y.sum(), (x, w, b))
+    gr = torch.autograd.grad(ref.sum(), (x, w, b))
+    for a, r in zip(g, gr):
+        assert torch.allclose(a, r, atol=1e-4)
+
+    y2 = _RMSNormFn.apply(x, w, 1e-6)
+    ref2 = torch.nn.functional.rms_norm(x, (32,), w, 1e-6)
+    assert torch.allclose(y2, ref2, atol=1e-5)
+
+    xb = torch.randn(8, 16, requires_grad=True)
+    bb = torch.randn(16, requires_grad=True)
+    y3 = _BiasGeluFn.apply(xb, bb)
+    ref3 = torch.nn.functional.gelu(xb + bb, approximate="tanh")
+    assert torch.allclose(y3, ref3, atol=1e-5)
+
+    s = torch.randn(2, 8, 8)
+    p = ScaledUpperTriangMaskedSoftmax.apply(s, 0.5)
+    mask = torch.triu(torch.ones(8, 8, dtype=torch.bool), 1)
+    refp = torch.softmax((s * 0.5).masked_fill(mask, float("-inf")), -1)
+    assert torch.allclose(p, refp, atol=1e-6)
