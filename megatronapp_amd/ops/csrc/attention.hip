@@ -374,6 +374,270 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd_kernel(
   }
 }
 
+// ------------------------------------------------------------------
+// Transposed-S forward (EXPERIMENTAL, round-2 candidate; not the default
+// path).  Computes S^T = K*Q^T so the MFMA C layout leaves each lane
+// owning ONE query column: softmax state becomes per-lane scalars, the
+// same Q registers serve as the B operand (A-frag of Q == B-frag of
+// Q^T), and P^T reaches the O^T MFMA through ds_bpermute lane exchanges
+// instead of a bf16-swizzled LDS round trip (the VALU/LDS hotspot of the
+// default kernel).  Correctness is tested (xfail-tolerant) before any
+// perf work.
+__device__ __forceinline__ unsigned pack_bf16x2(float lo, float hi) {
+  return (unsigned)f2bf(lo) | ((unsigned)f2bf(hi) << 16);
+}
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(FWD_BLOCK) void attn_fwd_t_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, unsigned short* __restrict__ o,
+    float* __restrict__ lse, int sq, int sk, int b, int nh, int ng,
+    float scale) {
+  const int qtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / nh;
+  const int h = bh % nh;
+  const int hkv = h / (nh / ng);
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lrow = lane >> 4;
+  const int lcol = lane & 15;
+
+  const long q_ss = (long)b * nh * D;
+  const long k_ss = (long)b * ng * D;
+  const unsigned short* qp = q + ((long)bi * nh + h) * D;
+  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
+  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* k_lds = (unsigned short*)smem;               // KVBLK*D
+  unsigned short* vt_lds = k_lds + KVBLK * D;                  // D*TSTRIDE
+  unsigned short* stage = vt_lds + (long)D * TSTRIDE;          // FQBLK*D
+
+  const int q0 = qtile * FQBLK + wid * QW;
+  const int qrow = q0 + lcol;              // THIS lane's query row
+
+  constexpr int DF = D / 32;
+  bf16x8 qfrag[DF];
+  {
+    const unsigned short* src = qp + (long)qrow * q_ss;
+#pragma unroll
+    for (int f = 0; f < DF; ++f) {
+      if (qrow < sq)
+        qfrag[f] = *(const bf16x8*)(src + f * 32 + lrow * 8);
+      else
+        qfrag[f] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  float m_run = -INFINITY;
+  float l_run = 0.f;
+  constexpr int DS = D / 16;
+  f32x4 oacc[DS];                          // O^T[d = dsub*16+lrow*4+r][q]
+#pragma unroll
+  for (int dsub = 0; dsub < DS; ++dsub) oacc[dsub] = f32x4{0, 0, 0, 0};
+
+  const int q_hi = qtile * FQBLK + FQBLK - 1;
+  int kv_end = sk;
+  if (CAUSAL) kv_end = min(sk, q_hi + 1 + (sk - sq));
+  const int n_kv_tiles = (kv_end + KVBLK - 1) / KVBLK;
+
+  constexpr int PIECES = KVBLK * D / 8 / FWD_BLOCK;
+  bf16x8 kreg[PIECES], vreg[PIECES];
+
+  auto stage_load = [&](int t) {
+    const int kv0 = t * KVBLK;
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int grow = kv0 + row;
+      kreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) =
+          kreg[pc];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[(long)(col + j) * TSTRIDE + row] = (unsigned short)vreg[pc][j];
+    }
+  };
+
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+
+  for (int t = 0; t < n_kv_tiles; ++t) {
+    const int kv0 = t * KVBLK;
+    if (t + 1 < n_kv_tiles) stage_load(t + 1);
+
+    // ---- S^T = K Q^T: C[kv = ksub*16+lrow*4+r][q = lcol]
+    f32x4 st[KVBLK / 16];
+#pragma unroll
+    for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+      st[ksub] = f32x4{0, 0, 0, 0};
+#pragma unroll
+      for (int f = 0; f < DF; ++f) {
+        const int krow = ksub * 16 + lcol;     // A-frag m = kv
+        bf16x8 afrag = *(const bf16x8*)((char*)(k_lds + (long)krow * D) +
+                                        swz(krow, (f * 32 + lrow * 8) * 2));
+        st[ksub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, qfrag[f], st[ksub], 0, 0, 0);
+      }
+    }
+
+    // ---- mask + per-lane (per-q) online softmax
+    float mtile = -INFINITY;
+#pragma unroll
+    for (int ksub = 0; ksub < KVBLK / 16; ++ksub) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kvcol = kv0 + ksub * 16 + lrow * 4 + r;
+        float sv = st[ksub][r] * scale;
+        bool valid = (kvcol < sk) && (qrow < sq);
+        if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
+        sv = valid ? sv : -INFINITY;
+        st[ksub][r] = sv;
+        mtile = fmaxf(mtile, sv);
+      }
+    }
+    // reduce across the 4 lane groups holding this q column
+    mtile = fmaxf(mtile, __shfl_xor(mtile, 16, WAVE));
+    mtile = fmaxf(mtile, __shfl_xor(mtile, 32, WAVE));
+
+    const float mn = fmaxf(m_run, mtile);
+    const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - mn);
+    m_run = mn;
+
+    float pf[KVBLK / 16][4];
+    float lt = 0.f;
+#pragma unroll
+    for (int ksub = 0; ksub < KVBLK / 16; ++ksub)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float sv = st[ksub][r];
+        float p = (sv == -INFINITY) ? 0.f : __expf(sv - m_run);
+        pf[ksub][r] = p;
+        lt += p;
+      }
+    lt += __shfl_xor(lt, 16, WAVE);
+    lt += __shfl_xor(lt, 32, WAVE);
+    l_run = l_run * alpha + lt;
+
+#pragma unroll
+    for (int dsub = 0; dsub < DS; ++dsub)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) oacc[dsub][r] *= alpha;
+
+    // ---- O^T += V^T P^T, P^T fragments assembled via ds_bpermute
+#pragma unroll
+    for (int ks = 0; ks < KVBLK / 32; ++ks) {
+      unsigned pk[2][2];
+#pragma unroll
+      for (int tsub = 0; tsub < 2; ++tsub) {
+        const int ksub = ks * 2 + tsub;
+        pk[tsub][0] = pack_bf16x2(pf[ksub][0], pf[ksub][1]);
+        pk[tsub][1] = pack_bf16x2(pf[ksub][2], pf[ksub][3]);
+      }
+      unsigned bp[4];
+#pragma unroll
+      for (int pp = 0; pp < 4; ++pp) {
+        const int src_lane =
+            (((lrow & 1) * 2 + (pp >> 1)) * 16 + lcol) << 2;
+        const int v0 = __builtin_amdgcn_ds_bpermute(src_lane,
+                                                    (int)pk[0][pp & 1]);
+        const int v1 = __builtin_amdgcn_ds_bpermute(src_lane,
+                                                    (int)pk[1][pp & 1]);
+        bp[pp] = (lrow >= 2) ? (unsigned)v1 : (unsigned)v0;
+      }
+      bf16x8 pfrag;
+#pragma unroll
+      for (int pp = 0; pp < 4; ++pp) {
+        pfrag[2 * pp] = (short)(bp[pp] & 0xffff);
+        pfrag[2 * pp + 1] = (short)(bp[pp] >> 16);
+      }
+#pragma unroll
+      for (int dsub = 0; dsub < DS; ++dsub) {
+        const int vrow = dsub * 16 + lcol;     // A-frag m = d
+        bf16x8 vfrag = *(const bf16x8*)(vt_lds + (long)vrow * TSTRIDE +
+                                        ks * 32 + lrow * 8);
+        oacc[dsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            vfrag, pfrag, oacc[dsub], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+    if (t + 1 < n_kv_tiles) {
+      stage_write();
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: lane owns q = qrow; O = O^T transposed through LDS
+  const float denom = l_run > 0.f ? 1.f / l_run : 0.f;
+  unsigned short* o_stage = stage + (long)wid * QW * D;
+#pragma unroll
+  for (int dsub = 0; dsub < DS; ++dsub)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      o_stage[(long)lcol * D + dsub * 16 + lrow * 4 + r] =
+          f2bf(oacc[dsub][r] * denom);
+  if (lrow == 0 && qrow < sq)
+    lse[((long)bi * nh + h) * sq + qrow] =
+        m_run + __logf(fmaxf(l_run, 1e-30f));
+  __syncthreads();
+  {
+    constexpr int pieces = QW * D / 8 / WAVE;
+#pragma unroll
+    for (int pc = 0; pc < pieces; ++pc) {
+      const int idx = (pc * WAVE + lane) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int qq = q0 + row;
+      if (qq < sq)
+        *(bf16x8*)(o + ((long)qq * b * nh + (long)bi * nh + h) * D + col) =
+            *(const bf16x8*)(o_stage + (long)row * D + col);
+    }
+  }
+}
+
+void launch_attn_fwd_t(const void* q, const void* k, const void* v, void* o,
+                       float* lse, int sq, int sk, int b, int nh, int ng,
+                       int d, float scale, bool causal, hipStream_t stream) {
+  if (sq % FQBLK != 0 || sk % KVBLK != 0)
+    throw std::runtime_error(
+        "attn_fwd_t: sq must be a multiple of 128, sk of 64");
+  if (d != 128) throw std::runtime_error("attn_fwd_t: d must be 128");
+  dim3 grid(sq / FQBLK, b * nh);
+  dim3 block(FWD_BLOCK);
+  const size_t lds =
+      (size_t)(KVBLK * d + d * TSTRIDE + FQBLK * d) * sizeof(unsigned short);
+  if (causal)
+    hipLaunchKernelGGL((attn_fwd_t_kernel<128, true>), grid, block, lds,
+                       stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+  else
+    hipLaunchKernelGGL((attn_fwd_t_kernel<128, false>), grid, block, lds,
+                       stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+  HIP_CHECK_LAUNCH();
+}
+
 // ablation entry (perf diagnosis only; outputs wrong for level>0)
 void launch_attn_fwd_ablate(const void* q, const void* k, const void* v,
                             void* o, float* lse, int sq, int sk, int b,
