@@ -25,6 +25,15 @@ def main():
         dq, dk, dv = ops.get_ops().attn_bwd(do, q, k, v, o, lse, scale, True)
     torch.cuda.synchronize(); dt = (time.perf_counter() - t0) / reps
     print(f"attn_bwd: {dt*1e6:.0f} us  {2.5*flops_fwd/dt/1e12:.0f} TF-equiv")
+    if hasattr(ops.get_ops(), "attn_fwd_t"):
+        ot, _ = ops.get_ops().attn_fwd_t(q, k, v, scale, True)
+        err = (ot.float() - o.float()).abs().max()
+        torch.cuda.synchronize(); t0 = time.perf_counter()
+        for _ in range(reps):
+            ops.get_ops().attn_fwd_t(q, k, v, scale, True)
+        torch.cuda.synchronize(); dt = (time.perf_counter() - t0) / reps
+        print(f"attn_fwd_t (transposed-S): {dt*1e6:.0f} us  "
+              f"{flops_fwd/dt/1e12:.0f} TF  maxerr-vs-default {float(err):.4f}")
     for lvl, name in ((1, "noPV"), (2, "noSM"), (3, "loader-only")):
         torch.cuda.synchronize(); t0 = time.perf_counter()
         for _ in range(reps):
