@@ -6,6 +6,8 @@ import enum
 class ModelType(enum.Enum):
     encoder_or_decoder = 1
     encoder_and_decoder = 2
+    retro_encoder = 3
+    retro_decoder = 4
 
 
 class AttnMaskType(enum.Enum):

@@ -58,8 +58,14 @@ class TransformerLayer(MegatronModule):
             self.cross_attention = build_module(
                 submodules.cross_attention, config=config,
                 layer_number=layer_number)
+            # custom bias-dropout-add operator (e.g. retro's chunked
+            # re-permute); None -> the standard fused bda
+            self.cross_attn_bda = (
+                build_module(submodules.cross_attn_bda, config=config)
+                if submodules.cross_attn_bda is not None else None)
         else:
             self.cross_attention = None
+            self.cross_attn_bda = None
         self.pre_mlp_layernorm = build_module(
             submodules.pre_mlp_layernorm, config=config,
             hidden_size=config.hidden_size, eps=config.layernorm_epsilon)
@@ -106,8 +112,14 @@ class TransformerLayer(MegatronModule):
                 cross_out_with_bias = self.cross_attention(
                     normed, attention_mask=context_mask,
                     key_value_states=context)
+                if isinstance(cross_out_with_bias, dict) and \
+                        "context" in cross_out_with_bias:
+                    context = cross_out_with_bias["context"]
+                bda = (self.cross_attn_bda(
+                           self.training, self.config.bias_dropout_fusion)
+                       if self.cross_attn_bda is not None else self._bda())
                 with self.bias_dropout_add_exec_handler():
-                    hidden_states = self._bda()(
+                    hidden_states = bda(
                         cross_out_with_bias, residual, self.hidden_dropout)
 
             # --- MLP block ---
