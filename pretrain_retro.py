@@ -4,11 +4,10 @@
 Retrieval-augmented GPT: each sequence chunk is paired with
 ``retro_num_neighbors`` retrieved neighbor chunks (plus continuations),
 encoded by a small transformer and attended to via chunked
-cross-attention.  The retrieval database harness of the reference
-(tools/retro preprocessing) requires a corpus + embedding index; in this
-offline environment neighbors are synthesized per sample, so the
-training path, model and losses are fully exercised end to end — plug a
-real neighbor loader into ``MockRetroDataset`` to use retrieved data.
+cross-attention.  With ``--retro-project-dir`` the neighbors come from a
+preprocessed retrieval database (tools/retro/preprocess.py: chunk db ->
+BERT embeddings -> exact MIPS neighbor search); without it a synthetic
+neighbor stream exercises the same training path.
 
   torchrun --nproc-per-node 1 --master-addr 127.0.0.1 pretrain_retro.py \
       --num-layers 12 --hidden-size 512 --num-attention-heads 8 \
@@ -22,6 +21,7 @@ import sys
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+import numpy as np
 import torch
 
 from megatronapp_amd.core import parallel_state
@@ -116,8 +116,57 @@ class MockRetroDataset(torch.utils.data.Dataset):
         }
 
 
+class RetroProjectDataset(torch.utils.data.Dataset):
+    """Samples from a preprocessed retro project directory
+    (tools/retro/preprocess.py): each sample is ``l`` consecutive chunks
+    with their retrieved neighbors, ordered (l, k) so the batch
+    flattens to the (b, l, k) layout the neighbor encoder reshapes."""
+
+    def __init__(self, project_dir, seq_length, num_neighbors,
+                 num_retrieved_chunks):
+        from tools.retro.preprocess import (
+            load_retro_project, load_neighbor_tokens)
+        self._load_neighbor_tokens = load_neighbor_tokens
+        self.chunks, self.doc_ids, self.neighbors, self.meta = \
+            load_retro_project(project_dir)
+        self.m = self.meta["chunk_length"]
+        assert seq_length % self.m == 0
+        self.l = seq_length // self.m
+        self.k = min(num_neighbors, self.neighbors.shape[1])
+        self.num_retrieved_chunks = num_retrieved_chunks
+        self.n = max(len(self.chunks) // self.l, 1)
+
+    def __len__(self):
+        return self.n
+
+    def __getitem__(self, idx):
+        c0 = idx * self.l
+        sel = [min(c0 + j, len(self.chunks) - 1) for j in range(self.l)]
+        tokens = torch.as_tensor(
+            np.concatenate([self.chunks[c] for c in sel]),
+            dtype=torch.long)
+        labels = torch.roll(tokens, -1)
+        labels[-1] = 0
+        nts = [self._load_neighbor_tokens(
+                   self.chunks, self.doc_ids, self.neighbors[c, :self.k],
+                   self.meta["pad_id"], self.num_retrieved_chunks)
+               for c in sel]                     # l × [k, r]
+        neighbor_tokens = torch.as_tensor(
+            np.concatenate(nts, axis=0), dtype=torch.long)  # [l*k, r]
+        return {"tokens": tokens, "labels": labels,
+                "loss_mask": torch.ones(len(tokens)),
+                "position_ids": torch.arange(len(tokens)),
+                "neighbor_tokens": neighbor_tokens}
+
+
 def train_valid_test_datasets_provider(train_val_test_num_samples):
     args = get_args()
+    if args.retro_project_dir:
+        import numpy  # noqa: F401 - RetroProjectDataset uses np
+        ds = RetroProjectDataset(
+            args.retro_project_dir, args.seq_length,
+            args.retro_num_neighbors, args.retro_num_retrieved_chunks)
+        return ds, ds, ds
     r = args.retro_num_retrieved_chunks * args.retro_chunk_length
     mk = lambda n, seed: MockRetroDataset(
         max(n or 0, 1), args.seq_length, args.padded_vocab_size,
