@@ -36,6 +36,9 @@ def get_tasks_args(parser):
     g.add_argument("--valid-data", nargs="*", default=None)
     g.add_argument("--overlapping-eval", type=int, default=32)
     g.add_argument("--pretrained-checkpoint", type=str, default=None)
+    g.add_argument("--qa-data", type=str, default=None)
+    g.add_argument("--evidence-data", type=str, default=None)
+    g.add_argument("--biencoder-projection-dim", type=int, default=0)
     return parser
 
 
@@ -69,10 +72,13 @@ def main():
     elif task in ("WIKITEXT103", "LAMBADA"):
         from tasks.zeroshot_gpt.evaluate import main as zeroshot_main
         zeroshot_main(tokenizer)
+    elif task in ("ORQA", "NQ"):
+        from tasks.orqa.evaluate import main as orqa_main
+        orqa_main(tokenizer)
     else:
         raise NotImplementedError(f"task {args.task} is not implemented "
                                   "(available: MNLI QQP RACE WIKITEXT103 "
-                                  "LAMBADA)")
+                                  "LAMBADA ORQA)")
 
 
 if __name__ == "__main__":

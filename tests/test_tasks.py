@@ -168,3 +168,35 @@ def test_tasks_main_zeroshot_eval(tmp_path):
         capture_output=True, text=True, cwd=REPO, env=env, timeout=420)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "ppl" in out.stdout
+
+
+def test_orqa_retriever_eval(tmp_path):
+    """Retrieval-accuracy machinery: with k == all blocks every question
+    whose answer exists in evidence must be a hit."""
+    qa = tmp_path / "qa.jsonl"
+    ev = tmp_path / "evidence.jsonl"
+    qa.write_text("\n".join(json.dumps(d) for d in [
+        {"question": "capital of france", "answers": ["Paris"]},
+        {"question": "largest planet", "answers": ["Jupiter"]},
+        {"question": "unanswerable", "answers": ["zzz-not-present"]},
+    ]))
+    ev.write_text("\n".join(json.dumps({"text": t}) for t in [
+        "paris is the capital of france",
+        "jupiter is the largest planet",
+        "unrelated text block",
+    ]))
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29704",
+               RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tasks", "main.py"),
+         "--task", "ORQA", "--qa-data", str(qa),
+         "--evidence-data", str(ev),
+         "--num-layers", "2", "--hidden-size", "64",
+         "--num-attention-heads", "4", "--seq-length", "32",
+         "--max-position-embeddings", "64", "--micro-batch-size", "2",
+         "--global-batch-size", "2", "--lr", "1e-4",
+         "--hidden-dropout", "0", "--attention-dropout", "0"],
+        capture_output=True, text=True, cwd=REPO, env=env, timeout=420)
+    assert out.returncode == 0, out.stderr[-2000:]
+    # top20 covers all 3 blocks: exactly the 2 answerable hit
+    assert "top20_accuracy: 66.67%" in out.stdout
