@@ -137,10 +137,21 @@ def load_hf_gpt2(path):
     return impl(path)
 
 
+def save_hf_llama(full, common, path):
+    from saver_hf_llama import save_hf_llama as impl
+    impl(full, common, path)
+
+
+def load_hf_llama(path):
+    from saver_hf_llama import load_hf_llama as impl
+    return impl(path)
+
+
 LOADERS = {"torch_dist": load_torch_dist, "legacy": load_legacy,
-           "consolidated": load_consolidated, "hf_gpt2": load_hf_gpt2}
+           "consolidated": load_consolidated, "hf_gpt2": load_hf_gpt2,
+           "hf_llama": load_hf_llama}
 SAVERS = {"torch_dist": save_torch_dist, "consolidated": save_consolidated,
-          "hf_gpt2": save_hf_gpt2}
+          "hf_gpt2": save_hf_gpt2, "hf_llama": save_hf_llama}
 
 
 def main():
