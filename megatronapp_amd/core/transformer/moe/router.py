@@ -96,6 +96,8 @@ class TopKRouter(torch.nn.Module):
                 aux = switch_load_balancing_loss(
                     scores, tokens_per_expert, self.topk,
                     self.config.moe_aux_loss_coeff)
+        if getattr(self.config, "moe_router_renormalize", False):
+            probs = probs / probs.sum(dim=-1, keepdim=True)
         if self.config.moe_z_loss_coeff:
             z = z_loss_func(logits, self.config.moe_z_loss_coeff)
             aux = z if aux is None else aux + z

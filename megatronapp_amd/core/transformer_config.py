@@ -112,6 +112,9 @@ class TransformerConfig(ModelParallelConfig):
     num_moe_experts: Optional[int] = None
     moe_router_topk: int = 2
     moe_router_load_balancing_type: str = "aux_loss"  # | "sinkhorn" | "none"
+    # renormalize the selected top-k probs to sum to 1 (Mixtral-style
+    # routing; reference moe_router_topk_scaling / norm_topk_prob)
+    moe_router_renormalize: bool = False
     moe_aux_loss_coeff: float = 0.0
     moe_z_loss_coeff: Optional[float] = None
     moe_token_dispatcher_type: str = "alltoall"   # | "allgather"

@@ -147,11 +147,22 @@ def load_hf_llama(path):
     return impl(path)
 
 
+def save_hf_mixtral(full, common, path):
+    from saver_hf_mixtral import save_hf_mixtral as impl
+    impl(full, common, path)
+
+
+def load_hf_mixtral(path):
+    from saver_hf_mixtral import load_hf_mixtral as impl
+    return impl(path)
+
+
 LOADERS = {"torch_dist": load_torch_dist, "legacy": load_legacy,
            "consolidated": load_consolidated, "hf_gpt2": load_hf_gpt2,
-           "hf_llama": load_hf_llama}
+           "hf_llama": load_hf_llama, "hf_mixtral": load_hf_mixtral}
 SAVERS = {"torch_dist": save_torch_dist, "consolidated": save_consolidated,
-          "hf_gpt2": save_hf_gpt2, "hf_llama": save_hf_llama}
+          "hf_gpt2": save_hf_gpt2, "hf_llama": save_hf_llama,
+          "hf_mixtral": save_hf_mixtral}
 
 
 def main():
