@@ -39,6 +39,13 @@ def get_tasks_args(parser):
     g.add_argument("--qa-data", type=str, default=None)
     g.add_argument("--evidence-data", type=str, default=None)
     g.add_argument("--biencoder-projection-dim", type=int, default=0)
+    g.add_argument("--prompt-type", choices=["knowledge", "response"],
+                   default="knowledge")
+    g.add_argument("--sample-input-file", type=str, default=None)
+    g.add_argument("--sample-output-file", type=str, default=None)
+    g.add_argument("--guess-file", type=str, default=None)
+    g.add_argument("--answer-file", type=str, default=None)
+    g.add_argument("--out-seq-length", type=int, default=64)
     return parser
 
 
@@ -75,10 +82,25 @@ def main():
     elif task in ("ORQA", "NQ"):
         from tasks.orqa.evaluate import main as orqa_main
         orqa_main(tokenizer)
+    elif task in ("MSDP-PROMPT",):
+        import json
+        from tasks.msdp.prompt import run_prompting
+        from tasks.zeroshot_gpt.evaluate import build_gpt
+        model, device = build_gpt(args)
+        with open(args.sample_input_file, encoding="utf-8") as f:
+            samples = [json.loads(line) for line in f if line.strip()]
+        run_prompting(model, tokenizer, samples, args.prompt_type,
+                      args.out_seq_length, device,
+                      args.sample_output_file)
+        print(f"wrote {len(samples)} generations to "
+              f"{args.sample_output_file}", flush=True)
+    elif task in ("MSDP-EVAL-F1",):
+        from tasks.msdp.evaluate import evaluate_f1
+        evaluate_f1(args.guess_file, args.answer_file)
     else:
         raise NotImplementedError(f"task {args.task} is not implemented "
                                   "(available: MNLI QQP RACE WIKITEXT103 "
-                                  "LAMBADA ORQA)")
+                                  "LAMBADA ORQA MSDP-PROMPT MSDP-EVAL-F1)")
 
 
 if __name__ == "__main__":

@@ -200,3 +200,56 @@ def test_orqa_retriever_eval(tmp_path):
     assert out.returncode == 0, out.stderr[-2000:]
     # top20 covers all 3 blocks: exactly the 2 answerable hit
     assert "top20_accuracy: 66.67%" in out.stdout
+
+
+def test_msdp_f1_metric():
+    from tasks.msdp.metrics import F1Metric
+    p, r, f1 = F1Metric.compute_each_pair("the cat sat", "a cat sat down")
+    # normalized: {cat, sat} vs {cat, sat, down}
+    assert abs(p - 1.0) < 1e-9 and abs(r - 2 / 3) < 1e-9
+    assert abs(f1 - 0.8) < 1e-9
+    p, r, f1 = F1Metric.compute_all_pairs(["cat"], ["cat"])
+    assert f1 == 1.0
+
+
+def test_msdp_prompt_and_eval(tmp_path):
+    inp = tmp_path / "dialog.jsonl"
+    inp.write_text("\n".join(json.dumps(d) for d in [
+        {"turns": ["hello there", "hi how are you"], "topic": "greeting"},
+        {"turns": ["what is rust"], "topic": "programming",
+         "knowledge": "rust is a language"},
+    ]))
+    out = tmp_path / "knwl.txt"
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29706",
+               RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tasks", "main.py"),
+         "--task", "MSDP-PROMPT", "--prompt-type", "knowledge",
+         "--sample-input-file", str(inp),
+         "--sample-output-file", str(out),
+         "--out-seq-length", "8",
+         "--num-layers", "2", "--hidden-size", "64",
+         "--num-attention-heads", "4", "--seq-length", "128",
+         "--max-position-embeddings", "256", "--micro-batch-size", "1",
+         "--global-batch-size", "1", "--lr", "1e-4",
+         "--hidden-dropout", "0", "--attention-dropout", "0"],
+        capture_output=True, text=True, cwd=REPO, env=env, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = out.read_text().splitlines()
+    assert len(lines) == 2
+    # F1 eval stage on known files
+    guess = tmp_path / "guess.txt"
+    ans = tmp_path / "ans.txt"
+    guess.write_text("the cat sat\n")
+    ans.write_text("a cat sat down\n")
+    r2 = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tasks", "main.py"),
+         "--task", "MSDP-EVAL-F1", "--guess-file", str(guess),
+         "--answer-file", str(ans),
+         "--num-layers", "2", "--hidden-size", "64",
+         "--num-attention-heads", "4", "--seq-length", "32",
+         "--max-position-embeddings", "64", "--micro-batch-size", "1",
+         "--global-batch-size", "1", "--lr", "1e-4"],
+        capture_output=True, text=True, cwd=REPO, env=env, timeout=420)
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    assert "f1: 0.8000" in r2.stdout
