@@ -93,6 +93,9 @@ def _add_model_args(p):
     # MoE
     g.add_argument("--num-experts", type=int, default=None)
     g.add_argument("--moe-router-topk", type=int, default=2)
+    g.add_argument("--moe-router-renormalize", action="store_true",
+                   help="renormalize top-k routing probs to sum to 1 "
+                        "(Mixtral-style)")
     g.add_argument("--moe-router-load-balancing-type", default="aux_loss",
                    choices=["aux_loss", "sinkhorn", "none"])
     g.add_argument("--moe-aux-loss-coeff", type=float, default=0.0)
@@ -458,6 +461,7 @@ def core_transformer_config_from_args(args, config_class=None):
         mtp_num_layers=args.mtp_num_layers,
         mtp_loss_scaling_factor=args.mtp_loss_scaling_factor,
         moe_router_topk=args.moe_router_topk,
+        moe_router_renormalize=args.moe_router_renormalize,
         moe_router_load_balancing_type=args.moe_router_load_balancing_type,
         moe_aux_loss_coeff=args.moe_aux_loss_coeff,
         moe_token_dispatcher_type=args.moe_token_dispatcher_type,
