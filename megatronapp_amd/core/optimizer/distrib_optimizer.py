@@ -225,6 +225,22 @@ class DistributedOptimizer:
                 if not torch.isfinite(g).all():
                     found_inf = True
                 g.mul_(inv)
+            # Each rank only sees its own ZeRO shard (and its own TP/PP
+            # slice of the model), so the skip decision must be agreed
+            # globally or DynamicGradScaler state diverges per rank.
+            if dist.is_initialized():
+                dev = (torch.device("cuda", torch.cuda.current_device())
+                       if torch.cuda.is_available() else torch.device("cpu"))
+                flag = torch.tensor(
+                    [1.0 if found_inf else 0.0], device=dev)
+                dist.all_reduce(
+                    flag, op=dist.ReduceOp.MAX,
+                    group=parallel_state.get_model_parallel_group())
+                dist.all_reduce(
+                    flag, op=dist.ReduceOp.MAX,
+                    group=parallel_state.get_data_parallel_group(
+                        with_context_parallel=True))
+                found_inf = bool(flag.item())
             self.grad_scaler.update(found_inf)
             if found_inf:
                 for i in range(len(self.buffers)):

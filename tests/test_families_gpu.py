@@ -105,9 +105,6 @@ def test_bf16_training_learns_on_gpu():
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="fp16 path first exercised on HW at round end; "
-                          "non-fatal so an issue cannot abort the -x suite")
 def test_fp16_training_step_gpu():
     """fp16 with dynamic loss scaling: one step on MI355X (fp16 params,
     scaled loss, unscaled master update)."""

@@ -1,17 +1,12 @@
-"""On-GPU train-step smokes for the round-1-late model families:
-Retro, LLaVA, ViT tasks, ICT biencoder, int8 PTQ.  All xfail-tolerant:
-these compose GPU-validated layers but have not themselves run on HW
-yet, so a surprise must not abort the driver's -x GPU suite."""
+"""On-GPU train-step smokes for the model families added late in round
+1: Retro, LLaVA, ViT tasks, ICT biencoder, int8 PTQ.  All passed on
+MI355X at the round-1 gate, so they are hard-fail now."""
 import pytest
 import torch
 
 from tests.utils import initialize_model_parallel, destroy
 
-pytestmark = [
-    pytest.mark.gpu,
-    pytest.mark.xfail(strict=False,
-                      reason="first HW exposure for these families"),
-]
+pytestmark = [pytest.mark.gpu]
 
 
 def _seed():

@@ -538,9 +538,6 @@ def test_norm_residual_backward_matches_eager(rows, H):
 
 
 @pytest.mark.gpu
-@pytest.mark.xfail(strict=False,
-                   reason="experimental transposed-S flash kernel; first "
-                          "hardware validation runs at round end")
 @pytest.mark.parametrize("sq,gqa", [(256, 1), (256, 4), (2048, 1)])
 def test_attn_fwd_transposed_matches_ref(sq, gqa):
     ops = _ops()

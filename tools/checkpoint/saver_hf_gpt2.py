@@ -124,7 +124,7 @@ def load_hf_gpt2(path):
     hn = cfg["n_embd"] // nh
     bin_path = os.path.join(path, "pytorch_model.bin")
     if os.path.exists(bin_path):
-        sd = torch.load(bin_path, map_location="cpu", weights_only=False)
+        sd = torch.load(bin_path, map_location="cpu", weights_only=True)
     else:
         from safetensors.torch import load_file
         sd = load_file(os.path.join(path, "model.safetensors"))

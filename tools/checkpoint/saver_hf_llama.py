@@ -62,9 +62,15 @@ def save_hf_llama(full: dict, common: dict, path: str) -> None:
 
     sd = {"model.embed_tokens.weight": wte,
           "model.norm.weight": take("decoder.final_layernorm.weight")}
-    out_key = "model.output_layer.weight"
-    sd["lm_head.weight"] = full[out_key].float() if out_key in full \
-        else wte
+    def take_opt(name):
+        for prefix in ("model.", "model0."):
+            if prefix + name in full:
+                return full[prefix + name].float()
+        return None
+
+    out_w = take_opt("output_layer.weight")
+    tied = out_w is None
+    sd["lm_head.weight"] = wte if tied else out_w
     ffn = None
     for i in layers:
         p = f"decoder.layers.{i}."
@@ -97,7 +103,7 @@ def save_hf_llama(full: dict, common: dict, path: str) -> None:
            "rope_theta": args.get("rotary_base", 10000.0),
            "rms_norm_eps": args.get("norm_epsilon", 1e-5),
            "hidden_act": "silu",
-           "tie_word_embeddings": out_key not in full,
+           "tie_word_embeddings": tied,
            "attention_bias": False, "mlp_bias": False}
     with open(os.path.join(path, "config.json"), "w") as f:
         json.dump(cfg, f, indent=2)
@@ -115,7 +121,7 @@ def load_hf_llama(path):
     rep = nh // ng
     bin_path = os.path.join(path, "pytorch_model.bin")
     if os.path.exists(bin_path):
-        sd = torch.load(bin_path, map_location="cpu", weights_only=False)
+        sd = torch.load(bin_path, map_location="cpu", weights_only=True)
     else:
         from safetensors.torch import load_file
         sd = {}

@@ -50,8 +50,10 @@ def save_hf_mixtral(full: dict, common: dict, path: str) -> None:
 
     sd = {"model.embed_tokens.weight": wte,
           "model.norm.weight": take("decoder.final_layernorm.weight"),
-          "lm_head.weight": full.get("model.output_layer.weight",
-                                     wte).float()}
+          "lm_head.weight": next(
+              (full[pref + "output_layer.weight"].float()
+               for pref in ("model.", "model0.")
+               if pref + "output_layer.weight" in full), wte)}
     n_exp = ffn = None
     for i in layers:
         p = f"decoder.layers.{i}."
@@ -104,7 +106,7 @@ def load_hf_mixtral(path):
     rep = nh // ng
     bin_path = os.path.join(path, "pytorch_model.bin")
     if os.path.exists(bin_path):
-        sd = torch.load(bin_path, map_location="cpu", weights_only=False)
+        sd = torch.load(bin_path, map_location="cpu", weights_only=True)
     else:
         from safetensors.torch import load_file
         sd = {}
