@@ -638,6 +638,340 @@ void launch_attn_fwd_t(const void* q, const void* k, const void* v, void* o,
   HIP_CHECK_LAUNCH();
 }
 
+// ===========================================================================
+// Round-2 forward (attn_fwd2): 8 waves x 32 q-rows (FQBLK2 = 256),
+// mfma_f32_32x32x16_bf16 with swapped QK^T, per-lane online softmax,
+// P^T fragments assembled fully in-register (v_cvt_pk_bf16_f32 +
+// v_permlane32_swap), V consumed from a ROW-MAJOR LDS image through the
+// gfx950 ds_read_b64_tr_b16 hardware transpose read — no scalar
+// transpose writes anywhere (the round-1 loader hotspot), no P LDS
+// round trip (the round-1 softmax/LDS hotspot).
+//
+// Lane maps (verified by tools/probe_tr16.hip on MI355X and locked by
+// tests/test_attn_fwd2_sim.py):
+//   mfma 32x32x16  A[m][k]: lane l -> A[l&31][(l>>5)*8 + j]
+//                  B[k][n]: lane l -> B[(l>>5)*8 + j][l&31]
+//                  C[m][n]: reg r  -> C[(r&3) + 8*(r>>2) + 4*(l>>5)][l&31]
+//   ds_read_b64_tr_b16 (per 16-lane group g): out(l, j) = element (l&3)
+//                  of the 4-element read of lane 16g + 4j + ((l>>2)&3)
+//   permlane32_swap(a, b) -> (a.lo|b.lo, a.hi|b.hi)
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+#define FQBLK2 256     // q rows per workgroup (8 waves x 32)
+#define QBLK2_ROWS 32  // q rows per wave
+#define VRS 160      // V LDS row stride (elems): (a/4)%64 banks of the tr
+                     // reads = 16*((l>>2)&3) + 8*((l>>4)&1) + 2*(l&3) per
+                     // half-wave -> conflict-free
+#define OSTRIDE 136  // epilogue O staging row stride (16B-aligned rows)
+
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+__device__ __forceinline__ short4v tr16_read(const unsigned short* lds_ptr) {
+  short4v v;
+  asm volatile("ds_read_b64_tr_b16 %0, %1"
+               : "=v"(v)
+               : "v"((unsigned)(unsigned long long)(uintptr_t)lds_ptr));
+  return v;
+}
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, unsigned short* __restrict__ o,
+    float* __restrict__ lse, int sq, int sk, int b, int nh, int ng,
+    float scale) {
+  static_assert(D == 128, "attn_fwd2 is specialized for head dim 128");
+  const int qtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / nh;
+  const int h = bh % nh;
+  const int hkv = h / (nh / ng);
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lh = lane >> 5;        // half-wave (MFMA k-half)
+  const int lq = lane & 31;        // this lane's q column / MFMA n
+
+  const long q_ss = (long)b * nh * D;
+  const long k_ss = (long)b * ng * D;
+  const unsigned short* qp = q + ((long)bi * nh + h) * D;
+  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
+  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* k_lds = (unsigned short*)smem;          // KVBLK*D swizzled
+  unsigned short* v_lds = k_lds + KVBLK * D;              // KVBLK*VRS row-major
+  unsigned short* o_stage = v_lds + KVBLK * VRS;          // 8*32*OSTRIDE
+
+  const int q0 = qtile * FQBLK2 + wid * QBLK2_ROWS;
+  const int qrow = q0 + lq;        // this lane's q row
+
+  // ---- Q B-fragments: qfrag[f] = Q[qrow][16f + 8*lh + j]
+  constexpr int NF = D / 16;
+  bf16x8 qfrag[NF];
+  {
+    const unsigned short* src = qp + (long)qrow * q_ss;
+#pragma unroll
+    for (int f = 0; f < NF; ++f)
+      qfrag[f] = (qrow < sq) ? *(const bf16x8*)(src + 16 * f + 8 * lh)
+                             : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+
+  float m_run = -INFINITY;
+  float l_run = 0.f;
+  constexpr int NDSUB = D / 32;
+  f32x16 oacc[NDSUB];
+#pragma unroll
+  for (int s = 0; s < NDSUB; ++s)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[s][r] = 0.f;
+
+  const int q_hi = qtile * FQBLK2 + FQBLK2 - 1;
+  int kv_end = sk;
+  if (CAUSAL) kv_end = min(sk, q_hi + 1 + (sk - sq));
+  const int n_kv_tiles = (kv_end + KVBLK - 1) / KVBLK;
+  // this wave's own causal end (waves past it idle through barriers)
+  int wave_kv_end = sk;
+  if (CAUSAL) wave_kv_end = min(sk, q0 + QBLK2_ROWS + (sk - sq));
+
+  // T14 async staging split
+  constexpr int PIECES = KVBLK * D / 8 / FWD_BLOCK;  // 2 at D=128
+  bf16x8 kreg[PIECES], vreg[PIECES];
+  auto stage_load = [&](int t) {
+    const int kv0 = t * KVBLK;
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int grow = kv0 + row;
+      kreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      *(bf16x8*)((char*)(k_lds + (long)row * D) + swz(row, col * 2)) =
+          kreg[pc];
+      *(bf16x8*)(v_lds + (long)row * VRS + col) = vreg[pc];
+    }
+  };
+
+  // per-lane tr-read base for V fragments:
+  //   row = 8*lh + ((l>>2)&3), col = 16*((l>>4)&1) + 4*(l&3)
+  const unsigned short* v_tr_base =
+      v_lds + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
+      16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+
+  for (int t = 0; t < n_kv_tiles; ++t) {
+    const int kv0 = t * KVBLK;
+    if (t + 1 < n_kv_tiles) stage_load(t + 1);
+
+    if (kv0 < wave_kv_end) {
+      // ---- S^T = K Q^T per 32-kv sub-block
+      f32x16 st[2];
+#pragma unroll
+      for (int ksub = 0; ksub < 2; ++ksub) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) st[ksub][r] = 0.f;
+        const int krow = 32 * ksub + lq;
+#pragma unroll
+        for (int f = 0; f < NF; ++f) {
+          bf16x8 afrag = *(const bf16x8*)(
+              (char*)(k_lds + (long)krow * D) +
+              swz(krow, (16 * f + 8 * lh) * 2));
+          st[ksub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              afrag, qfrag[f], st[ksub], 0, 0, 0);
+        }
+      }
+
+      // ---- mask + per-lane online softmax (lane owns q = qrow)
+      float mtile = -INFINITY;
+      const bool full_tile =
+          (qrow < sq) &&
+          (!CAUSAL ? (kv0 + KVBLK <= sk)
+                   : (kv0 + KVBLK - 1 <= q0 + (sk - sq)));
+      if (full_tile) {
+#pragma unroll
+        for (int ksub = 0; ksub < 2; ++ksub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            float s = st[ksub][r] * scale;
+            st[ksub][r] = s;
+            mtile = fmaxf(mtile, s);
+          }
+      } else {
+#pragma unroll
+        for (int ksub = 0; ksub < 2; ++ksub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kv = kv0 + 32 * ksub + (r & 3) + 8 * (r >> 2) + 4 * lh;
+            float s = st[ksub][r] * scale;
+            bool valid = (kv < sk) && (qrow < sq);
+            if (CAUSAL) valid = valid && (kv <= qrow + (sk - sq));
+            s = valid ? s : -INFINITY;
+            st[ksub][r] = s;
+            mtile = fmaxf(mtile, s);
+          }
+      }
+      mtile = fmaxf(mtile, __shfl_xor(mtile, 32, WAVE));
+
+      const float mn = fmaxf(m_run, mtile);
+      const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - mn);
+      m_run = mn;
+      float lt = 0.f;
+#pragma unroll
+      for (int ksub = 0; ksub < 2; ++ksub)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float s = st[ksub][r];
+          float p = (s == -INFINITY) ? 0.f : __expf(s - m_run);
+          st[ksub][r] = p;
+          lt += p;
+        }
+      lt += __shfl_xor(lt, 32, WAVE);
+      l_run = l_run * alpha + lt;
+#pragma unroll
+      for (int s = 0; s < NDSUB; ++s)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[s][r] *= alpha;
+
+      // ---- P^T fragments in-register: cvt_pk pairs + permlane32_swap.
+      // pfrag[ks] holds B[k = 16ks + 8*lh + jj][n = lq] as 4 u32 words.
+      unsigned pw[4][4];  // [ks][word]
+#pragma unroll
+      for (int ksub = 0; ksub < 2; ++ksub) {
+        unsigned u0[4], u1[4];
+#pragma unroll
+        for (int i4 = 0; i4 < 4; ++i4) {
+          u0[i4] = cvt_pk_bf16(st[ksub][4 * i4], st[ksub][4 * i4 + 1]);
+          u1[i4] = cvt_pk_bf16(st[ksub][4 * i4 + 2], st[ksub][4 * i4 + 3]);
+        }
+#pragma unroll
+        for (int K = 0; K < 2; ++K) {
+          auto s0 = __builtin_amdgcn_permlane32_swap(
+              (int)u0[2 * K], (int)u0[2 * K + 1], false, false);
+          auto s1 = __builtin_amdgcn_permlane32_swap(
+              (int)u1[2 * K], (int)u1[2 * K + 1], false, false);
+          const int ks = 2 * ksub + K;
+          pw[ks][0] = (unsigned)s0[0];
+          pw[ks][1] = (unsigned)s1[0];
+          pw[ks][2] = (unsigned)s0[1];
+          pw[ks][3] = (unsigned)s1[1];
+        }
+      }
+
+      // ---- O^T += V^T P^T : A fragments via hardware transpose reads
+#pragma unroll
+      for (int dsub = 0; dsub < NDSUB; ++dsub) {
+        short4v vr[8];
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+#pragma unroll
+          for (int rr = 0; rr < 2; ++rr)
+            vr[2 * ks + rr] = tr16_read(
+                v_tr_base + (long)(16 * ks + 4 * rr) * VRS + 32 * dsub);
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          bf16x8 vfrag, pfrag;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            vfrag[j] = vr[2 * ks][j];
+            vfrag[4 + j] = vr[2 * ks + 1][j];
+          }
+#pragma unroll
+          for (int w = 0; w < 4; ++w) {
+            pfrag[2 * w] = (short)(pw[ks][w] & 0xffff);
+            pfrag[2 * w + 1] = (short)(pw[ks][w] >> 16);
+          }
+          oacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              vfrag, pfrag, oacc[dsub], 0, 0, 0);
+        }
+      }
+    }
+
+    __syncthreads();
+    if (t + 1 < n_kv_tiles) {
+      stage_write();
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: normalize, transpose O^T -> O rows via LDS, store
+  const float denom = l_run > 0.f ? 1.f / l_run : 0.f;
+  if (lh == 0 && qrow < sq)
+    lse[((long)bi * nh + h) * sq + qrow] =
+        m_run + __logf(fmaxf(l_run, 1e-30f));
+  unsigned short* ows = o_stage + (long)wid * QBLK2_ROWS * OSTRIDE;
+#pragma unroll
+  for (int dsub = 0; dsub < NDSUB; ++dsub) {
+#pragma unroll
+    for (int r = 0; r < 16; r += 2) {
+      const int d_ = 32 * dsub + (r & 3) + 8 * (r >> 2) + 4 * lh;
+      *(unsigned*)(ows + (long)lq * OSTRIDE + d_) =
+          cvt_pk_bf16(oacc[dsub][r] * denom, oacc[dsub][r + 1] * denom);
+    }
+  }
+  __syncthreads();
+  {
+    constexpr int pieces = QBLK2_ROWS * D / 8 / WAVE;  // 8 at D=128
+#pragma unroll
+    for (int pc = 0; pc < pieces; ++pc) {
+      const int idx = (pc * WAVE + lane) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int qq = q0 + row;
+      if (qq < sq)
+        *(bf16x8*)(o + ((long)qq * b * nh + (long)bi * nh + h) * D + col) =
+            *(const bf16x8*)(ows + (long)row * OSTRIDE + col);
+    }
+  }
+}
+
+void launch_attn_fwd2(const void* q, const void* k, const void* v, void* o,
+                      float* lse, int sq, int sk, int b, int nh, int ng,
+                      int d, float scale, bool causal, hipStream_t stream) {
+  if (sq % FQBLK2 != 0 || sk % KVBLK != 0)
+    throw std::runtime_error(
+        "attn_fwd2: sq must be a multiple of 256, sk of 64");
+  if (d != 128) throw std::runtime_error("attn_fwd2: d must be 128");
+  dim3 grid(sq / FQBLK2, b * nh);
+  dim3 block(FWD_BLOCK);
+  const size_t lds =
+      (size_t)(KVBLK * d + KVBLK * VRS + 8 * 32 * OSTRIDE) *
+      sizeof(unsigned short);
+  if (causal)
+    hipLaunchKernelGGL((attn_fwd2_kernel<128, true>), grid, block, lds,
+                       stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+  else
+    hipLaunchKernelGGL((attn_fwd2_kernel<128, false>), grid, block, lds,
+                       stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+  HIP_CHECK_LAUNCH();
+}
+
 // ablation entry (perf diagnosis only; outputs wrong for level>0)
 void launch_attn_fwd_ablate(const void* q, const void* k, const void* v,
                             void* o, float* lse, int sq, int sk, int b,

@@ -56,6 +56,8 @@ torch::Tensor gemm_nn(torch::Tensor, torch::Tensor);
 void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
 void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
+void launch_attn_fwd2(const void*, const void*, const void*, void*, float*,
+                      int, int, int, int, int, int, float, bool, hipStream_t);
 void launch_attn_fwd_t(const void*, const void*, const void*, void*, float*,
                        int, int, int, int, int, int, float, bool,
                        hipStream_t);
@@ -377,6 +379,23 @@ std::vector<torch::Tensor> attn_fwd_t(torch::Tensor q, torch::Tensor k,
   return {o, lse};
 }
 
+std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
+                                     torch::Tensor v, double scale,
+                                     bool causal) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(v, "v");
+  const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
+            d = (int)q.size(3);
+  const int sk = (int)k.size(0), ng = (int)k.size(2);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({b, nh, sq}, q.options().dtype(torch::kFloat32));
+  launch_attn_fwd2(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                   lse.data_ptr<float>(), sq, sk, b, nh, ng, d, (float)scale,
+                   causal, cur_stream());
+  return {o, lse};
+}
+
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
@@ -531,6 +550,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adamw_flat", &adamw_flat);
   mod.def("selective_scan_fwd", &selective_scan_fwd);
   mod.def("attn_fwd_t", &attn_fwd_t);
+  mod.def("attn_fwd2", &attn_fwd2);
   mod.def("ce_rowmax", &ce_rowmax);
   mod.def("gemm_nt", &gemm_nt);
   mod.def("gemm_nn", &gemm_nn);

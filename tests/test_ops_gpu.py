@@ -555,3 +555,27 @@ def test_attn_fwd_transposed_matches_ref(sq, gqa):
     lse_got = lse.view(b, nh, sq)
     assert torch.allclose(lse_got, lse_ref.float(), atol=1e-3), \
         (lse_got - lse_ref.float()).abs().max()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("sq,gqa,causal", [(256, 1, True), (256, 4, True),
+                                           (2048, 1, True), (512, 1, False),
+                                           (2048, 4, True)])
+def test_attn_fwd2_matches_ref(sq, gqa, causal):
+    """Round-2 32x32-MFMA forward (swapped QK^T, tr16 V reads, in-register
+    P^T) vs the fp32 reference."""
+    ops = _ops()
+    torch.manual_seed(29)
+    b, nh, d = 2, 8, 128
+    ng = nh // gqa
+    q = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(sq, b, ng, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn_like(k)
+    scale = d ** -0.5
+    o_ref, lse_ref = _sdpa_ref(q, k, v, scale, causal)
+    o, lse = ops.attn_fwd2(q, k, v, scale, causal)
+    assert torch.allclose(o.float(), o_ref.float(), atol=2e-2, rtol=2e-2), \
+        (o.float() - o_ref.float()).abs().max()
+    lse_got = lse.view(b, nh, sq)
+    assert torch.allclose(lse_got, lse_ref.float(), atol=1e-3), \
+        (lse_got - lse_ref.float()).abs().max()
