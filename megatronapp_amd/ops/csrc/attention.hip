@@ -889,7 +889,13 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
             vr[2 * ks + rr] = tr16_read(
                 v_tr_base + (long)(16 * ks + 4 * rr) * VRS + 32 * dsub);
         }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        // The waitcnt must DATA-DEPEND on the tr-read outputs ("+v" ties)
+        // or the compiler hoists the register extracts above it and reads
+        // in-flight garbage (observed: deterministic 7/8-mass outputs).
+        asm volatile("s_waitcnt lgkmcnt(0)"
+                     : "+v"(vr[0]), "+v"(vr[1]), "+v"(vr[2]), "+v"(vr[3]),
+                       "+v"(vr[4]), "+v"(vr[5]), "+v"(vr[6]), "+v"(vr[7])
+                     :: "memory");
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
           bf16x8 vfrag, pfrag;
