@@ -978,6 +978,522 @@ void launch_attn_fwd2(const void* q, const void* k, const void* v, void* o,
   HIP_CHECK_LAUNCH();
 }
 
+// ===========================================================================
+// Round-2 backward: dq2 (grid over q tiles) + dkv2 (grid over kv tiles),
+// both on the fwd2 toolbox — swapped MFMAs put lse/Drow on lane-local
+// rows, dS/P are re-packed to fragments with the cvt_pk+permlane32_swap
+// recipe (A and B fragment maps are mutual transposes, so the identical
+// assembly serves both kernels), and the "transposed" operands (K^T for
+// dQ^T, Q^T/dO^T for dK/dV) come from row-major LDS images via
+// ds_read_b64_tr_b16.  Index math locked by tests/test_attn_bwd2_sim.py.
+
+// assemble the two 16-k fragments of one 32-row C tile (16 regs/lane)
+// into packed bf16 pairs; out[K][w] = word w of fragment K.
+#define SWAP_ASSEMBLE(VALS, OUT)                                            \
+  do {                                                                      \
+    unsigned u0_[4], u1_[4];                                                \
+    _Pragma("unroll") for (int i4_ = 0; i4_ < 4; ++i4_) {                   \
+      u0_[i4_] = cvt_pk_bf16((VALS)[4 * i4_], (VALS)[4 * i4_ + 1]);         \
+      u1_[i4_] = cvt_pk_bf16((VALS)[4 * i4_ + 2], (VALS)[4 * i4_ + 3]);     \
+    }                                                                       \
+    _Pragma("unroll") for (int K_ = 0; K_ < 2; ++K_) {                      \
+      auto s0_ = __builtin_amdgcn_permlane32_swap(                          \
+          (int)u0_[2 * K_], (int)u0_[2 * K_ + 1], false, false);            \
+      auto s1_ = __builtin_amdgcn_permlane32_swap(                          \
+          (int)u1_[2 * K_], (int)u1_[2 * K_ + 1], false, false);            \
+      (OUT)[K_][0] = (unsigned)s0_[0];                                      \
+      (OUT)[K_][1] = (unsigned)s1_[0];                                      \
+      (OUT)[K_][2] = (unsigned)s0_[1];                                      \
+      (OUT)[K_][3] = (unsigned)s1_[1];                                      \
+    }                                                                       \
+  } while (0)
+
+__device__ __forceinline__ bf16x8 frag_from_words(const unsigned* w) {
+  bf16x8 f;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    f[2 * i] = (short)(w[i] & 0xffff);
+    f[2 * i + 1] = (short)(w[i] >> 16);
+  }
+  return f;
+}
+
+// ------------------------------------------------------------------ dq2
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    unsigned short* __restrict__ dq, int sq, int sk, int b, int nh, int ng,
+    float scale) {
+  static_assert(D == 128, "dq2 is specialized for head dim 128");
+  const int qtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / nh;
+  const int h = bh % nh;
+  const int hkv = h / (nh / ng);
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lh = lane >> 5;
+  const int lq = lane & 31;
+
+  const long q_ss = (long)b * nh * D;
+  const long k_ss = (long)b * ng * D;
+  const unsigned short* qp = q + ((long)bi * nh + h) * D;
+  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
+  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+  const unsigned short* dop = dout + ((long)bi * nh + h) * D;
+  const float* lse_row = lse + ((long)bi * nh + h) * sq;
+  const long dr_ss = (long)b * nh;
+  const float* dr_base = drow + (long)bi * nh + h;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* k_swz = (unsigned short*)smem;          // KVBLK*D
+  unsigned short* v_swz = k_swz + KVBLK * D;              // KVBLK*D
+  unsigned short* k_rm = v_swz + KVBLK * D;               // KVBLK*VRS
+  unsigned short* o_stage = k_rm + KVBLK * VRS;           // 8*32*OSTRIDE
+
+  const int q0 = qtile * FQBLK2 + wid * QBLK2_ROWS;
+  const int qrow = q0 + lq;
+
+  constexpr int NF = D / 16;
+  bf16x8 qfrag[NF], dofrag[NF];
+  float lse_l, dr_l;
+  {
+    const unsigned short* src = qp + (long)qrow * q_ss;
+    const unsigned short* dsrc = dop + (long)qrow * q_ss;
+#pragma unroll
+    for (int f = 0; f < NF; ++f) {
+      qfrag[f] = *(const bf16x8*)(src + 16 * f + 8 * lh);
+      dofrag[f] = *(const bf16x8*)(dsrc + 16 * f + 8 * lh);
+    }
+    lse_l = lse_row[qrow];
+    dr_l = dr_base[(long)qrow * dr_ss];
+  }
+
+  constexpr int NDSUB = D / 32;
+  f32x16 dqacc[NDSUB];
+#pragma unroll
+  for (int s = 0; s < NDSUB; ++s)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dqacc[s][r] = 0.f;
+
+  const int q_hi = qtile * FQBLK2 + FQBLK2 - 1;
+  int kv_end = sk;
+  if (CAUSAL) kv_end = min(sk, q_hi + 1 + (sk - sq));
+  const int n_kv_tiles = (kv_end + KVBLK - 1) / KVBLK;
+  int wave_kv_end = sk;
+  if (CAUSAL) wave_kv_end = min(sk, q0 + QBLK2_ROWS + (sk - sq));
+
+  constexpr int PIECES = KVBLK * D / 8 / FWD_BLOCK;
+  bf16x8 kreg[PIECES], vreg[PIECES];
+  auto stage_load = [&](int t) {
+    const int kv0 = t * KVBLK;
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int grow = kv0 + row;
+      kreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vreg[pc] = (grow < sk)
+                     ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      *(bf16x8*)((char*)(k_swz + (long)row * D) + swz(row, col * 2)) =
+          kreg[pc];
+      *(bf16x8*)((char*)(v_swz + (long)row * D) + swz(row, col * 2)) =
+          vreg[pc];
+      *(bf16x8*)(k_rm + (long)row * VRS + col) = kreg[pc];
+    }
+  };
+
+  const unsigned short* k_tr_base =
+      k_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
+      16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+
+  for (int t = 0; t < n_kv_tiles; ++t) {
+    const int kv0 = t * KVBLK;
+    if (t + 1 < n_kv_tiles) stage_load(t + 1);
+
+    if (kv0 < wave_kv_end) {
+      unsigned dsw[4][4];  // packed dS^T fragments [ks][word]
+#pragma unroll
+      for (int ksub = 0; ksub < 2; ++ksub) {
+        f32x16 st, dpt;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          st[r] = 0.f;
+          dpt[r] = 0.f;
+        }
+        const int krow = 32 * ksub + lq;
+#pragma unroll
+        for (int f = 0; f < NF; ++f) {
+          bf16x8 ka = *(const bf16x8*)(
+              (char*)(k_swz + (long)krow * D) +
+              swz(krow, (16 * f + 8 * lh) * 2));
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[f], st,
+                                                       0, 0, 0);
+          bf16x8 va = *(const bf16x8*)(
+              (char*)(v_swz + (long)krow * D) +
+              swz(krow, (16 * f + 8 * lh) * 2));
+          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[f], dpt,
+                                                        0, 0, 0);
+        }
+        // dS^T = P^T * (dP^T - Drow) * scale, all per-lane
+        float dsv[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kv = kv0 + 32 * ksub + (r & 3) + 8 * (r >> 2) + 4 * lh;
+          bool valid = kv < sk;
+          if (CAUSAL) valid = valid && (kv <= qrow + (sk - sq));
+          float p = valid ? __expf(st[r] * scale - lse_l) : 0.f;
+          dsv[r] = p * (dpt[r] - dr_l) * scale;
+        }
+        unsigned out2[2][4];
+        SWAP_ASSEMBLE(dsv, out2);
+#pragma unroll
+        for (int K = 0; K < 2; ++K)
+#pragma unroll
+          for (int w = 0; w < 4; ++w) dsw[2 * ksub + K][w] = out2[K][w];
+      }
+
+      // dQ^T += K^T dS^T
+#pragma unroll
+      for (int dsub = 0; dsub < NDSUB; ++dsub) {
+        short4v kr[8];
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks)
+#pragma unroll
+          for (int rr = 0; rr < 2; ++rr)
+            kr[2 * ks + rr] = tr16_read(
+                k_tr_base + (long)(16 * ks + 4 * rr) * VRS + 32 * dsub);
+        asm volatile("s_waitcnt lgkmcnt(0)"
+                     : "+v"(kr[0]), "+v"(kr[1]), "+v"(kr[2]), "+v"(kr[3]),
+                       "+v"(kr[4]), "+v"(kr[5]), "+v"(kr[6]), "+v"(kr[7])
+                     :: "memory");
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          bf16x8 kfrag;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            kfrag[j] = kr[2 * ks][j];
+            kfrag[4 + j] = kr[2 * ks + 1][j];
+          }
+          dqacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kfrag, frag_from_words(dsw[ks]), dqacc[dsub], 0, 0, 0);
+        }
+      }
+    }
+
+    __syncthreads();
+    if (t + 1 < n_kv_tiles) {
+      stage_write();
+      __syncthreads();
+    }
+  }
+
+  // epilogue: dQ^T -> dQ rows via LDS
+  unsigned short* ows = o_stage + (long)wid * QBLK2_ROWS * OSTRIDE;
+#pragma unroll
+  for (int dsub = 0; dsub < NDSUB; ++dsub)
+#pragma unroll
+    for (int r = 0; r < 16; r += 2) {
+      const int d_ = 32 * dsub + (r & 3) + 8 * (r >> 2) + 4 * lh;
+      *(unsigned*)(ows + (long)lq * OSTRIDE + d_) =
+          cvt_pk_bf16(dqacc[dsub][r], dqacc[dsub][r + 1]);
+    }
+  __syncthreads();
+  {
+    constexpr int pieces = QBLK2_ROWS * D / 8 / WAVE;
+#pragma unroll
+    for (int pc = 0; pc < pieces; ++pc) {
+      const int idx = (pc * WAVE + lane) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int qq = q0 + row;
+      *(bf16x8*)(dq + ((long)qq * b * nh + (long)bi * nh + h) * D + col) =
+          *(const bf16x8*)(ows + (long)row * OSTRIDE + col);
+    }
+  }
+}
+
+// ------------------------------------------------------------------ dkv2
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    unsigned short* __restrict__ dk, unsigned short* __restrict__ dv, int sq,
+    int sk, int b, int nh, int ng, float scale) {
+  static_assert(D == 128, "dkv2 is specialized for head dim 128");
+  const int kvtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / ng;
+  const int hkv = bh % ng;
+  const int group = nh / ng;
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lh = lane >> 5;
+  const int ln = lane & 31;      // this lane's kv column (MFMA n)
+
+  const long q_ss = (long)b * nh * D;
+  const long k_ss = (long)b * ng * D;
+  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
+  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* q_swz = (unsigned short*)smem;           // 32*D
+  unsigned short* do_swz = q_swz + 32 * D;                 // 32*D
+  unsigned short* q_rm = do_swz + 32 * D;                  // 32*VRS
+  unsigned short* do_rm = q_rm + 32 * VRS;                 // 32*VRS
+  float* lse_lds = (float*)(do_rm + 32 * VRS);             // 32
+  float* dr_lds = lse_lds + 32;                            // 32
+
+  const int kv0w = kvtile * FQBLK2 + wid * QBLK2_ROWS;
+  const int kvcol = kv0w + ln;   // this lane's kv row in K/V
+
+  // K/V B-fragments are re-read from GLOBAL each q tile (the block's K/V
+  // rows are 128 KB and stay L2-resident across the whole q loop):
+  // keeping them in registers cost 64 VGPRs and pushed the kernel into
+  // 256 B/lane scratch spills, which is worse than L2 reads.
+  constexpr int NF = D / 16;
+  const unsigned short* ksrc = kp + (long)min(kvcol, sk - 1) * k_ss + 8 * lh;
+  const unsigned short* vsrc = vp + (long)min(kvcol, sk - 1) * k_ss + 8 * lh;
+  const bool kv_in_range = kvcol < sk;
+
+  constexpr int NDSUB = D / 32;
+  f32x16 dkacc[NDSUB], dvacc[NDSUB];
+#pragma unroll
+  for (int s = 0; s < NDSUB; ++s)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dkacc[s][r] = 0.f;
+      dvacc[s][r] = 0.f;
+    }
+
+  const int kv_lo = kvtile * FQBLK2;
+  int q_start = 0;
+  if (CAUSAL)
+    q_start = max(0, (kv_lo - (sk - sq)) / QBLK2_ROWS * QBLK2_ROWS);
+
+  const unsigned short* q_tr_base =
+      q_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
+      16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+  const unsigned short* do_tr_base =
+      do_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
+      16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+
+  for (int hq = hkv * group; hq < (hkv + 1) * group; ++hq) {
+    const unsigned short* qp = q + ((long)bi * nh + hq) * D;
+    const unsigned short* dop = dout + ((long)bi * nh + hq) * D;
+    const float* lse_row = lse + ((long)bi * nh + hq) * sq;
+    const long dr_ss = (long)b * nh;
+    const float* dr_base = drow + (long)bi * nh + hq;
+
+    for (int qt = q_start; qt < sq; qt += QBLK2_ROWS) {
+      // stage Q/dO (swizzled rows + row-major for tr reads) + lse/drow
+      {
+        constexpr int pieces = 32 * D / 8 / FWD_BLOCK;  // 1 at D=128
+#pragma unroll
+        for (int pc = 0; pc < pieces; ++pc) {
+          const int idx = (pc * FWD_BLOCK + tid) * 8;
+          const int row = idx / D;
+          const int col = idx % D;
+          const int grow = qt + row;
+          bf16x8 q8 = (grow < sq)
+                          ? *(const bf16x8*)(qp + (long)grow * q_ss + col)
+                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+          bf16x8 d8 = (grow < sq)
+                          ? *(const bf16x8*)(dop + (long)grow * q_ss + col)
+                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+          *(bf16x8*)((char*)(q_swz + (long)row * D) + swz(row, col * 2)) = q8;
+          *(bf16x8*)((char*)(do_swz + (long)row * D) + swz(row, col * 2)) = d8;
+          *(bf16x8*)(q_rm + (long)row * VRS + col) = q8;
+          *(bf16x8*)(do_rm + (long)row * VRS + col) = d8;
+        }
+        if (tid < 32) {
+          const int grow = qt + tid;
+          lse_lds[tid] = (grow < sq) ? lse_row[grow] : 0.f;
+          dr_lds[tid] = (grow < sq) ? dr_base[(long)grow * dr_ss] : 0.f;
+        }
+      }
+      __syncthreads();
+
+      const bool active = !CAUSAL || (kv0w <= qt + QBLK2_ROWS - 1 + (sk - sq));
+      if (active) {
+        // S and dP in C[m=q][n=kv]
+        f32x16 sc, dpc;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          sc[r] = 0.f;
+          dpc[r] = 0.f;
+        }
+        const int qrow_a = lane & 31;  // A-frag row within the q tile
+#pragma unroll
+        for (int f = 0; f < NF; ++f) {
+          const bf16x8 z8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+          bf16x8 kf = kv_in_range ? *(const bf16x8*)(ksrc + 16 * f) : z8;
+          bf16x8 vf = kv_in_range ? *(const bf16x8*)(vsrc + 16 * f) : z8;
+          bf16x8 qa = *(const bf16x8*)(
+              (char*)(q_swz + (long)qrow_a * D) +
+              swz(qrow_a, (16 * f + 8 * lh) * 2));
+          sc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf, sc, 0, 0, 0);
+          bf16x8 da = *(const bf16x8*)(
+              (char*)(do_swz + (long)qrow_a * D) +
+              swz(qrow_a, (16 * f + 8 * lh) * 2));
+          dpc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf, dpc,
+                                                        0, 0, 0);
+        }
+        // P and dS per lane (lse/drow indexed by the reg's q row)
+        float pv[16], dsv[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qr = (r & 3) + 8 * (r >> 2) + 4 * lh;  // within tile
+          const int qrow = qt + qr;
+          bool valid = (qrow < sq) && (kvcol < sk);
+          if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
+          float p = valid ? __expf(sc[r] * scale - lse_lds[qr]) : 0.f;
+          pv[r] = p;
+          dsv[r] = p * (dpc[r] - dr_lds[qr]) * scale;
+        }
+        unsigned pwds[2][4], dswds[2][4];
+        SWAP_ASSEMBLE(pv, pwds);
+        SWAP_ASSEMBLE(dsv, dswds);
+
+        // dV += P^T dO ; dK += dS^T Q  (tr fragments of dO / Q)
+#pragma unroll
+        for (int dsub = 0; dsub < NDSUB; ++dsub) {
+          short4v r8[4];
+#pragma unroll
+          for (int K = 0; K < 2; ++K)
+#pragma unroll
+            for (int rr = 0; rr < 2; ++rr)
+              r8[2 * K + rr] = tr16_read(
+                  do_tr_base + (long)(16 * K + 4 * rr) * VRS + 32 * dsub);
+          asm volatile("s_waitcnt lgkmcnt(0)"
+                       : "+v"(r8[0]), "+v"(r8[1]), "+v"(r8[2]), "+v"(r8[3])
+                       :: "memory");
+#pragma unroll
+          for (int K = 0; K < 2; ++K) {
+            bf16x8 df;
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              df[j] = r8[2 * K][j];
+              df[4 + j] = r8[2 * K + 1][j];
+            }
+            dvacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                frag_from_words(pwds[K]), df, dvacc[dsub], 0, 0, 0);
+          }
+#pragma unroll
+          for (int K = 0; K < 2; ++K)
+#pragma unroll
+            for (int rr = 0; rr < 2; ++rr)
+              r8[2 * K + rr] = tr16_read(
+                  q_tr_base + (long)(16 * K + 4 * rr) * VRS + 32 * dsub);
+          asm volatile("s_waitcnt lgkmcnt(0)"
+                       : "+v"(r8[0]), "+v"(r8[1]), "+v"(r8[2]), "+v"(r8[3])
+                       :: "memory");
+#pragma unroll
+          for (int K = 0; K < 2; ++K) {
+            bf16x8 qf;
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              qf[j] = r8[2 * K][j];
+              qf[4 + j] = r8[2 * K + 1][j];
+            }
+            dkacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                frag_from_words(dswds[K]), qf, dkacc[dsub], 0, 0, 0);
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // epilogue: direct coalesced b16 stores (lanes hold consecutive d)
+#pragma unroll
+  for (int dsub = 0; dsub < NDSUB; ++dsub)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kvrow = kv0w + (r & 3) + 8 * (r >> 2) + 4 * lh;
+      const int d_ = 32 * dsub + ln;
+      if (kvrow < sk) {
+        dk[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + d_] =
+            f2bf(dkacc[dsub][r]);
+        dv[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + d_] =
+            f2bf(dvacc[dsub][r]);
+      }
+    }
+}
+
+__global__ void attn_bwd_pre_kernel(const unsigned short* __restrict__ do_,
+                                    const unsigned short* __restrict__ o,
+                                    float* __restrict__ drow, long rows,
+                                    int d);
+
+void launch_attn_bwd2(const void* dout, const void* q, const void* k,
+                      const void* v, const void* o, const float* lse,
+                      float* drow, void* dq, void* dk, void* dv, int sq,
+                      int sk, int b, int nh, int ng, int d, float scale,
+                      bool causal, hipStream_t stream) {
+  if (sq % FQBLK2 != 0 || sk % FQBLK2 != 0 || d != 128)
+    throw std::runtime_error("attn_bwd2: sq/sk must be multiples of 256, d=128");
+  {
+    const long rows = (long)sq * b * nh;
+    const int waves_per_block = ATT_BLOCK / WAVE;
+    const long blocks = (rows + waves_per_block - 1) / waves_per_block;
+    hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3((unsigned)blocks),
+                       dim3(ATT_BLOCK), 0, stream,
+                       (const unsigned short*)dout, (const unsigned short*)o,
+                       drow, rows, d);
+    HIP_CHECK_LAUNCH();
+  }
+  const size_t lds_dq = (size_t)(2 * KVBLK * 128 + KVBLK * VRS +
+                                 8 * 32 * OSTRIDE) * sizeof(unsigned short);
+  const size_t lds_dkv = (size_t)(2 * 32 * 128 + 2 * 32 * VRS) *
+                             sizeof(unsigned short) +
+                         64 * sizeof(float);
+#define ATT_BWD2_LAUNCH(CC)                                                   \
+  do {                                                                        \
+    hipLaunchKernelGGL((attn_bwd_dq2_kernel<128, CC>),                        \
+                       dim3(sq / FQBLK2, b * nh), dim3(FWD_BLOCK), lds_dq,    \
+                       stream, (const unsigned short*)q,                      \
+                       (const unsigned short*)k, (const unsigned short*)v,    \
+                       (const unsigned short*)dout, lse, drow,                \
+                       (unsigned short*)dq, sq, sk, b, nh, ng, scale);        \
+    HIP_CHECK_LAUNCH();                                                       \
+    hipLaunchKernelGGL((attn_bwd_dkv2_kernel<128, CC>),                       \
+                       dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dkv,   \
+                       stream, (const unsigned short*)q,                      \
+                       (const unsigned short*)k, (const unsigned short*)v,    \
+                       (const unsigned short*)dout, lse, drow,                \
+                       (unsigned short*)dk, (unsigned short*)dv, sq, sk, b,   \
+                       nh, ng, scale);                                        \
+    HIP_CHECK_LAUNCH();                                                       \
+  } while (0)
+  if (causal) ATT_BWD2_LAUNCH(true);
+  else ATT_BWD2_LAUNCH(false);
+#undef ATT_BWD2_LAUNCH
+}
+
 // ablation entry (perf diagnosis only; outputs wrong for level>0)
 void launch_attn_fwd_ablate(const void* q, const void* k, const void* v,
                             void* o, float* lse, int sq, int sk, int b,

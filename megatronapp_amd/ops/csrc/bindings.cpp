@@ -58,6 +58,9 @@ void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
 void launch_attn_fwd2(const void*, const void*, const void*, void*, float*,
                       int, int, int, int, int, int, float, bool, hipStream_t);
+void launch_attn_bwd2(const void*, const void*, const void*, const void*,
+                      const void*, const float*, float*, void*, void*, void*,
+                      int, int, int, int, int, int, float, bool, hipStream_t);
 void launch_attn_fwd_t(const void*, const void*, const void*, void*, float*,
                        int, int, int, int, int, int, float, bool,
                        hipStream_t);
@@ -356,9 +359,14 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   const int sk = (int)k.size(0), ng = (int)k.size(2);
   auto o = torch::empty_like(q);
   auto lse = torch::empty({b, nh, sq}, q.options().dtype(torch::kFloat32));
-  launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                  lse.data_ptr<float>(), sq, sk, b, nh, ng, d, (float)scale,
-                  causal, cur_stream());
+  if (d == 128 && sq % 256 == 0 && sk % 64 == 0)
+    launch_attn_fwd2(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                     lse.data_ptr<float>(), sq, sk, b, nh, ng, d,
+                     (float)scale, causal, cur_stream());
+  else
+    launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                    lse.data_ptr<float>(), sq, sk, b, nh, ng, d, (float)scale,
+                    causal, cur_stream());
   return {o, lse};
 }
 
@@ -401,6 +409,34 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor o, torch::Tensor lse,
                                     double scale, bool causal) {
   check_bf16(dout, "dout");
+  const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
+            d = (int)q.size(3);
+  const int sk = (int)k.size(0), ng = (int)k.size(2);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto drow = torch::empty({(long)sq * b * nh},
+                           q.options().dtype(torch::kFloat32));
+  if (d == 128 && sq % 256 == 0 && sk % 256 == 0)
+    launch_attn_bwd2(dout.data_ptr(), q.data_ptr(), k.data_ptr(),
+                     v.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),
+                     drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                     dv.data_ptr(), sq, sk, b, nh, ng, d, (float)scale,
+                     causal, cur_stream());
+  else
+    launch_attn_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(),
+                    v.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),
+                    drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                    dv.data_ptr(), sq, sk, b, nh, ng, d, (float)scale,
+                    causal, cur_stream());
+  return {dq, dk, dv};
+}
+
+// always-v1 backward, kept for A/B benchmarking
+std::vector<torch::Tensor> attn_bwd_v1(torch::Tensor dout, torch::Tensor q,
+                                       torch::Tensor k, torch::Tensor v,
+                                       torch::Tensor o, torch::Tensor lse,
+                                       double scale, bool causal) {
   const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
             d = (int)q.size(3);
   const int sk = (int)k.size(0), ng = (int)k.size(2);
@@ -551,6 +587,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("selective_scan_fwd", &selective_scan_fwd);
   mod.def("attn_fwd_t", &attn_fwd_t);
   mod.def("attn_fwd2", &attn_fwd2);
+  mod.def("attn_bwd_v1", &attn_bwd_v1);
   mod.def("ce_rowmax", &ce_rowmax);
   mod.def("gemm_nt", &gemm_nt);
   mod.def("gemm_nn", &gemm_nn);

@@ -74,8 +74,12 @@ def bench_shape(sq, b, nh, d, label):
               f"{flops_fwd/dt/1e12:5.0f} TF  maxerr {float(err):.4f}")
 
     dt = timeit(lambda: O.attn_bwd(do, q, k, v, o, lse, scale, True))
-    print(f"attn_bwd:              {dt*1e6:7.0f} us  "
+    print(f"attn_bwd (dispatch):   {dt*1e6:7.0f} us  "
           f"{2.5*flops_fwd/dt/1e12:5.0f} TF-equiv")
+    if hasattr(O, "attn_bwd_v1"):
+        dt = timeit(lambda: O.attn_bwd_v1(do, q, k, v, o, lse, scale, True))
+        print(f"attn_bwd_v1:           {dt*1e6:7.0f} us  "
+              f"{2.5*flops_fwd/dt/1e12:5.0f} TF-equiv")
 
     # unfused forward
     dt = timeit(lambda: unfused_fwd(q, k, v, scale, True))
