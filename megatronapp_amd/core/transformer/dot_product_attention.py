@@ -179,7 +179,8 @@ class FlashAttention(nn.Module):
             return self._fallback(query, key, value, attention_mask,
                                   attn_mask_type, attention_bias, packed_seq_params)
         sq, b, np_, hn = query.shape
-        o = _FlashAttnFn.apply(query.contiguous(), key.contiguous(),
-                               value.contiguous(), self.softmax_scale, True,
+        # the kernels take strided views (d contiguous) directly — no
+        # .contiguous() copies of the QKV-split outputs
+        o = _FlashAttnFn.apply(query, key, value, self.softmax_scale, True,
                                self.dropout_p)
         return o.reshape(sq, b, np_ * hn)

@@ -678,12 +678,15 @@ __device__ __forceinline__ short4v tr16_read(const unsigned short* lds_ptr) {
   return v;
 }
 
+// qS/qB/qH etc. are element strides of the (possibly strided-view)
+// inputs along (seq, batch, head); d must be contiguous.
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, unsigned short* __restrict__ o,
     float* __restrict__ lse, int sq, int sk, int b, int nh, int ng,
-    float scale) {
+    float scale, long qS, long qB, long qH, long kS, long kB, long kH,
+    long vS, long vB, long vH) {
   static_assert(D == 128, "attn_fwd2 is specialized for head dim 128");
   const int qtile = blockIdx.x;
   const int bh = blockIdx.y;
@@ -697,11 +700,10 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
   const int lh = lane >> 5;        // half-wave (MFMA k-half)
   const int lq = lane & 31;        // this lane's q column / MFMA n
 
-  const long q_ss = (long)b * nh * D;
-  const long k_ss = (long)b * ng * D;
-  const unsigned short* qp = q + ((long)bi * nh + h) * D;
-  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
-  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+  const long q_ss = qS;
+  const unsigned short* qp = q + (long)bi * qB + (long)h * qH;
+  const unsigned short* kp = k + (long)bi * kB + (long)hkv * kH;
+  const unsigned short* vp = v + (long)bi * vB + (long)hkv * vH;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* k_lds = (unsigned short*)smem;          // KVBLK*D swizzled
@@ -751,10 +753,10 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
       const int col = idx % D;
       const int grow = kv0 + row;
       kreg[pc] = (grow < sk)
-                     ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                     ? *(const bf16x8*)(kp + (long)grow * kS + col)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       vreg[pc] = (grow < sk)
-                     ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                     ? *(const bf16x8*)(vp + (long)grow * vS + col)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   };
@@ -955,7 +957,9 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
 
 void launch_attn_fwd2(const void* q, const void* k, const void* v, void* o,
                       float* lse, int sq, int sk, int b, int nh, int ng,
-                      int d, float scale, bool causal, hipStream_t stream) {
+                      int d, float scale, bool causal, const long* qstr,
+                      const long* kstr, const long* vstr,
+                      hipStream_t stream) {
   if (sq % FQBLK2 != 0 || sk % KVBLK != 0)
     throw std::runtime_error(
         "attn_fwd2: sq must be a multiple of 256, sk of 64");
@@ -969,12 +973,16 @@ void launch_attn_fwd2(const void* q, const void* k, const void* v, void* o,
     hipLaunchKernelGGL((attn_fwd2_kernel<128, true>), grid, block, lds,
                        stream, (const unsigned short*)q,
                        (const unsigned short*)k, (const unsigned short*)v,
-                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale,
+                       qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],
+                       vstr[0], vstr[1], vstr[2]);
   else
     hipLaunchKernelGGL((attn_fwd2_kernel<128, false>), grid, block, lds,
                        stream, (const unsigned short*)q,
                        (const unsigned short*)k, (const unsigned short*)v,
-                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale);
+                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale,
+                       qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],
+                       vstr[0], vstr[1], vstr[2]);
   HIP_CHECK_LAUNCH();
 }
 
@@ -1026,7 +1034,8 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
     const unsigned short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
     unsigned short* __restrict__ dq, int sq, int sk, int b, int nh, int ng,
-    float scale) {
+    float scale, long qS, long qB, long qH, long kS, long kB, long kH,
+    long vS, long vB, long vH) {
   static_assert(D == 128, "dq2 is specialized for head dim 128");
   const int qtile = blockIdx.x;
   const int bh = blockIdx.y;
@@ -1039,11 +1048,11 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
   const int lh = lane >> 5;
   const int lq = lane & 31;
 
-  const long q_ss = (long)b * nh * D;
-  const long k_ss = (long)b * ng * D;
-  const unsigned short* qp = q + ((long)bi * nh + h) * D;
-  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
-  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+  const long q_ss = qS;
+  const long do_ss = (long)b * nh * D;
+  const unsigned short* qp = q + (long)bi * qB + (long)h * qH;
+  const unsigned short* kp = k + (long)bi * kB + (long)hkv * kH;
+  const unsigned short* vp = v + (long)bi * vB + (long)hkv * vH;
   const unsigned short* dop = dout + ((long)bi * nh + h) * D;
   const float* lse_row = lse + ((long)bi * nh + h) * sq;
   const long dr_ss = (long)b * nh;
@@ -1063,7 +1072,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
   float lse_l, dr_l;
   {
     const unsigned short* src = qp + (long)qrow * q_ss;
-    const unsigned short* dsrc = dop + (long)qrow * q_ss;
+    const unsigned short* dsrc = dop + (long)qrow * do_ss;
 #pragma unroll
     for (int f = 0; f < NF; ++f) {
       qfrag[f] = *(const bf16x8*)(src + 16 * f + 8 * lh);
@@ -1098,10 +1107,10 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
       const int col = idx % D;
       const int grow = kv0 + row;
       kreg[pc] = (grow < sk)
-                     ? *(const bf16x8*)(kp + (long)grow * k_ss + col)
+                     ? *(const bf16x8*)(kp + (long)grow * kS + col)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       vreg[pc] = (grow < sk)
-                     ? *(const bf16x8*)(vp + (long)grow * k_ss + col)
+                     ? *(const bf16x8*)(vp + (long)grow * vS + col)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   };
@@ -1241,7 +1250,8 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
     const unsigned short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
     unsigned short* __restrict__ dk, unsigned short* __restrict__ dv, int sq,
-    int sk, int b, int nh, int ng, float scale) {
+    int sk, int b, int nh, int ng, float scale, long qS, long qB, long qH,
+    long kS, long kB, long kH, long vS, long vB, long vH) {
   static_assert(D == 128, "dkv2 is specialized for head dim 128");
   const int kvtile = blockIdx.x;
   const int bh = blockIdx.y;
@@ -1254,10 +1264,9 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
   const int lh = lane >> 5;
   const int ln = lane & 31;      // this lane's kv column (MFMA n)
 
-  const long q_ss = (long)b * nh * D;
-  const long k_ss = (long)b * ng * D;
-  const unsigned short* kp = k + ((long)bi * ng + hkv) * D;
-  const unsigned short* vp = v + ((long)bi * ng + hkv) * D;
+  const long do_ss = (long)b * nh * D;
+  const unsigned short* kp = k + (long)bi * kB + (long)hkv * kH;
+  const unsigned short* vp = v + (long)bi * vB + (long)hkv * vH;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* q_swz = (unsigned short*)smem;           // 32*D
@@ -1275,8 +1284,8 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
   // keeping them in registers cost 64 VGPRs and pushed the kernel into
   // 256 B/lane scratch spills, which is worse than L2 reads.
   constexpr int NF = D / 16;
-  const unsigned short* ksrc = kp + (long)min(kvcol, sk - 1) * k_ss + 8 * lh;
-  const unsigned short* vsrc = vp + (long)min(kvcol, sk - 1) * k_ss + 8 * lh;
+  const unsigned short* ksrc = kp + (long)min(kvcol, sk - 1) * kS + 8 * lh;
+  const unsigned short* vsrc = vp + (long)min(kvcol, sk - 1) * vS + 8 * lh;
   const bool kv_in_range = kvcol < sk;
 
   constexpr int NDSUB = D / 32;
@@ -1302,7 +1311,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
       16 * ((lane >> 4) & 1) + 4 * (lane & 3);
 
   for (int hq = hkv * group; hq < (hkv + 1) * group; ++hq) {
-    const unsigned short* qp = q + ((long)bi * nh + hq) * D;
+    const unsigned short* qp = q + (long)bi * qB + (long)hq * qH;
     const unsigned short* dop = dout + ((long)bi * nh + hq) * D;
     const float* lse_row = lse + ((long)bi * nh + hq) * sq;
     const long dr_ss = (long)b * nh;
@@ -1319,10 +1328,10 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
           const int col = idx % D;
           const int grow = qt + row;
           bf16x8 q8 = (grow < sq)
-                          ? *(const bf16x8*)(qp + (long)grow * q_ss + col)
+                          ? *(const bf16x8*)(qp + (long)grow * qS + col)
                           : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
           bf16x8 d8 = (grow < sq)
-                          ? *(const bf16x8*)(dop + (long)grow * q_ss + col)
+                          ? *(const bf16x8*)(dop + (long)grow * do_ss + col)
                           : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
           *(bf16x8*)((char*)(q_swz + (long)row * D) + swz(row, col * 2)) = q8;
           *(bf16x8*)((char*)(do_swz + (long)row * D) + swz(row, col * 2)) = d8;
@@ -1453,7 +1462,8 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                       const void* v, const void* o, const float* lse,
                       float* drow, void* dq, void* dk, void* dv, int sq,
                       int sk, int b, int nh, int ng, int d, float scale,
-                      bool causal, hipStream_t stream) {
+                      bool causal, const long* qstr, const long* kstr,
+                      const long* vstr, hipStream_t stream) {
   if (sq % FQBLK2 != 0 || sk % FQBLK2 != 0 || d != 128)
     throw std::runtime_error("attn_bwd2: sq/sk must be multiples of 256, d=128");
   {
@@ -1478,7 +1488,9 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k, (const unsigned short*)v,    \
                        (const unsigned short*)dout, lse, drow,                \
-                       (unsigned short*)dq, sq, sk, b, nh, ng, scale);        \
+                       (unsigned short*)dq, sq, sk, b, nh, ng, scale,         \
+                       qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
+                       vstr[0], vstr[1], vstr[2]);                            \
     HIP_CHECK_LAUNCH();                                                       \
     hipLaunchKernelGGL((attn_bwd_dkv2_kernel<128, CC>),                       \
                        dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dkv,   \
@@ -1486,7 +1498,9 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        (const unsigned short*)k, (const unsigned short*)v,    \
                        (const unsigned short*)dout, lse, drow,                \
                        (unsigned short*)dk, (unsigned short*)dv, sq, sk, b,   \
-                       nh, ng, scale);                                        \
+                       nh, ng, scale,                                         \
+                       qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
+                       vstr[0], vstr[1], vstr[2]);                            \
     HIP_CHECK_LAUNCH();                                                       \
   } while (0)
   if (causal) ATT_BWD2_LAUNCH(true);

@@ -57,10 +57,23 @@ void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
 void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
 void launch_attn_fwd2(const void*, const void*, const void*, void*, float*,
-                      int, int, int, int, int, int, float, bool, hipStream_t);
+                      int, int, int, int, int, int, float, bool,
+                      const long*, const long*, const long*, hipStream_t);
 void launch_attn_bwd2(const void*, const void*, const void*, const void*,
                       const void*, const float*, float*, void*, void*, void*,
-                      int, int, int, int, int, int, float, bool, hipStream_t);
+                      int, int, int, int, int, int, float, bool,
+                      const long*, const long*, const long*, hipStream_t);
+
+// The round-2 kernels take (seq, batch, head) element strides so the
+// strided QKV-split views feed them without .contiguous() copies; d
+// must be unit-stride.
+static bool attn_strides(const torch::Tensor& t, long* out) {
+  if (t.stride(3) != 1) return false;
+  out[0] = (long)t.stride(0);
+  out[1] = (long)t.stride(1);
+  out[2] = (long)t.stride(2);
+  return true;
+}
 void launch_attn_fwd_t(const void*, const void*, const void*, void*, float*,
                        int, int, int, int, int, int, float, bool,
                        hipStream_t);
@@ -83,6 +96,12 @@ void check_bf16(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// attention inputs may be strided views (d contiguous); no contiguity check
+void check_bf16_any(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
 }
 
 // ------------------------------------------------------------------- norms
@@ -351,31 +370,35 @@ void embedding_bwd_accum(torch::Tensor dy, torch::Tensor tokens,
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, double scale,
                                     bool causal) {
-  check_bf16(q, "q");
-  check_bf16(k, "k");
-  check_bf16(v, "v");
+  check_bf16_any(q, "q");
+  check_bf16_any(k, "k");
+  check_bf16_any(v, "v");
   const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
             d = (int)q.size(3);
   const int sk = (int)k.size(0), ng = (int)k.size(2);
-  auto o = torch::empty_like(q);
+  auto o = torch::empty({sq, b, nh, d}, q.options());
   auto lse = torch::empty({b, nh, sq}, q.options().dtype(torch::kFloat32));
-  if (d == 128 && sq % 256 == 0 && sk % 64 == 0)
+  long qs[3], ks[3], vs[3];
+  if (d == 128 && sq % 256 == 0 && sk % 64 == 0 && attn_strides(q, qs) &&
+      attn_strides(k, ks) && attn_strides(v, vs)) {
     launch_attn_fwd2(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                      lse.data_ptr<float>(), sq, sk, b, nh, ng, d,
-                     (float)scale, causal, cur_stream());
-  else
-    launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                     (float)scale, causal, qs, ks, vs, cur_stream());
+  } else {
+    auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+    launch_attn_fwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(), o.data_ptr(),
                     lse.data_ptr<float>(), sq, sk, b, nh, ng, d, (float)scale,
                     causal, cur_stream());
+  }
   return {o, lse};
 }
 
 std::vector<torch::Tensor> attn_fwd_t(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, double scale,
                                     bool causal) {
-  check_bf16(q, "q");
-  check_bf16(k, "k");
-  check_bf16(v, "v");
+  check_bf16_any(q, "q");
+  check_bf16_any(k, "k");
+  check_bf16_any(v, "v");
   const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
             d = (int)q.size(3);
   const int sk = (int)k.size(0), ng = (int)k.size(2);
@@ -390,17 +413,20 @@ std::vector<torch::Tensor> attn_fwd_t(torch::Tensor q, torch::Tensor k,
 std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
                                      torch::Tensor v, double scale,
                                      bool causal) {
-  check_bf16(q, "q");
-  check_bf16(k, "k");
-  check_bf16(v, "v");
+  check_bf16_any(q, "q");
+  check_bf16_any(k, "k");
+  check_bf16_any(v, "v");
   const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
             d = (int)q.size(3);
   const int sk = (int)k.size(0), ng = (int)k.size(2);
-  auto o = torch::empty_like(q);
+  auto o = torch::empty({sq, b, nh, d}, q.options());
   auto lse = torch::empty({b, nh, sq}, q.options().dtype(torch::kFloat32));
+  long qs[3], ks[3], vs[3];
+  TORCH_CHECK(attn_strides(q, qs) && attn_strides(k, ks) &&
+              attn_strides(v, vs), "attn_fwd2: d must be contiguous");
   launch_attn_fwd2(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                    lse.data_ptr<float>(), sq, sk, b, nh, ng, d, (float)scale,
-                   causal, cur_stream());
+                   causal, qs, ks, vs, cur_stream());
   return {o, lse};
 }
 
@@ -408,7 +434,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     double scale, bool causal) {
-  check_bf16(dout, "dout");
+  check_bf16_any(dout, "dout");
+  check_bf16_any(q, "q");
+  check_bf16_any(k, "k");
+  check_bf16_any(v, "v");
   const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
             d = (int)q.size(3);
   const int sk = (int)k.size(0), ng = (int)k.size(2);
@@ -417,18 +446,23 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   auto dv = torch::empty_like(v);
   auto drow = torch::empty({(long)sq * b * nh},
                            q.options().dtype(torch::kFloat32));
-  if (d == 128 && sq % 256 == 0 && sk % 256 == 0)
-    launch_attn_bwd2(dout.data_ptr(), q.data_ptr(), k.data_ptr(),
+  long qs[3], ks[3], vs[3];
+  auto doc = dout.contiguous();
+  if (d == 128 && sq % 256 == 0 && sk % 256 == 0 && attn_strides(q, qs) &&
+      attn_strides(k, ks) && attn_strides(v, vs)) {
+    launch_attn_bwd2(doc.data_ptr(), q.data_ptr(), k.data_ptr(),
                      v.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),
                      drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                      dv.data_ptr(), sq, sk, b, nh, ng, d, (float)scale,
-                     causal, cur_stream());
-  else
-    launch_attn_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(),
-                    v.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),
+                     causal, qs, ks, vs, cur_stream());
+  } else {
+    auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+    launch_attn_bwd(doc.data_ptr(), qc.data_ptr(), kc.data_ptr(),
+                    vc.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),
                     drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                     dv.data_ptr(), sq, sk, b, nh, ng, d, (float)scale,
                     causal, cur_stream());
+  }
   return {dq, dk, dv};
 }
 
