@@ -1243,6 +1243,12 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
 }
 
 // ------------------------------------------------------------------ dkv2
+// 64-row q tiles (2 MFMA subtiles per staging round) with T14 register
+// prefetch across the flattened (head, q-tile) loop: the round-1 version
+// staged 32-row tiles synchronously and was barrier/latency-bound at
+// 3.6 ms/call (80% of the whole backward).
+#define QT2 64
+
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
@@ -1269,20 +1275,17 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
   const unsigned short* vp = v + (long)bi * vB + (long)hkv * vH;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  unsigned short* q_swz = (unsigned short*)smem;           // 32*D
-  unsigned short* do_swz = q_swz + 32 * D;                 // 32*D
-  unsigned short* q_rm = do_swz + 32 * D;                  // 32*VRS
-  unsigned short* do_rm = q_rm + 32 * VRS;                 // 32*VRS
-  float* lse_lds = (float*)(do_rm + 32 * VRS);             // 32
-  float* dr_lds = lse_lds + 32;                            // 32
+  unsigned short* q_swz = (unsigned short*)smem;           // QT2*D
+  unsigned short* do_swz = q_swz + QT2 * D;                // QT2*D
+  unsigned short* q_rm = do_swz + QT2 * D;                 // QT2*VRS
+  unsigned short* do_rm = q_rm + QT2 * VRS;                // QT2*VRS
+  float* lse_lds = (float*)(do_rm + QT2 * VRS);            // QT2
+  float* dr_lds = lse_lds + QT2;                           // QT2
 
   const int kv0w = kvtile * FQBLK2 + wid * QBLK2_ROWS;
-  const int kvcol = kv0w + ln;   // this lane's kv row in K/V
+  const int kvcol = kv0w + ln;
 
-  // K/V B-fragments are re-read from GLOBAL each q tile (the block's K/V
-  // rows are 128 KB and stay L2-resident across the whole q loop):
-  // keeping them in registers cost 64 VGPRs and pushed the kernel into
-  // 256 B/lane scratch spills, which is worse than L2 reads.
+  // K/V B-fragments are re-read from L2 per subtile (registers spill)
   constexpr int NF = D / 16;
   const unsigned short* ksrc = kp + (long)min(kvcol, sk - 1) * kS + 8 * lh;
   const unsigned short* vsrc = vp + (long)min(kvcol, sk - 1) * vS + 8 * lh;
@@ -1300,8 +1303,9 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
 
   const int kv_lo = kvtile * FQBLK2;
   int q_start = 0;
-  if (CAUSAL)
-    q_start = max(0, (kv_lo - (sk - sq)) / QBLK2_ROWS * QBLK2_ROWS);
+  if (CAUSAL) q_start = max(0, (kv_lo - (sk - sq)) / QT2 * QT2);
+  const int nqt = (sq - q_start + QT2 - 1) / QT2;
+  const int n_it = group * nqt;   // flattened (hq, q-tile) iterations
 
   const unsigned short* q_tr_base =
       q_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
@@ -1310,129 +1314,159 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
       do_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
       16 * ((lane >> 4) & 1) + 4 * (lane & 3);
 
-  for (int hq = hkv * group; hq < (hkv + 1) * group; ++hq) {
-    const unsigned short* qp = q + (long)bi * qB + (long)hq * qH;
-    const unsigned short* dop = dout + ((long)bi * nh + hq) * D;
-    const float* lse_row = lse + ((long)bi * nh + hq) * sq;
-    const long dr_ss = (long)b * nh;
-    const float* dr_base = drow + (long)bi * nh + hq;
+  // T14 staging registers for the next tile (2 pieces per tensor)
+  constexpr int PIECES = QT2 * D / 8 / FWD_BLOCK;  // 2 at D=128
+  bf16x8 qreg[PIECES], dreg[PIECES];
+  float lreg = 0.f, rreg = 0.f;
 
-    for (int qt = q_start; qt < sq; qt += QBLK2_ROWS) {
-      // stage Q/dO (swizzled rows + row-major for tr reads) + lse/drow
-      {
-        constexpr int pieces = 32 * D / 8 / FWD_BLOCK;  // 1 at D=128
+  auto stage_load = [&](int it) {
+    const int hq = hkv * group + it / nqt;
+    const int qt = q_start + (it % nqt) * QT2;
+    const unsigned short* qph = q + (long)bi * qB + (long)hq * qH;
+    const unsigned short* doph = dout + ((long)bi * nh + hq) * D;
 #pragma unroll
-        for (int pc = 0; pc < pieces; ++pc) {
-          const int idx = (pc * FWD_BLOCK + tid) * 8;
-          const int row = idx / D;
-          const int col = idx % D;
-          const int grow = qt + row;
-          bf16x8 q8 = (grow < sq)
-                          ? *(const bf16x8*)(qp + (long)grow * qS + col)
-                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-          bf16x8 d8 = (grow < sq)
-                          ? *(const bf16x8*)(dop + (long)grow * do_ss + col)
-                          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-          *(bf16x8*)((char*)(q_swz + (long)row * D) + swz(row, col * 2)) = q8;
-          *(bf16x8*)((char*)(do_swz + (long)row * D) + swz(row, col * 2)) = d8;
-          *(bf16x8*)(q_rm + (long)row * VRS + col) = q8;
-          *(bf16x8*)(do_rm + (long)row * VRS + col) = d8;
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int grow = qt + row;
+      qreg[pc] = (grow < sq)
+                     ? *(const bf16x8*)(qph + (long)grow * qS + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      dreg[pc] = (grow < sq)
+                     ? *(const bf16x8*)(doph + (long)grow * do_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    if (tid < QT2) {
+      const int grow = qt + tid;
+      const float* lse_row = lse + ((long)bi * nh + hq) * sq;
+      const float* dr_base = drow + (long)bi * nh + hq;
+      lreg = (grow < sq) ? lse_row[grow] : 0.f;
+      rreg = (grow < sq) ? dr_base[(long)grow * (long)b * nh] : 0.f;
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      *(bf16x8*)((char*)(q_swz + (long)row * D) + swz(row, col * 2)) =
+          qreg[pc];
+      *(bf16x8*)((char*)(do_swz + (long)row * D) + swz(row, col * 2)) =
+          dreg[pc];
+      *(bf16x8*)(q_rm + (long)row * VRS + col) = qreg[pc];
+      *(bf16x8*)(do_rm + (long)row * VRS + col) = dreg[pc];
+    }
+    if (tid < QT2) {
+      lse_lds[tid] = lreg;
+      dr_lds[tid] = rreg;
+    }
+  };
+
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+
+  for (int it = 0; it < n_it; ++it) {
+    const int qt = q_start + (it % nqt) * QT2;
+    if (it + 1 < n_it) stage_load(it + 1);
+
+#pragma clang loop unroll(disable)
+    for (int qs2 = 0; qs2 < 2; ++qs2) {
+      const int qt0 = qt + 32 * qs2;
+      const bool active =
+          !CAUSAL || (kv0w <= qt0 + 31 + (sk - sq));
+      if (!active) continue;
+      // S and dP in C[m=q][n=kv]
+      f32x16 sc, dpc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        sc[r] = 0.f;
+        dpc[r] = 0.f;
+      }
+      const int qrow_a = 32 * qs2 + ln;
+#pragma unroll
+      for (int f = 0; f < NF; ++f) {
+        const bf16x8 z8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        bf16x8 kf = kv_in_range ? *(const bf16x8*)(ksrc + 16 * f) : z8;
+        bf16x8 vf = kv_in_range ? *(const bf16x8*)(vsrc + 16 * f) : z8;
+        bf16x8 qa = *(const bf16x8*)(
+            (char*)(q_swz + (long)qrow_a * D) +
+            swz(qrow_a, (16 * f + 8 * lh) * 2));
+        sc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf, sc, 0, 0, 0);
+        bf16x8 da = *(const bf16x8*)(
+            (char*)(do_swz + (long)qrow_a * D) +
+            swz(qrow_a, (16 * f + 8 * lh) * 2));
+        dpc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf, dpc, 0, 0, 0);
+      }
+      // P overwrites sc, dS overwrites dpc (keeps transients in the
+      // existing f32x16 registers instead of two extra arrays)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qr = 32 * qs2 + (r & 3) + 8 * (r >> 2) + 4 * lh;
+        const int qrow = qt + qr;
+        bool valid = (qrow < sq) && kv_in_range;
+        if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
+        float pp = valid ? __expf(sc[r] * scale - lse_lds[qr]) : 0.f;
+        sc[r] = pp;
+        dpc[r] = pp * (dpc[r] - dr_lds[qr]) * scale;
+      }
+      unsigned pwds[2][4], dswds[2][4];
+      SWAP_ASSEMBLE(sc, pwds);
+      SWAP_ASSEMBLE(dpc, dswds);
+
+#pragma unroll
+      for (int dsub = 0; dsub < NDSUB; ++dsub) {
+        short4v r8[4];
+#pragma unroll
+        for (int K = 0; K < 2; ++K)
+#pragma unroll
+          for (int rr = 0; rr < 2; ++rr)
+            r8[2 * K + rr] = tr16_read(
+                do_tr_base + (long)(32 * qs2 + 16 * K + 4 * rr) * VRS +
+                32 * dsub);
+        asm volatile("s_waitcnt lgkmcnt(0)"
+                     : "+v"(r8[0]), "+v"(r8[1]), "+v"(r8[2]), "+v"(r8[3])
+                     :: "memory");
+#pragma unroll
+        for (int K = 0; K < 2; ++K) {
+          bf16x8 df;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            df[j] = r8[2 * K][j];
+            df[4 + j] = r8[2 * K + 1][j];
+          }
+          dvacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              frag_from_words(pwds[K]), df, dvacc[dsub], 0, 0, 0);
         }
-        if (tid < 32) {
-          const int grow = qt + tid;
-          lse_lds[tid] = (grow < sq) ? lse_row[grow] : 0.f;
-          dr_lds[tid] = (grow < sq) ? dr_base[(long)grow * dr_ss] : 0.f;
+#pragma unroll
+        for (int K = 0; K < 2; ++K)
+#pragma unroll
+          for (int rr = 0; rr < 2; ++rr)
+            r8[2 * K + rr] = tr16_read(
+                q_tr_base + (long)(32 * qs2 + 16 * K + 4 * rr) * VRS +
+                32 * dsub);
+        asm volatile("s_waitcnt lgkmcnt(0)"
+                     : "+v"(r8[0]), "+v"(r8[1]), "+v"(r8[2]), "+v"(r8[3])
+                     :: "memory");
+#pragma unroll
+        for (int K = 0; K < 2; ++K) {
+          bf16x8 qf;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            qf[j] = r8[2 * K][j];
+            qf[4 + j] = r8[2 * K + 1][j];
+          }
+          dkacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              frag_from_words(dswds[K]), qf, dkacc[dsub], 0, 0, 0);
         }
       }
-      __syncthreads();
+    }
 
-      const bool active = !CAUSAL || (kv0w <= qt + QBLK2_ROWS - 1 + (sk - sq));
-      if (active) {
-        // S and dP in C[m=q][n=kv]
-        f32x16 sc, dpc;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          sc[r] = 0.f;
-          dpc[r] = 0.f;
-        }
-        const int qrow_a = lane & 31;  // A-frag row within the q tile
-#pragma unroll
-        for (int f = 0; f < NF; ++f) {
-          const bf16x8 z8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-          bf16x8 kf = kv_in_range ? *(const bf16x8*)(ksrc + 16 * f) : z8;
-          bf16x8 vf = kv_in_range ? *(const bf16x8*)(vsrc + 16 * f) : z8;
-          bf16x8 qa = *(const bf16x8*)(
-              (char*)(q_swz + (long)qrow_a * D) +
-              swz(qrow_a, (16 * f + 8 * lh) * 2));
-          sc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf, sc, 0, 0, 0);
-          bf16x8 da = *(const bf16x8*)(
-              (char*)(do_swz + (long)qrow_a * D) +
-              swz(qrow_a, (16 * f + 8 * lh) * 2));
-          dpc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf, dpc,
-                                                        0, 0, 0);
-        }
-        // P and dS per lane (lse/drow indexed by the reg's q row)
-        float pv[16], dsv[16];
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int qr = (r & 3) + 8 * (r >> 2) + 4 * lh;  // within tile
-          const int qrow = qt + qr;
-          bool valid = (qrow < sq) && (kvcol < sk);
-          if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
-          float p = valid ? __expf(sc[r] * scale - lse_lds[qr]) : 0.f;
-          pv[r] = p;
-          dsv[r] = p * (dpc[r] - dr_lds[qr]) * scale;
-        }
-        unsigned pwds[2][4], dswds[2][4];
-        SWAP_ASSEMBLE(pv, pwds);
-        SWAP_ASSEMBLE(dsv, dswds);
-
-        // dV += P^T dO ; dK += dS^T Q  (tr fragments of dO / Q)
-#pragma unroll
-        for (int dsub = 0; dsub < NDSUB; ++dsub) {
-          short4v r8[4];
-#pragma unroll
-          for (int K = 0; K < 2; ++K)
-#pragma unroll
-            for (int rr = 0; rr < 2; ++rr)
-              r8[2 * K + rr] = tr16_read(
-                  do_tr_base + (long)(16 * K + 4 * rr) * VRS + 32 * dsub);
-          asm volatile("s_waitcnt lgkmcnt(0)"
-                       : "+v"(r8[0]), "+v"(r8[1]), "+v"(r8[2]), "+v"(r8[3])
-                       :: "memory");
-#pragma unroll
-          for (int K = 0; K < 2; ++K) {
-            bf16x8 df;
-#pragma unroll
-            for (int j = 0; j < 4; ++j) {
-              df[j] = r8[2 * K][j];
-              df[4 + j] = r8[2 * K + 1][j];
-            }
-            dvacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                frag_from_words(pwds[K]), df, dvacc[dsub], 0, 0, 0);
-          }
-#pragma unroll
-          for (int K = 0; K < 2; ++K)
-#pragma unroll
-            for (int rr = 0; rr < 2; ++rr)
-              r8[2 * K + rr] = tr16_read(
-                  q_tr_base + (long)(16 * K + 4 * rr) * VRS + 32 * dsub);
-          asm volatile("s_waitcnt lgkmcnt(0)"
-                       : "+v"(r8[0]), "+v"(r8[1]), "+v"(r8[2]), "+v"(r8[3])
-                       :: "memory");
-#pragma unroll
-          for (int K = 0; K < 2; ++K) {
-            bf16x8 qf;
-#pragma unroll
-            for (int j = 0; j < 4; ++j) {
-              qf[j] = r8[2 * K][j];
-              qf[4 + j] = r8[2 * K + 1][j];
-            }
-            dkacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                frag_from_words(dswds[K]), qf, dkacc[dsub], 0, 0, 0);
-          }
-        }
-      }
+    __syncthreads();
+    if (it + 1 < n_it) {
+      stage_write();
       __syncthreads();
     }
   }
@@ -1452,6 +1486,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
       }
     }
 }
+
 
 __global__ void attn_bwd_pre_kernel(const unsigned short* __restrict__ do_,
                                     const unsigned short* __restrict__ o,
@@ -1478,9 +1513,9 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
   }
   const size_t lds_dq = (size_t)(2 * KVBLK * 128 + KVBLK * VRS +
                                  8 * 32 * OSTRIDE) * sizeof(unsigned short);
-  const size_t lds_dkv = (size_t)(2 * 32 * 128 + 2 * 32 * VRS) *
+  const size_t lds_dkv = (size_t)(2 * QT2 * 128 + 2 * QT2 * VRS) *
                              sizeof(unsigned short) +
-                         64 * sizeof(float);
+                         2 * QT2 * sizeof(float);
 #define ATT_BWD2_LAUNCH(CC)                                                   \
   do {                                                                        \
     hipLaunchKernelGGL((attn_bwd_dq2_kernel<128, CC>),                        \
