@@ -68,7 +68,7 @@ def parse_args():
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--trace", action="store_true", help="enable MegaScan tracing")
     p.add_argument("--trace-dir", default="trace_out")
-    p.add_argument("--attention", default="fused", choices=["flash", "fused"])
+    p.add_argument("--attention", default="flash", choices=["flash", "fused"])
     p.add_argument("--seq-length", type=int, default=None,
                    help="override the model's sequence length")
     p.add_argument("--no-overlap-grad-reduce", action="store_true")

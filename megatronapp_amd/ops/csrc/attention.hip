@@ -1242,23 +1242,25 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
   }
 }
 
-// ------------------------------------------------------------------ dkv2
-// 64-row q tiles (2 MFMA subtiles per staging round) with T14 register
-// prefetch across the flattened (head, q-tile) loop: the round-1 version
-// staged 32-row tiles synchronously and was barrier/latency-bound at
-// 3.6 ms/call (80% of the whole backward).
+// ------------------------------------------------------- dv2 / dk2
+// The kv-side backward, split by OUTPUT into two dq2-shaped kernels:
+// one 64-register accumulator set each plus per-lane-resident K (and V)
+// fragments, streaming 64-row q tiles through LDS with T14 prefetch.
+// The combined dK+dV kernel could not keep K/V resident (256 B/lane
+// spills) and re-read them from L2 every subtile (~10 GB/call); the
+// split recomputes S (one extra matmul of five) but keeps every operand
+// where it belongs.
 #define QT2 64
 
+// dV[kv][d] = P^T[kv][q] dO[q][d];  P = exp(S - lse), S = Q K^T
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
+__global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dv2_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
-    const unsigned short* __restrict__ v,
     const unsigned short* __restrict__ dout,
-    const float* __restrict__ lse, const float* __restrict__ drow,
-    unsigned short* __restrict__ dk, unsigned short* __restrict__ dv, int sq,
-    int sk, int b, int nh, int ng, float scale, long qS, long qB, long qH,
-    long kS, long kB, long kH, long vS, long vB, long vH) {
-  static_assert(D == 128, "dkv2 is specialized for head dim 128");
+    const float* __restrict__ lse,
+    unsigned short* __restrict__ dv, int sq, int sk, int b, int nh, int ng,
+    float scale, long qS, long qB, long qH, long kS, long kB, long kH) {
+  static_assert(D == 128, "dv2 is specialized for head dim 128");
   const int kvtile = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / ng;
@@ -1268,57 +1270,51 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
   const int wid = tid / WAVE;
   const int lane = tid % WAVE;
   const int lh = lane >> 5;
-  const int ln = lane & 31;      // this lane's kv column (MFMA n)
+  const int ln = lane & 31;
 
   const long do_ss = (long)b * nh * D;
   const unsigned short* kp = k + (long)bi * kB + (long)hkv * kH;
-  const unsigned short* vp = v + (long)bi * vB + (long)hkv * vH;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* q_swz = (unsigned short*)smem;           // QT2*D
-  unsigned short* do_swz = q_swz + QT2 * D;                // QT2*D
-  unsigned short* q_rm = do_swz + QT2 * D;                 // QT2*VRS
-  unsigned short* do_rm = q_rm + QT2 * VRS;                // QT2*VRS
+  unsigned short* do_rm = q_swz + QT2 * D;                 // QT2*VRS
   float* lse_lds = (float*)(do_rm + QT2 * VRS);            // QT2
-  float* dr_lds = lse_lds + QT2;                           // QT2
 
   const int kv0w = kvtile * FQBLK2 + wid * QBLK2_ROWS;
   const int kvcol = kv0w + ln;
-
-  // K/V B-fragments are re-read from L2 per subtile (registers spill)
-  constexpr int NF = D / 16;
-  const unsigned short* ksrc = kp + (long)min(kvcol, sk - 1) * kS + 8 * lh;
-  const unsigned short* vsrc = vp + (long)min(kvcol, sk - 1) * vS + 8 * lh;
   const bool kv_in_range = kvcol < sk;
 
+  // K fragments resident: B[k=d][n=kv] -> K[kvcol][16f + 8lh + j]
+  constexpr int NF = D / 16;
+  bf16x8 kfrag[NF];
+  {
+    const unsigned short* ksrc = kp + (long)min(kvcol, sk - 1) * kS + 8 * lh;
+#pragma unroll
+    for (int f = 0; f < NF; ++f)
+      kfrag[f] = kv_in_range ? *(const bf16x8*)(ksrc + 16 * f)
+                             : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+
   constexpr int NDSUB = D / 32;
-  f32x16 dkacc[NDSUB], dvacc[NDSUB];
+  f32x16 dvacc[NDSUB];
 #pragma unroll
   for (int s = 0; s < NDSUB; ++s)
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      dkacc[s][r] = 0.f;
-      dvacc[s][r] = 0.f;
-    }
+    for (int r = 0; r < 16; ++r) dvacc[s][r] = 0.f;
 
   const int kv_lo = kvtile * FQBLK2;
   int q_start = 0;
   if (CAUSAL) q_start = max(0, (kv_lo - (sk - sq)) / QT2 * QT2);
   const int nqt = (sq - q_start + QT2 - 1) / QT2;
-  const int n_it = group * nqt;   // flattened (hq, q-tile) iterations
+  const int n_it = group * nqt;
 
-  const unsigned short* q_tr_base =
-      q_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
-      16 * ((lane >> 4) & 1) + 4 * (lane & 3);
   const unsigned short* do_tr_base =
       do_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
       16 * ((lane >> 4) & 1) + 4 * (lane & 3);
 
-  // T14 staging registers for the next tile (2 pieces per tensor)
-  constexpr int PIECES = QT2 * D / 8 / FWD_BLOCK;  // 2 at D=128
+  constexpr int PIECES = QT2 * D / 8 / FWD_BLOCK;
   bf16x8 qreg[PIECES], dreg[PIECES];
-  float lreg = 0.f, rreg = 0.f;
-
+  float lreg = 0.f;
   auto stage_load = [&](int it) {
     const int hq = hkv * group + it / nqt;
     const int qt = q_start + (it % nqt) * QT2;
@@ -1339,10 +1335,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
     }
     if (tid < QT2) {
       const int grow = qt + tid;
-      const float* lse_row = lse + ((long)bi * nh + hq) * sq;
-      const float* dr_base = drow + (long)bi * nh + hq;
-      lreg = (grow < sq) ? lse_row[grow] : 0.f;
-      rreg = (grow < sq) ? dr_base[(long)grow * (long)b * nh] : 0.f;
+      lreg = (grow < sq) ? lse[((long)bi * nh + hq) * sq + grow] : 0.f;
     }
   };
   auto stage_write = [&]() {
@@ -1353,15 +1346,9 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
       const int col = idx % D;
       *(bf16x8*)((char*)(q_swz + (long)row * D) + swz(row, col * 2)) =
           qreg[pc];
-      *(bf16x8*)((char*)(do_swz + (long)row * D) + swz(row, col * 2)) =
-          dreg[pc];
-      *(bf16x8*)(q_rm + (long)row * VRS + col) = qreg[pc];
       *(bf16x8*)(do_rm + (long)row * VRS + col) = dreg[pc];
     }
-    if (tid < QT2) {
-      lse_lds[tid] = lreg;
-      dr_lds[tid] = rreg;
-    }
+    if (tid < QT2) lse_lds[tid] = lreg;
   };
 
   stage_load(0);
@@ -1375,47 +1362,29 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
 #pragma clang loop unroll(disable)
     for (int qs2 = 0; qs2 < 2; ++qs2) {
       const int qt0 = qt + 32 * qs2;
-      const bool active =
-          !CAUSAL || (kv0w <= qt0 + 31 + (sk - sq));
-      if (!active) continue;
-      // S and dP in C[m=q][n=kv]
-      f32x16 sc, dpc;
+      if (CAUSAL && kv0w > qt0 + 31 + (sk - sq)) continue;
+      f32x16 sc;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        sc[r] = 0.f;
-        dpc[r] = 0.f;
-      }
+      for (int r = 0; r < 16; ++r) sc[r] = 0.f;
       const int qrow_a = 32 * qs2 + ln;
 #pragma unroll
       for (int f = 0; f < NF; ++f) {
-        const bf16x8 z8 = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 kf = kv_in_range ? *(const bf16x8*)(ksrc + 16 * f) : z8;
-        bf16x8 vf = kv_in_range ? *(const bf16x8*)(vsrc + 16 * f) : z8;
         bf16x8 qa = *(const bf16x8*)(
             (char*)(q_swz + (long)qrow_a * D) +
             swz(qrow_a, (16 * f + 8 * lh) * 2));
-        sc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf, sc, 0, 0, 0);
-        bf16x8 da = *(const bf16x8*)(
-            (char*)(do_swz + (long)qrow_a * D) +
-            swz(qrow_a, (16 * f + 8 * lh) * 2));
-        dpc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf, dpc, 0, 0, 0);
+        sc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[f], sc,
+                                                     0, 0, 0);
       }
-      // P overwrites sc, dS overwrites dpc (keeps transients in the
-      // existing f32x16 registers instead of two extra arrays)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qr = 32 * qs2 + (r & 3) + 8 * (r >> 2) + 4 * lh;
         const int qrow = qt + qr;
         bool valid = (qrow < sq) && kv_in_range;
         if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
-        float pp = valid ? __expf(sc[r] * scale - lse_lds[qr]) : 0.f;
-        sc[r] = pp;
-        dpc[r] = pp * (dpc[r] - dr_lds[qr]) * scale;
+        sc[r] = valid ? __expf(sc[r] * scale - lse_lds[qr]) : 0.f;
       }
-      unsigned pwds[2][4], dswds[2][4];
+      unsigned pwds[2][4];
       SWAP_ASSEMBLE(sc, pwds);
-      SWAP_ASSEMBLE(dpc, dswds);
-
 #pragma unroll
       for (int dsub = 0; dsub < NDSUB; ++dsub) {
         short4v r8[4];
@@ -1440,6 +1409,187 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
           dvacc[dsub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               frag_from_words(pwds[K]), df, dvacc[dsub], 0, 0, 0);
         }
+      }
+    }
+    __syncthreads();
+    if (it + 1 < n_it) {
+      stage_write();
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int dsub = 0; dsub < NDSUB; ++dsub)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kvrow = kv0w + (r & 3) + 8 * (r >> 2) + 4 * lh;
+      if (kvrow < sk)
+        dv[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + 32 * dsub +
+           ln] = f2bf(dvacc[dsub][r]);
+    }
+}
+
+// dK[kv][d] = dS^T[kv][q] Q[q][d];  dS = P (dP - Drow) scale
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dk2_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    unsigned short* __restrict__ dk, int sq, int sk, int b, int nh, int ng,
+    float scale, long qS, long qB, long qH, long kS, long kB, long kH,
+    long vS, long vB, long vH) {
+  static_assert(D == 128, "dk2 is specialized for head dim 128");
+  const int kvtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int bi = bh / ng;
+  const int hkv = bh % ng;
+  const int group = nh / ng;
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int lh = lane >> 5;
+  const int ln = lane & 31;
+
+  const long do_ss = (long)b * nh * D;
+  const unsigned short* kp = k + (long)bi * kB + (long)hkv * kH;
+  const unsigned short* vp = v + (long)bi * vB + (long)hkv * vH;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* q_swz = (unsigned short*)smem;           // QT2*D
+  unsigned short* do_swz = q_swz + QT2 * D;                // QT2*D
+  unsigned short* q_rm = do_swz + QT2 * D;                 // QT2*VRS
+  float* lse_lds = (float*)(q_rm + QT2 * VRS);             // QT2
+  float* dr_lds = lse_lds + QT2;                           // QT2
+
+  const int kv0w = kvtile * FQBLK2 + wid * QBLK2_ROWS;
+  const int kvcol = kv0w + ln;
+  const bool kv_in_range = kvcol < sk;
+
+  constexpr int NF = D / 16;
+  bf16x8 kfrag[NF], vfrag[NF];
+  {
+    const unsigned short* ksrc = kp + (long)min(kvcol, sk - 1) * kS + 8 * lh;
+    const unsigned short* vsrc = vp + (long)min(kvcol, sk - 1) * vS + 8 * lh;
+#pragma unroll
+    for (int f = 0; f < NF; ++f) {
+      kfrag[f] = kv_in_range ? *(const bf16x8*)(ksrc + 16 * f)
+                             : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vfrag[f] = kv_in_range ? *(const bf16x8*)(vsrc + 16 * f)
+                             : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  constexpr int NDSUB = D / 32;
+  f32x16 dkacc[NDSUB];
+#pragma unroll
+  for (int s = 0; s < NDSUB; ++s)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dkacc[s][r] = 0.f;
+
+  const int kv_lo = kvtile * FQBLK2;
+  int q_start = 0;
+  if (CAUSAL) q_start = max(0, (kv_lo - (sk - sq)) / QT2 * QT2);
+  const int nqt = (sq - q_start + QT2 - 1) / QT2;
+  const int n_it = group * nqt;
+
+  const unsigned short* q_tr_base =
+      q_rm + (long)(8 * lh + ((lane >> 2) & 3)) * VRS +
+      16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+
+  constexpr int PIECES = QT2 * D / 8 / FWD_BLOCK;
+  bf16x8 qreg[PIECES], dreg[PIECES];
+  float lreg = 0.f, rreg = 0.f;
+  auto stage_load = [&](int it) {
+    const int hq = hkv * group + it / nqt;
+    const int qt = q_start + (it % nqt) * QT2;
+    const unsigned short* qph = q + (long)bi * qB + (long)hq * qH;
+    const unsigned short* doph = dout + ((long)bi * nh + hq) * D;
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      const int grow = qt + row;
+      qreg[pc] = (grow < sq)
+                     ? *(const bf16x8*)(qph + (long)grow * qS + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      dreg[pc] = (grow < sq)
+                     ? *(const bf16x8*)(doph + (long)grow * do_ss + col)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    if (tid < QT2) {
+      const int grow = qt + tid;
+      lreg = (grow < sq) ? lse[((long)bi * nh + hq) * sq + grow] : 0.f;
+      rreg = (grow < sq)
+                 ? drow[(long)grow * (long)b * nh + (long)bi * nh + hq]
+                 : 0.f;
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int idx = (pc * FWD_BLOCK + tid) * 8;
+      const int row = idx / D;
+      const int col = idx % D;
+      *(bf16x8*)((char*)(q_swz + (long)row * D) + swz(row, col * 2)) =
+          qreg[pc];
+      *(bf16x8*)((char*)(do_swz + (long)row * D) + swz(row, col * 2)) =
+          dreg[pc];
+      *(bf16x8*)(q_rm + (long)row * VRS + col) = qreg[pc];
+    }
+    if (tid < QT2) {
+      lse_lds[tid] = lreg;
+      dr_lds[tid] = rreg;
+    }
+  };
+
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+
+  for (int it = 0; it < n_it; ++it) {
+    const int qt = q_start + (it % nqt) * QT2;
+    if (it + 1 < n_it) stage_load(it + 1);
+
+#pragma clang loop unroll(disable)
+    for (int qs2 = 0; qs2 < 2; ++qs2) {
+      const int qt0 = qt + 32 * qs2;
+      if (CAUSAL && kv0w > qt0 + 31 + (sk - sq)) continue;
+      f32x16 sc, dpc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        sc[r] = 0.f;
+        dpc[r] = 0.f;
+      }
+      const int qrow_a = 32 * qs2 + ln;
+#pragma unroll
+      for (int f = 0; f < NF; ++f) {
+        bf16x8 qa = *(const bf16x8*)(
+            (char*)(q_swz + (long)qrow_a * D) +
+            swz(qrow_a, (16 * f + 8 * lh) * 2));
+        sc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[f], sc,
+                                                     0, 0, 0);
+        bf16x8 da = *(const bf16x8*)(
+            (char*)(do_swz + (long)qrow_a * D) +
+            swz(qrow_a, (16 * f + 8 * lh) * 2));
+        dpc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vfrag[f], dpc,
+                                                      0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qr = 32 * qs2 + (r & 3) + 8 * (r >> 2) + 4 * lh;
+        const int qrow = qt + qr;
+        bool valid = (qrow < sq) && kv_in_range;
+        if (CAUSAL) valid = valid && (kvcol <= qrow + (sk - sq));
+        float pp = valid ? __expf(sc[r] * scale - lse_lds[qr]) : 0.f;
+        dpc[r] = pp * (dpc[r] - dr_lds[qr]) * scale;
+      }
+      unsigned dswds[2][4];
+      SWAP_ASSEMBLE(dpc, dswds);
+#pragma unroll
+      for (int dsub = 0; dsub < NDSUB; ++dsub) {
+        short4v r8[4];
 #pragma unroll
         for (int K = 0; K < 2; ++K)
 #pragma unroll
@@ -1463,7 +1613,6 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
         }
       }
     }
-
     __syncthreads();
     if (it + 1 < n_it) {
       stage_write();
@@ -1471,21 +1620,17 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dkv2_kernel(
     }
   }
 
-  // epilogue: direct coalesced b16 stores (lanes hold consecutive d)
 #pragma unroll
   for (int dsub = 0; dsub < NDSUB; ++dsub)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kvrow = kv0w + (r & 3) + 8 * (r >> 2) + 4 * lh;
-      const int d_ = 32 * dsub + ln;
-      if (kvrow < sk) {
-        dk[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + d_] =
-            f2bf(dkacc[dsub][r]);
-        dv[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + d_] =
-            f2bf(dvacc[dsub][r]);
-      }
+      if (kvrow < sk)
+        dk[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + 32 * dsub +
+           ln] = f2bf(dkacc[dsub][r]);
     }
 }
+
 
 
 __global__ void attn_bwd_pre_kernel(const unsigned short* __restrict__ do_,
@@ -1513,9 +1658,12 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
   }
   const size_t lds_dq = (size_t)(2 * KVBLK * 128 + KVBLK * VRS +
                                  8 * 32 * OSTRIDE) * sizeof(unsigned short);
-  const size_t lds_dkv = (size_t)(2 * QT2 * 128 + 2 * QT2 * VRS) *
-                             sizeof(unsigned short) +
-                         2 * QT2 * sizeof(float);
+  const size_t lds_dv = (size_t)(QT2 * 128 + QT2 * VRS) *
+                            sizeof(unsigned short) +
+                        QT2 * sizeof(float);
+  const size_t lds_dk = (size_t)(2 * QT2 * 128 + QT2 * VRS) *
+                            sizeof(unsigned short) +
+                        2 * QT2 * sizeof(float);
 #define ATT_BWD2_LAUNCH(CC)                                                   \
   do {                                                                        \
     hipLaunchKernelGGL((attn_bwd_dq2_kernel<128, CC>),                        \
@@ -1527,13 +1675,21 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
                        vstr[0], vstr[1], vstr[2]);                            \
     HIP_CHECK_LAUNCH();                                                       \
-    hipLaunchKernelGGL((attn_bwd_dkv2_kernel<128, CC>),                       \
-                       dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dkv,   \
+    hipLaunchKernelGGL((attn_bwd_dv2_kernel<128, CC>),                        \
+                       dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dv,    \
+                       stream, (const unsigned short*)q,                      \
+                       (const unsigned short*)k,                              \
+                       (const unsigned short*)dout, lse,                      \
+                       (unsigned short*)dv, sq, sk, b, nh, ng, scale,         \
+                       qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],           \
+                       kstr[2]);                                              \
+    HIP_CHECK_LAUNCH();                                                       \
+    hipLaunchKernelGGL((attn_bwd_dk2_kernel<128, CC>),                        \
+                       dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dk,    \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k, (const unsigned short*)v,    \
                        (const unsigned short*)dout, lse, drow,                \
-                       (unsigned short*)dk, (unsigned short*)dv, sq, sk, b,   \
-                       nh, ng, scale,                                         \
+                       (unsigned short*)dk, sq, sk, b, nh, ng, scale,         \
                        qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
                        vstr[0], vstr[1], vstr[2]);                            \
     HIP_CHECK_LAUNCH();                                                       \
