@@ -62,24 +62,26 @@ class CLIPViTModel(MegatronModule):
             (class_token_len if add_class_token else 0)
 
         eps = transformer_config.layernorm_epsilon
+        pdt = transformer_config.params_dtype
         self.ln_pre = self.ln_post = None
         if model_subtype == "clip":
-            self.ln_pre = torch.nn.LayerNorm(h, eps=eps)
+            self.ln_pre = torch.nn.LayerNorm(h, eps=eps, dtype=pdt)
             conv_bias = False
         else:  # siglip
-            self.ln_post = torch.nn.LayerNorm(h, eps=eps)
+            self.ln_post = torch.nn.LayerNorm(h, eps=eps, dtype=pdt)
             conv_bias = True
 
         self.conv1 = torch.nn.Conv2d(
             in_channels=3, out_channels=h, kernel_size=patch_dim,
-            stride=patch_dim, bias=conv_bias)
-        self.position_embeddings = torch.nn.Embedding(self.seq_length, h)
+            stride=patch_dim, bias=conv_bias, dtype=pdt)
+        self.position_embeddings = torch.nn.Embedding(self.seq_length, h,
+                                                      dtype=pdt)
         self.register_buffer(
             "position_ids",
             torch.arange(self.seq_length).unsqueeze(0), persistent=False)
         if add_class_token:
             self.class_token = torch.nn.Parameter(
-                torch.randn(1, class_token_len, h))
+                torch.randn(1, class_token_len, h, dtype=pdt))
 
         self.model_type = ModelType.encoder_or_decoder
         self.decoder = TransformerBlock(
@@ -91,7 +93,7 @@ class CLIPViTModel(MegatronModule):
 
     def forward(self, x: torch.Tensor,
                 attention_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
-        x = self.conv1(x)                        # [b, h, gh, gw]
+        x = self.conv1(x.to(self.conv1.weight.dtype))  # [b, h, gh, gw]
         x = x.flatten(2).permute(0, 2, 1)        # [b, patches, h]
         if self.add_class_token:
             cls = self.class_token.expand(x.shape[0], -1, -1)

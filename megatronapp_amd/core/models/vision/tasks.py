@@ -23,8 +23,9 @@ from .clip_vit import CLIPViTModel
 class VitMlpHead(MegatronModule):
     def __init__(self, config, hidden_size: int, num_classes: int):
         super().__init__(config=config)
-        self.dense_in = torch.nn.Linear(hidden_size, hidden_size)
-        self.dense_out = torch.nn.Linear(hidden_size, num_classes)
+        pdt = config.params_dtype
+        self.dense_in = torch.nn.Linear(hidden_size, hidden_size, dtype=pdt)
+        self.dense_out = torch.nn.Linear(hidden_size, num_classes, dtype=pdt)
         torch.nn.init.constant_(self.dense_out.bias, -10)
 
     def forward(self, hidden_states):
@@ -44,7 +45,8 @@ class VitClassificationModel(MegatronModule):
             config, transformer_layer_spec,
             patch_dim=patch_dim, img_h=img_h, img_w=img_w)
         if finetune:
-            self.head = torch.nn.Linear(config.hidden_size, num_classes)
+            self.head = torch.nn.Linear(config.hidden_size, num_classes,
+                                        dtype=config.params_dtype)
             torch.nn.init.zeros_(self.head.weight)
         else:
             self.head = VitMlpHead(config, config.hidden_size, num_classes)
@@ -72,7 +74,8 @@ class VitInpaintingModel(MegatronModule):
             img_w=img_w, model_subtype="siglip")
         self.flatten_dim = 3 * patch_dim * patch_dim
         self.linear_decoder = torch.nn.Linear(
-            config.hidden_size, self.flatten_dim)
+            config.hidden_size, self.flatten_dim,
+            dtype=config.params_dtype)
         torch.nn.init.zeros_(self.linear_decoder.weight)
 
     def set_input_tensor(self, input_tensor):
