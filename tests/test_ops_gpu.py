@@ -609,3 +609,36 @@ def test_attn_bwd2_matches_ref(sq, gqa, causal):
         err = (got.float() - ref).abs().max().item()
         rel = err / (ref.abs().max().item() + 1e-6)
         assert rel < 5e-2, f"{name} max err {err} rel {rel}"
+
+
+@pytest.mark.gpu
+def test_grouped_mlp_grouped_mm_matches_loop():
+    """GroupedMLP's single-launch torch._grouped_mm path (fwd + custom
+    autograd bwd) vs the segmented per-expert loop."""
+    from megatronapp_amd.core.transformer.moe.experts import (
+        GroupedMLP, _GroupedMMFn)
+    torch.manual_seed(41)
+    E, h, f, T = 4, 256, 512, 1024
+    x = torch.randn(T, h, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = (torch.randn(E, h, f, device="cuda", dtype=torch.bfloat16) *
+         0.05).requires_grad_(True)
+    sizes = torch.tensor([300, 200, 324, 200], device="cuda")
+    offs = torch.cumsum(sizes, 0).to(torch.int32)
+    y = _GroupedMMFn.apply(x, w, offs)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    outs, start = [], 0
+    for e in range(E):
+        n = int(sizes[e])
+        outs.append(xr[start:start + n] @ wr[e])
+        start += n
+    yr = torch.cat(outs)
+    yr.backward(dy.float())
+    assert torch.allclose(y.float(), yr, atol=1e-1, rtol=5e-2)
+    for got, ref, name in ((x.grad, xr.grad, "dx"), (w.grad, wr.grad, "dw")):
+        rel = (got.float() - ref).abs().max() / (ref.abs().max() + 1e-6)
+        assert rel < 5e-2, (name, float(rel))
