@@ -16,10 +16,15 @@ from ...core.enums import AttnMaskType
 from ... import ops as _ops
 
 
+def _sk_ok(t):
+    # the wave softmax kernels vectorize rows as short8: sk % 8 == 0
+    return t.shape[-1] % 8 == 0
+
+
 class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, inputs, scale):
-        if _ops.fused_enabled(inputs, "softmax"):
+        if _ops.fused_enabled(inputs, "softmax") and _sk_ok(inputs):
             probs = _ops.get_ops().scaled_upper_triang_masked_softmax_fwd(
                 inputs, scale)
         else:
@@ -36,7 +41,7 @@ class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (probs,) = ctx.saved_tensors
-        if _ops.fused_enabled(dy, "softmax"):
+        if _ops.fused_enabled(dy, "softmax") and _sk_ok(dy):
             # causal-aware bwd: only the valid row prefix is read
             dx = _ops.get_ops().scaled_upper_triang_masked_softmax_bwd(
                 dy.contiguous(), probs, ctx.scale)
@@ -50,7 +55,7 @@ class ScaledUpperTriangMaskedSoftmax(torch.autograd.Function):
 class ScaledMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, inputs, mask, scale):
-        if _ops.fused_enabled(inputs, "softmax"):
+        if _ops.fused_enabled(inputs, "softmax") and _sk_ok(inputs):
             probs = _ops.get_ops().scaled_masked_softmax_fwd(inputs, mask, scale)
         else:
             x = inputs.float() * scale
@@ -64,7 +69,7 @@ class ScaledMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (probs,) = ctx.saved_tensors
-        if _ops.fused_enabled(dy, "softmax"):
+        if _ops.fused_enabled(dy, "softmax") and _sk_ok(dy):
             dx = _ops.get_ops().scaled_softmax_bwd(dy.contiguous(), probs, ctx.scale)
         else:
             dyf = dy.float()
