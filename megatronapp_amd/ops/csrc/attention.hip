@@ -687,7 +687,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
     float* __restrict__ lse, int sq, int sk, int b, int nh, int ng,
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
     long vS, long vB, long vH) {
-  static_assert(D == 128, "attn_fwd2 is specialized for head dim 128");
+  static_assert(D == 128 || D == 64, "attn_fwd2: head dim 64/128");
   const int qtile = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / nh;
@@ -963,26 +963,26 @@ void launch_attn_fwd2(const void* q, const void* k, const void* v, void* o,
   if (sq % FQBLK2 != 0 || sk % KVBLK != 0)
     throw std::runtime_error(
         "attn_fwd2: sq must be a multiple of 256, sk of 64");
-  if (d != 128) throw std::runtime_error("attn_fwd2: d must be 128");
+  if (d != 128 && d != 64)
+    throw std::runtime_error("attn_fwd2: d must be 64 or 128");
   dim3 grid(sq / FQBLK2, b * nh);
   dim3 block(FWD_BLOCK);
   const size_t lds =
       (size_t)(KVBLK * d + KVBLK * VRS + 8 * 32 * OSTRIDE) *
       sizeof(unsigned short);
-  if (causal)
-    hipLaunchKernelGGL((attn_fwd2_kernel<128, true>), grid, block, lds,
-                       stream, (const unsigned short*)q,
-                       (const unsigned short*)k, (const unsigned short*)v,
-                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale,
-                       qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],
-                       vstr[0], vstr[1], vstr[2]);
-  else
-    hipLaunchKernelGGL((attn_fwd2_kernel<128, false>), grid, block, lds,
-                       stream, (const unsigned short*)q,
-                       (const unsigned short*)k, (const unsigned short*)v,
-                       (unsigned short*)o, lse, sq, sk, b, nh, ng, scale,
-                       qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],
-                       vstr[0], vstr[1], vstr[2]);
+#define FWD2_LAUNCH(DD, CC)                                                 \
+  hipLaunchKernelGGL((attn_fwd2_kernel<DD, CC>), grid, block, lds,          \
+                     stream, (const unsigned short*)q,                      \
+                     (const unsigned short*)k, (const unsigned short*)v,    \
+                     (unsigned short*)o, lse, sq, sk, b, nh, ng, scale,     \
+                     qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
+                     vstr[0], vstr[1], vstr[2])
+  if (d == 128) {
+    if (causal) FWD2_LAUNCH(128, true); else FWD2_LAUNCH(128, false);
+  } else {
+    if (causal) FWD2_LAUNCH(64, true); else FWD2_LAUNCH(64, false);
+  }
+#undef FWD2_LAUNCH
   HIP_CHECK_LAUNCH();
 }
 
@@ -1036,7 +1036,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
     unsigned short* __restrict__ dq, int sq, int sk, int b, int nh, int ng,
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
     long vS, long vB, long vH) {
-  static_assert(D == 128, "dq2 is specialized for head dim 128");
+  static_assert(D == 128 || D == 64, "dq2: head dim 64/128");
   const int qtile = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / nh;
@@ -1260,7 +1260,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dv2_kernel(
     const float* __restrict__ lse,
     unsigned short* __restrict__ dv, int sq, int sk, int b, int nh, int ng,
     float scale, long qS, long qB, long qH, long kS, long kB, long kH) {
-  static_assert(D == 128, "dv2 is specialized for head dim 128");
+  static_assert(D == 128 || D == 64, "dv2: head dim 64/128");
   const int kvtile = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / ng;
@@ -1439,7 +1439,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dk2_kernel(
     unsigned short* __restrict__ dk, int sq, int sk, int b, int nh, int ng,
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
     long vS, long vB, long vH) {
-  static_assert(D == 128, "dk2 is specialized for head dim 128");
+  static_assert(D == 128 || D == 64, "dk2: head dim 64/128");
   const int kvtile = blockIdx.x;
   const int bh = blockIdx.y;
   const int bi = bh / ng;
@@ -1644,8 +1644,9 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                       int sk, int b, int nh, int ng, int d, float scale,
                       bool causal, const long* qstr, const long* kstr,
                       const long* vstr, hipStream_t stream) {
-  if (sq % FQBLK2 != 0 || sk % FQBLK2 != 0 || d != 128)
-    throw std::runtime_error("attn_bwd2: sq/sk must be multiples of 256, d=128");
+  if (sq % FQBLK2 != 0 || sk % FQBLK2 != 0 || (d != 128 && d != 64))
+    throw std::runtime_error(
+        "attn_bwd2: sq/sk must be multiples of 256, d 64/128");
   {
     const long rows = (long)sq * b * nh;
     const int waves_per_block = ATT_BLOCK / WAVE;
@@ -1664,9 +1665,9 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
   const size_t lds_dk = (size_t)(2 * QT2 * 128 + QT2 * VRS) *
                             sizeof(unsigned short) +
                         2 * QT2 * sizeof(float);
-#define ATT_BWD2_LAUNCH(CC)                                                   \
+#define ATT_BWD2_LAUNCH(DD, CC)                                                   \
   do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dq2_kernel<128, CC>),                        \
+    hipLaunchKernelGGL((attn_bwd_dq2_kernel<DD, CC>),                        \
                        dim3(sq / FQBLK2, b * nh), dim3(FWD_BLOCK), lds_dq,    \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k, (const unsigned short*)v,    \
@@ -1675,7 +1676,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
                        vstr[0], vstr[1], vstr[2]);                            \
     HIP_CHECK_LAUNCH();                                                       \
-    hipLaunchKernelGGL((attn_bwd_dv2_kernel<128, CC>),                        \
+    hipLaunchKernelGGL((attn_bwd_dv2_kernel<DD, CC>),                        \
                        dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dv,    \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k,                              \
@@ -1684,7 +1685,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],           \
                        kstr[2]);                                              \
     HIP_CHECK_LAUNCH();                                                       \
-    hipLaunchKernelGGL((attn_bwd_dk2_kernel<128, CC>),                        \
+    hipLaunchKernelGGL((attn_bwd_dk2_kernel<DD, CC>),                        \
                        dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dk,    \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k, (const unsigned short*)v,    \
@@ -1694,8 +1695,13 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        vstr[0], vstr[1], vstr[2]);                            \
     HIP_CHECK_LAUNCH();                                                       \
   } while (0)
-  if (causal) ATT_BWD2_LAUNCH(true);
-  else ATT_BWD2_LAUNCH(false);
+  if (d == 128) {
+    if (causal) ATT_BWD2_LAUNCH(128, true);
+    else ATT_BWD2_LAUNCH(128, false);
+  } else {
+    if (causal) ATT_BWD2_LAUNCH(64, true);
+    else ATT_BWD2_LAUNCH(64, false);
+  }
 #undef ATT_BWD2_LAUNCH
 }
 

@@ -379,7 +379,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   auto o = torch::empty({sq, b, nh, d}, q.options());
   auto lse = torch::empty({b, nh, sq}, q.options().dtype(torch::kFloat32));
   long qs[3], ks[3], vs[3];
-  if (d == 128 && sq % 256 == 0 && sk % 64 == 0 && attn_strides(q, qs) &&
+  if ((d == 128 || d == 64) && sq % 256 == 0 && sk % 64 == 0 &&
+      attn_strides(q, qs) &&
       attn_strides(k, ks) && attn_strides(v, vs)) {
     launch_attn_fwd2(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                      lse.data_ptr<float>(), sq, sk, b, nh, ng, d,
@@ -448,7 +449,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                            q.options().dtype(torch::kFloat32));
   long qs[3], ks[3], vs[3];
   auto doc = dout.contiguous();
-  if (d == 128 && sq % 256 == 0 && sk % 256 == 0 && attn_strides(q, qs) &&
+  if ((d == 128 || d == 64) && sq % 256 == 0 && sk % 256 == 0 &&
+      attn_strides(q, qs) &&
       attn_strides(k, ks) && attn_strides(v, vs)) {
     launch_attn_bwd2(doc.data_ptr(), q.data_ptr(), k.data_ptr(),
                      v.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),

@@ -558,15 +558,19 @@ def test_attn_fwd_transposed_matches_ref(sq, gqa):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("sq,gqa,causal", [(256, 1, True), (256, 4, True),
-                                           (2048, 1, True), (512, 1, False),
-                                           (2048, 4, True)])
-def test_attn_fwd2_matches_ref(sq, gqa, causal):
+@pytest.mark.parametrize("sq,gqa,causal,d", [(256, 1, True, 128),
+                                             (256, 4, True, 128),
+                                             (2048, 1, True, 128),
+                                             (512, 1, False, 128),
+                                             (2048, 4, True, 128),
+                                             (256, 1, True, 64),
+                                             (2048, 2, True, 64)])
+def test_attn_fwd2_matches_ref(sq, gqa, causal, d):
     """Round-2 32x32-MFMA forward (swapped QK^T, tr16 V reads, in-register
     P^T) vs the fp32 reference."""
     ops = _ops()
     torch.manual_seed(29)
-    b, nh, d = 2, 8, 128
+    b, nh = 2, 8
     ng = nh // gqa
     q = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
     k = torch.randn(sq, b, ng, d, device="cuda", dtype=torch.bfloat16)
@@ -582,15 +586,19 @@ def test_attn_fwd2_matches_ref(sq, gqa, causal):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("sq,gqa,causal", [(256, 1, True), (256, 4, True),
-                                           (512, 1, False), (2048, 1, True),
-                                           (2048, 4, True)])
-def test_attn_bwd2_matches_ref(sq, gqa, causal):
-    """Round-2 backward (dq2 + dkv2 kernels; dispatched by attn_bwd for
-    sq,sk multiples of 256 at d=128) vs autograd through the fp32 ref."""
+@pytest.mark.parametrize("sq,gqa,causal,d", [(256, 1, True, 128),
+                                             (256, 4, True, 128),
+                                             (512, 1, False, 128),
+                                             (2048, 1, True, 128),
+                                             (2048, 4, True, 128),
+                                             (256, 1, True, 64),
+                                             (2048, 2, True, 64)])
+def test_attn_bwd2_matches_ref(sq, gqa, causal, d):
+    """Round-2 backward (dq2 + dv2/dk2 kernels; dispatched by attn_bwd
+    for sq,sk multiples of 256 at d in {64,128}) vs fp32 autograd."""
     ops = _ops()
     torch.manual_seed(31)
-    b, nh, d = 2, 8, 128
+    b, nh = 2, 8
     ng = nh // gqa
     scale = d ** -0.5
     q = torch.randn(sq, b, nh, d, device="cuda", dtype=torch.bfloat16)
