@@ -89,8 +89,10 @@ def main():
         dist.init_process_group(backend=backend, rank=rank,
                                 world_size=world_size)
     if on_gpu:
+        # modulo so oversubscribed runs (4 ranks on 1 GPU) stay valid
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
-    device = torch.device("cuda", local_rank) if on_gpu else torch.device("cpu")
+    device = (torch.device("cuda", local_rank % torch.cuda.device_count())
+              if on_gpu else torch.device("cpu"))
 
     n = world_size
     if args.pp is None:
