@@ -1,17 +1,21 @@
 """Sharded checkpoint save/load with cross-topology resharding.
 
 Reference: core/dist_checkpointing/serialization.py + strategies
-(torch_dist default, fully_parallel dedup, resharding).  Layout:
+(torch_dist default, fully_parallel dedup, resharding).
 
-  <dir>/shards_rank{R:05d}.pt   {key: {"offset", "global_shape", "tensor"}}
-  <dir>/index.json              {key: [{file, offset, shape, global_shape}]}
-  <dir>/common.pt               non-sharded objects (args, iteration, rng)
+Default on-disk format (round 2): **torch-DCP** — ``.metadata`` +
+``__N_M.distcp`` data files written through torch.distributed.checkpoint
+(see torch_dcp.py), byte-compatible with the reference's ``torch_dist``
+strategy and loadable by upstream tooling.  Non-sharded objects live in
+``common.pt`` (same as the reference).
+
+The round-1 private layout (``shards_rank*.pt`` + ``index.json``) is
+still READ transparently for old checkpoints; new saves are DCP-only.
 
 Save: every rank persists the shards whose ``replica_id == 0`` (the
 fully-parallel dedup — each unique shard written exactly once).
-Load: each rank scans the index for pieces overlapping its local shard's
-(offset, shape) window and copies the intersecting slices — so a
-checkpoint saved at TP=a, PP=b loads at TP=a', PP=b' unchanged.
+Load: each rank fills its local (offset, shape) window from whatever
+sharding the checkpoint was saved at (cross-topology reshard).
 """
 
 from __future__ import annotations
@@ -32,6 +36,13 @@ def _rank() -> int:
 
 def save(sharded_state_dict: Dict[str, ShardedTensor], ckpt_dir: str,
          common_state: dict = None) -> None:
+    """Write a torch-DCP sharded checkpoint (upstream-compatible)."""
+    from .torch_dcp import save_dcp
+    save_dcp(sharded_state_dict, ckpt_dir, common_state=common_state)
+
+
+def _save_legacy(sharded_state_dict: Dict[str, ShardedTensor],
+                 ckpt_dir: str, common_state: dict = None) -> None:
     os.makedirs(ckpt_dir, exist_ok=True)
     rank = _rank()
     mine = {k: st for k, st in sharded_state_dict.items()
@@ -76,9 +87,24 @@ def load_common(ckpt_dir: str) -> dict:
 
 def load(sharded_state_dict: Dict[str, ShardedTensor], ckpt_dir: str,
          strict: bool = True) -> Dict[str, "torch.Tensor"]:
-    """Fill each local shard in-place from overlapping saved pieces and
-    return {key: filled tensor} (reference API: feedable to
-    ``module.load_state_dict``)."""
+    """Fill each local shard in-place (from a torch-DCP checkpoint, or a
+    round-1 legacy index.json one) and return {key: filled tensor}
+    (reference API: feedable to ``module.load_state_dict``)."""
+    from .torch_dcp import is_dcp_checkpoint, load_dcp
+    if is_dcp_checkpoint(ckpt_dir):
+        sd = sharded_state_dict
+        if not strict:
+            import torch.distributed.checkpoint as _dcp
+            md = _dcp.FileSystemReader(ckpt_dir).read_metadata()
+            sd = {k: st for k, st in sd.items()
+                  if k in md.state_dict_metadata}
+        load_dcp(sd, ckpt_dir)
+        return {k: st.data for k, st in sd.items()}
+    return _load_legacy(sharded_state_dict, ckpt_dir, strict)
+
+
+def _load_legacy(sharded_state_dict: Dict[str, ShardedTensor],
+                 ckpt_dir: str, strict: bool = True):
     with open(os.path.join(ckpt_dir, "index.json")) as f:
         index = json.load(f)
     cache: Dict[str, dict] = {}

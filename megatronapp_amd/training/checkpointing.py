@@ -199,10 +199,12 @@ def load_checkpoint(model: List, optimizer, opt_param_scheduler,
             print(f"  no checkpoint tracker at {tracker}, starting fresh")
         return 0, 0
     iteration, release = read_metadata(tracker)
-    # sharded (torch_dist) checkpoints are recognised by their index file
+    # sharded (torch_dist) checkpoints: torch-DCP (.metadata) or the
+    # round-1 legacy index.json layout
     base = get_checkpoint_name(load_dir, iteration, release,
                                return_base_dir=True)
-    if os.path.exists(os.path.join(base, "index.json")):
+    if os.path.exists(os.path.join(base, ".metadata")) or \
+            os.path.exists(os.path.join(base, "index.json")):
         return _load_checkpoint_torch_dist(
             model, optimizer, opt_param_scheduler, base, iteration, strict)
     name = get_checkpoint_name(load_dir, iteration, release)

@@ -82,7 +82,9 @@ def test_checkpoint_converter(tmp_path):
     w = torch.randn(8, 6)
     # fake a TP=2 save: two shards along dim 0, written in one process
     sd0 = {"model.w": ShardedTensor("model.w", w[:4].clone(), (8, 6), (0, 0))}
-    dist_save(sd0, str(src), common_state={"iteration": 7})
+    from megatronapp_amd.core.dist_checkpointing.serialization import \
+        _save_legacy
+    _save_legacy(sd0, str(src), common_state={"iteration": 7})
     # second shard appended by hand (single-process test)
     import json, os
     shard = {"model.w": {"offset": (4, 0), "global_shape": (8, 6),
@@ -149,7 +151,7 @@ def test_async_save_torch_dist(tmp_path):
     save = str(tmp_path / "ckd")
     _run(["--train-iters", "3", "--save", save, "--save-interval", "3",
           "--ckpt-format", "torch_dist", "--async-save"], 29643)
-    assert os.path.exists(os.path.join(save, "iter_0000003", "index.json"))
+    assert os.path.exists(os.path.join(save, "iter_0000003", ".metadata"))
     out = _run(["--train-iters", "5", "--save", save, "--load", save,
                 "--ckpt-format", "torch_dist", "--save-interval", "100"],
                29644)

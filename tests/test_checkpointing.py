@@ -68,7 +68,7 @@ def test_save_and_resume_torch_dist(tmp_path):
     save = str(tmp_path / "ckpt_dist")
     out1 = _run(["--train-iters", "3", "--save", save, "--save-interval", "3",
                  "--ckpt-format", "torch_dist"], 29633)
-    assert os.path.exists(os.path.join(save, "iter_0000003", "index.json"))
+    assert os.path.exists(os.path.join(save, "iter_0000003", ".metadata"))
     out2 = _run(["--train-iters", "5", "--save", save, "--load", save,
                  "--ckpt-format", "torch_dist", "--save-interval", "100"],
                 29634)
@@ -142,12 +142,11 @@ def test_torch_dist_save_resume_pp2(tmp_path):
                          capture_output=True, text=True, cwd=REPO,
                          timeout=420)
     assert out.returncode == 0, out.stderr[-3000:]
-    idx = os.path.join(save, "iter_0000003", "index.json")
-    assert os.path.exists(idx)
-    import json
-    index = json.load(open(idx))
-    files = {m["file"] for metas in index.values() for m in metas}
-    assert len(files) == 2, files            # both stages wrote shards
+    base_dir = os.path.join(save, "iter_0000003")
+    assert os.path.exists(os.path.join(base_dir, ".metadata"))
+    # both PP stages wrote DCP shard files
+    distcp = [f for f in os.listdir(base_dir) if f.endswith(".distcp")]
+    assert len(distcp) == 2, distcp
     out2 = subprocess.run(base_args + ["--train-iters", "5", "--load", save,
                                        "--save-interval", "100"],
                           capture_output=True, text=True, cwd=REPO,

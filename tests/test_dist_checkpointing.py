@@ -174,3 +174,41 @@ def test_reshard_property_random_grids(tmp_path):
                     out, full[r0:r0 + rows // br, c0:c0 + cols // bc])
 
     check()
+
+
+def test_torch_dcp_interchange(tmp_path):
+    """The sharded format IS torch-DCP: plain torch.distributed.checkpoint
+    loads our checkpoints, and we load checkpoints produced by plain DCP
+    (what reference/upstream tooling writes)."""
+    import torch.distributed.checkpoint as dcp
+    from megatronapp_amd.core.dist_checkpointing.mapping import ShardedTensor
+    from megatronapp_amd.core.dist_checkpointing.torch_dcp import (
+        is_dcp_checkpoint, load_dcp, load_dcp_consolidated, save_dcp)
+    initialize_model_parallel()
+    d = str(tmp_path / "ours")
+    t1 = torch.randn(8, 16)
+    t2 = torch.arange(24.0).view(4, 6)
+    sd = {"model.a.weight": ShardedTensor("model.a.weight", t1, (8, 16),
+                                          (0, 0), 0),
+          "model.b.bias": ShardedTensor("model.b.bias", t2, (4, 6),
+                                        (0, 0), 0)}
+    save_dcp(sd, d, common_state={"iteration": 7})
+    assert is_dcp_checkpoint(d)
+    # upstream direction: vanilla DCP reads our checkpoint
+    target = {"model.a.weight": torch.empty(8, 16),
+              "model.b.bias": torch.empty(4, 6)}
+    dcp.load(target, storage_reader=dcp.FileSystemReader(d))
+    assert torch.equal(target["model.a.weight"], t1)
+    assert torch.equal(target["model.b.bias"], t2)
+    # import direction: we read a vanilla-DCP (reference-style) checkpoint
+    d2 = str(tmp_path / "theirs")
+    dcp.save({"x": t1}, storage_writer=dcp.FileSystemWriter(d2))
+    full = load_dcp_consolidated(d2)
+    assert torch.equal(full["x"], t1)
+    # resharded load: two half-windows of our own checkpoint
+    h1, h2 = torch.empty(4, 16), torch.empty(4, 16)
+    load_dcp({"model.a.weight": ShardedTensor("model.a.weight", h1,
+                                              (8, 16), (0, 0), 0)}, d)
+    load_dcp({"model.a.weight": ShardedTensor("model.a.weight", h2,
+                                              (8, 16), (4, 0), 0)}, d)
+    assert torch.equal(torch.cat([h1, h2]), t1)

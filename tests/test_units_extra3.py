@@ -114,8 +114,10 @@ def test_converter_legacy_loader(tmp_path):
                         "--saver", "torch_dist"],
                        capture_output=True, text=True)
     assert r.returncode == 0, r.stderr
-    idx = json.load(open(tmp_path / "out" / "index.json"))
-    assert "model.w" in idx and "model.b" in idx
+    from megatronapp_amd.core.dist_checkpointing.torch_dcp import (
+        load_dcp_consolidated)
+    full = load_dcp_consolidated(str(tmp_path / "out"))
+    assert "model.w" in full and "model.b" in full
 
 
 def test_selective_recompute_grads_exact():

@@ -40,7 +40,19 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 
 def load_torch_dist(path):
-    """Merge a sharded checkpoint's pieces into full tensors."""
+    """Merge a sharded checkpoint's pieces into full tensors.  Reads the
+    torch-DCP layout (.metadata + __N_M.distcp — ours or
+    reference/upstream-produced) and the round-1 legacy index.json."""
+    if os.path.exists(os.path.join(path, ".metadata")):
+        from megatronapp_amd.core.dist_checkpointing.torch_dcp import (
+            load_dcp_consolidated)
+        full = load_dcp_consolidated(path)
+        common = {}
+        cpath = os.path.join(path, "common.pt")
+        if os.path.exists(cpath):
+            common = torch.load(cpath, map_location="cpu",
+                                weights_only=False)
+        return full, common
     with open(os.path.join(path, "index.json")) as f:
         index = json.load(f)
     cache = {}
@@ -104,15 +116,11 @@ def load_consolidated(path):
 
 def save_torch_dist(full, common, path):
     os.makedirs(path, exist_ok=True)
-    payload = {k: {"offset": [0] * t.dim(), "global_shape": list(t.shape),
-                   "tensor": t} for k, t in full.items()}
-    fname = "shards_rank00000.pt"
-    torch.save(payload, os.path.join(path, fname))
-    index = {k: [{"file": fname, "offset": [0] * t.dim(),
-                  "shape": list(t.shape), "global_shape": list(t.shape)}]
-             for k, t in full.items()}
-    with open(os.path.join(path, "index.json"), "w") as f:
-        json.dump(index, f)
+    from megatronapp_amd.core.dist_checkpointing.mapping import ShardedTensor
+    from megatronapp_amd.core.dist_checkpointing.torch_dcp import save_dcp
+    sd = {k: ShardedTensor(k, t, tuple(t.shape), (0,) * t.dim(), 0)
+          for k, t in full.items()}
+    save_dcp(sd, path)
     torch.save(common, os.path.join(path, "common.pt"))
     # tracker so load_checkpoint finds it when pointed at the parent dir
     parent, leaf = os.path.split(os.path.normpath(path))
