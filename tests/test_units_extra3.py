@@ -326,3 +326,48 @@ def test_preprocess_data_cli(tmp_path):
     assert r.returncode == 0, r.stderr
     assert os.path.exists(prefix + "_text_document.bin") or \
         os.path.exists(prefix + ".bin"), os.listdir(tmp_path)
+
+
+def test_legacy_loader_merges_tp_pp_shards(tmp_path, monkeypatch):
+    """Reference-style legacy layout at TP=2 PP=2: qkv/fc1 merge on dim 0,
+    proj/fc2 on dim 1, norms replicated, PP layer indices globalized."""
+    import subprocess, sys
+    h = 8
+    def shard(tp, pp):
+        sd = {"model": {}}
+        li = 0  # local layer index
+        sd["model"][f"decoder.layers.{li}.self_attention.linear_qkv.weight"] = \
+            torch.full((3 * h // 2, h), float(tp + 10 * pp))
+        sd["model"][f"decoder.layers.{li}.self_attention.linear_proj.weight"] = \
+            torch.full((h, h // 2), float(tp + 10 * pp))
+        sd["model"][f"decoder.layers.{li}.input_layernorm.weight"] = \
+            torch.full((h,), float(pp))
+        if pp == 0:
+            sd["model"]["embedding.word_embeddings.weight"] = \
+                torch.full((16 // 2, h), float(tp))
+        sd["iteration"] = 5
+        return sd
+    base = tmp_path / "iter_0000005"
+    for tp in range(2):
+        for pp in range(2):
+            d = base / f"mp_rank_{tp:02d}_{pp:03d}"
+            os.makedirs(d)
+            torch.save(shard(tp, pp), d / "model_optim_rng.pt")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.join(repo, "tools", "checkpoint"))
+    try:
+        import importlib
+        import convert as conv
+        importlib.reload(conv)
+        full, common = conv.load_legacy(str(base))
+    finally:
+        sys.path.pop(0)
+    qkv0 = full["model.decoder.layers.0.self_attention.linear_qkv.weight"]
+    assert qkv0.shape == (3 * h, h)
+    assert qkv0[0, 0] == 0 and qkv0[-1, 0] == 1          # tp merge dim 0
+    proj1 = full["model.decoder.layers.1.self_attention.linear_proj.weight"]
+    assert proj1.shape == (h, h)
+    assert proj1[0, 0] == 10 and proj1[0, -1] == 11      # tp merge dim 1
+    assert full["model.decoder.layers.1.input_layernorm.weight"][0] == 1
+    assert full["model.embedding.word_embeddings.weight"].shape == (16, h)
+    assert common.get("iteration") == 5

@@ -31,6 +31,7 @@ Typical uses:
 import argparse
 import json
 import os
+import re
 import sys
 
 import torch
@@ -85,25 +86,87 @@ def load_torch_dist(path):
     return full, common
 
 
+# TP shard axis by parameter name (Megatron-core GPT family), for
+# merging reference-produced legacy checkpoints saved at TP>1:
+#   dim 0 (column-parallel outputs / vocab rows): qkv, fc1, embeddings,
+#          output layer; dim 1 (row-parallel inputs): proj, fc2.
+#   replicated: norms and all biases of row-parallel layers.
+def _legacy_tp_axis(name):
+    if name.endswith("bias"):
+        if any(t in name for t in ("linear_qkv", "linear_fc1", "linear_q",
+                                   "linear_kv")):
+            return 0
+        return None
+    if any(t in name for t in ("word_embeddings.weight",
+                               "output_layer.weight",
+                               "linear_qkv.weight", "linear_fc1.weight",
+                               "linear_q_proj", "linear_q_down",
+                               "linear_kv_down")):
+        return 0
+    if any(t in name for t in ("linear_proj.weight", "linear_fc2.weight")):
+        return 1
+    return None  # norms, rotary inv_freq, router, etc. are replicated
+
+
+def _legacy_rank_dirs(path):
+    """{(tp, pp): dir} from mp_rank_XX[_YYY] directory names."""
+    out = {}
+    for d in os.listdir(path):
+        if not d.startswith("mp_rank_"):
+            continue
+        parts = d[len("mp_rank_"):].split("_")
+        tp = int(parts[0])
+        pp = int(parts[1]) if len(parts) > 1 else 0
+        out[(tp, pp)] = d
+    return out
+
+
 def load_legacy(path):
-    """mp_rank_00/model_optim_rng.pt (TP=PP=1 checkpoints only)."""
-    ranks = sorted(d for d in os.listdir(path) if d.startswith("mp_rank_"))
-    if ranks != ["mp_rank_00"]:
-        raise SystemExit(
-            f"legacy loader handles TP=PP=1 checkpoints; found {ranks}. "
-            f"Re-save the run with --ckpt-format torch_dist and use "
-            f"--loader torch_dist instead.")
-    sd = torch.load(os.path.join(path, "mp_rank_00", "model_optim_rng.pt"),
-                    map_location="cpu", weights_only=False)
-    model_keys = [k for k in sd if k == "model" or k.startswith("model")]
+    """mp_rank_XX[_YYY]/model_optim_rng.pt — reference legacy layout at
+    any TP (name-based shard-axis merge) and any PP (disjoint layer
+    namespaces merged with globalized layer indices)."""
+    ranks = _legacy_rank_dirs(path)
+    if not ranks:
+        raise SystemExit(f"no mp_rank_* dirs under {path}")
+    tps = sorted({t for t, _ in ranks})
+    pps = sorted({p for _, p in ranks})
+
+    def load_rank(tp, pp):
+        return torch.load(os.path.join(path, ranks[(tp, pp)],
+                                       "model_optim_rng.pt"),
+                          map_location="cpu", weights_only=False)
+
     full = {}
-    for mk in model_keys:
-        prefix = "model." if mk == "model" else f"{mk}."
-        for name, t in sd[mk].items():
-            if torch.is_tensor(t):
-                full[prefix + name] = t
-    common = {k: v for k, v in sd.items() if k not in model_keys
-              and k not in ("optimizer", "rng_state")}
+    common = {}
+    layer_base = 0
+    for pp in pps:
+        shards = [load_rank(tp, pp) for tp in tps]
+        sd0 = shards[0]
+        model_keys = [k for k in sd0 if k == "model" or
+                      (k.startswith("model") and isinstance(sd0[k], dict))]
+        n_local_layers = 0
+        for mk in model_keys:
+            prefix = "model." if mk == "model" else f"{mk}."
+            for name, t in sd0[mk].items():
+                if not torch.is_tensor(t):
+                    continue
+                m = re.search(r"layers\.(\d+)\.", name)
+                if m:
+                    n_local_layers = max(n_local_layers, int(m.group(1)) + 1)
+                gname = prefix + (re.sub(
+                    r"layers\.(\d+)\.",
+                    lambda mm: f"layers.{int(mm.group(1)) + layer_base}.",
+                    name) if m else name)
+                axis = _legacy_tp_axis(name)
+                if axis is None or len(tps) == 1:
+                    full[gname] = t
+                else:
+                    full[gname] = torch.cat(
+                        [sh[mk][name] for sh in shards], dim=axis)
+        layer_base += n_local_layers
+        if pp == pps[0]:
+            common = {k: v for k, v in sd0.items() if k not in model_keys
+                      and k not in ("optimizer", "rng_state")}
     return full, common
 
 
