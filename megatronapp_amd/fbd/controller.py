@@ -91,6 +91,16 @@ class Controller:
         self.table = ReadinessTable(world_size)
         self._thread = None
         self._stop = threading.Event()
+        # server-rank-local posts can't dist.send to self (gloo
+        # self-send deadlocks): they go straight into the table under a
+        # lock and wait on an event the serve thread also fires.
+        self._lock = threading.Lock()
+        self._local_events = {}
+
+    def _fire_local(self, key):
+        ev = self._local_events.get(key)
+        if ev is not None:
+            ev.set()
 
     def start_server(self):
         if dist.get_rank() != self.server_rank:
@@ -101,7 +111,9 @@ class Controller:
     def _serve(self):
         msg = torch.zeros(2 + self.world, dtype=torch.int64)
         while not self._stop.is_set():
-            # any rank may post; receive one message at a time
+            # blocking any-source recv (gloo irecv completion polling
+            # never fires); shutdown arrives as a kind=2 sentinel from a
+            # designated non-server rank
             try:
                 dist.recv(msg, group=self.group)
             except RuntimeError:
@@ -113,13 +125,15 @@ class Controller:
                 return
             if kind == 0:
                 group = tuple(i for i in range(self.world) if bits[i])
-                if self.table.post_collective(src, group):
+                with self._lock:
+                    fired = self.table.post_collective(src, group)
+                if fired:
                     go = torch.ones(1, dtype=torch.int64)
                     for m in group:
                         if m == self.server_rank:
+                            self._fire_local(frozenset(group))
                             continue
                         dist.send(go, dst=m, group=self.group)
-                    self._local_go = True
             else:
                 dsts = [i for i in range(self.world) if bits[i]]
                 for a, b in self.table.post_p2p(src, dsts):
@@ -130,22 +144,43 @@ class Controller:
 
     def gate_collective(self, group_ranks: List[int]):
         """Block until the controller clears this collective."""
+        rank = dist.get_rank()
+        if rank == self.server_rank:
+            key = frozenset(group_ranks)
+            with self._lock:
+                fired = self.table.post_collective(rank, group_ranks)
+                if not fired:
+                    ev = self._local_events.setdefault(key,
+                                                       threading.Event())
+                    ev.clear()
+            if fired:
+                go = torch.ones(1, dtype=torch.int64)
+                for m in group_ranks:
+                    if m != self.server_rank:
+                        dist.send(go, dst=m, group=self.group)
+            else:
+                self._local_events[key].wait()
+            return
         msg = torch.zeros(2 + self.world, dtype=torch.int64)
         msg[0] = 0
-        msg[1] = dist.get_rank()
+        msg[1] = rank
         for r in group_ranks:
             msg[2 + r] = 1
         dist.send(msg, dst=self.server_rank, group=self.group)
-        if dist.get_rank() != self.server_rank:
-            go = torch.zeros(1, dtype=torch.int64)
-            dist.recv(go, src=self.server_rank, group=self.group)
+        go = torch.zeros(1, dtype=torch.int64)
+        dist.recv(go, src=self.server_rank, group=self.group)
 
     def shutdown(self):
-        self._stop.set()
-        if dist.get_rank() != self.server_rank and False:
-            pass
-        if self._thread is not None:
+        """Call on every rank.  The highest non-server rank sends the
+        kind=2 sentinel (gloo cannot self-send); the server joins its
+        thread after the sentinel lands."""
+        rank = dist.get_rank()
+        sentinel_rank = self.world - 1 if self.world - 1 != self.server_rank \
+            else self.world - 2
+        if rank == sentinel_rank and sentinel_rank >= 0:
             msg = torch.zeros(2 + self.world, dtype=torch.int64)
             msg[0] = 2
             dist.send(msg, dst=self.server_rank, group=self.group)
+        self._stop.set()
+        if self._thread is not None:
             self._thread.join(timeout=5)

@@ -28,15 +28,23 @@ from ..core.pipeline_parallel.schedules import (
 from ..core.utils import get_model_config
 
 
-def _dual_send(tensor: torch.Tensor, dual_rank: int):
-    dist.send(tensor.detach().contiguous(), dst=dual_rank)
+def _dual_isend(tensor: torch.Tensor, dual_rank: int, pending: list):
+    """Async ship of a stage input to the dual backward rank; the
+    forward pipe keeps running ahead instead of blocking per microbatch
+    (round-1 used blocking dist.send)."""
+    t = tensor.detach().contiguous()
+    pending.append((dist.isend(t, dst=dual_rank), t))
+    # bound the in-flight queue so activations don't pile up unboundedly
+    while len(pending) > 4:
+        req, _ = pending.pop(0)
+        req.wait()
 
 
-def _dual_recv(shape, dtype, dual_rank: int):
+def _dual_irecv(shape, dtype, dual_rank: int):
     device = torch.cuda.current_device() if torch.cuda.is_available() else "cpu"
     t = torch.empty(shape, dtype=dtype, device=device, requires_grad=True)
-    dist.recv(t, src=dual_rank)
-    return t
+    req = dist.irecv(t, src=dual_rank)
+    return req, t
 
 
 def forward_backward_disaggregated(
@@ -65,6 +73,7 @@ def forward_backward_disaggregated(
 
     if parallel_state.is_forward_stage():
         # ---------------- forward instance: no autograd, run ahead -------
+        pending: list = []
         with torch.no_grad():
             for i in range(num_microbatches):
                 input_tensor = p2p_communication.recv_forward(
@@ -77,7 +86,9 @@ def forward_backward_disaggregated(
                 # ship the stage input to the dual backward instance
                 # (first stage reads tokens from its own iterator instead)
                 if not is_first:
-                    _dual_send(input_tensor, dual)
+                    _dual_isend(input_tensor, dual, pending)
+        for req, _ in pending:
+            req.wait()
         return forward_data_store
 
     # -------------------- backward instance: recompute + backward --------
@@ -86,11 +97,19 @@ def forward_backward_disaggregated(
         contextlib.nullcontext()
     no_sync.__enter__()
     sync_open = True
+    # prefetch the first shipped input; post the next irecv before
+    # computing so transfer overlaps the recompute+backward
+    next_req = None
+    if not is_first:
+        next_req = _dual_irecv(tensor_shape, dtype, dual)
     for i in range(num_microbatches):
         if is_first:
             input_tensor = None
         else:
-            input_tensor = _dual_recv(tensor_shape, dtype, dual)
+            req, input_tensor = next_req
+            req.wait()
+            if i + 1 < num_microbatches:
+                next_req = _dual_irecv(tensor_shape, dtype, dual)
         output_tensor, _ = forward_step(
             forward_step_func, data_iterator, model, num_microbatches,
             input_tensor, forward_data_store, config, collect_non_loss_data,
