@@ -275,6 +275,15 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
 def linear_with_grad_accumulation_and_async_allreduce(
         input, weight, bias, gradient_accumulation_fusion,
         async_grad_allreduce, sequence_parallel):
+    # fp8 serving fast path: weights carry e4m3 data + per-row scales
+    # (inference/fp8.py quantize_model_fp8); 2x MFMA peak on gfx950
+    if (not torch.is_grad_enabled() and input.is_cuda
+            and hasattr(weight, "fp8_data")):
+        from ...inference.fp8 import fp8_linear
+        out = fp8_linear(input, weight)
+        if bias is not None:
+            out = out + bias
+        return out
     return LinearWithGradAccumulationAndAsyncCommunication.apply(
         input, weight, bias, gradient_accumulation_fusion,
         async_grad_allreduce, sequence_parallel)

@@ -12,12 +12,29 @@ import torch
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
+def _set_fp8(model, enabled):
+    """Toggle the serving fast path by hiding/restoring fp8 attrs."""
+    for mod in model.modules():
+        w = getattr(mod, "weight", None)
+        if w is None:
+            continue
+        if enabled and hasattr(w, "_fp8_data_stash"):
+            w.fp8_data, w.fp8_scale = w._fp8_data_stash
+            del w._fp8_data_stash
+        elif not enabled and hasattr(w, "fp8_data"):
+            w._fp8_data_stash = (w.fp8_data, w.fp8_scale)
+            del w.fp8_data
+            del w.fp8_scale
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="gpt3-1.3b")
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--prompt", type=int, default=128)
     ap.add_argument("--new", type=int, default=128)
+    ap.add_argument("--fp8", action="store_true",
+                    help="also time the fp8 (e4m3 _scaled_mm) weights path")
     args = ap.parse_args()
 
     import bench
@@ -62,7 +79,16 @@ def main():
                for i in range(args.batch)]
     sp = SamplingParams(num_tokens_to_generate=args.new, top_k=1)
 
-    for use_graphs, label in ((False, "eager"), (True, "hipgraph")):
+    modes = [(False, False, "eager"), (True, False, "hipgraph")]
+    if args.fp8:
+        from megatronapp_amd.inference.fp8 import quantize_model_fp8
+        n = quantize_model_fp8(model)
+        print(f"fp8: quantized {n} linears (e4m3, per-row scales)")
+        modes.append((False, True, "eager+fp8"))
+        modes.append((True, True, "hipgraph+fp8"))
+    ref_text = None
+    for use_graphs, use_fp8, label in modes:
+        _set_fp8(model, use_fp8)
         engine = get_inference_engine(model, tok, max_batch_size=args.batch)
         engine.controller.use_hip_graphs = use_graphs
         engine.generate(prompts[:2], SamplingParams(num_tokens_to_generate=8,
@@ -73,8 +99,14 @@ def main():
         torch.cuda.synchronize()
         dt = time.perf_counter() - t0
         tps = args.batch * args.new / dt
-        print(f"{label:9s}: {dt:.2f}s  {tps:8.1f} tokens/s "
-              f"({dt / args.new * 1000:.2f} ms/step) sample={out[0].generated_text[:40]!r}")
+        txt = out[0].generated_text
+        match = "" if ref_text is None else \
+            f" match={'Y' if txt == ref_text else 'n'}"
+        if ref_text is None:
+            ref_text = txt
+        print(f"{label:13s}: {dt:.2f}s  {tps:8.1f} tokens/s "
+              f"({dt / args.new * 1000:.2f} ms/step){match} "
+              f"sample={txt[:40]!r}")
 
 
 if __name__ == "__main__":

@@ -650,3 +650,57 @@ def test_grouped_mlp_grouped_mm_matches_loop():
     for got, ref, name in ((x.grad, xr.grad, "dx"), (w.grad, wr.grad, "dw")):
         rel = (got.float() - ref).abs().max() / (ref.abs().max() + 1e-6)
         assert rel < 5e-2, (name, float(rel))
+
+
+@pytest.mark.gpu
+def test_fp8_linear_matches_bf16():
+    """fp8 serving fast path (e4m3 _scaled_mm, per-row weight scales +
+    per-token act scales) vs the bf16 linear."""
+    from megatronapp_amd.inference.fp8 import (_quantize_weight, fp8_linear)
+    torch.manual_seed(51)
+    M, K, N = 64, 2048, 4096
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    w = torch.nn.Parameter(
+        torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.02)
+    ref = x @ w.t()
+    w.fp8_data, w.fp8_scale = _quantize_weight(w.data)
+    y = fp8_linear(x, w)
+    rel = (y.float() - ref.float()).norm() / ref.float().norm()
+    assert rel < 0.02, float(rel)
+
+
+@pytest.mark.gpu
+def test_fp8_model_decode_close_to_bf16():
+    """Whole-model fp8 serving: logits stay close and greedy tokens
+    mostly agree with bf16 on a small GPT."""
+    from tests.utils import initialize_model_parallel
+    initialize_model_parallel()
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    from megatronapp_amd.core.tensor_parallel.random import (
+        model_parallel_cuda_manual_seed)
+    from megatronapp_amd.inference.fp8 import (dequantize_model_fp8,
+                                               quantize_model_fp8)
+    model_parallel_cuda_manual_seed(7)
+    cfg = TransformerConfig(
+        num_layers=4, hidden_size=256, num_attention_heads=4,
+        ffn_hidden_size=512, hidden_dropout=0.0, attention_dropout=0.0,
+        bf16=True, params_dtype=torch.bfloat16)
+    with torch.device("cuda"):
+        m = GPTModel(config=cfg,
+                     transformer_layer_spec=get_gpt_layer_local_spec(),
+                     vocab_size=512, max_sequence_length=64,
+                     pre_process=True, post_process=True).eval()
+    toks = torch.randint(0, 512, (2, 64), device="cuda")
+    pos = torch.arange(64, device="cuda").expand(2, 64)
+    mask = None
+    with torch.no_grad():
+        ref = m(toks, pos, mask).float()
+        n = quantize_model_fp8(m)
+        assert n > 0
+        got = m(toks, pos, mask).float()
+        dequantize_model_fp8(m)
+    agree = (ref.argmax(-1) == got.argmax(-1)).float().mean()
+    assert agree > 0.9, float(agree)
