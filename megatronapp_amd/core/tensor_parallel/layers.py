@@ -281,9 +281,10 @@ def linear_with_grad_accumulation_and_async_allreduce(
             and hasattr(weight, "fp8_data")):
         from ...inference.fp8 import fp8_linear
         out = fp8_linear(input, weight)
-        if bias is not None:
-            out = out + bias
-        return out
+        if out is not None:    # None: below the fp8 token threshold
+            if bias is not None:
+                out = out + bias
+            return out
     return LinearWithGradAccumulationAndAsyncCommunication.apply(
         input, weight, bias, gradient_accumulation_fusion,
         async_grad_allreduce, sequence_parallel)

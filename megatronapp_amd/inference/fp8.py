@@ -18,6 +18,13 @@ import torch
 
 FP8_MAX = 448.0  # e4m3fn max normal
 
+# Below this many tokens the GEMM is weight-read/latency bound and the
+# dynamic activation-quant kernels cost more than fp8 saves (measured on
+# MI355X: batch-32 decode 2.2k tok/s fp8 vs 3.8k bf16, while prefill-
+# sized GEMMs hit 2249 TF vs 1199).  fp8 therefore engages for prefill
+# and any batched forward >= this row count; decode stays bf16.
+FP8_MIN_TOKENS = 256
+
 
 def _quantize_weight(w: torch.Tensor):
     """Per-output-row symmetric quantization to e4m3."""
@@ -37,6 +44,8 @@ def fp8_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     ws = weight.fp8_scale
     shp = x.shape
     x2 = x.reshape(-1, shp[-1])
+    if x2.shape[0] < FP8_MIN_TOKENS:
+        return None  # caller falls back to the bf16 path
     xs = x2.abs().amax(dim=1, keepdim=True).float() / FP8_MAX
     xs = torch.clamp(xs, min=1e-12)
     x8 = (x2.float() / xs).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
