@@ -15,3 +15,7 @@ class SamplingParams:
     add_BOS: bool = False
     termination_id: int = -1
     prevent_newline_after_colon: bool = False
+    # beam search (reference text_generation_server.py:243-267): > 0
+    # switches generation to beam decoding with this many hypotheses
+    beam_width: int = 0
+    length_penalty: float = 1.0

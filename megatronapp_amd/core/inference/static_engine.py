@@ -33,6 +33,22 @@ class StaticInferenceEngine:
                  report_step=None) -> List[InferenceRequest]:
         sampling_params = sampling_params or SamplingParams()
         requests: List[InferenceRequest] = []
+        if sampling_params.beam_width > 0:
+            # beam decoding runs one prompt at a time (hypotheses take
+            # the batch dim)
+            for prompt in prompts:
+                toks = self.controller.tokenize_prompts(
+                    [prompt], sampling_params.add_BOS)[0]
+                best, score, _ = self.controller.generate_beam_search(
+                    toks, sampling_params)
+                gen = torch.tensor(best[len(toks):])
+                req = InferenceRequest(
+                    prompt=prompt, prompt_tokens=toks,
+                    generated_tokens=gen, generated_log_probs=None)
+                req.generated_text = self.controller.detokenize(gen)
+                req.score = score
+                requests.append(req)
+            return requests
         for i in range(0, len(prompts), self.max_batch_size):
             chunk = prompts[i:i + self.max_batch_size]
             toks = self.controller.tokenize_prompts(chunk,
@@ -65,11 +81,13 @@ def get_inference_engine(model, tokenizer, max_batch_size=8):
 
 def run_mcore_engine(engine: StaticInferenceEngine, prompts: List[str],
                      temperature=1.0, top_k=0, top_p=0.0,
-                     logprobs=False, tokens_to_generate=64, report_step=None):
+                     logprobs=False, tokens_to_generate=64, report_step=None,
+                     beam_width=0, length_penalty=1.0):
     """REST/WS-facing wrapper (reference mcore_engine_server.py)."""
     sp = SamplingParams(num_tokens_to_generate=tokens_to_generate,
                         temperature=temperature, top_k=top_k, top_p=top_p,
-                        return_log_probs=logprobs)
+                        return_log_probs=logprobs, beam_width=beam_width,
+                        length_penalty=length_penalty)
     reqs = engine.generate(prompts, sp, report_step=report_step)
     return {
         "text": [r.prompt + r.generated_text for r in reqs],

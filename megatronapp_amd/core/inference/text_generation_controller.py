@@ -147,5 +147,80 @@ class TextGenerationController:
             logprobs = torch.stack(logprobs, dim=1)
         return batch, logprobs
 
+    # ------------------------------------------------------------ beam search
+    @torch.no_grad()
+    def generate_beam_search(self, prompt_tokens: List[int],
+                             sampling: SamplingParams):
+        """Single-prompt beam search (reference
+        text_generation_server.py:243-267 / beam_search API): maintains
+        ``beam_width`` hypotheses in the batch dim, reorders the KV cache
+        on every step, scores finished hypotheses with a GNMT-style
+        length penalty, and returns (best_tokens, best_score,
+        all_beams)."""
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+        beams = sampling.beam_width
+        assert beams > 0
+        eos = sampling.termination_id if sampling.termination_id >= 0 \
+            else getattr(self.tokenizer, "eod", -1)
+        plen = len(prompt_tokens)
+        total = plen + sampling.num_tokens_to_generate
+        inference_params = InferenceParams(beams, total)
+
+        tokens = torch.tensor(prompt_tokens, device=device).unsqueeze(0) \
+            .expand(beams, plen).contiguous()
+        scores = torch.full((beams,), float("-inf"), device=device)
+        scores[0] = 0.0          # all beams identical at start: keep one
+        finished: list = []      # (score, tokens list)
+
+        def lp(length: int) -> float:
+            return ((5.0 + length) / 6.0) ** sampling.length_penalty
+
+        step_tokens = tokens
+        pos = 0
+        for step in range(sampling.num_tokens_to_generate):
+            cur_len = step_tokens.shape[1]
+            position_ids = torch.arange(
+                pos, pos + cur_len, device=device).unsqueeze(0).expand(
+                    beams, -1)
+            logits = self.model(step_tokens, position_ids,
+                                inference_context=inference_params)
+            inference_params.increment_sequence_len_offset(cur_len)
+            pos += cur_len
+            last = logits[:, -1, :]
+            if parallel_state.get_tensor_model_parallel_world_size() > 1:
+                last = gather_from_tensor_model_parallel_region(last)
+            logprobs = torch.log_softmax(last.float(), dim=-1)   # [beams, v]
+            v = logprobs.shape[-1]
+            cand = scores.unsqueeze(-1) + logprobs               # [beams, v]
+            top_scores, top_idx = cand.view(-1).topk(beams)
+            beam_idx = top_idx // v
+            tok_idx = top_idx % v
+            inference_params.reorder_batch(beam_idx)
+            tokens = torch.cat(
+                [tokens.index_select(0, beam_idx),
+                 tok_idx.unsqueeze(-1)], dim=1)
+            scores = top_scores
+            # retire finished hypotheses, keep the rest going
+            if eos >= 0:
+                done = tok_idx == eos
+                for i in torch.nonzero(done).flatten().tolist():
+                    gen_len = tokens.shape[1] - plen
+                    finished.append((float(scores[i]) / lp(gen_len),
+                                     tokens[i].tolist()))
+                    scores[i] = float("-inf")
+                if len(finished) >= beams or bool(
+                        torch.isinf(scores).all()):
+                    break
+            step_tokens = tok_idx.unsqueeze(-1)
+
+        for i in range(beams):
+            if not torch.isinf(scores[i]):
+                gen_len = tokens.shape[1] - plen
+                finished.append((float(scores[i]) / lp(gen_len),
+                                 tokens[i].tolist()))
+        finished.sort(key=lambda x: -x[0])
+        best_score, best_tokens = finished[0]
+        return best_tokens, best_score, finished
+
     def detokenize(self, tokens_row: torch.Tensor) -> str:
         return self.tokenizer.detokenize([int(t) for t in tokens_row])

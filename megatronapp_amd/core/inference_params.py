@@ -46,6 +46,14 @@ class InferenceParams:
     def increment_sequence_len_offset(self, n: int):
         self.sequence_len_offset += n
 
+    def reorder_batch(self, indices: torch.Tensor):
+        """Permute the cache's batch dim (beam-search hypothesis
+        reordering): cache[:, i] <- cache[:, indices[i]]."""
+        end = self.sequence_len_offset
+        for layer, (k, v) in self.key_value_memory_dict.items():
+            k[:end] = k[:end].index_select(1, indices)
+            v[:end] = v[:end].index_select(1, indices)
+
 
 # alias matching newer reference naming
 class StaticInferenceContext(InferenceParams):

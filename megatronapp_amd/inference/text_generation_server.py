@@ -102,7 +102,9 @@ class InferenceGenerate:
             top_p=float(request.get("top_p", 0.0)),
             logprobs=bool(request.get("logprobs", False)),
             tokens_to_generate=int(request.get("tokens_to_generate", 64)),
-            report_step=report_step if websocket is not None else None)
+            report_step=report_step if websocket is not None else None,
+            beam_width=int(request.get("beam_width", 0)),
+            length_penalty=float(request.get("length_penalty", 1.0)))
         tt.tik_end()
         if websocket is not None:
             websocket.send({"type": "finish", **result})

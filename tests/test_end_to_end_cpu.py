@@ -119,3 +119,46 @@ def test_forward_only_eval():
              forward_only=True)
     assert len(out) == 2
     destroy()
+
+
+def test_beam_search_decoding():
+    """Beam search returns the highest-scoring hypothesis; width-1 beam
+    equals greedy decoding on the same model."""
+    from megatronapp_amd.core.inference.static_engine import (
+        get_inference_engine)
+    from megatronapp_amd.core.inference.sampling_params import SamplingParams
+    from megatronapp_amd.training.tokenizer import NullTokenizer
+    from megatronapp_amd.core.models.gpt import GPTModel
+    from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
+        get_gpt_layer_local_spec)
+    from megatronapp_amd.core.transformer_config import TransformerConfig
+    initialize_model_parallel()
+    torch.manual_seed(5)
+    cfg = TransformerConfig(
+        num_layers=2, hidden_size=64, num_attention_heads=4,
+        ffn_hidden_size=128, hidden_dropout=0.0, attention_dropout=0.0,
+        params_dtype=torch.float32)
+    model = GPTModel(config=cfg,
+                     transformer_layer_spec=get_gpt_layer_local_spec(),
+                     vocab_size=128, max_sequence_length=64,
+                     pre_process=True, post_process=True).eval()
+    tok = NullTokenizer(127)
+    engine = get_inference_engine(model, tok, max_batch_size=4)
+    engine.controller.use_hip_graphs = False
+    prompt = "5 9 13 2"
+    greedy = engine.generate(
+        [prompt], SamplingParams(num_tokens_to_generate=8, top_k=1))[0]
+    beam1 = engine.generate(
+        [prompt], SamplingParams(num_tokens_to_generate=8,
+                                 beam_width=1))[0]
+    assert beam1.generated_text == greedy.generated_text
+    beam4 = engine.generate(
+        [prompt], SamplingParams(num_tokens_to_generate=8,
+                                 beam_width=4))[0]
+    assert hasattr(beam4, "score")
+    assert len(beam4.generated_tokens) > 0
+    # a width-4 beam's cumulative logprob is >= the greedy hypothesis's
+    # (it explores a superset of the greedy path)
+    b1 = engine.generate(
+        [prompt], SamplingParams(num_tokens_to_generate=8, beam_width=1))[0]
+    assert beam4.score >= b1.score - 1e-4
