@@ -51,6 +51,12 @@ struct SlotHeader {
 struct ChannelHeader {
   int nslots;
   long slot_bytes;  // payload bytes per slot
+  // device fast path: the creator hipMalloc's nslots*slot_bytes of GPU
+  // memory and publishes its hipIpcMemHandle here; the opener maps it
+  // and GPU tensors move device-to-device (xGMI peer / same-GPU copy)
+  // while the control plane (tags + semaphores) stays in this shm.
+  int device_ok;
+  unsigned char ipc_handle[64];  // sizeof(hipIpcMemHandle_t)
 };
 
 struct Channel {
@@ -59,12 +65,16 @@ struct Channel {
   char* base{};
   size_t total_bytes{};
   bool creator{};
+  char* dev_base{};   // mapped device payload area (or null)
 
   SlotHeader* slot(int i) {
     return (SlotHeader*)(base + sizeof(ChannelHeader) +
                          (size_t)i * (sizeof(SlotHeader) + hdr->slot_bytes));
   }
   char* payload(int i) { return (char*)slot(i) + sizeof(SlotHeader); }
+  char* dev_payload(int i) {
+    return dev_base ? dev_base + (size_t)i * hdr->slot_bytes : nullptr;
+  }
 };
 
 std::map<std::string, Channel>& channels() {
@@ -84,10 +94,15 @@ Channel& get_channel(const std::string& dir, int src, int dst) {
   return it->second;
 }
 
-void copy_in(char* dst, const torch::Tensor& t) {
+void copy_in(char* dst, char* dev_dst, const torch::Tensor& t) {
   const long n = t.numel() * t.element_size();
 #ifdef __HIP_PLATFORM_AMD__
   if (t.is_cuda()) {
+    if (dev_dst) {
+      auto st = hipMemcpy(dev_dst, t.data_ptr(), n, hipMemcpyDeviceToDevice);
+      TORCH_CHECK(st == hipSuccess, "hipMemcpy D2D failed");
+      return;
+    }
     auto st = hipMemcpy(dst, t.data_ptr(), n, hipMemcpyDeviceToHost);
     TORCH_CHECK(st == hipSuccess, "hipMemcpy D2H failed");
     return;
@@ -96,9 +111,16 @@ void copy_in(char* dst, const torch::Tensor& t) {
   std::memcpy(dst, t.data_ptr(), n);
 }
 
-void copy_out(torch::Tensor& t, const char* src, long n) {
+void copy_out(torch::Tensor& t, const char* src, const char* dev_src,
+              long n) {
 #ifdef __HIP_PLATFORM_AMD__
   if (t.is_cuda()) {
+    if (dev_src) {
+      auto st = hipMemcpy(t.data_ptr(), dev_src, n,
+                          hipMemcpyDeviceToDevice);
+      TORCH_CHECK(st == hipSuccess, "hipMemcpy D2D failed");
+      return;
+    }
     auto st = hipMemcpy(t.data_ptr(), src, n, hipMemcpyHostToDevice);
     TORCH_CHECK(st == hipSuccess, "hipMemcpy H2D failed");
     return;
@@ -112,7 +134,7 @@ void copy_out(torch::Tensor& t, const char* src, long n) {
 // Create (or open) a channel.  Exactly one side passes create=true and the
 // pair must agree on nslots/slot_bytes.
 void init_channel(const std::string& dir, int src, int dst, long slot_bytes,
-                  int nslots, bool create) {
+                  int nslots, bool create, bool use_device) {
   const std::string name = chan_name(dir, src, dst);
   if (channels().count(name)) return;
   const size_t total = sizeof(ChannelHeader) +
@@ -151,6 +173,23 @@ void init_channel(const std::string& dir, int src, int dst, long slot_bytes,
   ch.creator = create;
   if (create) {
     ch.hdr->slot_bytes = slot_bytes;
+    ch.hdr->device_ok = 0;
+#ifdef __HIP_PLATFORM_AMD__
+    if (use_device) {
+      void* dptr = nullptr;
+      if (hipMalloc(&dptr, (size_t)nslots * slot_bytes) == hipSuccess) {
+        hipIpcMemHandle_t h;
+        if (hipIpcGetMemHandle(&h, dptr) == hipSuccess) {
+          static_assert(sizeof(h) <= 64, "ipc handle too large");
+          std::memcpy(ch.hdr->ipc_handle, &h, sizeof(h));
+          ch.dev_base = (char*)dptr;
+          ch.hdr->device_ok = 1;
+        } else {
+          hipFree(dptr);
+        }
+      }
+    }
+#endif
     for (int i = 0; i < nslots; ++i) {
       SlotHeader* s = (SlotHeader*)(ch.base + sizeof(ChannelHeader) +
                                     (size_t)i * (sizeof(SlotHeader) + slot_bytes));
@@ -168,6 +207,16 @@ void init_channel(const std::string& dir, int src, int dst, long slot_bytes,
     for (int tries = 0; tries < 6000 && ch.hdr->nslots == 0; ++tries)
       usleep(10000);
     TORCH_CHECK(ch.hdr->nslots == nslots, "channel slot mismatch");
+#ifdef __HIP_PLATFORM_AMD__
+    if (use_device && ch.hdr->device_ok) {
+      hipIpcMemHandle_t h;
+      std::memcpy(&h, ch.hdr->ipc_handle, sizeof(h));
+      void* dptr = nullptr;
+      if (hipIpcOpenMemHandle(&dptr, h, hipIpcMemLazyEnablePeerAccess) ==
+          hipSuccess)
+        ch.dev_base = (char*)dptr;
+    }
+#endif
   }
   channels().emplace(name, ch);
 }
@@ -187,7 +236,7 @@ void put_tensor(const std::string& dir, int src, int dst, int chunk,
     for (int i = 0; i < ch.hdr->nslots; ++i) {
       SlotHeader* s = ch.slot(i);
       if (sem_trywait(&s->sem_free) == 0) {
-        copy_in(ch.payload(i), t);
+        copy_in(ch.payload(i), ch.dev_payload(i), t);
         s->chunk = chunk;
         s->microbatch = microbatch;
         s->nbytes = n;
@@ -215,7 +264,7 @@ void get_tensor(const std::string& dir, int src, int dst, int chunk,
             const long n = s->nbytes;
             TORCH_CHECK(out.numel() * out.element_size() == n,
                         "DPP get size mismatch");
-            copy_out(out, ch.payload(i), n);
+            copy_out(out, ch.payload(i), ch.dev_payload(i), n);
             s->state = 0;
             s->chunk = -1;
             s->microbatch = -1;
@@ -245,6 +294,12 @@ bool probe_tensor(const std::string& dir, int src, int dst, int chunk,
 
 void clean_channels() {
   for (auto& [name, ch] : channels()) {
+#ifdef __HIP_PLATFORM_AMD__
+    if (ch.dev_base) {
+      if (ch.creator) hipFree(ch.dev_base);
+      else hipIpcCloseMemHandle(ch.dev_base);
+    }
+#endif
     munmap(ch.base, ch.total_bytes);
     if (ch.creator) shm_unlink(name.c_str());
   }
@@ -255,7 +310,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("init_channel", &init_channel, "create/open a DPP shm channel",
         pybind11::arg("dir"), pybind11::arg("src"), pybind11::arg("dst"),
         pybind11::arg("slot_bytes"), pybind11::arg("nslots"),
-        pybind11::arg("create"));
+        pybind11::arg("create"), pybind11::arg("use_device") = true);
   // GIL released during blocking copies/spins so Python worker threads
   // (the DPP policy scheduler) keep running
   m.def("put_tensor", &put_tensor,
