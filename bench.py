@@ -69,6 +69,8 @@ def parse_args():
     p.add_argument("--trace", action="store_true", help="enable MegaScan tracing")
     p.add_argument("--trace-dir", default="trace_out")
     p.add_argument("--attention", default="flash", choices=["flash", "fused"])
+    p.add_argument("--fp8", action="store_true",
+                   help="fp8 (e4m3) forward+dgrad GEMMs, bf16 wgrad")
     p.add_argument("--seq-length", type=int, default=None,
                    help="override the model's sequence length")
     p.add_argument("--no-overlap-grad-reduce", action="store_true")
@@ -183,6 +185,11 @@ def main():
         overlap_grad_reduce=not args.no_overlap_grad_reduce,
         use_distributed_optimizer=(dp > 1),
         grad_reduce_in_fp32=True)
+    if args.fp8 and on_gpu:
+        from megatronapp_amd.core.fp8 import enable_fp8_training
+        n8 = enable_fp8_training(model)
+        if rank == 0:
+            print(f"fp8 training: {n8} linears flagged")
     model = DistributedDataParallel(config, ddp_config, model)
     optimizer = get_megatron_optimizer(
         OptimizerConfig(lr=1e-4, weight_decay=0.1, clip_grad=1.0, bf16=bf16,

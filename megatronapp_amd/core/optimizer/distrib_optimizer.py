@@ -214,6 +214,8 @@ class DistributedOptimizer:
     @torch.no_grad()
     def step(self):
         self.step_count += 1
+        from ..fp8 import bump_step
+        bump_step()   # invalidate per-step fp8 weight-quantization caches
         lr = self.param_groups[0]["lr"]
         wd = self.param_groups[0].get("weight_decay", self.weight_decay)
 

@@ -188,7 +188,11 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         else:
             total_input = input
         ctx.save_for_backward(input, weight)
-        if bias is not None:
+        from ..fp8 import fp8_forward, fp8_train_enabled
+        ctx.fp8 = fp8_train_enabled(weight, total_input)
+        if ctx.fp8:
+            output = fp8_forward(total_input, weight, bias)
+        elif bias is not None:
             # F.linear -> addmm: hipBLASLt fuses the bias in the GEMM
             # epilogue (a separate [s*b, out] add costs ~166 us/layer at
             # mbs16 on the QKV projection)
@@ -207,7 +211,11 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         else:
             total_input = input
 
-        grad_input = grad_output.matmul(weight)
+        if getattr(ctx, "fp8", False):
+            from ..fp8 import fp8_dgrad
+            grad_input = fp8_dgrad(grad_output, weight)
+        else:
+            grad_input = grad_output.matmul(weight)
 
         handle = None
         if ctx.sequence_parallel:

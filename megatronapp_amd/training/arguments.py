@@ -175,6 +175,10 @@ def _add_mixed_precision_args(p):
     g = p.add_argument_group("mixed precision")
     g.add_argument("--fp16", action="store_true")
     g.add_argument("--bf16", action="store_true")
+    g.add_argument("--fp8", default=None, choices=["e4m3", "hybrid"],
+                   help="fp8 GEMMs for forward+dgrad (bf16 wgrad); "
+                        "native hipBLASLt scaled-GEMM path (reference: "
+                        "TE --fp8-format)")
     g.add_argument("--loss-scale", type=float, default=None,
                    help="static fp16 loss scale (default: dynamic)")
     g.add_argument("--initial-loss-scale", type=float, default=2 ** 32)

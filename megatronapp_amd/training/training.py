@@ -80,6 +80,11 @@ def get_model(model_provider_func, model_type=ModelType.encoder_or_decoder,
                  f"{parallel_state.get_pipeline_model_parallel_rank()}): "
                  f"{num_params}")
 
+    if getattr(args, "fp8", None) and torch.cuda.is_available():
+        from ..core.fp8 import enable_fp8_training
+        n8 = sum(enable_fp8_training(chunk) for chunk in model)
+        print_rank_0(f" > fp8 training: {n8} linears flagged ({args.fp8})")
+
     if wrap_with_ddp:
         config = core_transformer_config_from_args(args)
         ddp_config = DistributedDataParallelConfig(

@@ -54,6 +54,8 @@ def main():
             torch.cuda.is_available()
             and os.environ.get("SANITY_FP32") != "1"):
         cmd.append("--bf16")
+    if os.environ.get("SANITY_FP8") == "1":
+        cmd.extend(["--fp8", "hybrid"])
     env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29673",
                RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
     out = subprocess.run(cmd, capture_output=True, text=True, env=env)

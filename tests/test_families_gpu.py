@@ -145,3 +145,14 @@ def test_fp16_training_step_gpu():
     ok, norm, _ = opt.step()
     assert ok and norm is not None and torch.isfinite(loss)
     destroy()
+
+
+@pytest.mark.gpu
+def test_fp8_training_learns_on_gpu():
+    """fp8 (e4m3 forward+dgrad, bf16 wgrad) end-to-end convergence on the
+    structured corpus — the fp8 analogue of the bf16 learning guard."""
+    import sys
+    r = subprocess.run([sys.executable, "scripts/train_sanity.py",
+                        "--iters", "120"], capture_output=True, text=True,
+                       timeout=400, env={**os.environ, "SANITY_FP8": "1"})
+    assert "LEARNING SANITY OK" in r.stdout, r.stdout[-1500:] + r.stderr[-800:]
