@@ -151,6 +151,8 @@ def test_fp16_training_step_gpu():
 def test_fp8_training_learns_on_gpu():
     """fp8 (e4m3 forward+dgrad, bf16 wgrad) end-to-end convergence on the
     structured corpus — the fp8 analogue of the bf16 learning guard."""
+    import os
+    import subprocess
     import sys
     r = subprocess.run([sys.executable, "scripts/train_sanity.py",
                         "--iters", "120"], capture_output=True, text=True,
