@@ -33,7 +33,13 @@ def bump_step() -> None:
 
 
 def _quant_rows(t2d: torch.Tensor):
-    """Per-row e4m3 quantization of a 2D tensor."""
+    """Per-row e4m3 quantization of a 2D tensor — one fused HIP kernel
+    on GPU (the eager 6-kernel version was 24% of fp8 step time)."""
+    if t2d.is_cuda and t2d.dtype == torch.bfloat16:
+        from .. import ops as _ops
+        if _ops.have_ops() and hasattr(_ops.get_ops(), "quantize_rows_e4m3"):
+            q, s = _ops.get_ops().quantize_rows_e4m3(t2d.contiguous())
+            return q, s
     s = t2d.abs().amax(dim=1, keepdim=True).float() / FP8_MAX
     s = torch.clamp(s, min=1e-12)
     q = (t2d.float() / s).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)

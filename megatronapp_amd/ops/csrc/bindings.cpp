@@ -56,6 +56,8 @@ torch::Tensor gemm_nn(torch::Tensor, torch::Tensor);
 void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
 void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
+void launch_quant_rows_e4m3(const void*, void*, float*, long, int,
+                            hipStream_t);
 void launch_attn_fwd2(const void*, const void*, const void*, void*, float*,
                       int, int, int, int, int, int, float, bool,
                       const long*, const long*, const long*, hipStream_t);
@@ -366,6 +368,20 @@ void embedding_bwd_accum(torch::Tensor dy, torch::Tensor tokens,
                              cur_stream());
 }
 
+// ---------------------------------------------------------------- fp8 quant
+std::vector<torch::Tensor> quantize_rows_e4m3(torch::Tensor x) {
+  check_bf16(x, "x");
+  auto x2 = x.reshape({-1, x.size(-1)}).contiguous();
+  const long M = x2.size(0);
+  const int K = (int)x2.size(1);
+  auto q = torch::empty({M, (long)K},
+                        x.options().dtype(torch::kFloat8_e4m3fn));
+  auto s = torch::empty({M, 1}, x.options().dtype(torch::kFloat32));
+  launch_quant_rows_e4m3(x2.data_ptr(), q.data_ptr(), s.data_ptr<float>(),
+                         M, K, cur_stream());
+  return {q, s};
+}
+
 // ---------------------------------------------------------------- attention
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, double scale,
@@ -623,6 +639,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("selective_scan_fwd", &selective_scan_fwd);
   mod.def("attn_fwd_t", &attn_fwd_t);
   mod.def("attn_fwd2", &attn_fwd2);
+  mod.def("quantize_rows_e4m3", &quantize_rows_e4m3);
   mod.def("attn_bwd_v1", &attn_bwd_v1);
   mod.def("ce_rowmax", &ce_rowmax);
   mod.def("gemm_nt", &gemm_nt);

@@ -22,7 +22,7 @@ SOURCES = [
     "elementwise.hip",
     "rope.hip",
     "softmax.hip",
-    "adam.hip", "scan.hip", "ce.hip",
+    "adam.hip", "scan.hip", "ce.hip", "fp8quant.hip",
     "wgrad.cpp",
     "attention.hip",
 ]

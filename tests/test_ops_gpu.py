@@ -708,3 +708,21 @@ def test_fp8_model_decode_close_to_bf16():
     # bar is rough agreement plus bounded logit error
     rel = (got - ref).norm() / ref.norm()
     assert agree > 0.6 and rel < 0.1, (float(agree), float(rel))
+
+
+@pytest.mark.gpu
+def test_quantize_rows_e4m3_matches_eager():
+    ops = _ops()
+    torch.manual_seed(61)
+    for M, K in ((128, 2048), (1000, 8192)):
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 3
+        q, s = ops.quantize_rows_e4m3(x)
+        s_ref = torch.clamp(x.abs().amax(dim=1, keepdim=True).float() / 448.0,
+                            min=1e-12)
+        q_ref = (x.float() / s_ref).clamp(-448, 448).to(torch.float8_e4m3fn)
+        assert torch.allclose(s, s_ref, rtol=1e-3)
+        # dequantized values must match the eager recipe closely
+        deq = q.float() * s
+        deq_ref = q_ref.float() * s_ref
+        err = (deq - deq_ref).abs().max()
+        assert err < 0.1, float(err)
