@@ -164,6 +164,120 @@ class SentencePieceTokenizer(MegatronTokenizer):
         return self._tok.eos_id()
 
 
+class BertWordPieceTokenizer(MegatronTokenizer):
+    """WordPiece over a BERT vocab.txt (reference tokenizer.py
+    BertWordPieceLowerCase / BertWordPieceCase)."""
+
+    def __init__(self, vocab_file: str, lower_case: bool = True):
+        from tokenizers import BertWordPieceTokenizer as _BWP
+        self._tok = _BWP(vocab_file, lowercase=lower_case)
+        v = self._tok.get_vocab()
+        self._cls = v.get("[CLS]", 0)
+        self._sep = v.get("[SEP]", 0)
+        self._pad = v.get("[PAD]", 0)
+        self._mask = v.get("[MASK]", 0)
+
+    def tokenize(self, text):
+        return self._tok.encode(text, add_special_tokens=False).ids
+
+    def detokenize(self, ids):
+        return self._tok.decode(ids)
+
+    @property
+    def vocab_size(self):
+        return self._tok.get_vocab_size()
+
+    @property
+    def vocab(self):
+        return self._tok.get_vocab()
+
+    @property
+    def cls(self):
+        return self._cls
+
+    @property
+    def sep(self):
+        return self._sep
+
+    @property
+    def pad(self):
+        return self._pad
+
+    @property
+    def mask(self):
+        return self._mask
+
+    @property
+    def eod(self):
+        return self._sep
+
+
+class TikTokenizer(MegatronTokenizer):
+    """Native byte-level BPE over a .tiktoken mergeable-ranks file
+    ("<base64> <rank>" lines — the reference's TikTokenizer via the
+    tiktoken package, reimplemented so no extra dependency is needed)."""
+
+    PATTERN = (r"""'(?i:[sdmt]|ll|ve|re)|[^\r\n\p{L}\p{N}]?+\p{L}+"""
+               r"""|\p{N}{1,3}| ?[^\s\p{L}\p{N}]++[\r\n]*"""
+               r"""|\s*[\r\n]|\s+(?!\S)|\s+""")
+
+    def __init__(self, model_file: str, special_tokens=None):
+        import base64
+        import regex
+        self._ranks = {}
+        with open(model_file, "rb") as f:
+            for line in f:
+                line = line.strip()
+                if not line:
+                    continue
+                tok_b64, rank = line.split()
+                self._ranks[base64.b64decode(tok_b64)] = int(rank)
+        self._pat = regex.compile(self.PATTERN)
+        n = len(self._ranks)
+        self._specials = {}
+        for i, name in enumerate(special_tokens or ["<|endoftext|>"]):
+            self._specials[name] = n + i
+        self._eod = self._specials.get("<|endoftext|>", n)
+        self._id_to_bytes = {r: b for b, r in self._ranks.items()}
+        for name, i in self._specials.items():
+            self._id_to_bytes[i] = name.encode()
+
+    def _bpe(self, piece: bytes):
+        parts = [piece[i:i + 1] for i in range(len(piece))]
+        while len(parts) > 1:
+            best, best_rank = None, None
+            for i in range(len(parts) - 1):
+                r = self._ranks.get(parts[i] + parts[i + 1])
+                if r is not None and (best_rank is None or r < best_rank):
+                    best, best_rank = i, r
+            if best is None:
+                break
+            parts[best:best + 2] = [parts[best] + parts[best + 1]]
+        return [self._ranks[p] for p in parts if p in self._ranks]
+
+    def tokenize(self, text):
+        out = []
+        for piece in self._pat.findall(text):
+            b = piece.encode("utf-8")
+            if b in self._ranks:
+                out.append(self._ranks[b])
+            else:
+                out.extend(self._bpe(b))
+        return out
+
+    def detokenize(self, ids):
+        data = b"".join(self._id_to_bytes.get(i, b"") for i in ids)
+        return data.decode("utf-8", errors="replace")
+
+    @property
+    def vocab_size(self):
+        return len(self._ranks) + len(self._specials)
+
+    @property
+    def eod(self):
+        return self._eod
+
+
 def build_tokenizer(args):
     t = args.tokenizer_type
     if t == "NullTokenizer":
@@ -174,4 +288,10 @@ def build_tokenizer(args):
         return SentencePieceTokenizer(args.tokenizer_model)
     if t == "HuggingFaceTokenizer":
         return HuggingFaceTokenizer(args.tokenizer_model)
+    if t == "BertWordPieceLowerCase":
+        return BertWordPieceTokenizer(args.vocab_file, lower_case=True)
+    if t == "BertWordPieceCase":
+        return BertWordPieceTokenizer(args.vocab_file, lower_case=False)
+    if t == "TikTokenizer":
+        return TikTokenizer(args.tokenizer_model)
     raise ValueError(f"unknown tokenizer type {t}")

@@ -371,3 +371,33 @@ def test_legacy_loader_merges_tp_pp_shards(tmp_path, monkeypatch):
     assert full["model.decoder.layers.1.input_layernorm.weight"][0] == 1
     assert full["model.embedding.word_embeddings.weight"].shape == (16, h)
     assert common.get("iteration") == 5
+
+
+def test_bert_wordpiece_tokenizer(tmp_path):
+    vocab = ["[PAD]", "[UNK]", "[CLS]", "[SEP]", "[MASK]",
+             "the", "quick", "brown", "fox", "##es", "jump", "##ing"]
+    vf = tmp_path / "vocab.txt"
+    vf.write_text("\n".join(vocab))
+    from megatronapp_amd.training.tokenizer import BertWordPieceTokenizer
+    t = BertWordPieceTokenizer(str(vf))
+    ids = t.tokenize("the quick foxes")
+    assert ids == [5, 6, 8, 9]
+    assert "fox" in t.detokenize(ids)
+    assert t.cls == 2 and t.sep == 3 and t.mask == 4 and t.pad == 0
+
+
+def test_tiktoken_tokenizer(tmp_path):
+    import base64
+    # bytes for 'h','e','l','o',' ' plus merges 'he','ll','llo','hello'
+    toks = [b"h", b"e", b"l", b"o", b" ", b"he", b"ll", b"llo", b"hello"]
+    mf = tmp_path / "toy.tiktoken"
+    mf.write_text("\n".join(
+        f"{base64.b64encode(t).decode()} {i}" for i, t in enumerate(toks)))
+    from megatronapp_amd.training.tokenizer import TikTokenizer
+    t = TikTokenizer(str(mf))
+    ids = t.tokenize("hello")
+    assert ids == [8]              # fully merged
+    assert t.detokenize(ids) == "hello"
+    assert t.tokenize("he") == [5]
+    assert t.vocab_size == 10      # 9 ranks + <|endoftext|>
+    assert t.eod == 9
