@@ -117,12 +117,44 @@ class RankGenerator:
         return groups
 
 
-def _new_group(ranks, backend=None, gloo=False):
+# per-group RCCL tuning from --nccl-communicator-config-path
+_PG_COMM_CONFIG = {}
+
+
+def set_pg_comm_config(cfg: dict):
+    global _PG_COMM_CONFIG
+    _PG_COMM_CONFIG = cfg or {}
+
+
+def _pg_options(kind):
+    """ProcessGroupNCCL.Options for a named group kind (tp/dp/pp/cp/ep),
+    honoring min_ctas/max_ctas/cga_cluster_size from the comm config."""
+    cfg = _PG_COMM_CONFIG.get(kind)
+    if not cfg:
+        return None
+    try:
+        opts = dist.ProcessGroupNCCL.Options()
+        nccl_cfg = opts.config
+        if "min_ctas" in cfg:
+            nccl_cfg.min_ctas = int(cfg["min_ctas"])
+        if "max_ctas" in cfg:
+            nccl_cfg.max_ctas = int(cfg["max_ctas"])
+        if "cga_cluster_size" in cfg:
+            nccl_cfg.cga_cluster_size = int(cfg["cga_cluster_size"])
+        return opts
+    except Exception:
+        return None
+
+
+def _new_group(ranks, backend=None, gloo=False, kind=None):
     """Create a group; every rank must call this in the same order."""
     if gloo:
         if dist.is_gloo_available():
             return dist.new_group(ranks, backend="gloo")
         return None
+    opts = _pg_options(kind) if kind else None
+    if opts is not None:
+        return dist.new_group(ranks, backend=backend, pg_options=opts)
     return dist.new_group(ranks, backend=backend)
 
 
@@ -183,16 +215,16 @@ def initialize_model_parallel(
         "model parallel already initialized (call destroy_model_parallel first)"
 
     for ranks in gen.get_ranks("tp"):
-        group = _new_group(ranks)
+        group = _new_group(ranks, kind="tp")
         if rank in ranks:
             _TENSOR_MODEL_PARALLEL_GROUP = group
             _TENSOR_MODEL_PARALLEL_GLOBAL_RANKS = ranks
 
     global _PIPELINE_FWD_GROUP, _PIPELINE_BWD_GROUP
     for ranks in gen.get_ranks("pp"):
-        group = _new_group(ranks)
-        fwd_group = _new_group(ranks)
-        bwd_group = _new_group(ranks)
+        group = _new_group(ranks, kind="pp")
+        fwd_group = _new_group(ranks, kind="pp")
+        bwd_group = _new_group(ranks, kind="pp")
         if rank in ranks:
             _PIPELINE_MODEL_PARALLEL_GROUP = group
             _PIPELINE_FWD_GROUP = fwd_group
@@ -216,7 +248,7 @@ def initialize_model_parallel(
             _POSITION_EMBEDDING_GLOBAL_RANKS = pos_ranks
 
     for ranks in gen.get_ranks("dp"):
-        group = _new_group(ranks)
+        group = _new_group(ranks, kind="dp")
         group_gloo = _new_group(ranks, gloo=True) if create_gloo_process_groups else None
         if rank in ranks:
             _DATA_PARALLEL_GROUP = group
@@ -224,7 +256,7 @@ def initialize_model_parallel(
             _DATA_PARALLEL_GLOBAL_RANKS = ranks
 
     for ranks in gen.get_ranks("dp-cp"):
-        group = _new_group(ranks)
+        group = _new_group(ranks, kind="dp")
         group_gloo = _new_group(ranks, gloo=True) if create_gloo_process_groups else None
         if rank in ranks:
             _DATA_PARALLEL_GROUP_WITH_CP = group
@@ -232,7 +264,7 @@ def initialize_model_parallel(
             _DATA_PARALLEL_GLOBAL_RANKS_WITH_CP = ranks
 
     for ranks in gen.get_ranks("cp"):
-        group = _new_group(ranks)
+        group = _new_group(ranks, kind="cp")
         if rank in ranks:
             _CONTEXT_PARALLEL_GROUP = group
             _CONTEXT_PARALLEL_GLOBAL_RANKS = ranks

@@ -143,6 +143,14 @@ def _add_training_args(p):
                    help="overlap the ZeRO-1 param all-gather into the next step")
     g.add_argument("--use-distributed-optimizer", action="store_true")
     g.add_argument("--ddp-bucket-size", type=int, default=None)
+    g.add_argument("--deterministic-mode", action="store_true",
+                   help="bitwise-reproducible runs: disables nondeterministic "
+                        "kernels (torch.use_deterministic_algorithms) and "
+                        "autotuned GEMM selection variance")
+    g.add_argument("--nccl-communicator-config-path", type=str, default=None,
+                   help="YAML of per-process-group RCCL tuning "
+                        "(min_ctas/max_ctas/cga_cluster_size per group name); "
+                        "applied as pg options at group creation")
     g.add_argument("--check-weight-hash-across-dp-replicas-interval",
                    type=int, default=None)
     g.add_argument("--calculate-per-token-loss", action="store_true")

@@ -32,6 +32,16 @@ def initialize_megatron(extra_args_provider=None, args_defaults={},
     validate_args(args, args_defaults)
     set_global_variables(args)
 
+    if getattr(args, "deterministic_mode", False):
+        # bitwise reproducibility: deterministic kernel selection plus a
+        # pinned hipBLASLt workspace config (reference --deterministic-mode)
+        torch.use_deterministic_algorithms(True, warn_only=True)
+        os.environ.setdefault("CUBLAS_WORKSPACE_CONFIG", ":4096:8")
+        os.environ.setdefault("HIPBLASLT_ALLOW_TF32", "0")
+        if torch.cuda.is_available():
+            torch.backends.cudnn.deterministic = True
+            torch.backends.cudnn.benchmark = False
+
     _initialize_distributed(args)
     _set_random_seed(args.seed)
 
@@ -55,7 +65,26 @@ def initialize_megatron(extra_args_provider=None, args_defaults={},
     return args
 
 
+def _load_comm_config(path):
+    import yaml
+    with open(path) as f:
+        return yaml.safe_load(f) or {}
+
+
 def _initialize_distributed(args):
+    # per-group RCCL tuning (reference --nccl-communicator-config-path):
+    # supported keys per group name (tp/dp/pp/ep/cp or 'default'):
+    # NCCL_MIN_CTAS / NCCL_MAX_CTAS-style knobs exported before init so
+    # RCCL picks them up; per-group ProcessGroupNCCL.Options are applied
+    # for groups created after init through parallel_state.
+    if getattr(args, "nccl_communicator_config_path", None):
+        cfg = _load_comm_config(args.nccl_communicator_config_path)
+        base = cfg.get("default", {})
+        if "min_ctas" in base:
+            os.environ.setdefault("NCCL_MIN_CTAS", str(base["min_ctas"]))
+        if "max_ctas" in base:
+            os.environ.setdefault("NCCL_MAX_CTAS", str(base["max_ctas"]))
+        parallel_state.set_pg_comm_config(cfg)
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "6000")

@@ -401,3 +401,43 @@ def test_tiktoken_tokenizer(tmp_path):
     assert t.tokenize("he") == [5]
     assert t.vocab_size == 10      # 9 ranks + <|endoftext|>
     assert t.eod == 9
+
+
+def test_nccl_comm_config_and_deterministic_flags(tmp_path):
+    """--nccl-communicator-config-path parses into per-group pg options
+    and --deterministic-mode switches torch into deterministic
+    algorithms (both reference flags)."""
+    import yaml
+    from megatronapp_amd.core import parallel_state as ps
+    cfg = {"default": {"min_ctas": 2, "max_ctas": 16},
+           "tp": {"min_ctas": 4}, "dp": {"max_ctas": 32}}
+    f = tmp_path / "comm.yaml"
+    f.write_text(yaml.safe_dump(cfg))
+    from megatronapp_amd.training.initialize import _load_comm_config
+    loaded = _load_comm_config(str(f))
+    assert loaded["tp"]["min_ctas"] == 4
+    ps.set_pg_comm_config(loaded)
+    try:
+        opts = ps._pg_options("tp")
+        # gloo-only environments may not expose ProcessGroupNCCL options
+        if opts is not None:
+            assert opts.config.min_ctas == 4
+    finally:
+        ps.set_pg_comm_config({})
+
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "pretrain_gpt.py"),
+         "--num-layers", "2", "--hidden-size", "64",
+         "--num-attention-heads", "4", "--seq-length", "32",
+         "--micro-batch-size", "2", "--global-batch-size", "2",
+         "--mock-data", "--train-iters", "2", "--lr", "1e-4",
+         "--vocab-size", "128", "--eval-iters", "0",
+         "--hidden-dropout", "0", "--attention-dropout", "0",
+         "--deterministic-mode",
+         "--nccl-communicator-config-path", str(f)],
+        capture_output=True, text=True, cwd=repo, timeout=300,
+        env={**os.environ, "MASTER_ADDR": "127.0.0.1",
+             "MASTER_PORT": "29741", "RANK": "0", "WORLD_SIZE": "1"})
+    assert r.returncode == 0, r.stderr[-2500:]
