@@ -143,6 +143,13 @@ def _add_training_args(p):
                    help="overlap the ZeRO-1 param all-gather into the next step")
     g.add_argument("--use-distributed-optimizer", action="store_true")
     g.add_argument("--ddp-bucket-size", type=int, default=None)
+    g.add_argument("--non-persistent-save-interval", type=int, default=None,
+                   help="iterations between LOCAL (node-scratch) "
+                        "checkpoints for fast restart; kept in rotation "
+                        "of one (reference local non-persistent ckpts)")
+    g.add_argument("--non-persistent-ckpt-dir", type=str, default=None,
+                   help="node-local directory for non-persistent "
+                        "checkpoints (default <save>/local_ckpt)")
     g.add_argument("--deterministic-mode", action="store_true",
                    help="bitwise-reproducible runs: disables nondeterministic "
                         "kernels (torch.use_deterministic_algorithms) and "

@@ -91,6 +91,82 @@ def finalize_async_save(blocking: bool = True):
         _async_save_thread = None
 
 
+def save_local_checkpoint(iteration, model: List, optimizer,
+                          opt_param_scheduler):
+    """Non-persistent node-local checkpoint (reference
+    nvidia_resiliency LocalCheckpointManager, training.py:702-726):
+    a fast legacy-format save into local scratch, rotation depth 1, for
+    quick in-place restart after a crash.  load_checkpoint prefers it
+    when it is NEWER than the persistent checkpoint."""
+    import shutil
+    args = get_args()
+    base = args.non_persistent_ckpt_dir or (
+        os.path.join(args.save, "local_ckpt") if args.save else None)
+    if base is None:
+        return
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    new_dir = os.path.join(base, f"iter_{iteration:07d}")
+    os.makedirs(new_dir, exist_ok=True)
+    sd = {"iteration": iteration, "checkpoint_version": 3.0}
+    if len(model) == 1:
+        sd["model"] = model[0].state_dict_for_save_checkpoint()
+    else:
+        for i, chunk in enumerate(model):
+            sd[f"model{i}"] = chunk.state_dict_for_save_checkpoint()
+    if optimizer is not None:
+        sd["optimizer"] = optimizer.state_dict()
+    if opt_param_scheduler is not None:
+        sd["opt_param_scheduler"] = opt_param_scheduler.state_dict()
+    sd["rng_state"] = _rng_state()
+    torch.save(sd, os.path.join(new_dir, f"rank_{rank:05d}.pt"))
+    if dist.is_initialized():
+        dist.barrier()
+    if rank == 0:
+        with open(os.path.join(base, "latest"), "w") as f:
+            f.write(str(iteration))
+        # rotation depth 1: drop older local checkpoints
+        for d in os.listdir(base):
+            if d.startswith("iter_") and d != f"iter_{iteration:07d}":
+                shutil.rmtree(os.path.join(base, d), ignore_errors=True)
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def load_local_checkpoint_if_newer(model: List, optimizer,
+                                   opt_param_scheduler, persistent_iter):
+    """Return the local iteration if a newer local checkpoint was
+    loaded, else None."""
+    args = get_args()
+    base = args.non_persistent_ckpt_dir or (
+        os.path.join(args.save, "local_ckpt") if args.save else None)
+    if base is None or not os.path.exists(os.path.join(base, "latest")):
+        return None
+    with open(os.path.join(base, "latest")) as f:
+        local_iter = int(f.read().strip())
+    if local_iter <= persistent_iter:
+        return None
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    path = os.path.join(base, f"iter_{local_iter:07d}",
+                        f"rank_{rank:05d}.pt")
+    if not os.path.exists(path):
+        return None
+    sd = torch.load(path, map_location="cpu", weights_only=False)
+    if len(model) == 1:
+        model[0].load_state_dict(sd["model"], strict=True)
+    else:
+        for i, chunk in enumerate(model):
+            chunk.load_state_dict(sd[f"model{i}"], strict=True)
+    if optimizer is not None and "optimizer" in sd:
+        optimizer.load_state_dict(sd["optimizer"])
+    if opt_param_scheduler is not None and "opt_param_scheduler" in sd:
+        opt_param_scheduler.load_state_dict(sd["opt_param_scheduler"])
+    if sd.get("rng_state"):
+        _restore_rng_state(sd["rng_state"])
+    print(f"  loaded LOCAL (non-persistent) checkpoint at iteration "
+          f"{local_iter}", flush=True)
+    return local_iter
+
+
 def save_checkpoint(iteration, model: List, optimizer, opt_param_scheduler,
                     num_floating_point_operations_so_far=0, checkpointing_context=None,
                     train_data_iterator=None, **kwargs):

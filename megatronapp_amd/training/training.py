@@ -147,6 +147,15 @@ def setup_model_and_optimizer(model_provider_func, model_type,
     else:
         args.iteration = 0
         args.num_floating_point_operations_so_far = 0
+    # a NEWER node-local non-persistent checkpoint (crash restart) wins
+    if getattr(args, "non_persistent_save_interval", None) or \
+            getattr(args, "non_persistent_ckpt_dir", None):
+        from .checkpointing import load_local_checkpoint_if_newer
+        li = load_local_checkpoint_if_newer(model, optimizer,
+                                            opt_param_scheduler,
+                                            args.iteration)
+        if li is not None:
+            args.iteration = li
     return model, optimizer, opt_param_scheduler
 
 
@@ -440,6 +449,12 @@ def train(forward_step_func, model, optimizer, opt_param_scheduler,
                 iteration % args.save_interval == 0:
             save_checkpoint(iteration, model, optimizer, opt_param_scheduler,
                             args.num_floating_point_operations_so_far)
+
+        if getattr(args, "non_persistent_save_interval", None) and \
+                iteration % args.non_persistent_save_interval == 0:
+            from .checkpointing import save_local_checkpoint
+            save_local_checkpoint(iteration, model, optimizer,
+                                  opt_param_scheduler)
 
         if args.exit_interval and iteration % args.exit_interval == 0:
             break

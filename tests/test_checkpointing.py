@@ -182,3 +182,42 @@ def test_torch_dist_save_resume_pp2(tmp_path):
                  "--load", save, "--save-interval", "100",
                  "--eval-iters", "0"], 29668)
     assert "loaded checkpoint (torch_dist)" in out3
+
+
+def test_non_persistent_local_checkpoint_resume(tmp_path):
+    """Local (node-scratch) checkpoints save on their own interval and a
+    NEWER local checkpoint wins over the persistent one at resume."""
+    save = str(tmp_path / "persist")
+    local = str(tmp_path / "scratch")
+    args = ["--num-layers", "2", "--hidden-size", "64",
+            "--num-attention-heads", "4", "--seq-length", "32",
+            "--micro-batch-size", "2", "--global-batch-size", "4",
+            "--mock-data", "--lr", "1e-3", "--log-interval", "1",
+            "--vocab-size", "128", "--eval-iters", "0",
+            "--hidden-dropout", "0", "--attention-dropout", "0",
+            "--save", save, "--save-interval", "4",
+            "--non-persistent-save-interval", "2",
+            "--non-persistent-ckpt-dir", local]
+    out = _run(["--train-iters", "6"] + args[:0] + args, 29761) \
+        if False else None
+    import subprocess, sys
+    env = {**os.environ, "MASTER_ADDR": "127.0.0.1",
+           "MASTER_PORT": "29761", "RANK": "0", "WORLD_SIZE": "1"}
+    r = subprocess.run([sys.executable, "pretrain_gpt.py",
+                        "--train-iters", "6"] + args,
+                       capture_output=True, text=True, cwd=REPO,
+                       timeout=420, env=env)
+    assert r.returncode == 0, r.stderr[-2500:]
+    assert open(os.path.join(local, "latest")).read().strip() == "6"
+    # simulate a crash after the local save but before the final
+    # persistent save: roll the persistent tracker back to 4
+    with open(os.path.join(save,
+              "latest_checkpointed_iteration.txt"), "w") as f:
+        f.write("4")
+    r2 = subprocess.run([sys.executable, "pretrain_gpt.py",
+                         "--train-iters", "8", "--load", save] + args,
+                        capture_output=True, text=True, cwd=REPO,
+                        timeout=420, env=env)
+    assert r2.returncode == 0, r2.stderr[-2500:]
+    assert "loaded LOCAL (non-persistent) checkpoint at iteration 6" \
+        in r2.stdout
