@@ -30,9 +30,43 @@ def permute(tokens: torch.Tensor, indices: torch.Tensor):
     return tokens.index_select(0, tok_idx), sort_idx
 
 
+class _FusedCombineFn(torch.autograd.Function):
+    """unpermute + topk-weighted sum as ONE kernel each way
+    (ops/csrc/moe.hip) — the eager version is 3 full passes over
+    [n*topk, h]."""
+
+    @staticmethod
+    def forward(ctx, permuted, sort_idx, probs, n_tokens):
+        from .... import ops as _ops
+        inv_pos = torch.empty_like(sort_idx)
+        inv_pos[sort_idx] = torch.arange(sort_idx.numel(),
+                                         device=sort_idx.device)
+        probs_f = probs.reshape(-1).float().contiguous()
+        ctx.save_for_backward(permuted, sort_idx, probs_f)
+        ctx.topk = probs.shape[1]
+        ctx.probs_dtype = probs.dtype
+        return _ops.get_ops().moe_combine_fwd(
+            permuted.contiguous(), inv_pos.contiguous(), probs_f, n_tokens)
+
+    @staticmethod
+    def backward(ctx, dout):
+        from .... import ops as _ops
+        permuted, sort_idx, probs_f = ctx.saved_tensors
+        dpermuted, dprobs = _ops.get_ops().moe_combine_bwd(
+            dout, permuted.contiguous(), sort_idx.contiguous(), probs_f,
+            ctx.topk)
+        dprobs = dprobs.reshape(-1, ctx.topk).to(ctx.probs_dtype)
+        return dpermuted, None, dprobs, None
+
+
 def unpermute(permuted: torch.Tensor, sort_idx: torch.Tensor,
               probs: torch.Tensor, n_tokens: int):
     """Inverse of permute + weighted combine over topk copies."""
+    from .... import ops as _ops
+    if (permuted.is_cuda and permuted.dtype == torch.bfloat16
+            and permuted.shape[-1] % 8 == 0 and _ops.have_ops()
+            and hasattr(_ops.get_ops(), "moe_combine_fwd")):
+        return _FusedCombineFn.apply(permuted, sort_idx, probs, n_tokens)
     topk = probs.shape[1]
     h = permuted.shape[-1]
     # out-of-place index_copy keeps the autograd graph to `permuted`

@@ -58,6 +58,11 @@ void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
 void launch_quant_rows_e4m3(const void*, void*, float*, long, int,
                             hipStream_t);
+void launch_moe_combine_fwd(const void*, const long*, const float*, void*,
+                            long, int, int, hipStream_t);
+void launch_moe_combine_bwd(const void*, const void*, const long*,
+                            const float*, void*, float*, long, int, int,
+                            hipStream_t);
 void launch_attn_fwd2(const void*, const void*, const void*, void*, float*,
                       int, int, int, int, int, int, float, bool,
                       const long*, const long*, const long*, hipStream_t);
@@ -368,6 +373,36 @@ void embedding_bwd_accum(torch::Tensor dy, torch::Tensor tokens,
                              cur_stream());
 }
 
+// ---------------------------------------------------------------- moe
+torch::Tensor moe_combine_fwd(torch::Tensor permuted, torch::Tensor inv_pos,
+                              torch::Tensor probs, long n_tokens) {
+  check_bf16(permuted, "permuted");
+  const int h = (int)permuted.size(-1);
+  const int topk = (int)(permuted.size(0) / n_tokens);
+  auto out = torch::empty({n_tokens, (long)h}, permuted.options());
+  launch_moe_combine_fwd(permuted.data_ptr(), inv_pos.data_ptr<long>(),
+                         probs.data_ptr<float>(), out.data_ptr(), n_tokens,
+                         topk, h, cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> moe_combine_bwd(torch::Tensor dout,
+                                           torch::Tensor permuted,
+                                           torch::Tensor sort_idx,
+                                           torch::Tensor probs, long topk) {
+  check_bf16(dout, "dout");
+  check_bf16(permuted, "permuted");
+  const int h = (int)permuted.size(-1);
+  const long n_rows = permuted.size(0);
+  auto dpermuted = torch::empty_like(permuted);
+  auto dprobs = torch::empty({n_rows}, probs.options());
+  launch_moe_combine_bwd(dout.contiguous().data_ptr(), permuted.data_ptr(),
+                         sort_idx.data_ptr<long>(), probs.data_ptr<float>(),
+                         dpermuted.data_ptr(), dprobs.data_ptr<float>(),
+                         n_rows, (int)topk, h, cur_stream());
+  return {dpermuted, dprobs};
+}
+
 // ---------------------------------------------------------------- fp8 quant
 std::vector<torch::Tensor> quantize_rows_e4m3(torch::Tensor x) {
   check_bf16(x, "x");
@@ -640,6 +675,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd_t", &attn_fwd_t);
   mod.def("attn_fwd2", &attn_fwd2);
   mod.def("quantize_rows_e4m3", &quantize_rows_e4m3);
+  mod.def("moe_combine_fwd", &moe_combine_fwd);
+  mod.def("moe_combine_bwd", &moe_combine_bwd);
   mod.def("attn_bwd_v1", &attn_bwd_v1);
   mod.def("ce_rowmax", &ce_rowmax);
   mod.def("gemm_nt", &gemm_nt);

@@ -731,3 +731,35 @@ def test_quantize_rows_e4m3_matches_eager():
         deq_ref = q_ref.float() * s_ref
         err_ref = (deq_ref - x.float()).abs().max()
         assert err < err_ref * 1.5 + 1e-3
+
+
+@pytest.mark.gpu
+def test_moe_fused_combine_matches_eager():
+    """Fused unpermute+combine (moe.hip) vs the eager 3-pass version,
+    forward and gradients."""
+    from megatronapp_amd.core.transformer.moe.token_dispatcher import (
+        _FusedCombineFn, permute)
+    torch.manual_seed(71)
+    n, topk, h, E = 512, 2, 256, 8
+    tokens = torch.randn(n, h, device="cuda", dtype=torch.bfloat16)
+    indices = torch.randint(0, E, (n, topk), device="cuda")
+    probs = torch.softmax(torch.randn(n, topk, device="cuda"), dim=-1)
+    permuted, sort_idx = permute(tokens, indices)
+
+    pg = permuted.detach().requires_grad_(True)
+    prg = probs.detach().requires_grad_(True)
+    out = _FusedCombineFn.apply(pg, sort_idx, prg, n)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+
+    pe = permuted.detach().float().requires_grad_(True)
+    pre = probs.detach().float().requires_grad_(True)
+    unsorted = torch.zeros_like(pe).index_copy(0, sort_idx, pe)
+    ref = (unsorted.reshape(n, topk, h) * pre.unsqueeze(-1)).sum(1)
+    ref.backward(dout.float())
+
+    assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2)
+    rel = (pg.grad.float() - pe.grad).abs().max() / pe.grad.abs().max()
+    assert rel < 3e-2, float(rel)
+    relp = (prg.grad.float() - pre.grad).abs().max() / pre.grad.abs().max()
+    assert relp < 3e-2, float(relp)
