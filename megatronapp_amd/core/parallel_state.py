@@ -129,7 +129,7 @@ def set_pg_comm_config(cfg: dict):
 def _pg_options(kind):
     """ProcessGroupNCCL.Options for a named group kind (tp/dp/pp/cp/ep),
     honoring min_ctas/max_ctas/cga_cluster_size from the comm config."""
-    cfg = _PG_COMM_CONFIG.get(kind)
+    cfg = (_PG_COMM_CONFIG or {}).get(kind)
     if not cfg:
         return None
     try:
