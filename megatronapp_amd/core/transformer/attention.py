@@ -171,6 +171,13 @@ class SelfAttention(MegatronModule):
                            inference_context._arange[:query.shape[0]])
                     q_pos_emb = q_pos_emb.index_select(0, idx)
                     k_pos_emb = k_pos_emb.index_select(0, idx)
+                elif getattr(inference_context, "is_dynamic", False) and \
+                        query.shape[0] == 1:
+                    # continuous batching: per-ROW positions
+                    idx = inference_context.row_positions()
+                    q_pos_emb = q_pos_emb.index_select(0, idx).permute(
+                        1, 0, 2, 3)             # [1, b, 1, rot]
+                    k_pos_emb = q_pos_emb
                 else:
                     offset = inference_context.sequence_len_offset
                     q_pos_emb = q_pos_emb[offset:offset + query.shape[0]]
@@ -187,6 +194,11 @@ class SelfAttention(MegatronModule):
                 # fixed-shape window + padding mask (hipGraph decode)
                 attention_mask = inference_context.decode_padding_mask(
                     query.shape[0], query.shape[1])
+                attn_mask_type = AttnMaskType.padding
+            elif getattr(inference_context, "is_dynamic", False) and \
+                    query.shape[0] == 1:
+                # continuous batching: per-row window mask
+                attention_mask = inference_context.decode_padding_mask()
                 attn_mask_type = AttnMaskType.padding
             elif query.shape[0] == 1:
                 attn_mask_type = AttnMaskType.no_mask
