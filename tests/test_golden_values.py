@@ -14,9 +14,13 @@ GOLDEN = os.path.join(REPO, "tests", "golden", "golden_values_cpu.json")
 
 
 def test_golden_loss_curves_match():
+    # strip inherited rendezvous vars (earlier tests mutate os.environ)
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+                        "MASTER_PORT")}
     r = subprocess.run([sys.executable, "scripts/golden_values.py"],
                        capture_output=True, text=True, cwd=REPO,
-                       timeout=600)
+                       timeout=600, env=env)
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     assert "GOLDEN OK" in r.stdout
 
