@@ -688,8 +688,13 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_fwd2_kernel(
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
     long vS, long vB, long vH) {
   static_assert(D == 128 || D == 64, "attn_fwd2: head dim 64/128");
-  const int qtile = blockIdx.x;
-  const int bh = blockIdx.y;
+  // grid is (bh, qtile): consecutive blockIdx.x round-robin over XCDs,
+  // so every q-tile of one (batch, head) lands on the SAME XCD and its
+  // K/V tiles are fetched into that XCD's L2 once instead of 8 times
+  const int qtile = blockIdx.y;
+  const int bh = blockIdx.x;
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= FWD_BLOCK / 2)
+    __builtin_amdgcn_s_setprio(1);   // older-half priority (guide T9)
   const int bi = bh / nh;
   const int h = bh % nh;
   const int hkv = h / (nh / ng);
@@ -965,7 +970,7 @@ void launch_attn_fwd2(const void* q, const void* k, const void* v, void* o,
         "attn_fwd2: sq must be a multiple of 256, sk of 64");
   if (d != 128 && d != 64)
     throw std::runtime_error("attn_fwd2: d must be 64 or 128");
-  dim3 grid(sq / FQBLK2, b * nh);
+  dim3 grid(b * nh, sq / FQBLK2);
   dim3 block(FWD_BLOCK);
   const size_t lds =
       (size_t)(KVBLK * d + KVBLK * VRS + 8 * 32 * OSTRIDE) *
@@ -1037,8 +1042,10 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
     long vS, long vB, long vH) {
   static_assert(D == 128 || D == 64, "dq2: head dim 64/128");
-  const int qtile = blockIdx.x;
-  const int bh = blockIdx.y;
+  const int qtile = blockIdx.y;   // grid (bh, qtile): K/V per-XCD L2
+  const int bh = blockIdx.x;
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= FWD_BLOCK / 2)
+    __builtin_amdgcn_s_setprio(1);
   const int bi = bh / nh;
   const int h = bh % nh;
   const int hkv = h / (nh / ng);
@@ -1261,8 +1268,10 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dv2_kernel(
     unsigned short* __restrict__ dv, int sq, int sk, int b, int nh, int ng,
     float scale, long qS, long qB, long qH, long kS, long kB, long kH) {
   static_assert(D == 128 || D == 64, "dv2: head dim 64/128");
-  const int kvtile = blockIdx.x;
-  const int bh = blockIdx.y;
+  const int kvtile = blockIdx.y;  // grid (bh, kvtile): Q/dO per-XCD L2
+  const int bh = blockIdx.x;
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= FWD_BLOCK / 2)
+    __builtin_amdgcn_s_setprio(1);
   const int bi = bh / ng;
   const int hkv = bh % ng;
   const int group = nh / ng;
@@ -1440,8 +1449,10 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dk2_kernel(
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
     long vS, long vB, long vH) {
   static_assert(D == 128 || D == 64, "dk2: head dim 64/128");
-  const int kvtile = blockIdx.x;
-  const int bh = blockIdx.y;
+  const int kvtile = blockIdx.y;  // grid (bh, kvtile): Q/dO per-XCD L2
+  const int bh = blockIdx.x;
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= FWD_BLOCK / 2)
+    __builtin_amdgcn_s_setprio(1);
   const int bi = bh / ng;
   const int hkv = bh % ng;
   const int group = nh / ng;
@@ -1668,7 +1679,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
 #define ATT_BWD2_LAUNCH(DD, CC)                                                   \
   do {                                                                        \
     hipLaunchKernelGGL((attn_bwd_dq2_kernel<DD, CC>),                        \
-                       dim3(sq / FQBLK2, b * nh), dim3(FWD_BLOCK), lds_dq,    \
+                       dim3(b * nh, sq / FQBLK2), dim3(FWD_BLOCK), lds_dq,    \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k, (const unsigned short*)v,    \
                        (const unsigned short*)dout, lse, drow,                \
@@ -1677,7 +1688,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        vstr[0], vstr[1], vstr[2]);                            \
     HIP_CHECK_LAUNCH();                                                       \
     hipLaunchKernelGGL((attn_bwd_dv2_kernel<DD, CC>),                        \
-                       dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dv,    \
+                       dim3(b * ng, sk / FQBLK2), dim3(FWD_BLOCK), lds_dv,    \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k,                              \
                        (const unsigned short*)dout, lse,                      \
@@ -1686,7 +1697,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        kstr[2]);                                              \
     HIP_CHECK_LAUNCH();                                                       \
     hipLaunchKernelGGL((attn_bwd_dk2_kernel<DD, CC>),                        \
-                       dim3(sk / FQBLK2, b * ng), dim3(FWD_BLOCK), lds_dk,    \
+                       dim3(b * ng, sk / FQBLK2), dim3(FWD_BLOCK), lds_dk,    \
                        stream, (const unsigned short*)q,                      \
                        (const unsigned short*)k, (const unsigned short*)v,    \
                        (const unsigned short*)dout, lse, drow,                \
