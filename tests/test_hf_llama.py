@@ -76,3 +76,50 @@ def test_hf_llama_export_logit_parity(tmp_path, ng):
                                   atol=1e-6), k
     finally:
         destroy()
+
+
+def test_gguf_export_round_trip(tmp_path):
+    """GGUF writer: llama-family checkpoint -> .gguf, verified by an
+    independent minimal reader (header, metadata, aligned tensor data)."""
+    import os
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.join(repo, "tools", "checkpoint"))
+    try:
+        import importlib
+        import saver_gguf
+        importlib.reload(saver_gguf)
+        h, nh, ng, ffn, vocab, L = 64, 4, 2, 128, 256, 2
+        torch.manual_seed(3)
+        full = {"model.embedding.word_embeddings.weight":
+                torch.randn(vocab, h)}
+        full["model.decoder.final_layernorm.weight"] = torch.randn(h)
+        full["model.output_layer.weight"] = torch.randn(vocab, h)
+        for i in range(L):
+            p = f"model.decoder.layers.{i}."
+            full[p + "self_attention.linear_qkv.weight"] = torch.randn(
+                (nh + 2 * ng) * (h // nh), h)
+            full[p + "self_attention.linear_proj.weight"] = torch.randn(h, h)
+            full[p + "input_layernorm.weight"] = torch.randn(h)
+            full[p + "pre_mlp_layernorm.weight"] = torch.randn(h)
+            full[p + "mlp.linear_fc1.weight"] = torch.randn(2 * ffn, h)
+            full[p + "mlp.linear_fc2.weight"] = torch.randn(h, ffn)
+        common = {"args": {"num_attention_heads": nh,
+                           "num_query_groups": ng}}
+        out = str(tmp_path / "model.gguf")
+        saver_gguf.save_gguf(full, common, out, dtype="f32")
+        meta, tensors = saver_gguf.read_gguf(out)
+        assert meta["general.architecture"] == "llama"
+        assert meta["llama.block_count"] == L
+        assert meta["llama.attention.head_count_kv"] == ng
+        assert tensors["token_embd.weight"].shape == (vocab, h)
+        assert torch.allclose(
+            tensors["token_embd.weight"],
+            full["model.embedding.word_embeddings.weight"])
+        assert tensors["blk.1.ffn_down.weight"].shape == (h, ffn)
+        assert torch.allclose(tensors["blk.1.ffn_down.weight"],
+                              full["model.decoder.layers.1.mlp."
+                                   "linear_fc2.weight"])
+        assert tensors["output.weight"].shape == (vocab, h)
+    finally:
+        sys.path.pop(0)

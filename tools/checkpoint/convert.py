@@ -231,7 +231,13 @@ def load_hf_mixtral(path):
 LOADERS = {"torch_dist": load_torch_dist, "legacy": load_legacy,
            "consolidated": load_consolidated, "hf_gpt2": load_hf_gpt2,
            "hf_llama": load_hf_llama, "hf_mixtral": load_hf_mixtral}
+def save_gguf(full, common, path):
+    from saver_gguf import save_gguf as impl
+    impl(full, common, path)
+
+
 SAVERS = {"torch_dist": save_torch_dist, "consolidated": save_consolidated,
+          "gguf": save_gguf,
           "hf_gpt2": save_hf_gpt2, "hf_llama": save_hf_llama,
           "hf_mixtral": save_hf_mixtral}
 
