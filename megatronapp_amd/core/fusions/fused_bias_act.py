@@ -91,6 +91,39 @@ class _BiasSwigluFn(torch.autograd.Function):
         return dx, dbias
 
 
+class _BiasGegluFn(torch.autograd.Function):
+    """y = gelu(x1 + b1) * (x2 + b2) on interleaved halves [..., 2F]
+    (reference fused_bias_geglu.py)."""
+
+    @staticmethod
+    def forward(ctx, x, bias):
+        ctx.save_for_backward(x, bias)
+        if _ops.fused_enabled(x, "bias_act"):
+            return _ops.get_ops().bias_geglu_fwd(x, bias)
+        xf = (x.float() + bias.float()) if bias is not None else x.float()
+        x1, x2 = xf.chunk(2, dim=-1)
+        return (torch.nn.functional.gelu(x1, approximate="tanh") * x2
+                ).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, bias = ctx.saved_tensors
+        if _ops.fused_enabled(dy, "bias_act"):
+            dx = _ops.get_ops().bias_geglu_bwd(dy.contiguous(), x, bias)
+        else:
+            xf = (x.float() + bias.float()) if bias is not None else x.float()
+            x1, x2 = xf.chunk(2, dim=-1)
+            c0, c1 = 0.7978845608028654, 0.044715
+            t = torch.tanh(c0 * (x1 + c1 * x1 ** 3))
+            g = 0.5 * x1 * (1 + t)
+            dgelu = 0.5 * (1 + t) + \
+                0.5 * x1 * (1 - t * t) * c0 * (1 + 3 * c1 * x1 ** 2)
+            dyf = dy.float()
+            dx = torch.cat([dyf * x2 * dgelu, dyf * g], dim=-1).to(x.dtype)
+        dbias = _bias_grad(bias, dx)
+        return dx, dbias
+
+
 class _SquaredReluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, bias):
@@ -114,6 +147,10 @@ def bias_gelu_impl(x, bias=None):
 
 def bias_swiglu_impl(x, bias=None):
     return _BiasSwigluFn.apply(x, bias)
+
+
+def bias_geglu_impl(x, bias=None):
+    return _BiasGegluFn.apply(x, bias)
 
 
 def bias_squared_relu_impl(x, bias=None):

@@ -56,7 +56,11 @@ class ScaledMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, inputs, mask, scale):
         if _ops.fused_enabled(inputs, "softmax") and _sk_ok(inputs):
-            probs = _ops.get_ops().scaled_masked_softmax_fwd(inputs, mask, scale)
+            if mask is None:
+                probs = _ops.get_ops().scaled_softmax_fwd(inputs, scale)
+            else:
+                probs = _ops.get_ops().scaled_masked_softmax_fwd(inputs, mask,
+                                                                 scale)
         else:
             x = inputs.float() * scale
             if mask is not None:

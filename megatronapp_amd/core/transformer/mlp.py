@@ -49,7 +49,10 @@ class MLP(MegatronModule):
             bias=config.add_bias_linear, input_is_parallel=True,
             skip_bias_add=True, is_expert=is_expert)
 
-        if config.gated_linear_unit:
+        if config.gated_linear_unit and config.activation_func == "gelu":
+            from ..fusions.fused_bias_act import bias_geglu_impl
+            self.activation = bias_geglu_impl     # GeGLU
+        elif config.gated_linear_unit:
             self.activation = bias_swiglu_impl
         elif config.activation_func == "squared_relu":
             self.activation = bias_squared_relu_impl

@@ -27,6 +27,10 @@ void launch_bias_swiglu_fwd(const void*, const void*, void*, long, int,
                             hipStream_t);
 void launch_bias_swiglu_bwd(const void*, const void*, const void*, void*, long,
                             int, hipStream_t);
+void launch_bias_geglu_fwd(const void*, const void*, void*, long, int,
+                           hipStream_t);
+void launch_bias_geglu_bwd(const void*, const void*, const void*, void*,
+                           long, int, hipStream_t);
 void launch_rope(const void*, const float*, const float*, void*, long, int,
                  int, bool, hipStream_t);
 void launch_softmax_causal_fwd(const void*, void*, long, int, int, float,
@@ -267,6 +271,34 @@ torch::Tensor bias_swiglu_bwd(torch::Tensor dy, torch::Tensor x,
   return dx;
 }
 
+torch::Tensor bias_geglu_fwd(torch::Tensor x, c10::optional<torch::Tensor> bias) {
+  check_bf16(x, "x");
+  const int F2 = (int)x.size(-1);
+  TORCH_CHECK(F2 % 2 == 0, "last dim must be even (gated)");
+  const int F = F2 / 2;
+  const long N = x.numel() / F2;
+  auto sizes = x.sizes().vec();
+  sizes.back() = F;
+  auto y = torch::empty(sizes, x.options());
+  launch_bias_geglu_fwd(x.data_ptr(),
+                         bias.has_value() ? bias->data_ptr() : nullptr,
+                         y.data_ptr(), N, F, cur_stream());
+  return y;
+}
+
+torch::Tensor bias_geglu_bwd(torch::Tensor dy, torch::Tensor x,
+                              c10::optional<torch::Tensor> bias) {
+  check_bf16(dy, "dy");
+  const int F2 = (int)x.size(-1);
+  const int F = F2 / 2;
+  const long N = x.numel() / F2;
+  auto dx = torch::empty_like(x);
+  launch_bias_geglu_bwd(dy.data_ptr(), x.data_ptr(),
+                         bias.has_value() ? bias->data_ptr() : nullptr,
+                         dx.data_ptr(), N, F, cur_stream());
+  return dx;
+}
+
 // --------------------------------------------------------------------- rope
 torch::Tensor rope_apply(torch::Tensor t, torch::Tensor cs, torch::Tensor sn,
                          bool bwd) {
@@ -292,6 +324,19 @@ torch::Tensor rope_bwd(torch::Tensor dy, torch::Tensor cs, torch::Tensor sn) {
 }
 
 // ------------------------------------------------------------------ softmax
+torch::Tensor scaled_softmax_fwd(torch::Tensor x, double scale) {
+  // no-mask row softmax (reference scaled_softmax_cuda): the masked
+  // kernel with a null mask pointer
+  check_bf16(x, "x");
+  const int sk = (int)x.size(-1);
+  const int sq = (int)x.size(-2);
+  const long rows = x.numel() / sk;
+  auto y = torch::empty_like(x);
+  launch_softmax_masked_fwd(x.data_ptr(), nullptr, y.data_ptr(), rows,
+                            1, sq, sk, (float)scale, cur_stream());
+  return y;
+}
+
 torch::Tensor scaled_upper_triang_masked_softmax_fwd(torch::Tensor x,
                                                      double scale) {
   check_bf16(x, "x");
@@ -664,10 +709,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bias_gelu_bwd", &bias_gelu_bwd);
   mod.def("bias_swiglu_fwd", &bias_swiglu_fwd);
   mod.def("bias_swiglu_bwd", &bias_swiglu_bwd);
+  mod.def("bias_geglu_fwd", &bias_geglu_fwd);
+  mod.def("bias_geglu_bwd", &bias_geglu_bwd);
   mod.def("rope_fwd", &rope_fwd);
   mod.def("rope_bwd", &rope_bwd);
   mod.def("scaled_upper_triang_masked_softmax_fwd",
           &scaled_upper_triang_masked_softmax_fwd);
+  mod.def("scaled_softmax_fwd", &scaled_softmax_fwd);
   mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
   mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
   mod.def("adamw_flat", &adamw_flat);

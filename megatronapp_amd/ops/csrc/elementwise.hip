@@ -142,6 +142,77 @@ __global__ void bias_swiglu_bwd_kernel(const unsigned short* __restrict__ dy,
   }
 }
 
+// geglu: gelu(x1 + b1) * (x2 + b2) — the GeGLU gated MLP
+// (reference fused_bias_geglu.py)
+__global__ void bias_geglu_fwd_kernel(const unsigned short* __restrict__ x,
+                                      const unsigned short* __restrict__ bias,
+                                      unsigned short* __restrict__ y, long N,
+                                      int F) {
+  long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
+  const long total = N * (F / VEC);
+  const long stride = (long)gridDim.x * BLOCK;
+  for (; idx < total; idx += stride) {
+    const long row = idx / (F / VEC);
+    const int col = (int)(idx % (F / VEC)) * VEC;
+    const unsigned short* x1 = x + row * 2L * F + col;
+    const unsigned short* x2 = x1 + F;
+    short8v v1 = *(const short8v*)x1;
+    short8v v2 = *(const short8v*)x2;
+    short8v o;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float a = bf2f((unsigned short)v1[j]);
+      float b = bf2f((unsigned short)v2[j]);
+      if (bias != nullptr) {
+        a += bf2f((unsigned short)bias[col + j]);
+        b += bf2f((unsigned short)bias[F + col + j]);
+      }
+      o[j] = (short)f2bf(gelu_tanh(a) * b);
+    }
+    *(short8v*)(y + row * F + col) = o;
+  }
+}
+
+__global__ void bias_geglu_bwd_kernel(const unsigned short* __restrict__ dy,
+                                      const unsigned short* __restrict__ x,
+                                      const unsigned short* __restrict__ bias,
+                                      unsigned short* __restrict__ dx, long N,
+                                      int F) {
+  long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
+  const long total = N * (F / VEC);
+  const long stride = (long)gridDim.x * BLOCK;
+  for (; idx < total; idx += stride) {
+    const long row = idx / (F / VEC);
+    const int col = (int)(idx % (F / VEC)) * VEC;
+    const unsigned short* x1 = x + row * 2L * F + col;
+    const unsigned short* x2 = x1 + F;
+    short8v v1 = *(const short8v*)x1;
+    short8v v2 = *(const short8v*)x2;
+    short8v d = *(const short8v*)(dy + row * F + col);
+    short8v o1, o2;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float a = bf2f((unsigned short)v1[j]);
+      float b = bf2f((unsigned short)v2[j]);
+      if (bias != nullptr) {
+        a += bf2f((unsigned short)bias[col + j]);
+        b += bf2f((unsigned short)bias[F + col + j]);
+      }
+      float dyf = bf2f((unsigned short)d[j]);
+      // d/da gelu_tanh(a)
+      const float c0 = 0.7978845608028654f, c1 = 0.044715f;
+      float u = c0 * (a + c1 * a * a * a);
+      float t = tanhf(u);
+      float dgelu = 0.5f * (1.f + t) +
+                    0.5f * a * (1.f - t * t) * c0 * (1.f + 3.f * c1 * a * a);
+      o1[j] = (short)f2bf(dyf * b * dgelu);
+      o2[j] = (short)f2bf(dyf * gelu_tanh(a));
+    }
+    *(short8v*)(dx + row * 2L * F + col) = o1;
+    *(short8v*)(dx + row * 2L * F + F + col) = o2;
+  }
+}
+
 // ------------------------------------------------------------------ launchers
 static int ew_grid(long work_items) {
   long blocks = (work_items + BLOCK - 1) / BLOCK;
@@ -224,6 +295,25 @@ void launch_bias_swiglu_bwd(const void* dy, const void* x, const void* bias,
 // each lane owns 8 adjacent columns (one bf16x8 = 16 B load, fully
 // coalesced across the 256-thread block = 4 KB per row) and a block strip
 // of rows; per-column fp32 partials land with one atomicAdd each.
+void launch_bias_geglu_fwd(const void* x, const void* bias, void* y, long N,
+                           int F, hipStream_t stream) {
+  const long total = N * (F / VEC);
+  hipLaunchKernelGGL(bias_geglu_fwd_kernel, dim3(ew_grid(total)),
+                     dim3(BLOCK), 0, stream, (const unsigned short*)x,
+                     (const unsigned short*)bias, (unsigned short*)y, N, F);
+  HIP_CHECK_LAUNCH();
+}
+
+void launch_bias_geglu_bwd(const void* dy, const void* x, const void* bias,
+                           void* dx, long N, int F, hipStream_t stream) {
+  const long total = N * (F / VEC);
+  hipLaunchKernelGGL(bias_geglu_bwd_kernel, dim3(ew_grid(total)),
+                     dim3(BLOCK), 0, stream, (const unsigned short*)dy,
+                     (const unsigned short*)x, (const unsigned short*)bias,
+                     (unsigned short*)dx, N, F);
+  HIP_CHECK_LAUNCH();
+}
+
 __global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
                                     float* __restrict__ out, long R, int F,
                                     int rows_per_block) {

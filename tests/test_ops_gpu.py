@@ -763,3 +763,37 @@ def test_moe_fused_combine_matches_eager():
     assert rel < 3e-2, float(rel)
     relp = (prg.grad.float() - pre.grad).abs().max() / pre.grad.abs().max()
     assert relp < 3e-2, float(relp)
+
+
+@pytest.mark.gpu
+def test_bias_geglu_matches_eager():
+    """GeGLU fused kernels vs the fp32 eager formula, fwd + bwd."""
+    from megatronapp_amd.core.fusions.fused_bias_act import bias_geglu_impl
+    torch.manual_seed(81)
+    N, F = 512, 256
+    x = torch.randn(N, 2 * F, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(2 * F, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = bias_geglu_impl(x, b)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().requires_grad_(True)
+    br = b.detach().float().requires_grad_(True)
+    x1, x2 = (xr + br).chunk(2, dim=-1)
+    ref = torch.nn.functional.gelu(x1, approximate="tanh") * x2
+    ref.backward(dy.float())
+    assert torch.allclose(y.float(), ref, atol=5e-2, rtol=5e-2)
+    for got, want in ((x.grad, xr.grad), (b.grad, br.grad)):
+        rel = (got.float() - want).abs().max() / (want.abs().max() + 1e-6)
+        assert rel < 5e-2, float(rel)
+
+
+@pytest.mark.gpu
+def test_scaled_softmax_no_mask_fwd():
+    ops = _ops()
+    x = torch.randn(4, 64, 72, device="cuda", dtype=torch.bfloat16)
+    y = ops.scaled_softmax_fwd(x, 0.5)
+    ref = torch.softmax(x.float() * 0.5, dim=-1)
+    assert torch.allclose(y.float(), ref, atol=2e-2)
