@@ -39,8 +39,36 @@ class _BiasAddResidualFn(torch.autograd.Function):
         return grad, dummy, grad
 
 
+class _FusedBiasDropoutAddFn(torch.autograd.Function):
+    """One philox kernel for residual + dropout(x + bias)/(1-p) with the
+    byte mask saved for a single-pass backward (ops/csrc/dropout.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, bias, residual, prob):
+        seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+        out, mask = _ops.get_ops().bias_dropout_add_fwd(
+            x.contiguous(),
+            bias if bias is not None else torch.Tensor(),
+            residual.contiguous(), prob, seed)
+        ctx.save_for_backward(mask)
+        ctx.prob = prob
+        ctx.has_bias = bias is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        dx = _ops.get_ops().dropout_bwd(dy, mask, ctx.prob)
+        dbias = dx.reshape(-1, dx.shape[-1]).sum(0) if ctx.has_bias else None
+        return dx, dbias, dy, None
+
+
 def _bias_dropout_add_func(x_with_bias, residual, prob, training):
     x, bias = x_with_bias
+    if (training and 0.0 < prob < 1.0 and x.is_cuda
+            and x.dtype == torch.bfloat16 and _ops.have_ops()
+            and hasattr(_ops.get_ops(), "bias_dropout_add_fwd")):
+        return _FusedBiasDropoutAddFn.apply(x, bias, residual, prob)
     if (prob == 0.0 and bias is not None and x.is_cuda
             and x.dtype == torch.bfloat16 and _ops.have_ops()):
         if (torch.is_grad_enabled() and hasattr(bias, "main_grad")

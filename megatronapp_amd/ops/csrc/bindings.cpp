@@ -62,6 +62,11 @@ void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
 void launch_quant_rows_e4m3(const void*, void*, float*, long, int,
                             hipStream_t);
+void launch_bias_dropout_add_fwd(const void*, const void*, const void*,
+                                 void*, unsigned char*, long, int, float,
+                                 unsigned long long, hipStream_t);
+void launch_dropout_bwd(const void*, const unsigned char*, void*, long,
+                        float, hipStream_t);
 void launch_moe_combine_fwd(const void*, const long*, const float*, void*,
                             long, int, int, hipStream_t);
 void launch_moe_combine_bwd(const void*, const void*, const long*,
@@ -418,6 +423,32 @@ void embedding_bwd_accum(torch::Tensor dy, torch::Tensor tokens,
                              cur_stream());
 }
 
+// ------------------------------------------------------------- dropout
+std::vector<torch::Tensor> bias_dropout_add_fwd(
+    torch::Tensor x, torch::Tensor bias, torch::Tensor residual, double p,
+    long seed) {
+  check_bf16(x, "x");
+  check_bf16(residual, "residual");
+  auto out = torch::empty_like(x);
+  auto mask = torch::empty(x.sizes(), x.options().dtype(torch::kUInt8));
+  const long n = x.numel();
+  const int F = (int)x.size(-1);
+  launch_bias_dropout_add_fwd(
+      x.data_ptr(), bias.defined() ? bias.data_ptr() : nullptr,
+      residual.data_ptr(), out.data_ptr(), mask.data_ptr<unsigned char>(),
+      n, F, (float)p, (unsigned long long)seed, cur_stream());
+  return {out, mask};
+}
+
+torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
+  check_bf16(dy, "dy");
+  auto dx = torch::empty_like(dy);
+  launch_dropout_bwd(dy.contiguous().data_ptr(),
+                     mask.data_ptr<unsigned char>(), dx.data_ptr(),
+                     dy.numel(), (float)p, cur_stream());
+  return dx;
+}
+
 // ---------------------------------------------------------------- moe
 torch::Tensor moe_combine_fwd(torch::Tensor permuted, torch::Tensor inv_pos,
                               torch::Tensor probs, long n_tokens) {
@@ -723,6 +754,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd_t", &attn_fwd_t);
   mod.def("attn_fwd2", &attn_fwd2);
   mod.def("quantize_rows_e4m3", &quantize_rows_e4m3);
+  mod.def("bias_dropout_add_fwd", &bias_dropout_add_fwd);
+  mod.def("dropout_bwd", &dropout_bwd);
   mod.def("moe_combine_fwd", &moe_combine_fwd);
   mod.def("moe_combine_bwd", &moe_combine_bwd);
   mod.def("attn_bwd_v1", &attn_bwd_v1);

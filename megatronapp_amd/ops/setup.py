@@ -22,7 +22,7 @@ SOURCES = [
     "elementwise.hip",
     "rope.hip",
     "softmax.hip",
-    "adam.hip", "scan.hip", "ce.hip", "fp8quant.hip", "moe.hip",
+    "adam.hip", "scan.hip", "ce.hip", "fp8quant.hip", "moe.hip", "dropout.hip",
     "wgrad.cpp",
     "attention.hip",
 ]

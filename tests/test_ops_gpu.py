@@ -797,3 +797,34 @@ def test_scaled_softmax_no_mask_fwd():
     y = ops.scaled_softmax_fwd(x, 0.5)
     ref = torch.softmax(x.float() * 0.5, dim=-1)
     assert torch.allclose(y.float(), ref, atol=2e-2)
+
+
+@pytest.mark.gpu
+def test_fused_bias_dropout_add_statistics_and_grads():
+    """Philox fused bias-dropout-add: keep-rate within tolerance, kept
+    elements exactly (x+bias)/(1-p)+residual, backward masks dy
+    identically."""
+    from megatronapp_amd.core.fusions.fused_bias_dropout import (
+        _FusedBiasDropoutAddFn)
+    torch.manual_seed(91)
+    n, F, p = 4096, 256, 0.3
+    x = torch.randn(n, F, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    bias = torch.randn(F, device="cuda", dtype=torch.bfloat16,
+                       requires_grad=True)
+    res = torch.randn(n, F, device="cuda", dtype=torch.bfloat16)
+    out = _FusedBiasDropoutAddFn.apply(x, bias, res, p)
+    pre = (x + bias).detach()
+    kept = (out - res).abs() > 1e-6
+    rate = kept.float().mean().item()
+    assert abs(rate - (1 - p)) < 0.02, rate
+    scaled = (pre / (1 - p) + res).to(torch.bfloat16)
+    assert torch.allclose(out[kept].float(), scaled[kept].float(),
+                          atol=3e-2, rtol=3e-2)
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    # dx = dy*mask/(1-p): zero where dropped, scaled dy where kept
+    assert torch.allclose(x.grad[kept].float(),
+                          (dy[kept].float() / (1 - p)), atol=3e-2,
+                          rtol=3e-2)
+    assert (x.grad[~kept] == 0).all()
