@@ -815,16 +815,18 @@ def test_fused_bias_dropout_add_statistics_and_grads():
     res = torch.randn(n, F, device="cuda", dtype=torch.bfloat16)
     out = _FusedBiasDropoutAddFn.apply(x, bias, res, p)
     pre = (x + bias).detach()
-    kept = (out - res).abs() > 1e-6
+    dy = torch.randn_like(out)
+    dy[dy.abs() < 1e-2] = 0.1   # make dy nonzero so grads reveal the mask
+    out.backward(dy)
+    # ground-truth keep mask comes from the backward (dx != 0 iff kept)
+    kept = x.grad != 0
     rate = kept.float().mean().item()
     assert abs(rate - (1 - p)) < 0.02, rate
     scaled = (pre / (1 - p) + res).to(torch.bfloat16)
     assert torch.allclose(out[kept].float(), scaled[kept].float(),
                           atol=3e-2, rtol=3e-2)
-    dy = torch.randn_like(out)
-    out.backward(dy)
-    # dx = dy*mask/(1-p): zero where dropped, scaled dy where kept
+    assert torch.allclose(out[~kept].float(), res[~kept].float(),
+                          atol=1e-6)
     assert torch.allclose(x.grad[kept].float(),
                           (dy[kept].float() / (1 - p)), atol=3e-2,
                           rtol=3e-2)
-    assert (x.grad[~kept] == 0).all()
