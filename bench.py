@@ -115,7 +115,10 @@ def main():
         # with a pipeline keep enough microbatches to bound the bubble
         spd = gbs // dp  # samples per dp rank
         if pp == 1:
-            mbs = min(16, spd)
+            # 8B-class models (h >= 4096) OOM 288 GB at mbs 16
+            # (weights+optimizer ~150 GB + ~4 GB activations/layer)
+            cap = 8 if spec["hidden_size"] >= 4096 else 16
+            mbs = min(cap, spd)
         elif spd % 4 == 0 and spd // 4 >= pp:
             mbs = 4
         else:
