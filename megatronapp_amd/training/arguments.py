@@ -70,6 +70,18 @@ def _add_model_args(p):
     g.add_argument("--add-qkv-bias", action="store_true")
     g.add_argument("--qk-layernorm", action="store_true")
     g.add_argument("--untie-embeddings-and-output-weights", action="store_true")
+    g.add_argument("--no-position-embedding", action="store_false",
+                   dest="add_position_embedding",
+                   help="reference flag: drop learned absolute positions")
+    g.add_argument("--transformer-impl", default="local",
+                   choices=["local", "transformer_engine"],
+                   help="reference compatibility: this framework's native "
+                        "HIP kernels serve BOTH spec names (there is no TE "
+                        "on the MI355X path; transformer_engine maps to the "
+                        "same local modules)")
+    g.add_argument("--use-mcore-models", action="store_true",
+                   help="reference flag; mcore models are the only path "
+                        "here (accepted for script compatibility)")
     g.add_argument("--attention-dropout", type=float, default=0.1)
     g.add_argument("--hidden-dropout", type=float, default=0.1)
     g.add_argument("--layernorm-epsilon", type=float, default=1e-5)
@@ -131,6 +143,8 @@ def _add_training_args(p):
     g.add_argument("--adam-beta1", type=float, default=0.9)
     g.add_argument("--adam-beta2", type=float, default=0.999)
     g.add_argument("--adam-eps", type=float, default=1e-8)
+    g.add_argument("--recompute-activations", action="store_true",
+                   help="reference alias for selective recompute")
     g.add_argument("--recompute-granularity", default=None,
                    choices=[None, "selective", "full"])
     g.add_argument("--recompute-method", default=None,
@@ -240,6 +254,11 @@ def _add_data_args(p):
     g.add_argument("--decoder-seq-length", type=int, default=None,
                    help="decoder sequence length (encoder-decoder models)")
     g.add_argument("--vocab-size", type=int, default=None)
+    g.add_argument("--vocab-extra-ids", type=int, default=0,
+                   help="extra sentinel tokens (T5 span masking)")
+    g.add_argument("--encoder-num-layers", type=int, default=None)
+    g.add_argument("--decoder-num-layers", type=int, default=None)
+    g.add_argument("--encoder-seq-length", type=int, default=None)
     g.add_argument("--padded-vocab-size", type=int, default=None)
     g.add_argument("--make-vocab-size-divisible-by", type=int, default=128)
     g.add_argument("--vocab-file", default=None)
@@ -349,6 +368,17 @@ def validate_args(args, defaults={}):
     for key, value in defaults.items():
         if getattr(args, key, None) is None:
             setattr(args, key, value)
+
+    # reference-flag aliases
+    if getattr(args, "recompute_activations", False) and \
+            args.recompute_granularity is None:
+        args.recompute_granularity = "selective"
+    # T5-style scripts set encoder-* instead of the generic names;
+    # when given, they win (the generic flags keep argparse defaults)
+    if getattr(args, "encoder_num_layers", None):
+        args.num_layers = args.encoder_num_layers
+    if getattr(args, "encoder_seq_length", None):
+        args.seq_length = args.encoder_seq_length
 
     args.rank = int(os.getenv("RANK", "0"))
     args.world_size = int(os.getenv("WORLD_SIZE", "1"))
