@@ -38,6 +38,7 @@ class GraphDecodeContext(InferenceParams):
     """Inference KV-cache whose append path is hipGraph-replayable."""
 
     is_graph_context = True
+    window = None   # None = whole cache (class default survives __new__)
 
     def __init__(self, max_batch_size: int, max_sequence_length: int):
         super().__init__(max_batch_size, max_sequence_length)
@@ -47,6 +48,10 @@ class GraphDecodeContext(InferenceParams):
         self.cur_len = torch.zeros(1, dtype=torch.long, device=self.device)
         self._arange = torch.arange(max_sequence_length, device=self.device)
         self.graph_mode = False  # True once shapes must stay fixed
+        # attention window for the CURRENT capture/replay: a graph
+        # captured at window W only ever reads cache[:W], so short
+        # contexts stop paying for the full max-length window
+        self.window = max_sequence_length
 
     # -- offset bookkeeping -------------------------------------------------
     @property
@@ -84,15 +89,52 @@ class GraphDecodeContext(InferenceParams):
         idx = self._arange[:s_new] + self.cur_len
         k_cache.index_copy_(0, idx, key)
         v_cache.index_copy_(0, idx, value)
-        # full fixed-shape window; DotProductAttention masks the tail
-        return k_cache, v_cache
+        # fixed-shape window slice (= whole cache unless bucketed);
+        # DotProductAttention masks the tail
+        w = self.window or k_cache.shape[0]
+        return k_cache[:w], v_cache[:w]
 
     def decode_padding_mask(self, sq: int, batch: int) -> torch.Tensor:
-        """[b, 1, sq, max_len] bool, True = masked.  Valid keys for query
+        """[b, 1, sq, window] bool, True = masked.  Valid keys for query
         row r (global position cur_len + r) are positions <= cur_len + r."""
         pos_q = (self.cur_len + self._arange[:sq]).view(1, 1, sq, 1)
-        pos_k = self._arange.view(1, 1, 1, -1)
+        w = self.window or self.max_sequence_length
+        pos_k = self._arange[:w].view(1, 1, 1, -1)
         return (pos_k > pos_q).expand(batch, 1, sq, -1)
+
+
+class BucketedGraphedDecodeStep:
+    """Lengths-bucketed graph set: one captured graph per power-of-two
+    attention window, selected by the current offset — long-context
+    decode stops attending over the full max-length window (the cost
+    that shrank the round-1 2.3x win as context grew)."""
+
+    MIN_BUCKET = 256
+
+    def __init__(self, model, ctx: GraphDecodeContext, batch_size: int):
+        self.model = model
+        self.ctx = ctx
+        self.batch_size = batch_size
+        self.graphs = {}
+
+    @staticmethod
+    def _bucket_for(needed: int, max_len: int, min_bucket: int) -> int:
+        w = min_bucket
+        while w < needed:
+            w *= 2
+        return min(w, max_len)
+
+    @torch.no_grad()
+    def __call__(self, tokens: torch.Tensor, pos: torch.Tensor):
+        needed = self.ctx._offset_int + 1
+        w = self._bucket_for(needed, self.ctx.max_sequence_length,
+                             self.MIN_BUCKET)
+        if w not in self.graphs:
+            self.ctx.window = w
+            self.graphs[w] = GraphedDecodeStep(self.model, self.ctx,
+                                               self.batch_size)
+        self.ctx.window = w
+        return self.graphs[w](tokens, pos)
 
 
 class GraphedDecodeStep:

@@ -98,9 +98,9 @@ class TextGenerationController:
                 pos, pos + cur_len, device=device).unsqueeze(0).expand(b, -1)
             if use_graphs and cur_len == 1:
                 if graph_step is None:
-                    from ..hip_graphs import GraphedDecodeStep
-                    graph_step = GraphedDecodeStep(self.model,
-                                                   inference_params, b)
+                    from ..hip_graphs import BucketedGraphedDecodeStep
+                    graph_step = BucketedGraphedDecodeStep(
+                        self.model, inference_params, b)
                 logits = graph_step(step_tokens, position_ids)
             else:
                 logits = self.model(step_tokens, position_ids,

@@ -166,3 +166,43 @@ def test_graphed_decode_rope_model_gpu():
     for a, b in zip(got, ref):
         assert torch.equal(a, b), (got, ref)
     destroy()
+
+
+@pytest.mark.gpu
+def test_bucketed_graph_decode_crosses_buckets():
+    """Lengths-bucketed graphs: decoding across a bucket boundary (256 ->
+    512 window) matches eager decode token for token."""
+    from megatronapp_amd.core.hip_graphs import (BucketedGraphedDecodeStep,
+                                                 GraphDecodeContext)
+    from megatronapp_amd.core.inference_params import InferenceParams
+    initialize_model_parallel()
+    with torch.device("cuda"):
+        m = _tiny(torch.bfloat16).eval()
+    b, plen, n_new = 2, 250, 16    # crosses 256 at step 6
+    tok = torch.randint(0, 256, (b, plen), device="cuda")
+    pos = torch.arange(plen, device="cuda").unsqueeze(0).expand(b, -1)
+
+    def run(graphed):
+        torch.manual_seed(0)
+        ctx = GraphDecodeContext(b, 1024) if graphed else \
+            InferenceParams(b, 1024)
+        with torch.no_grad():
+            logits = m(tok, pos, None, inference_context=ctx)
+            ctx.increment_sequence_len_offset(plen)
+            out = [logits[:, -1].argmax(-1)]
+            step = BucketedGraphedDecodeStep(m, ctx, b) if graphed else None
+            for i in range(n_new - 1):
+                t = out[-1].unsqueeze(1)
+                p = torch.full((b, 1), plen + i, device="cuda",
+                               dtype=torch.long)
+                if graphed:
+                    lg = step(t, p)
+                else:
+                    lg = m(t, p, None, inference_context=ctx)
+                    ctx.increment_sequence_len_offset(1)
+                out.append(lg[:, -1].argmax(-1))
+        return torch.stack(out, 1)
+
+    eager = run(False)
+    graphed = run(True)
+    assert torch.equal(eager, graphed), (eager, graphed)
