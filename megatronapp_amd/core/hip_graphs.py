@@ -131,8 +131,13 @@ class BucketedGraphedDecodeStep:
                              self.MIN_BUCKET)
         if w not in self.graphs:
             self.ctx.window = w
+            # all buckets share ONE memory pool: capturing a second
+            # graph against a fresh pool after replaying the first
+            # faults the allocator (observed as a device abort)
+            pool = next(iter(self.graphs.values())).graph.pool() \
+                if self.graphs else None
             self.graphs[w] = GraphedDecodeStep(self.model, self.ctx,
-                                               self.batch_size)
+                                               self.batch_size, pool=pool)
         self.ctx.window = w
         return self.graphs[w](tokens, pos)
 
@@ -141,7 +146,7 @@ class GraphedDecodeStep:
     """Capture model(tokens, pos) single-token decode into one hipGraph."""
 
     def __init__(self, model, ctx: GraphDecodeContext, batch_size: int,
-                 warmup_iters: int = 3):
+                 warmup_iters: int = 3, pool=None):
         self.model = model
         self.ctx = ctx
         device = ctx.device
@@ -164,7 +169,7 @@ class GraphedDecodeStep:
 
         ctx.cur_len.fill_(start_len)
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        with torch.cuda.graph(self.graph, pool=pool):
             self.static_logits = self.model(
                 self.static_tokens, self.static_pos, None,
                 inference_context=ctx)

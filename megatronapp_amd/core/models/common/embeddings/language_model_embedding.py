@@ -52,6 +52,14 @@ class LanguageModelEmbedding(MegatronModule):
         else:
             embeddings = word_embeddings.transpose(0, 1).contiguous()  # [s, b, h]
         if self.add_position_embedding:
+            # out-of-range lookups are an async DEVICE FAULT (core dump
+            # with no traceback) — fail with a real error instead
+            if position_ids.numel() and int(position_ids.max()) >= \
+                    self.position_embeddings.num_embeddings:
+                raise ValueError(
+                    f"position id {int(position_ids.max())} exceeds "
+                    f"max_sequence_length "
+                    f"{self.position_embeddings.num_embeddings}")
             embeddings = embeddings + self.position_embeddings(
                 position_ids).transpose(0, 1)
         if tokentype_ids is not None and self.tokentype_embeddings is not None:

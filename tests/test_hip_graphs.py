@@ -5,7 +5,7 @@ import torch
 from tests.utils import initialize_model_parallel, destroy
 
 
-def _tiny(dtype=torch.float32):
+def _tiny(dtype=torch.float32, max_seq=64):
     from megatronapp_amd.core.models.gpt import GPTModel
     from megatronapp_amd.core.models.gpt.gpt_layer_specs import (
         get_gpt_layer_local_spec)
@@ -22,7 +22,7 @@ def _tiny(dtype=torch.float32):
     return GPTModel(config=cfg,
                     transformer_layer_spec=get_gpt_layer_local_spec(
                         use_flash=False),
-                    vocab_size=256, max_sequence_length=64,
+                    vocab_size=256, max_sequence_length=max_seq,
                     pre_process=True, post_process=True)
 
 
@@ -177,7 +177,7 @@ def test_bucketed_graph_decode_crosses_buckets():
     from megatronapp_amd.core.inference_params import InferenceParams
     initialize_model_parallel()
     with torch.device("cuda"):
-        m = _tiny(torch.bfloat16).eval()
+        m = _tiny(torch.bfloat16, max_seq=1024).eval()
     b, plen, n_new = 2, 250, 16    # crosses 256 at step 6
     tok = torch.randint(0, 256, (b, plen), device="cuda")
     pos = torch.arange(plen, device="cuda").unsqueeze(0).expand(b, -1)
