@@ -17,6 +17,7 @@ MI355X design notes:
 
 from __future__ import annotations
 
+import os
 import warnings
 from typing import Callable, Optional
 
@@ -234,13 +235,32 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         # wgrad GEMM overlaps the collective above
         grad_output_2d = grad_output.reshape(-1, grad_output.shape[-1])
         total_input_2d = total_input.reshape(-1, total_input.shape[-1])
+        fused_dbias = None
         if ctx.gradient_accumulation_fusion and hasattr(weight, "main_grad"):
             if (grad_output.is_cuda and _ops.have_ops()
                     and grad_output.dtype == torch.bfloat16):
-                # fp32-accumulating hipblasLt wgrad straight into main_grad
-                _ops.get_ops().wgrad_accum(
-                    grad_output_2d.contiguous(), total_input_2d.contiguous(),
-                    weight.main_grad)
+                # fp32-accumulating hipblasLt wgrad straight into main_grad;
+                # when this linear also owns a fused bias grad, ride the
+                # BGRADB epilogue so dbias costs no extra HBM pass
+                go2 = grad_output_2d.contiguous()
+                ti2 = total_input_2d.contiguous()
+                lt = _ops.get_ops()
+                bias_param = ctx.bias_param if use_bias else None
+                want_bgrad = (
+                    bias_param is not None
+                    and hasattr(bias_param, "main_grad")
+                    and hasattr(bias_param, "grad_added_to_main_grad")
+                    and "bgrad" not in os.environ.get(
+                        "MEGATRONAPP_DISABLE_FUSED", ""))
+                if want_bgrad:
+                    dbias_tmp = torch.empty_like(bias_param.main_grad)
+                    if lt.wgrad_accum_bgrad(go2, ti2, weight.main_grad,
+                                            dbias_tmp):
+                        fused_dbias = dbias_tmp
+                    else:          # no epilogue algo for this shape
+                        lt.wgrad_accum(go2, ti2, weight.main_grad)
+                else:
+                    lt.wgrad_accum(go2, ti2, weight.main_grad)
             else:
                 weight.main_grad.add_(
                     torch.matmul(grad_output_2d.t(), total_input_2d))
@@ -256,7 +276,14 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
         else:
             grad_weight = grad_output_2d.t().matmul(total_input_2d)
         grad_bias = None
-        if use_bias:
+        if use_bias and fused_dbias is not None:
+            bias_param = ctx.bias_param
+            bias_param.main_grad.add_(fused_dbias)
+            bias_param.grad_added_to_main_grad = True
+            grad_bias = torch.empty(bias_param.shape,
+                                    dtype=bias_param.dtype,
+                                    device=bias_param.device)
+        elif use_bias:
             bias_param = ctx.bias_param
             if (grad_output.is_cuda and _ops.have_ops() and
                     grad_output.dtype == torch.bfloat16 and

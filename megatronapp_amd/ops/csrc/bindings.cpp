@@ -55,6 +55,8 @@ void launch_selective_scan_fwd(const void*, const void*, const float*,
 void launch_adamw_flat(float*, const float*, float*, float*, long, float,
                        float, float, float, float, int, hipStream_t);
 void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
+bool wgrad_accum_bgrad(torch::Tensor, torch::Tensor, torch::Tensor,
+                       torch::Tensor);
 torch::Tensor gemm_nt(torch::Tensor, torch::Tensor);
 torch::Tensor gemm_nn(torch::Tensor, torch::Tensor);
 void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
@@ -216,6 +218,8 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, c10::optional<torch::Tensor> bias) 
   check_bf16(x, "x");
   auto y = torch::empty_like(x);
   const int F = (int)x.size(-1);
+  TORCH_CHECK(x.numel() % 8 == 0 && (!bias.has_value() || F % 8 == 0),
+              "bias_gelu: numel and F must be multiples of 8");
   launch_bias_gelu_fwd(x.data_ptr(),
                        bias.has_value() ? bias->data_ptr() : nullptr,
                        y.data_ptr(), x.numel(), F, cur_stream());
@@ -242,6 +246,8 @@ torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   check_bf16(dy, "dy");
   auto dx = torch::empty_like(x);
   const int F = (int)x.size(-1);
+  TORCH_CHECK(x.numel() % 8 == 0 && (!bias.has_value() || F % 8 == 0),
+              "bias_gelu: numel and F must be multiples of 8");
   launch_bias_gelu_bwd(dy.data_ptr(), x.data_ptr(),
                        bias.has_value() ? bias->data_ptr() : nullptr,
                        dx.data_ptr(), x.numel(), F, cur_stream());
@@ -769,6 +775,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bias_add_residual", &bias_add_residual);
   mod.def("adamw_flat_ranged", &adamw_flat_ranged);
   mod.def("wgrad_accum", &wgrad_accum);
+  mod.def("wgrad_accum_bgrad", &wgrad_accum_bgrad);
   mod.def("attn_fwd", &attn_fwd);
   mod.def("attn_fwd_ablate", &attn_fwd_ablate);
   mod.def("attn_bwd", &attn_bwd);

@@ -63,11 +63,16 @@ __global__ void bias_gelu_bwd_kernel(const unsigned short* __restrict__ dy,
   for (; i < n; i += stride) {
     short8v v = *(const short8v*)(x + i);
     short8v d = *(const short8v*)(dy + i);
+    // one vector bias load + one i64 modulo per 8 elements (a per-element
+    // `bias[(i+j) % F]` modulo made this kernel VALU-bound: 454 us vs the
+    // ~200 us HBM bound at mbs16, profiles/r02_flash_final_stats.csv)
+    short8v bv{};
+    if (bias != nullptr) bv = *(const short8v*)(bias + (i % F));
     short8v o;
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       float xf = bf2f((unsigned short)v[j]);
-      if (bias != nullptr) xf += bf2f((unsigned short)bias[(i + j) % F]);
+      if (bias != nullptr) xf += bf2f((unsigned short)bv[j]);
       o[j] = (short)f2bf(bf2f((unsigned short)d[j]) * gelu_tanh_grad(xf));
     }
     *(short8v*)(dx + i) = o;
@@ -322,6 +327,10 @@ __global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
   const long r0 = (long)blockIdx.y * rows_per_block;
   const long r1 = min(R, r0 + rows_per_block);
   float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  // 4 rows in flight per thread: the rolled loop kept one outstanding
+  // HBM load per lane and the kernel ran latency-bound at 3-4x the
+  // bandwidth bound (121 us avg, profiles/r02_flash_final_stats.csv)
+#pragma unroll 4
   for (long r = r0; r < r1; ++r) {
     short8v v = *(const short8v*)(dy + r * F + col);
 #pragma unroll
@@ -355,12 +364,13 @@ void launch_colsum_accum(const void* dy, float* out, long R, int F,
     return;
   }
   const int gx = (F / 8 + 255) / 256;
-  // atomic traffic per column == grid.y.  Keep each block's row strip
-  // >=128 rows (low atomic rate) but scale grid.y with R so large
-  // activations (mbs16: R=32k) still fill the 256 CUs.
-  long target_y = R / 128;
+  // atomic traffic per column == grid.y; the reduction is HBM-latency
+  // bound, so oversubscribe: ~32-row strips give 4 blocks/CU at F=2048
+  // (16 waves/CU of loads in flight) and the extra fp32 atomics are
+  // noise next to the streamed reads.
+  long target_y = R / 32;
   if (target_y < 64) target_y = 64;
-  if (target_y > 256) target_y = 256;
+  if (target_y > 1024) target_y = 1024;
   int rpb = (int)((R + target_y - 1) / target_y);
   if (rpb < 8) rpb = 8;
   dim3 grid(gx, (unsigned)((R + rpb - 1) / rpb));

@@ -830,3 +830,33 @@ def test_fused_bias_dropout_add_statistics_and_grads():
     assert torch.allclose(x.grad[kept].float(),
                           (dy[kept].float() / (1 - p)), atol=3e-2,
                           rtol=3e-2)
+
+
+@pytest.mark.gpu
+def test_wgrad_accum_bgrad_matches_eager():
+    """BGRADB-epilogue wgrad: main_grad += g^T @ x AND dbias = colsum(g)
+    in one hipblasLt call (tensor_parallel/layers.py backward)."""
+    torch.manual_seed(5)
+    ops = _ops()
+    supported = 0
+    for rows, out, in_ in [(4096, 1536, 512), (512, 768, 256),
+                           (333, 128, 96)]:
+        g = torch.randn(rows, out, device="cuda", dtype=torch.bfloat16)
+        x = torch.randn(rows, in_, device="cuda", dtype=torch.bfloat16)
+        mg = torch.randn(out, in_, device="cuda", dtype=torch.float32)
+        mg0 = mg.clone()
+        dbias = torch.full((out,), 7.0, device="cuda", dtype=torch.float32)
+        if not ops.wgrad_accum_bgrad(g, x, mg, dbias):
+            # no epilogue algo for this shape: main_grad must be untouched
+            assert torch.equal(mg, mg0), "failed call corrupted main_grad"
+            continue
+        supported += 1
+        ref_w = mg0 + g.float().t() @ x.float()
+        ref_b = g.float().sum(0)
+        werr = (mg - ref_w).abs().max().item()
+        wrel = werr / (ref_w.abs().max().item() + 1e-6)
+        berr = (dbias - ref_b).abs().max().item()
+        brel = berr / (ref_b.abs().max().item() + 1e-6)
+        assert wrel < 2e-2, f"wgrad rel err {wrel} at {rows}x{out}x{in_}"
+        assert brel < 2e-2, f"dbias rel err {brel} at {rows}x{out}x{in_}"
+    assert supported > 0, "BGRADB epilogue unsupported for every shape"
