@@ -59,7 +59,9 @@ bool wgrad_accum_bgrad(torch::Tensor, torch::Tensor, torch::Tensor,
                        torch::Tensor);
 torch::Tensor gemm_nt(torch::Tensor, torch::Tensor);
 torch::Tensor gemm_nn(torch::Tensor, torch::Tensor);
-void launch_colsum_accum(const void*, float*, long, int, hipStream_t);
+int colsum_grid_y(long, int);
+void launch_colsum_accum(const void*, float*, float*, int, long, int,
+                         hipStream_t);
 void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
 void launch_quant_rows_e4m3(const void*, void*, float*, long, int,
@@ -410,7 +412,15 @@ torch::Tensor colsum_accum(torch::Tensor dy, torch::Tensor out) {
   TORCH_CHECK(out.scalar_type() == torch::kFloat32 && out.is_contiguous());
   const int F = (int)dy.size(-1);
   const long R = dy.numel() / F;
-  launch_colsum_accum(dy.data_ptr(), out.data_ptr<float>(), R, F,
+  const int gy = colsum_grid_y(R, F);
+  torch::Tensor part;
+  float* part_p = nullptr;
+  if (gy > 0) {
+    part = torch::empty({(long)gy, (long)F},
+                        out.options().dtype(torch::kFloat32));
+    part_p = part.data_ptr<float>();
+  }
+  launch_colsum_accum(dy.data_ptr(), out.data_ptr<float>(), part_p, gy, R, F,
                       cur_stream());
   return out;
 }

@@ -60,6 +60,10 @@ __global__ void bias_gelu_bwd_kernel(const unsigned short* __restrict__ dy,
                                      int F) {
   long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC;
   const long stride = (long)gridDim.x * BLOCK * VEC;
+  // unroll 2: two (x, dy) load pairs in flight per lane — the rolled
+  // loop left ~20% of HBM bandwidth on the table (362 us vs ~290
+  // achievable for 3 passes at mbs16)
+#pragma unroll 2
   for (; i < n; i += stride) {
     short8v v = *(const short8v*)(x + i);
     short8v d = *(const short8v*)(dy + i);
@@ -319,17 +323,19 @@ void launch_bias_geglu_bwd(const void* dy, const void* x, const void* bias,
   HIP_CHECK_LAUNCH();
 }
 
-__global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
-                                    float* __restrict__ out, long R, int F,
-                                    int rows_per_block) {
+// Stage 1: per-block partial colsums into scratch[gy, F] (plain stores).
+// The single-kernel atomicAdd version serialized grid.y fp32 atomics on
+// every column (measured 121-259 us for a 17-67 us HBM-bound read) and
+// made bias grads nondeterministic; two deterministic stages are both
+// faster and reproducible.
+__global__ void colsum_part_kernel(const unsigned short* __restrict__ dy,
+                                   float* __restrict__ part, long R, int F,
+                                   int rows_per_block) {
   const int col = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
   if (col >= F) return;
   const long r0 = (long)blockIdx.y * rows_per_block;
   const long r1 = min(R, r0 + rows_per_block);
   float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  // 4 rows in flight per thread: the rolled loop kept one outstanding
-  // HBM load per lane and the kernel ran latency-bound at 3-4x the
-  // bandwidth bound (121 us avg, profiles/r02_flash_final_stats.csv)
 #pragma unroll 4
   for (long r = r0; r < r1; ++r) {
     short8v v = *(const short8v*)(dy + r * F + col);
@@ -337,7 +343,17 @@ __global__ void colsum_accum_kernel(const unsigned short* __restrict__ dy,
     for (int j = 0; j < 8; ++j) acc[j] += bf2f((unsigned short)v[j]);
   }
 #pragma unroll
-  for (int j = 0; j < 8; ++j) atomicAdd(out + col + j, acc[j]);
+  for (int j = 0; j < 8; ++j) part[(long)blockIdx.y * F + col + j] = acc[j];
+}
+
+// Stage 2: out[col] += sum_y part[y, col] (one writer per column).
+__global__ void colsum_reduce_kernel(const float* __restrict__ part,
+                                     float* __restrict__ out, int gy, int F) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= F) return;
+  float acc = 0.f;
+  for (int y = 0; y < gy; ++y) acc += part[(long)y * F + col];
+  out[col] += acc;
 }
 
 
@@ -354,8 +370,18 @@ __global__ void colsum_accum_scalar_kernel(const unsigned short* __restrict__ dy
   atomicAdd(out + col, acc);
 }
 
-void launch_colsum_accum(const void* dy, float* out, long R, int F,
-                         hipStream_t s) {
+int colsum_grid_y(long R, int F) {
+  if (F % 8 != 0) return 0;  // scalar/atomic fallback path
+  long target_y = R / 16;    // short strips: many loads in flight
+  if (target_y < 64) target_y = 64;
+  if (target_y > 1024) target_y = 1024;
+  int rpb = (int)((R + target_y - 1) / target_y);
+  if (rpb < 8) rpb = 8;
+  return (int)((R + rpb - 1) / rpb);
+}
+
+void launch_colsum_accum(const void* dy, float* out, float* part, int gy,
+                         long R, int F, hipStream_t s) {
   if (F % 8 != 0) {
     dim3 grid((F + 255) / 256, (unsigned)((R + 127) / 128));
     hipLaunchKernelGGL(colsum_accum_scalar_kernel, grid, dim3(256), 0, s,
@@ -364,18 +390,13 @@ void launch_colsum_accum(const void* dy, float* out, long R, int F,
     return;
   }
   const int gx = (F / 8 + 255) / 256;
-  // atomic traffic per column == grid.y; the reduction is HBM-latency
-  // bound, so oversubscribe: ~32-row strips give 4 blocks/CU at F=2048
-  // (16 waves/CU of loads in flight) and the extra fp32 atomics are
-  // noise next to the streamed reads.
-  long target_y = R / 32;
-  if (target_y < 64) target_y = 64;
-  if (target_y > 1024) target_y = 1024;
-  int rpb = (int)((R + target_y - 1) / target_y);
-  if (rpb < 8) rpb = 8;
-  dim3 grid(gx, (unsigned)((R + rpb - 1) / rpb));
-  hipLaunchKernelGGL(colsum_accum_kernel, grid, dim3(256), 0, s,
-                     (const unsigned short*)dy, out, R, F, rpb);
+  const int rpb = (int)((R + gy - 1) / gy);
+  dim3 grid(gx, (unsigned)gy);
+  hipLaunchKernelGGL(colsum_part_kernel, grid, dim3(256), 0, s,
+                     (const unsigned short*)dy, part, R, F, rpb);
+  HIP_CHECK_LAUNCH();
+  hipLaunchKernelGGL(colsum_reduce_kernel, dim3((F + 255) / 256), dim3(256),
+                     0, s, part, out, gy, F);
   HIP_CHECK_LAUNCH();
 }
 

@@ -3,6 +3,11 @@ gpt3-1.3b bench shapes (mbs16: rows = 32768).
 
   python tools/bench_eltwise.py          # on an MI355X via gpurun
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
 import torch
 
 from megatronapp_amd import ops
