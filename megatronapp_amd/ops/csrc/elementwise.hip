@@ -60,10 +60,6 @@ __global__ void bias_gelu_bwd_kernel(const unsigned short* __restrict__ dy,
                                      int F) {
   long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * VEC;
   const long stride = (long)gridDim.x * BLOCK * VEC;
-  // unroll 2: two (x, dy) load pairs in flight per lane — the rolled
-  // loop left ~20% of HBM bandwidth on the table (362 us vs ~290
-  // achievable for 3 passes at mbs16)
-#pragma unroll 2
   for (; i < n; i += stride) {
     short8v v = *(const short8v*)(x + i);
     short8v d = *(const short8v*)(dy + i);
@@ -346,14 +342,29 @@ __global__ void colsum_part_kernel(const unsigned short* __restrict__ dy,
   for (int j = 0; j < 8; ++j) part[(long)blockIdx.y * F + col + j] = acc[j];
 }
 
-// Stage 2: out[col] += sum_y part[y, col] (one writer per column).
+// Stage 2: out[col] += sum_y part[y, col].  One block per 8 columns,
+// 32 y-lanes each, LDS tree at the end: F/8 blocks keep the whole chip
+// busy (a flat one-thread-per-column version left only F lanes walking
+// gy strided loads each and ran slower than the atomics it replaced).
 __global__ void colsum_reduce_kernel(const float* __restrict__ part,
                                      float* __restrict__ out, int gy, int F) {
-  const int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= F) return;
+  __shared__ float red[256];
+  const int c = threadIdx.x & 7;
+  const int ty = threadIdx.x >> 3;
+  const int col = blockIdx.x * 8 + c;
   float acc = 0.f;
-  for (int y = 0; y < gy; ++y) acc += part[(long)y * F + col];
-  out[col] += acc;
+  if (col < F) {
+#pragma unroll 4
+    for (int y = ty; y < gy; y += 32) acc += part[(long)y * F + col];
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  if (ty == 0 && col < F) {
+    float s = 0.f;
+#pragma unroll
+    for (int t = 0; t < 32; ++t) s += red[t * 8 + c];
+    out[col] += s;
+  }
 }
 
 
@@ -372,9 +383,9 @@ __global__ void colsum_accum_scalar_kernel(const unsigned short* __restrict__ dy
 
 int colsum_grid_y(long R, int F) {
   if (F % 8 != 0) return 0;  // scalar/atomic fallback path
-  long target_y = R / 16;    // short strips: many loads in flight
+  long target_y = R / 32;    // short strips: many loads in flight
   if (target_y < 64) target_y = 64;
-  if (target_y > 1024) target_y = 1024;
+  if (target_y > 512) target_y = 512;
   int rpb = (int)((R + target_y - 1) / target_y);
   if (rpb < 8) rpb = 8;
   return (int)((R + rpb - 1) / rpb);
@@ -395,7 +406,7 @@ void launch_colsum_accum(const void* dy, float* out, float* part, int gy,
   hipLaunchKernelGGL(colsum_part_kernel, grid, dim3(256), 0, s,
                      (const unsigned short*)dy, part, R, F, rpb);
   HIP_CHECK_LAUNCH();
-  hipLaunchKernelGGL(colsum_reduce_kernel, dim3((F + 255) / 256), dim3(256),
+  hipLaunchKernelGGL(colsum_reduce_kernel, dim3((F + 7) / 8), dim3(256),
                      0, s, part, out, gy, F);
   HIP_CHECK_LAUNCH();
 }

@@ -865,3 +865,25 @@ def test_wgrad_accum_bgrad_matches_eager():
         # path auto-activates if a future build adds fp32-D support
         pytest.skip("no fp32-D BGRADB algos in this hipblaslt; "
                     "fallback verified")
+
+
+@pytest.mark.gpu
+def test_colsum_accum_matches_and_deterministic():
+    """Two-stage colsum: numerics vs fp32 sum and bitwise run-to-run
+    determinism (the atomicAdd version it replaced was neither fast nor
+    deterministic)."""
+    ops = _ops()
+    torch.manual_seed(9)
+    for R, F in [(32768, 2048), (4096, 8192), (1000, 120), (777, 333)]:
+        dy = torch.randn(R, F, device="cuda", dtype=torch.bfloat16)
+        out1 = torch.randn(F, device="cuda", dtype=torch.float32)
+        base = out1.clone()
+        ops.colsum_accum(dy, out1)
+        ref = base + dy.float().sum(0)
+        rel = ((out1 - ref).abs().max() /
+               (ref.abs().max() + 1e-6)).item()
+        assert rel < 2e-2, f"colsum rel err {rel} at {R}x{F}"
+        if F % 8 == 0:  # vector path is deterministic by construction
+            out2 = base.clone()
+            ops.colsum_accum(dy, out2)
+            assert torch.equal(out1, out2)
