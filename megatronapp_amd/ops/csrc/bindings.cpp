@@ -151,7 +151,9 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dw = mg_w.has_value()
                 ? *mg_w
                 : torch::zeros({H}, x.options().dtype(torch::kFloat32));
-  const int grid = (int)std::min<long>(N, 512);
+  // 2048 blocks = up to 8/CU: a block sits at two block-wide barrier
+  // reductions per row, so a 512-block launch left the CUs idle
+  const int grid = (int)std::min<long>(N, 2048);
   auto dw_part = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
   const void* dres_p = nullptr;
   torch::Tensor dres_c;
@@ -197,7 +199,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto db = mg_b.has_value()
                 ? *mg_b
                 : torch::zeros({H}, x.options().dtype(torch::kFloat32));
-  const int grid = (int)std::min<long>(N, 512);
+  // 2048 blocks = up to 8/CU: a block sits at two block-wide barrier
+  // reductions per row, so a 512-block launch left the CUs idle
+  const int grid = (int)std::min<long>(N, 2048);
   auto part = torch::empty({2, grid, H}, x.options().dtype(torch::kFloat32));
   const void* dres_p = nullptr;
   torch::Tensor dres_c;

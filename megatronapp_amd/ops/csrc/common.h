@@ -60,6 +60,40 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* lds) {
   return out;
 }
 
+// two sums in one barrier round-trip (callers with paired row statistics
+// pay 3 __syncthreads instead of 6); lds must hold 2*BLOCK/WAVE floats
+template <int BLOCK>
+__device__ __forceinline__ float2 block_reduce_sum2(float a, float b,
+                                                    float* lds) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  a = wave_reduce_sum(a);
+  b = wave_reduce_sum(b);
+  constexpr int NW = BLOCK / WAVE;
+  if (lane == 0) {
+    lds[wid] = a;
+    lds[NW + wid] = b;
+  }
+  __syncthreads();
+  float ra = (threadIdx.x < NW) ? lds[threadIdx.x] : 0.f;
+  float rb = (threadIdx.x < NW) ? lds[NW + threadIdx.x] : 0.f;
+  if (wid == 0) {
+#pragma unroll
+    for (int off = NW / 2; off > 0; off >>= 1) {
+      ra += __shfl_down(ra, off, WAVE);
+      rb += __shfl_down(rb, off, WAVE);
+    }
+    if (lane == 0) {
+      lds[0] = ra;
+      lds[1] = rb;
+    }
+  }
+  __syncthreads();
+  float2 out = make_float2(lds[0], lds[1]);
+  __syncthreads();
+  return out;
+}
+
 template <int BLOCK>
 __device__ __forceinline__ float block_reduce_max(float v, float* lds) {
   const int lane = threadIdx.x & (WAVE - 1);

@@ -79,13 +79,17 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
     unsigned short* dxr = dx + (long)row * H;
     const float r = invrms[row];
     float dot = 0.f;
+    // row tiles stay in registers between the two passes (<= 48 VGPRs
+    // at CHUNKS=4): re-reading x/dy/w for the dx pass doubled the HBM
+    // traffic of this kernel
+    short8v xs[CHUNKS], dvs[CHUNKS], wvs[CHUNKS];
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
       const int base = threadIdx.x * VEC + c * BLOCK * VEC;
       if (base >= H) break;
-      short8v xv = *(const short8v*)(xr + base);
-      short8v dv = *(const short8v*)(dyr + base);
-      short8v wv = *(const short8v*)(w + base);
+      short8v xv = xs[c] = *(const short8v*)(xr + base);
+      short8v dv = dvs[c] = *(const short8v*)(dyr + base);
+      short8v wv = wvs[c] = *(const short8v*)(w + base);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         float xh = bf2f((unsigned short)xv[j]) * r;
@@ -99,9 +103,9 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
     for (int c = 0; c < CHUNKS; ++c) {
       const int base = threadIdx.x * VEC + c * BLOCK * VEC;
       if (base >= H) break;
-      short8v xv = *(const short8v*)(xr + base);
-      short8v dv = *(const short8v*)(dyr + base);
-      short8v wv = *(const short8v*)(w + base);
+      short8v xv = xs[c];
+      short8v dv = dvs[c];
+      short8v wv = wvs[c];
       short8v o;
       // residual-join fusion: dx += the BDA residual's grad in the same
       // pass (saves a standalone [rows, H] add kernel per layer)
@@ -152,7 +156,7 @@ __global__ void layernorm_fwd_kernel(const unsigned short* __restrict__ x,
                                      float* __restrict__ mean,
                                      float* __restrict__ invstd, int N, int H,
                                      float eps) {
-  __shared__ float lds[BLOCK / WAVE];
+  __shared__ float lds[2 * BLOCK / WAVE];
   for (int row = blockIdx.x; row < N; row += gridDim.x) {
     const unsigned short* xr = x + (long)row * H;
     unsigned short* yr = y + (long)row * H;
@@ -166,8 +170,9 @@ __global__ void layernorm_fwd_kernel(const unsigned short* __restrict__ x,
         ss += f * f;
       }
     }
-    s = block_reduce_sum<BLOCK>(s, lds);
-    ss = block_reduce_sum<BLOCK>(ss, lds);
+    float2 sss = block_reduce_sum2<BLOCK>(s, ss, lds);
+    s = sss.x;
+    ss = sss.y;
     float mu = s / H;
     float var = ss / H - mu * mu;
     float r = rsqrtf(var + eps);
@@ -203,7 +208,7 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
                                      float* __restrict__ db_part,
                                      const unsigned short* __restrict__ dres,
                                      int N, int H) {
-  __shared__ float lds[BLOCK / WAVE];
+  __shared__ float lds[2 * BLOCK / WAVE];
   float dwacc[CHUNKS * VEC];
   float dbacc[CHUNKS * VEC];
 #pragma unroll
@@ -219,13 +224,15 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
     const float mu = mean[row];
     const float r = invstd[row];
     float sum1 = 0.f, sum2 = 0.f;
+    // register-resident row tiles between the passes (see rmsnorm_bwd)
+    short8v xs[CHUNKS], dvs[CHUNKS], wvs[CHUNKS];
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
       const int base = threadIdx.x * VEC + c * BLOCK * VEC;
       if (base >= H) break;
-      short8v xv = *(const short8v*)(xr + base);
-      short8v dv = *(const short8v*)(dyr + base);
-      short8v wv = *(const short8v*)(w + base);
+      short8v xv = xs[c] = *(const short8v*)(xr + base);
+      short8v dv = dvs[c] = *(const short8v*)(dyr + base);
+      short8v wv = wvs[c] = *(const short8v*)(w + base);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         float xh = (bf2f((unsigned short)xv[j]) - mu) * r;
@@ -237,15 +244,16 @@ __global__ void layernorm_bwd_kernel(const unsigned short* __restrict__ dy,
         dbacc[c * VEC + j] += dyf;
       }
     }
-    sum1 = block_reduce_sum<BLOCK>(sum1, lds) / H;
-    sum2 = block_reduce_sum<BLOCK>(sum2, lds) / H;
+    float2 s12 = block_reduce_sum2<BLOCK>(sum1, sum2, lds);
+    sum1 = s12.x / H;
+    sum2 = s12.y / H;
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
       const int base = threadIdx.x * VEC + c * BLOCK * VEC;
       if (base >= H) break;
-      short8v xv = *(const short8v*)(xr + base);
-      short8v dv = *(const short8v*)(dyr + base);
-      short8v wv = *(const short8v*)(w + base);
+      short8v xv = xs[c];
+      short8v dv = dvs[c];
+      short8v wv = wvs[c];
       short8v o;
       short8v rv;
       if (dres != nullptr)

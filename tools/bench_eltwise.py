@@ -49,6 +49,23 @@ def main():
     us = timeit(lambda: o.bias_gelu_bwd(dy, x, b))
     print(f"  {us:7.1f} us   (HBM bound {R * F * 6 / 8e12 * 1e6:5.1f} us)")
 
+    print("== layernorm_bwd [32768, 2048] + dres (bound ~ 4 bf16 passes) ==")
+    H = 2048
+    x2 = torch.randn(R, H, device="cuda", dtype=torch.bfloat16)
+    dy2 = torch.randn(R, H, device="cuda", dtype=torch.bfloat16)
+    w2 = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    b2 = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    dres = torch.randn(R, H, device="cuda", dtype=torch.bfloat16)
+    y2, mean, invstd = o.layernorm_fwd(x2, w2, b2, 1e-5)
+    us = timeit(lambda: o.layernorm_bwd(dy2, x2, w2, mean, invstd,
+                                    dres=dres))
+    print(f"  {us:7.1f} us   (HBM bound {R * H * 8 / 8e12 * 1e6:5.1f} us)")
+
+    print("== rmsnorm_bwd [32768, 2048] + dres ==")
+    yr, invrms = o.rmsnorm_fwd(x2, w2, 1e-5)
+    us = timeit(lambda: o.rmsnorm_bwd(dy2, x2, w2, invrms, dres=dres))
+    print(f"  {us:7.1f} us   (HBM bound {R * H * 8 / 8e12 * 1e6:5.1f} us)")
+
     print("== qkv wgrad: separate (wgrad_accum + colsum) vs BGRADB ==")
     out_f, in_f = 6144, 2048
     g = torch.randn(R, out_f, device="cuda", dtype=torch.bfloat16)
