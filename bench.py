@@ -67,6 +67,10 @@ def parse_args():
                    help="pipeline parallel size (default min(gpus,4) for >=4 GPUs)")
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--trace", action="store_true", help="enable MegaScan tracing")
+    p.add_argument("--torch-profile", default=None, metavar="OUT",
+                   help="run 2 extra steps under torch.profiler (with python "
+                        "stacks) after the timed loop and write the table to "
+                        "OUT (kernel attribution, not part of the metric)")
     p.add_argument("--trace-dir", default="trace_out")
     p.add_argument("--attention", default="flash", choices=["flash", "fused"])
     p.add_argument("--fp8", action="store_true",
@@ -256,6 +260,20 @@ def main():
 
     if tracer is not None:
         tracer.shutdown()
+
+    if args.torch_profile and rank == 0 and on_gpu:
+        from torch.profiler import ProfilerActivity, profile
+        cfg = torch._C._profiler._ExperimentalConfig(verbose=True)
+        with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                     with_stack=True, experimental_config=cfg) as prof:
+            for it in range(2):
+                one_step(args.warmup + args.steps + it)
+        with open(args.torch_profile, "w") as f:
+            f.write(prof.key_averages(group_by_stack_n=6).table(
+                sort_by="self_cuda_time_total", row_limit=60,
+                max_src_column_width=160))
+        prof.export_stacks(args.torch_profile + ".stacks",
+                           "self_cuda_time_total")
 
     # max elapsed over ranks
     t = torch.tensor([elapsed], dtype=torch.float64, device=device

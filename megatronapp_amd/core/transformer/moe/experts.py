@@ -22,21 +22,51 @@ def _grouped_mm_available(t: torch.Tensor) -> bool:
 
 class _GroupedMMFn(torch.autograd.Function):
     """y[seg_e] = x[seg_e] @ w[e] for jagged segments given by offs
-    (int32 inclusive cumsum of tokens_per_expert)."""
+    (int32 inclusive cumsum of tokens_per_expert).
+
+    When the stacked weight is a DDP-managed param (fp32 main_grad) and
+    host_counts is provided, the backward skips the grouped bf16 dw and
+    instead runs one fp32-accumulating hipBLASLt wgrad per expert slice
+    straight into main_grad[e] — removing both the [E, in, out] bf16 dw
+    materialization and the DDP hook's fp32 += bf16 pass over every
+    expert weight (together ~2 extra full passes over the expert params
+    per microbatch), and matching the dense linears' fp32 wgrad
+    accumulation precision."""
 
     @staticmethod
-    def forward(ctx, x, w, offs):
+    def forward(ctx, x, w, offs, host_counts=None):
         ctx.save_for_backward(x, w, offs)
+        ctx.host_counts = host_counts
+        ctx.wparam = w if host_counts is not None else None
         return torch._grouped_mm(x, w, offs=offs)
 
     @staticmethod
     def backward(ctx, dy):
+        from .... import ops as _ops
+
         x, w, offs = ctx.saved_tensors
         dy = dy.contiguous()
         dx = torch._grouped_mm(dy, w.transpose(1, 2), offs=offs)
-        # dw[e] = x[seg]^T @ dy[seg]: (2D, 2D) -> 3D grouped over tokens
-        dw = torch._grouped_mm(x.t().contiguous(), dy, offs=offs)
-        return dx, dw, None
+        wp = ctx.wparam
+        if (wp is not None and hasattr(wp, "main_grad")
+                and hasattr(wp, "grad_added_to_main_grad")
+                and wp.main_grad.dtype == torch.float32
+                and _ops.have_ops()):
+            lt = _ops.get_ops()
+            mg = wp.main_grad
+            start = 0
+            for e, n in enumerate(ctx.host_counts):
+                if n > 0:
+                    # mg[e][i, o] += sum_r x[r, i] * dy[r, o]
+                    lt.wgrad_accum(x[start:start + n], dy[start:start + n],
+                                   mg[e])
+                start += n
+            wp.grad_added_to_main_grad = True
+            dw = torch.empty_like(w)   # dummy so the DDP hook still fires
+        else:
+            # dw[e] = x[seg]^T @ dy[seg]: (2D, 2D) -> 3D grouped
+            dw = torch._grouped_mm(x.t().contiguous(), dy, offs=offs)
+        return dx, dw, None, None
 
 from ...fusions.fused_bias_act import bias_gelu_impl, bias_swiglu_impl
 from ...transformer_config import TransformerConfig
@@ -105,9 +135,21 @@ class GroupedMLP(nn.Module):
         if _grouped_mm_available(permuted_tokens):
             offs = torch.cumsum(tokens_per_expert.to(
                 device=permuted_tokens.device), 0).to(torch.int32)
-            inter = _GroupedMMFn.apply(permuted_tokens, self.weight1, offs)
+            # host token counts enable the per-expert fp32 wgrad in
+            # backward; the one-time .tolist() sync per layer is ~50 us
+            # against the ~2 full expert-weight passes it removes
+            import os as _os
+            dis = _os.environ.get("MEGATRONAPP_DISABLE_FUSED", "")
+            counts = None
+            if (torch.is_grad_enabled()
+                    and permuted_tokens.dtype == torch.bfloat16
+                    and hasattr(self.weight1, "main_grad")
+                    and not ("all" in dis or "moe_wgrad" in dis)):
+                counts = tokens_per_expert.tolist()
+            inter = _GroupedMMFn.apply(permuted_tokens, self.weight1, offs,
+                                       counts)
             inter = self.activation(inter, None)
-            return _GroupedMMFn.apply(inter, self.weight2, offs)
+            return _GroupedMMFn.apply(inter, self.weight2, offs, counts)
         outputs = []
         start = 0
         for e, n in enumerate(tokens_per_expert.tolist()):

@@ -887,3 +887,41 @@ def test_colsum_accum_matches_and_deterministic():
             out2 = base.clone()
             ops.colsum_accum(dy, out2)
             assert torch.equal(out1, out2)
+
+
+@pytest.mark.gpu
+def test_grouped_mm_fused_main_grad_wgrad():
+    """_GroupedMMFn with host_counts + a DDP-style fp32 main_grad: the
+    per-expert hipBLASLt wgrad must accumulate the same dw the grouped
+    bf16 path produces (and mark grad_added_to_main_grad)."""
+    from megatronapp_amd.core.transformer.moe.experts import _GroupedMMFn
+    torch.manual_seed(43)
+    E, h, f, T = 4, 256, 512, 1024
+    x = torch.randn(T, h, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = (torch.randn(E, h, f, device="cuda", dtype=torch.bfloat16) *
+         0.05).requires_grad_(True)
+    counts = [300, 200, 324, 200]
+    offs = torch.cumsum(torch.tensor(counts, device="cuda"), 0).to(torch.int32)
+    w.main_grad = torch.randn(E, h, f, device="cuda", dtype=torch.float32)
+    w.grad_added_to_main_grad = False
+    base = w.main_grad.clone()
+
+    y = _GroupedMMFn.apply(x, w, offs, counts)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    assert w.grad_added_to_main_grad
+
+    ref = base.clone()
+    start = 0
+    for e, n in enumerate(counts):
+        ref[e] += (x[start:start + n].float().t()
+                   @ dy[start:start + n].float())
+        start += n
+    rel = ((w.main_grad - ref).abs().max() / (ref.abs().max() + 1e-6)).item()
+    assert rel < 5e-2, rel
+    # dx unchanged vs the no-fusion path
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = _GroupedMMFn.apply(x2, w.detach().clone().requires_grad_(True), offs)
+    y2.backward(dy)
+    assert torch.equal(x.grad, x2.grad)
