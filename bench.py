@@ -67,6 +67,9 @@ def parse_args():
                    help="pipeline parallel size (default min(gpus,4) for >=4 GPUs)")
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--trace", action="store_true", help="enable MegaScan tracing")
+    p.add_argument("--moe-sequential-experts", action="store_true",
+                   help="use SequentialMLP instead of the grouped-GEMM "
+                        "experts (A/B debugging)")
     p.add_argument("--torch-profile", default=None, metavar="OUT",
                    help="run 2 extra steps under torch.profiler (with python "
                         "stacks) after the timed loop and write the table to "
@@ -183,6 +186,7 @@ def main():
             transformer_layer_spec=get_gpt_layer_local_spec(
                 normalization=config.normalization,
                 num_experts=spec.get("num_moe_experts"),
+                moe_grouped_gemm=not args.moe_sequential_experts,
                 use_flash=(args.attention == "flash")),
             vocab_size=vocab, max_sequence_length=seq,
             position_embedding_type=config.position_embedding_type,
