@@ -67,6 +67,8 @@ def parse_args():
                    help="pipeline parallel size (default min(gpus,4) for >=4 GPUs)")
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--trace", action="store_true", help="enable MegaScan tracing")
+    p.add_argument("--precision-aware-optimizer", action="store_true",
+                   help="bf16 Adam exp_avg/exp_avg_sq (fp32 math in-kernel)")
     p.add_argument("--moe-sequential-experts", action="store_true",
                    help="use SequentialMLP instead of the grouped-GEMM "
                         "experts (A/B debugging)")
@@ -204,7 +206,11 @@ def main():
     model = DistributedDataParallel(config, ddp_config, model)
     optimizer = get_megatron_optimizer(
         OptimizerConfig(lr=1e-4, weight_decay=0.1, clip_grad=1.0, bf16=bf16,
-                        use_distributed_optimizer=(dp > 1)), [model])
+                        use_distributed_optimizer=(dp > 1),
+                        use_precision_aware_optimizer=args.precision_aware_optimizer,
+                        exp_avg_dtype="bf16" if args.precision_aware_optimizer else "fp32",
+                        exp_avg_sq_dtype="bf16" if args.precision_aware_optimizer else "fp32"),
+        [model])
 
     tracer = None
     if args.trace:

@@ -20,6 +20,7 @@ giving one unified code path.
 from __future__ import annotations
 
 import math
+import warnings
 from typing import Dict, List, Optional
 
 import torch
@@ -36,7 +37,9 @@ from .optimizer_config import OptimizerConfig
 def _adam_step_flat(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
                     v: torch.Tensor, lr: float, beta1: float, beta2: float,
                     eps: float, weight_decay: float, step: int):
-    """One Adam(W) step over flat fp32 tensors; HIP kernel on GPU."""
+    """One Adam(W) step over a flat fp32 master/grad span; HIP kernel on
+    GPU.  States may be bf16 (precision-aware optimizer) — math is fp32
+    either way, bf16 states round on store."""
     if master.is_cuda and _ops.have_ops() and hasattr(_ops.get_ops(), "adamw_flat"):
         _ops.get_ops().adamw_flat(master, grad, m, v, lr, beta1, beta2, eps,
                                   weight_decay, step)
@@ -45,10 +48,16 @@ def _adam_step_flat(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
     bias_correction2 = 1 - beta2 ** step
     if weight_decay != 0:
         master.mul_(1 - lr * weight_decay)
-    m.mul_(beta1).add_(grad, alpha=1 - beta1)
-    v.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
-    denom = (v / bias_correction2).sqrt_().add_(eps)
-    master.addcdiv_(m, denom, value=-lr / bias_correction1)
+    mf = m.float() if m.dtype != torch.float32 else m
+    vf = v.float() if v.dtype != torch.float32 else v
+    mf.mul_(beta1).add_(grad, alpha=1 - beta1)
+    vf.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
+    denom = (vf / bias_correction2).sqrt_().add_(eps)
+    master.addcdiv_(mf, denom, value=-lr / bias_correction1)
+    if mf is not m:
+        m.copy_(mf)
+    if vf is not v:
+        v.copy_(vf)
 
 
 class ParamGroup:
@@ -117,8 +126,20 @@ class DistributedOptimizer:
                         master[os_ - lo:oe - lo].copy_(
                             flat[os_ - s:oe - s].float())
             self.shard_master.append(master)
-            self.shard_m.append(torch.zeros_like(master))
-            self.shard_v.append(torch.zeros_like(master))
+            st_dt = torch.float32
+            if getattr(config, "use_precision_aware_optimizer", False):
+                a = getattr(config, "exp_avg_dtype", "fp32")
+                b = getattr(config, "exp_avg_sq_dtype", "fp32")
+                if a == "bf16" and b == "bf16":
+                    st_dt = torch.bfloat16
+                elif "bf16" in (a, b):
+                    # the fused flat kernel keeps one dtype for both
+                    # states; mixed requests stay fp32 (the safe side)
+                    warnings.warn("precision-aware optimizer: exp_avg and "
+                                  "exp_avg_sq dtypes differ; keeping both "
+                                  "fp32")
+            self.shard_m.append(torch.zeros_like(master, dtype=st_dt))
+            self.shard_v.append(torch.zeros_like(master, dtype=st_dt))
 
             no_wd, norm_r = [], []
             for p, (s, e) in buf.param_index_map.items():

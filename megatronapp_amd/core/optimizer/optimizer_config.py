@@ -29,3 +29,9 @@ class OptimizerConfig:
     clip_grad: float = 1.0
     log_num_zeros_in_grad: bool = False
     barrier_with_L1_time: bool = False
+    # precision-aware optimizer (reference --use-precision-aware-optimizer):
+    # store Adam exp_avg / exp_avg_sq in bf16 (fp32 math in-kernel) —
+    # halves optimizer-state memory and the optimizer HBM stream
+    use_precision_aware_optimizer: bool = False
+    exp_avg_dtype: str = "fp32"
+    exp_avg_sq_dtype: str = "fp32"

@@ -1,16 +1,75 @@
-// Fused AdamW over flat fp32 buffers — the "multi-tensor Adam" of the
+// Fused AdamW over flat buffers — the "multi-tensor Adam" of the
 // reference (apex multi_tensor_apply) collapsed to ONE
 // contiguous span per DDP buffer by the buffer-aligned optimizer design
-// (core/optimizer/distrib_optimizer.py).  Pure HBM-bound float4 stream.
+// (core/optimizer/distrib_optimizer.py).  Pure HBM-bound stream.
+//
+// Optimizer states (exp_avg / exp_avg_sq) are fp32 by default; the
+// precision-aware mode (reference --use-precision-aware-optimizer)
+// stores them as bf16 — math stays fp32 in registers, states are
+// round-to-nearest-even on store, and the 24 B/element stream drops to
+// 16 B/element (plus half the state memory).
 
 #include "common.h"
 
 #define BLOCK 256
 
+typedef __attribute__((ext_vector_type(4))) short short4v_a;
+
+// 4-wide state load/store, fp32 or bf16 storage
+struct St4F32 {
+  using T = float;
+  static __device__ __forceinline__ void load(const float* s, long i,
+                                              float o[4]) {
+    float4v t = *(const float4v*)(s + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = t[j];
+  }
+  static __device__ __forceinline__ void store(float* s, long i,
+                                               const float in[4]) {
+    float4v t;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) t[j] = in[j];
+    *(float4v*)(s + i) = t;
+  }
+  static __device__ __forceinline__ float ld1(const float* s, long i) {
+    return s[i];
+  }
+  static __device__ __forceinline__ void st1(float* s, long i, float v) {
+    s[i] = v;
+  }
+};
+
+struct St4BF16 {
+  using T = unsigned short;
+  static __device__ __forceinline__ void load(const unsigned short* s, long i,
+                                              float o[4]) {
+    short4v_a t = *(const short4v_a*)(s + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = bf2f((unsigned short)t[j]);
+  }
+  static __device__ __forceinline__ void store(unsigned short* s, long i,
+                                               const float in[4]) {
+    short4v_a t;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) t[j] = (short)f2bf(in[j]);
+    *(short4v_a*)(s + i) = t;
+  }
+  static __device__ __forceinline__ float ld1(const unsigned short* s,
+                                              long i) {
+    return bf2f(s[i]);
+  }
+  static __device__ __forceinline__ void st1(unsigned short* s, long i,
+                                             float v) {
+    s[i] = f2bf(v);
+  }
+};
+
+template <typename ST>
 __global__ void adamw_flat_kernel(float* __restrict__ p,
                                   const float* __restrict__ g,
-                                  float* __restrict__ m, float* __restrict__ v,
-                                  long n, float lr, float beta1, float beta2,
+                                  typename ST::T* __restrict__ m,
+                                  typename ST::T* __restrict__ v, long n,
+                                  float lr, float beta1, float beta2,
                                   float eps, float wd, float bc1, float bc2) {
   long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * 4;
   const long stride = (long)gridDim.x * BLOCK * 4;
@@ -19,8 +78,9 @@ __global__ void adamw_flat_kernel(float* __restrict__ p,
   for (; i + 3 < n; i += stride) {
     float4v pv = *(float4v*)(p + i);
     float4v gv = *(const float4v*)(g + i);
-    float4v mv = *(float4v*)(m + i);
-    float4v vv = *(float4v*)(v + i);
+    float mv[4], vv[4];
+    ST::load(m, i, mv);
+    ST::load(v, i, vv);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       float pj = pv[j] * decay;
@@ -32,8 +92,8 @@ __global__ void adamw_flat_kernel(float* __restrict__ p,
       vv[j] = vj;
     }
     *(float4v*)(p + i) = pv;
-    *(float4v*)(m + i) = mv;
-    *(float4v*)(v + i) = vv;
+    ST::store(m, i, mv);
+    ST::store(v, i, vv);
   }
   // tail
   if (blockIdx.x == 0 && threadIdx.x < 4) {
@@ -41,11 +101,11 @@ __global__ void adamw_flat_kernel(float* __restrict__ p,
     long k = start + threadIdx.x;
     if (k < n) {
       float pj = p[k] * decay;
-      float mj = beta1 * m[k] + (1.f - beta1) * g[k];
-      float vj = beta2 * v[k] + (1.f - beta2) * g[k] * g[k];
+      float mj = beta1 * ST::ld1(m, k) + (1.f - beta1) * g[k];
+      float vj = beta2 * ST::ld1(v, k) + (1.f - beta2) * g[k] * g[k];
       p[k] = pj - step_size * mj / (sqrtf(vj / bc2) + eps);
-      m[k] = mj;
-      v[k] = vj;
+      ST::st1(m, k, mj);
+      ST::st1(v, k, vj);
     }
   }
 }
@@ -53,12 +113,14 @@ __global__ void adamw_flat_kernel(float* __restrict__ p,
 // Range-table variant: identical update but weight decay is zeroed inside
 // the no-wd ranges (biases/norms).  One launch per buffer instead of one
 // per param slice; ~600 ranges resolved by a per-thread binary search on
-// cached range tables (negligible next to the 24 B/element HBM stream).
+// cached range tables (negligible next to the HBM stream).
+template <typename ST>
 __global__ void adamw_flat_ranged_kernel(
-    float* __restrict__ p, const float* __restrict__ g, float* __restrict__ m,
-    float* __restrict__ v, const long* __restrict__ nw_s,
-    const long* __restrict__ nw_e, int n_ranges, long n, float lr, float beta1,
-    float beta2, float eps, float wd, float bc1, float bc2) {
+    float* __restrict__ p, const float* __restrict__ g,
+    typename ST::T* __restrict__ m, typename ST::T* __restrict__ v,
+    const long* __restrict__ nw_s, const long* __restrict__ nw_e,
+    int n_ranges, long n, float lr, float beta1, float beta2, float eps,
+    float wd, float bc1, float bc2) {
   long i = ((long)blockIdx.x * BLOCK + threadIdx.x) * 4;
   const long stride = (long)gridDim.x * BLOCK * 4;
   const float step_size = lr / bc1;
@@ -74,8 +136,9 @@ __global__ void adamw_flat_ranged_kernel(
     const float decay = no_wd ? 1.f : 1.f - lr * wd;
     float4v pv = *(float4v*)(p + i);
     float4v gv = *(const float4v*)(g + i);
-    float4v mv = *(float4v*)(m + i);
-    float4v vv = *(float4v*)(v + i);
+    float mv[4], vv[4];
+    ST::load(m, i, mv);
+    ST::load(v, i, vv);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       float pj = pv[j] * decay;
@@ -87,8 +150,8 @@ __global__ void adamw_flat_ranged_kernel(
       vv[j] = vj;
     }
     *(float4v*)(p + i) = pv;
-    *(float4v*)(m + i) = mv;
-    *(float4v*)(v + i) = vv;
+    ST::store(m, i, mv);
+    ST::store(v, i, vv);
   }
   if (blockIdx.x == 0 && threadIdx.x < 4) {
     long start = (n / 4) * 4;
@@ -102,37 +165,55 @@ __global__ void adamw_flat_ranged_kernel(
       const bool no_wd = lo > 0 && k < nw_e[lo - 1];
       const float decay = no_wd ? 1.f : 1.f - lr * wd;
       float pj = p[k] * decay;
-      float mj = beta1 * m[k] + (1.f - beta1) * g[k];
-      float vj = beta2 * v[k] + (1.f - beta2) * g[k] * g[k];
+      float mj = beta1 * ST::ld1(m, k) + (1.f - beta1) * g[k];
+      float vj = beta2 * ST::ld1(v, k) + (1.f - beta2) * g[k] * g[k];
       p[k] = pj - step_size * mj / (sqrtf(vj / bc2) + eps);
-      m[k] = mj;
-      v[k] = vj;
+      ST::st1(m, k, mj);
+      ST::st1(v, k, vj);
     }
   }
 }
 
-void launch_adamw_flat_ranged(float* p, const float* g, float* m, float* v,
-                              const long* nw_s, const long* nw_e, int n_ranges,
-                              long n, float lr, float beta1, float beta2,
-                              float eps, float wd, int step, hipStream_t s) {
+static int adam_grid(long n) {
+  long blocks = (n / 4 + BLOCK - 1) / BLOCK;
+  return (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
+}
+
+void launch_adamw_flat_ranged(float* p, const float* g, void* m, void* v,
+                              bool states_bf16, const long* nw_s,
+                              const long* nw_e, int n_ranges, long n,
+                              float lr, float beta1, float beta2, float eps,
+                              float wd, int step, hipStream_t s) {
   float bc1 = 1.f - powf(beta1, (float)step);
   float bc2 = 1.f - powf(beta2, (float)step);
-  long blocks = (n / 4 + BLOCK - 1) / BLOCK;
-  int grid = (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
-  hipLaunchKernelGGL(adamw_flat_ranged_kernel, dim3(grid), dim3(BLOCK), 0, s,
-                     p, g, m, v, nw_s, nw_e, n_ranges, n, lr, beta1, beta2,
-                     eps, wd, bc1, bc2);
+  int grid = adam_grid(n);
+  if (states_bf16)
+    hipLaunchKernelGGL(adamw_flat_ranged_kernel<St4BF16>, dim3(grid),
+                       dim3(BLOCK), 0, s, p, g, (unsigned short*)m,
+                       (unsigned short*)v, nw_s, nw_e, n_ranges, n, lr, beta1,
+                       beta2, eps, wd, bc1, bc2);
+  else
+    hipLaunchKernelGGL(adamw_flat_ranged_kernel<St4F32>, dim3(grid),
+                       dim3(BLOCK), 0, s, p, g, (float*)m, (float*)v, nw_s,
+                       nw_e, n_ranges, n, lr, beta1, beta2, eps, wd, bc1,
+                       bc2);
   HIP_CHECK_LAUNCH();
 }
 
-void launch_adamw_flat(float* p, const float* g, float* m, float* v, long n,
-                       float lr, float beta1, float beta2, float eps, float wd,
-                       int step, hipStream_t s) {
+void launch_adamw_flat(float* p, const float* g, void* m, void* v,
+                       bool states_bf16, long n, float lr, float beta1,
+                       float beta2, float eps, float wd, int step,
+                       hipStream_t s) {
   float bc1 = 1.f - powf(beta1, (float)step);
   float bc2 = 1.f - powf(beta2, (float)step);
-  long blocks = (n / 4 + BLOCK - 1) / BLOCK;
-  int grid = (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
-  hipLaunchKernelGGL(adamw_flat_kernel, dim3(grid), dim3(BLOCK), 0, s, p, g, m,
-                     v, n, lr, beta1, beta2, eps, wd, bc1, bc2);
+  int grid = adam_grid(n);
+  if (states_bf16)
+    hipLaunchKernelGGL(adamw_flat_kernel<St4BF16>, dim3(grid), dim3(BLOCK),
+                       0, s, p, g, (unsigned short*)m, (unsigned short*)v, n,
+                       lr, beta1, beta2, eps, wd, bc1, bc2);
+  else
+    hipLaunchKernelGGL(adamw_flat_kernel<St4F32>, dim3(grid), dim3(BLOCK), 0,
+                       s, p, g, (float*)m, (float*)v, n, lr, beta1, beta2,
+                       eps, wd, bc1, bc2);
   HIP_CHECK_LAUNCH();
 }

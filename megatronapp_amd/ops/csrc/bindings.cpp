@@ -41,9 +41,10 @@ void launch_softmax_causal_bwd(const void*, const void*, void*, long, int,
                                int, float, hipStream_t);
 void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
                         hipStream_t);
-void launch_adamw_flat_ranged(float*, const float*, float*, float*,
-                              const long*, const long*, int, long, float,
-                              float, float, float, float, int, hipStream_t);
+void launch_adamw_flat_ranged(float*, const float*, void*, void*,
+                              bool, const long*, const long*, int,
+                              long, float, float, float, float,
+                              float, int, hipStream_t);
 void launch_ce_rowmax(const void*, float*, long, int, hipStream_t);
 void launch_ce_fwd(const void*, const float*, const int*, float*, float*,
                    long, int, hipStream_t);
@@ -52,8 +53,9 @@ void launch_ce_bwd(const void*, const float*, const float*, const int*,
 void launch_selective_scan_fwd(const void*, const void*, const float*,
                                const void*, const void*, const float*, float*,
                                void*, int, int, int, int, hipStream_t);
-void launch_adamw_flat(float*, const float*, float*, float*, long, float,
-                       float, float, float, float, int, hipStream_t);
+void launch_adamw_flat(float*, const float*, void*, void*, bool,
+                       long, float, float, float, float, float,
+                       int, hipStream_t);
 void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
 bool wgrad_accum_bgrad(torch::Tensor, torch::Tensor, torch::Tensor,
                        torch::Tensor);
@@ -650,12 +652,22 @@ std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
 }
 
 // --------------------------------------------------------------------- adam
+// states fp32 OR bf16 (precision-aware optimizer); params/grads fp32
+static bool adam_states_bf16(const torch::Tensor& m, const torch::Tensor& v) {
+  TORCH_CHECK(m.scalar_type() == v.scalar_type(),
+              "adam m/v dtypes must match");
+  TORCH_CHECK(m.scalar_type() == torch::kFloat32 ||
+              m.scalar_type() == torch::kBFloat16,
+              "adam states must be fp32 or bf16");
+  return m.scalar_type() == torch::kBFloat16;
+}
+
 void adamw_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double beta1, double beta2,
                 double eps, double wd, long step) {
   TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
-  launch_adamw_flat(p.data_ptr<float>(), g.data_ptr<float>(),
-                    m.data_ptr<float>(), v.data_ptr<float>(), p.numel(),
+  launch_adamw_flat(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr(),
+                    v.data_ptr(), adam_states_bf16(m, v), p.numel(),
                     (float)lr, (float)beta1, (float)beta2, (float)eps,
                     (float)wd, (int)step, cur_stream());
 }
@@ -734,8 +746,8 @@ void adamw_flat_ranged(torch::Tensor p, torch::Tensor g, torch::Tensor m,
               nw_s.is_contiguous());
   TORCH_CHECK(nw_e.sizes() == nw_s.sizes());
   launch_adamw_flat_ranged(
-      p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
-      v.data_ptr<float>(), nw_s.data_ptr<long>(), nw_e.data_ptr<long>(),
+      p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr(), v.data_ptr(),
+      adam_states_bf16(m, v), nw_s.data_ptr<long>(), nw_e.data_ptr<long>(),
       (int)nw_s.numel(), p.numel(), (float)lr, (float)beta1, (float)beta2,
       (float)eps, (float)wd, (int)step,
       at::cuda::getCurrentCUDAStream());

@@ -156,6 +156,15 @@ def _add_training_args(p):
     g.add_argument("--overlap-param-gather", action="store_true",
                    help="overlap the ZeRO-1 param all-gather into the next step")
     g.add_argument("--use-distributed-optimizer", action="store_true")
+    g.add_argument("--use-precision-aware-optimizer", action="store_true",
+                   help="store Adam exp_avg/exp_avg_sq in the dtype given by "
+                        "--exp-avg-dtype/--exp-avg-sq-dtype (bf16 halves "
+                        "optimizer state memory and the optimizer HBM "
+                        "stream; update math stays fp32 in-kernel)")
+    g.add_argument("--exp-avg-dtype", default="fp32",
+                   choices=["fp32", "bf16"])
+    g.add_argument("--exp-avg-sq-dtype", default="fp32",
+                   choices=["fp32", "bf16"])
     g.add_argument("--ddp-bucket-size", type=int, default=None)
     g.add_argument("--non-persistent-save-interval", type=int, default=None,
                    help="iterations between LOCAL (node-scratch) "

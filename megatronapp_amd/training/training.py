@@ -130,7 +130,11 @@ def setup_model_and_optimizer(model_provider_func, model_type,
         hysteresis=args.hysteresis,
         use_distributed_optimizer=args.use_distributed_optimizer,
         overlap_param_gather=args.overlap_param_gather,
-        log_num_zeros_in_grad=args.log_num_zeros_in_grad)
+        log_num_zeros_in_grad=args.log_num_zeros_in_grad,
+        use_precision_aware_optimizer=getattr(
+            args, "use_precision_aware_optimizer", False),
+        exp_avg_dtype=getattr(args, "exp_avg_dtype", "fp32"),
+        exp_avg_sq_dtype=getattr(args, "exp_avg_sq_dtype", "fp32"))
     optimizer = get_megatron_optimizer(opt_config, model)
     if args.fp16 and hasattr(optimizer, "scale_loss"):
         # fp16: schedules scale the loss before backward; the optimizer

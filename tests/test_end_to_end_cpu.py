@@ -209,3 +209,55 @@ def test_dynamic_engine_matches_static_greedy():
         dyn.step()
     got = [dyn.finished[i].generated_text for i in ids]
     assert got == want, list(zip(got, want))
+
+
+def test_precision_aware_optimizer_learns():
+    """bf16 Adam states (reference --use-precision-aware-optimizer): the
+    eager fallback stores exp_avg/exp_avg_sq in bf16 and the loss curve
+    stays close to the fp32-state run."""
+    from tests.utils import initialize_model_parallel, destroy
+    initialize_model_parallel()
+
+    def run(precision_aware):
+        torch.manual_seed(21)
+        config = TransformerConfig(
+            num_layers=2, hidden_size=64, num_attention_heads=4,
+            ffn_hidden_size=128, pipeline_dtype=torch.float32,
+            hidden_dropout=0.0, attention_dropout=0.0,
+            finalize_model_grads_func=finalize_model_grads)
+        model = GPTModel(
+            config=config,
+            transformer_layer_spec=get_gpt_layer_local_spec(use_flash=False),
+            vocab_size=VOCAB, max_sequence_length=SEQ,
+            share_embeddings_and_output_weights=True, parallel_output=True)
+        model = DistributedDataParallel(
+            config, DistributedDataParallelConfig(), model)
+        opt = get_megatron_optimizer(OptimizerConfig(
+            lr=1e-3, weight_decay=0.01, clip_grad=1.0,
+            use_precision_aware_optimizer=precision_aware,
+            exp_avg_dtype="bf16" if precision_aware else "fp32",
+            exp_avg_sq_dtype="bf16" if precision_aware else "fp32"), [model])
+        torch.manual_seed(5)
+        tok = torch.randint(0, VOCAB, (4, SEQ + 1))
+        inp, lbl = tok[:, :-1], tok[:, 1:]
+        pos = torch.arange(SEQ).unsqueeze(0).expand(4, -1)
+        losses = []
+        for _ in range(8):
+            model.zero_grad_buffer()
+            loss = model(input_ids=inp, position_ids=pos,
+                         attention_mask=None, labels=lbl).float().mean()
+            loss.backward()
+            model.finish_grad_sync()
+            opt.step()
+            losses.append(float(loss.detach()))
+        return losses, opt
+
+    base, _ = run(False)
+    pa, opt = run(True)
+    assert opt.shard_m[0].dtype == torch.bfloat16
+    assert opt.shard_v[0].dtype == torch.bfloat16
+    assert pa[-1] < pa[0] - 0.05, pa
+    # curves track within a few percent of the loss scale
+    for a, b in zip(base, pa):
+        assert abs(a - b) < 0.05 * max(abs(a), 1.0), (base, pa)
+    destroy()

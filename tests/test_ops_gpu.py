@@ -925,3 +925,33 @@ def test_grouped_mm_fused_main_grad_wgrad():
     y2 = _GroupedMMFn.apply(x2, w.detach().clone().requires_grad_(True), offs)
     y2.backward(dy)
     assert torch.equal(x.grad, x2.grad)
+
+
+@pytest.mark.gpu
+def test_adamw_flat_bf16_states_matches_fp32_math():
+    """Precision-aware optimizer: bf16 exp_avg/exp_avg_sq storage with
+    fp32 in-kernel math tracks the fp32-state reference to bf16 rounding
+    over several steps."""
+    torch.manual_seed(8)
+    n = 10007
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda")
+    m = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
+    v = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
+    p_ref = p.clone()
+    mt = torch.zeros(n, device="cuda")
+    vt = torch.zeros(n, device="cuda")
+    lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-8, 0.01
+    for step in range(1, 5):
+        _ops().adamw_flat(p, g, m, v, lr, b1, b2, eps, wd, step)
+        # reference with bf16 state rounding applied the same way
+        p_ref.mul_(1 - lr * wd)
+        mf = mt.bfloat16().float().mul_(b1).add_(g, alpha=1 - b1)
+        vf = vt.bfloat16().float().mul_(b2).addcmul_(g, g, value=1 - b2)
+        bc1, bc2 = 1 - b1 ** step, 1 - b2 ** step
+        p_ref.addcdiv_(mf, (vf / bc2).sqrt().add(eps), value=-lr / bc1)
+        mt, vt = mf, vf
+    assert torch.allclose(m.float(), mt.bfloat16().float()), "m mismatch"
+    assert torch.allclose(v.float(), vt.bfloat16().float()), "v mismatch"
+    err = (p - p_ref).abs().max().item()
+    assert err < 1e-5, err
