@@ -53,8 +53,14 @@ class LanguageModelEmbedding(MegatronModule):
             embeddings = word_embeddings.transpose(0, 1).contiguous()  # [s, b, h]
         if self.add_position_embedding:
             # out-of-range lookups are an async DEVICE FAULT (core dump
-            # with no traceback) — fail with a real error instead
-            if position_ids.numel() and int(position_ids.max()) >= \
+            # with no traceback) — fail with a real error instead.  The
+            # check is a device->host sync, which is forbidden while a
+            # hipGraph is being captured (decode capture passes fixed
+            # in-range position buffers, so skipping it there is safe).
+            capturing = (position_ids.is_cuda
+                         and torch.cuda.is_current_stream_capturing())
+            if not capturing and position_ids.numel() and \
+                    int(position_ids.max()) >= \
                     self.position_embeddings.num_embeddings:
                 raise ValueError(
                     f"position id {int(position_ids.max())} exceeds "
