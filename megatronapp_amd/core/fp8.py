@@ -8,10 +8,11 @@ GEMM expresses):
   dgrad   : dx = dy8 @ w8c     dy per-token rows, W per-INPUT-channel
                                scales (a second cached quantization of W
                                along dim 1, column-major for _scaled_mm)
-  wgrad   : bf16               unchanged — accumulates exactly into the
-                               fp32 main_grad buffer (the fused
-                               wgrad_accum path), so optimizer numerics
-                               are identical to bf16 training.
+  wgrad   : fp8 e4m3           both operands transpose-quantized with
+                               per-tensor scales, fp32 GEMM output added
+                               into the fp32 main_grad (fp8_wgrad;
+                               MEGATRONAPP_DISABLE_FUSED=fp8_wgrad
+                               falls back to the exact bf16 wgrad).
 
 Weight quantizations are cached per optimizer step (`bump_step()` is
 called from DistributedOptimizer.step); activations/grads quantize
@@ -87,6 +88,35 @@ def fp8_dgrad(dy: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     dx = torch._scaled_mm(dy8, c["w8c"], scale_a=ds, scale_b=c["wcs"],
                           out_dtype=dy.dtype)
     return dx.reshape(*shp[:-1], weight.shape[1])
+
+
+def fp8_wgrad_enabled() -> bool:
+    import os
+    return "fp8_wgrad" not in os.environ.get("MEGATRONAPP_DISABLE_FUSED", "")
+
+
+def fp8_wgrad(g2d: torch.Tensor, x2d: torch.Tensor,
+              main_grad: torch.Tensor) -> bool:
+    """main_grad[out, in] += g2d[rows, out]^T @ x2d[rows, in] through an
+    fp8 e4m3 GEMM: both operands transpose-quantized (token dim
+    innermost, per-tensor just-in-time scales), fp32 _scaled_mm output
+    added into main_grad.  The per-tensor quantization of the GRADIENT
+    operand is the accuracy risk — guarded by the fp8 convergence test
+    (tests/test_families_gpu.py --fp8 twin).  Returns False when the
+    path does not apply (caller falls back to the exact bf16 wgrad)."""
+    from .. import ops as _ops
+    if not (g2d.is_cuda and g2d.dtype == torch.bfloat16
+            and main_grad.dtype == torch.float32 and _ops.have_ops()
+            and hasattr(_ops.get_ops(), "quantize_transpose_e4m3")
+            and fp8_wgrad_enabled()):
+        return False
+    lt = _ops.get_ops()
+    qg, sg = lt.quantize_transpose_e4m3(g2d)   # [out, rows]
+    qx, sx = lt.quantize_transpose_e4m3(x2d)   # [in, rows]
+    out = torch._scaled_mm(qg, qx.t(), scale_a=sg, scale_b=sx,
+                           out_dtype=torch.float32)
+    main_grad.add_(out)
+    return True
 
 
 def enable_fp8_training(model: torch.nn.Module) -> int:

@@ -245,9 +245,16 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
                 go2 = grad_output_2d.contiguous()
                 ti2 = total_input_2d.contiguous()
                 lt = _ops.get_ops()
+                wgrad_done = False
+                if getattr(ctx, "fp8", False):
+                    # fp8 wgrad (transpose-quantized e4m3 GEMM, fp32 out
+                    # accumulated); bias grad stays on the colsum path
+                    from ..fp8 import fp8_wgrad
+                    wgrad_done = fp8_wgrad(go2, ti2, weight.main_grad)
                 bias_param = ctx.bias_param if use_bias else None
                 want_bgrad = (
-                    bias_param is not None
+                    not wgrad_done
+                    and bias_param is not None
                     and hasattr(bias_param, "main_grad")
                     and hasattr(bias_param, "grad_added_to_main_grad")
                     and "bgrad" not in os.environ.get(
@@ -259,7 +266,7 @@ class LinearWithGradAccumulationAndAsyncCommunication(torch.autograd.Function):
                         fused_dbias = dbias_tmp
                     else:          # no epilogue algo for this shape
                         lt.wgrad_accum(go2, ti2, weight.main_grad)
-                else:
+                elif not wgrad_done:
                     lt.wgrad_accum(go2, ti2, weight.main_grad)
             else:
                 weight.main_grad.add_(

@@ -66,6 +66,8 @@ void launch_colsum_accum(const void*, float*, float*, int, long, int,
                          hipStream_t);
 void launch_embedding_bwd_accum(const void*, const int*, float*, long, int,
                                 hipStream_t);
+void launch_transpose_quant_e4m3(const void*, void*, unsigned int*,
+                                 float*, long, long, hipStream_t);
 void launch_quant_rows_e4m3(const void*, void*, float*, long, int,
                             hipStream_t);
 void launch_bias_dropout_add_fwd(const void*, const void*, const void*,
@@ -502,6 +504,21 @@ std::vector<torch::Tensor> moe_combine_bwd(torch::Tensor dout,
 }
 
 // ---------------------------------------------------------------- fp8 quant
+// [R, C] bf16 -> ([C, R] e4m3, [1] f32 per-tensor scale): wgrad operands
+// (fp8 MFMA wants the token dim innermost on both sides)
+std::vector<torch::Tensor> quantize_transpose_e4m3(torch::Tensor x) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 2, "expects 2D");
+  const long R = x.size(0), C = x.size(1);
+  auto q = torch::empty({C, R}, x.options().dtype(torch::kFloat8_e4m3fn));
+  auto s = torch::empty({1}, x.options().dtype(torch::kFloat32));
+  auto amax = torch::zeros({1}, x.options().dtype(torch::kInt32));
+  launch_transpose_quant_e4m3(x.data_ptr(), q.data_ptr(),
+                              (unsigned int*)amax.data_ptr(),
+                              s.data_ptr<float>(), R, C, cur_stream());
+  return {q, s};
+}
+
 std::vector<torch::Tensor> quantize_rows_e4m3(torch::Tensor x) {
   check_bf16(x, "x");
   auto x2 = x.reshape({-1, x.size(-1)}).contiguous();
@@ -786,6 +803,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd_t", &attn_fwd_t);
   mod.def("attn_fwd2", &attn_fwd2);
   mod.def("quantize_rows_e4m3", &quantize_rows_e4m3);
+  mod.def("quantize_transpose_e4m3", &quantize_transpose_e4m3);
   mod.def("bias_dropout_add_fwd", &bias_dropout_add_fwd);
   mod.def("dropout_bwd", &dropout_bwd);
   mod.def("moe_combine_fwd", &moe_combine_fwd);

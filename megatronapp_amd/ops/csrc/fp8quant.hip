@@ -73,3 +73,116 @@ void launch_quant_rows_e4m3(const void* x, void* q, float* scale, long M,
                      K);
   HIP_CHECK_LAUNCH();
 }
+
+// ---------------------------------------------------------------------------
+// Transpose-quantize for the fp8 wgrad GEMM: both wgrad operands need
+// their token (K) dimension innermost for the MFMA fp8 path, so the
+// quantization pass does the transpose through an LDS tile for free.
+//
+//   x [R, C] bf16  ->  q [C, R] e4m3,  scale [1] f32 = amax/448
+//
+// (per-tensor just-in-time scaling: the amax is of THIS tensor, not a
+// delayed-scaling history)
+
+#define TQ_TILE 64
+
+__global__ void amax_abs_kernel(const unsigned short* __restrict__ x, long n,
+                                unsigned int* __restrict__ amax_bits) {
+  __shared__ float lds[FP8Q_BLOCK / WAVE];
+  float amax = 0.f;
+  long i = ((long)blockIdx.x * FP8Q_BLOCK + threadIdx.x) * 8;
+  const long stride = (long)gridDim.x * FP8Q_BLOCK * 8;
+  for (; i + 7 < n; i += stride) {
+    bf16x8 v = *(const bf16x8*)(x + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      amax = fmaxf(amax, fabsf(bf2f((unsigned short)v[j])));
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (long k = (n / 8) * 8; k < n; ++k)
+      amax = fmaxf(amax, fabsf(bf2f(x[k])));
+  amax = block_reduce_max<FP8Q_BLOCK>(amax, lds);
+  // positive-float bit patterns order like uints
+  if (threadIdx.x == 0) atomicMax(amax_bits, __float_as_uint(amax));
+}
+
+__global__ void transpose_quant_e4m3_kernel(
+    const unsigned short* __restrict__ x, unsigned char* __restrict__ q,
+    const unsigned int* __restrict__ amax_bits, float* __restrict__ scale,
+    long R, long C) {
+  __shared__ unsigned short lds[TQ_TILE][TQ_TILE + 8];
+  const float amax = __uint_as_float(*amax_bits);
+  const float s = fmaxf(amax / 448.0f, 1e-12f);
+  const float inv = 1.0f / s;
+  if (blockIdx.x == 0 && blockIdx.y == 0 && threadIdx.x == 0) scale[0] = s;
+
+  const long r0 = (long)blockIdx.y * TQ_TILE;
+  const long c0 = (long)blockIdx.x * TQ_TILE;
+  // load 64x64 bf16: 8 lanes x 8 elements per row, 32 rows per pass
+  const int lx = threadIdx.x % 8;        // 8-wide column group
+  const int ly = threadIdx.x / 8;        // 32 rows
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const long r = r0 + ly + p * 32;
+    const long c = c0 + lx * 8;
+    if (r < R) {
+      if (c + 7 < C) {
+        bf16x8 v = *(const bf16x8*)(x + r * C + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds[ly + p * 32][lx * 8 + j] = (unsigned short)v[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          lds[ly + p * 32][lx * 8 + j] =
+              (c + j < C) ? x[r * C + c + j] : 0;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds[ly + p * 32][lx * 8 + j] = 0;
+    }
+  }
+  __syncthreads();
+  // store 64x64 e4m3: 16 lanes x uchar4 per output row, 16 rows per pass
+  const int ox = threadIdx.x % 16;       // 16-wide row-offset group
+  const int oy = threadIdx.x / 16;       // 16 output rows
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const long oc = c0 + oy + p * 16;    // output row == input column
+    const long orr = r0 + ox * 4;        // output col == input row
+    if (oc >= C) continue;
+    unsigned char packed[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float v = bf2f(lds[ox * 4 + j][oy + p * 16]) * inv;
+      __hip_fp8_e4m3 qv(v);
+      packed[j] = qv.__x;
+    }
+    if (orr + 3 < R) {
+      *(uchar4*)(q + oc * R + orr) = make_uchar4(packed[0], packed[1],
+                                                packed[2], packed[3]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        if (orr + j < R) q[oc * R + orr + j] = packed[j];
+    }
+  }
+}
+
+void launch_transpose_quant_e4m3(const void* x, void* q,
+                                 unsigned int* amax_bits, float* scale,
+                                 long R, long C, hipStream_t stream) {
+  long blocks = (R * C / 8 + FP8Q_BLOCK - 1) / FP8Q_BLOCK;
+  if (blocks > 4096) blocks = 4096;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(amax_abs_kernel, dim3((unsigned)blocks),
+                     dim3(FP8Q_BLOCK), 0, stream,
+                     (const unsigned short*)x, R * C, amax_bits);
+  HIP_CHECK_LAUNCH();
+  dim3 grid((unsigned)((C + TQ_TILE - 1) / TQ_TILE),
+            (unsigned)((R + TQ_TILE - 1) / TQ_TILE));
+  hipLaunchKernelGGL(transpose_quant_e4m3_kernel, grid, dim3(FP8Q_BLOCK), 0,
+                     stream, (const unsigned short*)x, (unsigned char*)q,
+                     amax_bits, scale, R, C);
+  HIP_CHECK_LAUNCH();
+}

@@ -960,3 +960,40 @@ def test_adamw_flat_bf16_states_matches_fp32_math():
                           rtol=2e-2, atol=1e-10), "v mismatch"
     err = (p - p_ref).abs().max().item()
     assert err < 1e-4, err
+
+
+@pytest.mark.gpu
+def test_quantize_transpose_e4m3():
+    """[R, C] bf16 -> [C, R] e4m3 with a per-tensor scale: dequantized
+    transpose must reconstruct the input to e4m3 precision, including
+    non-multiple-of-64 edges."""
+    ops = _ops()
+    torch.manual_seed(11)
+    for R, C in [(256, 192), (1000, 120), (130, 70)]:
+        x = torch.randn(R, C, device="cuda", dtype=torch.bfloat16) * 3
+        q, s = ops.quantize_transpose_e4m3(x)
+        assert q.shape == (C, R) and q.dtype == torch.float8_e4m3fn
+        rec = q.float().t() * s
+        err = (rec - x.float()).abs().max().item()
+        amax = x.float().abs().max().item()
+        # e4m3 mantissa: 3 bits -> rel step 1/16 of the top bin
+        assert err <= amax / 448 * 32, (R, C, err, amax)
+
+
+@pytest.mark.gpu
+def test_fp8_wgrad_accumulates_close_to_bf16():
+    """fp8_wgrad (transpose-quantized e4m3 GEMM, fp32 accumulate) tracks
+    the exact wgrad within fp8 tolerance."""
+    from megatronapp_amd.core.fp8 import fp8_wgrad
+    torch.manual_seed(12)
+    rows, out, in_ = 4096, 1536, 512
+    g = torch.randn(rows, out, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(rows, in_, device="cuda", dtype=torch.bfloat16)
+    mg = torch.randn(out, in_, device="cuda", dtype=torch.float32)
+    base = mg.clone()
+    assert fp8_wgrad(g, x, mg)
+    ref = base + g.float().t() @ x.float()
+    delta = ref - base          # the wgrad contribution itself
+    err = (mg - ref).abs().max().item()
+    rel = err / (delta.abs().max().item() + 1e-6)
+    assert rel < 8e-2, rel
