@@ -79,7 +79,8 @@ def parse_args():
     p.add_argument("--trace-dir", default="trace_out")
     p.add_argument("--attention", default="flash", choices=["flash", "fused"])
     p.add_argument("--fp8", action="store_true",
-                   help="fp8 (e4m3) forward+dgrad GEMMs, bf16 wgrad")
+                   help="fp8 (e4m3) forward+dgrad+wgrad GEMMs (wgrad "
+                        "accumulates fp32; NOT the headline dtype)")
     p.add_argument("--seq-length", type=int, default=None,
                    help="override the model's sequence length")
     p.add_argument("--no-overlap-grad-reduce", action="store_true")
@@ -314,7 +315,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(tokens_per_s / BASELINE_TOKENS_PER_S, 3),
-            "dtype": "bf16" if bf16 else "fp32",
+            "dtype": ("fp8" if args.fp8 and on_gpu else
+                      "bf16" if bf16 else "fp32"),
             "data": "synthetic",
             "config": {
                 "model": args.model,
