@@ -152,7 +152,17 @@ class DistributedOptimizer:
                 if param_is_not_tensor_parallel_duplicate(p):
                     norm_r.append(r)
             self.no_wd_ranges.append(no_wd)
-            self.norm_ranges.append(norm_r)
+            # coalesce adjacent ranges: at TP=1 every param counts, so
+            # the whole shard collapses to ONE span and the norm/clip
+            # pass is a single flat kernel instead of a foreach over
+            # hundreds of per-param slices (1.7% of a MoE step)
+            merged = []
+            for r in sorted(norm_r):
+                if merged and r[0] <= merged[-1][1]:
+                    merged[-1] = (merged[-1][0], max(merged[-1][1], r[1]))
+                else:
+                    merged.append(r)
+            self.norm_ranges.append(merged)
 
         # interface compat: param_groups for LR schedulers
         self.param_groups = [
