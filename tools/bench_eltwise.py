@@ -84,5 +84,24 @@ def main():
         print(f"  separate {us_sep:7.1f} us   (fused unavailable)")
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and "--adam" not in __import__("sys").argv:
     main()
+
+
+def bench_adam():
+    assert ops.have_ops()
+    o = ops.get_ops()
+    n = 500_000_000
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda")
+    for st_dt, label in ((torch.float32, "fp32"), (torch.bfloat16, "bf16")):
+        m = torch.zeros(n, device="cuda", dtype=st_dt)
+        v = torch.zeros(n, device="cuda", dtype=st_dt)
+        us = timeit(lambda: o.adamw_flat(p, g, m, v, 1e-3, 0.9, 0.999,
+                                         1e-8, 0.01, 2), iters=10)
+        bw = n * (16 + (8 if st_dt == torch.float32 else 4) * 2) / (us / 1e6) / 1e12
+        print(f"  adamw_flat {label} states: {us:8.1f} us  ({bw:5.2f} TB/s)")
+
+
+if __name__ == "__main__" and "--adam" in __import__("sys").argv:
+    bench_adam()
