@@ -208,6 +208,7 @@ def main():
     optimizer = get_megatron_optimizer(
         OptimizerConfig(lr=1e-4, weight_decay=0.1, clip_grad=1.0, bf16=bf16,
                         use_distributed_optimizer=(dp > 1),
+                        overlap_param_gather=(dp > 1),
                         use_precision_aware_optimizer=args.precision_aware_optimizer,
                         exp_avg_dtype="bf16" if args.precision_aware_optimizer else "fp32",
                         exp_avg_sq_dtype="bf16" if args.precision_aware_optimizer else "fp32"),
@@ -245,6 +246,8 @@ def main():
     def one_step(it):
         if tracer is not None:
             tracer.iteration_begin(it)
+        if hasattr(optimizer, "finish_param_sync"):
+            optimizer.finish_param_sync()   # overlapped ZeRO-1 gather
         model.zero_grad_buffer()
         optimizer.zero_grad()
         out = fb(forward_step_func=forward_step, data_iterator=None,
