@@ -61,6 +61,19 @@ class _SplitQKV(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dq, dk, dv):
         sq, b, ng, rep, hn = ctx.dims
+        # zero-copy path: the flash backward (attn_bwd_into) already wrote
+        # the three grads into one fused buffer with exactly this layout
+        if (rep == 1 and dq._base is not None and dq._base is dk._base
+                and dk._base is dv._base):
+            base = dq._base
+            if (tuple(base.shape) == (sq, b, ng, 3 * hn)
+                    and dq.storage_offset() == base.storage_offset()
+                    and dk.storage_offset() == base.storage_offset() + hn
+                    and dv.storage_offset() == base.storage_offset() + 2 * hn
+                    and dq.stride() == dk.stride() == dv.stride()
+                    and dq.stride() == base[..., :hn].stride()
+                    and base.is_contiguous()):
+                return base.view(sq, b, -1), None, None, None
         dm = torch.empty((sq, b, ng, (rep + 2) * hn), dtype=dq.dtype,
                          device=dq.device)
         dm[..., :rep * hn].copy_(dq.reshape(sq, b, ng, rep * hn))

@@ -134,7 +134,28 @@ class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = _ops.get_ops().attn_bwd(
+        ops = _ops.get_ops()
+        nh, hn = q.shape[2], q.shape[3]
+        ng = k.shape[2]
+        # rep==1: write dQ/dK/dV straight into one fused [s, b, g, 3*hn]
+        # buffer laid out like the QKV GEMM output — when the grads flow
+        # back to _SplitQKV unchanged (no rope in between), its backward
+        # recognizes the shared base and skips three slice copies
+        import os
+        if (nh == ng and q.shape[0] == k.shape[0]
+                and hasattr(ops, "attn_bwd_into")
+                and "qkv_fuse" not in os.environ.get(
+                    "MEGATRONAPP_DISABLE_FUSED", "")):
+            sq, b = q.shape[0], q.shape[1]
+            dm = torch.empty(sq, b, ng, 3 * hn, dtype=q.dtype,
+                             device=q.device)
+            dq_v = dm[..., :hn]
+            dk_v = dm[..., hn:2 * hn]
+            dv_v = dm[..., 2 * hn:]
+            if ops.attn_bwd_into(do.contiguous(), q, k, v, o, lse,
+                                 ctx.scale, ctx.causal, dq_v, dk_v, dv_v):
+                return dq_v, dk_v, dv_v, None, None, None
+        dq, dk, dv = ops.attn_bwd(
             do.contiguous(), q, k, v, o, lse, ctx.scale, ctx.causal)
         return dq, dk, dv, None, None, None
 

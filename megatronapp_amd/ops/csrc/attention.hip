@@ -1040,7 +1040,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
     const float* __restrict__ lse, const float* __restrict__ drow,
     unsigned short* __restrict__ dq, int sq, int sk, int b, int nh, int ng,
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
-    long vS, long vB, long vH) {
+    long vS, long vB, long vH, long oS, long oB, long oH) {
   static_assert(D == 128 || D == 64, "dq2: head dim 64/128");
   const int qtile = blockIdx.y;   // grid (bh, qtile): K/V per-XCD L2
   const int bh = blockIdx.x;
@@ -1243,7 +1243,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dq2_kernel(
       const int row = idx / D;
       const int col = idx % D;
       const int qq = q0 + row;
-      *(bf16x8*)(dq + ((long)qq * b * nh + (long)bi * nh + h) * D + col) =
+      *(bf16x8*)(dq + (long)qq * oS + (long)bi * oB + (long)h * oH + col) =
           *(const bf16x8*)(ows + (long)row * OSTRIDE + col);
     }
   }
@@ -1266,7 +1266,8 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dv2_kernel(
     const unsigned short* __restrict__ dout,
     const float* __restrict__ lse,
     unsigned short* __restrict__ dv, int sq, int sk, int b, int nh, int ng,
-    float scale, long qS, long qB, long qH, long kS, long kB, long kH) {
+    float scale, long qS, long qB, long qH, long kS, long kB, long kH,
+    long oS, long oB, long oH) {
   static_assert(D == 128 || D == 64, "dv2: head dim 64/128");
   const int kvtile = blockIdx.y;  // grid (bh, kvtile): Q/dO per-XCD L2
   const int bh = blockIdx.x;
@@ -1433,7 +1434,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dv2_kernel(
     for (int r = 0; r < 16; ++r) {
       const int kvrow = kv0w + (r & 3) + 8 * (r >> 2) + 4 * lh;
       if (kvrow < sk)
-        dv[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + 32 * dsub +
+        dv[(long)kvrow * oS + (long)bi * oB + (long)hkv * oH + 32 * dsub +
            ln] = f2bf(dvacc[dsub][r]);
     }
 }
@@ -1447,7 +1448,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dk2_kernel(
     const float* __restrict__ lse, const float* __restrict__ drow,
     unsigned short* __restrict__ dk, int sq, int sk, int b, int nh, int ng,
     float scale, long qS, long qB, long qH, long kS, long kB, long kH,
-    long vS, long vB, long vH) {
+    long vS, long vB, long vH, long oS, long oB, long oH) {
   static_assert(D == 128 || D == 64, "dk2: head dim 64/128");
   const int kvtile = blockIdx.y;  // grid (bh, kvtile): Q/dO per-XCD L2
   const int bh = blockIdx.x;
@@ -1637,7 +1638,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void attn_bwd_dk2_kernel(
     for (int r = 0; r < 16; ++r) {
       const int kvrow = kv0w + (r & 3) + 8 * (r >> 2) + 4 * lh;
       if (kvrow < sk)
-        dk[((long)kvrow * b * ng + (long)bi * ng + hkv) * D + 32 * dsub +
+        dk[(long)kvrow * oS + (long)bi * oB + (long)hkv * oH + 32 * dsub +
            ln] = f2bf(dkacc[dsub][r]);
     }
 }
@@ -1654,7 +1655,8 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                       float* drow, void* dq, void* dk, void* dv, int sq,
                       int sk, int b, int nh, int ng, int d, float scale,
                       bool causal, const long* qstr, const long* kstr,
-                      const long* vstr, hipStream_t stream) {
+                      const long* vstr, const long* dqs, const long* dks,
+                      const long* dvs, hipStream_t stream) {
   if (sq % FQBLK2 != 0 || sk % FQBLK2 != 0 || (d != 128 && d != 64))
     throw std::runtime_error(
         "attn_bwd2: sq/sk must be multiples of 256, d 64/128");
@@ -1685,7 +1687,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        (const unsigned short*)dout, lse, drow,                \
                        (unsigned short*)dq, sq, sk, b, nh, ng, scale,         \
                        qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
-                       vstr[0], vstr[1], vstr[2]);                            \
+                       vstr[0], vstr[1], vstr[2], dqs[0], dqs[1], dqs[2]);    \
     HIP_CHECK_LAUNCH();                                                       \
     hipLaunchKernelGGL((attn_bwd_dv2_kernel<DD, CC>),                        \
                        dim3(b * ng, sk / FQBLK2), dim3(FWD_BLOCK), lds_dv,    \
@@ -1694,7 +1696,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        (const unsigned short*)dout, lse,                      \
                        (unsigned short*)dv, sq, sk, b, nh, ng, scale,         \
                        qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],           \
-                       kstr[2]);                                              \
+                       kstr[2], dvs[0], dvs[1], dvs[2]);                      \
     HIP_CHECK_LAUNCH();                                                       \
     hipLaunchKernelGGL((attn_bwd_dk2_kernel<DD, CC>),                        \
                        dim3(b * ng, sk / FQBLK2), dim3(FWD_BLOCK), lds_dk,    \
@@ -1703,7 +1705,7 @@ void launch_attn_bwd2(const void* dout, const void* q, const void* k,
                        (const unsigned short*)dout, lse, drow,                \
                        (unsigned short*)dk, sq, sk, b, nh, ng, scale,         \
                        qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
-                       vstr[0], vstr[1], vstr[2]);                            \
+                       vstr[0], vstr[1], vstr[2], dks[0], dks[1], dks[2]);    \
     HIP_CHECK_LAUNCH();                                                       \
   } while (0)
   if (d == 128) {

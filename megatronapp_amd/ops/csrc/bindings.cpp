@@ -85,8 +85,9 @@ void launch_attn_fwd2(const void*, const void*, const void*, void*, float*,
                       const long*, const long*, const long*, hipStream_t);
 void launch_attn_bwd2(const void*, const void*, const void*, const void*,
                       const void*, const float*, float*, void*, void*, void*,
-                      int, int, int, int, int, int, float, bool,
-                      const long*, const long*, const long*, hipStream_t);
+                      int, int, int, int, int, int, float, bool, const long*,
+                      const long*, const long*, const long*, const long*,
+                      const long*, hipStream_t);
 
 // The round-2 kernels take (seq, batch, head) element strides so the
 // strided QKV-split views feed them without .contiguous() copies; d
@@ -608,9 +609,11 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
             d = (int)q.size(3);
   const int sk = (int)k.size(0), ng = (int)k.size(2);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  // q/k/v may be gapped views of the fused QKV GEMM output; the
+  // gradients are plain contiguous [s, b, h, d] tensors
+  auto dq = torch::empty({(long)sq, b, nh, d}, q.options());
+  auto dk = torch::empty({(long)sk, b, ng, d}, q.options());
+  auto dv = torch::empty({(long)sk, b, ng, d}, q.options());
   auto drow = torch::empty({(long)sq * b * nh},
                            q.options().dtype(torch::kFloat32));
   long qs[3], ks[3], vs[3];
@@ -618,11 +621,13 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   if ((d == 128 || d == 64) && sq % 256 == 0 && sk % 256 == 0 &&
       attn_strides(q, qs) &&
       attn_strides(k, ks) && attn_strides(v, vs)) {
+    const long dqs[3] = {(long)b * nh * d, (long)nh * d, (long)d};
+    const long dks[3] = {(long)b * ng * d, (long)ng * d, (long)d};
     launch_attn_bwd2(doc.data_ptr(), q.data_ptr(), k.data_ptr(),
                      v.data_ptr(), o.data_ptr(), lse.data_ptr<float>(),
                      drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                      dv.data_ptr(), sq, sk, b, nh, ng, d, (float)scale,
-                     causal, qs, ks, vs, cur_stream());
+                     causal, qs, ks, vs, dqs, dks, dks, cur_stream());
   } else {
     auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
     launch_attn_bwd(doc.data_ptr(), qc.data_ptr(), kc.data_ptr(),
@@ -632,6 +637,50 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                     causal, cur_stream());
   }
   return {dq, dk, dv};
+}
+
+// attn_bwd writing dQ/dK/dV INTO caller-provided (possibly strided)
+// buffers — the flash backward can then fill the fused [s, b, g,
+// (rep+2)*hn] QKV-grad buffer directly and _SplitQKV.backward skips its
+// three slice copies (~1% of a GPT step).  Requires the round-2 kernel
+// path; returns false when the shape would fall back to v1 (caller uses
+// attn_bwd + copies instead).  Out tensors must have d contiguous.
+bool attn_bwd_into(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                   torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                   double scale, bool causal, torch::Tensor dq,
+                   torch::Tensor dk, torch::Tensor dv) {
+  check_bf16_any(dout, "dout");
+  check_bf16_any(q, "q");
+  check_bf16_any(k, "k");
+  check_bf16_any(v, "v");
+  check_bf16_any(dq, "dq");
+  check_bf16_any(dk, "dk");
+  check_bf16_any(dv, "dv");
+  const int sq = (int)q.size(0), b = (int)q.size(1), nh = (int)q.size(2),
+            d = (int)q.size(3);
+  const int sk = (int)k.size(0), ng = (int)k.size(2);
+  long qs[3], ks[3], vs[3], dqs[3], dks[3], dvs[3];
+  if (!((d == 128 || d == 64) && sq % 256 == 0 && sk % 256 == 0 &&
+        attn_strides(q, qs) && attn_strides(k, ks) && attn_strides(v, vs) &&
+        attn_strides(dq, dqs) && attn_strides(dk, dks) &&
+        attn_strides(dv, dvs)))
+    return false;
+  // dq is written with 16-byte vector stores: its strides and base
+  // pointer must be 8-element aligned
+  if (dqs[0] % 8 || dqs[1] % 8 || dqs[2] % 8 ||
+      (reinterpret_cast<uintptr_t>(dq.data_ptr()) & 15))
+    return false;
+  TORCH_CHECK(dq.sizes() == q.sizes() && dk.sizes() == k.sizes() &&
+              dv.sizes() == v.sizes(), "attn_bwd_into: out shape mismatch");
+  auto drow = torch::empty({(long)sq * b * nh},
+                           q.options().dtype(torch::kFloat32));
+  auto doc = dout.contiguous();
+  launch_attn_bwd2(doc.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                   o.data_ptr(), lse.data_ptr<float>(),
+                   drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                   dv.data_ptr(), sq, sk, b, nh, ng, d, (float)scale, causal,
+                   qs, ks, vs, dqs, dks, dvs, cur_stream());
+  return true;
 }
 
 // always-v1 backward, kept for A/B benchmarking
@@ -809,6 +858,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("moe_combine_fwd", &moe_combine_fwd);
   mod.def("moe_combine_bwd", &moe_combine_bwd);
   mod.def("attn_bwd_v1", &attn_bwd_v1);
+  mod.def("attn_bwd_into", &attn_bwd_into);
   mod.def("ce_rowmax", &ce_rowmax);
   mod.def("gemm_nt", &gemm_nt);
   mod.def("gemm_nn", &gemm_nn);
