@@ -36,14 +36,18 @@ from .optimizer_config import OptimizerConfig
 
 def _adam_step_flat(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
                     v: torch.Tensor, lr: float, beta1: float, beta2: float,
-                    eps: float, weight_decay: float, step: int):
+                    eps: float, weight_decay: float, step: int,
+                    p16: torch.Tensor = None) -> bool:
     """One Adam(W) step over a flat fp32 master/grad span; HIP kernel on
     GPU.  States may be bf16 (precision-aware optimizer) — math is fp32
-    either way, bf16 states round on store."""
+    either way, bf16 states round on store.  p16 (optional): the bf16
+    model-param shard, written by the kernel in the same pass (replaces
+    the separate master->param cast copy).  Returns True iff the fused
+    kernel ran (and consumed p16)."""
     if master.is_cuda and _ops.have_ops() and hasattr(_ops.get_ops(), "adamw_flat"):
         _ops.get_ops().adamw_flat(master, grad, m, v, lr, beta1, beta2, eps,
-                                  weight_decay, step)
-        return
+                                  weight_decay, step, p16)
+        return True
     bias_correction1 = 1 - beta1 ** step
     bias_correction2 = 1 - beta2 ** step
     if weight_decay != 0:
@@ -58,6 +62,7 @@ def _adam_step_flat(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
         m.copy_(mf)
     if vf is not v:
         v.copy_(vf)
+    return False
 
 
 class ParamGroup:
@@ -296,10 +301,20 @@ class DistributedOptimizer:
                 grad = self._shard_grad(i)
                 master, m, v = self.shard_master[i], self.shard_m[i], self.shard_v[i]
                 no_wd = self.no_wd_ranges[i]
+                lo, hi = self.shard_bounds[i]
+                # bf16 param shard written by the adam kernel itself
+                p16 = None
+                if (buf.param_data is not None and master.is_cuda
+                        and buf.param_data.dtype == torch.bfloat16):
+                    p16 = buf.param_data[lo:hi]
+                wrote_p16 = False
                 if wd == 0 or not no_wd:
-                    _adam_step_flat(master, grad, m, v, lr,
-                                    self.config.adam_beta1, self.config.adam_beta2,
-                                    self.config.adam_eps, wd, self.step_count)
+                    fused = _adam_step_flat(master, grad, m, v, lr,
+                                            self.config.adam_beta1,
+                                            self.config.adam_beta2,
+                                            self.config.adam_eps, wd,
+                                            self.step_count, p16=p16)
+                    wrote_p16 = fused and p16 is not None
                 elif (master.is_cuda and _ops.have_ops()
                       and hasattr(_ops.get_ops(), "adamw_flat_ranged")
                       and self._ranges_aligned(i)):
@@ -307,7 +322,8 @@ class DistributedOptimizer:
                     _ops.get_ops().adamw_flat_ranged(
                         master, grad, m, v, nw_s, nw_e, lr,
                         self.config.adam_beta1, self.config.adam_beta2,
-                        self.config.adam_eps, wd, self.step_count)
+                        self.config.adam_eps, wd, self.step_count, p16)
+                    wrote_p16 = p16 is not None
                 else:
                     # two-pass: run with wd over the whole shard is wrong for
                     # no-wd params, so stitch: wd pass on full shard minus
@@ -330,9 +346,11 @@ class DistributedOptimizer:
                                             self.step_count)
                         cursor = max(cursor, e)
 
-                # cast master back into model params
-                lo, hi = self.shard_bounds[i]
-                if buf.param_data is not None:
+                # cast master back into model params (skipped when the
+                # adam kernel already wrote the bf16 shard)
+                if wrote_p16:
+                    pass
+                elif buf.param_data is not None:
                     buf.param_data[lo:hi].copy_(master)
                 else:
                     for p, (s, e) in buf.param_index_map.items():

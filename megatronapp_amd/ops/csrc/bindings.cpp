@@ -42,8 +42,8 @@ void launch_softmax_causal_bwd(const void*, const void*, void*, long, int,
 void launch_softmax_bwd(const void*, const void*, void*, long, int, float,
                         hipStream_t);
 void launch_adamw_flat_ranged(float*, const float*, void*, void*,
-                              bool, const long*, const long*, int,
-                              long, float, float, float, float,
+                              bool, void*, const long*, const long*,
+                              int, long, float, float, float, float,
                               float, int, hipStream_t);
 void launch_ce_rowmax(const void*, float*, long, int, hipStream_t);
 void launch_ce_fwd(const void*, const float*, const int*, float*, float*,
@@ -54,8 +54,8 @@ void launch_selective_scan_fwd(const void*, const void*, const float*,
                                const void*, const void*, const float*, float*,
                                void*, int, int, int, int, hipStream_t);
 void launch_adamw_flat(float*, const float*, void*, void*, bool,
-                       long, float, float, float, float, float,
-                       int, hipStream_t);
+                       void*, long, float, float, float, float,
+                       float, int, hipStream_t);
 void wgrad_accum(torch::Tensor, torch::Tensor, torch::Tensor);
 bool wgrad_accum_bgrad(torch::Tensor, torch::Tensor, torch::Tensor,
                        torch::Tensor);
@@ -728,12 +728,23 @@ static bool adam_states_bf16(const torch::Tensor& m, const torch::Tensor& v) {
   return m.scalar_type() == torch::kBFloat16;
 }
 
+static void* adam_p16_ptr(const c10::optional<torch::Tensor>& p16,
+                          const torch::Tensor& p) {
+  if (!p16.has_value()) return nullptr;
+  TORCH_CHECK(p16->scalar_type() == torch::kBFloat16 &&
+              p16->is_contiguous() && p16->numel() == p.numel(),
+              "adam p16 must be a contiguous bf16 mirror of the shard");
+  return p16->data_ptr();
+}
+
 void adamw_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double beta1, double beta2,
-                double eps, double wd, long step) {
+                double eps, double wd, long step,
+                c10::optional<torch::Tensor> p16) {
   TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
   launch_adamw_flat(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr(),
-                    v.data_ptr(), adam_states_bf16(m, v), p.numel(),
+                    v.data_ptr(), adam_states_bf16(m, v),
+                    adam_p16_ptr(p16, p), p.numel(),
                     (float)lr, (float)beta1, (float)beta2, (float)eps,
                     (float)wd, (int)step, cur_stream());
 }
@@ -806,14 +817,16 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor rowmax,
 void adamw_flat_ranged(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        torch::Tensor v, torch::Tensor nw_s, torch::Tensor nw_e,
                        double lr, double beta1, double beta2, double eps,
-                       double wd, long step) {
+                       double wd, long step,
+                       c10::optional<torch::Tensor> p16) {
   TORCH_CHECK(p.is_cuda() && p.dtype() == torch::kFloat32);
   TORCH_CHECK(nw_s.is_cuda() && nw_s.dtype() == torch::kInt64 &&
               nw_s.is_contiguous());
   TORCH_CHECK(nw_e.sizes() == nw_s.sizes());
   launch_adamw_flat_ranged(
       p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr(), v.data_ptr(),
-      adam_states_bf16(m, v), nw_s.data_ptr<long>(), nw_e.data_ptr<long>(),
+      adam_states_bf16(m, v), adam_p16_ptr(p16, p), nw_s.data_ptr<long>(),
+      nw_e.data_ptr<long>(),
       (int)nw_s.numel(), p.numel(), (float)lr, (float)beta1, (float)beta2,
       (float)eps, (float)wd, (int)step,
       at::cuda::getCurrentCUDAStream());
@@ -847,7 +860,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("scaled_softmax_fwd", &scaled_softmax_fwd);
   mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
   mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
-  mod.def("adamw_flat", &adamw_flat);
+  mod.def("adamw_flat", &adamw_flat, pybind11::arg("p"),
+          pybind11::arg("g"), pybind11::arg("m"), pybind11::arg("v"),
+          pybind11::arg("lr"), pybind11::arg("beta1"),
+          pybind11::arg("beta2"), pybind11::arg("eps"),
+          pybind11::arg("wd"), pybind11::arg("step"),
+          pybind11::arg("p16") = pybind11::none());
   mod.def("selective_scan_fwd", &selective_scan_fwd);
   mod.def("attn_fwd_t", &attn_fwd_t);
   mod.def("attn_fwd2", &attn_fwd2);
@@ -867,7 +885,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("scaled_upper_triang_masked_softmax_bwd",
           &scaled_upper_triang_masked_softmax_bwd);
   mod.def("bias_add_residual", &bias_add_residual);
-  mod.def("adamw_flat_ranged", &adamw_flat_ranged);
+  mod.def("adamw_flat_ranged", &adamw_flat_ranged, pybind11::arg("p"),
+          pybind11::arg("g"), pybind11::arg("m"), pybind11::arg("v"),
+          pybind11::arg("nw_s"), pybind11::arg("nw_e"),
+          pybind11::arg("lr"), pybind11::arg("beta1"),
+          pybind11::arg("beta2"), pybind11::arg("eps"),
+          pybind11::arg("wd"), pybind11::arg("step"),
+          pybind11::arg("p16") = pybind11::none());
   mod.def("wgrad_accum", &wgrad_accum);
   mod.def("wgrad_accum_bgrad", &wgrad_accum_bgrad);
   mod.def("attn_fwd", &attn_fwd);

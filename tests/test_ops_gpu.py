@@ -180,8 +180,11 @@ def test_adamw_flat_matches_torch():
     p_ref = p.clone()
 
     lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-8, 0.01
+    p16 = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
     for step in range(1, 4):
-        _ops().adamw_flat(p, g, m, v, lr, b1, b2, eps, wd, step)
+        _ops().adamw_flat(p, g, m, v, lr, b1, b2, eps, wd, step, p16)
+    # the fused bf16 param mirror must equal the cast of the master
+    assert torch.equal(p16, p.bfloat16())
 
     mt = torch.zeros(n, device="cuda")
     vt = torch.zeros(n, device="cuda")
