@@ -105,9 +105,13 @@ def fp8_wgrad(g2d: torch.Tensor, x2d: torch.Tensor,
     (tests/test_families_gpu.py --fp8 twin).  Returns False when the
     path does not apply (caller falls back to the exact bf16 wgrad)."""
     from .. import ops as _ops
+    rows = g2d.shape[0]
     if not (g2d.is_cuda and g2d.dtype == torch.bfloat16
             and main_grad.dtype == torch.float32 and _ops.have_ops()
             and hasattr(_ops.get_ops(), "quantize_transpose_e4m3")
+            # _scaled_mm wants 16-aligned dims
+            and rows % 16 == 0 and g2d.shape[1] % 16 == 0
+            and x2d.shape[1] % 16 == 0
             and fp8_wgrad_enabled()):
         return False
     lt = _ops.get_ops()
