@@ -12,6 +12,7 @@
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 from torch import nn
@@ -141,7 +142,6 @@ class _FlashAttnFn(torch.autograd.Function):
         # buffer laid out like the QKV GEMM output — when the grads flow
         # back to _SplitQKV unchanged (no rope in between), its backward
         # recognizes the shared base and skips three slice copies
-        import os
         if (nh == ng and q.shape[0] == k.shape[0]
                 and hasattr(ops, "attn_bwd_into")
                 and "qkv_fuse" not in os.environ.get(

@@ -12,8 +12,12 @@ segmented loop off-GPU.
 
 from __future__ import annotations
 
+import os
+
 import torch
 from torch import nn
+
+from .... import ops as _ops
 
 
 def _grouped_mm_available(t: torch.Tensor) -> bool:
@@ -42,8 +46,6 @@ class _GroupedMMFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        from .... import ops as _ops
-
         x, w, offs = ctx.saved_tensors
         dy = dy.contiguous()
         dx = torch._grouped_mm(dy, w.transpose(1, 2), offs=offs)
@@ -138,10 +140,9 @@ class GroupedMLP(nn.Module):
             # host token counts enable the per-expert fp32 wgrad in
             # backward; the one-time .tolist() sync per layer is ~50 us
             # against the ~2 full expert-weight passes it removes
-            import os as _os
-            dis = _os.environ.get("MEGATRONAPP_DISABLE_FUSED", "")
+            dis = os.environ.get("MEGATRONAPP_DISABLE_FUSED", "")
             counts = None
-            if (torch.is_grad_enabled()
+            if (torch.is_grad_enabled() and _ops.have_ops()
                     and permuted_tokens.dtype == torch.bfloat16
                     and hasattr(self.weight1, "main_grad")
                     and not ("all" in dis or "moe_wgrad" in dis)):

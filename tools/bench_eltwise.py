@@ -84,8 +84,11 @@ def main():
         print(f"  separate {us_sep:7.1f} us   (fused unavailable)")
 
 
-if __name__ == "__main__" and "--adam" not in __import__("sys").argv:
-    main()
+if __name__ == "__main__":
+    if "--adam" in sys.argv:
+        bench_adam()
+    else:
+        main()
 
 
 def bench_adam():
@@ -102,6 +105,3 @@ def bench_adam():
         bw = n * (16 + (8 if st_dt == torch.float32 else 4) * 2) / (us / 1e6) / 1e12
         print(f"  adamw_flat {label} states: {us:8.1f} us  ({bw:5.2f} TB/s)")
 
-
-if __name__ == "__main__" and "--adam" in __import__("sys").argv:
-    bench_adam()
