@@ -84,13 +84,16 @@ __global__ void bias_gelu_bwd_kernel(const unsigned short* __restrict__ dy,
 __global__ void bias_swiglu_fwd_kernel(const unsigned short* __restrict__ x,
                                        const unsigned short* __restrict__ bias,
                                        unsigned short* __restrict__ y, long N,
-                                       int F) {
+                                       int F, int fshift) {
   long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
   const long total = N * (F / VEC);
   const long stride = (long)gridDim.x * BLOCK;
   for (; idx < total; idx += stride) {
-    const long row = idx / (F / VEC);
-    const int col = (int)(idx % (F / VEC)) * VEC;
+    // fshift >= 0: F/VEC is a power of two — shift/mask instead of the
+    // ~80-VALU-op i64 div+mod pair per 8 elements
+    const long row = fshift >= 0 ? idx >> fshift : idx / (F / VEC);
+    const int col = (int)(fshift >= 0 ? (idx & ((1L << fshift) - 1))
+                                      : idx % (F / VEC)) * VEC;
     const unsigned short* x1 = x + row * 2L * F + col;
     const unsigned short* x2 = x1 + F;
     short8v v1 = *(const short8v*)x1;
@@ -115,13 +118,16 @@ __global__ void bias_swiglu_bwd_kernel(const unsigned short* __restrict__ dy,
                                        const unsigned short* __restrict__ x,
                                        const unsigned short* __restrict__ bias,
                                        unsigned short* __restrict__ dx, long N,
-                                       int F) {
+                                       int F, int fshift) {
   long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
   const long total = N * (F / VEC);
   const long stride = (long)gridDim.x * BLOCK;
   for (; idx < total; idx += stride) {
-    const long row = idx / (F / VEC);
-    const int col = (int)(idx % (F / VEC)) * VEC;
+    // fshift >= 0: F/VEC is a power of two — shift/mask instead of the
+    // ~80-VALU-op i64 div+mod pair per 8 elements
+    const long row = fshift >= 0 ? idx >> fshift : idx / (F / VEC);
+    const int col = (int)(fshift >= 0 ? (idx & ((1L << fshift) - 1))
+                                      : idx % (F / VEC)) * VEC;
     const unsigned short* x1 = x + row * 2L * F + col;
     const unsigned short* x2 = x1 + F;
     short8v v1 = *(const short8v*)x1;
@@ -152,13 +158,16 @@ __global__ void bias_swiglu_bwd_kernel(const unsigned short* __restrict__ dy,
 __global__ void bias_geglu_fwd_kernel(const unsigned short* __restrict__ x,
                                       const unsigned short* __restrict__ bias,
                                       unsigned short* __restrict__ y, long N,
-                                      int F) {
+                                      int F, int fshift) {
   long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
   const long total = N * (F / VEC);
   const long stride = (long)gridDim.x * BLOCK;
   for (; idx < total; idx += stride) {
-    const long row = idx / (F / VEC);
-    const int col = (int)(idx % (F / VEC)) * VEC;
+    // fshift >= 0: F/VEC is a power of two — shift/mask instead of the
+    // ~80-VALU-op i64 div+mod pair per 8 elements
+    const long row = fshift >= 0 ? idx >> fshift : idx / (F / VEC);
+    const int col = (int)(fshift >= 0 ? (idx & ((1L << fshift) - 1))
+                                      : idx % (F / VEC)) * VEC;
     const unsigned short* x1 = x + row * 2L * F + col;
     const unsigned short* x2 = x1 + F;
     short8v v1 = *(const short8v*)x1;
@@ -182,13 +191,16 @@ __global__ void bias_geglu_bwd_kernel(const unsigned short* __restrict__ dy,
                                       const unsigned short* __restrict__ x,
                                       const unsigned short* __restrict__ bias,
                                       unsigned short* __restrict__ dx, long N,
-                                      int F) {
+                                      int F, int fshift) {
   long idx = (long)blockIdx.x * BLOCK + threadIdx.x;
   const long total = N * (F / VEC);
   const long stride = (long)gridDim.x * BLOCK;
   for (; idx < total; idx += stride) {
-    const long row = idx / (F / VEC);
-    const int col = (int)(idx % (F / VEC)) * VEC;
+    // fshift >= 0: F/VEC is a power of two — shift/mask instead of the
+    // ~80-VALU-op i64 div+mod pair per 8 elements
+    const long row = fshift >= 0 ? idx >> fshift : idx / (F / VEC);
+    const int col = (int)(fshift >= 0 ? (idx & ((1L << fshift) - 1))
+                                      : idx % (F / VEC)) * VEC;
     const unsigned short* x1 = x + row * 2L * F + col;
     const unsigned short* x2 = x1 + F;
     short8v v1 = *(const short8v*)x1;
@@ -219,6 +231,12 @@ __global__ void bias_geglu_bwd_kernel(const unsigned short* __restrict__ dy,
 }
 
 // ------------------------------------------------------------------ launchers
+// log2(F/VEC) when it is a power of two, else -1 (runtime fast path)
+static int glu_fshift(int F) {
+  const int fv = F / VEC;
+  return (fv > 0 && (fv & (fv - 1)) == 0) ? __builtin_ctz(fv) : -1;
+}
+
 static int ew_grid(long work_items) {
   long blocks = (work_items + BLOCK - 1) / BLOCK;
   return (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
@@ -279,7 +297,7 @@ void launch_bias_swiglu_fwd(const void* x, const void* bias, void* y, long N,
   if (F % VEC != 0) throw std::runtime_error("F must be divisible by 8");
   hipLaunchKernelGGL(bias_swiglu_fwd_kernel, dim3(ew_grid(N * (F / VEC))),
                      dim3(BLOCK), 0, s, (const unsigned short*)x,
-                     (const unsigned short*)bias, (unsigned short*)y, N, F);
+                     (const unsigned short*)bias, (unsigned short*)y, N, F, glu_fshift(F));
   HIP_CHECK_LAUNCH();
 }
 
@@ -289,7 +307,7 @@ void launch_bias_swiglu_bwd(const void* dy, const void* x, const void* bias,
   hipLaunchKernelGGL(bias_swiglu_bwd_kernel, dim3(ew_grid(N * (F / VEC))),
                      dim3(BLOCK), 0, s, (const unsigned short*)dy,
                      (const unsigned short*)x, (const unsigned short*)bias,
-                     (unsigned short*)dx, N, F);
+                     (unsigned short*)dx, N, F, glu_fshift(F));
   HIP_CHECK_LAUNCH();
 }
 
@@ -305,7 +323,7 @@ void launch_bias_geglu_fwd(const void* x, const void* bias, void* y, long N,
   const long total = N * (F / VEC);
   hipLaunchKernelGGL(bias_geglu_fwd_kernel, dim3(ew_grid(total)),
                      dim3(BLOCK), 0, stream, (const unsigned short*)x,
-                     (const unsigned short*)bias, (unsigned short*)y, N, F);
+                     (const unsigned short*)bias, (unsigned short*)y, N, F, glu_fshift(F));
   HIP_CHECK_LAUNCH();
 }
 
@@ -315,7 +333,7 @@ void launch_bias_geglu_bwd(const void* dy, const void* x, const void* bias,
   hipLaunchKernelGGL(bias_geglu_bwd_kernel, dim3(ew_grid(total)),
                      dim3(BLOCK), 0, stream, (const unsigned short*)dy,
                      (const unsigned short*)x, (const unsigned short*)bias,
-                     (unsigned short*)dx, N, F);
+                     (unsigned short*)dx, N, F, glu_fshift(F));
   HIP_CHECK_LAUNCH();
 }
 
