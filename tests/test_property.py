@@ -76,3 +76,30 @@ def test_rampup_microbatch_calculator_invariants(mbs, dp, gbs_mult,
         prev = gbs
     calc.update(ramup, consistency_check=False)
     assert calc.get_current_global_batch_size() == final
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    weights=st.lists(st.floats(min_value=0.01, max_value=1.0),
+                     min_size=1, max_size=6),
+    size=st.integers(min_value=1, max_value=500),
+)
+def test_blending_indices_invariants(weights, size):
+    """Greedy blending: per-dataset sample indices are 0..k-1 dense in
+    order, every position assigned, and each realized share stays within
+    one sample of its normalized weight."""
+    from megatronapp_amd.core.datasets.build_helpers import load_helpers
+
+    w = np.array(weights, dtype=np.float64)
+    w = w / w.sum()
+    di, dsi = load_helpers().build_blending_indices(w, size)
+    di, dsi = np.asarray(di), np.asarray(dsi)
+    assert di.shape == (size,) and dsi.shape == (size,)
+    counts = np.zeros(len(w), dtype=np.int64)
+    for d, s in zip(di, dsi):
+        assert s == counts[d], "per-dataset indices must be dense in order"
+        counts[d] += 1
+    assert counts.sum() == size
+    # realized share within 1 sample of target at the end
+    for d in range(len(w)):
+        assert abs(counts[d] - w[d] * size) <= len(w), (counts, w * size)
